@@ -1,0 +1,114 @@
+"""Checkpointer: DCP sharded save/load + consolidated HF-safetensors export.
+
+Reference behavior: nemo_automodel/components/checkpoint/checkpointing.py:479
+(Checkpointer.save_model/load_model: torch.distributed.checkpoint sharded
+save, HF-safetensors consolidation modes, retention policy lifecycle.py).
+"""
+
+from __future__ import annotations
+
+import os
+import shutil
+
+import torch
+import torch.distributed as dist
+
+
+def _sd_options(full: bool = False):
+    from torch.distributed.checkpoint.state_dict import StateDictOptions
+
+    return StateDictOptions(full_state_dict=full, cpu_offload=full)
+
+
+class Checkpointer:
+    def __init__(
+        self,
+        checkpoint_dir: str = "checkpoints",
+        model_save_format: str = "safetensors",
+        save_consolidated: bool = False,
+        keep_last_n: int | None = None,
+    ):
+        self.checkpoint_dir = checkpoint_dir
+        self.model_save_format = model_save_format
+        self.save_consolidated = save_consolidated
+        self.keep_last_n = keep_last_n
+
+    # ---------------------------------------------------------------- save
+    def save(self, path: str, model=None, optimizer=None, extra_state: dict | None = None,
+             rank: int = 0) -> None:
+        import torch.distributed.checkpoint as dcp
+        from torch.distributed.checkpoint.state_dict import (
+            get_model_state_dict,
+            get_optimizer_state_dict,
+        )
+
+        os.makedirs(path, exist_ok=True)
+        state: dict = {}
+        if model is not None:
+            state["model"] = get_model_state_dict(model)
+        if optimizer is not None:
+            state["optimizer"] = get_optimizer_state_dict(model, optimizer)
+        if state:
+            dcp.save(state, checkpoint_id=os.path.join(path, "dcp"))
+        if extra_state and rank == 0:
+            torch.save(extra_state, os.path.join(path, "aux_state.pt"))
+        if self.save_consolidated and model is not None:
+            self.export_hf_safetensors(model, os.path.join(path, "hf"), rank=rank)
+        if rank == 0:
+            self._apply_retention()
+        if dist.is_initialized():
+            dist.barrier()
+
+    def load(self, path: str, model=None, optimizer=None, rank: int = 0) -> dict:
+        import torch.distributed.checkpoint as dcp
+        from torch.distributed.checkpoint.state_dict import (
+            get_model_state_dict,
+            get_optimizer_state_dict,
+            set_model_state_dict,
+            set_optimizer_state_dict,
+        )
+
+        state: dict = {}
+        if model is not None:
+            state["model"] = get_model_state_dict(model)
+        if optimizer is not None:
+            state["optimizer"] = get_optimizer_state_dict(model, optimizer)
+        if state:
+            dcp.load(state, checkpoint_id=os.path.join(path, "dcp"))
+            if model is not None:
+                set_model_state_dict(model, state["model"])
+            if optimizer is not None:
+                set_optimizer_state_dict(model, optimizer, state["optimizer"])
+        aux_path = os.path.join(path, "aux_state.pt")
+        if os.path.exists(aux_path):
+            return torch.load(aux_path, weights_only=False)
+        return {}
+
+    # ------------------------------------------------- consolidated HF export
+    def export_hf_safetensors(self, model, out_dir: str, rank: int = 0) -> None:
+        """Gather full state dict and write HF-layout safetensors on rank 0."""
+        from torch.distributed.checkpoint.state_dict import get_model_state_dict
+
+        full_sd = get_model_state_dict(model, options=_sd_options(full=True))
+        if rank != 0:
+            return
+        from safetensors.torch import save_file
+
+        os.makedirs(out_dir, exist_ok=True)
+        adapter = getattr(model, "state_dict_adapter", None)
+        if adapter is not None:
+            full_sd = adapter.to_hf(full_sd)
+        full_sd = {k: v.contiguous() for k, v in full_sd.items() if isinstance(v, torch.Tensor)}
+        save_file(full_sd, os.path.join(out_dir, "model.safetensors"),
+                  metadata={"format": "pt"})
+
+    # ------------------------------------------------------------- retention
+    def _apply_retention(self) -> None:
+        if not self.keep_last_n or not os.path.isdir(self.checkpoint_dir):
+            return
+        steps = sorted(
+            (d for d in os.listdir(self.checkpoint_dir) if d.startswith("step_")),
+            key=lambda d: int(d.split("_")[1]),
+        )
+        for d in steps[: -self.keep_last_n]:
+            shutil.rmtree(os.path.join(self.checkpoint_dir, d), ignore_errors=True)

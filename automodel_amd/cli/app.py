@@ -1,0 +1,53 @@
+"""``automodel`` CLI: YAML in, recipe out.
+
+Reference behavior: nemo_automodel/cli/app.py:95-159 (parse YAML, resolve the
+``recipe:`` target, dispatch to a launcher). Usage:
+
+    automodel <cfg.yaml> [--nproc-per-node N] [--a.b.c=v ...]
+"""
+
+from __future__ import annotations
+
+import sys
+
+from automodel_amd.config.loader import (
+    apply_overrides,
+    load_yaml_config,
+    parse_cli_overrides,
+    resolve_target,
+)
+
+RECIPE_ALIASES = {
+    "llm_finetune": "automodel_amd.recipes.llm.train_ft.TrainFinetuneRecipeForNextTokenPrediction",
+    "llm_pretrain": "automodel_amd.recipes.llm.train_ft.TrainFinetuneRecipeForNextTokenPrediction",
+    "llm_benchmark": "automodel_amd.recipes.llm.benchmark.BenchmarkingRecipeForNextTokenPrediction",
+    "vlm_finetune": "automodel_amd.recipes.vlm.finetune.FinetuneRecipeForVLM",
+}
+
+
+def main(argv: list[str] | None = None) -> None:
+    argv = argv if argv is not None else sys.argv[1:]
+    if not argv or argv[0] in ("-h", "--help"):
+        print(__doc__)
+        return
+    cfg_path = argv[0]
+    rest = argv[1:]
+    nproc = 1
+    if "--nproc-per-node" in rest:
+        i = rest.index("--nproc-per-node")
+        nproc = int(rest[i + 1])
+        rest = rest[:i] + rest[i + 2:]
+    overrides = parse_cli_overrides(rest)
+
+    cfg = load_yaml_config(cfg_path)
+    apply_overrides(cfg, overrides)
+    recipe_name = cfg.get("recipe", "llm_finetune")
+    target = RECIPE_ALIASES.get(recipe_name, recipe_name)
+
+    from automodel_amd.launcher.interactive import InteractiveLauncher
+
+    InteractiveLauncher(nproc_per_node=nproc).launch(cfg_path, target, rest)
+
+
+if __name__ == "__main__":
+    main()

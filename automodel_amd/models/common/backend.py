@@ -1,0 +1,49 @@
+"""Per-module kernel backend selection.
+
+Reference behavior: nemo_automodel/components/models/common/utils.py:282-370
+(BackendConfig selects attn/linear/rms_norm/rope/experts implementations per
+model). On MI355X the choices are: hand-written HIP/CDNA4 kernels ("hip"),
+plain torch eager ("torch"), or torch SDPA for attention ("sdpa").
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+
+
+@dataclass
+class BackendConfig:
+    # attention: "hip" = in-tree CDNA4 flash kernel, "sdpa" = torch SDPA,
+    # "eager" = explicit matmul+softmax reference.
+    attn: str = "hip"
+    rms_norm: str = "hip"
+    rope: str = "hip"
+    # cross-entropy: "hip_fused" = fused lm_head-GEMM+CE without materializing
+    # logits (cut-cross-entropy equivalent), "chunked" = chunked logits CE,
+    # "torch" = plain CE.
+    loss: str = "hip_fused"
+    linear: str = "torch"          # plain GEMMs ride hipBLASLt via torch.linear
+    experts: str = "hip_grouped"   # MoE expert compute
+    dispatcher: str = "rccl_a2a"   # MoE token dispatch
+
+    def for_cpu(self) -> "BackendConfig":
+        """CPU-safe variant (unit tests run without a GPU)."""
+        return BackendConfig(
+            attn="sdpa",
+            rms_norm="torch",
+            rope="torch",
+            loss="chunked",
+            linear="torch",
+            experts="torch",
+            dispatcher="torch",
+        )
+
+    @classmethod
+    def resolve(cls, cfg: "BackendConfig | dict | None", device_type: str) -> "BackendConfig":
+        if cfg is None:
+            cfg = cls()
+        elif isinstance(cfg, dict):
+            cfg = cls(**cfg)
+        if device_type != "cuda":
+            return cfg.for_cpu()
+        return cfg

@@ -1,0 +1,242 @@
+"""Llama-family causal LM, MI355X-native.
+
+Custom DTensor-friendly implementation in the spirit of the reference's
+in-tree model (reference: nemo_automodel/components/models/llama/model.py),
+with kernels selected by BackendConfig: HIP flash attention, HIP RMSNorm,
+fused HIP RoPE, and (via the recipe's loss) fused linear cross-entropy.
+Parameter names match the HF checkpoint layout exactly
+(model.layers.N.self_attn.q_proj.weight, ...) so the state-dict adapter is
+the identity for this family.
+
+Also covers Qwen2-style variants (attention bias) via LlamaConfig flags.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any
+
+import torch
+import torch.nn as nn
+
+from automodel_amd.models.common.backend import BackendConfig
+from automodel_amd.ops.attention import flash_attention
+from automodel_amd.ops.rms_norm import RMSNorm
+from automodel_amd.ops.rope import apply_rope, build_rope_cache
+from automodel_amd.ops.swiglu import swiglu
+
+
+@dataclass
+class LlamaConfig:
+    vocab_size: int = 128256
+    hidden_size: int = 4096
+    intermediate_size: int = 14336
+    num_hidden_layers: int = 32
+    num_attention_heads: int = 32
+    num_key_value_heads: int = 8
+    head_dim: int | None = None
+    max_position_embeddings: int = 8192
+    rms_norm_eps: float = 1e-5
+    rope_theta: float = 500000.0
+    rope_scaling: dict | None = None
+    tie_word_embeddings: bool = False
+    attention_bias: bool = False
+    mlp_bias: bool = False
+    initializer_range: float = 0.02
+
+    def __post_init__(self):
+        if self.head_dim is None:
+            self.head_dim = self.hidden_size // self.num_attention_heads
+
+    @classmethod
+    def from_hf_config(cls, hf: Any) -> "LlamaConfig":
+        """Build from an HF config object or dict (llama / qwen2 / mistral)."""
+        if hasattr(hf, "to_dict"):
+            hf = hf.to_dict()
+        get = hf.get
+        return cls(
+            vocab_size=get("vocab_size", 32000),
+            hidden_size=get("hidden_size", 4096),
+            intermediate_size=get("intermediate_size", 11008),
+            num_hidden_layers=get("num_hidden_layers", 32),
+            num_attention_heads=get("num_attention_heads", 32),
+            num_key_value_heads=get("num_key_value_heads", get("num_attention_heads", 32)),
+            head_dim=get("head_dim", None),
+            max_position_embeddings=get("max_position_embeddings", 8192),
+            rms_norm_eps=get("rms_norm_eps", 1e-5),
+            rope_theta=get("rope_theta", 10000.0),
+            rope_scaling=get("rope_scaling", None),
+            tie_word_embeddings=get("tie_word_embeddings", False),
+            attention_bias=get("attention_bias", get("qkv_bias", False)),
+            mlp_bias=get("mlp_bias", False),
+            initializer_range=get("initializer_range", 0.02),
+        )
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.backend = backend
+        H, D = cfg.num_attention_heads, cfg.head_dim
+        Hk = cfg.num_key_value_heads
+        self.num_heads, self.num_kv_heads, self.head_dim = H, Hk, D
+        bias = cfg.attention_bias
+        self.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=bias)
+        self.k_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
+        self.v_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
+        self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        B, S, _ = x.shape
+        q = self.q_proj(x).view(B, S, -1, self.head_dim)
+        k = self.k_proj(x).view(B, S, -1, self.head_dim)
+        v = self.v_proj(x).view(B, S, -1, self.head_dim)
+        q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
+        o = flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+        return self.o_proj(o.reshape(B, S, -1))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+        super().__init__()
+        self.backend = backend
+        bias = cfg.mlp_bias
+        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg, backend)
+        self.mlp = LlamaMLP(cfg, backend)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class LlamaModel(nn.Module):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        self.layers = nn.ModuleList(
+            LlamaDecoderLayer(cfg, backend) for _ in range(cfg.num_hidden_layers)
+        )
+        self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        cos, sin = build_rope_cache(
+            cfg.head_dim, cfg.max_position_embeddings, cfg.rope_theta, cfg.rope_scaling
+        )
+        self.register_buffer("rope_cos", cos, persistent=False)
+        self.register_buffer("rope_sin", sin, persistent=False)
+
+    def forward(self, input_ids: torch.Tensor, position_ids: torch.Tensor | None = None) -> torch.Tensor:
+        x = self.embed_tokens(input_ids)
+        S = input_ids.shape[1]
+        if position_ids is None:
+            cos, sin = self.rope_cos[:S], self.rope_sin[:S]
+        else:
+            cos, sin = self.rope_cos[position_ids[0]], self.rope_sin[position_ids[0]]
+        for layer in self.layers:
+            x = layer(x, cos, sin)
+        return self.norm(x)
+
+
+class LlamaForCausalLM(nn.Module):
+    hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM")
+    config_class = LlamaConfig
+
+    @staticmethod
+    def config_from_hf(hf_cfg) -> LlamaConfig:
+        return LlamaConfig.from_hf_config(hf_cfg)
+
+    def __init__(self, config: LlamaConfig | dict, backend: BackendConfig | dict | None = None):
+        super().__init__()
+        if isinstance(config, dict):
+            config = LlamaConfig(**config)
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+        backend = BackendConfig.resolve(
+            backend if not isinstance(backend, dict) else BackendConfig(**backend), device_type
+        )
+        self.config = config
+        self.backend = backend
+        self.model = LlamaModel(config, backend)
+        self.lm_head = nn.Linear(config.hidden_size, config.vocab_size, bias=False)
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        # set by the recipe; called as loss_fn(hidden, lm_head_weight, labels).
+        # Computing the loss INSIDE forward keeps lm_head.weight unsharded
+        # under FSDP2 (reference computes it via lm-weight gather instead,
+        # loss/linear_ce.py:147).
+        self.loss_fn = None
+
+    # -- forward paths ---------------------------------------------------------
+    def forward(
+        self,
+        input_ids: torch.Tensor,
+        labels: torch.Tensor | None = None,
+        position_ids: torch.Tensor | None = None,
+        return_hidden: bool = False,
+        **_: Any,
+    ) -> torch.Tensor:
+        """labels given -> token-sum loss; return_hidden -> [B,S,H]; else logits.
+
+        ``labels`` must already be shifted (labels[t] is the target of
+        position t), matching the reference's dataset convention.
+        """
+        hidden = self.model(input_ids, position_ids)
+        if labels is not None:
+            assert self.loss_fn is not None, "set model.loss_fn before passing labels"
+            return self.loss_fn(hidden, self.lm_head.weight, labels)
+        if return_hidden:
+            return hidden
+        return self.lm_head(hidden)
+
+    # -- init ------------------------------------------------------------------
+    @torch.no_grad()
+    def init_weights(self, device=None) -> None:
+        """Random init (used for synthetic benchmarks and tests)."""
+        std = self.config.initializer_range
+
+        def _init(m: nn.Module):
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+                if m.bias is not None:
+                    nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, mean=0.0, std=std)
+            elif isinstance(m, RMSNorm):
+                nn.init.ones_(m.weight)
+
+        if device is not None:
+            self.to_empty(device=device)
+            # re-register non-persistent rope buffers lost by to_empty on meta
+            cos, sin = build_rope_cache(
+                self.config.head_dim,
+                self.config.max_position_embeddings,
+                self.config.rope_theta,
+                self.config.rope_scaling,
+                device=device,
+            )
+            self.model.rope_cos.copy_(cos)
+            self.model.rope_sin.copy_(sin)
+        self.apply(_init)
+        if self.config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+
+    def num_parameters(self) -> int:
+        seen, total = set(), 0
+        for p in self.parameters():
+            if id(p) not in seen:
+                seen.add(id(p))
+                total += p.numel()
+        return total

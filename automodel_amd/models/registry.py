@@ -1,0 +1,96 @@
+"""Model registry + builder: HF architecture name -> in-tree MI355X model.
+
+Reference behavior: nemo_automodel/_transformers/registry.py:485 (maps HF
+``architectures`` to custom model classes) and auto_model.py:380-643
+(meta-device init -> weight load / random init).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+from typing import Any
+
+import torch
+
+_REGISTRY: dict[str, Any] = {}
+
+
+def register_architecture(names: tuple[str, ...]):
+    def deco(cls):
+        for n in names:
+            _REGISTRY[n] = cls
+        return cls
+
+    return deco
+
+
+def get_model_class(architecture: str):
+    _ensure_builtin()
+    if architecture not in _REGISTRY:
+        raise KeyError(
+            f"architecture '{architecture}' not registered; known: {sorted(_REGISTRY)}"
+        )
+    return _REGISTRY[architecture]
+
+
+def _ensure_builtin() -> None:
+    if _REGISTRY:
+        return
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    for name in LlamaForCausalLM.hf_architectures:
+        _REGISTRY[name] = LlamaForCausalLM
+    try:
+        from automodel_amd.moe.model import MoEForCausalLM
+
+        for name in MoEForCausalLM.hf_architectures:
+            _REGISTRY[name] = MoEForCausalLM
+    except ImportError:
+        pass
+
+
+def build_model(
+    config: dict | None = None,
+    pretrained_path: str | None = None,
+    architecture: str | None = None,
+    backend: dict | None = None,
+    dtype: str = "bfloat16",
+    device: str | None = None,
+    meta_init: bool = True,
+) -> torch.nn.Module:
+    """Construct a model from an arch-config dict or a local HF checkpoint dir.
+
+    ``pretrained_path``: directory with HF ``config.json`` (+ safetensors —
+    loaded by the checkpointer after sharding, reference checkpointing.py:1228).
+    Random init happens on ``device`` when no weights are loaded.
+    """
+    if pretrained_path:
+        with open(os.path.join(pretrained_path, "config.json")) as f:
+            hf_cfg = json.load(f)
+        architecture = architecture or hf_cfg.get("architectures", ["LlamaForCausalLM"])[0]
+        cls = get_model_class(architecture)
+        model_cfg = cls.config_from_hf(hf_cfg) if hasattr(cls, "config_from_hf") else None
+        if model_cfg is None:
+            from automodel_amd.models.llama.model import LlamaConfig
+
+            model_cfg = LlamaConfig.from_hf_config(hf_cfg)
+    else:
+        assert config is not None, "build_model needs config= or pretrained_path="
+        cls = get_model_class(architecture or "LlamaForCausalLM")
+        cfg_cls = getattr(cls, "config_class", None)
+        if cfg_cls is None:
+            from automodel_amd.models.llama.model import LlamaConfig
+
+            cfg_cls = LlamaConfig
+        model_cfg = cfg_cls(**dict(config))
+
+    torch_dtype = getattr(torch, dtype) if isinstance(dtype, str) else dtype
+    init_device = torch.device("meta") if meta_init else None
+    if init_device is not None:
+        with init_device:
+            model = cls(model_cfg, backend=backend)
+    else:
+        model = cls(model_cfg, backend=backend)
+    model = model.to(dtype=torch_dtype)
+    return model

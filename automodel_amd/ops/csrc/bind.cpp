@@ -1,0 +1,44 @@
+// TORCH_LIBRARY registration for the amd_ops namespace.
+// Loaded from python via torch.ops.load_library (automodel_amd/ops/_backend.py).
+
+#include <torch/library.h>
+
+#include "ops_api.h"
+
+TORCH_LIBRARY(amd_ops, m) {
+  m.def("rms_norm_fwd(Tensor x, Tensor w, float eps) -> (Tensor, Tensor)");
+  m.impl("rms_norm_fwd", &amd_ops::rms_norm_fwd);
+  m.def("rms_norm_bwd(Tensor dy, Tensor x, Tensor w, Tensor invrms) -> (Tensor, Tensor)");
+  m.impl("rms_norm_bwd", &amd_ops::rms_norm_bwd);
+
+  m.def("rope_fwd(Tensor q, Tensor k, Tensor cos, Tensor sin, bool backward) -> (Tensor, Tensor)");
+  m.impl("rope_fwd", &amd_ops::rope_fwd);
+
+  m.def("swiglu_fwd(Tensor g, Tensor u) -> Tensor");
+  m.impl("swiglu_fwd", &amd_ops::swiglu_fwd);
+  m.def("swiglu_bwd(Tensor dy, Tensor g, Tensor u) -> (Tensor, Tensor)");
+  m.impl("swiglu_bwd", &amd_ops::swiglu_bwd);
+
+  m.def(
+      "adamw_step(Tensor(a!) param, Tensor grad, Tensor(b!) master, Tensor(c!) m, "
+      "Tensor(d!) v, int step, float lr, float beta1, float beta2, float eps, "
+      "float weight_decay) -> ()");
+  m.impl("adamw_step", &amd_ops::adamw_step);
+
+  m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, float scale, bool causal) -> (Tensor, Tensor)");
+  m.impl("flash_attn_fwd", &amd_ops::flash_attn_fwd);
+  m.def(
+      "flash_attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v, Tensor o, Tensor lse, "
+      "float scale, bool causal) -> (Tensor, Tensor, Tensor)");
+  m.impl("flash_attn_bwd", &amd_ops::flash_attn_bwd);
+
+  m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
+  m.impl("mfma_probe", &amd_ops::mfma_probe);
+
+  m.def("fused_ce_fwd(Tensor hidden, Tensor weight, Tensor labels) -> (Tensor, Tensor)");
+  m.impl("fused_ce_fwd", &amd_ops::fused_ce_fwd);
+  m.def(
+      "fused_ce_bwd(Tensor hidden, Tensor weight, Tensor labels, Tensor lse, Tensor dloss)"
+      " -> (Tensor, Tensor)");
+  m.impl("fused_ce_bwd", &amd_ops::fused_ce_bwd);
+}

@@ -1,0 +1,44 @@
+// Declarations of all HIP op entry points (implemented in the sibling .hip TUs).
+#pragma once
+
+#include <ATen/ATen.h>
+#include <tuple>
+
+namespace amd_ops {
+
+std::tuple<at::Tensor, at::Tensor> rms_norm_fwd(const at::Tensor& x, const at::Tensor& w,
+                                                double eps);
+std::tuple<at::Tensor, at::Tensor> rms_norm_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                                const at::Tensor& w, const at::Tensor& invrms);
+
+std::tuple<at::Tensor, at::Tensor> rope_fwd(const at::Tensor& q, const at::Tensor& k,
+                                            const at::Tensor& cosb, const at::Tensor& sinb,
+                                            bool backward);
+
+at::Tensor swiglu_fwd(const at::Tensor& g, const at::Tensor& u);
+std::tuple<at::Tensor, at::Tensor> swiglu_bwd(const at::Tensor& dy, const at::Tensor& g,
+                                              const at::Tensor& u);
+
+void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor master, at::Tensor m,
+                at::Tensor v, int64_t step, double lr, double beta1, double beta2,
+                double eps, double weight_decay);
+
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at::Tensor& k,
+                                                  const at::Tensor& v, double scale,
+                                                  bool causal);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
+    const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal);
+
+at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b);
+
+std::tuple<at::Tensor, at::Tensor> fused_ce_fwd(const at::Tensor& hidden,
+                                                const at::Tensor& weight,
+                                                const at::Tensor& labels);
+std::tuple<at::Tensor, at::Tensor> fused_ce_bwd(const at::Tensor& hidden,
+                                                const at::Tensor& weight,
+                                                const at::Tensor& labels,
+                                                const at::Tensor& lse,
+                                                const at::Tensor& dloss);
+
+}  // namespace amd_ops

@@ -1,0 +1,98 @@
+"""Rotary position embedding: fused HIP q+k application, host-precomputed tables.
+
+Replaces the reference's fused-RoPE backends (TE/QuACK/Liger).
+Convention matches HF Llama rotate-half: y = x*cos + rotate_half(x)*sin with
+rotate_half([x1, x2]) = [-x2, x1] over the two halves of head_dim. cos/sin are
+precomputed on host (guide Appendix B: never per-element device trig) with
+shape [S, head_dim] (half-table duplicated, fp32).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from automodel_amd.ops._backend import hip_ops
+
+
+def build_rope_cache(
+    head_dim: int,
+    max_seq_len: int,
+    base: float = 10000.0,
+    scaling: dict | None = None,
+    device=None,
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Return (cos, sin) of shape [max_seq_len, head_dim], fp32.
+
+    ``scaling`` supports HF rope_scaling dicts: llama3-style
+    {rope_type: "llama3", factor, low_freq_factor, high_freq_factor,
+    original_max_position_embeddings} and linear {rope_type: "linear", factor}.
+    """
+    inv_freq = 1.0 / (base ** (torch.arange(0, head_dim, 2, dtype=torch.float32) / head_dim))
+    if scaling:
+        rope_type = scaling.get("rope_type", scaling.get("type", "default"))
+        if rope_type == "linear":
+            inv_freq = inv_freq / float(scaling["factor"])
+        elif rope_type == "llama3":
+            factor = float(scaling["factor"])
+            low = float(scaling.get("low_freq_factor", 1.0))
+            high = float(scaling.get("high_freq_factor", 4.0))
+            orig = float(scaling.get("original_max_position_embeddings", 8192))
+            wavelen = 2 * torch.pi / inv_freq
+            ratio = orig / wavelen
+            smooth = ((ratio - low) / (high - low)).clamp(0.0, 1.0)
+            scaled = inv_freq / factor
+            inv_freq = torch.where(
+                wavelen > orig / low,
+                torch.where(wavelen < orig / high, inv_freq, (1 - smooth) * scaled + smooth * inv_freq),
+                inv_freq,
+            )
+    t = torch.arange(max_seq_len, dtype=torch.float32)
+    freqs = torch.outer(t, inv_freq)
+    emb = torch.cat([freqs, freqs], dim=-1)
+    cos, sin = emb.cos(), emb.sin()
+    if device is not None:
+        cos, sin = cos.to(device), sin.to(device)
+    return cos, sin
+
+
+def _rotate_half(x: torch.Tensor) -> torch.Tensor:
+    x1, x2 = x.chunk(2, dim=-1)
+    return torch.cat([-x2, x1], dim=-1)
+
+
+def apply_rope_ref(
+    q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """q,k: [B, S, H, D]; cos/sin: [S, D] fp32."""
+    c = cos.to(q.dtype)[None, :, None, :]
+    s = sin.to(q.dtype)[None, :, None, :]
+    return q * c + _rotate_half(q) * s, k * c + _rotate_half(k) * s
+
+
+class _RopeHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, cos, sin):
+        q, k = q.contiguous(), k.contiguous()
+        qo, ko = hip_ops().rope_fwd(q, k, cos, sin, False)
+        ctx.save_for_backward(cos, sin)
+        return qo, ko
+
+    @staticmethod
+    def backward(ctx, dq, dk):
+        cos, sin = ctx.saved_tensors
+        # d/dx of rotate is rotate by -theta: cos stays, sin negates.
+        dqi, dki = hip_ops().rope_fwd(dq.contiguous(), dk.contiguous(), cos, sin, True)
+        return dqi, dki, None, None
+
+
+def apply_rope(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    backend: str = "hip",
+) -> tuple[torch.Tensor, torch.Tensor]:
+    """Apply RoPE to q [B,S,Hq,D] and k [B,S,Hk,D] with position-sliced tables."""
+    if backend == "hip" and q.is_cuda:
+        return _RopeHip.apply(q, k, cos, sin)
+    return apply_rope_ref(q, k, cos, sin)

@@ -1,0 +1,47 @@
+"""FSDP2 (fully_shard) sharding tuned for MI355X.
+
+Reference behavior: nemo_automodel/components/distributed/parallelizer.py:1192
+(apply_fsdp2_sharding_recursively: per-decoder-layer fully_shard, root
+unsharded params, explicit prefetch) and fsdp2.py:85 (FSDP2Manager).
+
+MI355X-first choices (SURVEY §7 Phase 1): with 288 GB HBM3E per GPU,
+``reshard_after_forward=False`` is the default (params stay gathered between
+forward and backward — xGMI all-gather happens once per step, not twice), and
+the per-layer bucket IS the whole decoder layer (large collectives suit the
+per-link-bound ring over 7x153 GB/s xGMI).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+from torch.distributed.device_mesh import DeviceMesh
+from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
+
+
+def apply_fsdp(
+    model: nn.Module,
+    mesh: DeviceMesh,
+    layer_cls_names: tuple[str, ...] = ("LlamaDecoderLayer",),
+    param_dtype: torch.dtype = torch.bfloat16,
+    reduce_dtype: torch.dtype = torch.float32,
+    reshard_after_forward: bool = False,
+) -> nn.Module:
+    """Shard each decoder layer, then the root; wire backward prefetch."""
+    mp = MixedPrecisionPolicy(param_dtype=param_dtype, reduce_dtype=reduce_dtype)
+    layers = [
+        m for m in model.modules() if type(m).__name__ in layer_cls_names
+    ]
+    for layer in layers:
+        fully_shard(layer, mesh=mesh, mp_policy=mp, reshard_after_forward=reshard_after_forward)
+    fully_shard(model, mesh=mesh, mp_policy=mp, reshard_after_forward=reshard_after_forward)
+
+    # explicit prefetch: each layer prefetches the next (fwd) / previous (bwd)
+    # at depth 2, mirroring the reference's DefaultParallelizationStrategy
+    # (parallelizer.py:290-291); FSDP2 does implicit depth-1 already.
+    for i, layer in enumerate(layers):
+        if i + 2 < len(layers):
+            layer.set_modules_to_forward_prefetch([layers[i + 1], layers[i + 2]])
+        if i - 2 >= 0:
+            layer.set_modules_to_backward_prefetch([layers[i - 1], layers[i - 2]])
+    return model

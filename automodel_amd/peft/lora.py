@@ -1,0 +1,94 @@
+"""LoRA / PEFT: LinearLoRA wrapper, wildcard module matching, merge.
+
+Reference behavior: nemo_automodel/components/_peft/lora.py:45-570
+(PeftConfig, LinearLoRA with frozen base + A/B adapters, wildcard
+target_modules matching, apply_lora_to_linear_modules freezes the base model).
+The fused SGMV HIP kernels (reference _peft/lora_kernel.py) are a later
+optimization; the adapter math here is two small GEMMs riding hipBLASLt.
+"""
+
+from __future__ import annotations
+
+import fnmatch
+import math
+from dataclasses import dataclass, field
+
+import torch
+import torch.nn as nn
+
+
+@dataclass
+class PeftConfig:
+    target_modules: list[str] = field(
+        default_factory=lambda: ["*q_proj", "*k_proj", "*v_proj", "*o_proj"]
+    )
+    dim: int = 8                      # rank
+    alpha: float = 16.0
+    dropout: float = 0.0
+    use_dora: bool = False
+
+    @classmethod
+    def from_config(cls, cfg) -> "PeftConfig":
+        if isinstance(cfg, cls):
+            return cfg
+        d = dict(cfg.items()) if hasattr(cfg, "items") else dict(cfg)
+        d.pop("_target_", None)
+        return cls(**{k: (list(v) if k == "target_modules" else v) for k, v in d.items()})
+
+
+class LinearLoRA(nn.Module):
+    """y = base(x) + (dropout(x) @ A^T) @ B^T * (alpha / r); base frozen."""
+
+    def __init__(self, base: nn.Linear, dim: int, alpha: float, dropout: float = 0.0):
+        super().__init__()
+        self.base = base
+        self.dim = dim
+        self.scale = alpha / dim
+        dtype = base.weight.dtype
+        dev = base.weight.device
+        self.lora_A = nn.Linear(base.in_features, dim, bias=False, dtype=dtype, device=dev)
+        self.lora_B = nn.Linear(dim, base.out_features, bias=False, dtype=dtype, device=dev)
+        self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        self.reset_lora_parameters()
+        self.base.weight.requires_grad_(False)
+        if self.base.bias is not None:
+            self.base.bias.requires_grad_(False)
+
+    def reset_lora_parameters(self) -> None:
+        if not self.lora_A.weight.is_meta:
+            nn.init.kaiming_uniform_(self.lora_A.weight, a=math.sqrt(5))
+            nn.init.zeros_(self.lora_B.weight)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.base(x) + self.lora_B(self.lora_A(self.dropout(x))) * self.scale
+
+    @torch.no_grad()
+    def merge(self) -> nn.Linear:
+        """Fold the adapter into the base weight and return the plain Linear."""
+        delta = (self.lora_B.weight @ self.lora_A.weight) * self.scale
+        self.base.weight += delta.to(self.base.weight.dtype)
+        return self.base
+
+
+def apply_lora_to_linear_modules(model: nn.Module, cfg) -> int:
+    """Freeze the model; replace matching nn.Linear with LinearLoRA. Returns
+    the number of adapted modules (reference lora.py:570)."""
+    peft = PeftConfig.from_config(cfg)
+    for p in model.parameters():
+        p.requires_grad_(False)
+
+    replaced = 0
+    for name, module in list(model.named_modules()):
+        for child_name, child in list(module.named_children()):
+            full = f"{name}.{child_name}" if name else child_name
+            if isinstance(child, nn.Linear) and not isinstance(child, LinearLoRA):
+                if any(fnmatch.fnmatch(full, pat) for pat in peft.target_modules):
+                    setattr(module, child_name,
+                            LinearLoRA(child, peft.dim, peft.alpha, peft.dropout))
+                    replaced += 1
+    return replaced
+
+
+def lora_state_dict(model: nn.Module) -> dict[str, torch.Tensor]:
+    """Adapter-only state dict (reference checkpoint/addons.py PEFT saves)."""
+    return {k: v for k, v in model.state_dict().items() if "lora_" in k}

@@ -1,0 +1,275 @@
+"""Flagship LLM SFT / pretrain recipe (next-token prediction).
+
+Reference behavior: nemo_automodel/recipes/llm/train_ft.py:436-1601
+(TrainFinetuneRecipeForNextTokenPrediction: setup() builds mesh -> model ->
+optimizer -> dataloaders -> schedulers; run_train_validation_loop() drives
+grad-accum steps with token-count-normalized token-sum loss, grad clip,
+metrics, checkpoint cadence).
+
+MI355X layout: one process per GPU over RCCL; FSDP2 sharding with
+reshard_after_forward=False (288 GB HBM3E); hot ops on in-tree HIP kernels.
+"""
+
+from __future__ import annotations
+
+import os
+import sys
+import time
+from typing import Any
+
+import torch
+import torch.distributed as dist
+
+from automodel_amd.config.loader import ConfigNode, load_yaml_config, parse_cli_overrides, apply_overrides
+from automodel_amd.datasets.loader import build_dataloader
+from automodel_amd.datasets.mock import MockDataset, MockIterableDataset
+from automodel_amd.loggers.metric_logger import MetricLogger, setup_logging
+from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
+from automodel_amd.models.registry import build_model
+from automodel_amd.optim.adamw import build_adamw
+from automodel_amd.optim.lr_scheduler import WarmupDecayLR
+from automodel_amd.parallel.fsdp import apply_fsdp
+from automodel_amd.parallel.mesh import build_mesh, init_distributed
+from automodel_amd.recipes.base import BaseRecipe
+from automodel_amd.training.rng import StatefulRNG
+from automodel_amd.training.step_scheduler import StepScheduler
+from automodel_amd.training.utils import clip_grad_norm_, count_label_tokens, prepare_for_grad_accumulation
+
+
+class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
+    def __init__(self, cfg: ConfigNode):
+        super().__init__(cfg)
+        self.logger = setup_logging()
+        self.device = torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+
+    # ------------------------------------------------------------------ setup
+    def setup(self) -> None:
+        cfg = self.cfg
+        self.rank_id, self.local_rank, self.world = init_distributed()
+        self.rng = StatefulRNG(seed=cfg.get("seed", 42), ranked=True)
+
+        dist_cfg = cfg.get("distributed", ConfigNode())
+        self.mesh = build_mesh(
+            dp_replicate=dist_cfg.get("dp_replicate", 1),
+            dp_shard=dist_cfg.get("dp_shard", -1),
+            tp=dist_cfg.get("tp", 1),
+            pp=dist_cfg.get("pp", 1),
+            cp=dist_cfg.get("cp", 1),
+        )
+
+        # ---- model (meta init -> shard -> materialize/load)
+        mcfg = cfg.model
+        if "_target_" in mcfg:
+            self.model = mcfg.instantiate()
+        else:
+            self.model = build_model(
+                config=mcfg.get("config") and mcfg.config.to_dict(),
+                pretrained_path=mcfg.get("pretrained_path"),
+                architecture=mcfg.get("architecture"),
+                backend=mcfg.get("backend") and mcfg.backend.to_dict(),
+                dtype=mcfg.get("dtype", "bfloat16"),
+            )
+
+        # loss lives inside forward so FSDP keeps lm_head unsharded at use
+        loss_cfg = cfg.get("loss_fn", ConfigNode())
+        self.loss_fn = loss_cfg.maybe_instantiate(
+            default=FusedLinearCrossEntropy(
+                backend=loss_cfg.get("backend", "hip_fused"),
+                chunk_size=loss_cfg.get("chunk_size", 2048),
+            )
+        )
+        self.model.loss_fn = self.loss_fn
+
+        if self.mesh.mesh is not None and self.mesh.dims["dp_shard"] > 1:
+            apply_fsdp(
+                self.model,
+                self.mesh["dp_shard"],
+                reshard_after_forward=dist_cfg.get("reshard_after_forward", False),
+            )
+
+        pretrained = mcfg.get("pretrained_path")
+        if pretrained and os.path.exists(os.path.join(pretrained, "model.safetensors.index.json")) or \
+           pretrained and any(f.endswith(".safetensors") for f in os.listdir(pretrained)):
+            from automodel_amd.checkpoint.hf_loader import load_hf_weights
+            load_hf_weights(self.model, pretrained, device=self.device)
+        else:
+            self.model.init_weights(device=self.device)
+
+        # ---- PEFT
+        peft_cfg = cfg.get("peft")
+        if peft_cfg:
+            from automodel_amd.peft.lora import apply_lora_to_linear_modules
+            apply_lora_to_linear_modules(self.model, peft_cfg)
+
+        # ---- optimizer / schedulers
+        opt_cfg = cfg.get("optimizer", ConfigNode())
+        self.optimizer = opt_cfg.maybe_instantiate(model=self.model) or build_adamw(
+            self.model,
+            lr=opt_cfg.get("lr", 2e-5),
+            weight_decay=opt_cfg.get("weight_decay", 0.01),
+            betas=tuple(opt_cfg.get("betas", (0.9, 0.999))),
+        )
+        sched_cfg = cfg.get("step_scheduler", ConfigNode())
+        self.step_scheduler = StepScheduler(
+            grad_acc_steps=sched_cfg.get("grad_acc_steps", 1),
+            ckpt_every_steps=sched_cfg.get("ckpt_every_steps", 0),
+            val_every_steps=sched_cfg.get("val_every_steps", 0),
+            max_steps=sched_cfg.get("max_steps"),
+            num_epochs=sched_cfg.get("num_epochs", 1),
+        )
+        lr_cfg = cfg.get("lr_scheduler", ConfigNode())
+        self.lr_scheduler = WarmupDecayLR(
+            self.optimizer,
+            warmup_steps=lr_cfg.get("warmup_steps", 0),
+            total_steps=lr_cfg.get("total_steps", sched_cfg.get("max_steps")),
+            decay=lr_cfg.get("decay", "constant"),
+            min_lr_ratio=lr_cfg.get("min_lr_ratio", 0.0),
+        )
+
+        # ---- data
+        self.train_loader = self._build_loader(cfg.get("dataloader", ConfigNode()))
+        self.step_scheduler.dataloader = self.train_loader
+        self.val_loader = None
+        if cfg.get("validation") and cfg.validation.get("dataloader"):
+            self.val_loader = self._build_loader(cfg.validation.dataloader)
+
+        # ---- checkpointing
+        ckpt_cfg = cfg.get("checkpoint")
+        if ckpt_cfg and ckpt_cfg.get("enabled", True) and ckpt_cfg.get("checkpoint_dir"):
+            from automodel_amd.checkpoint.checkpointing import Checkpointer
+            self.checkpointer = Checkpointer(
+                checkpoint_dir=ckpt_cfg.checkpoint_dir,
+                model_save_format=ckpt_cfg.get("model_save_format", "safetensors"),
+                save_consolidated=ckpt_cfg.get("save_consolidated", False),
+                keep_last_n=ckpt_cfg.get("keep_last_n"),
+            )
+
+        # ---- metric logging
+        out_dir = cfg.get("output_dir", "outputs")
+        self.metrics = MetricLogger(os.path.join(out_dir, "training.jsonl"))
+        self.max_grad_norm = cfg.get("max_grad_norm", 1.0)
+
+        restore = cfg.get("restore_from")
+        if restore:
+            self.load_checkpoint(restore)
+
+    def _build_loader(self, dcfg: ConfigNode):
+        ds_cfg = dcfg.get("dataset", ConfigNode())
+        if "_target_" in ds_cfg:
+            dataset = ds_cfg.instantiate()
+        else:
+            kind = ds_cfg.get("kind", "mock_iterable")
+            kwargs = {k: v for k, v in ds_cfg.items() if k != "kind"}
+            dataset = (MockIterableDataset if kind == "mock_iterable" else MockDataset)(**kwargs)
+        return build_dataloader(
+            dataset,
+            batch_size=dcfg.get("batch_size", 1),
+            shuffle=dcfg.get("shuffle", True),
+            num_workers=dcfg.get("num_workers", 0),
+            pad_token_id=dcfg.get("pad_token_id", 0),
+            dp_rank=self.mesh.dp_rank if self.mesh.mesh is not None else 0,
+            dp_world=self.mesh.dp_size,
+            seed=self.cfg.get("seed", 42),
+        )
+
+    # ------------------------------------------------------------- train step
+    def _forward_backward_step(self, batch: dict, loss_scale: float) -> torch.Tensor:
+        input_ids = batch["input_ids"].to(self.device, non_blocking=True)
+        labels = batch["labels"].to(self.device, non_blocking=True)
+        loss = self.model(input_ids, labels=labels)
+        (loss * loss_scale).backward()
+        return loss.detach()
+
+    def _run_train_optim_step(self, batches: list[dict]) -> dict[str, Any]:
+        t0 = time.perf_counter()
+        device = self.device
+        num_label_tokens = torch.zeros((), dtype=torch.long)
+        for b in batches:
+            num_label_tokens += count_label_tokens(b["labels"])
+        num_label_tokens = num_label_tokens.to(device)
+        if self.world > 1:
+            dist.all_reduce(num_label_tokens, group=self.mesh.dp_group())
+        global_tokens = max(1, int(num_label_tokens.item()))
+
+        # loss_sum / global_tokens * dp_world compensates FSDP's mean-reduce
+        loss_scale = self.mesh.dp_size / global_tokens
+        total_loss = torch.zeros((), dtype=torch.float32, device=device)
+        for i, batch in enumerate(batches):
+            prepare_for_grad_accumulation(self.model, is_final_microbatch=(i == len(batches) - 1))
+            total_loss += self._forward_backward_step(batch, loss_scale).float()
+
+        grad_norm = clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
+        self.optimizer.step()
+        self.optimizer.zero_grad(set_to_none=True)
+        self.lr_scheduler.step()
+
+        if self.world > 1:
+            dist.all_reduce(total_loss, group=self.mesh.dp_group())
+        step_time = time.perf_counter() - t0
+        ntok = int(num_label_tokens.item())
+        return {
+            "step": self.step_scheduler.step,
+            "loss": total_loss.item() / max(1, ntok),
+            "grad_norm": float(grad_norm),
+            "lr": self.lr_scheduler.get_last_lr()[0],
+            "num_label_tokens": ntok,
+            "step_time_s": step_time,
+            "tps": ntok / step_time,
+            "mem_gb": (torch.cuda.max_memory_allocated() / 2**30) if torch.cuda.is_available() else 0.0,
+        }
+
+    @torch.no_grad()
+    def _run_validation_epoch(self) -> dict[str, float]:
+        self.model.eval()
+        total_loss, total_tok = 0.0, 0
+        for batch in self.val_loader:
+            input_ids = batch["input_ids"].to(self.device, non_blocking=True)
+            labels = batch["labels"].to(self.device, non_blocking=True)
+            loss = self.model(input_ids, labels=labels)
+            total_loss += float(loss)
+            total_tok += int(count_label_tokens(labels))
+        self.model.train()
+        t = torch.tensor([total_loss, float(total_tok)], device=self.device)
+        if self.world > 1:
+            dist.all_reduce(t, group=self.mesh.dp_group())
+        return {"val_loss": float(t[0] / max(1.0, float(t[1])))}
+
+    # -------------------------------------------------------------- main loop
+    def run_train_validation_loop(self) -> None:
+        self.model.train()
+        for epoch in self.step_scheduler.epochs:
+            self.train_loader.set_epoch(epoch)
+            for batches in self.step_scheduler:
+                metrics = self._run_train_optim_step(batches)
+                self.metrics.log(metrics)
+                if self.rank == 0:
+                    self.logger.info(
+                        f"step {metrics['step']} | loss {metrics['loss']:.4f} | "
+                        f"gnorm {metrics['grad_norm']:.3f} | lr {metrics['lr']:.2e} | "
+                        f"tok {metrics['num_label_tokens']} | {metrics['step_time_s']*1e3:.0f} ms"
+                    )
+                if self.val_loader is not None and self.step_scheduler.is_val_step:
+                    vm = self._run_validation_epoch()
+                    self.metrics.log({"step": metrics["step"], **vm})
+                if self.step_scheduler.is_ckpt_step and hasattr(self, "checkpointer"):
+                    self.save_checkpoint(
+                        os.path.join(self.checkpointer.checkpoint_dir,
+                                     f"step_{self.step_scheduler.step}")
+                    )
+        self.metrics.close()
+
+
+def main(argv: list[str] | None = None) -> None:
+    argv = argv if argv is not None else sys.argv[1:]
+    assert argv, "usage: python -m automodel_amd.recipes.llm.train_ft <cfg.yaml> [--a.b=c ...]"
+    cfg = load_yaml_config(argv[0])
+    apply_overrides(cfg, parse_cli_overrides(argv[1:]))
+    recipe = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    recipe.setup()
+    recipe.run_train_validation_loop()
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

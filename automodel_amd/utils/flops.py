@@ -1,0 +1,46 @@
+"""Analytic FLOPs + MFU calculation.
+
+Reference behavior: nemo_automodel/components/utils/flops_utils.py:19-650
+(per-family analytic FLOPs; MFU = achieved / peak). MI355X denominator:
+peak dense bf16 ~= 2.5 PFLOP/s per GPU (CDNA4; AMD's larger headline figures
+include 2:1 structured sparsity — guide §5.4 rule 11).
+"""
+
+from __future__ import annotations
+
+MI355X_PEAK_BF16 = 2.5e15  # dense
+MI355X_PEAK_FP8 = 5.0e15
+
+
+def llama_flops_per_token(
+    hidden: int,
+    intermediate: int,
+    layers: int,
+    vocab: int,
+    seq_len: int,
+    num_heads: int,
+    num_kv_heads: int,
+    head_dim: int | None = None,
+) -> float:
+    """Training FLOPs per token (fwd+bwd = 3x fwd) for a Llama-style model.
+
+    Matches the reference's llama3 calculator structure (flops_utils.py:95):
+    attention projections + scores + MLP + LM head; causal attention counted
+    at S/2 average context.
+    """
+    d = head_dim or hidden // num_heads
+    q_size = num_heads * d
+    kv_size = num_kv_heads * d
+    # per-layer, per-token MACs
+    attn_proj = hidden * q_size + 2 * hidden * kv_size + q_size * hidden
+    attn_scores = 2 * (seq_len / 2) * d * num_heads  # QK^T + PV at avg causal ctx
+    mlp = 3 * hidden * intermediate
+    per_layer = attn_proj + attn_scores + mlp
+    lm_head = hidden * vocab
+    fwd = 2 * (layers * per_layer + lm_head)  # 2 FLOPs per MAC
+    return 3.0 * fwd  # fwd + bwd (2x)
+
+
+def mfu(tokens_per_sec_per_gpu: float, flops_per_token: float,
+        peak: float = MI355X_PEAK_BF16) -> float:
+    return tokens_per_sec_per_gpu * flops_per_token / peak

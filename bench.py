@@ -1,0 +1,173 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: Llama-3-8B full-param SFT, FSDP2, bf16, seq=4096.
+
+Measures the BASELINE.json metric — tokens/sec (whole job) + MFU at
+1/2/4/8 MI355X — on synthetic data with random-init weights (no network).
+Reference anchor: NVIDIA-NeMo/Automodel performance-summary row "Llama3 8B,
+1xH100, 12,472.87 tok/s/GPU" (that row is LoRA — this
+bench does the strictly-harder full-parameter SFT of BASELINE.json config #2).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W                 # single GPU
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=8)
+    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--mbs", type=int, default=4, help="micro-batch size per GPU")
+    p.add_argument("--seq-len", type=int, default=4096)
+    p.add_argument("--model", type=str, default="llama3_8b",
+                   choices=["llama3_8b", "llama3_1b_proxy"])
+    p.add_argument("--attn", type=str, default="hip")
+    p.add_argument("--loss", type=str, default="chunked")
+    p.add_argument("--profile-steps", type=int, default=0,
+                   help="if >0, run this many steps (no JSON contract) for rocprof")
+    return p.parse_args()
+
+
+MODEL_CONFIGS = {
+    "llama3_8b": dict(
+        vocab_size=128256, hidden_size=4096, intermediate_size=14336,
+        num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
+        max_position_embeddings=8192, rope_theta=500000.0, rms_norm_eps=1e-5,
+    ),
+    # small proxy for smoke/debug runs
+    "llama3_1b_proxy": dict(
+        vocab_size=128256, hidden_size=2048, intermediate_size=8192,
+        num_hidden_layers=16, num_attention_heads=16, num_key_value_heads=8,
+        head_dim=128,
+        max_position_embeddings=8192, rope_theta=500000.0, rms_norm_eps=1e-5,
+    ),
+}
+
+
+def main():
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", 1))
+    rank = int(os.environ.get("RANK", 0))
+
+    from automodel_amd.parallel.mesh import build_mesh, init_distributed
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    from automodel_amd.models.common.backend import BackendConfig
+    from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
+    from automodel_amd.optim.adamw import FusedAdamW
+    from automodel_amd.parallel.fsdp import apply_fsdp
+    from automodel_amd.utils.flops import llama_flops_per_token, mfu, MI355X_PEAK_BF16
+
+    init_distributed()
+    use_cuda = torch.cuda.is_available()
+    device = torch.device("cuda") if use_cuda else torch.device("cpu")
+    mesh = build_mesh(dp_shard=-1)
+
+    cfg = LlamaConfig(**MODEL_CONFIGS[args.model])
+    backend = BackendConfig(attn=args.attn, loss=args.loss)
+    torch.manual_seed(1234 + rank)
+    with torch.device("meta"):
+        model = LlamaForCausalLM(cfg, backend=backend)
+    model = model.to(dtype=torch.bfloat16)
+    model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=4096)
+    if world > 1:
+        apply_fsdp(model, mesh["dp_shard"], reshard_after_forward=False)
+    model.init_weights(device=device)
+    model.train()
+    opt = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.0)
+
+    # synthetic data (BASELINE measurement conditions: mock data)
+    g = torch.Generator(device="cpu").manual_seed(5678 + rank)
+    batches = []
+    for _ in range(2):
+        ids = torch.randint(0, cfg.vocab_size, (args.mbs, args.seq_len + 1), generator=g)
+        batches.append(
+            (ids[:, :-1].to(device), ids[:, 1:].contiguous().to(device))
+        )
+
+    def step(i: int):
+        input_ids, labels = batches[i % len(batches)]
+        loss = model(input_ids, labels=labels)
+        n_tok = labels.numel()
+        (loss / n_tok).backward()
+        opt.step()
+        opt.zero_grad(set_to_none=True)
+        return float(loss.detach()) / n_tok
+
+    n_steps = args.profile_steps if args.profile_steps > 0 else args.steps
+    for i in range(args.warmup):
+        step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    t0 = time.perf_counter()
+    last_loss = None
+    for i in range(n_steps):
+        last_loss = step(i)
+    if use_cuda:
+        torch.cuda.synchronize()
+    if dist.is_initialized():
+        dist.barrier()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    t = torch.tensor([elapsed], dtype=torch.float64,
+                     device=device if dist.is_initialized() and dist.get_backend() == "nccl" else "cpu")
+    if dist.is_initialized():
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t.item())
+
+    tokens_per_step = args.mbs * args.seq_len * world
+    tps = tokens_per_step * n_steps / elapsed
+    fpt = llama_flops_per_token(
+        cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
+        cfg.vocab_size, args.seq_len, cfg.num_attention_heads,
+        cfg.num_key_value_heads, cfg.head_dim,
+    )
+    achieved_mfu = mfu(tps / world, fpt) if use_cuda else None
+    baseline_tok_per_gpu = 12472.87
+
+    if rank == 0 and args.profile_steps == 0:
+        print(json.dumps({
+            "metric": "tokens/sec (whole node) + MFU, Llama-3-8B SFT FSDP2 at 1/2/4/8 MI355X",
+            "value": round(tps, 1),
+            "unit": "tokens/s",
+            "n_gpus": world,
+            "steps": n_steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / n_steps * 1e3, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(tps / (baseline_tok_per_gpu * world), 4),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": "llama3-8b" if args.model == "llama3_8b" else args.model,
+                "global_batch": args.mbs * world,
+                "seq_len": args.seq_len,
+                "parallelism": f"fsdp{world}" if world > 1 else "single",
+                "mfu": round(achieved_mfu, 4) if achieved_mfu is not None else None,
+                "flops_per_token": fpt,
+                "loss_per_token": last_loss,
+                "attn_backend": args.attn,
+                "loss_backend": args.loss,
+            },
+        }))
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
