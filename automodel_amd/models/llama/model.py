@@ -146,6 +146,9 @@ class LlamaModel(nn.Module):
             cos, sin = self.rope_cos[:S], self.rope_sin[:S]
         else:
             cos, sin = self.rope_cos[position_ids[0]], self.rope_sin[position_ids[0]]
+        # model.to(bf16) converts buffers; the HIP rope kernel wants f32 tables
+        if cos.dtype != torch.float32:
+            cos, sin = cos.float(), sin.float()
         for layer in self.layers:
             x = layer(x, cos, sin)
         return self.norm(x)
