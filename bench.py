@@ -90,7 +90,7 @@ def main():
     # synthetic data (BASELINE measurement conditions: mock data)
     g = torch.Generator(device="cpu").manual_seed(5678 + rank)
     batches = []
-    for _ in range(2):
+    for _ in range(8):
         ids = torch.randint(0, cfg.vocab_size, (args.mbs, args.seq_len + 1), generator=g)
         batches.append(
             (ids[:, :-1].to(device), ids[:, 1:].contiguous().to(device))

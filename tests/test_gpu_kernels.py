@@ -69,8 +69,8 @@ def test_rope_parity():
     assert torch.allclose(qo.float(), qr.float(), atol=2e-2, rtol=2e-2)
     assert torch.allclose(ko.float(), kr.float(), atol=2e-2, rtol=2e-2)
     dq, dk = torch.randn_like(qo), torch.randn_like(ko)
-    qo.backward(dq); ko.backward(dk)
-    qr.backward(dq); kr.backward(dk)
+    torch.autograd.backward([qo, ko], [dq, dk])
+    torch.autograd.backward([qr, kr], [dq, dk])
     assert torch.allclose(q.grad.float(), q2.grad.float(), atol=2e-2, rtol=2e-2)
     assert torch.allclose(k.grad.float(), k2.grad.float(), atol=2e-2, rtol=2e-2)
 
