@@ -35,6 +35,11 @@ TORCH_LIBRARY(amd_ops, m) {
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
   m.impl("mfma_probe", &amd_ops::mfma_probe);
 
+  m.def("ce_fwd_logits(Tensor logits, Tensor labels, Tensor(a!) loss_sum) -> (Tensor, Tensor)");
+  m.impl("ce_fwd_logits", &amd_ops::ce_fwd_logits);
+  m.def("ce_bwd_logits(Tensor(a!) logits, Tensor labels, Tensor lse, Tensor dloss) -> ()");
+  m.impl("ce_bwd_logits", &amd_ops::ce_bwd_logits);
+
   m.def("fused_ce_fwd(Tensor hidden, Tensor weight, Tensor labels) -> (Tensor, Tensor)");
   m.impl("fused_ce_fwd", &amd_ops::fused_ce_fwd);
   m.def(

@@ -32,6 +32,12 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
 
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b);
 
+std::tuple<at::Tensor, at::Tensor> ce_fwd_logits(const at::Tensor& logits,
+                                                 const at::Tensor& labels,
+                                                 at::Tensor loss_sum);
+void ce_bwd_logits(at::Tensor logits, const at::Tensor& labels, const at::Tensor& lse,
+                   const at::Tensor& dloss);
+
 std::tuple<at::Tensor, at::Tensor> fused_ce_fwd(const at::Tensor& hidden,
                                                 const at::Tensor& weight,
                                                 const at::Tensor& labels);

@@ -34,7 +34,7 @@ def parse_args():
     p.add_argument("--model", type=str, default="llama3_8b",
                    choices=["llama3_8b", "llama3_1b_proxy"])
     p.add_argument("--attn", type=str, default="hip")
-    p.add_argument("--loss", type=str, default="chunked")
+    p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="if >0, run this many steps (no JSON contract) for rocprof")
     return p.parse_args()

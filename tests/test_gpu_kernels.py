@@ -166,6 +166,33 @@ def test_flash_attention_bwd_grads():
         (v.grad.float() - v2.grad).abs().max()
 
 
+def test_hybrid_linear_ce_parity():
+    """hybrid (hipBLASLt GEMM + HIP CE epilogue) vs fp32 dense reference."""
+    from automodel_amd.loss.linear_ce import fused_linear_cross_entropy
+
+    torch.manual_seed(0)
+    T, H, V = 512, 256, 1024
+    hidden = torch.randn(T, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    weight = torch.randn(V, H, device="cuda", dtype=torch.bfloat16, requires_grad=True) * 0.05
+    weight = weight.detach().requires_grad_(True)
+    labels = torch.randint(0, V, (T,), device="cuda")
+    labels[7] = -100
+    loss = fused_linear_cross_entropy(hidden, weight, labels, backend="hybrid",
+                                      chunk_size=128)
+    loss.backward()
+
+    h2 = hidden.detach().float().requires_grad_(True)
+    w2 = weight.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(h2 @ w2.t(), labels,
+                                            ignore_index=-100, reduction="sum")
+    ref.backward()
+    assert torch.allclose(loss, ref, rtol=2e-2), (float(loss), float(ref))
+    assert torch.allclose(hidden.grad.float(), h2.grad, atol=5e-2, rtol=5e-2), \
+        (hidden.grad.float() - h2.grad).abs().max()
+    assert torch.allclose(weight.grad.float(), w2.grad, atol=5e-2, rtol=5e-2), \
+        (weight.grad.float() - w2.grad).abs().max()
+
+
 def test_model_train_step_cuda():
     """Tiny model full step on GPU with all HIP backends active."""
     from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
