@@ -110,8 +110,8 @@ class _FlashAttnHip(torch.autograd.Function):
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
-        dq, dk, dv = _attention_bwd_composite(
-            do, q, k, v, o, lse, ctx.causal, ctx.scale
+        dq, dk, dv = hip_ops().flash_attn_bwd(
+            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal
         )
         return dq, dk, dv, None, None
 

@@ -252,12 +252,400 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at:
   return {out, lse};
 }
 
+// ===========================================================================
+// Backward: flash-attention-2 style split.
+//   delta kernel: delta[b,h,s] = rowsum(dO * O)
+//   kernel A (kv-parallel): recompute S^T (swapped, lse/delta lane-local),
+//     accumulate dK, dV; P^T/dS^T transposed through a per-wave LDS buffer;
+//     f32 atomicAdd into dk/dv (GQA heads collapse onto the kv head).
+//   kernel B (q-parallel): recompute S in [q][k] layout, accumulate dQ
+//     (plain stores — q tiles are exclusive per block).
+// Replaces the GEMM-composite python backward (profiles/bench8b round-1
+// showed its masked_fill/exp/mul/f32-add chain at ~25% of step time).
+// ===========================================================================
+
+__global__ void fa_delta_kernel(const bf16* __restrict__ dout, const bf16* __restrict__ o,
+                                float* __restrict__ delta, int S, int Hq, long rows) {
+  // row r = ((b*S + s)*Hq + h); one wave per row, D=128 -> 2 elems/lane
+  const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE_SIZE;
+  if (row >= rows) return;
+  const int lane = threadIdx.x & (WAVE_SIZE - 1);
+  const bf16* dp = dout + row * 128 + lane * 2;
+  const bf16* op = o + row * 128 + lane * 2;
+  float acc = bf2f(dp[0]) * bf2f(op[0]) + bf2f(dp[1]) * bf2f(op[1]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) {
+    // delta layout [B,Hq,S]: row -> (b, s, h)
+    const long h = row % Hq;
+    const long bs = row / Hq;
+    const long b = bs / S, s = bs % S;
+    delta[(b * Hq + h) * S + s] = acc;
+  }
+}
+
+// q/do row images: [32][256 B], XOR swizzle ((row&7)<<4) (same as k_lds_off)
+// qt/dot images:   [128][64 B], XOR swizzle ((d&3)<<4)   (same as vt_lds_off)
+// per-wave transpose buffer: [32][64 B] with ((row&3)<<4)
+__device__ __forceinline__ int tb_off(int row, int byte_in_row) {
+  return row * 64 + (byte_in_row ^ ((row & 3) << 4));
+}
+
+#define FAB_WAVES 4
+
+template <int D>
+__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel(
+    const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
+    float* __restrict__ dk, float* __restrict__ dv, int B, int S, int Hq, int Hk,
+    float scale, bool causal) {
+  static_assert(D == 128);
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* q_rows = smem;                 // 32*256 = 8 KiB
+  char* do_rows = smem + 8 * 1024;     // 8 KiB
+  char* qt = smem + 16 * 1024;         // 128*64 = 8 KiB
+  char* dot = smem + 24 * 1024;        // 8 KiB
+  char* tbuf = smem + 32 * 1024;       // per-wave 2 KiB x4
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid / 64;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+
+  const int kvb = blockIdx.x * (FAB_WAVES * 32);
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int kvh = h / (Hq / Hk);
+  const int kv0 = kvb + wid * 32;      // this wave's kv rows
+
+  const long q_base = (((long)b * S) * Hq + h) * D;
+  const long kv_base = (((long)b * S) * Hk + kvh) * D;
+  const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
+  const float* lse_row = lse + ((long)b * Hq + h) * S;
+  const float* dlt_row = delta + ((long)b * Hq + h) * S;
+
+  // K/V rows of this wave -> A fragments in registers
+  bf16x8_v kfrag[D / 16], vfrag[D / 16];
+#pragma unroll
+  for (int c = 0; c < D / 16; ++c) {
+    const long off = kv_base + (long)(kv0 + col) * kv_rs + c * 16 + half * 8;
+    kfrag[c] = *reinterpret_cast<const bf16x8_v*>(k + off);
+    vfrag[c] = *reinterpret_cast<const bf16x8_v*>(v + off);
+  }
+
+  f32x16 dk_acc[D / 32], dv_acc[D / 32];
+#pragma unroll
+  for (int t = 0; t < D / 32; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) { dk_acc[t][r] = 0.f; dv_acc[t][r] = 0.f; }
+
+  char* tb = tbuf + wid * 2048;
+  const int jq_start = causal ? kvb / 32 : 0;
+
+  for (int jq = jq_start; jq < S / 32; ++jq) {
+    const int q0 = jq * 32;
+    // ---- cooperative stage: q/do rows + transposed images
+    {
+      const int r0 = tid / 16, c0 = (tid % 16) * 8;   // 16 threads per row of 128
+      for (int rr = r0; rr < 32; rr += 16) {
+        bf16x8 qv = *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
+        bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(dout + q_base + (long)(q0 + rr) * q_rs + c0);
+        *reinterpret_cast<bf16x8*>(q_rows + k_lds_off(rr, c0 * 2)) = qv;
+        *reinterpret_cast<bf16x8*>(do_rows + k_lds_off(rr, c0 * 2)) = dv8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          *reinterpret_cast<bf16*>(qt + vt_lds_off(c0 + j, rr * 2)) = qv.v[j];
+          *reinterpret_cast<bf16*>(dot + vt_lds_off(c0 + j, rr * 2)) = dv8.v[j];
+        }
+      }
+    }
+    __syncthreads();
+
+    const bool live = !causal || (q0 + 31 >= kv0);
+    if (live) {
+      const int qg = q0 + col;
+      const float lse_q = lse_row[qg];
+      const float dlt_q = dlt_row[qg];
+
+      // ---- S^T = K Q^T (D rows = kv, cols = q)
+      f32x16 st;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        bf16x8_v qb = *reinterpret_cast<const bf16x8_v*>(
+            q_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qb, st, 0, 0, 0);
+      }
+      // ---- dP^T = V dO^T
+      f32x16 dpt;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        bf16x8_v db = *reinterpret_cast<const bf16x8_v*>(
+            do_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], db, dpt, 0, 0, 0);
+      }
+
+      float p[16], ds[16];
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        const bool masked = (causal && kg > qg) || kg >= S;
+        p[r] = masked ? 0.f : __expf(st[r] * scale - lse_q);
+        ds[r] = p[r] * (dpt[r] - dlt_q) * scale;
+      }
+
+      // ---- transpose P^T -> A-frags via wave-local LDS buffer, accumulate dV
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(p[r]);
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int c2 = 0; c2 < 2; ++c2) {
+        bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
+            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+#pragma unroll
+        for (int t = 0; t < D / 32; ++t) {
+          bf16x8_v dob = *reinterpret_cast<const bf16x8_v*>(
+              dot + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
+          dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
+        }
+      }
+      // ---- transpose dS^T, accumulate dK
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(ds[r]);
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int c2 = 0; c2 < 2; ++c2) {
+        bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
+            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+#pragma unroll
+        for (int t = 0; t < D / 32; ++t) {
+          bf16x8_v qb2 = *reinterpret_cast<const bf16x8_v*>(
+              qt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
+          dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
+        }
+      }
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: atomic accumulate into f32 dk/dv [B,S,Hk,D]
+#pragma unroll
+  for (int t = 0; t < D / 32; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+      const int kg = kv0 + krow;
+      if (kg < S) {
+        const long off = kv_base + (long)kg * kv_rs + t * 32 + col;
+        atomicAdd(dk + off, dk_acc[t][r]);
+        atomicAdd(dv + off, dv_acc[t][r]);
+      }
+    }
+  }
+}
+
+template <int D>
+__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE) void flash_bwd_dq_kernel(
+    const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
+    const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
+    bf16* __restrict__ dq, int B, int S, int Hq, int Hk, float scale, bool causal) {
+  static_assert(D == 128);
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_rows = smem;                 // 8 KiB ([32][256B] swz)
+  char* v_rows = smem + 8 * 1024;      // 8 KiB
+  char* kt = smem + 16 * 1024;         // 8 KiB ([128][64B] swz)
+  char* tbuf = smem + 24 * 1024;       // per-wave 2 KiB x4
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid / 64;
+  const int col = lane & 31;
+  const int half = lane >> 5;
+
+  const int qb_blk = blockIdx.x * (FAB_WAVES * 32);
+  const int h = blockIdx.y;
+  const int b = blockIdx.z;
+  const int kvh = h / (Hq / Hk);
+  const int q0 = qb_blk + wid * 32;
+
+  const long q_base = (((long)b * S) * Hq + h) * D;
+  const long kv_base = (((long)b * S) * Hk + kvh) * D;
+  const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
+  const float* lse_row = lse + ((long)b * Hq + h) * S;
+  const float* dlt_row = delta + ((long)b * Hq + h) * S;
+
+  // Q/dO rows of this wave as A-fragments
+  bf16x8_v qfrag[D / 16], dofrag[D / 16];
+#pragma unroll
+  for (int c = 0; c < D / 16; ++c) {
+    const long off = q_base + (long)(q0 + col) * q_rs + c * 16 + half * 8;
+    qfrag[c] = *reinterpret_cast<const bf16x8_v*>(q + off);
+    dofrag[c] = *reinterpret_cast<const bf16x8_v*>(dout + off);
+  }
+  // per-lane lse/delta for this lane's 16 q rows (D-layout rows)
+  float lse_q[16], dlt_q[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+    lse_q[r] = lse_row[qrow];
+    dlt_q[r] = dlt_row[qrow];
+  }
+
+  f32x16 dq_acc[D / 32];
+#pragma unroll
+  for (int t = 0; t < D / 32; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
+
+  char* tb = tbuf + wid * 2048;
+  const int block_q_max = qb_blk + FAB_WAVES * 32 - 1;
+  const int n_tiles = causal ? (min(block_q_max, S - 1) / 32 + 1) : (S + 31) / 32;
+
+  for (int jk = 0; jk < n_tiles; ++jk) {
+    const int k0 = jk * 32;
+    // ---- stage K rows, V rows, K^T
+    {
+      const int r0 = tid / 16, c0 = (tid % 16) * 8;
+      for (int rr = r0; rr < 32; rr += 16) {
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)(k0 + rr) * kv_rs + c0);
+        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)(k0 + rr) * kv_rs + c0);
+        *reinterpret_cast<bf16x8*>(k_rows + k_lds_off(rr, c0 * 2)) = kv8;
+        *reinterpret_cast<bf16x8*>(v_rows + k_lds_off(rr, c0 * 2)) = vv8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<bf16*>(kt + vt_lds_off(c0 + j, rr * 2)) = kv8.v[j];
+      }
+    }
+    __syncthreads();
+
+    const bool live = !causal || (k0 <= q0 + 31);
+    if (live) {
+      // ---- S[q][k] = Q K^T : A=Q rows, B from k_rows (contiguous d)
+      f32x16 s;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) s[r] = 0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        bf16x8_v kb = *reinterpret_cast<const bf16x8_v*>(
+            k_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag[c], kb, s, 0, 0, 0);
+      }
+      // ---- dP[q][k] = dO V^T
+      f32x16 dp;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) dp[r] = 0.f;
+#pragma unroll
+      for (int c = 0; c < D / 16; ++c) {
+        bf16x8_v vb = *reinterpret_cast<const bf16x8_v*>(
+            v_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+        dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vb, dp, 0, 0, 0);
+      }
+
+      float ds[16];
+      const int kg = k0 + col;   // D cols = k here
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qg = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        const bool masked = (causal && kg > qg) || kg >= S;
+        const float p = masked ? 0.f : __expf(s[r] * scale - lse_q[r]);
+        ds[r] = p * (dp[r] - dlt_q[r]) * scale;
+      }
+
+      // ---- transpose dS (cols=k -> A-frag rows=q), accumulate dQ
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
+        *reinterpret_cast<bf16*>(tb + tb_off(qrow, col * 2)) = f2bf(ds[r]);
+      }
+      __builtin_amdgcn_wave_barrier();
+#pragma unroll
+      for (int c2 = 0; c2 < 2; ++c2) {
+        bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
+            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+#pragma unroll
+        for (int t = 0; t < D / 32; ++t) {
+          bf16x8_v ktb = *reinterpret_cast<const bf16x8_v*>(
+              kt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
+          dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, ktb, dq_acc[t], 0, 0, 0);
+        }
+      }
+      __builtin_amdgcn_wave_barrier();
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: plain bf16 stores (q rows exclusive to this block)
+#pragma unroll
+  for (int t = 0; t < D / 32; ++t) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      if (qrow < S) {
+        dq[q_base + (long)qrow * q_rs + t * 32 + col] = f2bf(dq_acc[t][r]);
+      }
+    }
+  }
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
     const at::Tensor& o, const at::Tensor& lse, double scale, bool causal) {
-  TORCH_CHECK(false,
-              "flash_attn_bwd HIP kernel not built yet — python wrapper uses the "
-              "GEMM-composite backward (ops/attention.py)");
+  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Hk = k.size(2);
+  TORCH_CHECK(D == 128 && S % 128 == 0, "flash_attn_bwd: D=128, S%128==0 required");
+  auto stream = c10::hip::getCurrentHIPStream();
+
+  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  {
+    const long rows = (long)B * S * Hq;
+    const int block = 256;
+    const long grid = (rows * WAVE_SIZE + block - 1) / block;
+    hipLaunchKernelGGL(fa_delta_kernel, dim3((unsigned)grid), dim3(block), 0, stream.stream(),
+                       reinterpret_cast<const bf16*>(dout.data_ptr()),
+                       reinterpret_cast<const bf16*>(o.data_ptr()),
+                       delta.data_ptr<float>(), S, Hq, rows);
+    HIP_CHECK_KERNEL();
+  }
+
+  auto dq = at::empty_like(q);
+  auto dk32 = at::zeros({B, S, Hk, D}, q.options().dtype(at::kFloat));
+  auto dv32 = at::zeros({B, S, Hk, D}, q.options().dtype(at::kFloat));
+
+  const dim3 grid_kv(S / 128, Hq, B);
+  const size_t smem_a = 40 * 1024;
+  hipLaunchKernelGGL((flash_bwd_dkv_kernel<128>), grid_kv, dim3(256), smem_a, stream.stream(),
+                     reinterpret_cast<const bf16*>(dout.data_ptr()),
+                     reinterpret_cast<const bf16*>(q.data_ptr()),
+                     reinterpret_cast<const bf16*>(k.data_ptr()),
+                     reinterpret_cast<const bf16*>(v.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     dk32.data_ptr<float>(), dv32.data_ptr<float>(),
+                     B, S, Hq, Hk, (float)scale, causal);
+  HIP_CHECK_KERNEL();
+
+  const dim3 grid_q(S / 128, Hq, B);
+  const size_t smem_b = 32 * 1024;
+  hipLaunchKernelGGL((flash_bwd_dq_kernel<128>), grid_q, dim3(256), smem_b, stream.stream(),
+                     reinterpret_cast<const bf16*>(dout.data_ptr()),
+                     reinterpret_cast<const bf16*>(q.data_ptr()),
+                     reinterpret_cast<const bf16*>(k.data_ptr()),
+                     reinterpret_cast<const bf16*>(v.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(dq.data_ptr()),
+                     B, S, Hq, Hk, (float)scale, causal);
+  HIP_CHECK_KERNEL();
+
+  return {dq, dk32.to(at::kBFloat16), dv32.to(at::kBFloat16)};
 }
 
 // ---- MFMA layout self-test: d[32,32] = a[32,16] @ b[16,32] via one mfma.
