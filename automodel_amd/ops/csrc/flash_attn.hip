@@ -47,7 +47,7 @@ __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
 }
 
 template <int D>
-__global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
+__global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ out, float* __restrict__ lse, int B, int S, int Hq, int Hk,
     float scale, bool causal) {
@@ -131,27 +131,27 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
         p = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[c], p, 0, 0, 0);
       }
 
-      // ---- masked online softmax (lane owns q row q0+col, ks (r&3)+8*(r>>2)+4*half)
+      // ---- masked online softmax, IN PLACE in the p accumulator (register
+      // budget: s_val/pv scratch arrays cost 32 VGPRs -> occupancy cliff)
       const int qg = q0 + col;
-      float s_val[16];
       float tile_max = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * half;
         float sv = p[r] * scale;
         if ((causal && kg > qg) || kg >= S) sv = -1e30f;
-        s_val[r] = sv;
+        p[r] = sv;
         tile_max = fmaxf(tile_max, sv);
       }
       tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32));
       const float m_new = fmaxf(m_run, tile_max);
       alpha = __expf(m_run - m_new);
       float psum = 0.f;
-      float pv[16];
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        pv[r] = __expf(s_val[r] - m_new);
-        psum += pv[r];
+        float e = __expf(p[r] - m_new);
+        p[r] = e;
+        psum += e;
       }
       psum += __shfl_xor(psum, 32);
       l_run = l_run * alpha + psum;
@@ -161,8 +161,8 @@ __global__ __launch_bounds__(FA_BLOCK) void flash_fwd_kernel(
       unsigned int pk[8];
 #pragma unroll
       for (int i = 0; i < 8; ++i) {
-        unsigned lo = __builtin_bit_cast(unsigned short, f2bf(pv[2 * i]));
-        unsigned hi = __builtin_bit_cast(unsigned short, f2bf(pv[2 * i + 1]));
+        unsigned lo = __builtin_bit_cast(unsigned short, f2bf(p[2 * i]));
+        unsigned hi = __builtin_bit_cast(unsigned short, f2bf(p[2 * i + 1]));
         pk[i] = lo | (hi << 16);
       }
       // fragment 0: k 0..15  <- regs 0..7 ; fragment 1: k 16..31 <- regs 8..15
@@ -388,20 +388,21 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
         dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], db, dpt, 0, 0, 0);
       }
 
-      float p[16], ds[16];
+      // in place: st becomes P^T, dpt becomes dS^T (register budget)
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
         const bool masked = (causal && kg > qg) || kg >= S;
-        p[r] = masked ? 0.f : __expf(st[r] * scale - lse_q);
-        ds[r] = p[r] * (dpt[r] - dlt_q) * scale;
+        const float pv = masked ? 0.f : __expf(st[r] * scale - lse_q);
+        st[r] = pv;
+        dpt[r] = pv * (dpt[r] - dlt_q) * scale;
       }
 
       // ---- transpose P^T -> A-frags via wave-local LDS buffer, accumulate dV
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(p[r]);
+        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(st[r]);
       }
       __builtin_amdgcn_wave_barrier();
 #pragma unroll
@@ -420,7 +421,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(ds[r]);
+        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(dpt[r]);
       }
       __builtin_amdgcn_wave_barrier();
 #pragma unroll
@@ -455,7 +456,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
 }
 
 template <int D>
-__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE) void flash_bwd_dq_kernel(
+__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, int B, int S, int Hq, int Hk, float scale, bool causal) {
@@ -465,6 +466,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE) void flash_bwd_dq_kernel(
   char* v_rows = smem + 8 * 1024;      // 8 KiB
   char* kt = smem + 16 * 1024;         // 8 KiB ([128][64B] swz)
   char* tbuf = smem + 24 * 1024;       // per-wave 2 KiB x4
+  float* stats = reinterpret_cast<float*>(smem + 32 * 1024);  // lse[128]+delta[128]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -492,13 +494,10 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE) void flash_bwd_dq_kernel(
     qfrag[c] = *reinterpret_cast<const bf16x8_v*>(q + off);
     dofrag[c] = *reinterpret_cast<const bf16x8_v*>(dout + off);
   }
-  // per-lane lse/delta for this lane's 16 q rows (D-layout rows)
-  float lse_q[16], dlt_q[16];
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
-    lse_q[r] = lse_row[qrow];
-    dlt_q[r] = dlt_row[qrow];
+  // lse/delta for the block's 128 q rows -> LDS (32 regs saved per lane)
+  if (tid < 128) {
+    stats[tid] = lse_row[qb_blk + tid];
+    stats[128 + tid] = dlt_row[qb_blk + tid];
   }
 
   f32x16 dq_acc[D / 32];
@@ -551,21 +550,21 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE) void flash_bwd_dq_kernel(
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vb, dp, 0, 0, 0);
       }
 
-      float ds[16];
       const int kg = k0 + col;   // D cols = k here
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        const int qg = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+        const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;  // block-local + wid*32
+        const int qg = q0 + qrow;
         const bool masked = (causal && kg > qg) || kg >= S;
-        const float p = masked ? 0.f : __expf(s[r] * scale - lse_q[r]);
-        ds[r] = p * (dp[r] - dlt_q[r]) * scale;
+        const float p = masked ? 0.f : __expf(s[r] * scale - stats[wid * 32 + qrow]);
+        s[r] = p * (dp[r] - stats[128 + wid * 32 + qrow]) * scale;  // dS in place
       }
 
       // ---- transpose dS (cols=k -> A-frag rows=q), accumulate dQ
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        *reinterpret_cast<bf16*>(tb + tb_off(qrow, col * 2)) = f2bf(ds[r]);
+        *reinterpret_cast<bf16*>(tb + tb_off(qrow, col * 2)) = f2bf(s[r]);
       }
       __builtin_amdgcn_wave_barrier();
 #pragma unroll
@@ -634,7 +633,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
   HIP_CHECK_KERNEL();
 
   const dim3 grid_q(S / 128, Hq, B);
-  const size_t smem_b = 32 * 1024;
+  const size_t smem_b = 33 * 1024;
   hipLaunchKernelGGL((flash_bwd_dq_kernel<128>), grid_q, dim3(256), smem_b, stream.stream(),
                      reinterpret_cast<const bf16*>(dout.data_ptr()),
                      reinterpret_cast<const bf16*>(q.data_ptr()),
