@@ -1,0 +1,180 @@
+"""MoE causal LM (Qwen3-MoE / Mixtral-class), MI355X-native.
+
+Reference behavior: nemo_automodel/components/models/qwen3_moe/model.py and
+components/moe/layers.py — llama-style attention + MoE FFN with stacked
+expert weights, HF key parity via state_dict_adapter.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass, field
+from typing import Any
+
+import torch
+import torch.nn as nn
+
+from automodel_amd.models.common.backend import BackendConfig
+from automodel_amd.models.llama.model import LlamaAttention, LlamaConfig
+from automodel_amd.moe.config import MoEConfig
+from automodel_amd.moe.layers import MoE, FakeBalancedGate, Gate
+from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+from automodel_amd.ops.rms_norm import RMSNorm
+from automodel_amd.ops.rope import build_rope_cache
+
+
+@dataclass
+class MoEModelConfig(LlamaConfig):
+    moe: MoEConfig = field(default_factory=MoEConfig)
+    hf_flavor: str = "qwen3_moe"   # qwen3_moe | mixtral
+
+    def __post_init__(self):
+        super().__post_init__()
+        if isinstance(self.moe, dict):
+            self.moe = MoEConfig(**self.moe)
+
+    @classmethod
+    def from_hf_config(cls, hf: Any) -> "MoEModelConfig":
+        if hasattr(hf, "to_dict"):
+            hf = hf.to_dict()
+        base = LlamaConfig.from_hf_config(hf).__dict__.copy()
+        arch = (hf.get("architectures") or ["Qwen3MoeForCausalLM"])[0]
+        if "Mixtral" in arch:
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_local_experts", 8),
+                n_activated_experts=hf.get("num_experts_per_tok", 2),
+                moe_intermediate_size=hf.get("intermediate_size", 14336),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                norm_topk_prob=True,
+            )
+            flavor = "mixtral"
+        else:
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_experts", 64),
+                n_activated_experts=hf.get("num_experts_per_tok", 8),
+                moe_intermediate_size=hf.get("moe_intermediate_size", 768),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                norm_topk_prob=hf.get("norm_topk_prob", True),
+            )
+            flavor = "qwen3_moe"
+        return cls(**base, moe=moe, hf_flavor=flavor)
+
+
+class MoEDecoderLayer(nn.Module):
+    def __init__(self, cfg: MoEModelConfig, backend: BackendConfig):
+        super().__init__()
+        self.self_attn = LlamaAttention(cfg, backend)
+        self.mlp = MoE(cfg.hidden_size, cfg.moe)
+        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+
+    def forward(self, x, cos, sin):
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+        x = x + self.mlp(self.post_attention_layernorm(x))
+        return x
+
+
+class MoEForCausalLM(nn.Module):
+    hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM")
+    config_class = MoEModelConfig
+
+    @staticmethod
+    def config_from_hf(hf_cfg) -> MoEModelConfig:
+        return MoEModelConfig.from_hf_config(hf_cfg)
+
+    def __init__(self, config: MoEModelConfig | dict, backend: BackendConfig | dict | None = None):
+        super().__init__()
+        if isinstance(config, dict):
+            config = MoEModelConfig(**config)
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+        backend = BackendConfig.resolve(
+            backend if not isinstance(backend, dict) else BackendConfig(**backend), device_type
+        )
+        self.config = config
+        self.backend = backend
+        self.model = nn.Module()
+        self.model.embed_tokens = nn.Embedding(config.vocab_size, config.hidden_size)
+        self.model.layers = nn.ModuleList(
+            MoEDecoderLayer(config, backend) for _ in range(config.num_hidden_layers)
+        )
+        self.model.norm = RMSNorm(config.hidden_size, config.rms_norm_eps, backend.rms_norm)
+        cos, sin = build_rope_cache(config.head_dim, config.max_position_embeddings,
+                                    config.rope_theta, config.rope_scaling)
+        self.model.register_buffer("rope_cos", cos, persistent=False)
+        self.model.register_buffer("rope_sin", sin, persistent=False)
+        self.lm_head = nn.Linear(config.hidden_size, config.vocab_size, bias=False)
+        if config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+        self.loss_fn = None
+        self.state_dict_adapter = MoEStateDictAdapter(config)
+
+    def forward(self, input_ids, labels=None, position_ids=None, return_hidden=False, **_):
+        x = self.model.embed_tokens(input_ids)
+        S = input_ids.shape[1]
+        cos, sin = self.model.rope_cos[:S], self.model.rope_sin[:S]
+        if cos.dtype != torch.float32:
+            cos, sin = cos.float(), sin.float()
+        for layer in self.model.layers:
+            x = layer(x, cos, sin)
+        x = self.model.norm(x)
+        if labels is not None:
+            assert self.loss_fn is not None
+            loss = self.loss_fn(x, self.lm_head.weight, labels)
+            aux = self.collect_aux_losses()
+            return loss + aux if aux is not None else loss
+        if return_hidden:
+            return x
+        return self.lm_head(x)
+
+    def collect_aux_losses(self):
+        total = None
+        for layer in self.model.layers:
+            aux = getattr(layer.mlp.gate, "last_aux_loss", None)
+            if aux is not None:
+                total = aux if total is None else total + aux
+                layer.mlp.gate.last_aux_loss = None
+        return total
+
+    @torch.no_grad()
+    def update_moe_gate_bias(self) -> None:
+        """Aux-free balancing bias update after each optim step
+        (reference train_ft.py update_moe_gate_bias)."""
+        for layer in self.model.layers:
+            load = layer.mlp.last_expert_load
+            if load is not None:
+                layer.mlp.gate.update_bias(load)
+
+    @torch.no_grad()
+    def init_weights(self, device=None) -> None:
+        std = self.config.initializer_range
+        if device is not None:
+            self.to_empty(device=device)
+            cos, sin = build_rope_cache(self.config.head_dim,
+                                        self.config.max_position_embeddings,
+                                        self.config.rope_theta,
+                                        self.config.rope_scaling, device=device)
+            self.model.rope_cos.copy_(cos)
+            self.model.rope_sin.copy_(sin)
+        for m in self.modules():
+            if isinstance(m, nn.Linear):
+                nn.init.normal_(m.weight, std=std)
+                if m.bias is not None:
+                    nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.Embedding):
+                nn.init.normal_(m.weight, std=std)
+            elif isinstance(m, RMSNorm):
+                nn.init.ones_(m.weight)
+            elif isinstance(m, Gate):
+                nn.init.normal_(m.weight, std=std)
+        for m in self.modules():
+            if type(m).__name__ == "GroupedExperts":
+                m.init_weights(std)
+        if self.config.tie_word_embeddings:
+            self.lm_head.weight = self.model.embed_tokens.weight
+
+    def num_parameters(self) -> int:
+        seen, total = set(), 0
+        for p in self.parameters():
+            if id(p) not in seen:
+                seen.add(id(p))
+                total += p.numel()
+        return total

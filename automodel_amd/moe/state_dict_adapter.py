@@ -1,0 +1,72 @@
+"""HF per-expert keys <-> stacked expert tensors.
+
+Reference behavior: nemo_automodel/components/moe/state_dict_mixin.py (953
+LoC: HF per-expert keys <-> stacked DTensor expert weights, EP-aware).
+
+Internal layout: model.layers.N.mlp.experts.{gate_proj,up_proj,down_proj}
+as stacked [E, out, in] parameters and model.layers.N.mlp.gate.weight.
+
+HF layouts:
+  qwen3_moe: model.layers.N.mlp.experts.E.{gate_proj,up_proj,down_proj}.weight
+             model.layers.N.mlp.gate.weight
+  mixtral:   model.layers.N.block_sparse_moe.experts.E.{w1,w3,w2}.weight
+             model.layers.N.block_sparse_moe.gate.weight
+             (w1=gate, w3=up, w2=down)
+"""
+
+from __future__ import annotations
+
+import re
+
+import torch
+
+_MIXTRAL_MAP = {"w1": "gate_proj", "w3": "up_proj", "w2": "down_proj"}
+_MIXTRAL_INV = {v: k for k, v in _MIXTRAL_MAP.items()}
+
+
+class MoEStateDictAdapter:
+    def __init__(self, config):
+        self.config = config
+        self.flavor = getattr(config, "hf_flavor", "qwen3_moe")
+        self.n_experts = config.moe.n_routed_experts
+
+    # ---- HF -> internal (stack per-expert tensors)
+    def from_hf(self, sd: dict[str, torch.Tensor]) -> dict[str, torch.Tensor]:
+        out: dict[str, torch.Tensor] = {}
+        pending: dict[str, dict[int, torch.Tensor]] = {}
+        pat_q = re.compile(r"^(model\.layers\.\d+)\.mlp\.experts\.(\d+)\.(gate_proj|up_proj|down_proj)\.weight$")
+        pat_m = re.compile(r"^(model\.layers\.\d+)\.block_sparse_moe\.experts\.(\d+)\.(w1|w2|w3)\.weight$")
+        for key, t in sd.items():
+            m = pat_q.match(key) or pat_m.match(key)
+            if m:
+                layer, e, proj = m.group(1), int(m.group(2)), m.group(3)
+                proj = _MIXTRAL_MAP.get(proj, proj)
+                pending.setdefault(f"{layer}.mlp.experts.{proj}", {})[e] = t
+            elif ".block_sparse_moe.gate." in key:
+                out[key.replace(".block_sparse_moe.gate.", ".mlp.gate.")] = t
+            elif ".block_sparse_moe." in key:
+                out[key.replace(".block_sparse_moe.", ".mlp.")] = t
+            else:
+                out[key] = t
+        for stacked_key, parts in pending.items():
+            out[stacked_key] = torch.stack([parts[e] for e in sorted(parts)], dim=0)
+        return out
+
+    # ---- internal -> HF (unstack)
+    def to_hf(self, sd: dict[str, torch.Tensor]) -> dict[str, torch.Tensor]:
+        out: dict[str, torch.Tensor] = {}
+        pat = re.compile(r"^(model\.layers\.\d+)\.mlp\.experts\.(gate_proj|up_proj|down_proj)$")
+        for key, t in sd.items():
+            m = pat.match(key)
+            if m:
+                layer, proj = m.group(1), m.group(2)
+                for e in range(t.shape[0]):
+                    if self.flavor == "mixtral":
+                        out[f"{layer}.block_sparse_moe.experts.{e}.{_MIXTRAL_INV[proj]}.weight"] = t[e]
+                    else:
+                        out[f"{layer}.mlp.experts.{e}.{proj}.weight"] = t[e]
+            elif self.flavor == "mixtral" and ".mlp.gate." in key:
+                out[key.replace(".mlp.gate.", ".block_sparse_moe.gate.")] = t
+            else:
+                out[key] = t
+        return out

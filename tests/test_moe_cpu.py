@@ -1,0 +1,189 @@
+"""MoE tests: gate routing, aux loss, fake gate, grouped experts vs dense
+reference, state-dict adapter roundtrip, model e2e, EP dispatcher parity."""
+
+import pytest
+import torch
+
+from automodel_amd.moe.config import MoEConfig
+from automodel_amd.moe.experts import GroupedExperts, permute_tokens, unpermute_tokens
+from automodel_amd.moe.layers import FakeBalancedGate, Gate, MoE
+from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+from tests.dist_utils import run_distributed
+
+torch.manual_seed(0)
+
+MOE_CFG = MoEConfig(n_routed_experts=4, n_activated_experts=2,
+                    moe_intermediate_size=32, aux_loss_coeff=0.01)
+
+
+def test_gate_topk_and_probs():
+    g = Gate(16, MOE_CFG)
+    torch.nn.init.normal_(g.weight)
+    g.train()
+    x = torch.randn(10, 16)
+    probs, idx = g(x)
+    assert probs.shape == (10, 2) and idx.shape == (10, 2)
+    assert torch.allclose(probs.sum(-1), torch.ones(10), atol=1e-5)  # normalized
+    assert idx.max() < 4
+    assert g.last_aux_loss is not None and g.last_aux_loss > 0
+
+
+def test_gate_expert_bias_update():
+    cfg = MoEConfig(n_routed_experts=4, n_activated_experts=1, expert_bias=True,
+                    bias_update_speed=0.1)
+    g = Gate(8, cfg)
+    torch.nn.init.zeros_(g.weight)
+    load = torch.tensor([10.0, 0.0, 0.0, 0.0])
+    g.update_bias(load)
+    assert g.e_score_bias[0] < 0 and g.e_score_bias[1] > 0  # push against overload
+
+
+def test_fake_balanced_gate_uniform():
+    g = FakeBalancedGate(16, MOE_CFG)
+    probs, idx = g(torch.randn(8, 16))
+    counts = torch.bincount(idx.reshape(-1), minlength=4)
+    assert counts.max() - counts.min() <= 0  # perfectly balanced (8*2/4 each)
+
+
+def test_permute_unpermute_roundtrip():
+    x = torch.randn(6, 8)
+    idx = torch.randint(0, 4, (6, 2))
+    probs = torch.ones(6, 2) * 0.5
+    xp, sort_idx, counts = permute_tokens(x, idx, 4)
+    assert counts.sum() == 12
+    y = unpermute_tokens(xp, sort_idx, probs)
+    assert torch.allclose(y, x, atol=1e-6)  # identity experts, probs sum to 1
+
+
+def test_grouped_experts_match_dense_reference():
+    E, H, I = 3, 8, 16
+    ge = GroupedExperts(E, H, I)
+    ge.init_weights()
+    x = torch.randn(10, H)
+    idx = torch.randint(0, E, (10, 2))
+    probs = torch.softmax(torch.randn(10, 2), dim=-1)
+    y = ge(x, probs, idx)
+    # dense reference
+    ref = torch.zeros_like(x)
+    for t in range(10):
+        for kk in range(2):
+            e = idx[t, kk]
+            h = torch.nn.functional.silu(x[t] @ ge.gate_proj[e].t()) * (x[t] @ ge.up_proj[e].t())
+            ref[t] += probs[t, kk] * (h @ ge.down_proj[e].t())
+    assert torch.allclose(y, ref, atol=1e-4), (y - ref).abs().max()
+
+
+def test_moe_layer_forward_backward():
+    moe = MoE(16, MOE_CFG)
+    moe.experts.init_weights()
+    torch.nn.init.normal_(moe.gate.weight)
+    x = torch.randn(2, 5, 16, requires_grad=True)
+    y = moe(x)
+    assert y.shape == x.shape
+    y.sum().backward()
+    assert moe.experts.gate_proj.grad is not None
+    assert moe.gate.weight.grad is not None
+    assert moe.last_expert_load.sum() == 2 * 5 * 2
+
+
+MODEL_CFG = dict(
+    vocab_size=128, hidden_size=64, intermediate_size=128, num_hidden_layers=2,
+    num_attention_heads=4, num_key_value_heads=2, max_position_embeddings=64,
+    moe={"n_routed_experts": 4, "n_activated_experts": 2,
+         "moe_intermediate_size": 32, "aux_loss_coeff": 0.01},
+)
+
+
+def test_moe_model_train_step():
+    from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
+
+    m = MoEForCausalLM(MODEL_CFG)
+    m.init_weights()
+    m.loss_fn = FusedLinearCrossEntropy(backend="chunked", chunk_size=32)
+    ids = torch.randint(0, 128, (2, 17))
+    loss = m(ids[:, :-1].contiguous(), labels=ids[:, 1:].contiguous())
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert m.model.layers[0].mlp.experts.gate_proj.grad is not None
+    assert m.model.layers[0].mlp.gate.weight.grad is not None  # aux flows
+    m.update_moe_gate_bias()
+
+
+def test_moe_state_dict_adapter_roundtrip():
+    m = MoEForCausalLM(MODEL_CFG)
+    m.init_weights()
+    sd = {k: v for k, v in m.state_dict().items()}
+    hf = m.state_dict_adapter.to_hf(sd)
+    assert "model.layers.0.mlp.experts.0.gate_proj.weight" in hf
+    assert "model.layers.0.mlp.experts.gate_proj" not in hf
+    back = m.state_dict_adapter.from_hf(hf)
+    for k in sd:
+        assert k in back, k
+        assert torch.equal(sd[k], back[k]), k
+
+
+def test_moe_mixtral_adapter():
+    cfg = MoEModelConfig.from_hf_config({
+        "architectures": ["MixtralForCausalLM"], "vocab_size": 128,
+        "hidden_size": 64, "intermediate_size": 128, "num_hidden_layers": 1,
+        "num_attention_heads": 4, "num_key_value_heads": 2,
+        "num_local_experts": 4, "num_experts_per_tok": 2,
+    })
+    assert cfg.moe.n_routed_experts == 4 and cfg.hf_flavor == "mixtral"
+    m = MoEForCausalLM(cfg)
+    m.init_weights()
+    hf = m.state_dict_adapter.to_hf(m.state_dict())
+    assert "model.layers.0.block_sparse_moe.experts.0.w1.weight" in hf
+    assert "model.layers.0.block_sparse_moe.gate.weight" in hf
+    back = m.state_dict_adapter.from_hf(hf)
+    assert "model.layers.0.mlp.experts.gate_proj" in back
+
+
+# ------------------------------------------------------------ EP dispatchers
+def _ep_fn(rank, world, dispatcher):
+    import torch.distributed as dist
+
+    from automodel_amd.moe.dispatch import AllGatherDispatcher, AllToAllDispatcher
+    from automodel_amd.moe.parallelizer import apply_ep
+    from automodel_amd.moe.layers import MoE
+
+    torch.manual_seed(0)  # same full model on both ranks
+    cfg = MoEConfig(n_routed_experts=4, n_activated_experts=2, moe_intermediate_size=32)
+    moe_ref = MoE(16, cfg)
+    moe_ref.experts.init_weights()
+    torch.nn.init.normal_(moe_ref.gate.weight)
+
+    torch.manual_seed(0)
+    moe_ep = MoE(16, cfg)
+    moe_ep.experts.init_weights()
+    torch.nn.init.normal_(moe_ep.gate.weight)
+
+    class _Mesh:
+        def get_group(self):
+            return dist.group.WORLD
+
+    apply_ep(moe_ep, _Mesh(), dispatcher=dispatcher)
+    assert moe_ep.experts.n_experts == 2
+
+    # each rank gets its own token slice; reference computes the same slice
+    torch.manual_seed(100 + rank)
+    x = torch.randn(2, 3, 16, requires_grad=True)
+    x_ref = x.detach().clone().requires_grad_(True)
+    y_ep = moe_ep(x)
+    y_ref = moe_ref(x_ref)
+    assert torch.allclose(y_ep, y_ref, atol=1e-4), (y_ep - y_ref).abs().max()
+    y_ep.sum().backward()
+    y_ref.sum().backward()
+    assert torch.allclose(x.grad, x_ref.grad, atol=1e-4)
+    # local expert grads match the reference's corresponding expert slice,
+    # up to the other rank's contribution (different tokens) — so just check
+    # they exist and are finite.
+    g = moe_ep.experts.gate_proj.grad
+    assert g is not None and torch.isfinite(g).all()
+    return float(y_ep.sum())
+
+
+@pytest.mark.parametrize("dispatcher", ["a2a", "allgather"])
+def test_ep2_dispatcher_matches_dense(dispatcher):
+    run_distributed(_ep_fn, world=2, args=(dispatcher,))
