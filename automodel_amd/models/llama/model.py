@@ -88,12 +88,17 @@ class LlamaAttention(nn.Module):
         self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        from automodel_amd.parallel.cp import active_cp, cp_flash_attention
+
         B, S, _ = x.shape
         q = self.q_proj(x).view(B, S, -1, self.head_dim)
         k = self.k_proj(x).view(B, S, -1, self.head_dim)
         v = self.v_proj(x).view(B, S, -1, self.head_dim)
         q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
-        o = flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+        if active_cp() is not None:
+            o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+        else:
+            o = flash_attention(q, k, v, causal=True, backend=self.backend.attn)
         return self.o_proj(o.reshape(B, S, -1))
 
 

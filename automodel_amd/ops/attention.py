@@ -2,11 +2,12 @@
 
 Replaces the reference's TE DotProductAttention / flash-attn / FlexAttention
 backends with one in-tree HIP flash kernel
-(csrc/flash_attn.hip): BSHD layout, GQA, online softmax, MFMA 16x16 tiles,
-LDS-staged K/V (guide Appendix B "Fused attention prefill").
+(csrc/flash_attn.hip): BSHD layout, GQA, online softmax, MFMA 32x32 tiles,
+LDS-staged K/V, fused backward (dKV kv-parallel + dQ q-parallel kernels).
 
-Shapes: q [B, S, Hq, D], k/v [B, S, Hk, D] with Hq % Hk == 0.
-Returns o [B, S, Hq, D].
+Shapes: q [B, Sq, Hq, D], k/v [B, Skv, Hk, D] with Hq % Hk == 0.
+``q_start`` offsets q positions against the kv sequence (context parallelism:
+each CP rank attends its q chunks over the gathered KV).
 """
 
 from __future__ import annotations
@@ -18,12 +19,19 @@ import torch
 from automodel_amd.ops._backend import hip_ops
 
 
-def attention_ref(q, k, v, causal: bool = True, scale: float | None = None):
+def _causal_mask(Sq: int, Skv: int, q_start: int, device) -> torch.Tensor:
+    qpos = torch.arange(q_start, q_start + Sq, device=device)
+    kpos = torch.arange(Skv, device=device)
+    return kpos[None, :] <= qpos[:, None]       # [Sq, Skv] True = keep
+
+
+def attention_ref(q, k, v, causal: bool = True, scale: float | None = None,
+                  q_start: int = 0):
     """Eager fp32 reference used by kernel parity tests."""
-    B, S, Hq, D = q.shape
-    Hk = k.shape[2]
+    B, Sq, Hq, D = q.shape
+    Skv, Hk = k.shape[1], k.shape[2]
     scale = scale or 1.0 / math.sqrt(D)
-    qf = q.permute(0, 2, 1, 3).float()              # B,Hq,S,D
+    qf = q.permute(0, 2, 1, 3).float()              # B,Hq,Sq,D
     kf = k.permute(0, 2, 1, 3).float()
     vf = v.permute(0, 2, 1, 3).float()
     if Hk != Hq:
@@ -32,88 +40,41 @@ def attention_ref(q, k, v, causal: bool = True, scale: float | None = None):
         vf = vf.repeat_interleave(rep, dim=1)
     scores = torch.matmul(qf, kf.transpose(-1, -2)) * scale
     if causal:
-        mask = torch.ones(S, S, dtype=torch.bool, device=q.device).tril()
-        scores = scores.masked_fill(~mask, float("-inf"))
+        scores = scores.masked_fill(~_causal_mask(Sq, Skv, q_start, q.device), float("-inf"))
     p = torch.softmax(scores, dim=-1)
     o = torch.matmul(p, vf)
     return o.permute(0, 2, 1, 3).to(q.dtype)
 
 
-def _sdpa(q, k, v, causal: bool, scale: float | None):
+def _sdpa(q, k, v, causal: bool, scale: float | None, q_start: int = 0):
     qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))   # B,H,S,D
-    o = torch.nn.functional.scaled_dot_product_attention(
-        qt, kt, vt, is_causal=causal, scale=scale, enable_gqa=q.shape[2] != k.shape[2]
-    )
+    kwargs = dict(scale=scale, enable_gqa=q.shape[2] != k.shape[2])
+    if causal and (q_start != 0 or q.shape[1] != k.shape[1]):
+        mask = _causal_mask(q.shape[1], k.shape[1], q_start, q.device)
+        o = torch.nn.functional.scaled_dot_product_attention(qt, kt, vt,
+                                                             attn_mask=mask, **kwargs)
+    else:
+        o = torch.nn.functional.scaled_dot_product_attention(qt, kt, vt,
+                                                             is_causal=causal, **kwargs)
     return o.transpose(1, 2)
-
-
-def _attention_bwd_composite(do, q, k, v, o, lse, causal: bool, scale: float,
-                             q_chunk: int = 1024):
-    """Backward as a chain of hipBLASLt GEMMs, q-chunked so the score matrix is
-    never fully materialized. Used until the fused HIP backward kernel lands;
-    all FLOPs ride MFMA through the GEMM library.
-
-    do/q/o: [B,S,Hq,D]; k/v: [B,S,Hk,D]; lse: [B,Hq,S] (natural log).
-    """
-    B, S, Hq, D = q.shape
-    Hk = k.shape[2]
-    rep = Hq // Hk
-    # [B*Hq, S, D] views
-    qb = q.permute(0, 2, 1, 3).reshape(B * Hq, S, D)
-    dob = do.permute(0, 2, 1, 3).reshape(B * Hq, S, D)
-    ob = o.permute(0, 2, 1, 3).reshape(B * Hq, S, D)
-    kb = k.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1).reshape(B * Hq, S, D)
-    vb = v.permute(0, 2, 1, 3).repeat_interleave(rep, dim=1).reshape(B * Hq, S, D)
-    lse_b = lse.reshape(B * Hq, S)
-
-    delta = (dob.float() * ob.float()).sum(-1)            # [BH, S]
-    dq = torch.empty_like(qb)
-    dk_acc = torch.zeros_like(kb, dtype=torch.float32)
-    dv_acc = torch.zeros_like(vb, dtype=torch.float32)
-
-    kt = kb.transpose(1, 2)                               # [BH, D, S]
-    vt = vb.transpose(1, 2)
-    arange_k = torch.arange(S, device=q.device)
-    for s0 in range(0, S, q_chunk):
-        s1 = min(s0 + q_chunk, S)
-        qc = qb[:, s0:s1]                                  # [BH, c, D]
-        kv_end = s1 if causal else S
-        scores = torch.bmm(qc, kt[:, :, :kv_end]).float() * scale
-        p = torch.exp(scores - lse_b[:, s0:s1, None])      # [BH, c, kv]
-        if causal:
-            mask = arange_k[None, None, :kv_end] > torch.arange(s0, s1, device=q.device)[None, :, None]
-            p = p.masked_fill(mask, 0.0)
-        pb = p.to(q.dtype)
-        dp = torch.bmm(dob[:, s0:s1], vt[:, :, :kv_end]).float()
-        ds = (p * (dp - delta[:, s0:s1, None]) * scale).to(q.dtype)
-        dq[:, s0:s1] = torch.bmm(ds, kb[:, :kv_end])
-        dk_acc[:, :kv_end] += torch.bmm(ds.transpose(1, 2), qc).float()
-        dv_acc[:, :kv_end] += torch.bmm(pb.transpose(1, 2), dob[:, s0:s1]).float()
-
-    dqo = dq.reshape(B, Hq, S, D).permute(0, 2, 1, 3).contiguous()
-    dk4 = dk_acc.reshape(B, Hk, rep, S, D).sum(2)
-    dv4 = dv_acc.reshape(B, Hk, rep, S, D).sum(2)
-    dko = dk4.permute(0, 2, 1, 3).to(k.dtype).contiguous()
-    dvo = dv4.permute(0, 2, 1, 3).to(v.dtype).contiguous()
-    return dqo, dko, dvo
 
 
 class _FlashAttnHip(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal: bool, scale: float):
+    def forward(ctx, q, k, v, causal: bool, scale: float, q_start: int):
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = hip_ops().flash_attn_fwd(q, k, v, scale, causal)
+        o, lse = hip_ops().flash_attn_fwd(q, k, v, scale, causal, q_start)
         ctx.save_for_backward(q, k, v, o, lse)
-        ctx.causal, ctx.scale = causal, scale
+        ctx.causal, ctx.scale, ctx.q_start = causal, scale, q_start
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
         dq, dk, dv = hip_ops().flash_attn_bwd(
-            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal
+            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal, ctx.q_start
         )
-        return dq, dk, dv, None, None
+        return dq, dk, dv, None, None, None
 
 
 def flash_attention(
@@ -123,10 +84,11 @@ def flash_attention(
     causal: bool = True,
     scale: float | None = None,
     backend: str = "hip",
+    q_start: int = 0,
 ) -> torch.Tensor:
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if backend == "hip" and q.is_cuda:
-        return _FlashAttnHip.apply(q, k, v, causal, scale)
+        return _FlashAttnHip.apply(q, k, v, causal, scale, q_start)
     if backend == "eager":
-        return attention_ref(q, k, v, causal, scale)
-    return _sdpa(q, k, v, causal, scale)
+        return attention_ref(q, k, v, causal, scale, q_start)
+    return _sdpa(q, k, v, causal, scale, q_start)

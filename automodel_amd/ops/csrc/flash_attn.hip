@@ -49,8 +49,8 @@ __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
 template <int D>
 __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
-    bf16* __restrict__ out, float* __restrict__ lse, int B, int S, int Hq, int Hk,
-    float scale, bool causal) {
+    bf16* __restrict__ out, float* __restrict__ lse, int B, int Sq, int Skv, int Hq,
+    int Hk, int q_start, float scale, bool causal) {
   static_assert(D == 128, "flash_fwd: D=128 only for now");
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                       // KVBLK*D*2 = 8 KiB
@@ -68,8 +68,8 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
   const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
 
-  const long q_base = (((long)b * S) * Hq + h) * D;        // + s*Hq*D
-  const long kv_base = (((long)b * S) * Hk + kvh) * D;     // + s*Hk*D
+  const long q_base = (((long)b * Sq) * Hq + h) * D;       // + s*Hq*D
+  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;   // + s*Hk*D
   const long q_row_stride = (long)Hq * D;
   const long kv_row_stride = (long)Hk * D;
 
@@ -91,8 +91,9 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = -1e30f, l_run = 0.f;
 
-  const int q_block_max = qblk0 + FA_WAVES * QBLK - 1;
-  const int n_tiles = causal ? (min(q_block_max, S - 1) / KVBLK + 1) : (S + KVBLK - 1) / KVBLK;
+  const int q_block_max = q_start + qblk0 + FA_WAVES * QBLK - 1;
+  const int n_tiles = causal ? (min(q_block_max, Skv - 1) / KVBLK + 1)
+                             : (Skv + KVBLK - 1) / KVBLK;
 
   for (int jt = 0; jt < n_tiles; ++jt) {
     const int k0 = jt * KVBLK;
@@ -103,7 +104,7 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
 #pragma unroll
       for (int rr = 0; rr < 2; ++rr) {
         const int row = r0 + rr * 16;
-        const int ks = k0 + row;
+        const int ks = min(k0 + row, Skv - 1);  // overhang rows masked later
         bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_row_stride + c0);
         *reinterpret_cast<bf16x8*>(k_lds + k_lds_off(row, c0 * 2)) = kv8;
         bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)ks * kv_row_stride + c0);
@@ -115,7 +116,7 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     }
     __syncthreads();
 
-    const bool tile_live = !causal || (k0 <= q0 + QBLK - 1);
+    const bool tile_live = !causal || (k0 <= q_start + q0 + QBLK - 1);
     float alpha = 1.f;
     bf16x8_v pa0, pa1;
     if (tile_live) {
@@ -133,13 +134,13 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
 
       // ---- masked online softmax, IN PLACE in the p accumulator (register
       // budget: s_val/pv scratch arrays cost 32 VGPRs -> occupancy cliff)
-      const int qg = q0 + col;
+      const int qg = q_start + q0 + col;
       float tile_max = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * half;
         float sv = p[r] * scale;
-        if ((causal && kg > qg) || kg >= S) sv = -1e30f;
+        if ((causal && kg > qg) || kg >= Skv) sv = -1e30f;
         p[r] = sv;
         tile_max = fmaxf(tile_max, sv);
       }
@@ -210,8 +211,8 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
   __syncthreads();
 
   const int qg = q0 + col;
-  if (lane < 32 && qg < S) {
-    lse[((long)b * Hq + h) * S + qg] = (l_run > 0.f) ? m_run + __logf(l_run) : -1e30f;
+  if (lane < 32 && qg < Sq) {
+    lse[((long)b * Hq + h) * Sq + qg] = (l_run > 0.f) ? m_run + __logf(l_run) : -1e30f;
   }
 #pragma unroll
   for (int t = 0; t < D / 32; ++t) {
@@ -219,7 +220,7 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     for (int r = 0; r < 16; ++r) {
       const int row = (r & 3) + 8 * (r >> 2) + 4 * half;  // q row in wave tile
       const int qrow = q0 + row;
-      if (qrow < S) {
+      if (qrow < Sq) {
         const float inv_l = bcast[wid * 32 + row];
         out[q_base + (long)qrow * q_row_stride + t * 32 + col] = f2bf(o_acc[t][r] * inv_l);
       }
@@ -229,17 +230,18 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
 
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                                   const at::Tensor& v, double scale,
-                                                  bool causal) {
+                                                  bool causal, int64_t q_start) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.scalar_type() == at::kBFloat16,
               "flash_attn_fwd: q must be [B,S,Hq,D] bf16");
-  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Hk = k.size(2);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hk = k.size(2);
   TORCH_CHECK(D == 128, "flash_attn_fwd: only D=128 supported, got ", D);
   TORCH_CHECK(Hq % Hk == 0, "flash_attn_fwd: Hq must be divisible by Hk");
-  TORCH_CHECK(S % (FA_WAVES * QBLK) == 0, "flash_attn_fwd: S must be a multiple of 128");
+  TORCH_CHECK(Sq % (FA_WAVES * QBLK) == 0, "flash_attn_fwd: Sq must be a multiple of 128");
+  TORCH_CHECK(Skv % KVBLK == 0, "flash_attn_fwd: Skv must be a multiple of 32");
   auto out = at::empty_like(q);
-  auto lse = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
-  const dim3 grid(S / (FA_WAVES * QBLK), Hq, B);
+  auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
+  const dim3 grid(Sq / (FA_WAVES * QBLK), Hq, B);
   const size_t smem = 2 * KVBLK * 128 * 2 + FA_WAVES * 32 * sizeof(float);
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL((flash_fwd_kernel<128>), grid, dim3(FA_BLOCK), smem, stream.stream(),
@@ -247,7 +249,7 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at:
                      reinterpret_cast<const bf16*>(k.data_ptr()),
                      reinterpret_cast<const bf16*>(v.data_ptr()),
                      reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
-                     B, S, Hq, Hk, (float)scale, causal);
+                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
   HIP_CHECK_KERNEL();
   return {out, lse};
 }
@@ -296,8 +298,8 @@ template <int D>
 __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dk, float* __restrict__ dv, int B, int S, int Hq, int Hk,
-    float scale, bool causal) {
+    float* __restrict__ dk, float* __restrict__ dv, int B, int Sq, int Skv, int Hq,
+    int Hk, int q_start, float scale, bool causal) {
   static_assert(D == 128);
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* q_rows = smem;                 // 32*256 = 8 KiB
@@ -318,11 +320,11 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
   const int kvh = h / (Hq / Hk);
   const int kv0 = kvb + wid * 32;      // this wave's kv rows
 
-  const long q_base = (((long)b * S) * Hq + h) * D;
-  const long kv_base = (((long)b * S) * Hk + kvh) * D;
+  const long q_base = (((long)b * Sq) * Hq + h) * D;
+  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;
   const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
-  const float* lse_row = lse + ((long)b * Hq + h) * S;
-  const float* dlt_row = delta + ((long)b * Hq + h) * S;
+  const float* lse_row = lse + ((long)b * Hq + h) * Sq;
+  const float* dlt_row = delta + ((long)b * Hq + h) * Sq;
 
   // K/V rows of this wave -> A fragments in registers
   bf16x8_v kfrag[D / 16], vfrag[D / 16];
@@ -340,9 +342,9 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
     for (int r = 0; r < 16; ++r) { dk_acc[t][r] = 0.f; dv_acc[t][r] = 0.f; }
 
   char* tb = tbuf + wid * 2048;
-  const int jq_start = causal ? kvb / 32 : 0;
+  const int jq_start = causal ? (kvb > q_start ? (kvb - q_start) / 32 : 0) : 0;
 
-  for (int jq = jq_start; jq < S / 32; ++jq) {
+  for (int jq = jq_start; jq < Sq / 32; ++jq) {
     const int q0 = jq * 32;
     // ---- cooperative stage: q/do rows + transposed images
     {
@@ -361,11 +363,12 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
     }
     __syncthreads();
 
-    const bool live = !causal || (q0 + 31 >= kv0);
+    const bool live = !causal || (q_start + q0 + 31 >= kv0);
     if (live) {
-      const int qg = q0 + col;
-      const float lse_q = lse_row[qg];
-      const float dlt_q = dlt_row[qg];
+      const int ql = q0 + col;          // local q row (lse/delta index)
+      const int qg = q_start + ql;      // global position (mask)
+      const float lse_q = lse_row[ql];
+      const float dlt_q = dlt_row[ql];
 
       // ---- S^T = K Q^T (D rows = kv, cols = q)
       f32x16 st;
@@ -392,7 +395,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
-        const bool masked = (causal && kg > qg) || kg >= S;
+        const bool masked = (causal && kg > qg) || kg >= Skv;
         const float pv = masked ? 0.f : __expf(st[r] * scale - lse_q);
         st[r] = pv;
         dpt[r] = pv * (dpt[r] - dlt_q) * scale;
@@ -446,7 +449,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel
     for (int r = 0; r < 16; ++r) {
       const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
       const int kg = kv0 + krow;
-      if (kg < S) {
+      if (kg < Skv) {
         const long off = kv_base + (long)kg * kv_rs + t * 32 + col;
         atomicAdd(dk + off, dk_acc[t][r]);
         atomicAdd(dv + off, dv_acc[t][r]);
@@ -459,7 +462,8 @@ template <int D>
 __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
-    bf16* __restrict__ dq, int B, int S, int Hq, int Hk, float scale, bool causal) {
+    bf16* __restrict__ dq, int B, int Sq, int Skv, int Hq, int Hk, int q_start,
+    float scale, bool causal) {
   static_assert(D == 128);
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_rows = smem;                 // 8 KiB ([32][256B] swz)
@@ -480,11 +484,11 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
   const int kvh = h / (Hq / Hk);
   const int q0 = qb_blk + wid * 32;
 
-  const long q_base = (((long)b * S) * Hq + h) * D;
-  const long kv_base = (((long)b * S) * Hk + kvh) * D;
+  const long q_base = (((long)b * Sq) * Hq + h) * D;
+  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;
   const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
-  const float* lse_row = lse + ((long)b * Hq + h) * S;
-  const float* dlt_row = delta + ((long)b * Hq + h) * S;
+  const float* lse_row = lse + ((long)b * Hq + h) * Sq;
+  const float* dlt_row = delta + ((long)b * Hq + h) * Sq;
 
   // Q/dO rows of this wave as A-fragments
   bf16x8_v qfrag[D / 16], dofrag[D / 16];
@@ -507,8 +511,8 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
   char* tb = tbuf + wid * 2048;
-  const int block_q_max = qb_blk + FAB_WAVES * 32 - 1;
-  const int n_tiles = causal ? (min(block_q_max, S - 1) / 32 + 1) : (S + 31) / 32;
+  const int block_q_max = q_start + qb_blk + FAB_WAVES * 32 - 1;
+  const int n_tiles = causal ? (min(block_q_max, Skv - 1) / 32 + 1) : (Skv + 31) / 32;
 
   for (int jk = 0; jk < n_tiles; ++jk) {
     const int k0 = jk * 32;
@@ -516,8 +520,9 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     {
       const int r0 = tid / 16, c0 = (tid % 16) * 8;
       for (int rr = r0; rr < 32; rr += 16) {
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)(k0 + rr) * kv_rs + c0);
-        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)(k0 + rr) * kv_rs + c0);
+        const int ks = min(k0 + rr, Skv - 1);  // overhang masked in compute
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_rs + c0);
+        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)ks * kv_rs + c0);
         *reinterpret_cast<bf16x8*>(k_rows + k_lds_off(rr, c0 * 2)) = kv8;
         *reinterpret_cast<bf16x8*>(v_rows + k_lds_off(rr, c0 * 2)) = vv8;
 #pragma unroll
@@ -527,7 +532,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     }
     __syncthreads();
 
-    const bool live = !causal || (k0 <= q0 + 31);
+    const bool live = !causal || (k0 <= q_start + q0 + 31);
     if (live) {
       // ---- S[q][k] = Q K^T : A=Q rows, B from k_rows (contiguous d)
       f32x16 s;
@@ -554,8 +559,8 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;  // block-local + wid*32
-        const int qg = q0 + qrow;
-        const bool masked = (causal && kg > qg) || kg >= S;
+        const int qg = q_start + q0 + qrow;
+        const bool masked = (causal && kg > qg) || kg >= Skv;
         const float p = masked ? 0.f : __expf(s[r] * scale - stats[wid * 32 + qrow]);
         s[r] = p * (dp[r] - stats[128 + wid * 32 + qrow]) * scale;  // dS in place
       }
@@ -589,7 +594,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
-      if (qrow < S) {
+      if (qrow < Sq) {
         dq[q_base + (long)qrow * q_rs + t * 32 + col] = f2bf(dq_acc[t][r]);
       }
     }
@@ -598,29 +603,31 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal) {
-  const int B = q.size(0), S = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Hk = k.size(2);
-  TORCH_CHECK(D == 128 && S % 128 == 0, "flash_attn_bwd: D=128, S%128==0 required");
+    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal,
+    int64_t q_start) {
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
+  const int Skv = k.size(1), Hk = k.size(2);
+  TORCH_CHECK(D == 128 && Sq % 128 == 0 && Skv % 128 == 0,
+              "flash_attn_bwd: D=128, Sq%128==0, Skv%128==0 required");
   auto stream = c10::hip::getCurrentHIPStream();
 
-  auto delta = at::empty({B, Hq, S}, q.options().dtype(at::kFloat));
+  auto delta = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
   {
-    const long rows = (long)B * S * Hq;
+    const long rows = (long)B * Sq * Hq;
     const int block = 256;
     const long grid = (rows * WAVE_SIZE + block - 1) / block;
     hipLaunchKernelGGL(fa_delta_kernel, dim3((unsigned)grid), dim3(block), 0, stream.stream(),
                        reinterpret_cast<const bf16*>(dout.data_ptr()),
                        reinterpret_cast<const bf16*>(o.data_ptr()),
-                       delta.data_ptr<float>(), S, Hq, rows);
+                       delta.data_ptr<float>(), Sq, Hq, rows);
     HIP_CHECK_KERNEL();
   }
 
   auto dq = at::empty_like(q);
-  auto dk32 = at::zeros({B, S, Hk, D}, q.options().dtype(at::kFloat));
-  auto dv32 = at::zeros({B, S, Hk, D}, q.options().dtype(at::kFloat));
+  auto dk32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
+  auto dv32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
 
-  const dim3 grid_kv(S / 128, Hq, B);
+  const dim3 grid_kv(Skv / 128, Hq, B);
   const size_t smem_a = 40 * 1024;
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<128>), grid_kv, dim3(256), smem_a, stream.stream(),
                      reinterpret_cast<const bf16*>(dout.data_ptr()),
@@ -629,10 +636,10 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
                      reinterpret_cast<const bf16*>(v.data_ptr()),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-                     B, S, Hq, Hk, (float)scale, causal);
+                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
   HIP_CHECK_KERNEL();
 
-  const dim3 grid_q(S / 128, Hq, B);
+  const dim3 grid_q(Sq / 128, Hq, B);
   const size_t smem_b = 33 * 1024;
   hipLaunchKernelGGL((flash_bwd_dq_kernel<128>), grid_q, dim3(256), smem_b, stream.stream(),
                      reinterpret_cast<const bf16*>(dout.data_ptr()),
@@ -641,7 +648,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
                      reinterpret_cast<const bf16*>(v.data_ptr()),
                      lse.data_ptr<float>(), delta.data_ptr<float>(),
                      reinterpret_cast<bf16*>(dq.data_ptr()),
-                     B, S, Hq, Hk, (float)scale, causal);
+                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
   HIP_CHECK_KERNEL();
 
   return {dq, dk32.to(at::kBFloat16), dv32.to(at::kBFloat16)};

@@ -25,10 +25,11 @@ void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor master, at::Tensor
 
 std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at::Tensor& k,
                                                   const at::Tensor& v, double scale,
-                                                  bool causal);
+                                                  bool causal, int64_t q_start);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal);
+    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal,
+    int64_t q_start);
 
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b);
 

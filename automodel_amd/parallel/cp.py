@@ -1,0 +1,144 @@
+"""Context parallelism: zigzag sequence sharding + all-gather-KV attention.
+
+Reference behavior (SURVEY §2.4 CP rows): the reference ships five CP
+mechanisms; the canonical MI355X one here is all-gather KV over the cp group
+(RCCL all-gather over xGMI — for a single node the 7 direct links make one
+large all-gather cheaper than per-step ring P2P) with zigzag (2-chunk
+round-robin) load balancing exactly like ContextParallelSharder's "striped"
+layout (context_parallel/sharder.py:116-143).
+
+Each rank holds 2 chunks of S/(2P): chunk r and chunk 2P-1-r, so causal work
+is balanced. Attention runs the in-tree flash kernel with ``q_start`` offsets
+against the gathered KV; backward reduce-scatters dK/dV to their owners.
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+import torch
+import torch.distributed as dist
+
+from automodel_amd.ops.attention import flash_attention
+
+
+@dataclass
+class CPContext:
+    group: object
+    rank: int
+    world: int
+
+
+_ACTIVE_CP: CPContext | None = None
+
+
+def enable_cp(mesh_axis) -> CPContext:
+    global _ACTIVE_CP
+    group = mesh_axis.get_group() if hasattr(mesh_axis, "get_group") else mesh_axis
+    _ACTIVE_CP = CPContext(group=group, rank=dist.get_rank(group),
+                           world=dist.get_world_size(group))
+    return _ACTIVE_CP
+
+
+def disable_cp() -> None:
+    global _ACTIVE_CP
+    _ACTIVE_CP = None
+
+
+def active_cp() -> CPContext | None:
+    return _ACTIVE_CP
+
+
+def zigzag_chunk_ids(rank: int, world: int) -> tuple[int, int]:
+    return rank, 2 * world - 1 - rank
+
+
+def shard_batch_cp(batch: dict, rank: int, world: int, seq_dim: int = 1) -> dict:
+    """Zigzag-shard input_ids/labels along seq; adds global position_ids."""
+    if world == 1:
+        return batch
+    out = dict(batch)
+    g0, g1 = zigzag_chunk_ids(rank, world)
+    for key in ("input_ids", "labels", "attention_mask"):
+        if key in batch and isinstance(batch[key], torch.Tensor):
+            t = batch[key]
+            S = t.shape[seq_dim]
+            assert S % (2 * world) == 0, f"S={S} not divisible by 2*cp={2*world}"
+            C = S // (2 * world)
+            c0 = t.narrow(seq_dim, g0 * C, C)
+            c1 = t.narrow(seq_dim, g1 * C, C)
+            out[key] = torch.cat([c0, c1], dim=seq_dim)
+    S = batch["input_ids"].shape[seq_dim]
+    C = S // (2 * world)
+    pos = torch.cat([
+        torch.arange(g0 * C, (g0 + 1) * C),
+        torch.arange(g1 * C, (g1 + 1) * C),
+    ])
+    B = batch["input_ids"].shape[0]
+    out["position_ids"] = pos.unsqueeze(0).expand(B, -1).contiguous()
+    return out
+
+
+class _GatherSeqZigzag(torch.autograd.Function):
+    """All-gather along seq and reorder zigzag chunks to global order.
+    Backward: inverse reorder + reduce-scatter (sum) back to owners."""
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, cp: CPContext):
+        ctx.cp = cp
+        P = cp.world
+        B, S2, *rest = x.shape
+        C = S2 // 2
+        parts = [torch.empty_like(x) for _ in range(P)]
+        dist.all_gather(parts, x.contiguous(), group=cp.group)
+        stacked = torch.stack(parts)                      # [P, B, 2C, ...]
+        chunks = stacked.view(P, B, 2, C, *rest)
+        # global chunk g -> (rank, local slot)
+        order = []
+        for g in range(2 * P):
+            r, slot = (g, 0) if g < P else (2 * P - 1 - g, 1)
+            order.append(chunks[r, :, slot])
+        return torch.cat(order, dim=1)                    # [B, 2P*C, ...]
+
+    @staticmethod
+    def backward(ctx, grad):
+        cp = ctx.cp
+        P = cp.world
+        B, S, *rest = grad.shape
+        C = S // (2 * P)
+        g_chunks = grad.view(B, 2 * P, C, *rest)
+        # rebuild per-rank zigzag layout [P, B, 2C, ...]
+        per_rank = []
+        for r in range(P):
+            g0, g1 = zigzag_chunk_ids(r, P)
+            per_rank.append(torch.cat([g_chunks[:, g0], g_chunks[:, g1]], dim=1))
+        flat = torch.stack(per_rank).contiguous()         # [P, B, 2C, ...]
+        if dist.get_backend(cp.group) == "gloo":
+            dist.all_reduce(flat, group=cp.group)
+            return flat[cp.rank], None
+        out = torch.empty_like(flat[0])
+        dist.reduce_scatter_tensor(out, flat, group=cp.group)
+        return out, None
+
+
+def cp_flash_attention(q, k, v, causal: bool = True, scale: float | None = None,
+                       backend: str = "hip") -> torch.Tensor:
+    """q/k/v local zigzag shards [B, 2C, H, D]; returns local O shard."""
+    cp = _ACTIVE_CP
+    assert cp is not None, "cp_flash_attention called without enable_cp"
+    P = cp.world
+    B, S2 = q.shape[0], q.shape[1]
+    C = S2 // 2
+    kg = _GatherSeqZigzag.apply(k, cp)
+    vg = _GatherSeqZigzag.apply(v, cp)
+    g0, g1 = zigzag_chunk_ids(cp.rank, P)
+    outs = []
+    for local_slot, g in ((0, g0), (1, g1)):
+        qc = q.narrow(1, local_slot * C, C)
+        kv_len = (g + 1) * C if causal else 2 * P * C
+        outs.append(
+            flash_attention(qc, kg.narrow(1, 0, kv_len), vg.narrow(1, 0, kv_len),
+                            causal=causal, scale=scale, backend=backend,
+                            q_start=g * C)
+        )
+    return torch.cat(outs, dim=1)

@@ -56,6 +56,10 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             pp=dist_cfg.get("pp", 1),
             cp=dist_cfg.get("cp", 1),
         )
+        self.cp_size = self.mesh.dims.get("cp", 1)
+        if self.cp_size > 1:
+            from automodel_amd.parallel.cp import enable_cp
+            enable_cp(self.mesh["cp"])
 
         # ---- model (meta init -> shard -> materialize/load)
         mcfg = cfg.model
@@ -80,10 +84,17 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         )
         self.model.loss_fn = self.loss_fn
 
-        if self.mesh.mesh is not None and self.mesh.dims["dp_shard"] > 1:
+        if self.mesh.mesh is not None and self.mesh.dims["tp"] > 1:
+            from automodel_amd.parallel.tp import apply_tp
+            apply_tp(self.model, self.mesh["tp"],
+                     sequence_parallel=dist_cfg.get("sequence_parallel", False))
+        if self.mesh.mesh is not None and (
+            self.mesh.dims["dp_shard"] > 1 or self.cp_size > 1
+        ):
+            fsdp_axis = "dp_shard_cp" if self.cp_size > 1 else "dp_shard"
             apply_fsdp(
                 self.model,
-                self.mesh["dp_shard"],
+                self.mesh[fsdp_axis],
                 reshard_after_forward=dist_cfg.get("reshard_after_forward", False),
             )
 
@@ -153,6 +164,13 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         if restore:
             self.load_checkpoint(restore)
 
+    def _dp_cp_group(self):
+        if self.mesh.mesh is None:
+            return None
+        if self.cp_size > 1:
+            return self.mesh.mesh["dp_cp"].get_group()
+        return self.mesh.dp_group()
+
     def _build_loader(self, dcfg: ConfigNode):
         ds_cfg = dcfg.get("dataset", ConfigNode())
         if "_target_" in ds_cfg:
@@ -174,9 +192,16 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
 
     # ------------------------------------------------------------- train step
     def _forward_backward_step(self, batch: dict, loss_scale: float) -> torch.Tensor:
+        if self.cp_size > 1:
+            from automodel_amd.parallel.cp import shard_batch_cp
+            cp_rank = self.mesh["cp"].get_local_rank()
+            batch = shard_batch_cp(batch, cp_rank, self.cp_size)
         input_ids = batch["input_ids"].to(self.device, non_blocking=True)
         labels = batch["labels"].to(self.device, non_blocking=True)
-        loss = self.model(input_ids, labels=labels)
+        position_ids = batch.get("position_ids")
+        if position_ids is not None:
+            position_ids = position_ids.to(self.device, non_blocking=True)
+        loss = self.model(input_ids, labels=labels, position_ids=position_ids)
         (loss * loss_scale).backward()
         return loss.detach()
 
@@ -188,11 +213,14 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             num_label_tokens += count_label_tokens(b["labels"])
         num_label_tokens = num_label_tokens.to(device)
         if self.world > 1:
+            # counted on the UNSHARDED batch (cp ranks share it): reduce over dp
             dist.all_reduce(num_label_tokens, group=self.mesh.dp_group())
         global_tokens = max(1, int(num_label_tokens.item()))
 
-        # loss_sum / global_tokens * dp_world compensates FSDP's mean-reduce
-        loss_scale = self.mesh.dp_size / global_tokens
+        # loss_sum / global_tokens * dp_cp_world compensates FSDP's mean-reduce
+        # (with CP, each rank sees a token subset and FSDP reduces over
+        # dp_shard_cp — reference train_ft.py:1186 scales by dp_cp_size)
+        loss_scale = self.mesh.dp_cp_size / global_tokens
         total_loss = torch.zeros((), dtype=torch.float32, device=device)
         for i, batch in enumerate(batches):
             prepare_for_grad_accumulation(self.model, is_final_microbatch=(i == len(batches) - 1))
@@ -204,7 +232,7 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         self.lr_scheduler.step()
 
         if self.world > 1:
-            dist.all_reduce(total_loss, group=self.mesh.dp_group())
+            dist.all_reduce(total_loss, group=self._dp_cp_group())
         step_time = time.perf_counter() - t0
         ntok = int(num_label_tokens.item())
         return {

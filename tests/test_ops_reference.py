@@ -1,8 +1,6 @@
 """CPU checks of the op reference implementations (the HIP kernels are
 parity-tested against these same references in test_gpu_kernels.py)."""
 
-import math
-
 import pytest
 import torch
 
@@ -89,33 +87,3 @@ def test_masked_ce_sum_semantics():
         logits.reshape(-1, 50).float(), labels.reshape(-1),
         ignore_index=-100, reduction="sum")
     assert torch.allclose(loss, ref)
-
-
-def test_attention_composite_bwd_matches_autograd():
-    """The GEMM-composite backward must match autograd through the eager ref."""
-    from automodel_amd.ops.attention import _attention_bwd_composite
-
-    torch.manual_seed(3)
-    B, S, Hq, Hk, D = 1, 32, 4, 2, 16
-    q = torch.randn(B, S, Hq, D, requires_grad=True)
-    k = torch.randn(B, S, Hk, D, requires_grad=True)
-    v = torch.randn(B, S, Hk, D, requires_grad=True)
-    o = attention_ref(q, k, v, causal=True)
-    do = torch.randn_like(o)
-    o.backward(do)
-
-    # lse from the reference path
-    scale = 1.0 / math.sqrt(D)
-    qf = q.detach().permute(0, 2, 1, 3)
-    kf = k.detach().permute(0, 2, 1, 3).repeat_interleave(2, dim=1)
-    scores = (qf @ kf.transpose(-1, -2)) * scale
-    mask = torch.ones(S, S, dtype=torch.bool).tril()
-    scores = scores.masked_fill(~mask, float("-inf"))
-    lse = torch.logsumexp(scores.float(), dim=-1)  # [B,Hq,S]
-
-    dq, dk, dv = _attention_bwd_composite(
-        do, q.detach(), k.detach(), v.detach(), o.detach(), lse,
-        causal=True, scale=scale, q_chunk=8)
-    assert torch.allclose(dq, q.grad, atol=1e-4), (dq - q.grad).abs().max()
-    assert torch.allclose(dk, k.grad, atol=1e-4), (dk - k.grad).abs().max()
-    assert torch.allclose(dv, v.grad, atol=1e-4), (dv - v.grad).abs().max()
