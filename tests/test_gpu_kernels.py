@@ -166,6 +166,30 @@ def test_flash_attention_bwd_grads():
         (v.grad.float() - v2.grad).abs().max()
 
 
+def test_flash_attention_q_start_parity():
+    """CP building block: q chunk attending a longer KV prefix with offset."""
+    from automodel_amd.ops.attention import attention_ref, flash_attention
+
+    torch.manual_seed(2)
+    B, Sq, Skv, Hq, Hk, D = 1, 128, 384, 4, 2, 128
+    q = torch.randn(B, Sq, Hq, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, Skv, Hk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, Skv, Hk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    o = flash_attention(q, k, v, causal=True, backend="hip", q_start=256)
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    o_ref = attention_ref(q2, k2, v2, causal=True, q_start=256)
+    assert torch.allclose(o.float(), o_ref, atol=3e-2, rtol=3e-2), \
+        (o.float() - o_ref).abs().max()
+    do = torch.randn_like(o)
+    o.backward(do)
+    o_ref.backward(do.float())
+    assert torch.allclose(q.grad.float(), q2.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(k.grad.float(), k2.grad, atol=5e-2, rtol=5e-2)
+    assert torch.allclose(v.grad.float(), v2.grad, atol=5e-2, rtol=5e-2)
+
+
 def test_hybrid_linear_ce_parity():
     """hybrid (hipBLASLt GEMM + HIP CE epilogue) vs fp32 dense reference."""
     from automodel_amd.loss.linear_ce import fused_linear_cross_entropy
