@@ -48,6 +48,13 @@ def _ensure_builtin() -> None:
             _REGISTRY[name] = MoEForCausalLM
     except ImportError:
         pass
+    try:
+        from automodel_amd.models.vlm.model import VLMForConditionalGeneration
+
+        for name in VLMForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = VLMForConditionalGeneration
+    except ImportError:
+        pass
 
 
 def build_model(
