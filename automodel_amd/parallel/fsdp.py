@@ -22,12 +22,18 @@ from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
 def apply_fsdp(
     model: nn.Module,
     mesh: DeviceMesh,
-    layer_cls_names: tuple[str, ...] = ("LlamaDecoderLayer",),
+    layer_cls_names: tuple[str, ...] = ("LlamaDecoderLayer", "MoEDecoderLayer",
+                                        "VisionBlock"),
     param_dtype: torch.dtype = torch.bfloat16,
     reduce_dtype: torch.dtype = torch.float32,
     reshard_after_forward: bool = False,
 ) -> nn.Module:
-    """Shard each decoder layer, then the root; wire backward prefetch."""
+    """Shard each decoder layer, then the root; wire backward prefetch.
+
+    ``mesh`` may be 1-D (pure FSDP over dp_shard / dp_shard_cp) or 2-D
+    (dp_replicate, dp_shard) for HSDP — fully_shard then replicates grads
+    over the first axis with all-reduce (reference HSDP via dp_replicate).
+    """
     mp = MixedPrecisionPolicy(param_dtype=param_dtype, reduce_dtype=reduce_dtype)
     layers = [
         m for m in model.modules() if type(m).__name__ in layer_cls_names

@@ -84,6 +84,31 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         )
         self.model.loss_fn = self.loss_fn
 
+        # ---- activation checkpointing (before sharding)
+        ac_cfg = cfg.get("activation_checkpointing")
+        if ac_cfg and ac_cfg.get("enabled", True):
+            from automodel_amd.parallel.activation_checkpointing import apply_ac
+            apply_ac(self.model, mode=ac_cfg.get("mode", "full"),
+                     every_n=ac_cfg.get("every_n", 1))
+
+        # ---- pipeline parallelism: split the model into this rank's stage
+        self.pipeline = None
+        if self.mesh.mesh is not None and self.mesh.pp_size > 1:
+            from automodel_amd.parallel.pp import AutoPipeline, PipelineConfig
+            assert cfg.get("step_scheduler", ConfigNode()).get("grad_acc_steps", 1) == 1, \
+                "PP handles microbatching via the schedule; use pp microbatches"
+            pp_cfg = dist_cfg.get("pipeline", ConfigNode())
+            if any(p.is_meta for p in self.model.parameters()):
+                self.model.init_weights(device="cpu")
+            self.pipeline = AutoPipeline(
+                self.model, self.mesh["pp"],
+                PipelineConfig(pp_size=self.mesh.pp_size,
+                               schedule=pp_cfg.get("schedule", "1f1b"),
+                               microbatches=pp_cfg.get("microbatches", self.mesh.pp_size)),
+                loss_fn=self.loss_fn, device=self.device,
+            )
+            self.model = self.pipeline.stage_module
+
         if self.mesh.mesh is not None and self.mesh.dims["tp"] > 1:
             from automodel_amd.parallel.tp import apply_tp
             apply_tp(self.model, self.mesh["tp"],
@@ -92,9 +117,13 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             self.mesh.dims["dp_shard"] > 1 or self.cp_size > 1
         ):
             fsdp_axis = "dp_shard_cp" if self.cp_size > 1 else "dp_shard"
+            if self.mesh.dims.get("dp_replicate", 1) > 1:
+                fsdp_mesh = self.mesh.mesh[("dp_replicate", fsdp_axis)]  # HSDP
+            else:
+                fsdp_mesh = self.mesh[fsdp_axis]
             apply_fsdp(
                 self.model,
-                self.mesh[fsdp_axis],
+                fsdp_mesh,
                 reshard_after_forward=dist_cfg.get("reshard_after_forward", False),
             )
 
@@ -103,8 +132,9 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
            pretrained and any(f.endswith(".safetensors") for f in os.listdir(pretrained)):
             from automodel_amd.checkpoint.hf_loader import load_hf_weights
             load_hf_weights(self.model, pretrained, device=self.device)
-        else:
+        elif self.pipeline is None:
             self.model.init_weights(device=self.device)
+        # (PP stages were materialized + initialized before the split)
 
         # ---- PEFT
         peft_cfg = cfg.get("peft")
@@ -191,6 +221,27 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         )
 
     # ------------------------------------------------------------- train step
+    def _pp_step(self, batches: list[dict], loss_scale: float) -> torch.Tensor:
+        """One optimizer step through the pipeline schedule (pp > 1).
+        The schedule splits the batch into microbatches; the loss is computed
+        on the last stage and broadcast for metrics (reference
+        train_ft.py:1188 PP loss broadcast)."""
+        batch = batches[0]
+        input_ids = batch["input_ids"].to(self.device, non_blocking=True)
+        labels = batch["labels"].to(self.device, non_blocking=True)
+        losses = self.pipeline.step(input_ids=input_ids, target=labels)
+        total = torch.zeros((), dtype=torch.float32, device=self.device)
+        if self.pipeline.is_last and losses:
+            total = sum(l.float() for l in losses)
+        # scale grads: schedule backwards sum-of-microbatch losses; match the
+        # non-PP loss_scale by scaling grads post-hoc
+        for p in self.model.parameters():
+            if p.grad is not None:
+                p.grad.mul_(loss_scale)
+        src = dist.get_process_group_ranks(self.mesh["pp"].get_group())[-1]
+        dist.broadcast(total, src=src, group=self.mesh["pp"].get_group())
+        return total
+
     def _forward_backward_step(self, batch: dict, loss_scale: float) -> torch.Tensor:
         if self.cp_size > 1:
             from automodel_amd.parallel.cp import shard_batch_cp
@@ -222,9 +273,12 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         # dp_shard_cp — reference train_ft.py:1186 scales by dp_cp_size)
         loss_scale = self.mesh.dp_cp_size / global_tokens
         total_loss = torch.zeros((), dtype=torch.float32, device=device)
-        for i, batch in enumerate(batches):
-            prepare_for_grad_accumulation(self.model, is_final_microbatch=(i == len(batches) - 1))
-            total_loss += self._forward_backward_step(batch, loss_scale).float()
+        if self.pipeline is not None:
+            total_loss = self._pp_step(batches, loss_scale).float()
+        else:
+            for i, batch in enumerate(batches):
+                prepare_for_grad_accumulation(self.model, is_final_microbatch=(i == len(batches) - 1))
+                total_loss += self._forward_backward_step(batch, loss_scale).float()
 
         grad_norm = clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
         self.optimizer.step()
