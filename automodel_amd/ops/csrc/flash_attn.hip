@@ -294,12 +294,8 @@ __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
 
 #define FAB_WAVES 4
 
-// Each block computes dK/dV for ONE 64-wide d-half (blockIdx.z encodes
-// (batch, d-half)): S^T/dP^T are recomputed per half, but the halved
-// accumulator footprint (64 vs 128 VGPRs) doubles occupancy -> net win
-// (profiles r1_step1: this kernel was 20% of step at 1 wave/SIMD).
 template <int D>
-__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
+__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dk, float* __restrict__ dv, int B, int Sq, int Skv, int Hq,
@@ -320,8 +316,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
 
   const int kvb = blockIdx.x * (FAB_WAVES * 32);
   const int h = blockIdx.y;
-  const int b = blockIdx.z >> 1;
-  const int d_base = (blockIdx.z & 1) * (D / 2);
+  const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
   const int kv0 = kvb + wid * 32;      // this wave's kv rows
 
@@ -340,9 +335,9 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
     vfrag[c] = *reinterpret_cast<const bf16x8_v*>(v + off);
   }
 
-  f32x16 dk_acc[D / 64], dv_acc[D / 64];
+  f32x16 dk_acc[D / 32], dv_acc[D / 32];
 #pragma unroll
-  for (int t = 0; t < D / 64; ++t)
+  for (int t = 0; t < D / 32; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) { dk_acc[t][r] = 0.f; dv_acc[t][r] = 0.f; }
 
@@ -418,9 +413,9 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
         bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
             tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-        for (int t = 0; t < D / 64; ++t) {
+        for (int t = 0; t < D / 32; ++t) {
           bf16x8_v dob = *reinterpret_cast<const bf16x8_v*>(
-              dot + vt_lds_off(d_base + t * 32 + col, (c2 * 16 + half * 8) * 2));
+              dot + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
           dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
         }
       }
@@ -437,9 +432,9 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
         bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
             tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-        for (int t = 0; t < D / 64; ++t) {
+        for (int t = 0; t < D / 32; ++t) {
           bf16x8_v qb2 = *reinterpret_cast<const bf16x8_v*>(
-              qt + vt_lds_off(d_base + t * 32 + col, (c2 * 16 + half * 8) * 2));
+              qt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
           dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
         }
       }
@@ -449,13 +444,13 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
 
   // ---- epilogue: atomic accumulate into f32 dk/dv [B,S,Hk,D]
 #pragma unroll
-  for (int t = 0; t < D / 64; ++t) {
+  for (int t = 0; t < D / 32; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
       const int kg = kv0 + krow;
       if (kg < Skv) {
-        const long off = kv_base + (long)kg * kv_rs + d_base + t * 32 + col;
+        const long off = kv_base + (long)kg * kv_rs + t * 32 + col;
         atomicAdd(dk + off, dk_acc[t][r]);
         atomicAdd(dv + off, dv_acc[t][r]);
       }
@@ -632,7 +627,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
   auto dk32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
   auto dv32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
 
-  const dim3 grid_kv(Skv / 128, Hq, B * 2);  // z = (batch, d-half)
+  const dim3 grid_kv(Skv / 128, Hq, B);
   const size_t smem_a = 40 * 1024;
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<128>), grid_kv, dim3(256), smem_a, stream.stream(),
                      reinterpret_cast<const bf16*>(dout.data_ptr()),
