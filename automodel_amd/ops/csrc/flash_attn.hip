@@ -295,7 +295,7 @@ __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
 #define FAB_WAVES 4
 
 template <int D>
-__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 1) void flash_bwd_dkv_kernel(
+__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dk, float* __restrict__ dv, int B, int Sq, int Skv, int Hq,
