@@ -30,6 +30,28 @@ class MockIterableDataset(IterableDataset):
             n += 1
 
 
+class MockClassificationDataset(Dataset):
+    """Sequence-classification samples: tokens + one class id."""
+
+    def __init__(self, num_samples: int = 64, seq_len: int = 32, vocab_size: int = 1000,
+                 num_labels: int = 2, seed: int = 0):
+        self.num_samples = num_samples
+        self.seq_len = seq_len
+        self.vocab_size = vocab_size
+        self.num_labels = num_labels
+        self.seed = seed
+
+    def __len__(self):
+        return self.num_samples
+
+    def __getitem__(self, idx: int):
+        g = torch.Generator().manual_seed(self.seed + idx)
+        return {
+            "input_ids": torch.randint(0, self.vocab_size, (self.seq_len,), generator=g),
+            "labels": torch.randint(0, self.num_labels, (1,), generator=g),
+        }
+
+
 class MockDataset(Dataset):
     """Finite map-style mock dataset (deterministic per index)."""
 
