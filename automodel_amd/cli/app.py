@@ -22,6 +22,8 @@ RECIPE_ALIASES = {
     "llm_pretrain": "automodel_amd.recipes.llm.train_ft.TrainFinetuneRecipeForNextTokenPrediction",
     "llm_benchmark": "automodel_amd.recipes.llm.benchmark.BenchmarkingRecipeForNextTokenPrediction",
     "vlm_finetune": "automodel_amd.recipes.vlm.finetune.FinetuneRecipeForVLM",
+    "llm_kd": "automodel_amd.recipes.llm.kd.KDRecipeForNextTokenPrediction",
+    "llm_seq_cls": "automodel_amd.recipes.llm.train_seq_cls.TrainFinetuneRecipeForSequenceClassification",
 }
 
 

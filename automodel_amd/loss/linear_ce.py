@@ -104,7 +104,7 @@ class _HybridLinearCE(torch.autograd.Function):
             h = hidden[s : s + chunk_size]
             y = labels[s : s + chunk_size]
             logits = h @ wt                       # recompute (bf16 GEMM)
-            ops.ce_bwd_logits(logits, y, lse_all[s : s + chunk_size], d)
+            ops.ce_bwd_logits(logits, y, lse_all[s : s + chunk_size], d, 0)
             dh[s : s + chunk_size] = logits @ weight
             dw += (logits.t() @ h).float()
         return dh, dw.to(weight.dtype), None, None

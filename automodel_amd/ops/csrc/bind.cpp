@@ -37,8 +37,10 @@ TORCH_LIBRARY(amd_ops, m) {
 
   m.def("ce_fwd_logits(Tensor logits, Tensor labels, Tensor(a!) loss_sum) -> (Tensor, Tensor)");
   m.impl("ce_fwd_logits", &amd_ops::ce_fwd_logits);
-  m.def("ce_bwd_logits(Tensor(a!) logits, Tensor labels, Tensor lse, Tensor dloss) -> ()");
+  m.def("ce_bwd_logits(Tensor(a!) logits, Tensor labels, Tensor lse, Tensor dloss, int vocab_offset=0) -> ()");
   m.impl("ce_bwd_logits", &amd_ops::ce_bwd_logits);
+  m.def("ce_stats_logits(Tensor logits) -> (Tensor, Tensor)");
+  m.impl("ce_stats_logits", &amd_ops::ce_stats_logits);
 
   m.def("fused_ce_fwd(Tensor hidden, Tensor weight, Tensor labels) -> (Tensor, Tensor)");
   m.impl("fused_ce_fwd", &amd_ops::fused_ce_fwd);

@@ -72,18 +72,54 @@ __global__ void ce_fwd_logits_kernel(const bf16* __restrict__ logits,
   }
 }
 
+// per-row online (max, sumexp) stats only — vocab-parallel CE exchanges
+// (m, d, X_y) across TP ranks (reference loss/te_parallel_ce.py:45-191).
+__global__ void ce_stats_logits_kernel(const bf16* __restrict__ logits,
+                                       float* __restrict__ m_out,
+                                       float* __restrict__ s_out, int V) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* red = reinterpret_cast<float*>(smem);
+  const long row = blockIdx.x;
+  const bf16* x = logits + row * (long)V;
+  float m = -1e30f, s = 0.f;
+  const int stride = blockDim.x * 8;
+  for (int i = threadIdx.x * 8; i + 7 < V; i += stride) {
+    bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + i);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float f = bf2f(xv.v[j]);
+      if (f > m) { s *= __expf(m - f); m = f; }
+      s += __expf(f - m);
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    combine_ms(m, s, __shfl_xor(m, off), __shfl_xor(s, off));
+  const int wid = threadIdx.x / WAVE_SIZE;
+  if ((threadIdx.x & 63) == 0) { red[2 * wid] = m; red[2 * wid + 1] = s; }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float M = red[0], S = red[1];
+    for (int w = 1; w < blockDim.x / WAVE_SIZE; ++w)
+      combine_ms(M, S, red[2 * w], red[2 * w + 1]);
+    m_out[row] = M;
+    s_out[row] = S;
+  }
+}
+
 // grad in place: g = (exp(x - lse) - onehot) * dloss ; ignored rows -> 0.
 __global__ void ce_bwd_logits_kernel(bf16* __restrict__ logits,
                                      const long* __restrict__ labels,
                                      const float* __restrict__ lse,
                                      const float* __restrict__ dloss,
-                                     int V, long ignore_index) {
+                                     int V, long ignore_index, long vocab_offset) {
   const long row = blockIdx.y;
-  const long y = labels[row];
+  const long y_raw = labels[row];
+  const long y = (y_raw == ignore_index) ? ignore_index : y_raw - vocab_offset;
   const float d = dloss[0];
   const float l = lse[row];
   bf16* x = logits + row * (long)V;
-  const bool ignored = (y == ignore_index);
+  const bool ignored = (y_raw == ignore_index);
   for (int i = (blockIdx.x * blockDim.x + threadIdx.x) * 8; i < V;
        i += gridDim.x * blockDim.x * 8) {
     bf16x8 xv = *reinterpret_cast<bf16x8*>(x + i);
@@ -121,7 +157,7 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd_logits(const at::Tensor& logits,
 }
 
 void ce_bwd_logits(at::Tensor logits, const at::Tensor& labels, const at::Tensor& lse,
-                   const at::Tensor& dloss) {
+                   const at::Tensor& dloss, int64_t vocab_offset) {
   const long T = logits.size(0);
   const int V = logits.size(1);
   TORCH_CHECK(V % 8 == 0, "ce_bwd_logits: V must be a multiple of 8");
@@ -131,8 +167,26 @@ void ce_bwd_logits(at::Tensor logits, const at::Tensor& labels, const at::Tensor
   hipLaunchKernelGGL(ce_bwd_logits_kernel, dim3(gx, T), dim3(block), 0, stream.stream(),
                      reinterpret_cast<bf16*>(logits.data_ptr()),
                      labels.data_ptr<long>(), lse.data_ptr<float>(),
-                     dloss.data_ptr<float>(), V, -100);
+                     dloss.data_ptr<float>(), V, -100, vocab_offset);
   HIP_CHECK_KERNEL();
+}
+
+std::tuple<at::Tensor, at::Tensor> ce_stats_logits(const at::Tensor& logits) {
+  TORCH_CHECK(logits.is_cuda() && logits.dim() == 2 &&
+              logits.scalar_type() == at::kBFloat16, "ce_stats_logits: [T,V] bf16");
+  const long T = logits.size(0);
+  const int V = logits.size(1);
+  TORCH_CHECK(V % 8 == 0, "ce_stats_logits: V must be a multiple of 8");
+  auto m = at::empty({T}, logits.options().dtype(at::kFloat));
+  auto s = at::empty({T}, logits.options().dtype(at::kFloat));
+  const int block = 512;
+  const size_t smem = (block / WAVE_SIZE) * 2 * sizeof(float);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(ce_stats_logits_kernel, dim3(T), dim3(block), smem, stream.stream(),
+                     reinterpret_cast<const bf16*>(logits.data_ptr()),
+                     m.data_ptr<float>(), s.data_ptr<float>(), V);
+  HIP_CHECK_KERNEL();
+  return {m, s};
 }
 
 }  // namespace amd_ops

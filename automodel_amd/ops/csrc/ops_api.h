@@ -37,7 +37,8 @@ std::tuple<at::Tensor, at::Tensor> ce_fwd_logits(const at::Tensor& logits,
                                                  const at::Tensor& labels,
                                                  at::Tensor loss_sum);
 void ce_bwd_logits(at::Tensor logits, const at::Tensor& labels, const at::Tensor& lse,
-                   const at::Tensor& dloss);
+                   const at::Tensor& dloss, int64_t vocab_offset);
+std::tuple<at::Tensor, at::Tensor> ce_stats_logits(const at::Tensor& logits);
 
 std::tuple<at::Tensor, at::Tensor> fused_ce_fwd(const at::Tensor& hidden,
                                                 const at::Tensor& weight,

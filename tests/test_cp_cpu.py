@@ -118,3 +118,39 @@ def _cp_model_fn(rank, world):
 def test_cp2_model_loss_and_grad_parity():
     out = run_distributed(_cp_model_fn, world=2)
     assert abs(out[0] - out[1]) < 1e-4
+
+
+# -------------------------------------------------- vocab-parallel CE (TP)
+def _vp_ce_fn(rank, world):
+    import torch.distributed as dist
+
+    from automodel_amd.loss.vocab_parallel_ce import vocab_parallel_cross_entropy
+
+    torch.manual_seed(0)  # same full logits everywhere
+    T, V = 32, 64
+    logits = torch.randn(T, V)
+    labels = torch.randint(0, V, (T,))
+    labels[3] = -100
+    ref = torch.nn.functional.cross_entropy(logits.float(), labels,
+                                            ignore_index=-100, reduction="sum")
+
+    Vl = V // world
+    shard = logits[:, rank * Vl : (rank + 1) * Vl].clone().requires_grad_(True)
+    loss = vocab_parallel_cross_entropy(shard, labels, vocab_offset=rank * Vl,
+                                        group=dist.group.WORLD)
+    assert torch.allclose(loss, ref, rtol=1e-4), (float(loss), float(ref))
+    loss.backward()
+
+    l2 = logits.clone().requires_grad_(True)
+    ref2 = torch.nn.functional.cross_entropy(l2.float(), labels,
+                                             ignore_index=-100, reduction="sum")
+    ref2.backward()
+    ref_shard_grad = l2.grad[:, rank * Vl : (rank + 1) * Vl]
+    assert torch.allclose(shard.grad.float(), ref_shard_grad, atol=1e-4), \
+        (shard.grad.float() - ref_shard_grad).abs().max()
+    return float(loss)
+
+
+def test_vocab_parallel_ce_tp2():
+    out = run_distributed(_vp_ce_fn, world=2)
+    assert abs(out[0] - out[1]) < 1e-4
