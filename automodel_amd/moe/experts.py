@@ -69,8 +69,23 @@ class GroupedExperts(nn.Module):
     def forward_permuted(self, x_perm: torch.Tensor, counts: torch.Tensor) -> torch.Tensor:
         """Compute experts over tokens already sorted by expert (counts[e] each).
 
-        counts covers THIS module's experts (post-EP-shard slice).
+        counts covers THIS module's experts (post-EP-shard slice). On GPU the
+        three projections run the in-tree grouped-GEMM HIP kernel.
         """
+        use_grouped = (
+            x_perm.is_cuda and self.backend in ("auto", "hip_grouped")
+            and x_perm.dtype == torch.bfloat16
+            and self.intermediate_size % 128 == 0 and self.hidden_size % 128 == 0
+            and x_perm.numel() > 0
+        )
+        if use_grouped:
+            from automodel_amd.ops.grouped_gemm import grouped_linear
+
+            cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+            g = grouped_linear(x_perm, self.gate_proj, cl)
+            u = grouped_linear(x_perm, self.up_proj, cl)
+            h = swiglu(g, u)
+            return grouped_linear(h, self.down_proj, cl)
         return self._expert_mlp_loop(x_perm, counts)
 
     def forward(self, x: torch.Tensor, probs: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:

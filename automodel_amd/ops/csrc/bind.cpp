@@ -48,4 +48,11 @@ TORCH_LIBRARY(amd_ops, m) {
       "fused_ce_bwd(Tensor hidden, Tensor weight, Tensor labels, Tensor lse, Tensor dloss)"
       " -> (Tensor, Tensor)");
   m.impl("fused_ce_bwd", &amd_ops::fused_ce_bwd);
+
+  m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
+  m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
+  m.def("permute_gather(Tensor x, Tensor src) -> Tensor");
+  m.impl("permute_gather", &amd_ops::permute_gather);
+  m.def("unpermute_combine(Tensor yp, Tensor pos, Tensor probs) -> Tensor");
+  m.impl("unpermute_combine", &amd_ops::unpermute_combine);
 }

@@ -1,0 +1,219 @@
+// MoE kernels (CDNA4): grouped GEMM (bf16, MFMA 16x16x32) + fused token
+// permute/unpermute.
+//
+// Replaces the reference's external grouped_gemm / torch._grouped_mm expert
+// path and the fused permute/unpermute utilities (#5/#7):
+//   grouped_gemm_nt: y[m, n] = sum_k x[m, k] * w[e(m), n, k] over variable-
+//     size expert groups; one 128x128x64 LDS-tiled MFMA block per output
+//     tile, tile list precomputed host-side (expert, m0).
+//   permute_gather / unpermute_combine: gather token replicas into expert-
+//     sorted order and combine top-k outputs with routing probs without
+//     materializing the repeat_interleave copy.
+
+#include <torch/library.h>
+#include <ATen/ATen.h>
+#include <c10/hip/HIPStream.h>
+
+#include "common.h"
+#include "ops_api.h"
+
+namespace amd_ops {
+
+typedef __bf16 bf16x8v __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define GG_BM 128
+#define GG_BN 128
+#define GG_BK 64
+
+// LDS tiles: [128 rows][64 cols] bf16, 128-B rows, XOR swizzle ((row&7)<<4).
+__device__ __forceinline__ int gg_off(int row, int byte_in_row) {
+  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
+}
+
+__global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w, bf16* __restrict__ y,
+    const int* __restrict__ tile_map,   // [n_mtiles][2]: (expert, row0)
+    const int* __restrict__ offs,       // [E+1] group row offsets
+    int K, int N) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* xa = smem;                  // 16 KiB
+  char* wb = smem + GG_BM * GG_BK * 2;
+
+  const int e = tile_map[2 * blockIdx.x];
+  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int m_end = offs[e + 1];
+  const int n0 = blockIdx.y * GG_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;          // 0..3 -> k chunk of 8
+
+  const long wbase = (long)e * N * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  for (int k0 = 0; k0 < K; k0 += GG_BK) {
+    // ---- stage x tile [128 m][64 k] and w tile [128 n][64 k]
+    {
+      // 256 threads x 4 pieces: piece p -> row tid/4 + p*64? use flat:
+      // elem = (tid*4 + p) * 8 over 128*64 elems
+      const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
+      const int c0 = (tid % 8) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + rr * 32;
+        const int m = m0 + row;
+        bf16x8 xv;
+        if (m < m_end) {
+          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
+        }
+        *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xv;
+        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+            w + wbase + (long)(n0 + row) * K + k0 + c0);
+        *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wv;
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < GG_BK / 32; ++kk) {
+      bf16x8v a[4], b[4];
+      const int arow = (wid >> 1) * 64;
+      const int brow = (wid & 1) * 64;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        a[i] = *reinterpret_cast<const bf16x8v*>(
+            xa + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+        b[i] = *reinterpret_cast<const bf16x8v*>(
+            wb + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: D 16x16 layout col=l&15, row=(l>>4)*4+r
+  const int mw = m0 + (wid >> 1) * 64;
+  const int nw = n0 + (wid & 1) * 64;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mw + i * 16 + kq * 4 + r;
+      if (m < m_end) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          y[(long)m * N + nw + j * 16 + l16] = f2bf(acc[i][j][r]);
+        }
+      }
+    }
+  }
+}
+
+at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
+                           const at::Tensor& offs, const at::Tensor& tile_map) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kBFloat16,
+              "grouped_gemm_nt: x [M,K] bf16");
+  TORCH_CHECK(w.dim() == 3 && w.scalar_type() == at::kBFloat16, "w [E,N,K] bf16");
+  const long M = x.size(0);
+  const int K = x.size(1), N = w.size(1);
+  TORCH_CHECK(K == w.size(2), "K mismatch");
+  TORCH_CHECK(K % GG_BK == 0 && N % GG_BN == 0, "need K%64==0, N%128==0");
+  TORCH_CHECK(tile_map.scalar_type() == at::kInt && offs.scalar_type() == at::kInt,
+              "tile_map/offs must be int32");
+  auto y = at::empty({M, (long)N}, x.options());
+  const int n_mtiles = tile_map.size(0);
+  if (n_mtiles == 0 || M == 0) return y;
+  const dim3 grid(n_mtiles, N / GG_BN);
+  const size_t smem = 2 * GG_BM * GG_BK * 2;
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(grouped_gemm_nt_kernel, grid, dim3(256), smem, stream.stream(),
+                     reinterpret_cast<const bf16*>(x.data_ptr()),
+                     reinterpret_cast<const bf16*>(w.data_ptr()),
+                     reinterpret_cast<bf16*>(y.data_ptr()),
+                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), K, N);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+// ---- fused permute / unpermute -------------------------------------------
+// gather: y[i, :] = x[src[i], :]
+__global__ void permute_gather_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,
+                                      const int* __restrict__ src, int H, long M) {
+  const long i = blockIdx.x;
+  const bf16* xr = x + (long)src[i] * H;
+  bf16* yr = y + i * (long)H;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8)
+    *reinterpret_cast<bf16x8*>(yr + c) = *reinterpret_cast<const bf16x8*>(xr + c);
+}
+
+// combine: out[t, :] = sum_k probs[t, k] * yp[pos[t*K + k], :]
+__global__ void unpermute_combine_kernel(const bf16* __restrict__ yp,
+                                         bf16* __restrict__ out,
+                                         const int* __restrict__ pos,
+                                         const float* __restrict__ probs,
+                                         int H, int topk, long T) {
+  const long t = blockIdx.x;
+  bf16* orow = out + t * (long)H;
+  for (int c = threadIdx.x * 8; c < H; c += blockDim.x * 8) {
+    float accv[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+    for (int k = 0; k < topk; ++k) {
+      const float p = probs[t * topk + k];
+      const bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          yp + (long)pos[t * topk + k] * H + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) accv[j] += p * bf2f(v.v[j]);
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o.v[j] = f2bf(accv[j]);
+    *reinterpret_cast<bf16x8*>(orow + c) = o;
+  }
+}
+
+at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16 && x.size(1) % 8 == 0,
+              "permute_gather: bf16 [T,H], H%8==0");
+  const long M = src.size(0);
+  const int H = x.size(1);
+  auto y = at::empty({M, (long)H}, x.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(permute_gather_kernel, dim3(M), dim3(std::min(256, H / 8)), 0,
+                     stream.stream(), reinterpret_cast<const bf16*>(x.data_ptr()),
+                     reinterpret_cast<bf16*>(y.data_ptr()), src.data_ptr<int>(), H, M);
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
+at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,
+                             const at::Tensor& probs) {
+  const long T = probs.size(0);
+  const int topk = probs.size(1);
+  const int H = yp.size(1);
+  auto out = at::empty({T, (long)H}, yp.options());
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(unpermute_combine_kernel, dim3(T), dim3(std::min(256, H / 8)), 0,
+                     stream.stream(), reinterpret_cast<const bf16*>(yp.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()), pos.data_ptr<int>(),
+                     probs.data_ptr<float>(), H, topk, T);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
+}  // namespace amd_ops

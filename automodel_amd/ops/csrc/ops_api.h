@@ -49,4 +49,10 @@ std::tuple<at::Tensor, at::Tensor> fused_ce_bwd(const at::Tensor& hidden,
                                                 const at::Tensor& lse,
                                                 const at::Tensor& dloss);
 
+at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
+                           const at::Tensor& offs, const at::Tensor& tile_map);
+at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);
+at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,
+                             const at::Tensor& probs);
+
 }  // namespace amd_ops
