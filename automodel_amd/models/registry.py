@@ -49,6 +49,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.deepseek_v3.model import DeepseekV3ForCausalLM
+
+        for name in DeepseekV3ForCausalLM.hf_architectures:
+            _REGISTRY[name] = DeepseekV3ForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.vlm.model import VLMForConditionalGeneration
 
         for name in VLMForConditionalGeneration.hf_architectures:
