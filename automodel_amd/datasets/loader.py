@@ -17,7 +17,12 @@ IGNORE_INDEX = -100
 
 
 def padded_collate(batch: list[dict], pad_token_id: int = 0) -> dict:
-    """Pad variable-length samples to the batch max; labels padded with -100."""
+    """Pad variable-length samples to the batch max; labels padded with -100.
+    Packed samples (with cu_seqlens) collate into one THD batch instead."""
+    if "cu_seqlens" in batch[0]:
+        from automodel_amd.datasets.llm.packed_sequence import thd_collate
+
+        return thd_collate(batch)
     keys = batch[0].keys()
     out = {}
     max_len = max(len(b["input_ids"]) for b in batch)

@@ -77,6 +77,54 @@ class _FlashAttnHip(torch.autograd.Function):
         return dq, dk, dv, None, None, None
 
 
+_VARLEN_CU: torch.Tensor | None = None
+
+
+def set_varlen_context(cu_seqlens: torch.Tensor | None) -> None:
+    """Install cu_seqlens for packed (THD) batches — attention then applies
+    block-diagonal causal masking per document (reference THD runtime,
+    distributed/thd_utils.py:85)."""
+    global _VARLEN_CU
+    _VARLEN_CU = cu_seqlens
+
+
+def flash_attention_varlen(q, k, v, cu_seqlens: torch.Tensor, scale: float | None = None,
+                           backend: str = "hip") -> torch.Tensor:
+    """Packed-sequence attention: q/k/v [1, T, H, D], cu_seqlens int32 [n+1].
+
+    GPU path: per-document HIP flash calls with zero-padding to the kernel's
+    128-row granularity — appended pad KV sits at positions > every real q
+    position, so causal masking excludes it; pad q rows are sliced off.
+    CPU path: one sdpa call with a dense block-causal mask.
+    """
+    assert q.shape[0] == 1, "varlen expects a packed THD batch [1, T, H, D]"
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    bounds = cu_seqlens.tolist()
+    if backend == "hip" and q.is_cuda:
+        outs = []
+        for a, b in zip(bounds[:-1], bounds[1:]):
+            L = b - a
+            if L == 0:
+                continue
+            pad = (-L) % 128
+            qc, kc, vc = (t[:, a:b] for t in (q, k, v))
+            if pad:
+                qc = torch.nn.functional.pad(qc, (0, 0, 0, 0, 0, pad))
+                kc = torch.nn.functional.pad(kc, (0, 0, 0, 0, 0, pad))
+                vc = torch.nn.functional.pad(vc, (0, 0, 0, 0, 0, pad))
+            o = _FlashAttnHip.apply(qc, kc, vc, True, scale, 0)
+            outs.append(o[:, :L])
+        return torch.cat(outs, dim=1)
+    from automodel_amd.datasets.llm.packed_sequence import block_causal_mask
+
+    mask = block_causal_mask(cu_seqlens.cpu(), q.shape[1]).to(q.device)
+    qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
+    o = torch.nn.functional.scaled_dot_product_attention(
+        qt, kt, vt, attn_mask=mask, scale=scale,
+        enable_gqa=q.shape[2] != k.shape[2])
+    return o.transpose(1, 2)
+
+
 def flash_attention(
     q: torch.Tensor,
     k: torch.Tensor,
@@ -86,6 +134,8 @@ def flash_attention(
     backend: str = "hip",
     q_start: int = 0,
 ) -> torch.Tensor:
+    if _VARLEN_CU is not None and causal and q_start == 0:
+        return flash_attention_varlen(q, k, v, _VARLEN_CU, scale, backend)
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if backend == "hip" and q.is_cuda:
         return _FlashAttnHip.apply(q, k, v, causal, scale, q_start)

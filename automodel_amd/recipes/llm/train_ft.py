@@ -252,8 +252,16 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         position_ids = batch.get("position_ids")
         if position_ids is not None:
             position_ids = position_ids.to(self.device, non_blocking=True)
-        loss = self.model(input_ids, labels=labels, position_ids=position_ids)
-        (loss * loss_scale).backward()
+        cu = batch.get("cu_seqlens")
+        if cu is not None:
+            from automodel_amd.ops.attention import set_varlen_context
+            set_varlen_context(cu.to(self.device))
+        try:
+            loss = self.model(input_ids, labels=labels, position_ids=position_ids)
+            (loss * loss_scale).backward()
+        finally:
+            if cu is not None:
+                set_varlen_context(None)
         return loss.detach()
 
     def _run_train_optim_step(self, batches: list[dict]) -> dict[str, Any]:
