@@ -1,0 +1,53 @@
+"""Model capability flags + mesh validation.
+
+Reference behavior: nemo_automodel/_transformers/capabilities.py:583
+(per-model capability flags validated against the mesh before training).
+"""
+
+from __future__ import annotations
+
+from dataclasses import dataclass
+
+
+@dataclass
+class ModelCapabilities:
+    supports_tp: bool = True
+    supports_pp: bool = True
+    supports_cp: bool = True
+    supports_ep: bool = False
+    supports_packed_sequences: bool = True
+    supports_lora: bool = True
+    flash_head_dims: tuple[int, ...] = (128,)
+
+
+_CAPS = {
+    "LlamaForCausalLM": ModelCapabilities(),
+    "MoEForCausalLM": ModelCapabilities(supports_ep=True, supports_cp=False),
+    "DeepseekV3ForCausalLM": ModelCapabilities(supports_ep=True, supports_tp=False,
+                                               supports_cp=False, flash_head_dims=()),
+    "VLMForConditionalGeneration": ModelCapabilities(supports_pp=False,
+                                                     supports_cp=False),
+    "LlamaForSequenceClassification": ModelCapabilities(supports_pp=False,
+                                                        supports_cp=False),
+}
+
+
+def get_capabilities(model) -> ModelCapabilities:
+    return _CAPS.get(type(model).__name__, ModelCapabilities())
+
+
+def validate_model_against_mesh(model, mesh_dims: dict) -> list[str]:
+    """Returns a list of violations (empty = valid). The recipe raises on any."""
+    caps = get_capabilities(model)
+    problems = []
+    if mesh_dims.get("tp", 1) > 1 and not caps.supports_tp:
+        problems.append(f"{type(model).__name__} has no TP plan")
+    if mesh_dims.get("pp", 1) > 1 and not caps.supports_pp:
+        problems.append(f"{type(model).__name__} does not support PP")
+    if mesh_dims.get("cp", 1) > 1 and not caps.supports_cp:
+        problems.append(f"{type(model).__name__} does not support CP")
+    heads = getattr(getattr(model, "config", None), "num_attention_heads", None)
+    tp = mesh_dims.get("tp", 1)
+    if heads and tp > 1 and heads % tp != 0:
+        problems.append(f"num_attention_heads {heads} not divisible by tp={tp}")
+    return problems

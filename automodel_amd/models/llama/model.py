@@ -42,6 +42,7 @@ class LlamaConfig:
     tie_word_embeddings: bool = False
     attention_bias: bool = False
     mlp_bias: bool = False
+    qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
     initializer_range: float = 0.02
 
     def __post_init__(self):
@@ -54,7 +55,9 @@ class LlamaConfig:
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
         get = hf.get
+        archs = " ".join(get("architectures", []) or [])
         return cls(
+            qk_norm="Qwen3" in archs,
             vocab_size=get("vocab_size", 32000),
             hidden_size=get("hidden_size", 4096),
             intermediate_size=get("intermediate_size", 11008),
@@ -86,6 +89,9 @@ class LlamaAttention(nn.Module):
         self.k_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
         self.v_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
         self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
+        if cfg.qk_norm:
+            self.q_norm = RMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
+            self.k_norm = RMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
         from automodel_amd.parallel.cp import active_cp, cp_flash_attention
@@ -94,6 +100,9 @@ class LlamaAttention(nn.Module):
         q = self.q_proj(x).view(B, S, -1, self.head_dim)
         k = self.k_proj(x).view(B, S, -1, self.head_dim)
         v = self.v_proj(x).view(B, S, -1, self.head_dim)
+        if self.cfg.qk_norm:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
         q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
         if active_cp() is not None:
             o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
@@ -160,7 +169,7 @@ class LlamaModel(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
-    hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM")
+    hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM", "Qwen3ForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

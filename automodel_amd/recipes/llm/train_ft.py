@@ -74,6 +74,12 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
                 dtype=mcfg.get("dtype", "bfloat16"),
             )
 
+        # validate model capabilities against the requested mesh
+        from automodel_amd.models.common.capabilities import validate_model_against_mesh
+        problems = validate_model_against_mesh(self.model, self.mesh.dims)
+        if problems:
+            raise ValueError("model/mesh incompatibility: " + "; ".join(problems))
+
         # loss lives inside forward so FSDP keeps lm_head unsharded at use
         loss_cfg = cfg.get("loss_fn", ConfigNode())
         self.loss_fn = loss_cfg.maybe_instantiate(

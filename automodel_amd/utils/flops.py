@@ -44,3 +44,21 @@ def llama_flops_per_token(
 def mfu(tokens_per_sec_per_gpu: float, flops_per_token: float,
         peak: float = MI355X_PEAK_BF16) -> float:
     return tokens_per_sec_per_gpu * flops_per_token / peak
+
+
+def moe_flops_per_token(
+    hidden: int, layers: int, vocab: int, seq_len: int,
+    num_heads: int, num_kv_heads: int,
+    moe_intermediate: int, n_activated: int, n_shared: int = 0,
+    shared_intermediate: int | None = None, head_dim: int | None = None,
+) -> float:
+    """Training FLOPs/token for MoE models: only ACTIVATED experts count
+    (reference flops_utils.py mixtral/deepseekv3 calculators)."""
+    d = head_dim or hidden // num_heads
+    attn_proj = hidden * num_heads * d * 2 + 2 * hidden * num_kv_heads * d
+    attn_scores = 2 * (seq_len / 2) * d * num_heads
+    expert_mlp = 3 * hidden * moe_intermediate * n_activated
+    shared_mlp = 3 * hidden * (shared_intermediate or moe_intermediate) * (1 if n_shared else 0)
+    per_layer = attn_proj + attn_scores + expert_mlp + shared_mlp
+    fwd = 2 * (layers * per_layer + hidden * vocab)
+    return 3.0 * fwd

@@ -113,3 +113,75 @@ def test_dataloader_dp_sharding_disjoint():
     ids1 = torch.cat([b["input_ids"] for b in l1])
     assert len(ids0) == 4 and len(ids1) == 4
     assert not torch.equal(ids0, ids1)
+
+
+def test_timers():
+    import time as _t
+
+    from automodel_amd.training.timers import Timers
+
+    tm = Timers(cuda_sync=False)
+    with tm("a"):
+        _t.sleep(0.01)
+    assert tm.mean("a") >= 0.009
+    assert "a" in tm.summary()
+
+
+def test_capabilities_validation():
+    from automodel_amd.models.common.capabilities import validate_model_against_mesh
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+
+    m = LlamaForCausalLM(LlamaConfig(vocab_size=64, hidden_size=32,
+                                     intermediate_size=64, num_hidden_layers=1,
+                                     num_attention_heads=3, num_key_value_heads=3,
+                                     max_position_embeddings=32))
+    assert validate_model_against_mesh(m, {"tp": 1}) == []
+    assert validate_model_against_mesh(m, {"tp": 2})  # 3 heads not divisible
+
+
+def test_slurm_launcher_render(tmp_path):
+    from automodel_amd.launcher.slurm import SlurmLauncher
+
+    l = SlurmLauncher(nodes=2, gpus_per_node=8, account="acct")
+    script = l.render("cfg.yaml", "pkg.Recipe", ["--a.b=1"])
+    assert "--nnodes=2" in script and "--nproc-per-node=8" in script
+    assert "HSA_ENABLE_IPC_MODE_LEGACY=0" in script
+    assert "#SBATCH --account=acct" in script
+    path = l.launch("cfg.yaml", "pkg.Recipe", [], script_path=str(tmp_path / "j.sub"),
+                    submit=False)
+    assert (tmp_path / "j.sub").exists()
+
+
+def test_async_checkpoint_writer(tmp_path):
+    import torch as _torch
+
+    from automodel_amd.checkpoint.async_save import AsyncCheckpointWriter
+
+    w = AsyncCheckpointWriter()
+    state = {"w": _torch.randn(4, 4), "nested": {"b": _torch.ones(2)}}
+    written = {}
+
+    def write_fn(staged):
+        written.update(staged)
+
+    w.save_async(state, write_fn)
+    w.wait()
+    assert _torch.equal(written["w"], state["w"])
+    assert written["w"].device.type == "cpu"
+
+
+def test_qwen3_qk_norm_model():
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+
+    cfg = LlamaConfig.from_hf_config({
+        "architectures": ["Qwen3ForCausalLM"], "vocab_size": 64,
+        "hidden_size": 32, "intermediate_size": 64, "num_hidden_layers": 1,
+        "num_attention_heads": 2, "num_key_value_heads": 1,
+        "max_position_embeddings": 32,
+    })
+    assert cfg.qk_norm
+    m = LlamaForCausalLM(cfg)
+    m.init_weights()
+    assert hasattr(m.model.layers[0].self_attn, "q_norm")
+    out = m(torch.randint(0, 64, (1, 8)))
+    assert out.shape == (1, 8, 64)
