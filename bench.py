@@ -35,6 +35,7 @@ def parse_args():
                    choices=["llama3_8b", "llama3_1b_proxy"])
     p.add_argument("--attn", type=str, default="hip")
     p.add_argument("--loss", type=str, default="hybrid")
+    p.add_argument("--loss-chunk", type=int, default=4096)
     p.add_argument("--profile-steps", type=int, default=0,
                    help="if >0, run this many steps (no JSON contract) for rocprof")
     return p.parse_args()
@@ -80,7 +81,7 @@ def main():
     with torch.device("meta"):
         model = LlamaForCausalLM(cfg, backend=backend)
     model = model.to(dtype=torch.bfloat16)
-    model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=4096)
+    model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=args.loss_chunk)
     if world > 1:
         apply_fsdp(model, mesh["dp_shard"], reshard_after_forward=False)
     model.init_weights(device=device)
