@@ -90,6 +90,14 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         )
         self.model.loss_fn = self.loss_fn
 
+        # ---- fp8 linear swap (before sharding)
+        q_cfg = cfg.get("quantization")
+        if q_cfg and q_cfg.get("fp8", False):
+            from automodel_amd.quantization.fp8 import apply_fp8_to_model
+            n = apply_fp8_to_model(self.model)
+            if self.rank == 0:
+                self.logger.info(f"fp8: swapped {n} linears to Float8Linear")
+
         # ---- activation checkpointing (before sharding)
         ac_cfg = cfg.get("activation_checkpointing")
         if ac_cfg and ac_cfg.get("enabled", True):

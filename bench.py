@@ -36,6 +36,7 @@ def parse_args():
     p.add_argument("--attn", type=str, default="hip")
     p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--loss-chunk", type=int, default=4096)
+    p.add_argument("--fp8", action="store_true", help="swap linears to Float8Linear")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="if >0, run this many steps (no JSON contract) for rocprof")
     return p.parse_args()
@@ -82,6 +83,11 @@ def main():
         model = LlamaForCausalLM(cfg, backend=backend)
     model = model.to(dtype=torch.bfloat16)
     model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=args.loss_chunk)
+    if args.fp8:
+        from automodel_amd.quantization.fp8 import apply_fp8_to_model
+        n = apply_fp8_to_model(model)
+        if rank == 0:
+            print(f"fp8: swapped {n} linears")
     if world > 1:
         apply_fsdp(model, mesh["dp_shard"], reshard_after_forward=False)
     model.init_weights(device=device)
@@ -152,7 +158,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": round(tps / (baseline_tok_per_gpu * world), 4),
-            "dtype": "bf16",
+            "dtype": "fp8" if args.fp8 else "bf16",
             "data": "synthetic",
             "config": {
                 "model": "llama3-8b" if args.model == "llama3_8b" else args.model,

@@ -338,3 +338,42 @@ def test_varlen_attention_gpu():
         assert torch.allclose(out[:, a:b].float(), ref, atol=3e-2, rtol=3e-2), (a, b)
     out.sum().backward()
     assert torch.isfinite(q.grad.float()).all()
+
+
+def test_fp8_linear_numerics_and_train():
+    """Float8Linear vs bf16 linear: loose forward parity + trainable."""
+    from automodel_amd.quantization.fp8 import Float8Linear, apply_fp8_to_model
+
+    torch.manual_seed(0)
+    lin = torch.nn.Linear(256, 512, bias=False, device="cuda", dtype=torch.bfloat16)
+    f8 = Float8Linear.from_linear(lin)
+    x = torch.randn(64, 256, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y8 = f8(x)
+    y16 = torch.nn.functional.linear(x.detach(), lin.weight)
+    rel = (y8.float() - y16.float()).abs().mean() / y16.float().abs().mean()
+    assert rel < 0.1, float(rel)
+    y8.sum().backward()
+    assert x.grad is not None and lin.weight.grad is not None
+    assert torch.isfinite(x.grad.float()).all()
+
+    # model swap: loss still decreases
+    from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    from automodel_amd.optim.adamw import FusedAdamW
+
+    m = LlamaForCausalLM(LlamaConfig(
+        vocab_size=512, hidden_size=512, intermediate_size=1024,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=4,
+        head_dim=128, max_position_embeddings=256)).to(torch.bfloat16)
+    m.loss_fn = FusedLinearCrossEntropy(backend="hybrid")
+    m.init_weights(device="cuda")
+    n = apply_fp8_to_model(m)
+    assert n == 2 * 7
+    opt = FusedAdamW(m.parameters(), lr=1e-3, weight_decay=0.0)
+    ids = torch.randint(0, 512, (2, 129), device="cuda")
+    losses = []
+    for _ in range(6):
+        loss = m(ids[:, :-1], labels=ids[:, 1:].contiguous()) / 256
+        opt.zero_grad(); loss.backward(); opt.step()
+        losses.append(float(loss.detach()))
+    assert losses[-1] < losses[0], losses
