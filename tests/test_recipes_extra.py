@@ -101,3 +101,28 @@ def test_benchmark_recipe(tmp_path):
     r.run_train_validation_loop()
     lines = [json.loads(x) for x in open(tmp_path / "bench" / "training.jsonl")]
     assert any("benchmark_summary" in m for m in lines)
+
+
+def test_infonce_loss():
+    from automodel_amd.loss.infonce import info_nce_loss
+
+    torch.manual_seed(0)
+    q = torch.randn(8, 16)
+    # positives close to queries -> low loss vs random
+    loss_aligned = info_nce_loss(q, q + 0.01 * torch.randn(8, 16))
+    loss_random = info_nce_loss(q, torch.randn(8, 16))
+    assert loss_aligned < loss_random
+
+
+def test_generation_greedy():
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    from automodel_amd.utils.generation import generate
+
+    m = LlamaForCausalLM(LlamaConfig(**{**TINY, "max_position_embeddings": 128}))
+    m.init_weights()
+    ids = torch.randint(0, 128, (2, 8))
+    out = generate(m, ids, max_new_tokens=5)
+    assert out.shape == (2, 13)
+    # deterministic
+    out2 = generate(m, ids, max_new_tokens=5)
+    assert torch.equal(out, out2)
