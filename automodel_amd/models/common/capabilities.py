@@ -46,6 +46,11 @@ def validate_model_against_mesh(model, mesh_dims: dict) -> list[str]:
         problems.append(f"{type(model).__name__} does not support PP")
     if mesh_dims.get("cp", 1) > 1 and not caps.supports_cp:
         problems.append(f"{type(model).__name__} does not support CP")
+    cfg = getattr(model, "config", None)
+    if mesh_dims.get("tp", 1) > 1 and (
+        getattr(cfg, "fused_qkv", False) or getattr(cfg, "fused_gate_up", False)
+    ):
+        problems.append("fused qkv/gate_up projections have no TP plan (disable fusion)")
     heads = getattr(getattr(model, "config", None), "num_attention_heads", None)
     tp = mesh_dims.get("tp", 1)
     if heads and tp > 1 and heads % tp != 0:

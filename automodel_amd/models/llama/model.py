@@ -23,7 +23,7 @@ from automodel_amd.models.common.backend import BackendConfig
 from automodel_amd.ops.attention import flash_attention
 from automodel_amd.ops.rms_norm import RMSNorm
 from automodel_amd.ops.rope import apply_rope, build_rope_cache
-from automodel_amd.ops.swiglu import swiglu
+from automodel_amd.ops.swiglu import swiglu, swiglu_cat
 
 
 @dataclass
@@ -43,6 +43,8 @@ class LlamaConfig:
     attention_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
+    fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
+    fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
     initializer_range: float = 0.02
 
     def __post_init__(self):
@@ -85,9 +87,12 @@ class LlamaAttention(nn.Module):
         Hk = cfg.num_key_value_heads
         self.num_heads, self.num_kv_heads, self.head_dim = H, Hk, D
         bias = cfg.attention_bias
-        self.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=bias)
-        self.k_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
-        self.v_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
+        if cfg.fused_qkv:
+            self.qkv_proj = nn.Linear(cfg.hidden_size, (H + 2 * Hk) * D, bias=bias)
+        else:
+            self.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=bias)
+            self.k_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
+            self.v_proj = nn.Linear(cfg.hidden_size, Hk * D, bias=bias)
         self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
         if cfg.qk_norm:
             self.q_norm = RMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
@@ -97,9 +102,18 @@ class LlamaAttention(nn.Module):
         from automodel_amd.parallel.cp import active_cp, cp_flash_attention
 
         B, S, _ = x.shape
-        q = self.q_proj(x).view(B, S, -1, self.head_dim)
-        k = self.k_proj(x).view(B, S, -1, self.head_dim)
-        v = self.v_proj(x).view(B, S, -1, self.head_dim)
+        if self.cfg.fused_qkv:
+            qkv = self.qkv_proj(x)
+            q, k, v = qkv.split([self.num_heads * self.head_dim,
+                                 self.num_kv_heads * self.head_dim,
+                                 self.num_kv_heads * self.head_dim], dim=-1)
+            q = q.view(B, S, -1, self.head_dim)
+            k = k.contiguous().view(B, S, -1, self.head_dim)
+            v = v.contiguous().view(B, S, -1, self.head_dim)
+        else:
+            q = self.q_proj(x).view(B, S, -1, self.head_dim)
+            k = self.k_proj(x).view(B, S, -1, self.head_dim)
+            v = self.v_proj(x).view(B, S, -1, self.head_dim)
         if self.cfg.qk_norm:
             q = self.q_norm(q)
             k = self.k_norm(k)
@@ -115,12 +129,19 @@ class LlamaMLP(nn.Module):
     def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
         super().__init__()
         self.backend = backend
+        self.fused = cfg.fused_gate_up
         bias = cfg.mlp_bias
-        self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
-        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
+        if self.fused:
+            self.gate_up_proj = nn.Linear(cfg.hidden_size, 2 * cfg.intermediate_size,
+                                          bias=bias)
+        else:
+            self.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
+            self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=bias)
         self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.fused:
+            return self.down_proj(swiglu_cat(self.gate_up_proj(x)))
         return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
@@ -190,6 +211,8 @@ class LlamaForCausalLM(nn.Module):
         self.lm_head = nn.Linear(config.hidden_size, config.vocab_size, bias=False)
         if config.tie_word_embeddings:
             self.lm_head.weight = self.model.embed_tokens.weight
+        if config.fused_qkv or config.fused_gate_up:
+            self.state_dict_adapter = LlamaFusedStateDictAdapter(config)
         # set by the recipe; called as loss_fn(hidden, lm_head_weight, labels).
         # Computing the loss INSIDE forward keeps lm_head.weight unsharded
         # under FSDP2 (reference computes it via lm-weight gather instead,
@@ -257,3 +280,52 @@ class LlamaForCausalLM(nn.Module):
                 seen.add(id(p))
                 total += p.numel()
         return total
+
+
+class LlamaFusedStateDictAdapter:
+    """HF per-projection keys <-> fused qkv / gate_up weights."""
+
+    def __init__(self, config: LlamaConfig):
+        self.cfg = config
+
+    def from_hf(self, sd: dict) -> dict:
+        import re
+
+        cfg = self.cfg
+        out = dict(sd)
+        for i in range(cfg.num_hidden_layers):
+            p = f"model.layers.{i}"
+            if cfg.fused_qkv:
+                parts = [out.pop(f"{p}.self_attn.{n}_proj.weight", None) for n in "qkv"]
+                if all(x is not None for x in parts):
+                    out[f"{p}.self_attn.qkv_proj.weight"] = torch.cat(parts, dim=0)
+                bparts = [out.pop(f"{p}.self_attn.{n}_proj.bias", None) for n in "qkv"]
+                if all(x is not None for x in bparts):
+                    out[f"{p}.self_attn.qkv_proj.bias"] = torch.cat(bparts, dim=0)
+            if cfg.fused_gate_up:
+                g = out.pop(f"{p}.mlp.gate_proj.weight", None)
+                u = out.pop(f"{p}.mlp.up_proj.weight", None)
+                if g is not None and u is not None:
+                    out[f"{p}.mlp.gate_up_proj.weight"] = torch.cat([g, u], dim=0)
+        return out
+
+    def to_hf(self, sd: dict) -> dict:
+        cfg = self.cfg
+        H, Hk, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
+        out = {}
+        for k, t in sd.items():
+            if k.endswith(".self_attn.qkv_proj.weight") or k.endswith(".self_attn.qkv_proj.bias"):
+                q, kk, v = t.split([H * D, Hk * D, Hk * D], dim=0)
+                kind = "weight" if k.endswith("weight") else "bias"
+                base = k.rsplit(".qkv_proj.", 1)[0]
+                out[f"{base}.q_proj.{kind}"] = q
+                out[f"{base}.k_proj.{kind}"] = kk
+                out[f"{base}.v_proj.{kind}"] = v
+            elif k.endswith(".mlp.gate_up_proj.weight"):
+                g, u = t.chunk(2, dim=0)
+                base = k.rsplit(".gate_up_proj.", 1)[0]
+                out[f"{base}.gate_proj.weight"] = g
+                out[f"{base}.up_proj.weight"] = u
+            else:
+                out[k] = t
+        return out

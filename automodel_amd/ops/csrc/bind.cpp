@@ -18,6 +18,10 @@ TORCH_LIBRARY(amd_ops, m) {
   m.impl("swiglu_fwd", &amd_ops::swiglu_fwd);
   m.def("swiglu_bwd(Tensor dy, Tensor g, Tensor u) -> (Tensor, Tensor)");
   m.impl("swiglu_bwd", &amd_ops::swiglu_bwd);
+  m.def("swiglu_cat_fwd(Tensor gu) -> Tensor");
+  m.impl("swiglu_cat_fwd", &amd_ops::swiglu_cat_fwd);
+  m.def("swiglu_cat_bwd(Tensor dy, Tensor gu) -> Tensor");
+  m.impl("swiglu_cat_bwd", &amd_ops::swiglu_cat_bwd);
 
   m.def(
       "adamw_step(Tensor(a!) param, Tensor grad, Tensor(b!) master, Tensor(c!) m, "

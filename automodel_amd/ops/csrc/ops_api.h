@@ -16,6 +16,8 @@ std::tuple<at::Tensor, at::Tensor> rope_fwd(const at::Tensor& q, const at::Tenso
                                             bool backward);
 
 at::Tensor swiglu_fwd(const at::Tensor& g, const at::Tensor& u);
+at::Tensor swiglu_cat_fwd(const at::Tensor& gu);
+at::Tensor swiglu_cat_bwd(const at::Tensor& dy, const at::Tensor& gu);
 std::tuple<at::Tensor, at::Tensor> swiglu_bwd(const at::Tensor& dy, const at::Tensor& g,
                                               const at::Tensor& u);
 

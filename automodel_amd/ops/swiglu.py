@@ -35,3 +35,26 @@ def swiglu(gate: torch.Tensor, up: torch.Tensor, backend: str = "hip") -> torch.
     if backend == "hip" and gate.is_cuda:
         return _SwiGLUHip.apply(gate, up)
     return swiglu_ref(gate, up)
+
+
+class _SwiGLUCatHip(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gu):
+        gu2 = gu.reshape(-1, gu.shape[-1]).contiguous()
+        ctx.save_for_backward(gu2)
+        ctx.shape = gu.shape
+        return hip_ops().swiglu_cat_fwd(gu2).view(*gu.shape[:-1], gu.shape[-1] // 2)
+
+    @staticmethod
+    def backward(ctx, dy):
+        (gu2,) = ctx.saved_tensors
+        dgu = hip_ops().swiglu_cat_bwd(dy.reshape(-1, dy.shape[-1]).contiguous(), gu2)
+        return dgu.view(ctx.shape)
+
+
+def swiglu_cat(gu: torch.Tensor, backend: str = "hip") -> torch.Tensor:
+    """SwiGLU over a concatenated [.., 2I] gate|up tensor (fused gate_up GEMM)."""
+    if backend == "hip" and gu.is_cuda:
+        return _SwiGLUCatHip.apply(gu)
+    g, u = gu.chunk(2, dim=-1)
+    return swiglu_ref(g, u)

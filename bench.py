@@ -32,11 +32,13 @@ def parse_args():
     p.add_argument("--mbs", type=int, default=4, help="micro-batch size per GPU")
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3_8b",
-                   choices=["llama3_8b", "llama3_1b_proxy"])
+                   choices=["llama3_8b", "llama3_1b_proxy", "tiny_proxy"])
     p.add_argument("--attn", type=str, default="hip")
     p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--loss-chunk", type=int, default=4096)
     p.add_argument("--fp8", action="store_true", help="swap linears to Float8Linear")
+    p.add_argument("--no-fused-proj", action="store_true",
+                   help="disable fused qkv/gate_up projections")
     p.add_argument("--profile-steps", type=int, default=0,
                    help="if >0, run this many steps (no JSON contract) for rocprof")
     return p.parse_args()
@@ -47,6 +49,12 @@ MODEL_CONFIGS = {
         vocab_size=128256, hidden_size=4096, intermediate_size=14336,
         num_hidden_layers=32, num_attention_heads=32, num_key_value_heads=8,
         max_position_embeddings=8192, rope_theta=500000.0, rms_norm_eps=1e-5,
+    ),
+    # minimal config for CPU contract smoke tests
+    "tiny_proxy": dict(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=2,
+        head_dim=128, max_position_embeddings=512, rope_theta=10000.0,
     ),
     # small proxy for smoke/debug runs
     "llama3_1b_proxy": dict(
@@ -76,7 +84,9 @@ def main():
     device = torch.device("cuda") if use_cuda else torch.device("cpu")
     mesh = build_mesh(dp_shard=-1)
 
-    cfg = LlamaConfig(**MODEL_CONFIGS[args.model])
+    cfg = LlamaConfig(**MODEL_CONFIGS[args.model],
+                      fused_qkv=not args.no_fused_proj,
+                      fused_gate_up=not args.no_fused_proj)
     backend = BackendConfig(attn=args.attn, loss=args.loss)
     torch.manual_seed(1234 + rank)
     with torch.device("meta"):

@@ -106,3 +106,21 @@ def test_tie_word_embeddings():
     # only counted once
     untied = make_model(tie_word_embeddings=False)
     assert m.num_parameters() < untied.num_parameters()
+
+
+def test_fused_projections_match_unfused():
+    """fused qkv/gate_up forward == separate projections (same weights via
+    the fused state-dict adapter)."""
+    torch.manual_seed(0)
+    base = make_model()
+    cfg_f = LlamaConfig(**{**TINY}, fused_qkv=True, fused_gate_up=True)
+    fused = LlamaForCausalLM(cfg_f, backend=BackendConfig().for_cpu())
+    fused.load_state_dict(fused.state_dict_adapter.from_hf(base.state_dict()),
+                          strict=False)
+    ids = torch.randint(0, 256, (2, 16))
+    assert torch.allclose(fused(ids), base(ids), atol=1e-5)
+    # roundtrip back to HF keys
+    hf = fused.state_dict_adapter.to_hf(fused.state_dict())
+    assert "model.layers.0.self_attn.q_proj.weight" in hf
+    assert torch.allclose(hf["model.layers.0.mlp.up_proj.weight"],
+                          base.state_dict()["model.layers.0.mlp.up_proj.weight"])
