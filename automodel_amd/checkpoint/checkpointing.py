@@ -27,11 +27,24 @@ class Checkpointer:
         model_save_format: str = "safetensors",
         save_consolidated: bool = False,
         keep_last_n: int | None = None,
+        async_save: bool = False,
     ):
         self.checkpoint_dir = checkpoint_dir
         self.model_save_format = model_save_format
         self.save_consolidated = save_consolidated
         self.keep_last_n = keep_last_n
+        self.async_save = async_save
+        self._async_writer = None
+        if async_save:
+            from automodel_amd.checkpoint.async_save import AsyncCheckpointWriter
+
+            self._async_writer = AsyncCheckpointWriter()
+
+    def maybe_wait_for_staging(self) -> None:
+        """Block before the optimizer step mutates weights while an async
+        save is still staging (reference train_ft.py:1251)."""
+        if self._async_writer is not None:
+            self._async_writer.wait()
 
     # ---------------------------------------------------------------- save
     def save(self, path: str, model=None, optimizer=None, extra_state: dict | None = None,

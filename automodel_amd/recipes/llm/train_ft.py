@@ -303,9 +303,18 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
                 total_loss += self._forward_backward_step(batch, loss_scale).float()
 
         grad_norm = clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
+        if hasattr(self, "checkpointer") and hasattr(self.checkpointer, "maybe_wait_for_staging"):
+            self.checkpointer.maybe_wait_for_staging()
         self.optimizer.step()
         self.optimizer.zero_grad(set_to_none=True)
         self.lr_scheduler.step()
+        # MoE: aux-free gate-bias update + load metrics (reference
+        # train_ft.py update_moe_gate_bias / load_balance_metrics)
+        moe_metrics = {}
+        if hasattr(self.model, "update_moe_gate_bias"):
+            self.model.update_moe_gate_bias()
+            from automodel_amd.moe.load_balance_metrics import load_balance_metrics
+            moe_metrics = load_balance_metrics(self.model, group=self.mesh.dp_group())
 
         if self.world > 1:
             dist.all_reduce(total_loss, group=self._dp_cp_group())
@@ -313,6 +322,7 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         ntok = int(num_label_tokens.item())
         return {
             "step": self.step_scheduler.step,
+            **moe_metrics,
             "loss": total_loss.item() / max(1, ntok),
             "grad_norm": float(grad_norm),
             "lr": self.lr_scheduler.get_last_lr()[0],

@@ -187,3 +187,62 @@ def _ep_fn(rank, world, dispatcher):
 @pytest.mark.parametrize("dispatcher", ["a2a", "allgather"])
 def test_ep2_dispatcher_matches_dense(dispatcher):
     run_distributed(_ep_fn, world=2, args=(dispatcher,))
+
+
+def test_mtp_loss():
+    from automodel_amd.loss.mtp import MTPHead, calculate_mtp_loss
+
+    torch.manual_seed(0)
+    B, S, H, V = 2, 16, 32, 64
+    hidden = torch.randn(B, S, H, requires_grad=True)
+    emb = torch.nn.Embedding(V, H)
+    lm_w = torch.randn(V, H, requires_grad=True)
+    head = MTPHead(H)
+    ids = torch.randint(0, V, (B, S))
+    labels = torch.randint(0, V, (B, S))
+    loss = calculate_mtp_loss(hidden, emb, lm_w, head, ids, labels)
+    assert torch.isfinite(loss) and loss > 0
+    loss.backward()
+    assert hidden.grad is not None and head.eh_proj.weight.grad is not None
+
+
+def test_load_balance_metrics():
+    from automodel_amd.moe.load_balance_metrics import load_balance_metrics
+    from automodel_amd.moe.model import MoEForCausalLM
+
+    m = MoEForCausalLM(MODEL_CFG)
+    m.init_weights()
+    ids = torch.randint(0, 128, (2, 16))
+    m(ids)
+    metrics = load_balance_metrics(m, detailed=True)
+    assert metrics["moe_imbalance_mean"] >= 1.0
+    assert "moe_layer0_max_frac" in metrics
+
+
+def test_moe_recipe_with_metrics(tmp_path):
+    import json
+
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.llm.train_ft import (
+        TrainFinetuneRecipeForNextTokenPrediction,
+    )
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"architecture": "Qwen3MoeForCausalLM", "config": MODEL_CFG,
+                  "dtype": "float32"},
+        "loss_fn": {"backend": "chunked", "chunk_size": 64},
+        "optimizer": {"lr": 1e-3},
+        "step_scheduler": {"grad_acc_steps": 1, "max_steps": 2},
+        "dataloader": {
+            "dataset": {"kind": "mock", "num_samples": 8, "seq_len": 16,
+                        "vocab_size": 128},
+            "batch_size": 2,
+        },
+        "output_dir": str(tmp_path / "moe"),
+    })
+    r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    r.setup()
+    r.run_train_validation_loop()
+    lines = [json.loads(x) for x in open(tmp_path / "moe" / "training.jsonl")]
+    assert any("moe_imbalance_mean" in m for m in lines)
