@@ -8,6 +8,28 @@ from __future__ import annotations
 
 import torch
 
+from automodel_amd.ops._backend import hip_ops
+
+
+class _SoftCEHip(torch.autograd.Function):
+    """HIP soft CE: -sum softmax(t) * log_softmax(s), token-sum over rows."""
+
+    @staticmethod
+    def forward(ctx, s, t):
+        s2 = s.reshape(-1, s.shape[-1]).contiguous()
+        t2 = t.reshape(-1, t.shape[-1]).contiguous()
+        loss, lse_s, lse_t = hip_ops().soft_ce_fwd(s2, t2)
+        ctx.save_for_backward(s2, t2, lse_s, lse_t)
+        ctx.shape = s.shape
+        return loss.sum()
+
+    @staticmethod
+    def backward(ctx, dloss):
+        s2, t2, lse_s, lse_t = ctx.saved_tensors
+        g = hip_ops().soft_ce_bwd(s2, t2, lse_s, lse_t,
+                                  dloss.reshape(1).float().contiguous())
+        return g.view(ctx.shape), None
+
 
 def soft_cross_entropy(student_logits: torch.Tensor, teacher_probs: torch.Tensor,
                        mask: torch.Tensor | None = None) -> torch.Tensor:
@@ -17,6 +39,15 @@ def soft_cross_entropy(student_logits: torch.Tensor, teacher_probs: torch.Tensor
     if mask is not None:
         per_tok = per_tok * mask.float()
     return per_tok.sum()
+
+
+def soft_cross_entropy_from_logits(student_logits: torch.Tensor,
+                                   teacher_logits: torch.Tensor) -> torch.Tensor:
+    """HIP one-pass soft CE on GPU bf16 (csrc/soft_ce.hip); torch fallback."""
+    if student_logits.is_cuda and student_logits.dtype == torch.bfloat16:
+        return _SoftCEHip.apply(student_logits, teacher_logits)
+    return soft_cross_entropy(student_logits,
+                              torch.softmax(teacher_logits.float(), dim=-1))
 
 
 def forward_kl(

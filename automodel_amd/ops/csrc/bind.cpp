@@ -53,6 +53,11 @@ TORCH_LIBRARY(amd_ops, m) {
       " -> (Tensor, Tensor)");
   m.impl("fused_ce_bwd", &amd_ops::fused_ce_bwd);
 
+  m.def("soft_ce_fwd(Tensor s, Tensor t) -> (Tensor, Tensor, Tensor)");
+  m.impl("soft_ce_fwd", &amd_ops::soft_ce_fwd);
+  m.def("soft_ce_bwd(Tensor s, Tensor t, Tensor lse_s, Tensor lse_t, Tensor dloss) -> Tensor");
+  m.impl("soft_ce_bwd", &amd_ops::soft_ce_bwd);
+
   m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
   m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");

@@ -51,6 +51,12 @@ std::tuple<at::Tensor, at::Tensor> fused_ce_bwd(const at::Tensor& hidden,
                                                 const at::Tensor& lse,
                                                 const at::Tensor& dloss);
 
+std::tuple<at::Tensor, at::Tensor, at::Tensor> soft_ce_fwd(const at::Tensor& s,
+                                                           const at::Tensor& t);
+at::Tensor soft_ce_bwd(const at::Tensor& s, const at::Tensor& t,
+                       const at::Tensor& lse_s, const at::Tensor& lse_t,
+                       const at::Tensor& dloss);
+
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);

@@ -377,3 +377,20 @@ def test_fp8_linear_numerics_and_train():
         opt.zero_grad(); loss.backward(); opt.step()
         losses.append(float(loss.detach()))
     assert losses[-1] < losses[0], losses
+
+
+def test_soft_ce_hip_parity():
+    from automodel_amd.loss.kd_loss import soft_cross_entropy_from_logits
+
+    torch.manual_seed(0)
+    T, V = 128, 1024
+    s = torch.randn(T, V, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    t = torch.randn(T, V, device="cuda", dtype=torch.bfloat16)
+    loss = soft_cross_entropy_from_logits(s, t)
+    s2 = s.detach().float().requires_grad_(True)
+    ref = -(torch.softmax(t.float(), -1) * torch.log_softmax(s2, -1)).sum()
+    assert torch.allclose(loss, ref, rtol=2e-2), (float(loss), float(ref))
+    loss.backward()
+    ref.backward()
+    assert torch.allclose(s.grad.float(), s2.grad, atol=5e-2, rtol=5e-2), \
+        (s.grad.float() - s2.grad).abs().max()
