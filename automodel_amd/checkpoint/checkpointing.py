@@ -112,8 +112,34 @@ class Checkpointer:
         if adapter is not None:
             full_sd = adapter.to_hf(full_sd)
         full_sd = {k: v.contiguous() for k, v in full_sd.items() if isinstance(v, torch.Tensor)}
-        save_file(full_sd, os.path.join(out_dir, "model.safetensors"),
-                  metadata={"format": "pt"})
+        total_bytes = sum(v.numel() * v.element_size() for v in full_sd.values())
+        max_shard = 4 * 2**30
+        if total_bytes <= max_shard:
+            save_file(full_sd, os.path.join(out_dir, "model.safetensors"),
+                      metadata={"format": "pt"})
+            return
+        # shard by size + write HF index (reference consolidate_hf_safetensors)
+        import json as _json
+
+        shards: list[dict] = [{}]
+        sizes = [0]
+        for k, v in full_sd.items():
+            b = v.numel() * v.element_size()
+            if sizes[-1] + b > max_shard and shards[-1]:
+                shards.append({})
+                sizes.append(0)
+            shards[-1][k] = v
+            sizes[-1] += b
+        n = len(shards)
+        weight_map = {}
+        for i, shard in enumerate(shards):
+            fn = f"model-{i+1:05d}-of-{n:05d}.safetensors"
+            save_file(shard, os.path.join(out_dir, fn), metadata={"format": "pt"})
+            for k in shard:
+                weight_map[k] = fn
+        with open(os.path.join(out_dir, "model.safetensors.index.json"), "w") as f:
+            _json.dump({"metadata": {"total_size": total_bytes},
+                        "weight_map": weight_map}, f)
 
     # ------------------------------------------------------------- retention
     def _apply_retention(self) -> None:

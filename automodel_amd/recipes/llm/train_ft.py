@@ -199,6 +199,22 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
                 keep_last_n=ckpt_cfg.get("keep_last_n"),
             )
 
+        # ---- EMA / NEFTune / stepped GC (training extras)
+        self.ema = None
+        if cfg.get("ema") and cfg.ema.get("enabled", True):
+            from automodel_amd.training.extras import EMA
+            self.ema = EMA(self.model, decay=cfg.ema.get("decay", 0.999))
+        if cfg.get("neftune") and cfg.neftune.get("enabled", True):
+            from automodel_amd.training.extras import apply_neftune
+            emb = getattr(getattr(self.model, "model", None), "embed_tokens", None)
+            if emb is not None:
+                apply_neftune(emb, alpha=cfg.neftune.get("alpha", 5.0))
+        self.gc = None
+        gc_cfg = cfg.get("garbage_collection")
+        if gc_cfg and gc_cfg.get("enabled", True):
+            from automodel_amd.training.extras import SteppedGarbageCollector
+            self.gc = SteppedGarbageCollector(gc_cfg.get("every_steps", 100))
+
         # ---- metric logging
         out_dir = cfg.get("output_dir", "outputs")
         self.metrics = MetricLogger(os.path.join(out_dir, "training.jsonl"))
@@ -310,6 +326,10 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         self.lr_scheduler.step()
         # MoE: aux-free gate-bias update + load metrics (reference
         # train_ft.py update_moe_gate_bias / load_balance_metrics)
+        if self.ema is not None:
+            self.ema.update(self.model)
+        if self.gc is not None:
+            self.gc.maybe_collect(self.step_scheduler.step)
         moe_metrics = {}
         if hasattr(self.model, "update_moe_gate_bias"):
             self.model.update_moe_gate_bias()

@@ -114,3 +114,40 @@ def test_recipe_lora(tmp_path):
     trainable = [n for n, p in r.model.named_parameters() if p.requires_grad]
     assert trainable and all("lora_" in n for n in trainable)
     r.run_train_validation_loop()
+
+
+def test_checkpoint_kill_resume_trajectory_parity(tmp_path):
+    """Checkpoint robustness (reference functional_tests/checkpoint_robustness):
+    an interrupted+resumed run must reproduce the uninterrupted loss
+    trajectory exactly (model+optimizer+scheduler+RNG+dataloader state)."""
+    import json
+
+    ckpt = str(tmp_path / "ck")
+
+    def run(max_steps, restore=None, tag="a"):
+        cfg = base_cfg(
+            tmp_path,
+            checkpoint={"enabled": True, "checkpoint_dir": ckpt},
+            step_scheduler={"grad_acc_steps": 1, "max_steps": max_steps,
+                            "ckpt_every_steps": 3},
+            dataloader={
+                "dataset": {"kind": "mock", "num_samples": 32, "seq_len": 16,
+                            "vocab_size": 128},
+                "batch_size": 2, "shuffle": False,
+            },
+        )
+        cfg["output_dir"] = str(tmp_path / f"out_{tag}_{max_steps}")
+        if restore:
+            cfg["restore_from"] = restore
+        r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+        r.setup()
+        r.run_train_validation_loop()
+        lines = [json.loads(x) for x in open(f"{cfg['output_dir']}/training.jsonl")]
+        return [m["loss"] for m in lines]
+
+    full = run(6, tag="full")                      # uninterrupted: steps 1..6
+    _ = run(3, tag="partial")                      # killed after step 3 (saved)
+    resumed = run(6, restore=f"{ckpt}/step_3", tag="resumed")  # steps 4..6
+    assert len(full) == 6 and len(resumed) == 3
+    for a, b in zip(full[3:], resumed):
+        assert abs(a - b) < 1e-4, (full, resumed)
