@@ -29,7 +29,7 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=8)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--mbs", type=int, default=4, help="micro-batch size per GPU")
+    p.add_argument("--mbs", type=int, default=8, help="micro-batch size per GPU")
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3_8b",
                    choices=["llama3_8b", "llama3_1b_proxy", "tiny_proxy"])
