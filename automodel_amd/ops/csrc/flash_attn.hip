@@ -37,13 +37,20 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 #define FA_WAVES 4
 #define FA_BLOCK (FA_WAVES * WAVE_SIZE)
 
-// K tile: [KVBLK][D] bf16 row-major, 256 B rows, XOR-swizzled by ((row&7)<<4).
+// K tile: [KVBLK][D] bf16 row-major, 256 B rows. A 256-B row starts every
+// 64 banks, so ALL rows share bank 0 without a swizzle; ((row&15)<<4)
+// spreads a 16-lane ds_read_b128 group over all 16 slots -> conflict-free
+// (guide Guideline 4). PMC before the fix: SQ_LDS_BANK_CONFLICT 2.6e11 in
+// the dkv kernel alone (profiles/r1_pmc_counters.csv).
 __device__ __forceinline__ int k_lds_off(int row, int byte_in_row) {
-  return row * 256 + (byte_in_row ^ ((row & 7) << 4));
+  return row * 256 + (byte_in_row ^ ((row & 15) << 4));
 }
-// V^T tile: [D][KVBLK] bf16, 64 B rows, XOR-swizzled by ((row&3)<<4).
+// V^T tile: [D][KVBLK] bf16, 64 B rows (16 words). Rows d and d+4 share the
+// same 16-word bank window (16*d mod 64); rotating the 16-B slot by
+// (d>>2)&3 separates the four colliding rows. (The previous (d&3) rotation
+// only moved rows that never collided.)
 __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
-  return d * 64 + (byte_in_row ^ ((d & 3) << 4));
+  return d * 64 + (byte_in_row ^ (((d >> 2) & 3) << 4));
 }
 
 template <int D>
@@ -289,7 +296,7 @@ __global__ void fa_delta_kernel(const bf16* __restrict__ dout, const bf16* __res
 // qt/dot images:   [128][64 B], XOR swizzle ((d&3)<<4)   (same as vt_lds_off)
 // per-wave transpose buffer: [32][64 B] with ((row&3)<<4)
 __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
-  return row * 64 + (byte_in_row ^ ((row & 3) << 4));
+  return row * 64 + (byte_in_row ^ (((row >> 2) & 3) << 4));
 }
 
 #define FAB_WAVES 4

@@ -26,9 +26,11 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define GG_BN 128
 #define GG_BK 64
 
-// LDS tiles: [128 rows][64 cols] bf16, 128-B rows, XOR swizzle ((row&7)<<4).
+// LDS tiles: [128 rows][64 cols] bf16, 128-B rows (32 words): rows of equal
+// parity share a 32-word bank window; rotating the 16-B slot by (row>>1)&7
+// separates the 8 colliding rows of a 16-lane b128 group -> conflict-free.
 __device__ __forceinline__ int gg_off(int row, int byte_in_row) {
-  return row * 128 + (byte_in_row ^ ((row & 7) << 4));
+  return row * 128 + (byte_in_row ^ (((row >> 1) & 7) << 4));
 }
 
 __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
