@@ -24,6 +24,7 @@ RECIPE_ALIASES = {
     "vlm_finetune": "automodel_amd.recipes.vlm.finetune.FinetuneRecipeForVLM",
     "llm_kd": "automodel_amd.recipes.llm.kd.KDRecipeForNextTokenPrediction",
     "llm_seq_cls": "automodel_amd.recipes.llm.train_seq_cls.TrainFinetuneRecipeForSequenceClassification",
+    "retrieval": "automodel_amd.recipes.llm.train_retrieval.TrainRecipeForRetrieval",
 }
 
 

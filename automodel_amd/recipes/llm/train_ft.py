@@ -224,6 +224,14 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         if restore:
             self.load_checkpoint(restore)
 
+    def _count_batch_tokens(self, batch: dict) -> torch.Tensor:
+        if "labels" in batch:
+            return count_label_tokens(batch["labels"])
+        for k in ("input_ids", "query_ids"):
+            if k in batch:
+                return torch.tensor(batch[k].numel())
+        return torch.tensor(1)
+
     def _dp_cp_group(self):
         if self.mesh.mesh is None:
             return None
@@ -299,7 +307,7 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         device = self.device
         num_label_tokens = torch.zeros((), dtype=torch.long)
         for b in batches:
-            num_label_tokens += count_label_tokens(b["labels"])
+            num_label_tokens += self._count_batch_tokens(b)
         num_label_tokens = num_label_tokens.to(device)
         if self.world > 1:
             # counted on the UNSHARDED batch (cp ranks share it): reduce over dp

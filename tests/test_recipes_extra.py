@@ -126,3 +126,22 @@ def test_generation_greedy():
     # deterministic
     out2 = generate(m, ids, max_new_tokens=5)
     assert torch.equal(out, out2)
+
+
+def test_retrieval_recipe(tmp_path):
+    from automodel_amd.recipes.llm.train_retrieval import TrainRecipeForRetrieval
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"config": TINY, "embedding_dim": 16, "dtype": "float32"},
+        "optimizer": {"lr": 1e-3},
+        "step_scheduler": {"grad_acc_steps": 1, "max_steps": 3},
+        "retrieval": {"temperature": 0.1},
+        "dataloader": {"dataset": {"num_samples": 16, "seq_len": 12,
+                                   "vocab_size": 128}, "batch_size": 4},
+        "output_dir": str(tmp_path / "ret"),
+    })
+    r = TrainRecipeForRetrieval(cfg)
+    r.setup()
+    r.run_train_validation_loop()
+    assert r.step_scheduler.step == 3
