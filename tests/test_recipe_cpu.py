@@ -151,3 +151,24 @@ def test_checkpoint_kill_resume_trajectory_parity(tmp_path):
     assert len(full) == 6 and len(resumed) == 3
     for a, b in zip(full[3:], resumed):
         assert abs(a - b) < 1e-4, (full, resumed)
+
+
+def test_validation_loop_cadence(tmp_path):
+    import json
+
+    cfg = base_cfg(
+        tmp_path,
+        step_scheduler={"grad_acc_steps": 1, "max_steps": 4, "val_every_steps": 2},
+        validation={"dataloader": {
+            "dataset": {"kind": "mock", "num_samples": 8, "seq_len": 16,
+                        "vocab_size": 128, "seed": 9},
+            "batch_size": 2, "shuffle": False,
+        }},
+    )
+    r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    r.setup()
+    r.run_train_validation_loop()
+    lines = [json.loads(x) for x in open(tmp_path / "out" / "training.jsonl")]
+    val_lines = [m for m in lines if "val_loss" in m]
+    assert len(val_lines) == 2          # steps 2 and 4
+    assert all(v["val_loss"] > 0 for v in val_lines)
