@@ -50,7 +50,10 @@ __device__ __forceinline__ int k_lds_off(int row, int byte_in_row) {
 // (d>>2)&3 separates the four colliding rows. (The previous (d&3) rotation
 // only moved rows that never collided.)
 __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
-  return d * 64 + (byte_in_row ^ (((d >> 2) & 3) << 4));
+  // ((d>>2)^(d>>3))&3: the read-colliding set {d, d+4, d+8, d+12} still maps
+  // to 4 distinct slots (conflict-free b128 reads) AND the transpose-store
+  // writers (d stepping 8) now spread over all 4 slots (8-way -> 4-way).
+  return d * 64 + (byte_in_row ^ ((((d >> 2) ^ (d >> 3)) & 3) << 4));
 }
 
 template <int D>
@@ -296,7 +299,7 @@ __global__ void fa_delta_kernel(const bf16* __restrict__ dout, const bf16* __res
 // qt/dot images:   [128][64 B], XOR swizzle ((d&3)<<4)   (same as vt_lds_off)
 // per-wave transpose buffer: [32][64 B] with ((row&3)<<4)
 __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
-  return row * 64 + (byte_in_row ^ (((row >> 2) & 3) << 4));
+  return row * 64 + (byte_in_row ^ ((((row >> 2) ^ (row >> 3)) & 3) << 4));
 }
 
 #define FAB_WAVES 4
