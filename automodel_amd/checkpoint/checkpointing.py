@@ -108,6 +108,7 @@ class Checkpointer:
         from safetensors.torch import save_file
 
         os.makedirs(out_dir, exist_ok=True)
+        self._export_config_json(model, out_dir)
         adapter = getattr(model, "state_dict_adapter", None)
         if adapter is not None:
             full_sd = adapter.to_hf(full_sd)
@@ -140,6 +141,25 @@ class Checkpointer:
         with open(os.path.join(out_dir, "model.safetensors.index.json"), "w") as f:
             _json.dump({"metadata": {"total_size": total_bytes},
                         "weight_map": weight_map}, f)
+
+    @staticmethod
+    def _export_config_json(model, out_dir: str) -> None:
+        """Write an HF-compatible config.json next to the weights so the
+        export round-trips through pretrained_path loading."""
+        import dataclasses
+        import json as _json
+
+        cfg = getattr(model, "config", None)
+        if cfg is None or not dataclasses.is_dataclass(cfg):
+            return
+        d = {}
+        for f in dataclasses.fields(cfg):
+            v = getattr(cfg, f.name)
+            if isinstance(v, (int, float, str, bool, type(None), list, dict)):
+                d[f.name] = v
+        d["architectures"] = [type(model).__name__]
+        with open(os.path.join(out_dir, "config.json"), "w") as fh:
+            _json.dump(d, fh, indent=1)
 
     # ------------------------------------------------------------- retention
     def _apply_retention(self) -> None:

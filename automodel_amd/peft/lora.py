@@ -37,18 +37,29 @@ class PeftConfig:
 
 
 class LinearLoRA(nn.Module):
-    """y = base(x) + (dropout(x) @ A^T) @ B^T * (alpha / r); base frozen."""
+    """y = base(x) + (dropout(x) @ A^T) @ B^T * (alpha / r); base frozen.
 
-    def __init__(self, base: nn.Linear, dim: int, alpha: float, dropout: float = 0.0):
+    use_dora: DoRA decomposition (reference _peft/lora.py:227) — a learned
+    per-output magnitude rescales the direction of W + BA.
+    """
+
+    def __init__(self, base: nn.Linear, dim: int, alpha: float, dropout: float = 0.0,
+                 use_dora: bool = False):
         super().__init__()
         self.base = base
         self.dim = dim
         self.scale = alpha / dim
+        self.use_dora = use_dora
         dtype = base.weight.dtype
         dev = base.weight.device
         self.lora_A = nn.Linear(base.in_features, dim, bias=False, dtype=dtype, device=dev)
         self.lora_B = nn.Linear(dim, base.out_features, bias=False, dtype=dtype, device=dev)
         self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
+        if use_dora:
+            with torch.no_grad():
+                mag = base.weight.float().norm(dim=1) if not base.weight.is_meta \
+                    else torch.ones(base.out_features)
+            self.lora_magnitude = nn.Parameter(mag.to(dtype=dtype, device=dev))
         self.reset_lora_parameters()
         self.base.weight.requires_grad_(False)
         if self.base.bias is not None:
@@ -60,6 +71,12 @@ class LinearLoRA(nn.Module):
             nn.init.zeros_(self.lora_B.weight)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.use_dora:
+            w = self.base.weight + (self.lora_B.weight @ self.lora_A.weight) * self.scale
+            col_norm = w.float().norm(dim=1).clamp_min(1e-6).to(w.dtype)
+            w = w * (self.lora_magnitude / col_norm).unsqueeze(1)
+            y = torch.nn.functional.linear(x, w, self.base.bias)
+            return y
         return self.base(x) + self.lora_B(self.lora_A(self.dropout(x))) * self.scale
 
     @torch.no_grad()
@@ -84,7 +101,8 @@ def apply_lora_to_linear_modules(model: nn.Module, cfg) -> int:
             if isinstance(child, nn.Linear) and not isinstance(child, LinearLoRA):
                 if any(fnmatch.fnmatch(full, pat) for pat in peft.target_modules):
                     setattr(module, child_name,
-                            LinearLoRA(child, peft.dim, peft.alpha, peft.dropout))
+                            LinearLoRA(child, peft.dim, peft.alpha, peft.dropout,
+                                       use_dora=peft.use_dora))
                     replaced += 1
     return replaced
 

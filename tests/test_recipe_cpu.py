@@ -172,3 +172,46 @@ def test_validation_loop_cadence(tmp_path):
     val_lines = [m for m in lines if "val_loss" in m]
     assert len(val_lines) == 2          # steps 2 and 4
     assert all(v["val_loss"] > 0 for v in val_lines)
+
+
+def test_hf_checkpoint_roundtrip(tmp_path):
+    """HF checkpoints in / HF checkpoints out (key architectural invariant):
+    consolidated safetensors export loads back through the pretrained_path
+    machinery into an identical model."""
+    import torch
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.checkpoint.hf_loader import load_hf_weights
+    from automodel_amd.models.registry import build_model
+
+    r = TrainFinetuneRecipeForNextTokenPrediction(base_cfg(tmp_path))
+    r.setup()
+    r.run_train_validation_loop()
+
+    out = str(tmp_path / "hf_export")
+    Checkpointer().export_hf_safetensors(r.model, out, rank=0)
+    m2 = build_model(pretrained_path=out, dtype="float32", meta_init=False)
+    load_hf_weights(m2, out, device="cpu")
+    ids = torch.randint(0, 128, (2, 8))
+    with torch.no_grad():
+        a = r.model(ids)
+        b = m2(ids)
+    assert torch.allclose(a, b, atol=1e-5), (a - b).abs().max()
+
+
+def test_dora_adapter():
+    import torch
+
+    from automodel_amd.peft.lora import LinearLoRA
+
+    base = torch.nn.Linear(16, 8, bias=False)
+    dora = LinearLoRA(base, dim=4, alpha=8, use_dora=True)
+    x = torch.randn(3, 16)
+    # B starts at zero and magnitude = ||W|| -> output == base initially
+    assert torch.allclose(dora(x), base(x), atol=1e-5)
+    assert dora.lora_magnitude.requires_grad
+    y = dora(x).sum()
+    y.backward()
+    assert dora.lora_A.weight.grad is not None
+    assert dora.lora_magnitude.grad is not None
+    assert base.weight.grad is None
