@@ -56,8 +56,8 @@ __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
   return d * 64 + (byte_in_row ^ ((((d >> 2) ^ (d >> 3)) & 3) << 4));
 }
 
-template <int D>
-__global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
+template <int D, int WAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
     bf16* __restrict__ out, float* __restrict__ lse, int B, int Sq, int Skv, int Hq,
     int Hk, int q_start, float scale, bool causal) {
@@ -73,7 +73,7 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
   const int col = lane & 31;           // q row within wave tile (QK layout)
   const int half = lane >> 5;
 
-  const int qblk0 = blockIdx.x * (FA_WAVES * QBLK);
+  const int qblk0 = blockIdx.x * (WAVES * QBLK);
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
@@ -101,7 +101,7 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = -1e30f, l_run = 0.f;
 
-  const int q_block_max = q_start + qblk0 + FA_WAVES * QBLK - 1;
+  const int q_block_max = q_start + qblk0 + WAVES * QBLK - 1;
   const int n_tiles = causal ? (min(q_block_max, Skv - 1) / KVBLK + 1)
                              : (Skv + KVBLK - 1) / KVBLK;
 
@@ -109,11 +109,10 @@ __global__ __launch_bounds__(FA_BLOCK, 2) void flash_fwd_kernel(
     const int k0 = jt * KVBLK;
     // ---- stage K tile (swizzled) and V^T tile cooperatively
     {
-      // 256 threads, tile = 32 rows x 128 cols: thread t -> row t/16+{0,16}, col (t%16)*8
-      const int r0 = tid / 16, c0 = (tid % 16) * 8;
+      // tile = 32 rows x 128 cols; 16 threads per row, WAVES*4 rows per pass
+      const int c0 = (tid % 16) * 8;
 #pragma unroll
-      for (int rr = 0; rr < 2; ++rr) {
-        const int row = r0 + rr * 16;
+      for (int row = tid / 16; row < KVBLK; row += WAVES * 4) {
         const int ks = min(k0 + row, Skv - 1);  // overhang rows masked later
         bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_row_stride + c0);
         *reinterpret_cast<bf16x8*>(k_lds + k_lds_off(row, c0 * 2)) = kv8;
@@ -251,15 +250,28 @@ std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at:
   TORCH_CHECK(Skv % KVBLK == 0, "flash_attn_fwd: Skv must be a multiple of 32");
   auto out = at::empty_like(q);
   auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
-  const dim3 grid(Sq / (FA_WAVES * QBLK), Hq, B);
-  const size_t smem = 2 * KVBLK * 128 * 2 + FA_WAVES * 32 * sizeof(float);
   auto stream = c10::hip::getCurrentHIPStream();
-  hipLaunchKernelGGL((flash_fwd_kernel<128>), grid, dim3(FA_BLOCK), smem, stream.stream(),
-                     reinterpret_cast<const bf16*>(q.data_ptr()),
-                     reinterpret_cast<const bf16*>(k.data_ptr()),
-                     reinterpret_cast<const bf16*>(v.data_ptr()),
-                     reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
-                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  // 8-wave blocks amortize the shared K/V staging over twice the waves;
+  // 4-wave fallback keeps small/CP-chunk shapes (Sq % 256 != 0) working.
+  if (Sq % (8 * QBLK) == 0) {
+    const dim3 grid(Sq / (8 * QBLK), Hq, B);
+    const size_t smem = 2 * KVBLK * 128 * 2 + 8 * 32 * sizeof(float);
+    hipLaunchKernelGGL((flash_fwd_kernel<128, 8>), grid, dim3(512), smem, stream.stream(),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  } else {
+    const dim3 grid(Sq / (4 * QBLK), Hq, B);
+    const size_t smem = 2 * KVBLK * 128 * 2 + 4 * 32 * sizeof(float);
+    hipLaunchKernelGGL((flash_fwd_kernel<128, 4>), grid, dim3(256), smem, stream.stream(),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  }
   HIP_CHECK_KERNEL();
   return {out, lse};
 }
@@ -304,8 +316,8 @@ __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
 
 #define FAB_WAVES 4
 
-template <int D>
-__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
+template <int D, int WAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     float* __restrict__ dk, float* __restrict__ dv, int B, int Sq, int Skv, int Hq,
@@ -324,7 +336,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
   const int col = lane & 31;
   const int half = lane >> 5;
 
-  const int kvb = blockIdx.x * (FAB_WAVES * 32);
+  const int kvb = blockIdx.x * (WAVES * 32);
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
@@ -358,8 +370,8 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
     const int q0 = jq * 32;
     // ---- cooperative stage: q/do rows + transposed images
     {
-      const int r0 = tid / 16, c0 = (tid % 16) * 8;   // 16 threads per row of 128
-      for (int rr = r0; rr < 32; rr += 16) {
+      const int c0 = (tid % 16) * 8;   // 16 threads per row of 128
+      for (int rr = tid / 16; rr < 32; rr += WAVES * 4) {
         bf16x8 qv = *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
         bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(dout + q_base + (long)(q0 + rr) * q_rs + c0);
         *reinterpret_cast<bf16x8*>(q_rows + k_lds_off(rr, c0 * 2)) = qv;
@@ -468,8 +480,8 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel
   }
 }
 
-template <int D>
-__global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
+template <int D, int WAVES>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, int B, int Sq, int Skv, int Hq, int Hk, int q_start,
@@ -479,8 +491,8 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
   char* k_rows = smem;                 // 8 KiB ([32][256B] swz)
   char* v_rows = smem + 8 * 1024;      // 8 KiB
   char* kt = smem + 16 * 1024;         // 8 KiB ([128][64B] swz)
-  char* tbuf = smem + 24 * 1024;       // per-wave 2 KiB x4
-  float* stats = reinterpret_cast<float*>(smem + 32 * 1024);  // lse[128]+delta[128]
+  char* tbuf = smem + 24 * 1024;       // per-wave 2 KiB x WAVES
+  float* stats = reinterpret_cast<float*>(smem + 24 * 1024 + WAVES * 2048);
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -488,7 +500,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
   const int col = lane & 31;
   const int half = lane >> 5;
 
-  const int qb_blk = blockIdx.x * (FAB_WAVES * 32);
+  const int qb_blk = blockIdx.x * (WAVES * 32);
   const int h = blockIdx.y;
   const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
@@ -508,10 +520,10 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     qfrag[c] = *reinterpret_cast<const bf16x8_v*>(q + off);
     dofrag[c] = *reinterpret_cast<const bf16x8_v*>(dout + off);
   }
-  // lse/delta for the block's 128 q rows -> LDS (32 regs saved per lane)
-  if (tid < 128) {
+  // lse/delta for the block's q rows -> LDS (32 regs saved per lane)
+  if (tid < WAVES * 32) {
     stats[tid] = lse_row[qb_blk + tid];
-    stats[128 + tid] = dlt_row[qb_blk + tid];
+    stats[WAVES * 32 + tid] = dlt_row[qb_blk + tid];
   }
 
   f32x16 dq_acc[D / 32];
@@ -521,15 +533,15 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
   char* tb = tbuf + wid * 2048;
-  const int block_q_max = q_start + qb_blk + FAB_WAVES * 32 - 1;
+  const int block_q_max = q_start + qb_blk + WAVES * 32 - 1;
   const int n_tiles = causal ? (min(block_q_max, Skv - 1) / 32 + 1) : (Skv + 31) / 32;
 
   for (int jk = 0; jk < n_tiles; ++jk) {
     const int k0 = jk * 32;
     // ---- stage K rows, V rows, K^T
     {
-      const int r0 = tid / 16, c0 = (tid % 16) * 8;
-      for (int rr = r0; rr < 32; rr += 16) {
+      const int c0 = (tid % 16) * 8;
+      for (int rr = tid / 16; rr < 32; rr += WAVES * 4) {
         const int ks = min(k0 + rr, Skv - 1);  // overhang masked in compute
         bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_rs + c0);
         bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)ks * kv_rs + c0);
@@ -572,7 +584,7 @@ __global__ __launch_bounds__(FAB_WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
         const int qg = q_start + q0 + qrow;
         const bool masked = (causal && kg > qg) || kg >= Skv;
         const float p = masked ? 0.f : __expf(s[r] * scale - stats[wid * 32 + qrow]);
-        s[r] = p * (dp[r] - stats[128 + wid * 32 + qrow]) * scale;  // dS in place
+        s[r] = p * (dp[r] - stats[WAVES * 32 + wid * 32 + qrow]) * scale;  // dS in place
       }
 
       // ---- transpose dS (cols=k -> A-frag rows=q), accumulate dQ
@@ -637,28 +649,58 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
   auto dk32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
   auto dv32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
 
-  const dim3 grid_kv(Skv / 128, Hq, B);
-  const size_t smem_a = 40 * 1024;
-  hipLaunchKernelGGL((flash_bwd_dkv_kernel<128>), grid_kv, dim3(256), smem_a, stream.stream(),
-                     reinterpret_cast<const bf16*>(dout.data_ptr()),
-                     reinterpret_cast<const bf16*>(q.data_ptr()),
-                     reinterpret_cast<const bf16*>(k.data_ptr()),
-                     reinterpret_cast<const bf16*>(v.data_ptr()),
-                     lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  if (Skv % 256 == 0) {
+    const dim3 grid_kv(Skv / 256, Hq, B);
+    const size_t smem_a = 32 * 1024 + 8 * 2048;
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<128, 8>), grid_kv, dim3(512), smem_a,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dout.data_ptr()),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       dk32.data_ptr<float>(), dv32.data_ptr<float>(),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  } else {
+    const dim3 grid_kv(Skv / 128, Hq, B);
+    const size_t smem_a = 32 * 1024 + 4 * 2048;
+    hipLaunchKernelGGL((flash_bwd_dkv_kernel<128, 4>), grid_kv, dim3(256), smem_a,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dout.data_ptr()),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       dk32.data_ptr<float>(), dv32.data_ptr<float>(),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  }
   HIP_CHECK_KERNEL();
 
-  const dim3 grid_q(Sq / 128, Hq, B);
-  const size_t smem_b = 33 * 1024;
-  hipLaunchKernelGGL((flash_bwd_dq_kernel<128>), grid_q, dim3(256), smem_b, stream.stream(),
-                     reinterpret_cast<const bf16*>(dout.data_ptr()),
-                     reinterpret_cast<const bf16*>(q.data_ptr()),
-                     reinterpret_cast<const bf16*>(k.data_ptr()),
-                     reinterpret_cast<const bf16*>(v.data_ptr()),
-                     lse.data_ptr<float>(), delta.data_ptr<float>(),
-                     reinterpret_cast<bf16*>(dq.data_ptr()),
-                     B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  if (Sq % 256 == 0) {
+    const dim3 grid_q(Sq / 256, Hq, B);
+    const size_t smem_b = 24 * 1024 + 8 * 2048 + 8 * 64 * sizeof(float);
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<128, 8>), grid_q, dim3(512), smem_b,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dout.data_ptr()),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       reinterpret_cast<bf16*>(dq.data_ptr()),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  } else {
+    const dim3 grid_q(Sq / 128, Hq, B);
+    const size_t smem_b = 24 * 1024 + 4 * 2048 + 4 * 64 * sizeof(float);
+    hipLaunchKernelGGL((flash_bwd_dq_kernel<128, 4>), grid_q, dim3(256), smem_b,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(dout.data_ptr()),
+                       reinterpret_cast<const bf16*>(q.data_ptr()),
+                       reinterpret_cast<const bf16*>(k.data_ptr()),
+                       reinterpret_cast<const bf16*>(v.data_ptr()),
+                       lse.data_ptr<float>(), delta.data_ptr<float>(),
+                       reinterpret_cast<bf16*>(dq.data_ptr()),
+                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  }
   HIP_CHECK_KERNEL();
 
   return {dq, dk32.to(at::kBFloat16), dv32.to(at::kBFloat16)};
