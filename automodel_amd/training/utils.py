@@ -13,27 +13,40 @@ import torch.distributed as dist
 from torch.distributed.tensor import DTensor
 
 
-def clip_grad_norm_(parameters, max_norm: float, group=None) -> torch.Tensor:
-    """Global 2-norm clip correct for DTensor (FSDP2-sharded) grads.
+def _grad_sq_contribution(g: torch.Tensor, world: int) -> torch.Tensor:
+    """This rank's contribution to the global sq-norm such that a WORLD
+    all-reduce counts every element exactly once (reference
+    training/utils.py:400: replication-aware grad-norm)."""
+    if isinstance(g, DTensor):
+        local = g.to_local()
+        shard_ranks = 1
+        for dim_size, placement in zip(g.device_mesh.shape, g.placements):
+            if placement.is_shard():
+                shard_ranks *= dim_size
+        replicas = max(1, world // shard_ranks)
+        return local.float().pow(2).sum() / replicas
+    # plain tensors are replicated on every rank
+    sq = g.float().pow(2).sum()
+    return sq / world if world > 1 else sq
 
-    Local shard sq-sums are all-reduced over ``group`` (default: WORLD, which
-    is correct for pure FSDP sharding where every rank holds a disjoint shard).
-    """
+
+def clip_grad_norm_(parameters, max_norm: float, group=None) -> torch.Tensor:
+    """Global 2-norm clip correct for DTensor grads across mixed shardings:
+    sharded elements are counted once via their owning shard; replicated
+    elements are divided by their replica count before the all-reduce."""
     params = [p for p in parameters if p.grad is not None]
     if not params:
         return torch.tensor(0.0)
     device = params[0].grad.device
+    world = dist.get_world_size(group) if dist.is_initialized() else 1
     total_sq = torch.zeros((), dtype=torch.float32, device=device)
     for p in params:
-        g = p.grad
-        local = g.to_local() if isinstance(g, DTensor) else g
-        total_sq += local.float().pow(2).sum()
-    if dist.is_initialized() and dist.get_world_size(group) > 1:
+        total_sq += _grad_sq_contribution(p.grad, world)
+    if dist.is_initialized() and world > 1:
         dist.all_reduce(total_sq, group=group)
     total_norm = total_sq.sqrt()
     if max_norm is not None and max_norm > 0:
-        clip_coef = max_norm / (total_norm + 1e-6)
-        clip_coef = torch.clamp(clip_coef, max=1.0)
+        clip_coef = torch.clamp(max_norm / (total_norm + 1e-6), max=1.0)
         for p in params:
             g = p.grad
             local = g.to_local() if isinstance(g, DTensor) else g

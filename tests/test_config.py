@@ -76,3 +76,21 @@ def test_secret_redaction():
     cfg = ConfigNode({"wandb": {"api_key": "s3cret"}, "lr": 1.0})
     red = cfg.redacted_dict()
     assert red["wandb"]["api_key"] == "***" and red["lr"] == 1.0
+
+
+def test_typed_recipe_config():
+    from automodel_amd.recipes.typed_config import RecipeConfig
+
+    cfg = ConfigNode({
+        "distributed": {"tp": 2, "cp": 1},
+        "optimizer": {"lr": 3e-4, "unknown_key_ignored": 1},
+        "step_scheduler": {"max_steps": 7},
+        "loss_fn": {"backend": "chunked", "chunk_size": 64},
+    })
+    rc = RecipeConfig(cfg)
+    assert rc.distributed.tp == 2 and rc.distributed.dp_shard == -1
+    assert rc.optimizer.lr == 3e-4
+    sched = rc.step_scheduler.build()
+    assert sched.max_steps == 7
+    loss = rc.loss.build()
+    assert loss.backend == "chunked"

@@ -243,3 +243,33 @@ def test_activation_checkpointing_grads_match():
     g1 = m1.model.layers[0].mlp.down_proj.weight.grad
     g2 = m2.model.layers[0].mlp.down_proj.weight.grad
     assert torch.allclose(g1, g2, atol=1e-5)
+
+
+def _gradnorm_tp_fn(rank, world):
+    from automodel_amd.parallel.mesh import build_mesh
+    from automodel_amd.parallel.tp import apply_tp
+    from automodel_amd.training.utils import clip_grad_norm_
+
+    model = _make_model(seed=13)
+    inp, lab = _make_batch(seed=14)
+    logits = model(inp)
+    loss = torch.nn.functional.cross_entropy(
+        logits.reshape(-1, 128), lab.reshape(-1), reduction="sum")
+    loss.backward()
+    ref_norm = float(clip_grad_norm_(model.parameters(), max_norm=0.0))
+
+    tp_model = _make_model(seed=13)
+    ctx = build_mesh(dp_shard=1, tp=2, device_type="cpu")
+    apply_tp(tp_model, ctx["tp"])
+    logits2 = tp_model(inp)
+    loss2 = torch.nn.functional.cross_entropy(
+        logits2.reshape(-1, 128), lab.reshape(-1), reduction="sum")
+    loss2.backward()
+    tp_norm = float(clip_grad_norm_(tp_model.parameters(), max_norm=0.0))
+    assert abs(tp_norm - ref_norm) / ref_norm < 1e-3, (tp_norm, ref_norm)
+    return tp_norm
+
+
+def test_grad_norm_tp2_matches_single_rank():
+    out = run_distributed(_gradnorm_tp_fn, world=2)
+    assert abs(out[0] - out[1]) < 1e-4
