@@ -43,6 +43,51 @@ def apply_ep(model: nn.Module, ep_mesh, dispatcher: str = "a2a") -> nn.Module:
     return model
 
 
+def parallelize_moe_model(
+    model: nn.Module,
+    ep: int,
+    dispatcher: str = "a2a",
+    device_type: str | None = None,
+    param_dtype=None,
+    reshard_after_forward: bool = False,
+) -> nn.Module:
+    """Compose EP with FSDP (reference moe/parallelizer.py:978 parallelize_model):
+
+      1. carve the MoE mesh (ep_shard, ep) from the world,
+      2. slice experts over the ep axis + attach the RCCL a2a dispatcher,
+      3. fully_shard expert modules over ep_shard (each expert shard is
+         further sharded only across ranks that hold the SAME experts),
+      4. fully_shard the decoder layers + root over the full dp world
+         (expert params are already managed by their inner FSDP module).
+    """
+    import torch.distributed as dist
+    from torch.distributed.device_mesh import init_device_mesh
+    from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
+
+    from automodel_amd.parallel.fsdp import apply_fsdp
+
+    world = dist.get_world_size()
+    assert world % ep == 0, f"world {world} not divisible by ep {ep}"
+    ep_shard = world // ep
+    if device_type is None:
+        device_type = "cuda" if torch.cuda.is_available() else "cpu"
+    moe_mesh = init_device_mesh(device_type, (ep_shard, ep),
+                                mesh_dim_names=("ep_shard", "ep"))
+    apply_ep(model, moe_mesh["ep"], dispatcher=dispatcher)
+
+    param_dtype = param_dtype or (torch.bfloat16 if device_type == "cuda" else torch.float32)
+    mp = MixedPrecisionPolicy(param_dtype=param_dtype, reduce_dtype=torch.float32)
+    if ep_shard > 1:
+        for m in model.modules():
+            if isinstance(m, MoE):
+                fully_shard(m.experts, mesh=moe_mesh["ep_shard"], mp_policy=mp,
+                            reshard_after_forward=reshard_after_forward)
+    dp_mesh = init_device_mesh(device_type, (world,), mesh_dim_names=("dp",))
+    apply_fsdp(model, dp_mesh["dp"], param_dtype=param_dtype,
+               reshard_after_forward=reshard_after_forward)
+    return model
+
+
 def moe_param_groups(model: nn.Module) -> tuple[list, list]:
     """(expert_params, dense_params) — expert grads are NOT reduced over dp
     when EP spans the dp group (each rank owns distinct experts)."""

@@ -62,3 +62,29 @@ def moe_flops_per_token(
     per_layer = attn_proj + attn_scores + expert_mlp + shared_mlp
     fwd = 2 * (layers * per_layer + hidden * vocab)
     return 3.0 * fwd
+
+
+def deepseek_v3_flops_per_token(cfg, seq_len: int) -> float:
+    """Training FLOPs/token for MLA + MoE models (reference flops_utils.py:437).
+    cfg: DeepseekV3Config."""
+    H = cfg.hidden_size
+    nh = cfg.num_attention_heads
+    qk, vd = cfg.qk_head_dim, cfg.v_head_dim
+    # MLA projections per token
+    q_proj = (H * cfg.q_lora_rank + cfg.q_lora_rank * nh * qk) if cfg.q_lora_rank \
+        else H * nh * qk
+    kv_proj = H * (cfg.kv_lora_rank + cfg.qk_rope_head_dim) + \
+        cfg.kv_lora_rank * nh * (cfg.qk_nope_head_dim + vd)
+    o_proj = nh * vd * H
+    scores = 2 * (seq_len / 2) * nh * (qk + vd) / 2  # QK^T + PV at avg ctx
+    moe = cfg.moe
+    dense_layers = cfg.first_k_dense_replace
+    moe_layers = cfg.num_hidden_layers - dense_layers
+    dense_mlp = 3 * H * cfg.intermediate_size
+    inter = moe.moe_intermediate_size or H
+    moe_mlp = 3 * H * inter * moe.n_activated_experts + \
+        3 * H * (moe.shared_expert_intermediate_size or inter) * (1 if moe.n_shared_experts else 0)
+    attn = q_proj + kv_proj + o_proj + scores
+    total = (dense_layers * (attn + dense_mlp) + moe_layers * (attn + moe_mlp)
+             + H * cfg.vocab_size)
+    return 3.0 * 2 * total
