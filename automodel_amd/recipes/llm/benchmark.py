@@ -74,11 +74,21 @@ class BenchmarkingRecipeForNextTokenPrediction(TrainFinetuneRecipeForNextTokenPr
             * self.seq_len
             * self.mesh.dp_size
         )
-        fpt = llama_flops_per_token(
-            cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
-            cfg.vocab_size, self.seq_len, cfg.num_attention_heads,
-            cfg.num_key_value_heads, cfg.head_dim,
-        )
+        if hasattr(cfg, "moe"):
+            from automodel_amd.utils.flops import moe_flops_per_token
+            fpt = moe_flops_per_token(
+                cfg.hidden_size, cfg.num_hidden_layers, cfg.vocab_size,
+                self.seq_len, cfg.num_attention_heads, cfg.num_key_value_heads,
+                cfg.moe.moe_intermediate_size or cfg.intermediate_size,
+                cfg.moe.n_activated_experts, cfg.moe.n_shared_experts,
+                cfg.moe.shared_expert_intermediate_size, cfg.head_dim,
+            )
+        else:
+            fpt = llama_flops_per_token(
+                cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
+                cfg.vocab_size, self.seq_len, cfg.num_attention_heads,
+                cfg.num_key_value_heads, cfg.head_dim,
+            )
         tps = tokens_per_step / mean_t
         summary = {
             "mean_step_time_s": round(mean_t, 4),
