@@ -58,6 +58,9 @@ TORCH_LIBRARY(amd_ops, m) {
   m.def("soft_ce_bwd(Tensor s, Tensor t, Tensor lse_s, Tensor lse_t, Tensor dloss) -> Tensor");
   m.impl("soft_ce_bwd", &amd_ops::soft_ce_bwd);
 
+  m.def("nf4_dequant(Tensor packed, Tensor absmax, int block_size, int rows, int cols) -> Tensor");
+  m.impl("nf4_dequant", &amd_ops::nf4_dequant);
+
   m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
   m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");

@@ -26,6 +26,8 @@ class PeftConfig:
     alpha: float = 16.0
     dropout: float = 0.0
     use_dora: bool = False
+    quantize_base: bool = False       # QLoRA: NF4-quantize matched base linears
+    quant_block_size: int = 64
 
     @classmethod
     def from_config(cls, cfg) -> "PeftConfig":
@@ -43,27 +45,35 @@ class LinearLoRA(nn.Module):
     per-output magnitude rescales the direction of W + BA.
     """
 
-    def __init__(self, base: nn.Linear, dim: int, alpha: float, dropout: float = 0.0,
+    def __init__(self, base: nn.Module, dim: int, alpha: float, dropout: float = 0.0,
                  use_dora: bool = False):
         super().__init__()
         self.base = base
         self.dim = dim
         self.scale = alpha / dim
         self.use_dora = use_dora
-        dtype = base.weight.dtype
-        dev = base.weight.device
+        base_w = self._base_weight(materialize=use_dora)
+        dtype = base_w.dtype if base_w is not None else torch.float32
+        dev = base_w.device if base_w is not None else next(iter(base.buffers())).device
         self.lora_A = nn.Linear(base.in_features, dim, bias=False, dtype=dtype, device=dev)
         self.lora_B = nn.Linear(dim, base.out_features, bias=False, dtype=dtype, device=dev)
         self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()
         if use_dora:
             with torch.no_grad():
-                mag = base.weight.float().norm(dim=1) if not base.weight.is_meta \
+                mag = base_w.float().norm(dim=1) if not base_w.is_meta \
                     else torch.ones(base.out_features)
             self.lora_magnitude = nn.Parameter(mag.to(dtype=dtype, device=dev))
         self.reset_lora_parameters()
-        self.base.weight.requires_grad_(False)
-        if self.base.bias is not None:
-            self.base.bias.requires_grad_(False)
+        for p in self.base.parameters():
+            p.requires_grad_(False)
+
+    def _base_weight(self, materialize: bool = True):
+        """The effective base weight. For an NF4Linear base (QLoRA) this is
+        the dequantized matrix (only materialized when needed)."""
+        w = getattr(self.base, "weight", None)
+        if w is not None:
+            return w
+        return self.base.dequantized_weight() if materialize else None
 
     def reset_lora_parameters(self) -> None:
         if not self.lora_A.weight.is_meta:
@@ -72,7 +82,7 @@ class LinearLoRA(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.use_dora:
-            w = self.base.weight + (self.lora_B.weight @ self.lora_A.weight) * self.scale
+            w = self._base_weight() + (self.lora_B.weight @ self.lora_A.weight) * self.scale
             col_norm = w.float().norm(dim=1).clamp_min(1e-6).to(w.dtype)
             w = w * (self.lora_magnitude / col_norm).unsqueeze(1)
             y = torch.nn.functional.linear(x, w, self.base.bias)
@@ -81,8 +91,18 @@ class LinearLoRA(nn.Module):
 
     @torch.no_grad()
     def merge(self) -> nn.Linear:
-        """Fold the adapter into the base weight and return the plain Linear."""
+        """Fold the adapter into the base weight and return the plain Linear.
+        A quantized (NF4) base is dequantized first — merge de-quantizes."""
         delta = (self.lora_B.weight @ self.lora_A.weight) * self.scale
+        if not isinstance(self.base, nn.Linear):
+            w = self._base_weight()
+            lin = nn.Linear(self.base.in_features, self.base.out_features,
+                            bias=self.base.bias is not None,
+                            dtype=w.dtype, device=w.device)
+            lin.weight.copy_(w)
+            if self.base.bias is not None:
+                lin.bias.copy_(self.base.bias)
+            self.base = lin
         self.base.weight += delta.to(self.base.weight.dtype)
         return self.base
 
@@ -100,6 +120,10 @@ def apply_lora_to_linear_modules(model: nn.Module, cfg) -> int:
             full = f"{name}.{child_name}" if name else child_name
             if isinstance(child, nn.Linear) and not isinstance(child, LinearLoRA):
                 if any(fnmatch.fnmatch(full, pat) for pat in peft.target_modules):
+                    if peft.quantize_base:
+                        from automodel_amd.quantization.nf4 import NF4Linear
+
+                        child = NF4Linear(child, peft.quant_block_size)
                     setattr(module, child_name,
                             LinearLoRA(child, peft.dim, peft.alpha, peft.dropout,
                                        use_dora=peft.use_dora))
