@@ -185,3 +185,33 @@ def test_qwen3_qk_norm_model():
     assert hasattr(m.model.layers[0].self_attn, "q_norm")
     out = m(torch.randint(0, 64, (1, 8)))
     assert out.shape == (1, 8, 64)
+
+
+def test_skypilot_launcher_renders_task(tmp_path):
+    from automodel_amd.launcher.skypilot import SkyPilotConfig, SkyPilotLauncher
+
+    l = SkyPilotLauncher(cloud="kubernetes", accelerators="MI355X:8", num_nodes=2,
+                         job_name="ft")
+    task = l.render_task("cfg.yaml", "automodel_amd.recipes.llm.train_ft",
+                         ["--optimizer.lr=1e-4"])
+    assert task["num_nodes"] == 2 and task["resources"]["accelerators"] == "MI355X:8"
+    assert task["envs"]["HSA_ENABLE_IPC_MODE_LEGACY"] == "0"
+    assert "--nproc-per-node=8" in task["run"]
+    assert "--node-rank=$SKYPILOT_NODE_RANK" in task["run"]
+    assert "--optimizer.lr=1e-4" in task["run"]
+
+    # single node -> standalone rendezvous
+    single = SkyPilotLauncher(accelerators="MI355X:4").render_task("c.yaml", "t")
+    assert "--standalone" in single["run"]
+
+    # task file written, valid YAML, no submission without `sky`
+    p = l.launch("cfg.yaml", "t", task_path=str(tmp_path / "task.yaml"), submit=False)
+    import yaml as _y
+
+    loaded = _y.safe_load(open(p))
+    assert loaded["name"] == "ft"
+
+    import pytest as _pt
+
+    with _pt.raises(ValueError):
+        SkyPilotConfig(cloud="notacloud")
