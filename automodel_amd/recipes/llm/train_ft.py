@@ -156,6 +156,15 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             from automodel_amd.peft.lora import apply_lora_to_linear_modules
             apply_lora_to_linear_modules(self.model, peft_cfg)
 
+        # ---- QAT (reference train_ft.py:218 applies after PEFT/model build)
+        qat_cfg = cfg.get("qat")
+        self.qat_enabled = bool(qat_cfg and qat_cfg.get("enabled", True))
+        if self.qat_enabled:
+            from automodel_amd.quantization.qat import prepare_qat
+            n_qat = prepare_qat(self.model, qat_cfg)
+            if self.rank == 0:
+                self.logger.info(f"QAT: fake-quantizing {n_qat} linear modules")
+
         # ---- optimizer / schedulers
         opt_cfg = cfg.get("optimizer", ConfigNode())
         self.optimizer = opt_cfg.maybe_instantiate(model=self.model) or build_adamw(
@@ -382,6 +391,9 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         for epoch in self.step_scheduler.epochs:
             self.train_loader.set_epoch(epoch)
             for batches in self.step_scheduler:
+                if getattr(self, "qat_enabled", False):
+                    from automodel_amd.quantization.qat import maybe_enable_delayed_fake_quant
+                    maybe_enable_delayed_fake_quant(self.model, self.step_scheduler.step)
                 metrics = self._run_train_optim_step(batches)
                 self.metrics.log(metrics)
                 if self.rank == 0:
