@@ -81,7 +81,7 @@ class NF4Linear(nn.Module):
         if self.weight_packed.is_cuda:
             from automodel_amd.ops._backend import require_ops
 
-            require_ops("nf4_dequant")
+            require_ops()
             return torch.ops.amd_ops.nf4_dequant(
                 self.weight_packed, self.weight_absmax, self.block_size,
                 self.out_features, self.in_features,
