@@ -241,6 +241,35 @@ class LlamaForCausalLM(nn.Module):
             return hidden
         return self.lm_head(hidden)
 
+    @torch.no_grad()
+    def forward_with_aux(
+        self, input_ids: torch.Tensor, aux_layers: tuple[int, ...] | None = None
+    ) -> tuple[torch.Tensor, list[torch.Tensor]]:
+        """-> (logits, [aux hidden [B,S,H] per requested layer]). Used by the
+        EAGLE draft trainer (speculative/train_draft.py); default aux layers
+        are low/mid/high like EAGLE-3's 3 auxiliary states."""
+        m = self.model
+        L = len(m.layers)
+        if aux_layers is None:
+            want = [max(0, L // 4), L // 2, max(0, L - 2)]
+            picked: list[int] = []
+            for w in want:  # dedupe for small L, keeping 3 distinct layers
+                while w in picked and w < L - 1:
+                    w += 1
+                picked.append(min(w, L - 1))
+            aux_layers = tuple(dict.fromkeys(picked))
+        x = m.embed_tokens(input_ids)
+        S = input_ids.shape[1]
+        cos, sin = m.rope_cos[:S], m.rope_sin[:S]
+        if cos.dtype != torch.float32:
+            cos, sin = cos.float(), sin.float()
+        aux = []
+        for i, layer in enumerate(m.layers):
+            x = layer(x, cos, sin)
+            if i in aux_layers:
+                aux.append(x)
+        return self.lm_head(m.norm(x)), aux
+
     # -- init ------------------------------------------------------------------
     @torch.no_grad()
     def init_weights(self, device=None) -> None:
