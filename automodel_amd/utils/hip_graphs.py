@@ -62,3 +62,28 @@ def apply_partial_graphs(model: nn.Module, cls_names: tuple[str, ...],
                 setattr(module, child_name, GraphedForward(child, warmup))
                 n += 1
     return n
+
+
+def apply_training_graphs(model: nn.Module, cls_names: tuple[str, ...],
+                          sample_input: torch.Tensor, warmup: int = 3) -> int:
+    """hipGraph-capture matching submodules for TRAINING (forward+backward
+    replay) via torch.cuda.make_graphed_callables — the reference's partial
+    CUDA-graph manager equivalent for the train loop. ``sample_input`` must
+    match the static per-module input shape/dtype/device. Returns count."""
+    targets = []
+    for name, module in model.named_modules():
+        for child_name, child in module.named_children():
+            if type(child).__name__ in cls_names:
+                targets.append((module, child_name, child))
+    if not targets:
+        return 0
+    graphed = torch.cuda.make_graphed_callables(
+        tuple(t[2] for t in targets),
+        tuple((sample_input.clone().requires_grad_(True),) for _ in targets),
+        num_warmup_iters=warmup,
+    )
+    if not isinstance(graphed, tuple):
+        graphed = (graphed,)
+    for (module, child_name, _), g in zip(targets, graphed):
+        setattr(module, child_name, g)
+    return len(targets)

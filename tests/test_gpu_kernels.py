@@ -419,3 +419,36 @@ def test_hip_graph_capture_replay():
     with torch.no_grad():
         y2 = g(x2).clone()
         assert torch.allclose(y2.float(), mod(x2).float(), atol=1e-2)
+
+
+def test_hip_graph_training_capture():
+    """Training-mode hipGraph capture (make_graphed_callables): fwd+bwd
+    replay matches eager gradients."""
+    from automodel_amd.utils.hip_graphs import apply_training_graphs
+
+    torch.manual_seed(1)
+
+    class Block(torch.nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.fc1 = torch.nn.Linear(64, 128, device="cuda")
+            self.fc2 = torch.nn.Linear(128, 64, device="cuda")
+
+        def forward(self, x):
+            return self.fc2(torch.nn.functional.gelu(self.fc1(x)))
+
+    model = torch.nn.Sequential(Block(), Block())
+    ref = torch.nn.Sequential(Block(), Block())
+    ref.load_state_dict(model.state_dict())
+
+    sample = torch.randn(8, 64, device="cuda")
+    n = apply_training_graphs(model, ("Block",), sample)
+    assert n == 2
+
+    x = torch.randn(8, 64, device="cuda", requires_grad=True)
+    y = model(x)
+    y.sum().backward()
+    xr = x.detach().clone().requires_grad_(True)
+    ref(xr).sum().backward()
+    assert torch.allclose(y.float(), ref(xr).float(), atol=1e-4)
+    assert torch.allclose(x.grad, xr.grad, atol=1e-4)
