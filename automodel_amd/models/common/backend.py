@@ -8,7 +8,7 @@ plain torch eager ("torch"), or torch SDPA for attention ("sdpa").
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass, field, replace
 
 
 @dataclass
@@ -38,12 +38,28 @@ class BackendConfig:
             dispatcher="torch",
         )
 
+    # head dims the in-tree flash kernels are tiled for (flash_attn.hip)
+    FLASH_HEAD_DIMS = (128,)
+
     @classmethod
-    def resolve(cls, cfg: "BackendConfig | dict | None", device_type: str) -> "BackendConfig":
+    def resolve(cls, cfg: "BackendConfig | dict | None", device_type: str,
+                head_dim: int | None = None) -> "BackendConfig":
         if cfg is None:
             cfg = cls()
         elif isinstance(cfg, dict):
             cfg = cls(**cfg)
         if device_type != "cuda":
             return cfg.for_cpu()
+        if (head_dim is not None and cfg.attn == "hip"
+                and head_dim not in cls.FLASH_HEAD_DIMS):
+            # explicit, visible downgrade at model build — NOT a silent
+            # runtime fallback (the kernel itself still fails loudly)
+            import warnings
+
+            warnings.warn(
+                f"head_dim {head_dim} not in flash kernel dims "
+                f"{cls.FLASH_HEAD_DIMS}; using sdpa attention for this model",
+                stacklevel=2,
+            )
+            cfg = replace(cfg, attn="sdpa")
         return cfg

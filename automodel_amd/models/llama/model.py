@@ -203,7 +203,8 @@ class LlamaForCausalLM(nn.Module):
             config = LlamaConfig(**config)
         device_type = "cuda" if torch.cuda.is_available() else "cpu"
         backend = BackendConfig.resolve(
-            backend if not isinstance(backend, dict) else BackendConfig(**backend), device_type
+            backend if not isinstance(backend, dict) else BackendConfig(**backend),
+            device_type, head_dim=config.head_dim,
         )
         self.config = config
         self.backend = backend

@@ -124,3 +124,18 @@ def test_fused_projections_match_unfused():
     assert "model.layers.0.self_attn.q_proj.weight" in hf
     assert torch.allclose(hf["model.layers.0.mlp.up_proj.weight"],
                           base.state_dict()["model.layers.0.mlp.up_proj.weight"])
+
+
+def test_backend_resolve_head_dim_guard():
+    """hd != 128 on GPU downgrades attn to sdpa at build time (visible
+    warning), never a silent runtime fallback inside the kernel."""
+    import warnings
+
+    from automodel_amd.models.common.backend import BackendConfig
+
+    with warnings.catch_warnings(record=True) as w:
+        warnings.simplefilter("always")
+        b = BackendConfig.resolve(None, "cuda", head_dim=64)
+    assert b.attn == "sdpa"
+    assert any("head_dim 64" in str(x.message) for x in w)
+    assert BackendConfig.resolve(None, "cuda", head_dim=128).attn == "hip"
