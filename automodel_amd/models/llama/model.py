@@ -121,7 +121,16 @@ class LlamaAttention(nn.Module):
         if active_cp() is not None:
             o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
         else:
-            o = flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+            from automodel_amd.utils.kv_cache import active_kv_cache, maybe_update_kv
+
+            cached = active_kv_cache() is not None
+            k, v, cache_pos = maybe_update_kv(k, v)
+            # cached decode: q is short (often 1) and prompts are arbitrary
+            # lengths — sdpa is the right tool (GEMV-bound; the flash
+            # kernel's 128-row tiling wants training shapes)
+            attn_backend = "sdpa" if cached else self.backend.attn
+            o = flash_attention(q, k, v, causal=True, backend=attn_backend,
+                                q_start=cache_pos)
         return self.o_proj(o.reshape(B, S, -1))
 
 

@@ -135,3 +135,35 @@ def test_train_draft_recipe_learns():
     # distillation loss must drop on this tiny repeated dataset
     assert logs[-1]["loss"] < logs[0]["loss"]
     assert all(not p.requires_grad for p in r.target.parameters())
+
+
+def test_generate_cached_matches_uncached():
+    from automodel_amd.utils.generation import generate_cached
+
+    target = _target()
+    prompt = torch.randint(0, 150, (2, 11))
+    a = generate(target, prompt, max_new_tokens=20)
+    b = generate_cached(target, prompt, max_new_tokens=20)
+    torch.testing.assert_close(a, b)
+
+
+def test_kv_cache_overflow_guard():
+    from automodel_amd.utils.kv_cache import KVCache, kv_cache_context
+
+    target = _target()
+    cache = KVCache.for_model(target, batch=1, max_len=8)
+    ids = torch.randint(0, 150, (1, 16))
+    with kv_cache_context(cache):
+        cache.begin_forward()
+        with pytest.raises(AssertionError):
+            target(ids)  # 16 > max_len 8
+
+
+def test_kv_cache_eos_early_stop():
+    from automodel_amd.utils.generation import generate_cached
+
+    target = _target()
+    prompt = torch.randint(0, 150, (1, 5))
+    ref = generate(target, prompt, max_new_tokens=10, eos_token_id=3)
+    out = generate_cached(target, prompt, max_new_tokens=10, eos_token_id=3)
+    torch.testing.assert_close(ref, out)
