@@ -79,11 +79,10 @@ class NF4Linear(nn.Module):
     def dequantized_weight(self, dtype=None) -> torch.Tensor:
         dtype = dtype or (torch.bfloat16 if self.weight_packed.is_cuda else torch.float32)
         if self.weight_packed.is_cuda:
-            from automodel_amd.ops._backend import require_ops
+            from automodel_amd.ops._backend import hip_ops
 
-            require_ops()
-            return torch.ops.amd_ops.nf4_dequant(
-                self.weight_packed, self.weight_absmax, self.block_size,
+            return hip_ops().nf4_dequant(
+                self.weight_packed, self.weight_absmax.float(), self.block_size,
                 self.out_features, self.in_features,
             ).to(dtype)
         return dequantize_nf4(
@@ -93,7 +92,8 @@ class NF4Linear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         w = self.dequantized_weight(dtype=x.dtype)
-        return F.linear(x, w, self.bias)
+        b = self.bias.to(x.dtype) if self.bias is not None else None
+        return F.linear(x, w, b)
 
     def extra_repr(self) -> str:
         return (f"in={self.in_features}, out={self.out_features}, "

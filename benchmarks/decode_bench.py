@@ -58,9 +58,9 @@ def main():
     if dev == "cuda":
         torch.cuda.synchronize()
 
-    # prefill timing
+    # prefill timing (no_grad: decode must not build an autograd graph)
     cache = KVCache.for_model(model, args.batch, args.prompt + args.new)
-    with kv_cache_context(cache):
+    with torch.no_grad(), kv_cache_context(cache):
         t0 = time.perf_counter()
         cache.begin_forward()
         logits = model(ids)

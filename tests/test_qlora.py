@@ -128,9 +128,9 @@ def test_nf4_dequant_kernel_parity():
         w = torch.randn(rows, cols)
         packed, absmax = quantize_nf4(w, 64)
         ref = dequantize_nf4(packed, absmax, (rows, cols), 64)
-        out = torch.ops.amd_ops.nf4_dequant(
-            packed.cuda(), absmax.cuda(), 64, rows, cols
-        )
+        from automodel_amd.ops._backend import hip_ops
+
+        out = hip_ops().nf4_dequant(packed.cuda(), absmax.cuda(), 64, rows, cols)
         torch.testing.assert_close(out.cpu().float(), ref, atol=1e-2, rtol=1e-2)
 
 
