@@ -56,6 +56,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.gemma.model import GemmaForCausalLM
+
+        for name in GemmaForCausalLM.hf_architectures:
+            _REGISTRY[name] = GemmaForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.vlm.model import VLMForConditionalGeneration
 
         for name in VLMForConditionalGeneration.hf_architectures:

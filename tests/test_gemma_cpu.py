@@ -1,0 +1,100 @@
+"""Gemma-2 family: logits parity against HF transformers on a tiny random
+config, plus architecture-specific behaviors (softcap, sliding window,
+(1+w) norm, embed scaling)."""
+
+import pytest
+import torch
+
+from automodel_amd.models.gemma.model import GemmaConfig, GemmaForCausalLM
+
+TINY = dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+            num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+            head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+            sliding_window=8, query_pre_attn_scalar=16.0,
+            attn_logit_softcapping=50.0, final_logit_softcapping=30.0)
+
+
+def _mine():
+    torch.manual_seed(0)
+    m = GemmaForCausalLM(TINY)
+    m.init_weights(device="cpu")
+    return m.eval()
+
+
+def test_gemma_hf_logits_parity():
+    transformers = pytest.importorskip("transformers")
+    hf_cfg = transformers.Gemma2Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+        sliding_window=8, query_pre_attn_scalar=16,
+        attn_logit_softcapping=50.0, final_logit_softcapping=30.0,
+        attn_implementation="eager",
+    )
+    torch.manual_seed(1)
+    hf = transformers.Gemma2ForCausalLM(hf_cfg).eval()
+    mine = GemmaForCausalLM(TINY).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing  # only rope buffers
+
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        out = mine(ids)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_gemma_sliding_window_masks_far_tokens():
+    m = _mine()
+    ids = torch.randint(0, 300, (1, 32))
+    with torch.no_grad():
+        base = m(ids)
+        # perturbing a token >window before the last position must NOT change
+        # the last logits when ALL layers are sliding (even layers here are;
+        # odd/global layers do see it, so compare through a single layer)
+        layer = m.model.layers[0]          # layer 0: sliding (window 8)
+        x = torch.randn(1, 32, 64)
+        cos, sin = m.model.rope_cos[:32], m.model.rope_sin[:32]
+        y0 = layer(x, cos, sin)[0, -1]
+        x2 = x.clone()
+        x2[0, 5] += 10.0                   # pos 5 is 26 back from pos 31 > 8
+        y1 = layer(x2, cos, sin)[0, -1]
+        torch.testing.assert_close(y0, y1)
+        y2 = layer(x2, cos, sin)[0, 6]     # pos 6 sees pos 5 (distance 1)
+        assert not torch.allclose(layer(x, cos, sin)[0, 6], y2)
+
+
+def test_gemma_softcap_bounds_logit_influence():
+    cfg = dict(TINY, attn_logit_softcapping=None, final_logit_softcapping=5.0)
+    torch.manual_seed(0)
+    m = GemmaForCausalLM(cfg)
+    m.init_weights(device="cpu")
+    with torch.no_grad():
+        out = m(torch.randint(0, 300, (1, 8)))
+    assert out.abs().max() <= 5.0 + 1e-5  # tanh cap
+
+
+def test_gemma_norm_zero_init_is_identity_scale():
+    from automodel_amd.models.gemma.model import GemmaRMSNorm
+
+    n = GemmaRMSNorm(16, 1e-6, "torch")
+    x = torch.randn(4, 16)
+    y = n(x)
+    rms = torch.rsqrt(x.float().pow(2).mean(-1, keepdim=True) + 1e-6)
+    torch.testing.assert_close(y, (x.float() * rms).to(x.dtype), atol=1e-5, rtol=1e-5)
+
+
+def test_gemma_generate_and_train_step():
+    from automodel_amd.loss.masked_ce import MaskedCrossEntropy
+    from automodel_amd.utils.generation import generate
+
+    m = _mine()
+    ids = torch.randint(0, 300, (1, 8))
+    out = generate(m, ids, max_new_tokens=4)
+    assert out.shape == (1, 12)
+    m.loss_fn = lambda h, w, l: MaskedCrossEntropy()(h @ w.t(), l)
+    loss = m(ids, labels=ids.clone())
+    loss.backward()
+    assert m.model.layers[0].mlp.gate_proj.weight.grad is not None
+    assert m.lm_head.weight is m.model.embed_tokens.weight  # tied
