@@ -38,13 +38,18 @@ def build_rope_cache(
             high = float(scaling.get("high_freq_factor", 4.0))
             orig = float(scaling.get("original_max_position_embeddings", 8192))
             wavelen = 2 * torch.pi / inv_freq
-            ratio = orig / wavelen
-            smooth = ((ratio - low) / (high - low)).clamp(0.0, 1.0)
+            smooth = ((orig / wavelen - low) / (high - low)).clamp(0.0, 1.0)
             scaled = inv_freq / factor
+            # three bands (HF llama3 rope): short wavelengths unchanged,
+            # long wavelengths fully scaled, middle band interpolated
             inv_freq = torch.where(
-                wavelen > orig / low,
-                torch.where(wavelen < orig / high, inv_freq, (1 - smooth) * scaled + smooth * inv_freq),
+                wavelen < orig / high,                       # high-freq: keep
                 inv_freq,
+                torch.where(
+                    wavelen > orig / low,                    # low-freq: /factor
+                    scaled,
+                    (1 - smooth) * scaled + smooth * inv_freq,
+                ),
             )
     t = torch.arange(max_seq_len, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)

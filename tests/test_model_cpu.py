@@ -139,3 +139,57 @@ def test_backend_resolve_head_dim_guard():
     assert b.attn == "sdpa"
     assert any("head_dim 64" in str(x.message) for x in w)
     assert BackendConfig.resolve(None, "cuda", head_dim=128).attn == "hip"
+
+
+def test_hf_logits_parity_llama():
+    """Bit-level check of the flagship architecture against HF transformers
+    (eager): GQA attention, rope, rms norms, swiglu, lm_head."""
+    transformers = pytest.importorskip("transformers")
+    hf_cfg = transformers.LlamaConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0,
+        attn_implementation="eager", tie_word_embeddings=False,
+    )
+    torch.manual_seed(3)
+    hf = transformers.LlamaForCausalLM(hf_cfg).eval()
+    mine = LlamaForCausalLM(dict(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+    )).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        ref = hf(ids).logits
+        out = mine(ids)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_llama3_rope_scaling():
+    """Llama-3.1-style rope scaling parity (the bench model's rope)."""
+    transformers = pytest.importorskip("transformers")
+    scaling = {"rope_type": "llama3", "factor": 8.0,
+               "low_freq_factor": 1.0, "high_freq_factor": 4.0,
+               "original_max_position_embeddings": 64}
+    hf_cfg = transformers.LlamaConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=256, rope_theta=10000.0,
+        rope_scaling=dict(scaling), attn_implementation="eager",
+        tie_word_embeddings=False,
+    )
+    torch.manual_seed(4)
+    hf = transformers.LlamaForCausalLM(hf_cfg).eval()
+    mine = LlamaForCausalLM(dict(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=256, rope_theta=10000.0, rms_norm_eps=1e-6,
+        rope_scaling=dict(scaling),
+    )).eval()
+    mine.load_state_dict(hf.state_dict(), strict=False)
+    ids = torch.randint(0, 300, (1, 100))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
