@@ -123,14 +123,22 @@ class LlamaAttention(nn.Module):
         else:
             from automodel_amd.utils.kv_cache import active_kv_cache, maybe_update_kv
 
-            cached = active_kv_cache() is not None
-            k, v, cache_pos = maybe_update_kv(k, v)
-            # cached decode: q is short (often 1) and prompts are arbitrary
-            # lengths — sdpa is the right tool (GEMV-bound; the flash
-            # kernel's 128-row tiling wants training shapes)
-            attn_backend = "sdpa" if cached else self.backend.attn
-            o = flash_attention(q, k, v, causal=True, backend=attn_backend,
-                                q_start=cache_pos)
+            cache = active_kv_cache()
+            if cache is not None and getattr(cache, "is_static", False):
+                # static-shape decode (hipGraph capture): full-buffer KV +
+                # additive mask from the device position counter
+                from automodel_amd.ops.attention import sdpa_masked
+
+                k, v, _ = maybe_update_kv(k, v)
+                o = sdpa_masked(q, k, v, cache.attn_mask())
+            else:
+                k, v, cache_pos = maybe_update_kv(k, v)
+                # cached decode: q is short (often 1) and prompts are
+                # arbitrary lengths — sdpa is the right tool (GEMV-bound;
+                # the flash kernel's 128-row tiling wants training shapes)
+                attn_backend = "sdpa" if cache is not None else self.backend.attn
+                o = flash_attention(q, k, v, causal=True, backend=attn_backend,
+                                    q_start=cache_pos)
         return self.o_proj(o.reshape(B, S, -1))
 
 

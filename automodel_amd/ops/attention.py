@@ -46,6 +46,17 @@ def attention_ref(q, k, v, causal: bool = True, scale: float | None = None,
     return o.permute(0, 2, 1, 3).to(q.dtype)
 
 
+def sdpa_masked(q, k, v, attn_mask, scale: float | None = None):
+    """sdpa with an explicit additive mask — used by the static-cache
+    (hipGraph) decode path where causal structure lives in the mask."""
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))
+    o = torch.nn.functional.scaled_dot_product_attention(
+        qt, kt, vt, attn_mask=attn_mask, scale=scale,
+        enable_gqa=q.shape[2] != k.shape[2])
+    return o.transpose(1, 2)
+
+
 def _sdpa(q, k, v, causal: bool, scale: float | None, q_start: int = 0):
     qt, kt, vt = (t.transpose(1, 2) for t in (q, k, v))   # B,H,S,D
     kwargs = dict(scale=scale, enable_gqa=q.shape[2] != k.shape[2])

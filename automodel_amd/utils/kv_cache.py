@@ -57,6 +57,62 @@ class KVCache:
         self.pos += n
 
 
+class StaticKVCache(KVCache):
+    """Static-shape variant for hipGraph decode: attention always runs over
+    the full preallocated [max_len] buffers with an additive mask computed
+    from a DEVICE position counter — every tensor shape in the decode step
+    is constant, so one torch.cuda.graph (hipGraph on ROCm) captures the
+    whole per-token forward and replays it with a single hipGraphLaunch
+    (guide: capture launch-bound inner loops in hipGraphs).
+
+    Prefill runs with host indexing (not captured); ``freeze_for_graph()``
+    moves the position to a device tensor for the captured decode loop —
+    the in-graph ``advance_device()`` increments it with no host sync.
+    """
+
+    def __init__(self, *args, **kwargs):
+        super().__init__(*args, **kwargs)
+        self.pos_dev: torch.Tensor | None = None
+
+    @property
+    def is_static(self) -> bool:
+        return self.pos_dev is not None
+
+    def freeze_for_graph(self) -> None:
+        self.pos_dev = torch.tensor(self.pos, device=self.k.device,
+                                    dtype=torch.long)
+
+    def update(self, k_new: torch.Tensor, v_new: torch.Tensor):
+        i = self._layer
+        self._layer += 1
+        if self.pos_dev is None:
+            S = k_new.shape[1]
+            assert self.pos + S <= self.max_len, "KV cache overflow"
+            self.k[i][:, self.pos : self.pos + S] = k_new
+            self.v[i][:, self.pos : self.pos + S] = v_new
+        else:  # graph mode: single-token write at a device index
+            assert k_new.shape[1] == 1, "static decode writes one token"
+            idx = self.pos_dev.reshape(1)
+            self.k[i].index_copy_(1, idx, k_new)
+            self.v[i].index_copy_(1, idx, v_new)
+        return self.k[i], self.v[i]          # full static buffers
+
+    def position_ids(self) -> torch.Tensor:
+        """[1, 1] device position for rope indexing inside the graph."""
+        return self.pos_dev.reshape(1, 1)
+
+    def attn_mask(self) -> torch.Tensor:
+        """[1, 1, 1, max_len] additive mask: positions <= pos are visible
+        (the current token was just written at index pos)."""
+        ar = torch.arange(self.max_len, device=self.k.device)
+        visible = ar <= self.pos_dev
+        return torch.where(visible, 0.0, float("-inf")).reshape(1, 1, 1, -1) \
+            .to(self.k.dtype)
+
+    def advance_device(self) -> None:
+        self.pos_dev.add_(1)
+
+
 @contextmanager
 def kv_cache_context(cache: KVCache):
     global _ACTIVE

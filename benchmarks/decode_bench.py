@@ -37,10 +37,12 @@ def main():
     ap.add_argument("--batch", type=int, default=8)
     ap.add_argument("--prompt", type=int, default=512)
     ap.add_argument("--new", type=int, default=64)
+    ap.add_argument("--graphed", action="store_true",
+                    help="decode via generate_graphed (hipGraph-captured step)")
     args = ap.parse_args()
 
     from automodel_amd.models.llama.model import LlamaForCausalLM
-    from automodel_amd.utils.generation import generate_cached
+    from automodel_amd.utils.generation import generate_cached, generate_graphed
     from automodel_amd.utils.kv_cache import KVCache, kv_cache_context
 
     dev = "cuda" if torch.cuda.is_available() else "cpu"
@@ -57,6 +59,28 @@ def main():
     generate_cached(model, ids[:, :64], max_new_tokens=4)
     if dev == "cuda":
         torch.cuda.synchronize()
+
+    if args.graphed:
+        t0 = time.perf_counter()
+        out = generate_graphed(model, ids, max_new_tokens=args.new)
+        if dev == "cuda":
+            torch.cuda.synchronize()
+        total_s = time.perf_counter() - t0
+        # time a second run (graph capture amortized) for steady-state decode
+        t0 = time.perf_counter()
+        out = generate_graphed(model, ids, max_new_tokens=args.new)
+        if dev == "cuda":
+            torch.cuda.synchronize()
+        total2_s = time.perf_counter() - t0
+        print(json.dumps({
+            "model": args.model, "batch": args.batch, "mode": "graphed",
+            "prompt_len": args.prompt, "new_tokens": args.new,
+            "first_run_s": round(total_s, 3),
+            "steady_ms_per_token": round(total2_s / args.new * 1e3, 2),
+            "steady_decode_tokens_per_s": round(args.batch * args.new / total2_s, 1),
+            "dtype": str(dtype).split(".")[-1], "device": dev,
+        }))
+        return
 
     # prefill timing (no_grad: decode must not build an autograd graph)
     cache = KVCache.for_model(model, args.batch, args.prompt + args.new)
