@@ -61,6 +61,9 @@ TORCH_LIBRARY(amd_ops, m) {
   m.def("nf4_dequant(Tensor packed, Tensor absmax, int block_size, int rows, int cols) -> Tensor");
   m.impl("nf4_dequant", &amd_ops::nf4_dequant);
 
+  m.def("gemv_bf16(Tensor x, Tensor w, Tensor? bias) -> Tensor");
+  m.impl("gemv_bf16", &amd_ops::gemv_bf16);
+
   m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
   m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");

@@ -60,6 +60,9 @@ at::Tensor soft_ce_bwd(const at::Tensor& s, const at::Tensor& t,
 at::Tensor nf4_dequant(const at::Tensor& packed, const at::Tensor& absmax,
                        int64_t block_size, int64_t rows, int64_t cols);
 
+at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
+                     const c10::optional<at::Tensor>& bias);
+
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);

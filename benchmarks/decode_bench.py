@@ -39,6 +39,8 @@ def main():
     ap.add_argument("--new", type=int, default=64)
     ap.add_argument("--graphed", action="store_true",
                     help="decode via generate_graphed (hipGraph-captured step)")
+    ap.add_argument("--gemv", action="store_true",
+                    help="swap linears for the in-tree decode GEMV kernel")
     args = ap.parse_args()
 
     from automodel_amd.models.llama.model import LlamaForCausalLM
@@ -60,6 +62,12 @@ def main():
     if dev == "cuda":
         torch.cuda.synchronize()
 
+    if args.gemv:
+        from automodel_amd.serving import swap_linears_for_decode
+
+        n = swap_linears_for_decode(model)
+        print(f"# swapped {n} linears for DecodeLinear", flush=True)
+
     if args.graphed:
         t0 = time.perf_counter()
         out = generate_graphed(model, ids, max_new_tokens=args.new)
@@ -73,7 +81,8 @@ def main():
             torch.cuda.synchronize()
         total2_s = time.perf_counter() - t0
         print(json.dumps({
-            "model": args.model, "batch": args.batch, "mode": "graphed",
+            "model": args.model, "batch": args.batch,
+            "mode": "graphed+gemv" if args.gemv else "graphed",
             "prompt_len": args.prompt, "new_tokens": args.new,
             "first_run_s": round(total_s, 3),
             "steady_ms_per_token": round(total2_s / args.new * 1e3, 2),

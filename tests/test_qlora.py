@@ -144,3 +144,34 @@ def test_nf4_linear_gpu_forward():
     w = dequantize_nf4(q.weight_packed.cpu(), q.weight_absmax.cpu(), (128, 256), 64)
     ref = torch.nn.functional.linear(x.cpu().float(), w, base.bias)
     torch.testing.assert_close(y.cpu().float(), ref, atol=5e-2, rtol=5e-2)
+
+
+def test_decode_linear_cpu_fallthrough():
+    """On CPU / under grad, DecodeLinear must behave exactly like Linear."""
+    from automodel_amd.serving import DecodeLinear, swap_linears_for_decode
+
+    torch.manual_seed(7)
+    m = nn.Sequential(nn.Linear(512, 64), nn.ReLU(), nn.Linear(64, 512))
+    ref = nn.Sequential(nn.Linear(512, 64), nn.ReLU(), nn.Linear(64, 512))
+    ref.load_state_dict(m.state_dict())
+    n = swap_linears_for_decode(m)
+    assert n == 2 and isinstance(m[0], DecodeLinear)
+    x = torch.randn(3, 512)
+    torch.testing.assert_close(m(x), ref(x))
+    m(x).sum().backward()  # grad path works
+
+
+@pytest.mark.gpu
+def test_gemv_kernel_parity():
+    from automodel_amd.serving import gemv_bf16
+
+    torch.manual_seed(8)
+    for B, K, N in [(1, 4096, 1024), (8, 512, 300), (5, 1024, 256), (16, 512, 128)]:
+        x = torch.randn(B, K, device="cuda", dtype=torch.bfloat16)
+        w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
+        bias = torch.randn(N, device="cuda", dtype=torch.bfloat16)
+        y = gemv_bf16(x, w, bias)
+        ref = torch.nn.functional.linear(x.float(), w.float(), bias.float())
+        torch.testing.assert_close(y.float(), ref, atol=5e-2, rtol=5e-2)
+        y2 = gemv_bf16(x, w, None)
+        torch.testing.assert_close(y2.float(), ref - bias.float(), atol=5e-2, rtol=5e-2)
