@@ -21,34 +21,47 @@
 
 namespace amd_ops {
 
+// Two rows per wave: doubles outstanding b128 loads per wave (latency
+// hiding for the short 8-iteration K sweep at K=4096).
 template <int B>
 __global__ __launch_bounds__(256, 4) void gemv_bf16_kernel(
     const bf16* __restrict__ W, const bf16* __restrict__ x,
     const bf16* __restrict__ bias, bf16* __restrict__ y, int N, int K) {
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const int row = blockIdx.x * (blockDim.x / WAVE_SIZE) + (threadIdx.x / WAVE_SIZE);
-  if (row >= N) return;
-  float acc[B];
+  const int wid = blockIdx.x * (blockDim.x / WAVE_SIZE) + (threadIdx.x / WAVE_SIZE);
+  const int row0 = 2 * wid;
+  if (row0 >= N) return;
+  const bool two = row0 + 1 < N;
+  float acc0[B], acc1[B];
 #pragma unroll
-  for (int b = 0; b < B; ++b) acc[b] = 0.f;
-  const bf16* wrow = W + (long)row * K;
+  for (int b = 0; b < B; ++b) acc0[b] = acc1[b] = 0.f;
+  const bf16* wrow0 = W + (long)row0 * K;
+  const bf16* wrow1 = wrow0 + (two ? K : 0);
   for (int k = lane * 8; k < K; k += WAVE_SIZE * 8) {
-    const bf16x8 w = *reinterpret_cast<const bf16x8*>(wrow + k);
+    const bf16x8 w0 = *reinterpret_cast<const bf16x8*>(wrow0 + k);
+    const bf16x8 w1 = *reinterpret_cast<const bf16x8*>(wrow1 + k);
 #pragma unroll
     for (int b = 0; b < B; ++b) {
       const bf16x8 xv = *reinterpret_cast<const bf16x8*>(x + (long)b * K + k);
-      float d = 0.f;
+      float d0 = 0.f, d1 = 0.f;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) d += bf2f(w.v[j]) * bf2f(xv.v[j]);
-      acc[b] += d;
+      for (int j = 0; j < 8; ++j) {
+        const float xf = bf2f(xv.v[j]);
+        d0 += bf2f(w0.v[j]) * xf;
+        d1 += bf2f(w1.v[j]) * xf;
+      }
+      acc0[b] += d0;
+      acc1[b] += d1;
     }
   }
 #pragma unroll
   for (int b = 0; b < B; ++b) {
-    const float total = wave_reduce_sum(acc[b]);
+    const float t0 = wave_reduce_sum(acc0[b]);
+    const float t1 = wave_reduce_sum(acc1[b]);
     if (lane == 0) {
-      const float bb = bias ? bf2f(bias[row]) : 0.f;
-      y[(long)b * N + row] = f2bf(total + bb);
+      y[(long)b * N + row0] = f2bf(t0 + (bias ? bf2f(bias[row0]) : 0.f));
+      if (two)
+        y[(long)b * N + row0 + 1] = f2bf(t1 + (bias ? bf2f(bias[row0 + 1]) : 0.f));
     }
   }
 }
@@ -72,7 +85,8 @@ at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
     bptr = reinterpret_cast<const bf16*>(bias->contiguous().data_ptr());
   }
   const int waves_per_block = 4;
-  const dim3 grid((N + waves_per_block - 1) / waves_per_block);
+  const int rows_per_block = 2 * waves_per_block;
+  const dim3 grid((N + rows_per_block - 1) / rows_per_block);
   auto stream = c10::hip::getCurrentHIPStream();
 #define LAUNCH(BB)                                                             \
   gemv_bf16_kernel<BB><<<grid, dim3(waves_per_block * WAVE_SIZE), 0, stream>>>( \

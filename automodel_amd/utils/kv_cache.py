@@ -73,6 +73,7 @@ class StaticKVCache(KVCache):
     def __init__(self, *args, **kwargs):
         super().__init__(*args, **kwargs)
         self.pos_dev: torch.Tensor | None = None
+        self._mask: torch.Tensor | None = None
 
     @property
     def is_static(self) -> bool:
@@ -81,6 +82,12 @@ class StaticKVCache(KVCache):
     def freeze_for_graph(self) -> None:
         self.pos_dev = torch.tensor(self.pos, device=self.k.device,
                                     dtype=torch.long)
+        # persistent additive mask, updated ONCE per step in-graph
+        # (building it per layer costs 5 elementwise kernels x n_layers)
+        ar = torch.arange(self.max_len, device=self.k.device)
+        self._mask = torch.where(ar <= self.pos, 0.0, float("-inf")) \
+            .reshape(1, 1, 1, -1).to(self.k.dtype).contiguous()
+        self._zero = torch.zeros(1, dtype=self.k.dtype, device=self.k.device)
 
     def update(self, k_new: torch.Tensor, v_new: torch.Tensor):
         i = self._layer
@@ -102,14 +109,14 @@ class StaticKVCache(KVCache):
         return self.pos_dev.reshape(1, 1)
 
     def attn_mask(self) -> torch.Tensor:
-        """[1, 1, 1, max_len] additive mask: positions <= pos are visible
-        (the current token was just written at index pos)."""
-        ar = torch.arange(self.max_len, device=self.k.device)
-        visible = ar <= self.pos_dev
-        return torch.where(visible, 0.0, float("-inf")).reshape(1, 1, 1, -1) \
-            .to(self.k.dtype)
+        """[1, 1, 1, max_len] additive mask: positions <= pos visible."""
+        return self._mask
 
     def advance_device(self) -> None:
+        """In-graph per-step bookkeeping: reveal the next slot, bump pos —
+        two tiny kernels per STEP (not per layer)."""
+        nxt = torch.clamp(self.pos_dev + 1, max=self.max_len - 1)
+        self._mask.view(-1).index_copy_(0, nxt, self._zero)
         self.pos_dev.add_(1)
 
 
