@@ -36,7 +36,9 @@ def _gemv_eligible(x: torch.Tensor, w: torch.Tensor) -> bool:
     if x.dtype != torch.bfloat16 or w.dtype != torch.bfloat16:
         return False
     rows = x.numel() // x.shape[-1]
-    return rows <= 16 and x.shape[-1] % 512 == 0
+    # measured on MI355X (profiles/README.md): the wave-per-row GEMV beats
+    # hipBLASLt up to ~4 rows; above that the library's m-tiled path wins
+    return rows <= 4 and x.shape[-1] % 512 == 0
 
 
 class DecodeLinear(nn.Linear):
