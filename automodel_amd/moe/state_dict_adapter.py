@@ -42,6 +42,12 @@ class MoEStateDictAdapter:
                 layer, e, proj = m.group(1), int(m.group(2)), m.group(3)
                 proj = _MIXTRAL_MAP.get(proj, proj)
                 pending.setdefault(f"{layer}.mlp.experts.{proj}", {})[e] = t
+            elif key.endswith(".mlp.experts.gate_up_proj"):
+                # transformers >= 4.56 stacked-expert layout: [E, 2I, H]
+                # fused as [gate; up] along dim 1 (chunk(2, dim=1))
+                gate, up = t.chunk(2, dim=1)
+                out[key.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
+                out[key.replace("gate_up_proj", "up_proj")] = up.contiguous()
             elif ".block_sparse_moe.gate." in key:
                 out[key.replace(".block_sparse_moe.gate.", ".mlp.gate.")] = t
             elif ".block_sparse_moe." in key:
