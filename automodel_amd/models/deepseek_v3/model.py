@@ -138,8 +138,14 @@ class MLAAttention(nn.Module):
             B, S, H, cfg.qk_nope_head_dim + cfg.v_head_dim)
         k_nope, v = kv.split([cfg.qk_nope_head_dim, cfg.v_head_dim], dim=-1)
 
-        # decoupled rope: q_rope per head, k_rope single shared head
+        # decoupled rope: q_rope per head, k_rope single shared head.
+        # DeepSeek checkpoints store the rope dims INTERLEAVED ([x0,y0,x1,y1..]);
+        # de-interleave to half-split order before the rotate-half kernel
+        # (HF apply_rotary_pos_emb_interleave equivalent).
         k_rope = k_rope.view(B, S, 1, cfg.qk_rope_head_dim)
+        d2 = cfg.qk_rope_head_dim // 2
+        q_rope = q_rope.view(B, S, H, d2, 2).transpose(-1, -2).reshape(B, S, H, -1)
+        k_rope = k_rope.view(B, S, 1, d2, 2).transpose(-1, -2).reshape(B, S, 1, -1)
         q_rope, k_rope = apply_rope(q_rope.contiguous(), k_rope.contiguous(),
                                     cos, sin, backend="torch")
         k_rope = k_rope.expand(B, S, H, cfg.qk_rope_head_dim)
@@ -270,7 +276,7 @@ class DeepseekV3ForCausalLM(nn.Module):
             elif isinstance(m, Gate):
                 nn.init.normal_(m.weight, std=std)
                 if m.cfg.expert_bias:
-                    m.e_score_bias.zero_()
+                    m.e_score_correction_bias.zero_()
         for m in self.modules():
             if type(m).__name__ == "GroupedExperts":
                 m.init_weights(std)

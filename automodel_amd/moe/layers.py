@@ -23,7 +23,7 @@ class Gate(nn.Module):
         self.weight = nn.Parameter(torch.empty(cfg.n_routed_experts, hidden_size))
         if cfg.expert_bias:
             # routing-only bias, updated out-of-band (aux-free balancing)
-            self.register_buffer("e_score_bias", torch.zeros(cfg.n_routed_experts))
+            self.register_buffer("e_score_correction_bias", torch.zeros(cfg.n_routed_experts))
         self.last_aux_loss: torch.Tensor | None = None
 
     def forward(self, x: torch.Tensor) -> tuple[torch.Tensor, torch.Tensor]:
@@ -37,7 +37,7 @@ class Gate(nn.Module):
 
         select_scores = scores
         if cfg.expert_bias:
-            select_scores = scores + self.e_score_bias
+            select_scores = scores + self.e_score_correction_bias
 
         if cfg.n_expert_groups > 1:
             T, E = scores.shape
@@ -78,7 +78,7 @@ class Gate(nn.Module):
             return
         mean_load = expert_load.float().mean()
         err = expert_load.float() - mean_load
-        self.e_score_bias -= self.cfg.bias_update_speed * err.sign()
+        self.e_score_correction_bias -= self.cfg.bias_update_speed * err.sign()
 
 
 class FakeBalancedGate(nn.Module):

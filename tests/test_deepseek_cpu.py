@@ -46,9 +46,9 @@ def test_mla_train_step_and_bias_update():
     loss.backward()
     assert m.model.layers[0].self_attn.kv_b_proj.weight.grad is not None
     assert m.model.layers[1].mlp.experts.gate_proj.grad is not None
-    bias_before = m.model.layers[1].mlp.gate.e_score_bias.clone()
+    bias_before = m.model.layers[1].mlp.gate.e_score_correction_bias.clone()
     m.update_moe_gate_bias()
-    assert not torch.equal(bias_before, m.model.layers[1].mlp.gate.e_score_bias)
+    assert not torch.equal(bias_before, m.model.layers[1].mlp.gate.e_score_correction_bias)
 
 
 def test_deepseek_hf_config_mapping():

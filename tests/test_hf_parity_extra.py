@@ -111,3 +111,35 @@ def test_hf_logits_parity_mixtral():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_deepseek_v3():
+    """MLA (q/kv LoRA projections, decoupled interleaved rope), dense-first
+    layers, sigmoid gate with e_score_correction_bias, shared expert."""
+    from automodel_amd.models.deepseek_v3.model import DeepseekV3ForCausalLM
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(9)
+    kw = dict(vocab_size=300, hidden_size=64, intermediate_size=96,
+              num_hidden_layers=3, num_attention_heads=4,
+              first_k_dense_replace=1, q_lora_rank=32, kv_lora_rank=16,
+              qk_nope_head_dim=16, qk_rope_head_dim=8, v_head_dim=16,
+              max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6)
+    hf_cfg = transformers.DeepseekV3Config(
+        moe_intermediate_size=48, num_key_value_heads=4, n_routed_experts=8,
+        n_shared_experts=1, num_experts_per_tok=2, n_group=1, topk_group=1,
+        norm_topk_prob=True, routed_scaling_factor=1.0,
+        attn_implementation="eager", tie_word_embeddings=False, **kw)
+    hf = transformers.DeepseekV3ForCausalLM(hf_cfg).eval()
+    mine = DeepseekV3ForCausalLM(dict(
+        moe=dict(n_routed_experts=8, n_shared_experts=1, n_activated_experts=2,
+                 moe_intermediate_size=48, shared_expert_intermediate_size=48,
+                 norm_topk_prob=True, score_func="sigmoid", route_scale=1.0,
+                 expert_bias=True), **kw)).eval()
+    sd = MoEStateDictAdapter(mine.config).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)

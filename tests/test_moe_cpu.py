@@ -36,7 +36,7 @@ def test_gate_expert_bias_update():
     torch.nn.init.zeros_(g.weight)
     load = torch.tensor([10.0, 0.0, 0.0, 0.0])
     g.update_bias(load)
-    assert g.e_score_bias[0] < 0 and g.e_score_bias[1] > 0  # push against overload
+    assert g.e_score_correction_bias[0] < 0 and g.e_score_correction_bias[1] > 0  # push against overload
 
 
 def test_fake_balanced_gate_uniform():
