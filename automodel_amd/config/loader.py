@@ -12,7 +12,6 @@ from __future__ import annotations
 import importlib
 import os
 import re
-import types
 from typing import Any, Iterator, Mapping
 
 import yaml

@@ -9,7 +9,6 @@ indexed_dataset.py (.bin/.idx mmap pair). Own on-disk format:
 
 from __future__ import annotations
 
-import os
 import struct
 
 import numpy as np

@@ -8,7 +8,7 @@ to the JSONL MetricLogger which is always on.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any
 
 from automodel_amd.loggers.metric_logger import get_rank

@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import torch
 
-from automodel_amd.ops._backend import hip_ops, ops_available
+from automodel_amd.ops._backend import hip_ops
 
 IGNORE_INDEX = -100
 

@@ -8,7 +8,7 @@ plain torch eager ("torch"), or torch SDPA for attention ("sdpa").
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field, replace
+from dataclasses import dataclass, replace
 
 
 @dataclass

@@ -13,7 +13,7 @@ Also covers Qwen2-style variants (attention bias) via LlamaConfig flags.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any
 
 import torch

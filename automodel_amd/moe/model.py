@@ -16,7 +16,7 @@ import torch.nn as nn
 from automodel_amd.models.common.backend import BackendConfig
 from automodel_amd.models.llama.model import LlamaAttention, LlamaConfig
 from automodel_amd.moe.config import MoEConfig
-from automodel_amd.moe.layers import MoE, FakeBalancedGate, Gate
+from automodel_amd.moe.layers import MoE, Gate
 from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
 from automodel_amd.ops.rms_norm import RMSNorm
 from automodel_amd.ops.rope import build_rope_cache

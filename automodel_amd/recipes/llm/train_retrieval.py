@@ -14,7 +14,6 @@ import torch.nn as nn
 
 from automodel_amd.config.loader import ConfigNode, apply_overrides, load_yaml_config, parse_cli_overrides
 from automodel_amd.loss.infonce import info_nce_loss
-from automodel_amd.models.common.backend import BackendConfig
 from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
 from automodel_amd.recipes.llm.train_ft import TrainFinetuneRecipeForNextTokenPrediction
 

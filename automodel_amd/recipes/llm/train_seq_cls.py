@@ -10,7 +10,7 @@ import sys
 import torch
 import torch.nn as nn
 
-from automodel_amd.config.loader import ConfigNode, apply_overrides, load_yaml_config, parse_cli_overrides
+from automodel_amd.config.loader import apply_overrides, load_yaml_config, parse_cli_overrides
 from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
 from automodel_amd.recipes.llm.train_ft import TrainFinetuneRecipeForNextTokenPrediction
 

@@ -9,7 +9,6 @@ components stay YAML-agnostic (reference invariant).
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Any
 
 from automodel_amd.config.loader import ConfigNode
 

@@ -14,7 +14,6 @@ import torch
 
 from automodel_amd.config.loader import ConfigNode, apply_overrides, load_yaml_config, parse_cli_overrides
 from automodel_amd.datasets.vlm.mock import MockVLMDataset, vlm_collate
-from automodel_amd.models.vlm.model import VLMConfig, VLMForConditionalGeneration
 from automodel_amd.recipes.llm.train_ft import TrainFinetuneRecipeForNextTokenPrediction
 
 
