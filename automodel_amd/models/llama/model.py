@@ -60,6 +60,10 @@ class LlamaConfig:
         archs = " ".join(get("architectures", []) or [])
         return cls(
             qk_norm="Qwen3" in archs,
+            # Phi-3 ships fused qkv_proj / gate_up_proj weights — exactly this
+            # family's fused layout (keys and math match; parity-tested)
+            fused_qkv="Phi3" in archs,
+            fused_gate_up="Phi3" in archs,
             vocab_size=get("vocab_size", 32000),
             hidden_size=get("hidden_size", 4096),
             intermediate_size=get("intermediate_size", 11008),
@@ -207,7 +211,8 @@ class LlamaModel(nn.Module):
 
 
 class LlamaForCausalLM(nn.Module):
-    hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM", "Qwen3ForCausalLM")
+    hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
+                        "Qwen3ForCausalLM", "Phi3ForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

@@ -143,3 +143,27 @@ def test_hf_logits_parity_deepseek_v3():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_phi3():
+    """Phi-3 rides the llama family's fused qkv/gate_up layout unchanged."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    hf_cfg = transformers.Phi3Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attn_implementation="eager", tie_word_embeddings=False,
+        pad_token_id=0, eos_token_id=1, bos_token_id=2,
+    )
+    torch.manual_seed(11)
+    hf = transformers.Phi3ForCausalLM(hf_cfg).eval()
+    mine = LlamaForCausalLM(
+        LlamaForCausalLM.config_from_hf(hf_cfg.to_dict() | {"architectures": ["Phi3ForCausalLM"]})
+    ).eval()
+    assert mine.config.fused_qkv and mine.config.fused_gate_up
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
