@@ -63,6 +63,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.gpt_oss.model import GptOssForCausalLM
+
+        for name in GptOssForCausalLM.hf_architectures:
+            _REGISTRY[name] = GptOssForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.vlm.model import VLMForConditionalGeneration
 
         for name in VLMForConditionalGeneration.hf_architectures:

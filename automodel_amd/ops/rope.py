@@ -32,6 +32,36 @@ def build_rope_cache(
         rope_type = scaling.get("rope_type", scaling.get("type", "default"))
         if rope_type == "linear":
             inv_freq = inv_freq / float(scaling["factor"])
+        elif rope_type == "yarn":
+            # YaRN (public formulas; HF _compute_yarn_parameters): NTK-by-parts
+            # interpolation with a wavelength ramp + attention temperature.
+            import math as _m
+
+            factor = float(scaling.get("factor", 1.0))
+            beta_fast = float(scaling.get("beta_fast", 32.0))
+            beta_slow = float(scaling.get("beta_slow", 1.0))
+            orig = float(scaling.get("original_max_position_embeddings", 4096))
+            attn_factor = scaling.get("attention_factor")
+            mscale_all = scaling.get("mscale")  # deepseek-style override
+            dim = head_dim
+
+            def find_dim(num_rot):
+                return (dim * _m.log(orig / (num_rot * 2 * _m.pi))) / (2 * _m.log(base))
+
+            low, high = find_dim(beta_fast), find_dim(beta_slow)
+            if scaling.get("truncate", True):
+                low, high = _m.floor(low), _m.ceil(high)
+            low, high = max(low, 0), min(high, dim - 1)
+            ramp = ((torch.arange(dim // 2, dtype=torch.float32) - low)
+                    / max(high - low, 1e-3)).clamp(0.0, 1.0)
+            extrap_mask = 1.0 - ramp
+            inv_freq = inv_freq * (1 - extrap_mask) / factor + inv_freq * extrap_mask
+            if attn_factor is not None:
+                attention_scaling = float(attn_factor)
+            elif mscale_all is not None:
+                attention_scaling = 0.1 * float(mscale_all) * _m.log(factor) + 1.0
+            else:
+                attention_scaling = 0.1 * _m.log(factor) + 1.0
         elif rope_type == "llama3":
             factor = float(scaling["factor"])
             low = float(scaling.get("low_freq_factor", 1.0))
@@ -54,7 +84,8 @@ def build_rope_cache(
     t = torch.arange(max_seq_len, dtype=torch.float32)
     freqs = torch.outer(t, inv_freq)
     emb = torch.cat([freqs, freqs], dim=-1)
-    cos, sin = emb.cos(), emb.sin()
+    scale = locals().get("attention_scaling", 1.0)
+    cos, sin = emb.cos() * scale, emb.sin() * scale
     if device is not None:
         cos, sin = cos.to(device), sin.to(device)
     return cos, sin
