@@ -27,12 +27,17 @@ class Checkpointer:
         model_save_format: str = "safetensors",
         save_consolidated: bool = False,
         keep_last_n: int | None = None,
+        keep_top_k: int | None = None,
+        metric_higher_is_better: bool = False,
         async_save: bool = False,
     ):
         self.checkpoint_dir = checkpoint_dir
         self.model_save_format = model_save_format
         self.save_consolidated = save_consolidated
         self.keep_last_n = keep_last_n
+        self.keep_top_k = keep_top_k             # by recorded metric (lifecycle.py)
+        self.metric_higher_is_better = metric_higher_is_better
+        self._step_metrics: dict = {}
         self.async_save = async_save
         self._async_writer = None
         if async_save:
@@ -162,12 +167,31 @@ class Checkpointer:
             _json.dump(d, fh, indent=1)
 
     # ------------------------------------------------------------- retention
+    def record_metric(self, step: int, value: float) -> None:
+        """Associate a validation metric with a saved step for keep_top_k
+        retention (reference checkpoint/lifecycle.py keep-top-k-by-metric)."""
+        self._step_metrics[step] = float(value)
+
     def _apply_retention(self) -> None:
-        if not self.keep_last_n or not os.path.isdir(self.checkpoint_dir):
+        if not os.path.isdir(self.checkpoint_dir):
             return
         steps = sorted(
             (d for d in os.listdir(self.checkpoint_dir) if d.startswith("step_")),
             key=lambda d: int(d.split("_")[1]),
         )
-        for d in steps[: -self.keep_last_n]:
+        drop: set = set()
+        if self.keep_last_n:
+            drop.update(steps[: -self.keep_last_n])
+        if self.keep_top_k:
+            scored = [d for d in steps if int(d.split("_")[1]) in self._step_metrics]
+            ranked = sorted(
+                scored,
+                key=lambda d: self._step_metrics[int(d.split("_")[1])],
+                reverse=self.metric_higher_is_better,
+            )
+            keep = set(ranked[: self.keep_top_k]) | ({steps[-1]} if steps else set())
+            drop.update(d for d in scored if d not in keep)
+            if self.keep_last_n:          # keep_last_n still protects the tail
+                drop.difference_update(steps[-self.keep_last_n:])
+        for d in drop:
             shutil.rmtree(os.path.join(self.checkpoint_dir, d), ignore_errors=True)

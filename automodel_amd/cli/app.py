@@ -29,10 +29,27 @@ RECIPE_ALIASES = {
 }
 
 
+def query_capabilities(arch: str | None = None) -> None:
+    """Print per-model capability flags (reference cli/query_capabilities.py)."""
+    from automodel_amd.models.common.capabilities import _CAPS, ModelCapabilities
+    from automodel_amd.models.registry import _REGISTRY, _ensure_builtin
+
+    _ensure_builtin()
+    rows = [arch] if arch else sorted(set(list(_REGISTRY) + list(_CAPS)))
+    for name in rows:
+        cls = _REGISTRY.get(name)
+        caps = _CAPS.get(cls.__name__ if cls else name, ModelCapabilities())
+        flags = ", ".join(f"{k}={v}" for k, v in vars(caps).items())
+        print(f"{name}: {flags}")
+
+
 def main(argv: list[str] | None = None) -> None:
     argv = argv if argv is not None else sys.argv[1:]
     if not argv or argv[0] in ("-h", "--help"):
         print(__doc__)
+        return
+    if argv[0] == "capabilities":
+        query_capabilities(argv[1] if len(argv) > 1 else None)
         return
     cfg_path = argv[0]
     rest = argv[1:]

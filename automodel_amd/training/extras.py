@@ -79,3 +79,21 @@ def memory_stats() -> dict:
         "max_allocated_gb": torch.cuda.max_memory_allocated() / 2**30,
         "reserved_gb": torch.cuda.memory_reserved() / 2**30,
     }
+
+
+@torch.no_grad()
+def repair_embedding_rows(embedding: nn.Embedding, reference_rows=None,
+                          std: float = 0.02) -> int:
+    """Re-init embedding rows containing NaN/Inf (reference
+    training/embedding_row_repair.py — broken rows appear in some released
+    checkpoints). Returns the number of rows repaired."""
+    w = embedding.weight
+    bad = ~torch.isfinite(w).all(dim=-1)
+    n = int(bad.sum())
+    if n:
+        if reference_rows is not None:
+            w[bad] = reference_rows.to(w.dtype)
+        else:
+            w[bad] = torch.randn(n, w.shape[1], device=w.device,
+                                 dtype=w.dtype) * std
+    return n
