@@ -53,6 +53,26 @@ class GroupedExperts(nn.Module):
         for p in (self.gate_proj, self.up_proj, self.down_proj):
             nn.init.normal_(p, std=std)
 
+    @staticmethod
+    def project(x_perm: torch.Tensor, w: torch.Tensor, counts) -> torch.Tensor:
+        """Grouped projection y[t] = x[t] @ w[expert(t)].T over expert-sorted
+        tokens — grouped-GEMM HIP kernel on GPU, per-expert loop on CPU.
+        (Also used by peft/lora_experts.py adapters.)"""
+        cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+        if (x_perm.is_cuda and x_perm.dtype == torch.bfloat16
+                and x_perm.numel() > 0):
+            from automodel_amd.ops.grouped_gemm import grouped_linear
+
+            return grouped_linear(x_perm, w, cl)
+        outs = []
+        start = 0
+        for e, n in enumerate(cl):
+            if n == 0:
+                continue
+            outs.append(x_perm[start : start + n] @ w[e].t())
+            start += n
+        return torch.cat(outs, dim=0) if outs else x_perm[:0]
+
     def _expert_mlp_loop(self, x_perm: torch.Tensor, counts: torch.Tensor) -> torch.Tensor:
         outs = []
         start = 0

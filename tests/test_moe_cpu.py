@@ -278,3 +278,42 @@ def _ep_fsdp_fn(rank, world):
 def test_ep2_fsdp_composition():
     out = run_distributed(_ep_fsdp_fn, world=2)
     assert all(v > 0 for v in out.values())
+
+
+def test_expert_lora():
+    """LoRA on stacked expert weights: zero-init B -> identity; adapters
+    train while the base stays frozen; swap preserves the MoE forward API."""
+    from automodel_amd.moe.experts import GroupedExperts
+    from automodel_amd.peft.lora_experts import (
+        GroupedExpertsLoRA,
+        apply_lora_to_grouped_experts,
+    )
+
+    torch.manual_seed(0)
+    base = GroupedExperts(n_experts=4, hidden_size=16, intermediate_size=32)
+    base.init_weights()
+    import copy
+
+    ref = copy.deepcopy(base)
+    holder = torch.nn.ModuleDict({"experts": base})
+    n = apply_lora_to_grouped_experts(holder, dim=4, alpha=8)
+    assert n == 1 and isinstance(holder["experts"], GroupedExpertsLoRA)
+    lora = holder["experts"]
+
+    T, K = 10, 2
+    x = torch.randn(T, 16)
+    probs = torch.softmax(torch.randn(T, K), dim=-1)
+    idx = torch.randint(0, 4, (T, K))
+    # B zero-init: output identical to the frozen base
+    torch.testing.assert_close(lora(x, probs, idx), ref(x, probs, idx))
+
+    trainable = [k for k, p in lora.named_parameters() if p.requires_grad]
+    assert all("lora_" in k for k in trainable) and len(trainable) == 6
+
+    with torch.no_grad():
+        torch.nn.init.normal_(lora.lora_B_gate, std=0.1)
+    out = lora(x, probs, idx)
+    assert not torch.allclose(out, ref(x, probs, idx))
+    out.sum().backward()
+    assert lora.lora_A_gate.grad is not None
+    assert lora.base.gate_proj.grad is None
