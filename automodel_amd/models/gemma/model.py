@@ -50,6 +50,11 @@ class GemmaConfig:
     sliding_window: int | None = 4096
     query_pre_attn_scalar: float | None = 256.0
     tie_word_embeddings: bool = True
+    # --- gemma-3 extensions (models/gemma/ covers both generations) ---
+    qk_norm: bool = False                 # per-head (1+w) q/k RMSNorm
+    layer_types: list | None = None       # explicit sliding/full pattern
+    rope_local_base_freq: float | None = None  # sliding layers' theta
+    rope_scaling: dict | None = None      # global layers' scaling (linear 8x)
 
     @classmethod
     def from_hf_config(cls, hf: dict) -> "GemmaConfig":
@@ -71,6 +76,28 @@ class GemmaConfig:
             query_pre_attn_scalar=get("query_pre_attn_scalar", 256.0),
             tie_word_embeddings=get("tie_word_embeddings", True),
         )
+
+    @classmethod
+    def from_hf_gemma3(cls, hf: dict) -> "GemmaConfig":
+        """Gemma3Text configs: per-layer-type rope_parameters, qk-norm, no
+        softcapping."""
+        if hasattr(hf, "to_dict"):
+            hf = hf.to_dict()
+        g = hf.get
+        rp = g("rope_parameters") or g("rope_scaling") or {}
+        local = (rp.get("sliding_attention") or {})
+        full = (rp.get("full_attention") or {})
+        cfg = cls.from_hf_config(hf)
+        cfg.qk_norm = True
+        cfg.attn_logit_softcapping = None
+        cfg.final_logit_softcapping = None
+        cfg.layer_types = g("layer_types")
+        cfg.rope_local_base_freq = local.get("rope_theta",
+                                             g("rope_local_base_freq", 10000.0))
+        cfg.rope_theta = full.get("rope_theta", g("rope_theta", 1000000.0))
+        if full.get("rope_type") == "linear":
+            cfg.rope_scaling = {"rope_type": "linear", "factor": full.get("factor", 8.0)}
+        return cfg
 
 
 class GemmaRMSNorm(nn.Module):
@@ -102,8 +129,15 @@ class GemmaAttention(nn.Module):
         self.o_proj = nn.Linear(H * D, cfg.hidden_size, bias=False)
         self.scale = (cfg.query_pre_attn_scalar or D) ** -0.5
         self.softcap = cfg.attn_logit_softcapping
-        # gemma-2: even layers are sliding-window, odd are global
-        self.window = cfg.sliding_window if layer_idx % 2 == 0 else None
+        if cfg.layer_types is not None:          # gemma-3 explicit pattern
+            self.is_sliding = cfg.layer_types[layer_idx] == "sliding_attention"
+        else:                                    # gemma-2: even layers sliding
+            self.is_sliding = layer_idx % 2 == 0
+        self.window = cfg.sliding_window if self.is_sliding else None
+        if cfg.qk_norm:
+            self.q_norm = GemmaRMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
+            self.k_norm = GemmaRMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
+        self.qk_norm = cfg.qk_norm
         self.backend = backend
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
@@ -111,6 +145,9 @@ class GemmaAttention(nn.Module):
         q = self.q_proj(x).view(B, S, -1, self.head_dim)
         k = self.k_proj(x).view(B, S, -1, self.head_dim)
         v = self.v_proj(x).view(B, S, -1, self.head_dim)
+        if self.qk_norm:
+            q = self.q_norm(q)
+            k = self.k_norm(k)
         q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
         if self.softcap is None and self.window is None:
             o = flash_attention(q, k, v, causal=True, scale=self.scale,
@@ -162,7 +199,9 @@ class GemmaDecoderLayer(nn.Module):
         self.pre_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
         self.post_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
 
-    def forward(self, x, cos, sin):
+    def forward(self, x, cos, sin, cos_local=None, sin_local=None):
+        if cos_local is not None and self.self_attn.is_sliding:
+            cos, sin = cos_local, sin_local   # gemma-3 dual-frequency rope
         x = x + self.post_attention_layernorm(self.self_attn(self.input_layernorm(x), cos, sin))
         x = x + self.post_feedforward_layernorm(self.mlp(self.pre_feedforward_layernorm(x)))
         return x
@@ -195,9 +234,18 @@ class GemmaForCausalLM(nn.Module):
         )
         inner.norm = GemmaRMSNorm(config.hidden_size, config.rms_norm_eps, backend.rms_norm)
         cos, sin = build_rope_cache(config.head_dim, config.max_position_embeddings,
-                                    config.rope_theta)
+                                    config.rope_theta, config.rope_scaling)
         inner.register_buffer("rope_cos", cos, persistent=False)
         inner.register_buffer("rope_sin", sin, persistent=False)
+        if config.rope_local_base_freq is not None:
+            cl, sl = build_rope_cache(config.head_dim,
+                                      config.max_position_embeddings,
+                                      config.rope_local_base_freq)
+            inner.register_buffer("rope_cos_local", cl, persistent=False)
+            inner.register_buffer("rope_sin_local", sl, persistent=False)
+        else:
+            inner.rope_cos_local = None
+            inner.rope_sin_local = None
         self.model = inner
         self.lm_head = nn.Linear(config.hidden_size, config.vocab_size, bias=False)
         if config.tie_word_embeddings:
@@ -215,8 +263,17 @@ class GemmaForCausalLM(nn.Module):
             cos, sin = m.rope_cos[position_ids[0]], m.rope_sin[position_ids[0]]
         if cos.dtype != torch.float32:
             cos, sin = cos.float(), sin.float()
+        cos_l = sin_l = None
+        if m.rope_cos_local is not None:
+            if position_ids is None:
+                cos_l, sin_l = m.rope_cos_local[:S], m.rope_sin_local[:S]
+            else:
+                cos_l = m.rope_cos_local[position_ids[0]]
+                sin_l = m.rope_sin_local[position_ids[0]]
+            if cos_l.dtype != torch.float32:
+                cos_l, sin_l = cos_l.float(), sin_l.float()
         for layer in m.layers:
-            x = layer(x, cos, sin)
+            x = layer(x, cos, sin, cos_l, sin_l)
         hidden = m.norm(x)
         if labels is not None:
             assert self.loss_fn is not None, "set model.loss_fn before passing labels"
@@ -249,3 +306,14 @@ class GemmaForCausalLM(nn.Module):
                 nn.init.zeros_(mod.weight)
         if self.config.tie_word_embeddings:
             self.lm_head.weight = self.model.embed_tokens.weight
+
+
+class Gemma3ForCausalLM(GemmaForCausalLM):
+    """Gemma-3 text model: Gemma-2 base + qk-norm, explicit layer_types,
+    dual-frequency rope (local theta on sliding layers), no softcapping."""
+
+    hf_architectures = ("Gemma3ForCausalLM", "Gemma3ForConditionalGeneration")
+
+    @staticmethod
+    def config_from_hf(hf_cfg) -> GemmaConfig:
+        return GemmaConfig.from_hf_gemma3(hf_cfg)

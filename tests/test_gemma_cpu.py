@@ -98,3 +98,30 @@ def test_gemma_generate_and_train_step():
     loss.backward()
     assert m.model.layers[0].mlp.gate_proj.weight.grad is not None
     assert m.lm_head.weight is m.model.embed_tokens.weight  # tied
+
+
+def test_gemma3_hf_logits_parity():
+    """Gemma-3: qk-norm, explicit layer_types, dual-frequency rope
+    (local theta on sliding layers, global theta on full layers)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.gemma.model import Gemma3ForCausalLM
+
+    hf_cfg = transformers.Gemma3TextConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=7, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rms_norm_eps=1e-6,
+        sliding_window=8, query_pre_attn_scalar=16,
+        attn_implementation="eager", tie_word_embeddings=False,
+    )
+    torch.manual_seed(13)
+    hf = transformers.Gemma3ForCausalLM(hf_cfg).eval()
+    mine = Gemma3ForCausalLM(Gemma3ForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    assert mine.config.qk_norm and mine.config.layer_types is not None
+    assert mine.config.rope_local_base_freq == 10000.0
+    assert mine.config.rope_theta == 1000000.0
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
