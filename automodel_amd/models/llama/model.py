@@ -43,6 +43,8 @@ class LlamaConfig:
     attention_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
+    bidirectional: bool = False    # no causal mask (retrieval embedding models,
+                                   # reference models/llama_bidirectional/)
     fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
     fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
     initializer_range: float = 0.02
@@ -122,7 +124,11 @@ class LlamaAttention(nn.Module):
             q = self.q_norm(q)
             k = self.k_norm(k)
         q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
-        if active_cp() is not None:
+        if self.cfg.bidirectional:
+            from automodel_amd.ops.attention import _sdpa
+
+            o = _sdpa(q, k, v, causal=False, scale=None)
+        elif active_cp() is not None:
             o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
         else:
             from automodel_amd.utils.kv_cache import active_kv_cache, maybe_update_kv

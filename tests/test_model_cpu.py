@@ -193,3 +193,28 @@ def test_hf_logits_parity_llama3_rope_scaling():
     ids = torch.randint(0, 300, (1, 100))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_bidirectional_llama_attends_to_future():
+    """bidirectional=True (retrieval embedding models): early positions see
+    later tokens; causal model's position-0 hidden is input-suffix-invariant."""
+    cfg = dict(vocab_size=100, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+               max_position_embeddings=64)
+    torch.manual_seed(0)
+    causal = LlamaForCausalLM(cfg)
+    causal.init_weights(device="cpu")
+    torch.manual_seed(0)
+    bidir = LlamaForCausalLM(dict(cfg, bidirectional=True))
+    bidir.init_weights(device="cpu")
+
+    a = torch.randint(0, 100, (1, 8))
+    b = a.clone()
+    b[0, -1] = (b[0, -1] + 1) % 100  # change only the LAST token
+    with torch.no_grad():
+        hc_a = causal(a, return_hidden=True)
+        hc_b = causal(b, return_hidden=True)
+        hb_a = bidir(a, return_hidden=True)
+        hb_b = bidir(b, return_hidden=True)
+    torch.testing.assert_close(hc_a[:, 0], hc_b[:, 0])      # causal: unchanged
+    assert not torch.allclose(hb_a[:, 0], hb_b[:, 0])       # bidir: changed
