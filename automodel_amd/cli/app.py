@@ -26,6 +26,7 @@ RECIPE_ALIASES = {
     "llm_seq_cls": "automodel_amd.recipes.llm.train_seq_cls.TrainFinetuneRecipeForSequenceClassification",
     "retrieval": "automodel_amd.recipes.llm.train_retrieval.TrainRecipeForRetrieval",
     "eagle_draft": "automodel_amd.speculative.train_draft.TrainEagleDraftRecipe",
+    "llm_dllm": "automodel_amd.recipes.llm.train_dllm.TrainDiffusionLMRecipe",
 }
 
 
