@@ -53,8 +53,11 @@ class LinearLoRA(nn.Module):
         self.scale = alpha / dim
         self.use_dora = use_dora
         base_w = self._base_weight(materialize=use_dora)
-        dtype = base_w.dtype if base_w is not None else torch.float32
-        dev = base_w.device if base_w is not None else next(iter(base.buffers())).device
+        if base_w is not None:
+            dtype, dev = base_w.dtype, base_w.device
+        else:  # NF4 base: match the dtype the base computed in pre-quantization
+            dtype = getattr(base, "compute_dtype", torch.float32)
+            dev = next(iter(base.buffers())).device
         self.lora_A = nn.Linear(base.in_features, dim, bias=False, dtype=dtype, device=dev)
         self.lora_B = nn.Linear(dim, base.out_features, bias=False, dtype=dtype, device=dev)
         self.dropout = nn.Dropout(dropout) if dropout > 0 else nn.Identity()

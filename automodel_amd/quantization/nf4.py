@@ -68,6 +68,7 @@ class NF4Linear(nn.Module):
         self.in_features = base.in_features
         self.out_features = base.out_features
         self.block_size = block_size
+        self.compute_dtype = base.weight.dtype   # adapters/dequant match this
         packed, absmax = quantize_nf4(base.weight, block_size)
         self.register_buffer("weight_packed", packed)
         self.register_buffer("weight_absmax", absmax)

@@ -175,3 +175,15 @@ def test_gemv_kernel_parity():
         torch.testing.assert_close(y.float(), ref, atol=5e-2, rtol=5e-2)
         y2 = gemv_bf16(x, w, None)
         torch.testing.assert_close(y2.float(), ref - bias.float(), atol=5e-2, rtol=5e-2)
+
+
+def test_qlora_adapter_dtype_matches_base():
+    """Adapters under an NF4 base must inherit the base's compute dtype
+    (bf16 base -> bf16 adapters), or decode-dtype GEMMs fail."""
+    from automodel_amd.peft.lora import LinearLoRA
+
+    base = nn.Linear(64, 32).to(torch.bfloat16)
+    lora = LinearLoRA(NF4Linear(base), dim=4, alpha=8)
+    assert lora.lora_A.weight.dtype == torch.bfloat16
+    x = torch.randn(2, 64, dtype=torch.bfloat16)
+    assert lora(x).dtype == torch.bfloat16
