@@ -129,7 +129,15 @@ class LlamaAttention(nn.Module):
 
             o = _sdpa(q, k, v, causal=False, scale=None)
         elif active_cp() is not None:
-            o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+            from automodel_amd.ops import attention as attn_mod
+
+            if attn_mod._VARLEN_CU is not None:
+                # packed documents under CP: block-diagonal-correct path
+                from automodel_amd.parallel.cp import cp_blockdiag_attention
+
+                o = cp_blockdiag_attention(q, k, v, attn_mod._VARLEN_CU)
+            else:
+                o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
         else:
             from automodel_amd.utils.kv_cache import active_kv_cache, maybe_update_kv
 

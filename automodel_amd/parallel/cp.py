@@ -142,3 +142,45 @@ def cp_flash_attention(q, k, v, causal: bool = True, scale: float | None = None,
                             q_start=g * C)
         )
     return torch.cat(outs, dim=1)
+
+
+def local_global_positions(rank: int, world: int, C: int,
+                           device=None) -> torch.Tensor:
+    """Global positions of this rank's 2C zigzag rows (chunk g0 then g1)."""
+    g0, g1 = zigzag_chunk_ids(rank, world)
+    return torch.cat([
+        torch.arange(g0 * C, (g0 + 1) * C, device=device),
+        torch.arange(g1 * C, (g1 + 1) * C, device=device),
+    ])
+
+
+def cp_blockdiag_attention(q, k, v, cu_seqlens: torch.Tensor,
+                           scale: float | None = None) -> torch.Tensor:
+    """Packed-document-correct CP attention (reference blockdiag_cp/
+    exchange.py all-gather strategy + runtime.py cp_blockdiag_sdpa): each
+    local (zigzag-sharded) q row attends only within its document AND
+    causally, against the all-gathered global K/V. Mask is built from the
+    GLOBAL cu_seqlens, so document boundaries are exact regardless of how
+    the shard cuts documents."""
+    import math
+
+    cp = _ACTIVE_CP
+    assert cp is not None, "cp_blockdiag_attention called without enable_cp"
+    B, S2 = q.shape[0], q.shape[1]
+    C = S2 // 2
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    kg = _GatherSeqZigzag.apply(k, cp)
+    vg = _GatherSeqZigzag.apply(v, cp)
+    T = kg.shape[1]
+    dev = q.device
+    bounds = cu_seqlens.to(dev)[1:-1]
+    doc_all = torch.bucketize(torch.arange(T, device=dev), bounds, right=True)
+    gpos = local_global_positions(cp.rank, cp.world, C, device=dev)
+    allowed = (doc_all[gpos][:, None] == doc_all[None, :]) \
+        & (torch.arange(T, device=dev)[None, :] <= gpos[:, None])
+    mask = torch.where(allowed, 0.0, float("-inf")) \
+        .to(q.dtype).reshape(1, 1, S2, T)
+
+    from automodel_amd.ops.attention import sdpa_masked
+
+    return sdpa_masked(q, kg, vg, mask, scale)
