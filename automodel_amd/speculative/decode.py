@@ -83,15 +83,19 @@ class EagleProposer:
 
     @torch.no_grad()
     def observe(self, ids: torch.Tensor) -> None:
-        """Refresh the carry from the target's aux hiddens over the full
-        verified prefix (called after every verify round)."""
+        """Invalidate the carry after a verify round; the refresh happens
+        lazily in propose() (ONE aux forward per round, not two)."""
+        self.carry = None
+
+    @torch.no_grad()
+    def _refresh(self, ids: torch.Tensor) -> None:
         _, aux = self.target.forward_with_aux(ids, self.aux_layers)
         self.carry = self.draft.fuse_aux(aux)
 
     @torch.no_grad()
     def propose(self, ids: torch.Tensor, gamma: int) -> torch.Tensor:
         if self.carry is None or self.carry.shape[1] != ids.shape[1]:
-            self.observe(ids)
+            self._refresh(ids)
         cur_ids, cur_carry = ids, self.carry
         out = []
         for _ in range(gamma):
