@@ -52,7 +52,7 @@ class QATConfig:
     quantizer_type: str = "int8_dynact_int4weight"
     group_size: int = 32
     delay_steps: int = 0
-    skip_modules: tuple = ("lm_head",)
+    skip_modules: tuple = ("lm_head", "lora_")  # never fake-quant adapters
 
     def __post_init__(self):
         valid = ("int8_dynact_int4weight", "int4_weight_only")

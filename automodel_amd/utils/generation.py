@@ -64,6 +64,8 @@ def generate_graphed(
     with kv_cache_context(cache):
         cache.begin_forward()
         logits = model(input_ids)                      # prefill (host-indexed)
+        assert cache._layer == model.config.num_hidden_layers, (
+            "model did not consult the KV cache — static decode unsupported")
         cache.advance(T)
         cache.freeze_for_graph()                       # device position from here
         nxt = logits[:, -1].argmax(-1, keepdim=True)   # [B, 1]
@@ -122,6 +124,9 @@ def generate_cached(
     with kv_cache_context(cache):
         cache.begin_forward()
         logits = model(input_ids)               # prefill
+        assert cache._layer == model.config.num_hidden_layers, (
+            "model did not consult the KV cache (attention lacks "
+            "maybe_update_kv) — cached decode would be silently wrong")
         cache.advance(T)
         nxt = _pick_next(logits[:, -1].float(), temperature)
         ids = torch.cat([ids, nxt], dim=1)

@@ -135,3 +135,14 @@ def test_qat_recipe_integration(tmp_path):
     n_qat = sum(1 for m in r.model.modules() if isinstance(m, QATLinear))
     assert n_qat > 0
     r.run_train_validation_loop()
+
+
+def test_prepare_qat_skips_lora_adapters():
+    from automodel_amd.peft.lora import LinearLoRA
+
+    m = nn.ModuleDict({"q_proj": LinearLoRA(nn.Linear(32, 32), dim=4, alpha=8)})
+    n = prepare_qat(m, {"group_size": 16})
+    # base linear inside the LoRA wrapper is quantized; adapters are NOT
+    assert n == 1
+    assert isinstance(m["q_proj"].base, QATLinear)
+    assert type(m["q_proj"].lora_A) is nn.Linear

@@ -3,6 +3,7 @@ model, and the EAGLE draft-training recipe."""
 
 import pytest
 import torch
+import torch.nn as nn
 
 from automodel_amd.config.loader import ConfigNode
 from automodel_amd.models.llama.model import LlamaForCausalLM
@@ -193,3 +194,24 @@ def test_generate_graphed_gpu_hipgraph():
     a = generate_graphed(m, prompt, max_new_tokens=12, use_hip_graph=False)
     b = generate_graphed(m, prompt, max_new_tokens=12, use_hip_graph=True)
     torch.testing.assert_close(a, b)
+
+
+def test_generate_cached_rejects_cache_blind_model():
+    """Models whose attention ignores the cache must fail loudly, not
+    silently mis-decode."""
+    from automodel_amd.utils.generation import generate_cached
+
+    class Blind(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.config = type("C", (), {"num_hidden_layers": 2,
+                                         "num_key_value_heads": 1,
+                                         "head_dim": 8})()
+            self.emb = nn.Embedding(50, 16)
+            self.head = nn.Linear(16, 50)
+
+        def forward(self, ids, **kw):
+            return self.head(self.emb(ids))
+
+    with pytest.raises(AssertionError, match="consult the KV cache"):
+        generate_cached(Blind(), torch.randint(0, 50, (1, 4)), max_new_tokens=2)
