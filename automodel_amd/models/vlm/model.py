@@ -90,7 +90,7 @@ class VisionTower(nn.Module):
 
 
 class VLMForConditionalGeneration(nn.Module):
-    hf_architectures = ("Qwen2VLForConditionalGeneration",)
+    hf_architectures = ("GenericVLMForConditionalGeneration",)  # qwen2_vl/ has the real Qwen2-VL
     config_class = VLMConfig
 
     def __init__(self, config: VLMConfig | dict, backend: BackendConfig | dict | None = None):

@@ -99,9 +99,13 @@ def _rotate_half(x: torch.Tensor) -> torch.Tensor:
 def apply_rope_ref(
     q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor
 ) -> tuple[torch.Tensor, torch.Tensor]:
-    """q,k: [B, S, H, D]; cos/sin: [S, D] fp32."""
-    c = cos.to(q.dtype)[None, :, None, :]
-    s = sin.to(q.dtype)[None, :, None, :]
+    """q,k: [B, S, H, D]; cos/sin: [S, D] (or [B, S, D] for m-rope) fp32."""
+    if cos.dim() == 3:                     # batch-shaped tables (qwen2-vl m-rope)
+        c = cos.to(q.dtype)[:, :, None, :]
+        s = sin.to(q.dtype)[:, :, None, :]
+    else:
+        c = cos.to(q.dtype)[None, :, None, :]
+        s = sin.to(q.dtype)[None, :, None, :]
     return q * c + _rotate_half(q) * s, k * c + _rotate_half(k) * s
 
 
@@ -129,6 +133,6 @@ def apply_rope(
     backend: str = "hip",
 ) -> tuple[torch.Tensor, torch.Tensor]:
     """Apply RoPE to q [B,S,Hq,D] and k [B,S,Hk,D] with position-sliced tables."""
-    if backend == "hip" and q.is_cuda:
+    if backend == "hip" and q.is_cuda and cos.dim() == 2:
         return _RopeHip.apply(q, k, cos, sin)
     return apply_rope_ref(q, k, cos, sin)

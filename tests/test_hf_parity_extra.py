@@ -227,3 +227,70 @@ def test_gpt_oss_trains():
     loss.backward()
     assert m.model.layers[0].self_attn.sinks.grad is not None
     assert m.model.layers[0].mlp.experts.gate_up_proj.grad is not None
+
+
+def test_hf_logits_parity_qwen2_vl():
+    """Full Qwen2-VL: ViT tower (conv3d patch embed, 2-axis rotary,
+    per-image attention, quick-gelu, 2x2 merger), image splice, and m-rope
+    text decoder — text-only AND text+image paths."""
+    from automodel_amd.models.qwen2_vl.model import Qwen2VLForConditionalGeneration
+
+    torch.manual_seed(20)
+    hf_cfg = transformers.Qwen2VLConfig(
+        text_config=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         num_key_value_heads=2, max_position_embeddings=256,
+                         rope_theta=10000.0, rms_norm_eps=1e-6,
+                         rope_scaling={"type": "mrope", "mrope_section": [2, 3, 3]},
+                         tie_word_embeddings=False),
+        vision_config=dict(embed_dim=32, depth=2, num_heads=2, hidden_size=64,
+                           patch_size=4, temporal_patch_size=2,
+                           spatial_merge_size=2, in_channels=3, mlp_ratio=4),
+        image_token_id=299, vision_start_token_id=298,
+    )
+    hf = transformers.Qwen2VLForConditionalGeneration(hf_cfg).eval()
+    mine = Qwen2VLForConditionalGeneration(
+        Qwen2VLForConditionalGeneration.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k or "rot_inv" in k for k in missing), missing
+
+    ids = torch.randint(0, 290, (2, 20))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=2e-4, rtol=2e-4)
+
+    grid = torch.tensor([[1, 4, 4]])
+    pix = torch.randn(16, 3 * 2 * 4 * 4)
+    seq = torch.cat([torch.randint(0, 290, (1, 3)), torch.tensor([[298]]),
+                     torch.full((1, 4), 299), torch.randint(0, 290, (1, 5))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix, image_grid_thw=grid,
+                 mm_token_type_ids=(seq == 299).int()).logits
+        out = mine(seq, pixel_values=pix, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_qwen2_vl_trains():
+    from automodel_amd.loss.masked_ce import MaskedCrossEntropy
+    from automodel_amd.models.qwen2_vl.model import Qwen2VLForConditionalGeneration
+
+    m = Qwen2VLForConditionalGeneration(dict(
+        text=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, max_position_embeddings=256,
+                  attention_bias=True),
+        vision=dict(embed_dim=32, depth=1, num_heads=2, hidden_size=64,
+                    patch_size=4, temporal_patch_size=2, spatial_merge_size=2),
+        mrope_section=(2, 3, 3), image_token_id=299,
+    ))
+    m.init_weights(device="cpu")
+    m.loss_fn = lambda h, w, l: MaskedCrossEntropy()(h @ w.t(), l)
+    seq = torch.cat([torch.randint(0, 290, (1, 4)), torch.full((1, 4), 299),
+                     torch.randint(0, 290, (1, 4))], dim=1)
+    pix = torch.randn(16, 3 * 2 * 4 * 4)
+    loss = m(seq, pixel_values=pix, image_grid_thw=torch.tensor([[1, 4, 4]]),
+             labels=seq.clone())
+    loss.backward()
+    assert m.model.visual.blocks[0].attn.qkv.weight.grad is not None
+    assert m.model.language_model.layers[0].self_attn.q_proj.weight.grad is not None

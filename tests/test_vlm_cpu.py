@@ -54,7 +54,7 @@ def test_vlm_recipe_end_to_end(tmp_path):
     cfg = ConfigNode({
         "seed": 1,
         "model": {
-            "architecture": "Qwen2VLForConditionalGeneration",
+            "architecture": "GenericVLMForConditionalGeneration",
             "config": {"text": TEXT, "vision": VISION},
             "dtype": "float32",
         },
