@@ -83,3 +83,32 @@ def test_async_checkpoint_writer(tmp_path):
     # second save waits for the first implicitly
     w.save_async({"x": torch.zeros(1)}, write_fn)
     w.wait()
+
+
+def test_generation_server():
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+    from automodel_amd.serving.server import build_app
+    from automodel_amd.utils.generation import generate
+
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(dict(vocab_size=150, hidden_size=32,
+                                  intermediate_size=64, num_hidden_layers=2,
+                                  num_attention_heads=2, num_key_value_heads=1,
+                                  max_position_embeddings=128))
+    model.init_weights(device="cpu")
+    model.eval()
+    app = build_app(model, tokenizer=None, use_cache=True)
+    client = TestClient(app)
+
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+
+    prompt = [5, 9, 23, 44]
+    r = client.post("/generate", json={"prompt_ids": prompt, "max_new_tokens": 8})
+    assert r.status_code == 200
+    out = r.json()["output_ids"]
+    ref = generate(model, torch.tensor([prompt]), max_new_tokens=8)
+    assert out == ref[0, 4:].tolist()   # server(KV-cached) == plain greedy
