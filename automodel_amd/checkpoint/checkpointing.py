@@ -118,6 +118,15 @@ class Checkpointer:
         if adapter is not None:
             full_sd = adapter.to_hf(full_sd)
         full_sd = {k: v.contiguous() for k, v in full_sd.items() if isinstance(v, torch.Tensor)}
+        # drop tied aliases (safetensors refuses shared storage; HF convention
+        # omits the tied lm_head — loaders re-tie from config)
+        seen: dict[int, str] = {}
+        for k in list(full_sd):
+            ptr = full_sd[k].untyped_storage().data_ptr()
+            if ptr in seen:
+                del full_sd[k]
+            else:
+                seen[ptr] = k
         total_bytes = sum(v.numel() * v.element_size() for v in full_sd.values())
         max_shard = 4 * 2**30
         if total_bytes <= max_shard:

@@ -198,3 +198,43 @@ def test_dllm_recipe_end_to_end(tmp_path):
     r.setup()
     assert r.model.config.bidirectional
     r.run_train_validation_loop()
+
+
+def test_hf_export_roundtrip_gemma(tmp_path):
+    """export_hf_safetensors -> build_model(pretrained_path) round-trips the
+    gemma family (newer families must survive the consolidation path too)."""
+    import torch
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.models.gemma.model import GemmaForCausalLM
+
+    torch.manual_seed(0)
+    cfg = dict(vocab_size=200, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=2,
+               num_key_value_heads=1, head_dim=16, max_position_embeddings=64,
+               sliding_window=8, query_pre_attn_scalar=16.0)
+    m = GemmaForCausalLM(cfg)
+    m.init_weights(device="cpu")
+    out = tmp_path / "export"
+    Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
+    import json
+    import os
+
+    assert os.path.exists(out / "model.safetensors.index.json") or \
+        any(f.endswith(".safetensors") for f in os.listdir(out))
+    cj = json.load(open(out / "config.json"))
+    assert cj["architectures"] == ["GemmaForCausalLM"]
+
+    from safetensors.torch import load_file
+
+    files = [f for f in os.listdir(out) if f.endswith(".safetensors")]
+    sd = {}
+    for f in files:
+        sd.update(load_file(str(out / f)))
+    m2 = GemmaForCausalLM(cfg)
+    m2.init_weights(device="cpu")
+    missing, unexpected = m2.load_state_dict(sd, strict=False)
+    assert not unexpected
+    x = torch.randint(0, 200, (1, 8))
+    with torch.no_grad():
+        torch.testing.assert_close(m(x), m2(x))
