@@ -452,3 +452,53 @@ def test_hip_graph_training_capture():
     ref(xr).sum().backward()
     assert torch.allclose(y.float(), ref(xr).float(), atol=1e-4)
     assert torch.allclose(x.grad, xr.grad, atol=1e-4)
+
+
+def test_new_families_gpu_smoke():
+    """Gemma-2, GPT-OSS and Qwen2-VL run forward+backward on MI355X bf16
+    (sdpa attention paths + HIP rms/rope where wired)."""
+    from automodel_amd.models.gemma.model import GemmaForCausalLM
+    from automodel_amd.models.gpt_oss.model import GptOssForCausalLM
+    from automodel_amd.models.qwen2_vl.model import Qwen2VLForConditionalGeneration
+
+    def bf16(m):
+        return m.to(torch.bfloat16)
+
+    g = GemmaForCausalLM(dict(vocab_size=200, hidden_size=64, intermediate_size=128,
+                              num_hidden_layers=2, num_attention_heads=4,
+                              num_key_value_heads=2, head_dim=16,
+                              max_position_embeddings=64, sliding_window=8,
+                              query_pre_attn_scalar=16.0))
+    g.init_weights(device="cuda")
+    g = bf16(g)
+    ids = torch.randint(0, 200, (2, 16), device="cuda")
+    g(ids).float().sum().backward()
+    assert g.model.layers[0].mlp.gate_proj.weight.grad is not None
+
+    o = GptOssForCausalLM(dict(vocab_size=200, hidden_size=64, intermediate_size=96,
+                               num_hidden_layers=2, num_attention_heads=4,
+                               num_key_value_heads=2, head_dim=16,
+                               num_local_experts=4, num_experts_per_tok=2,
+                               max_position_embeddings=64, sliding_window=8))
+    o.init_weights(device="cuda")
+    o = bf16(o)
+    o(ids).float().sum().backward()
+    assert o.model.layers[0].mlp.experts.gate_up_proj.grad is not None
+
+    v = Qwen2VLForConditionalGeneration(dict(
+        text=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, max_position_embeddings=64,
+                  attention_bias=True),
+        vision=dict(embed_dim=32, depth=1, num_heads=2, hidden_size=64,
+                    patch_size=4, temporal_patch_size=2, spatial_merge_size=2),
+        mrope_section=(2, 3, 3), image_token_id=299))
+    v.init_weights(device="cuda")
+    v = bf16(v)
+    seq = torch.cat([torch.randint(0, 290, (1, 4), device="cuda"),
+                     torch.full((1, 4), 299, device="cuda"),
+                     torch.randint(0, 290, (1, 4), device="cuda")], dim=1)
+    pix = torch.randn(16, 3 * 2 * 4 * 4, device="cuda", dtype=torch.bfloat16)
+    out = v(seq, pixel_values=pix, image_grid_thw=torch.tensor([[1, 4, 4]]))
+    out.float().sum().backward()
+    assert v.model.visual.blocks[0].attn.qkv.weight.grad is not None
