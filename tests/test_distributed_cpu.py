@@ -368,3 +368,48 @@ def _tp_fsdp_worker(rank, world):
 
 def test_tp2_x_dpshard2_world4():
     run_distributed(_tp_fsdp_worker, world=4)
+
+
+def _cp_dp_worker(rank, world):
+    """2D composition: CP=2 x dp_shard=2 — sharded loss matches single-run."""
+    import torch.distributed as dist
+
+    from automodel_amd.loss.masked_ce import MaskedCrossEntropy
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+    from automodel_amd.parallel.cp import disable_cp, enable_cp, shard_batch_cp
+    from automodel_amd.parallel.fsdp import apply_fsdp
+    from automodel_amd.parallel.mesh import build_mesh
+
+    cfg = dict(vocab_size=128, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=4,
+               num_key_value_heads=2, max_position_embeddings=64)
+    torch.manual_seed(0)
+    ref = LlamaForCausalLM(cfg)
+    ref.init_weights(device="cpu")
+    ids = torch.randint(0, 128, (4, 16), generator=torch.Generator().manual_seed(7))
+    ce = MaskedCrossEntropy()
+    ref_loss = ce(ref(ids), ids) / ids.numel()
+
+    ctx = build_mesh(dp_shard=2, cp=2, device_type="cpu")
+    torch.manual_seed(0)
+    model = LlamaForCausalLM(cfg)
+    model.init_weights(device="cpu")
+    apply_fsdp(model, ctx.mesh["dp_shard_cp"], param_dtype=torch.float32)
+    enable_cp(ctx.mesh["cp"])
+    try:
+        dp_rank = ctx.mesh["dp_shard"].get_local_rank()
+        cp_rank = ctx.mesh["cp"].get_local_rank()
+        shard = {"input_ids": ids[dp_rank * 2 : dp_rank * 2 + 2],
+                 "labels": ids[dp_rank * 2 : dp_rank * 2 + 2].clone()}
+        local = shard_batch_cp(shard, cp_rank, 2)
+        out = model(local["input_ids"], position_ids=local["position_ids"])
+        loss = ce(out, local["labels"]) / ids.numel()
+        t = loss.detach().clone()
+        dist.all_reduce(t)   # sum over all 4 ranks = full-batch loss
+        torch.testing.assert_close(t, ref_loss.detach(), atol=1e-5, rtol=1e-5)
+    finally:
+        disable_cp()
+
+
+def test_cp2_x_dpshard2_world4():
+    run_distributed(_cp_dp_worker, world=4)
