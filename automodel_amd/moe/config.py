@@ -20,4 +20,5 @@ class MoEConfig:
     bias_update_speed: float = 1e-3
     moe_intermediate_size: int | None = None
     shared_expert_intermediate_size: int | None = None
+    shared_expert_gate: bool = False   # qwen2-moe: sigmoid-gated shared expert
     fake_balanced_gate: bool = False      # benchmark ideal routing

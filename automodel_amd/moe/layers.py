@@ -128,6 +128,8 @@ class MoE(nn.Module):
         if cfg.n_shared_experts > 0:
             shared_inter = cfg.shared_expert_intermediate_size or inter * cfg.n_shared_experts
             self.shared_experts = SharedExpert(hidden_size, shared_inter)
+            if cfg.shared_expert_gate:   # qwen2-moe sigmoid token gate
+                self.shared_expert_gate = nn.Linear(hidden_size, 1, bias=False)
         # per-forward expert load (for metrics + bias update)
         self.last_expert_load: torch.Tensor | None = None
 
@@ -145,7 +147,10 @@ class MoE(nn.Module):
         else:
             y = self.experts(xf, probs, indices)
         if self.shared_experts is not None:
-            y = y + self.shared_experts(xf)
+            shared = self.shared_experts(xf)
+            if getattr(self, "shared_expert_gate", None) is not None:
+                shared = torch.sigmoid(self.shared_expert_gate(xf)) * shared
+            y = y + shared
         # gate.last_aux_loss stays attached to the graph; the model adds
         # sum-of-aux to the main loss before backward (reference
         # MoEAuxLossAutoScaler moe_utils.py:568 achieves the same coupling).

@@ -47,6 +47,19 @@ class MoEModelConfig(LlamaConfig):
                 norm_topk_prob=True,
             )
             flavor = "mixtral"
+        elif "Qwen2Moe" in arch:
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_experts", 60),
+                n_activated_experts=hf.get("num_experts_per_tok", 4),
+                n_shared_experts=1,
+                shared_expert_intermediate_size=hf.get(
+                    "shared_expert_intermediate_size", 5632),
+                shared_expert_gate=True,
+                moe_intermediate_size=hf.get("moe_intermediate_size", 1408),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                norm_topk_prob=hf.get("norm_topk_prob", False),
+            )
+            flavor = "qwen2_moe"
         else:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 64),

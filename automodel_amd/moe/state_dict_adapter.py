@@ -48,6 +48,8 @@ class MoEStateDictAdapter:
                 gate, up = t.chunk(2, dim=1)
                 out[key.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
                 out[key.replace("gate_up_proj", "up_proj")] = up.contiguous()
+            elif ".mlp.shared_expert." in key:
+                out[key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")] = t
             elif ".block_sparse_moe.gate." in key:
                 out[key.replace(".block_sparse_moe.gate.", ".mlp.gate.")] = t
             elif ".block_sparse_moe." in key:
@@ -61,6 +63,9 @@ class MoEStateDictAdapter:
     # ---- internal -> HF (unstack)
     def to_hf(self, sd: dict[str, torch.Tensor]) -> dict[str, torch.Tensor]:
         out: dict[str, torch.Tensor] = {}
+        if self.flavor == "qwen2_moe":
+            sd = {k.replace(".mlp.shared_experts.", ".mlp.shared_expert."): v
+                  for k, v in sd.items()}
         pat = re.compile(r"^(model\.layers\.\d+)\.mlp\.experts\.(gate_proj|up_proj|down_proj)$")
         for key, t in sd.items():
             m = pat.match(key)

@@ -294,3 +294,31 @@ def test_qwen2_vl_trains():
     loss.backward()
     assert m.model.visual.blocks[0].attn.qkv.weight.grad is not None
     assert m.model.language_model.layers[0].self_attn.q_proj.weight.grad is not None
+
+
+def test_hf_logits_parity_qwen2_moe():
+    """Qwen2-MoE: softmax-all routing without topk renorm + sigmoid-GATED
+    shared expert (shared_expert_gate) + qkv bias."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(30)
+    hf_cfg = transformers.Qwen2MoeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        num_experts=4, num_experts_per_tok=2, moe_intermediate_size=48,
+        shared_expert_intermediate_size=96, norm_topk_prob=False,
+        decoder_sparse_step=1, mlp_only_layers=[],
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.Qwen2MoeForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Qwen2MoeForCausalLM"]
+    mine = MoEForCausalLM(MoEModelConfig.from_hf_config(d)).eval()
+    assert mine.config.moe.shared_expert_gate
+    sd = MoEStateDictAdapter(mine.config).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
