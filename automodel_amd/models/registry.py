@@ -72,6 +72,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.nemotron.model import NemotronForCausalLM
+
+        for name in NemotronForCausalLM.hf_architectures:
+            _REGISTRY[name] = NemotronForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.qwen2_vl.model import Qwen2VLForConditionalGeneration
 
         for name in Qwen2VLForConditionalGeneration.hf_architectures:

@@ -322,3 +322,32 @@ def test_hf_logits_parity_qwen2_moe():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_nemotron():
+    """Nemotron: LayerNorm-1P (bias, (1+w) gain), squared-ReLU MLP,
+    partial rotary (factor 0.5)."""
+    from automodel_amd.models.nemotron.model import NemotronForCausalLM
+
+    torch.manual_seed(31)
+    hf_cfg = transformers.NemotronConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False)
+    hf = transformers.NemotronForCausalLM(hf_cfg).eval()
+    mine = NemotronForCausalLM(
+        NemotronForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+    # trains
+    from automodel_amd.loss.masked_ce import MaskedCrossEntropy
+
+    mine.loss_fn = lambda h, w, l: MaskedCrossEntropy()(h @ w.t(), l)
+    loss = mine(ids, labels=ids.clone())
+    loss.backward()
+    assert mine.model.layers[0].mlp.up_proj.weight.grad is not None
