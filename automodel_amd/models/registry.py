@@ -72,6 +72,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.glm4_moe.model import Glm4MoeForCausalLM
+
+        for name in Glm4MoeForCausalLM.hf_architectures:
+            _REGISTRY[name] = Glm4MoeForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.nemotron.model import NemotronForCausalLM
 
         for name in NemotronForCausalLM.hf_architectures:

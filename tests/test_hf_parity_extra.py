@@ -351,3 +351,29 @@ def test_hf_logits_parity_nemotron():
     loss = mine(ids, labels=ids.clone())
     loss.backward()
     assert mine.model.layers[0].mlp.up_proj.weight.grad is not None
+
+
+def test_hf_logits_parity_glm4_moe():
+    """GLM4-MoE: DeepSeek-style sigmoid+bias routing and shared expert
+    under plain GQA with partial rotary (0.5)."""
+    from automodel_amd.models.glm4_moe.model import Glm4MoeForCausalLM
+
+    torch.manual_seed(33)
+    hf_cfg = transformers.Glm4MoeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False, n_routed_experts=8, n_shared_experts=1,
+        num_experts_per_tok=2, moe_intermediate_size=48,
+        first_k_dense_replace=1, norm_topk_prob=True,
+        routed_scaling_factor=1.0, head_dim=16)
+    hf = transformers.Glm4MoeForCausalLM(hf_cfg).eval()
+    mine = Glm4MoeForCausalLM(
+        Glm4MoeForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    sd = mine.state_dict_adapter.from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
