@@ -45,6 +45,7 @@ class LlamaConfig:
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
     bidirectional: bool = False    # no causal mask (retrieval embedding models,
                                    # reference models/llama_bidirectional/)
+    no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
     fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
     fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
     initializer_range: float = 0.02
@@ -77,6 +78,7 @@ class LlamaConfig:
             rms_norm_eps=get("rms_norm_eps", 1e-5),
             rope_theta=get("rope_theta", 10000.0),
             rope_scaling=get("rope_scaling", None),
+            no_rope_layers=get("no_rope_layers", None),
             tie_word_embeddings=get("tie_word_embeddings", False),
             attention_bias=get("attention_bias", get("qkv_bias", False)),
             mlp_bias=get("mlp_bias", False),
@@ -85,10 +87,12 @@ class LlamaConfig:
 
 
 class LlamaAttention(nn.Module):
-    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig,
+                 use_rope: bool = True):
         super().__init__()
         self.cfg = cfg
         self.backend = backend
+        self.use_rope = use_rope
         H, D = cfg.num_attention_heads, cfg.head_dim
         Hk = cfg.num_key_value_heads
         self.num_heads, self.num_kv_heads, self.head_dim = H, Hk, D
@@ -123,7 +127,8 @@ class LlamaAttention(nn.Module):
         if self.cfg.qk_norm:
             q = self.q_norm(q)
             k = self.k_norm(k)
-        q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
+        if self.use_rope:
+            q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
         if self.cfg.bidirectional:
             from automodel_amd.ops.attention import _sdpa
 
@@ -181,9 +186,10 @@ class LlamaMLP(nn.Module):
 
 
 class LlamaDecoderLayer(nn.Module):
-    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig,
+                 use_rope: bool = True):
         super().__init__()
-        self.self_attn = LlamaAttention(cfg, backend)
+        self.self_attn = LlamaAttention(cfg, backend, use_rope=use_rope)
         self.mlp = LlamaMLP(cfg, backend)
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
@@ -199,8 +205,11 @@ class LlamaModel(nn.Module):
         super().__init__()
         self.cfg = cfg
         self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
+        nope = cfg.no_rope_layers
         self.layers = nn.ModuleList(
-            LlamaDecoderLayer(cfg, backend) for _ in range(cfg.num_hidden_layers)
+            LlamaDecoderLayer(cfg, backend,
+                              use_rope=(nope is None or bool(nope[i])))
+            for i in range(cfg.num_hidden_layers)
         )
         self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         cos, sin = build_rope_cache(
@@ -226,7 +235,7 @@ class LlamaModel(nn.Module):
 
 class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
-                        "Qwen3ForCausalLM", "Phi3ForCausalLM")
+                        "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

@@ -377,3 +377,26 @@ def test_hf_logits_parity_glm4_moe():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_smollm3():
+    """SmolLM3: llama family + NoPE layers (no_rope_layers mask)."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(35)
+    hf_cfg = transformers.SmolLM3Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attn_implementation="eager", tie_word_embeddings=False,
+        pad_token_id=0, bos_token_id=1, eos_token_id=2)
+    hf = transformers.SmolLM3ForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["SmolLM3ForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.no_rope_layers == [1, 1, 1, 0]
+    assert not mine.model.layers[3].self_attn.use_rope
+    mine.load_state_dict(hf.state_dict(), strict=False)
+    ids = torch.randint(0, 300, (2, 20))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
