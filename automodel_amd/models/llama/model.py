@@ -43,6 +43,7 @@ class LlamaConfig:
     attention_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
+    qk_norm_full: bool = False     # OLMoE-style full-projection q/k RMSNorm
     bidirectional: bool = False    # no causal mask (retrieval embedding models,
                                    # reference models/llama_bidirectional/)
     no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
@@ -107,6 +108,9 @@ class LlamaAttention(nn.Module):
         if cfg.qk_norm:
             self.q_norm = RMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
             self.k_norm = RMSNorm(D, cfg.rms_norm_eps, backend.rms_norm)
+        elif cfg.qk_norm_full:
+            self.q_norm = RMSNorm(H * D, cfg.rms_norm_eps, backend.rms_norm)
+            self.k_norm = RMSNorm(Hk * D, cfg.rms_norm_eps, backend.rms_norm)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
         from automodel_amd.parallel.cp import active_cp, cp_flash_attention
@@ -121,8 +125,13 @@ class LlamaAttention(nn.Module):
             k = k.contiguous().view(B, S, -1, self.head_dim)
             v = v.contiguous().view(B, S, -1, self.head_dim)
         else:
-            q = self.q_proj(x).view(B, S, -1, self.head_dim)
-            k = self.k_proj(x).view(B, S, -1, self.head_dim)
+            q = self.q_proj(x)
+            k = self.k_proj(x)
+            if self.cfg.qk_norm_full:   # OLMoE: norm over the flat projection
+                q = self.q_norm(q)
+                k = self.k_norm(k)
+            q = q.view(B, S, -1, self.head_dim)
+            k = k.view(B, S, -1, self.head_dim)
             v = self.v_proj(x).view(B, S, -1, self.head_dim)
         if self.cfg.qk_norm:
             q = self.q_norm(q)

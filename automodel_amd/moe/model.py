@@ -47,6 +47,16 @@ class MoEModelConfig(LlamaConfig):
                 norm_topk_prob=True,
             )
             flavor = "mixtral"
+        elif "Olmoe" in arch:
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_experts", 64),
+                n_activated_experts=hf.get("num_experts_per_tok", 8),
+                moe_intermediate_size=hf.get("intermediate_size", 1024),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                norm_topk_prob=hf.get("norm_topk_prob", False),
+            )
+            base["qk_norm_full"] = True
+            flavor = "qwen3_moe"   # same stacked-expert key layout
         elif "Qwen2Moe" in arch:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 60),
@@ -87,7 +97,8 @@ class MoEDecoderLayer(nn.Module):
 
 
 class MoEForCausalLM(nn.Module):
-    hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM")
+    hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
+                        "OlmoeForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod

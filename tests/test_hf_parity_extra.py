@@ -400,3 +400,29 @@ def test_hf_logits_parity_smollm3():
     ids = torch.randint(0, 300, (2, 20))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_olmoe():
+    """OLMoE: full-projection q/k RMSNorm + softmax routing without topk
+    renorm, stacked experts."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(36)
+    hf_cfg = transformers.OlmoeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_experts=4, num_experts_per_tok=2, max_position_embeddings=128,
+        rope_theta=10000.0, rms_norm_eps=1e-5, attn_implementation="eager",
+        tie_word_embeddings=False, pad_token_id=0, eos_token_id=2)
+    hf = transformers.OlmoeForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["OlmoeForCausalLM"]
+    mine = MoEForCausalLM(MoEModelConfig.from_hf_config(d)).eval()
+    assert mine.config.qk_norm_full
+    sd = MoEStateDictAdapter(mine.config).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
