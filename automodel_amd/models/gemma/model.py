@@ -55,11 +55,16 @@ class GemmaConfig:
     layer_types: list | None = None       # explicit sliding/full pattern
     rope_local_base_freq: float | None = None  # sliding layers' theta
     rope_scaling: dict | None = None      # global layers' scaling (linear 8x)
+    post_norms: bool = True               # gemma-2/3: extra post-attn/ffn norms
+                                          # (gemma-1 has only the two pre-norms)
 
     @classmethod
     def from_hf_config(cls, hf: dict) -> "GemmaConfig":
         get = hf.get
+        archs = " ".join(get("architectures", []) or [])
+        v1 = bool(archs) and "Gemma2" not in archs and "Gemma3" not in archs
         return cls(
+            post_norms=not v1,
             vocab_size=get("vocab_size", 256000),
             hidden_size=get("hidden_size", 2304),
             intermediate_size=get("intermediate_size", 9216),
@@ -70,10 +75,10 @@ class GemmaConfig:
             max_position_embeddings=get("max_position_embeddings", 8192),
             rope_theta=get("rope_theta", 10000.0),
             rms_norm_eps=get("rms_norm_eps", 1e-6),
-            attn_logit_softcapping=get("attn_logit_softcapping", 50.0),
-            final_logit_softcapping=get("final_logit_softcapping", 30.0),
-            sliding_window=get("sliding_window", 4096),
-            query_pre_attn_scalar=get("query_pre_attn_scalar", 256.0),
+            attn_logit_softcapping=None if v1 else get("attn_logit_softcapping", 50.0),
+            final_logit_softcapping=None if v1 else get("final_logit_softcapping", 30.0),
+            sliding_window=None if v1 else get("sliding_window", 4096),
+            query_pre_attn_scalar=None if v1 else get("query_pre_attn_scalar", 256.0),
             tie_word_embeddings=get("tie_word_embeddings", True),
         )
 
@@ -192,16 +197,21 @@ class GemmaDecoderLayer(nn.Module):
     def __init__(self, cfg: GemmaConfig, backend: BackendConfig, layer_idx: int):
         super().__init__()
         nb = backend.rms_norm
+        self.post_norms = cfg.post_norms
         self.self_attn = GemmaAttention(cfg, backend, layer_idx)
         self.mlp = GemmaMLP(cfg)
         self.input_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
         self.post_attention_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
-        self.pre_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
-        self.post_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
+        if cfg.post_norms:
+            self.pre_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
+            self.post_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
 
     def forward(self, x, cos, sin, cos_local=None, sin_local=None):
         if cos_local is not None and self.self_attn.is_sliding:
             cos, sin = cos_local, sin_local   # gemma-3 dual-frequency rope
+        if not self.post_norms:   # gemma-1: classic pre-norm residual order
+            x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+            return x + self.mlp(self.post_attention_layernorm(x))
         x = x + self.post_attention_layernorm(self.self_attn(self.input_layernorm(x), cos, sin))
         x = x + self.post_feedforward_layernorm(self.mlp(self.pre_feedforward_layernorm(x)))
         return x

@@ -125,3 +125,26 @@ def test_gemma3_hf_logits_parity():
     ids = torch.randint(0, 300, (2, 24))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_gemma1_hf_logits_parity():
+    """Gemma-1: two-norm pre-norm layers, no softcapping/sliding —
+    the same module in post_norms=False mode."""
+    transformers = pytest.importorskip("transformers")
+    torch.manual_seed(37)
+    hf_cfg = transformers.GemmaConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+        rms_norm_eps=1e-6, attn_implementation="eager",
+        tie_word_embeddings=False)
+    hf = transformers.GemmaForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["GemmaForCausalLM"]
+    mine = GemmaForCausalLM(GemmaForCausalLM.config_from_hf(d)).eval()
+    assert not mine.config.post_norms
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
