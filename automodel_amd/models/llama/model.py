@@ -47,6 +47,7 @@ class LlamaConfig:
     bidirectional: bool = False    # no causal mask (retrieval embedding models,
                                    # reference models/llama_bidirectional/)
     no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
+    sliding_window: int | None = None   # Mistral-style windowed causal attention
     fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
     fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
     initializer_range: float = 0.02
@@ -80,6 +81,10 @@ class LlamaConfig:
             rope_theta=get("rope_theta", 10000.0),
             rope_scaling=get("rope_scaling", None),
             no_rope_layers=get("no_rope_layers", None),
+            # mistral: sliding_window set => windowed; qwen2 gates it behind
+            # use_sliding_window
+            sliding_window=(get("sliding_window")
+                            if get("use_sliding_window", True) else None),
             tie_word_embeddings=get("tie_word_embeddings", False),
             attention_bias=get("attention_bias", get("qkv_bias", False)),
             mlp_bias=get("mlp_bias", False),
@@ -138,7 +143,18 @@ class LlamaAttention(nn.Module):
             k = self.k_norm(k)
         if self.use_rope:
             q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
-        if self.cfg.bidirectional:
+        if self.cfg.sliding_window is not None:
+            # windowed causal (mistral): explicit mask via sdpa — the flash
+            # kernel's full-causal tiling doesn't window (round-2 kernel)
+            from automodel_amd.ops.attention import sdpa_masked
+
+            i = torch.arange(S, device=x.device)
+            keep = (i[None, :] <= i[:, None]) \
+                & (i[None, :] > i[:, None] - self.cfg.sliding_window)
+            mask = torch.where(keep, 0.0, float("-inf")) \
+                .to(q.dtype).reshape(1, 1, S, S)
+            o = sdpa_masked(q, k, v, mask)
+        elif self.cfg.bidirectional:
             from automodel_amd.ops.attention import _sdpa
 
             o = _sdpa(q, k, v, causal=False, scale=None)

@@ -426,3 +426,31 @@ def test_hf_logits_parity_olmoe():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_mistral_sliding_window():
+    """Mistral with a real sliding window (the flagship flash path is
+    unaffected: window None keeps the fused kernel)."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(38)
+    hf_cfg = transformers.MistralConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        sliding_window=8, attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.MistralForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["MistralForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.sliding_window == 8
+    mine.load_state_dict(hf.state_dict(), strict=False)
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+    # flagship configs (no window) keep the flash path
+    assert LlamaForCausalLM(dict(vocab_size=100, hidden_size=32,
+                                 intermediate_size=64, num_hidden_layers=1,
+                                 num_attention_heads=2, num_key_value_heads=1,
+                                 max_position_embeddings=32)) \
+        .config.sliding_window is None
