@@ -65,6 +65,9 @@ class DeepseekV3Config:
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
         g = hf.get
+        archs = " ".join(g("architectures", []) or [])
+        # V2: softmax scoring, no aux-free bias; V3: sigmoid + bias
+        v2 = "DeepseekV2" in archs
         return cls(
             vocab_size=g("vocab_size", 129280),
             hidden_size=g("hidden_size", 7168),
@@ -87,9 +90,11 @@ class DeepseekV3Config:
                 n_activated_experts=g("num_experts_per_tok", 8),
                 n_expert_groups=g("n_group", 1),
                 n_limited_groups=g("topk_group", 1),
-                score_func=g("scoring_func", "sigmoid"),
+                score_func=g("scoring_func", "softmax" if v2 else "sigmoid"),
+                # HF V2 router never renormalizes the top-k weights
+                norm_topk_prob=False if v2 else g("norm_topk_prob", True),
                 route_scale=g("routed_scaling_factor", 1.0),
-                expert_bias=True,
+                expert_bias=not v2,
                 moe_intermediate_size=g("moe_intermediate_size", 2048),
                 shared_expert_intermediate_size=(
                     g("moe_intermediate_size", 2048) * g("n_shared_experts", 1)

@@ -454,3 +454,33 @@ def test_hf_logits_parity_mistral_sliding_window():
                                  num_attention_heads=2, num_key_value_heads=1,
                                  max_position_embeddings=32)) \
         .config.sliding_window is None
+
+
+def test_hf_logits_parity_deepseek_v2():
+    """DeepSeek-V2 through the V3 class: softmax routing (no bias, no topk
+    renorm in the HF V2 router) with the same MLA attention."""
+    from automodel_amd.models.deepseek_v3.model import DeepseekV3ForCausalLM
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(40)
+    hf_cfg = transformers.DeepseekV2Config(
+        vocab_size=300, hidden_size=64, intermediate_size=96,
+        moe_intermediate_size=48, num_hidden_layers=3, num_attention_heads=4,
+        num_key_value_heads=4, n_routed_experts=8, n_shared_experts=1,
+        num_experts_per_tok=2, first_k_dense_replace=1, q_lora_rank=32,
+        kv_lora_rank=16, qk_nope_head_dim=16, qk_rope_head_dim=8,
+        v_head_dim=16, n_group=1, topk_group=1, norm_topk_prob=True,
+        routed_scaling_factor=1.0, max_position_embeddings=128,
+        rope_theta=10000.0, rms_norm_eps=1e-6, attn_implementation="eager",
+        tie_word_embeddings=False)
+    hf = transformers.DeepseekV2ForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["DeepseekV2ForCausalLM"]
+    mine = DeepseekV3ForCausalLM(DeepseekV3ForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.moe.score_func == "softmax"
+    sd = MoEStateDictAdapter(mine.config).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
