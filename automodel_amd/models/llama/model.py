@@ -86,7 +86,9 @@ class LlamaConfig:
             sliding_window=(get("sliding_window")
                             if get("use_sliding_window", True) else None),
             tie_word_embeddings=get("tie_word_embeddings", False),
-            attention_bias=get("attention_bias", get("qkv_bias", False)),
+            # HF Qwen2 hardcodes qkv bias=True regardless of config fields
+            attention_bias=("Qwen2ForCausalLM" in archs or "Qwen2MoeForCausalLM" in archs
+                            or get("attention_bias", get("qkv_bias", False))),
             mlp_bias=get("mlp_bias", False),
             initializer_range=get("initializer_range", 0.02),
         )

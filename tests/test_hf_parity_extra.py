@@ -484,3 +484,30 @@ def test_hf_logits_parity_deepseek_v2():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_qwen2_biases():
+    """Qwen2: HF hardcodes qkv biases (True) regardless of config — biases
+    must load and contribute (randomized here so the test is meaningful)."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(41)
+    hf_cfg = transformers.Qwen2Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.Qwen2ForCausalLM(hf_cfg).eval()
+    with torch.no_grad():
+        for n, p in hf.named_parameters():
+            if n.endswith("bias"):
+                p.normal_(0, 0.1)
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Qwen2ForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.attention_bias
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
