@@ -104,6 +104,10 @@ def main():
     model.train()
     opt = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.0)
 
+    # proxy configs have short contexts; clamp so --model tiny_proxy works
+    # without an explicit --seq-len (driver default path is unaffected)
+    args.seq_len = min(args.seq_len, cfg.max_position_embeddings)
+
     # synthetic data (BASELINE measurement conditions: mock data)
     g = torch.Generator(device="cpu").manual_seed(5678 + rank)
     batches = []
