@@ -322,7 +322,7 @@ class Gemma3ForCausalLM(GemmaForCausalLM):
     """Gemma-3 text model: Gemma-2 base + qk-norm, explicit layer_types,
     dual-frequency rope (local theta on sliding layers), no softcapping."""
 
-    hf_architectures = ("Gemma3ForCausalLM", "Gemma3ForConditionalGeneration")
+    hf_architectures = ("Gemma3ForCausalLM",)  # the ConditionalGeneration VLM variant is round-2
 
     @staticmethod
     def config_from_hf(hf_cfg) -> GemmaConfig:

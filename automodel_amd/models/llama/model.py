@@ -48,6 +48,7 @@ class LlamaConfig:
                                    # reference models/llama_bidirectional/)
     no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
     sliding_window: int | None = None   # Mistral-style windowed causal attention
+    rope_interleaved: bool = False      # GPT-NeoX/Ernie pair-interleaved rope
     fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
     fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
     initializer_range: float = 0.02
@@ -63,6 +64,12 @@ class LlamaConfig:
             hf = hf.to_dict()
         get = hf.get
         archs = " ".join(get("architectures", []) or [])
+        # newer transformers store theta/scaling under rope_parameters
+        rp = get("rope_parameters") or {}
+        theta = rp.get("rope_theta", get("rope_theta", 10000.0))
+        scaling = get("rope_scaling")
+        if scaling is None and rp.get("rope_type", "default") != "default":
+            scaling = rp
         return cls(
             qk_norm="Qwen3" in archs,
             # Phi-3 ships fused qkv_proj / gate_up_proj weights — exactly this
@@ -78,13 +85,14 @@ class LlamaConfig:
             head_dim=get("head_dim", None),
             max_position_embeddings=get("max_position_embeddings", 8192),
             rms_norm_eps=get("rms_norm_eps", 1e-5),
-            rope_theta=get("rope_theta", 10000.0),
-            rope_scaling=get("rope_scaling", None),
+            rope_theta=theta,
+            rope_scaling=scaling,
             no_rope_layers=get("no_rope_layers", None),
             # mistral: sliding_window set => windowed; qwen2 gates it behind
             # use_sliding_window
             sliding_window=(get("sliding_window")
                             if get("use_sliding_window", True) else None),
+            rope_interleaved="Ernie" in archs,
             tie_word_embeddings=get("tie_word_embeddings", False),
             # HF Qwen2 hardcodes qkv bias=True regardless of config fields
             attention_bias=("Qwen2ForCausalLM" in archs or "Qwen2MoeForCausalLM" in archs
@@ -144,6 +152,14 @@ class LlamaAttention(nn.Module):
             q = self.q_norm(q)
             k = self.k_norm(k)
         if self.use_rope:
+            if self.cfg.rope_interleaved:
+                # interleaved convention: de-interleave to half-split order
+                # before the standard kernel. Attention scores are invariant
+                # to a head-dim permutation applied to BOTH q and k, so no
+                # re-interleave is needed (v is untouched).
+                d2 = self.head_dim // 2
+                q = q.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
+                k = k.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
             q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
         if self.cfg.sliding_window is not None:
             # windowed causal (mistral): explicit mask via sdpa — the flash
@@ -262,7 +278,8 @@ class LlamaModel(nn.Module):
 
 class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
-                        "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM")
+                        "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
+                        "Ernie4_5ForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

@@ -511,3 +511,27 @@ def test_hf_logits_parity_qwen2_biases():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_ernie4_5():
+    """Ernie-4.5: interleaved rope convention (handled by the permutation-
+    invariance trick) + rope_parameters theta source + decoupled head_dim."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(42)
+    hf_cfg = transformers.Ernie4_5Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False)
+    hf = transformers.Ernie4_5ForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Ernie4_5ForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.rope_interleaved
+    assert mine.config.rope_theta == 500000.0   # from rope_parameters
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
