@@ -137,3 +137,43 @@ def apply_lora_to_linear_modules(model: nn.Module, cfg) -> int:
 def lora_state_dict(model: nn.Module) -> dict[str, torch.Tensor]:
     """Adapter-only state dict (reference checkpoint/addons.py PEFT saves)."""
     return {k: v for k, v in model.state_dict().items() if "lora_" in k}
+
+
+def export_hf_peft_adapter(model: nn.Module, out_dir: str,
+                           base_model_name: str = "") -> None:
+    """Write the adapters in HF-PEFT layout (adapter_config.json +
+    adapter_model.safetensors with `base_model.model.` key prefixes and
+    .weight suffixes) so `peft.PeftModel.from_pretrained` can load them
+    (reference checkpoint/addons.py PEFT saves are HF-PEFT compatible)."""
+    import json
+    import os
+
+    from safetensors.torch import save_file
+
+    os.makedirs(out_dir, exist_ok=True)
+    sd = {}
+    target_modules: set[str] = set()
+    rank = alpha = None
+    for name, module in model.named_modules():
+        if not isinstance(module, LinearLoRA):
+            continue
+        rank, alpha = module.dim, module.scale * module.dim
+        target_modules.add(name.rsplit(".", 1)[-1])
+        sd[f"base_model.model.{name}.lora_A.weight"] = \
+            module.lora_A.weight.detach().contiguous()
+        sd[f"base_model.model.{name}.lora_B.weight"] = \
+            module.lora_B.weight.detach().contiguous()
+    save_file(sd, os.path.join(out_dir, "adapter_model.safetensors"),
+              metadata={"format": "pt"})
+    cfg = {
+        "peft_type": "LORA",
+        "base_model_name_or_path": base_model_name,
+        "r": rank,
+        "lora_alpha": alpha,
+        "lora_dropout": 0.0,
+        "target_modules": sorted(target_modules),
+        "bias": "none",
+        "task_type": "CAUSAL_LM",
+    }
+    with open(os.path.join(out_dir, "adapter_config.json"), "w") as f:
+        json.dump(cfg, f, indent=1)

@@ -215,3 +215,38 @@ def test_dora_adapter():
     assert dora.lora_A.weight.grad is not None
     assert dora.lora_magnitude.grad is not None
     assert base.weight.grad is None
+
+
+def test_export_hf_peft_adapter(tmp_path):
+    """Adapter export in HF-PEFT layout round-trips through the peft lib."""
+    import json
+
+    import torch.nn as nn
+
+    from automodel_amd.peft.lora import (
+        LinearLoRA,
+        apply_lora_to_linear_modules,
+        export_hf_peft_adapter,
+    )
+
+    class Tiny(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.q_proj = nn.Linear(16, 16)
+            self.v_proj = nn.Linear(16, 16)
+
+    m = Tiny()
+    apply_lora_to_linear_modules(m, {"target_modules": ["*q_proj", "*v_proj"],
+                                     "dim": 4, "alpha": 8})
+    with torch.no_grad():
+        nn.init.normal_(m.q_proj.lora_B.weight)
+    export_hf_peft_adapter(m, str(tmp_path), base_model_name="tiny")
+    cfg = json.load(open(tmp_path / "adapter_config.json"))
+    assert cfg["r"] == 4 and cfg["lora_alpha"] == 8
+    assert sorted(cfg["target_modules"]) == ["q_proj", "v_proj"]
+    from safetensors.torch import load_file
+
+    sd = load_file(str(tmp_path / "adapter_model.safetensors"))
+    assert "base_model.model.q_proj.lora_A.weight" in sd
+    torch.testing.assert_close(sd["base_model.model.q_proj.lora_B.weight"],
+                               m.q_proj.lora_B.weight)
