@@ -5,3 +5,4 @@ from automodel_amd.eval.tool_calling import (  # noqa: F401
     parse_tool_calls,
     score_tool_calls,
 )
+from automodel_amd.eval.perplexity import evaluate_nll  # noqa: F401

@@ -88,3 +88,27 @@ def test_capabilities_cli(capsys):
     assert "supports_ep=True" in out and "supports_tp=False" in out
     query_capabilities()  # all architectures
     assert "GptOssForCausalLM" in capsys.readouterr().out
+
+
+def test_evaluate_nll():
+    from automodel_amd.eval.perplexity import evaluate_nll
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(0)
+    m = LlamaForCausalLM(dict(vocab_size=100, hidden_size=32, intermediate_size=64,
+                              num_hidden_layers=2, num_attention_heads=2,
+                              num_key_value_heads=1, max_position_embeddings=64))
+    m.init_weights(device="cpu")
+    batches = [{"input_ids": torch.randint(0, 100, (2, 17))} for _ in range(3)]
+    out = evaluate_nll(m, batches)
+    assert out["n_tokens"] == 3 * 2 * 16
+    assert out["perplexity"] > 1.0
+    # random-init model on uniform-random tokens: nll near ln(vocab)
+    import math
+
+    assert abs(out["nll_per_token"] - math.log(100)) < 1.0
+    # ignore_index masking respected
+    b = {"input_ids": torch.randint(0, 100, (1, 9)),
+         "labels": torch.full((1, 9), -100)}
+    out2 = evaluate_nll(m, [b])
+    assert out2["n_tokens"] == 0
