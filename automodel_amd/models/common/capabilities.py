@@ -51,6 +51,8 @@ def validate_model_against_mesh(model, mesh_dims: dict) -> list[str]:
         getattr(cfg, "fused_qkv", False) or getattr(cfg, "fused_gate_up", False)
     ):
         problems.append("fused qkv/gate_up projections have no TP plan (disable fusion)")
+    if mesh_dims.get("cp", 1) > 1 and getattr(cfg, "sliding_window", None):
+        problems.append("sliding-window attention has no CP mechanism (disable cp or the window)")
     heads = getattr(getattr(model, "config", None), "num_attention_heads", None)
     tp = mesh_dims.get("tp", 1)
     if heads and tp > 1 and heads % tp != 0:

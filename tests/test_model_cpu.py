@@ -218,3 +218,15 @@ def test_bidirectional_llama_attends_to_future():
         hb_b = bidir(b, return_hidden=True)
     torch.testing.assert_close(hc_a[:, 0], hc_b[:, 0])      # causal: unchanged
     assert not torch.allclose(hb_a[:, 0], hb_b[:, 0])       # bidir: changed
+
+
+def test_capabilities_sliding_window_cp_guard():
+    from automodel_amd.models.common.capabilities import validate_model_against_mesh
+
+    m = LlamaForCausalLM(dict(vocab_size=100, hidden_size=32, intermediate_size=64,
+                              num_hidden_layers=1, num_attention_heads=2,
+                              num_key_value_heads=1, max_position_embeddings=64,
+                              sliding_window=16))
+    probs = validate_model_against_mesh(m, {"cp": 2})
+    assert any("sliding-window" in p for p in probs)
+    assert not validate_model_against_mesh(m, {"cp": 1})
