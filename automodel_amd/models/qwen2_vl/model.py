@@ -45,6 +45,11 @@ class VisionConfig:
     spatial_merge_size: int = 2
     in_channels: int = 3
     hidden_size: int = 3584          # text width (merger output)
+    # ---- qwen2.5-vl variant ----
+    variant: str = "v2"              # "v2" (LN + quick-gelu) | "v2_5" (RMS + swiglu + windows)
+    intermediate_size: int | None = None     # v2_5 explicit mlp width
+    window_size: int = 112
+    fullatt_block_indexes: tuple = (7, 15, 23, 31)
 
 
 @dataclass
@@ -151,20 +156,45 @@ class VisionAttention(nn.Module):
         return self.proj(torch.cat(outs, dim=0).reshape(N, -1))
 
 
+class _VisionRMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float = 1e-6):
+        super().__init__()
+        self.weight = nn.Parameter(torch.ones(dim))
+        self.eps = eps
+
+    def forward(self, x):
+        xf = x.float()
+        inv = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (xf * inv).to(x.dtype) * self.weight
+
+
 class VisionBlock(nn.Module):
     def __init__(self, cfg: VisionConfig):
         super().__init__()
-        self.norm1 = nn.LayerNorm(cfg.embed_dim, eps=1e-6)
-        self.norm2 = nn.LayerNorm(cfg.embed_dim, eps=1e-6)
+        self.v25 = cfg.variant == "v2_5"
+        norm_cls = _VisionRMSNorm if self.v25 else (lambda d: nn.LayerNorm(d, eps=1e-6))
+        self.norm1 = norm_cls(cfg.embed_dim)
+        self.norm2 = norm_cls(cfg.embed_dim)
         self.attn = VisionAttention(cfg)
-        hidden = int(cfg.embed_dim * cfg.mlp_ratio)
-        self.mlp = nn.Sequential()
-        self.mlp.fc1 = nn.Linear(cfg.embed_dim, hidden)
-        self.mlp.fc2 = nn.Linear(hidden, cfg.embed_dim)
+        if self.v25:   # swiglu with biases
+            hidden = cfg.intermediate_size or int(cfg.embed_dim * cfg.mlp_ratio)
+            self.mlp = nn.Module()
+            self.mlp.gate_proj = nn.Linear(cfg.embed_dim, hidden, bias=True)
+            self.mlp.up_proj = nn.Linear(cfg.embed_dim, hidden, bias=True)
+            self.mlp.down_proj = nn.Linear(hidden, cfg.embed_dim, bias=True)
+        else:
+            hidden = int(cfg.embed_dim * cfg.mlp_ratio)
+            self.mlp = nn.Sequential()
+            self.mlp.fc1 = nn.Linear(cfg.embed_dim, hidden)
+            self.mlp.fc2 = nn.Linear(hidden, cfg.embed_dim)
 
     def forward(self, x, cos, sin, cu):
         x = x + self.attn(self.norm1(x), cos, sin, cu)
-        h = self.mlp.fc1(self.norm2(x))
+        h = self.norm2(x)
+        if self.v25:
+            g = self.mlp.gate_proj(h)
+            return x + self.mlp.down_proj(torch.nn.functional.silu(g) * self.mlp.up_proj(h))
+        h = self.mlp.fc1(h)
         h = h * torch.sigmoid(1.702 * h)          # quick-gelu
         return x + self.mlp.fc2(h)
 
@@ -174,7 +204,8 @@ class PatchMerger(nn.Module):
         super().__init__()
         self.merge = cfg.spatial_merge_size
         dim = cfg.embed_dim * self.merge ** 2
-        self.ln_q = nn.LayerNorm(cfg.embed_dim, eps=1e-6)
+        self.ln_q = (_VisionRMSNorm(cfg.embed_dim) if cfg.variant == "v2_5"
+                     else nn.LayerNorm(cfg.embed_dim, eps=1e-6))
         self.mlp = nn.Sequential(nn.Linear(dim, dim), nn.GELU(),
                                  nn.Linear(dim, cfg.hidden_size))
 
@@ -214,9 +245,52 @@ class VisionTransformer(nn.Module):
         cos, sin = emb.cos(), emb.sin()
         cu = torch.cat([torch.zeros(1, dtype=torch.long),
                         grid_thw.prod(-1).cumsum(0)]).to(x.device)
+        if c.variant == "v2_5":
+            # window reorder (qwen2.5): merge-unit groups permuted so each
+            # attention window is contiguous; inverted after the merger
+            widx, cu_win = vision_window_index(grid_thw, c.spatial_merge_size,
+                                               c.window_size, c.patch_size)
+            widx = widx.to(x.device)
+            mu = c.spatial_merge_size ** 2
+            N = x.shape[0]
+            x = x.reshape(N // mu, mu, -1)[widx].reshape(N, -1)
+            cos = cos.reshape(N // mu, mu, -1)[widx].reshape(N, -1)
+            sin = sin.reshape(N // mu, mu, -1)[widx].reshape(N, -1)
+            cu_win = cu_win.to(x.device)
+            for i, blk in enumerate(self.blocks):
+                use_full = i in c.fullatt_block_indexes
+                x = blk(x, cos, sin, cu if use_full else cu_win)
+            out = self.merger(x)
+            return out[torch.argsort(widx)]
         for blk in self.blocks:
             x = blk(x, cos, sin, cu)
         return self.merger(x)
+
+
+def vision_window_index(grid_thw: torch.Tensor, merge: int, window: int,
+                        patch: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Window partition order over merge-unit groups (qwen2.5-vl): -> (index
+    [N/merge^2], cu_window_seqlens in PATCH tokens)."""
+    vw = window // merge // patch
+    mu = merge ** 2
+    idx_parts, cu = [], [0]
+    base = 0
+    for t, h, w in grid_thw.tolist():
+        gh, gw = h // merge, w // merge
+        index = torch.arange(t * gh * gw).reshape(t, gh, gw)
+        pad_h, pad_w = (-gh) % vw, (-gw) % vw
+        nh, nw = (gh + pad_h) // vw, (gw + pad_w) // vw
+        padded = torch.nn.functional.pad(index, (0, pad_w, 0, pad_h),
+                                         value=-100)
+        padded = padded.reshape(t, nh, vw, nw, vw).permute(0, 1, 3, 2, 4) \
+            .reshape(t, nh * nw, vw, vw)
+        seqlens = (padded != -100).sum([2, 3]).reshape(-1)
+        flat = padded.reshape(-1)
+        idx_parts.append(flat[flat != -100] + base)
+        cu.extend((seqlens.cumsum(0) * mu + cu[-1]).tolist())
+        base += t * gh * gw
+    cu_t = torch.unique_consecutive(torch.tensor(cu, dtype=torch.long))
+    return torch.cat(idx_parts), cu_t
 
 
 # ----------------------------------------------------------------- text model
@@ -352,3 +426,40 @@ class Qwen2VLForConditionalGeneration(nn.Module):
                     nn.init.zeros_(mod.bias)
         if self.config.text.tie_word_embeddings:
             self.lm_head.weight = self.model.language_model.embed_tokens.weight
+
+
+class Qwen2_5_VLForConditionalGeneration(Qwen2VLForConditionalGeneration):
+    """Qwen2.5-VL: the same m-rope text decoder with the v2_5 vision tower
+    (RMSNorm, biased SwiGLU MLP, windowed attention with token reorder)."""
+
+    hf_architectures = ("Qwen2_5_VLForConditionalGeneration",)
+
+    @staticmethod
+    def config_from_hf(hf_cfg) -> Qwen2VLConfig:
+        if hasattr(hf_cfg, "to_dict"):
+            hf_cfg = hf_cfg.to_dict()
+        tc, vc = hf_cfg.get("text_config", hf_cfg), hf_cfg.get("vision_config", {})
+        text = LlamaConfig.from_hf_config(dict(tc, architectures=["Qwen2ForCausalLM"]))
+        text.attention_bias = True
+        rs = tc.get("rope_scaling") or tc.get("rope_parameters") or {}
+        return Qwen2VLConfig(
+            text=text,
+            vision=VisionConfig(
+                variant="v2_5",
+                embed_dim=vc.get("hidden_size", 1280),     # 2.5 names it hidden_size
+                depth=vc.get("depth", 32),
+                num_heads=vc.get("num_heads", 16),
+                intermediate_size=vc.get("intermediate_size"),
+                patch_size=vc.get("patch_size", 14),
+                temporal_patch_size=vc.get("temporal_patch_size", 2),
+                spatial_merge_size=vc.get("spatial_merge_size", 2),
+                in_channels=vc.get("in_channels", 3),
+                hidden_size=vc.get("out_hidden_size", tc.get("hidden_size", 3584)),
+                window_size=vc.get("window_size", 112),
+                fullatt_block_indexes=tuple(vc.get("fullatt_block_indexes",
+                                                   (7, 15, 23, 31))),
+            ),
+            mrope_section=tuple(rs.get("mrope_section", (16, 24, 24))),
+            image_token_id=hf_cfg.get("image_token_id", 151655),
+            vision_start_token_id=hf_cfg.get("vision_start_token_id", 151652),
+        )

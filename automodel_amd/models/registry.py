@@ -86,10 +86,15 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
-        from automodel_amd.models.qwen2_vl.model import Qwen2VLForConditionalGeneration
+        from automodel_amd.models.qwen2_vl.model import (
+            Qwen2_5_VLForConditionalGeneration,
+            Qwen2VLForConditionalGeneration,
+        )
 
         for name in Qwen2VLForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen2VLForConditionalGeneration
+        for name in Qwen2_5_VLForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen2_5_VLForConditionalGeneration
     except ImportError:
         pass
     try:

@@ -535,3 +535,43 @@ def test_hf_logits_parity_ernie4_5():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_qwen2_5_vl():
+    """Qwen2.5-VL: RMSNorm ViT, biased-SwiGLU MLP, WINDOWED vision attention
+    with merge-unit token reorder — text and text+image paths."""
+    from automodel_amd.models.qwen2_vl.model import Qwen2_5_VLForConditionalGeneration
+
+    torch.manual_seed(50)
+    hf_cfg = transformers.Qwen2_5_VLConfig(
+        text_config=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         num_key_value_heads=2, max_position_embeddings=256,
+                         rope_theta=10000.0, rms_norm_eps=1e-6,
+                         rope_scaling={"type": "mrope", "mrope_section": [2, 3, 3]},
+                         tie_word_embeddings=False),
+        vision_config=dict(hidden_size=32, depth=2, num_heads=2,
+                           out_hidden_size=64, patch_size=4,
+                           temporal_patch_size=2, spatial_merge_size=2,
+                           in_channels=3, intermediate_size=64,
+                           window_size=16, fullatt_block_indexes=[1]),
+        image_token_id=299, vision_start_token_id=298)
+    hf = transformers.Qwen2_5_VLForConditionalGeneration(hf_cfg).eval()
+    mine = Qwen2_5_VLForConditionalGeneration(
+        Qwen2_5_VLForConditionalGeneration.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+
+    ids = torch.randint(0, 290, (2, 20))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=2e-4, rtol=2e-4)
+    grid = torch.tensor([[1, 8, 8]])
+    pix = torch.randn(64, 3 * 2 * 4 * 4)
+    seq = torch.cat([torch.randint(0, 290, (1, 3)), torch.tensor([[298]]),
+                     torch.full((1, 16), 299), torch.randint(0, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix, image_grid_thw=grid,
+                 mm_token_type_ids=(seq == 299).int()).logits
+        out = mine(seq, pixel_values=pix, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
