@@ -98,6 +98,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.llava.model import LlavaForConditionalGeneration
+
+        for name in LlavaForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = LlavaForConditionalGeneration
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.vlm.model import VLMForConditionalGeneration
 
         for name in VLMForConditionalGeneration.hf_architectures:

@@ -575,3 +575,41 @@ def test_hf_logits_parity_qwen2_5_vl():
                  mm_token_type_ids=(seq == 299).int()).logits
         out = mine(seq, pixel_values=pix, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_llava():
+    """LLaVA: CLIP tower (class token, learned positions, pre/post LN,
+    quick-GELU), feature layer -2 with class-token drop, GELU projector,
+    image splice — text and text+image paths."""
+    from automodel_amd.models.llava.model import LlavaForConditionalGeneration
+
+    torch.manual_seed(55)
+    hf_cfg = transformers.LlavaConfig(
+        text_config=dict(model_type="llama", vocab_size=300, hidden_size=64,
+                         intermediate_size=128, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         max_position_embeddings=128, rms_norm_eps=1e-6,
+                         rope_theta=10000.0, tie_word_embeddings=False),
+        vision_config=dict(model_type="clip_vision_model", hidden_size=32,
+                           intermediate_size=64, num_hidden_layers=2,
+                           num_attention_heads=2, image_size=16, patch_size=4,
+                           projection_dim=32),
+        image_token_id=299)
+    hf = transformers.LlavaForConditionalGeneration(hf_cfg).eval()
+    mine = LlavaForConditionalGeneration(
+        LlavaForConditionalGeneration.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert mine.config.image_token_id == 299
+
+    ids = torch.randint(0, 290, (2, 12))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=2e-4, rtol=2e-4)
+    pix = torch.randn(1, 3, 16, 16)
+    seq = torch.cat([torch.randint(0, 290, (1, 3)), torch.full((1, 16), 299),
+                     torch.randint(0, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix).logits
+        out = mine(seq, pixel_values=pix)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
