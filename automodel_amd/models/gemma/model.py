@@ -145,7 +145,8 @@ class GemmaAttention(nn.Module):
         self.qk_norm = cfg.qk_norm
         self.backend = backend
 
-    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                block_ids: torch.Tensor | None = None) -> torch.Tensor:
         B, S, _ = x.shape
         q = self.q_proj(x).view(B, S, -1, self.head_dim)
         k = self.k_proj(x).view(B, S, -1, self.head_dim)
@@ -154,14 +155,14 @@ class GemmaAttention(nn.Module):
             q = self.q_norm(q)
             k = self.k_norm(k)
         q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
-        if self.softcap is None and self.window is None:
+        if self.softcap is None and self.window is None and block_ids is None:
             o = flash_attention(q, k, v, causal=True, scale=self.scale,
                                 backend=self.backend.attn)
         else:
-            o = self._eager_capped(q, k, v)
+            o = self._eager_capped(q, k, v, block_ids)
         return self.o_proj(o.reshape(B, S, -1))
 
-    def _eager_capped(self, q, k, v):
+    def _eager_capped(self, q, k, v, block_ids=None):
         """Softcap/sliding-window need the scores; bandwidth-bound eager path
         (these models route here only for short eval shapes; long-context
         gemma training uses the global layers' flash path)."""
@@ -174,10 +175,19 @@ class GemmaAttention(nn.Module):
         scores = torch.einsum("bhqd,bhkd->bhqk", qt.float(), kt.float()) * self.scale
         scores = _softcap(scores, self.softcap)
         i = torch.arange(S, device=q.device)
-        mask = i[None, :] > i[:, None]                    # future
+        allowed = i[None, :] <= i[:, None]                # causal
+        if block_ids is not None:
+            # gemma-3 multimodal: image tokens attend bidirectionally
+            # within their own image block
+            same = (block_ids[:, :, None] == block_ids[:, None, :]) \
+                & (block_ids[:, :, None] >= 0)
+            allowed = allowed[None] | same
+        else:
+            allowed = allowed[None].expand(1, S, S)
         if self.window is not None:
-            mask |= i[None, :] <= (i[:, None] - self.window)
-        scores = scores.masked_fill(mask, float("-inf"))
+            within = (i[None, :] - i[:, None]).abs() < self.window
+            allowed = allowed & within[None]
+        scores = scores.masked_fill(~allowed.unsqueeze(1), float("-inf"))
         o = torch.einsum("bhqk,bhkd->bhqd", scores.softmax(-1), vt.float())
         return o.transpose(1, 2).to(q.dtype)
 
@@ -206,13 +216,15 @@ class GemmaDecoderLayer(nn.Module):
             self.pre_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
             self.post_feedforward_layernorm = GemmaRMSNorm(cfg.hidden_size, cfg.rms_norm_eps, nb)
 
-    def forward(self, x, cos, sin, cos_local=None, sin_local=None):
+    def forward(self, x, cos, sin, cos_local=None, sin_local=None,
+                block_ids=None):
         if cos_local is not None and self.self_attn.is_sliding:
             cos, sin = cos_local, sin_local   # gemma-3 dual-frequency rope
         if not self.post_norms:   # gemma-1: classic pre-norm residual order
-            x = x + self.self_attn(self.input_layernorm(x), cos, sin)
+            x = x + self.self_attn(self.input_layernorm(x), cos, sin, block_ids)
             return x + self.mlp(self.post_attention_layernorm(x))
-        x = x + self.post_attention_layernorm(self.self_attn(self.input_layernorm(x), cos, sin))
+        x = x + self.post_attention_layernorm(
+            self.self_attn(self.input_layernorm(x), cos, sin, block_ids))
         x = x + self.post_feedforward_layernorm(self.mlp(self.pre_feedforward_layernorm(x)))
         return x
 
@@ -263,9 +275,12 @@ class GemmaForCausalLM(nn.Module):
         self.loss_fn = None
 
     def forward(self, input_ids: torch.Tensor, labels: torch.Tensor | None = None,
-                position_ids: torch.Tensor | None = None, **_: Any) -> torch.Tensor:
+                position_ids: torch.Tensor | None = None,
+                inputs_embeds: torch.Tensor | None = None,
+                block_ids: torch.Tensor | None = None, **_: Any) -> torch.Tensor:
         m = self.model
-        x = m.embed_tokens(input_ids) * (self.config.hidden_size ** 0.5)
+        base = inputs_embeds if inputs_embeds is not None else m.embed_tokens(input_ids)
+        x = base * (self.config.hidden_size ** 0.5)
         S = input_ids.shape[1]
         if position_ids is None:
             cos, sin = m.rope_cos[:S], m.rope_sin[:S]
@@ -283,7 +298,7 @@ class GemmaForCausalLM(nn.Module):
             if cos_l.dtype != torch.float32:
                 cos_l, sin_l = cos_l.float(), sin_l.float()
         for layer in m.layers:
-            x = layer(x, cos, sin, cos_l, sin_l)
+            x = layer(x, cos, sin, cos_l, sin_l, block_ids)
         hidden = m.norm(x)
         if labels is not None:
             assert self.loss_fn is not None, "set model.loss_fn before passing labels"

@@ -57,11 +57,14 @@ def _ensure_builtin() -> None:
         pass
     try:
         from automodel_amd.models.gemma.model import Gemma3ForCausalLM, GemmaForCausalLM
+        from automodel_amd.models.gemma.vlm import Gemma3ForConditionalGeneration
 
         for name in GemmaForCausalLM.hf_architectures:
             _REGISTRY[name] = GemmaForCausalLM
         for name in Gemma3ForCausalLM.hf_architectures:
             _REGISTRY[name] = Gemma3ForCausalLM
+        for name in Gemma3ForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Gemma3ForConditionalGeneration
     except ImportError:
         pass
     try:

@@ -148,3 +148,50 @@ def test_gemma1_hf_logits_parity():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_gemma3_vlm_hf_logits_parity():
+    """Gemma-3 multimodal: SigLIP tower, avg-pool soft-token projector,
+    image-block bidirectional attention — text and text+image paths."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.gemma.vlm import Gemma3ForConditionalGeneration
+
+    torch.manual_seed(60)
+    hf_cfg = transformers.Gemma3Config(
+        text_config=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                         num_hidden_layers=3, num_attention_heads=4,
+                         num_key_value_heads=2, head_dim=16,
+                         max_position_embeddings=128, sliding_window=8,
+                         query_pre_attn_scalar=16, rms_norm_eps=1e-6,
+                         tie_word_embeddings=False),
+        vision_config=dict(hidden_size=32, intermediate_size=64,
+                           num_hidden_layers=2, num_attention_heads=2,
+                           image_size=16, patch_size=4),
+        mm_tokens_per_image=4, image_token_id=299)
+    hf = transformers.Gemma3ForConditionalGeneration(hf_cfg).eval()
+    mine = Gemma3ForConditionalGeneration(
+        Gemma3ForConditionalGeneration.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+
+    ids = torch.randint(0, 290, (2, 14))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=2e-4, rtol=2e-4)
+    pix = torch.randn(1, 3, 16, 16)
+    seq = torch.cat([torch.randint(0, 290, (1, 3)), torch.full((1, 4), 299),
+                     torch.randint(0, 290, (1, 5))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix,
+                 token_type_ids=(seq == 299).int()).logits
+        out = mine(seq, pixel_values=pix)
+    torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+    # image tokens see each other bidirectionally (HF zero-inits the
+    # projector, so randomize it first to make pixels matter)
+    with torch.no_grad():
+        torch.nn.init.normal_(
+            mine.model.multi_modal_projector.mm_input_projection_weight, std=0.1)
+        base = mine(seq, pixel_values=pix)
+        out2 = mine(seq, pixel_values=pix + 10.0)
+    assert not torch.allclose(base[0, 3], out2[0, 3])
