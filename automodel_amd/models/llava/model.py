@@ -212,6 +212,10 @@ class LlavaForConditionalGeneration(nn.Module):
             self.lm_head.weight = lm.embed_tokens.weight
         self.loss_fn = None
 
+    def freeze_vision_tower(self) -> None:
+        for p in self.model.vision_tower.parameters():
+            p.requires_grad_(False)
+
     def image_features(self, pixel_values: torch.Tensor) -> torch.Tensor:
         hiddens = self.model.vision_tower(pixel_values)
         feats = hiddens[self.config.vision_feature_layer]

@@ -73,3 +73,40 @@ def test_vlm_recipe_end_to_end(tmp_path):
     r.setup()
     r.run_train_validation_loop()
     assert r.step_scheduler.step == 2
+
+
+def test_llava_recipe_end_to_end(tmp_path):
+    """The VLM recipe drives the real LLaVA family (CLIP tower + splice)
+    with mock pixel data; frozen tower stays frozen through a step."""
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.vlm.finetune import FinetuneRecipeForVLM
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"architecture": "LlavaForConditionalGeneration",
+                  "config": {
+                      "text": {"vocab_size": 320, "hidden_size": 32,
+                               "intermediate_size": 64, "num_hidden_layers": 2,
+                               "num_attention_heads": 2, "num_key_value_heads": 1,
+                               "max_position_embeddings": 64},
+                      "vision": {"hidden_size": 16, "intermediate_size": 32,
+                                 "num_hidden_layers": 1, "num_attention_heads": 2,
+                                 "image_size": 16, "patch_size": 4},
+                      "image_token_id": 300,
+                  }, "dtype": "float32"},
+        "freeze": {"vision_tower": True},
+        "loss_fn": {"backend": "chunked", "chunk_size": 64},
+        "optimizer": {"lr": 1e-3},
+        "step_scheduler": {"max_steps": 2},
+        "dataloader": {"dataset": {"kind": "mock_vlm", "num_samples": 4,
+                                   "seq_len": 32, "vocab_size": 320,
+                                   "image_size": 16, "patch_size": 4,
+                                   "image_token_id": 300},
+                       "batch_size": 2},
+        "output_dir": str(tmp_path),
+    })
+    r = FinetuneRecipeForVLM(cfg)
+    r.setup()
+    assert all(not p.requires_grad
+               for p in r.model.model.vision_tower.parameters())
+    r.run_train_validation_loop()   # two steps complete without error
