@@ -20,7 +20,7 @@ path). State-dict keys match HF GptOssForCausalLM exactly.
 
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any
 
 import torch
@@ -28,7 +28,6 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from automodel_amd.models.common.backend import BackendConfig
-from automodel_amd.ops.attention import sdpa_masked
 from automodel_amd.ops.rms_norm import RMSNorm
 from automodel_amd.ops.rope import apply_rope, build_rope_cache
 
