@@ -101,6 +101,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.mistral3.model import Mistral3ForConditionalGeneration
+
+        for name in Mistral3ForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Mistral3ForConditionalGeneration
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.llava.model import LlavaForConditionalGeneration
 
         for name in LlavaForConditionalGeneration.hf_architectures:

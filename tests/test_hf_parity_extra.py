@@ -613,3 +613,43 @@ def test_hf_logits_parity_llava():
         ref = hf(input_ids=seq, pixel_values=pix).logits
         out = mine(seq, pixel_values=pix)
     torch.testing.assert_close(out, ref, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_mistral3():
+    """Mistral-3 VLM: Pixtral tower (RMSNorm blocks, 2-D interleaved-freq
+    rotary, bias-free swiglu), unfold patch merger, GELU projector."""
+    from automodel_amd.models.mistral3.model import Mistral3ForConditionalGeneration
+
+    torch.manual_seed(70)
+    hf_cfg = transformers.Mistral3Config(
+        text_config=dict(model_type="mistral", vocab_size=300, hidden_size=64,
+                         intermediate_size=128, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         max_position_embeddings=128, rms_norm_eps=1e-6,
+                         rope_theta=10000.0, sliding_window=None,
+                         tie_word_embeddings=False),
+        vision_config=dict(model_type="pixtral", hidden_size=32,
+                           intermediate_size=64, num_hidden_layers=2,
+                           num_attention_heads=2, image_size=16, patch_size=4,
+                           head_dim=16),
+        image_token_index=299, spatial_merge_size=2,
+        multimodal_projector_bias=False)
+    hf = transformers.Mistral3ForConditionalGeneration(hf_cfg).eval()
+    mine = Mistral3ForConditionalGeneration(
+        Mistral3ForConditionalGeneration.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+
+    ids = torch.randint(0, 290, (2, 12))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=2e-4, rtol=2e-4)
+    pix = torch.randn(1, 3, 16, 16)
+    seq = torch.cat([torch.randint(0, 290, (1, 3)), torch.full((1, 4), 299),
+                     torch.randint(0, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix,
+                 image_sizes=torch.tensor([[16, 16]])).logits
+        out = mine(seq, pixel_values=pix)
+    torch.testing.assert_close(out, ref, atol=3e-4, rtol=3e-4)
