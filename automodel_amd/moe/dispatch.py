@@ -49,7 +49,9 @@ class _AllReduceSum(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, group):
         ctx.group = group
-        x = x.contiguous()
+        # clone: all_reduce is in-place and the input must stay untouched
+        # for autograd correctness if the caller reuses it
+        x = x.contiguous().clone()
         dist.all_reduce(x, group=group)
         return x
 
