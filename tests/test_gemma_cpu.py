@@ -195,3 +195,28 @@ def test_gemma3_vlm_hf_logits_parity():
         base = mine(seq, pixel_values=pix)
         out2 = mine(seq, pixel_values=pix + 10.0)
     assert not torch.allclose(base[0, 3], out2[0, 3])
+
+
+def test_gemma_recipe_end_to_end(tmp_path):
+    """train_ft drives the gemma family (softcapped CE path) two steps."""
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.llm.train_ft import (
+        TrainFinetuneRecipeForNextTokenPrediction,
+    )
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"architecture": "Gemma2ForCausalLM",
+                  "config": dict(TINY, vocab_size=256),
+                  "dtype": "float32"},
+        "loss_fn": {"backend": "chunked", "chunk_size": 64},
+        "optimizer": {"lr": 1e-3},
+        "step_scheduler": {"max_steps": 2},
+        "dataloader": {"dataset": {"kind": "mock", "num_samples": 4,
+                                   "seq_len": 24, "vocab_size": 256},
+                       "batch_size": 2},
+        "output_dir": str(tmp_path),
+    })
+    r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    r.setup()
+    r.run_train_validation_loop()
