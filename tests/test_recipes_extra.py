@@ -238,3 +238,33 @@ def test_hf_export_roundtrip_gemma(tmp_path):
     x = torch.randint(0, 200, (1, 8))
     with torch.no_grad():
         torch.testing.assert_close(m(x), m2(x))
+
+
+def test_gpt_oss_recipe_end_to_end(tmp_path):
+    """train_ft drives GPT-OSS (sink attention + MoE experts) two steps."""
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.llm.train_ft import (
+        TrainFinetuneRecipeForNextTokenPrediction,
+    )
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"architecture": "GptOssForCausalLM",
+                  "config": dict(vocab_size=256, hidden_size=32,
+                                 intermediate_size=48, num_hidden_layers=2,
+                                 num_attention_heads=2, num_key_value_heads=1,
+                                 head_dim=16, num_local_experts=4,
+                                 num_experts_per_tok=2,
+                                 max_position_embeddings=64, sliding_window=8),
+                  "dtype": "float32"},
+        "loss_fn": {"backend": "chunked", "chunk_size": 64},
+        "optimizer": {"lr": 1e-3},
+        "step_scheduler": {"max_steps": 2},
+        "dataloader": {"dataset": {"kind": "mock", "num_samples": 4,
+                                   "seq_len": 24, "vocab_size": 256},
+                       "batch_size": 2},
+        "output_dir": str(tmp_path),
+    })
+    r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    r.setup()
+    r.run_train_validation_loop()
