@@ -29,6 +29,30 @@ _CAPS = {
                                                      supports_cp=False),
     "LlamaForSequenceClassification": ModelCapabilities(supports_pp=False,
                                                         supports_cp=False),
+    # families without a TP plan / with non-flash attention paths
+    "GemmaForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
+                                          flash_head_dims=()),
+    "Gemma3ForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
+                                           flash_head_dims=()),
+    "Gemma3ForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_cp=False, supports_pp=False,
+        flash_head_dims=()),
+    "GptOssForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
+                                           supports_ep=False, flash_head_dims=()),
+    "NemotronForCausalLM": ModelCapabilities(supports_tp=False,
+                                             supports_cp=False,
+                                             flash_head_dims=()),
+    "Glm4MoeForCausalLM": ModelCapabilities(supports_tp=False,
+                                            supports_cp=False,
+                                            flash_head_dims=()),
+    "Qwen2VLForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_cp=False, supports_pp=False),
+    "Qwen2_5_VLForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_cp=False, supports_pp=False),
+    "LlavaForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_cp=False, supports_pp=False),
+    "Mistral3ForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_cp=False, supports_pp=False),
 }
 
 

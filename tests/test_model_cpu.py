@@ -230,3 +230,15 @@ def test_capabilities_sliding_window_cp_guard():
     probs = validate_model_against_mesh(m, {"cp": 2})
     assert any("sliding-window" in p for p in probs)
     assert not validate_model_against_mesh(m, {"cp": 1})
+
+
+def test_capabilities_new_families_reject_tp():
+    from automodel_amd.models.common.capabilities import validate_model_against_mesh
+    from automodel_amd.models.gemma.model import GemmaForCausalLM
+
+    m = GemmaForCausalLM(dict(vocab_size=100, hidden_size=32, intermediate_size=64,
+                              num_hidden_layers=1, num_attention_heads=2,
+                              num_key_value_heads=1, head_dim=16,
+                              max_position_embeddings=64))
+    probs = validate_model_against_mesh(m, {"tp": 2})
+    assert any("TP" in p for p in probs)
