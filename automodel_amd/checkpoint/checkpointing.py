@@ -50,6 +50,9 @@ class Checkpointer:
         save is still staging (reference train_ft.py:1251)."""
         if self._async_writer is not None:
             self._async_writer.wait()
+        if getattr(self, "_async_future", None) is not None:
+            self._async_future.result()
+            self._async_future = None
 
     # ---------------------------------------------------------------- save
     def save(self, path: str, model=None, optimizer=None, extra_state: dict | None = None,
@@ -67,7 +70,16 @@ class Checkpointer:
         if optimizer is not None:
             state["optimizer"] = get_optimizer_state_dict(model, optimizer)
         if state:
-            dcp.save(state, checkpoint_id=os.path.join(path, "dcp"))
+            if self.async_save:
+                # staging happens inside async_save before it returns;
+                # the write completes on a background thread. The next
+                # optimizer step calls maybe_wait_for_staging().
+                if getattr(self, "_async_future", None) is not None:
+                    self._async_future.result()
+                self._async_future = dcp.async_save(
+                    state, checkpoint_id=os.path.join(path, "dcp"))
+            else:
+                dcp.save(state, checkpoint_id=os.path.join(path, "dcp"))
         if extra_state and rank == 0:
             torch.save(extra_state, os.path.join(path, "aux_state.pt"))
         if self.save_consolidated and model is not None:
