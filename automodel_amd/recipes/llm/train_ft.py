@@ -206,6 +206,9 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
                 model_save_format=ckpt_cfg.get("model_save_format", "safetensors"),
                 save_consolidated=ckpt_cfg.get("save_consolidated", False),
                 keep_last_n=ckpt_cfg.get("keep_last_n"),
+                keep_top_k=ckpt_cfg.get("keep_top_k"),
+                metric_higher_is_better=ckpt_cfg.get("metric_higher_is_better", False),
+                async_save=ckpt_cfg.get("async_save", False),
             )
 
         # ---- EMA / NEFTune / stepped GC (training extras)
