@@ -22,8 +22,12 @@ from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
 def apply_fsdp(
     model: nn.Module,
     mesh: DeviceMesh,
-    layer_cls_names: tuple[str, ...] = ("LlamaDecoderLayer", "MoEDecoderLayer",
-                                        "VisionBlock"),
+    layer_cls_names: tuple[str, ...] = (
+        "LlamaDecoderLayer", "MoEDecoderLayer", "VisionBlock",
+        "GemmaDecoderLayer", "GptOssDecoderLayer", "NemotronDecoderLayer",
+        "Glm4MoeDecoderLayer", "DeepseekV3DecoderLayer",
+        "SiglipEncoderLayer", "ClipEncoderLayer", "PixtralLayer",
+    ),
     param_dtype: torch.dtype = torch.bfloat16,
     reduce_dtype: torch.dtype = torch.float32,
     reshard_after_forward: bool = False,

@@ -446,3 +446,33 @@ def _tp_qknorm_worker(rank, world):
 
 def test_tp2_qk_norm_forward_parity():
     run_distributed(_tp_qknorm_worker, world=2)
+
+
+def _fsdp_gemma_worker(rank, world):
+    """apply_fsdp shards the NEW families per decoder layer (not one root
+    bucket) and a step runs."""
+    from torch.distributed.tensor import DTensor
+
+    from automodel_amd.models.gemma.model import GemmaForCausalLM
+    from automodel_amd.parallel.fsdp import apply_fsdp
+    from automodel_amd.parallel.mesh import build_mesh
+
+    ctx = build_mesh(dp_shard=world, device_type="cpu")
+    torch.manual_seed(0)
+    m = GemmaForCausalLM(dict(vocab_size=128, hidden_size=32,
+                              intermediate_size=64, num_hidden_layers=2,
+                              num_attention_heads=2, num_key_value_heads=1,
+                              head_dim=16, max_position_embeddings=64,
+                              sliding_window=8, query_pre_attn_scalar=16.0))
+    m.init_weights(device="cpu")
+    apply_fsdp(m, ctx.mesh["dp_shard"], param_dtype=torch.float32)
+    # per-layer sharding happened: decoder layers are FSDP modules themselves
+    from torch.distributed.fsdp import FSDPModule
+
+    assert all(isinstance(l, FSDPModule) for l in m.model.layers)
+    ids = torch.randint(0, 128, (2, 16))
+    m(ids).float().sum().backward()
+
+
+def test_fsdp_shards_new_families():
+    run_distributed(_fsdp_gemma_worker, world=2)
