@@ -22,7 +22,11 @@ from torch.distributed.algorithms._checkpoint.checkpoint_wrapper import (
 def apply_ac(
     model: nn.Module,
     mode: str = "full",                      # full | selective
-    layer_cls_names: tuple[str, ...] = ("LlamaDecoderLayer", "MoEDecoderLayer"),
+    layer_cls_names: tuple[str, ...] = (
+        "LlamaDecoderLayer", "MoEDecoderLayer", "GemmaDecoderLayer",
+        "GptOssDecoderLayer", "NemotronDecoderLayer", "Glm4MoeDecoderLayer",
+        "DeepseekV3DecoderLayer",
+    ),
     selective_cls_names: tuple[str, ...] = ("LlamaAttention", "LlamaMLP", "MoE"),
     every_n: int = 1,
 ) -> nn.Module:
