@@ -30,10 +30,12 @@ _CAPS = {
     "LlamaForSequenceClassification": ModelCapabilities(supports_pp=False,
                                                         supports_cp=False),
     # families without a TP plan / with non-flash attention paths
+    # PP stages replay a llama-shaped forward; gemma needs embed scaling +
+    # dual-frequency rope, so PP is not supported for the family
     "GemmaForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
-                                          flash_head_dims=()),
+                                          supports_pp=False, flash_head_dims=()),
     "Gemma3ForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
-                                           flash_head_dims=()),
+                                           supports_pp=False, flash_head_dims=()),
     "Gemma3ForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_cp=False, supports_pp=False,
         flash_head_dims=()),
