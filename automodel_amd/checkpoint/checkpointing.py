@@ -181,7 +181,11 @@ class Checkpointer:
         d = {}
         for f in dataclasses.fields(cfg):
             v = getattr(cfg, f.name)
-            if isinstance(v, (int, float, str, bool, type(None), list, dict)):
+            if dataclasses.is_dataclass(v):        # nested (moe/text/vision)
+                d[f.name] = dataclasses.asdict(v)
+            elif isinstance(v, tuple):
+                d[f.name] = list(v)
+            elif isinstance(v, (int, float, str, bool, type(None), list, dict)):
                 d[f.name] = v
         d["architectures"] = [type(model).__name__]
         with open(os.path.join(out_dir, "config.json"), "w") as fh:

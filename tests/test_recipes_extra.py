@@ -268,3 +268,26 @@ def test_gpt_oss_recipe_end_to_end(tmp_path):
     r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
     r.setup()
     r.run_train_validation_loop()
+
+
+def test_hf_export_nested_config(tmp_path):
+    """Config export serializes nested dataclass configs (MoE)."""
+    import json
+
+    import torch
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.models.registry import build_model
+
+    m = build_model(config=dict(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64,
+        moe=dict(n_routed_experts=4, n_activated_experts=2,
+                 moe_intermediate_size=48)),
+        architecture="Qwen3MoeForCausalLM", dtype="float32",
+        meta_init=False, device="cpu")
+    out = tmp_path / "hf"
+    Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
+    cj = json.load(open(out / "config.json"))
+    assert cj["moe"]["n_routed_experts"] == 4
