@@ -71,7 +71,13 @@ class Gemma3VLMConfig:
     def from_hf_config(cls, hf: Any) -> "Gemma3VLMConfig":
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
-        tc, vc = hf.get("text_config", {}), hf.get("vision_config", {})
+        if "text" in hf and "vision" in hf:   # our own exported config.json
+            import dataclasses as _dc
+
+            keys = {f.name for f in _dc.fields(cls)}
+            return cls(**{k: v for k, v in hf.items() if k in keys})
+        tc = hf.get("text_config", hf.get("text", {}))
+        vc = hf.get("vision_config", hf.get("vision", {}))
         return cls(
             text=GemmaConfig.from_hf_gemma3(tc),
             vision=SiglipVisionConfig(

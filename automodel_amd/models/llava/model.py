@@ -74,7 +74,13 @@ class LlavaConfig:
     def from_hf_config(cls, hf: Any) -> "LlavaConfig":
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
-        tc, vc = hf.get("text_config", {}), hf.get("vision_config", {})
+        if "text" in hf and "vision" in hf:   # our own exported config.json
+            import dataclasses as _dc
+
+            keys = {f.name for f in _dc.fields(cls)}
+            return cls(**{k: v for k, v in hf.items() if k in keys})
+        tc = hf.get("text_config", hf.get("text", {}))
+        vc = hf.get("vision_config", hf.get("vision", {}))
         return cls(
             text=LlamaConfig.from_hf_config(
                 dict(tc, architectures=[tc.get("model_type", "llama")])),

@@ -73,7 +73,13 @@ class Mistral3Config:
     def from_hf_config(cls, hf: Any) -> "Mistral3Config":
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
-        tc, vc = hf.get("text_config", {}), hf.get("vision_config", {})
+        if "text" in hf and "vision" in hf:   # our own exported config.json
+            import dataclasses as _dc
+
+            keys = {f.name for f in _dc.fields(cls)}
+            return cls(**{k: v for k, v in hf.items() if k in keys})
+        tc = hf.get("text_config", hf.get("text", {}))
+        vc = hf.get("vision_config", hf.get("vision", {}))
         vrp = vc.get("rope_parameters") or {}
         return cls(
             text=LlamaConfig.from_hf_config(

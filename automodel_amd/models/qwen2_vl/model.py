@@ -88,7 +88,13 @@ class Qwen2VLConfig:
     def from_hf_config(cls, hf: Any) -> "Qwen2VLConfig":
         if hasattr(hf, "to_dict"):
             hf = hf.to_dict()
-        tc, vc = hf.get("text_config", hf), hf.get("vision_config", {})
+        if "text" in hf and "vision" in hf:   # our own exported config.json
+            import dataclasses as _dc
+
+            keys = {f.name for f in _dc.fields(cls)}
+            return cls(**{k: v for k, v in hf.items() if k in keys})
+        tc = hf.get("text_config", hf.get("text", hf))
+        vc = hf.get("vision_config", hf.get("vision", {}))
         text = LlamaConfig.from_hf_config(dict(tc, architectures=["Qwen2ForCausalLM"]))
         text.attention_bias = True   # qwen2 qkv bias
         rs = tc.get("rope_scaling") or tc.get("rope_parameters") or {}
@@ -438,6 +444,8 @@ class Qwen2_5_VLForConditionalGeneration(Qwen2VLForConditionalGeneration):
     def config_from_hf(hf_cfg) -> Qwen2VLConfig:
         if hasattr(hf_cfg, "to_dict"):
             hf_cfg = hf_cfg.to_dict()
+        if "text" in hf_cfg and "vision" in hf_cfg:   # our exported dump
+            return Qwen2VLConfig.from_hf_config(hf_cfg)
         tc, vc = hf_cfg.get("text_config", hf_cfg), hf_cfg.get("vision_config", {})
         text = LlamaConfig.from_hf_config(dict(tc, architectures=["Qwen2ForCausalLM"]))
         text.attention_bias = True

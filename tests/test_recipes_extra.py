@@ -291,3 +291,33 @@ def test_hf_export_nested_config(tmp_path):
     Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
     cj = json.load(open(out / "config.json"))
     assert cj["moe"]["n_routed_experts"] == 4
+
+
+def test_vlm_export_reload_roundtrip(tmp_path):
+    """Composite (VLM) export -> build_model(pretrained_path) reconstructs
+    the same architecture and weights."""
+    import torch
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.checkpoint.hf_loader import load_hf_weights
+    from automodel_amd.models.registry import build_model
+
+    cfg = dict(text=dict(vocab_size=310, hidden_size=32, intermediate_size=64,
+                         num_hidden_layers=1, num_attention_heads=2,
+                         num_key_value_heads=1, max_position_embeddings=64,
+                         attention_bias=True),
+               vision=dict(embed_dim=16, depth=1, num_heads=2, hidden_size=32,
+                           patch_size=4, temporal_patch_size=2,
+                           spatial_merge_size=2),
+               mrope_section=(2, 3, 3), image_token_id=309)
+    torch.manual_seed(0)
+    m = build_model(config=cfg, architecture="Qwen2VLForConditionalGeneration",
+                    dtype="float32", meta_init=False, device="cpu")
+    out = tmp_path / "hf"
+    Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
+    m2 = build_model(pretrained_path=str(out), dtype="float32",
+                     meta_init=False, device="cpu")
+    load_hf_weights(m2, str(out), device="cpu")
+    ids = torch.randint(0, 300, (1, 8))
+    with torch.no_grad():
+        torch.testing.assert_close(m(ids), m2(ids))
