@@ -116,6 +116,10 @@ class TrainEagleDraftRecipe:
         self.optimizer.step()
         return {"loss": float(loss.detach()), "draft_top1_agreement": float(acc)}
 
+    def run_train_validation_loop(self) -> list[dict]:
+        """CLI/launcher entry point (same contract as the other recipes)."""
+        return self.run()
+
     def run(self) -> list[dict]:
         logs = []
         step = 0
