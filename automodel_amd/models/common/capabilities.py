@@ -32,17 +32,16 @@ _CAPS = {
     # families without a TP plan / with non-flash attention paths
     # PP stages replay a llama-shaped forward; gemma needs embed scaling +
     # dual-frequency rope, so PP is not supported for the family
-    "GemmaForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
+    "GemmaForCausalLM": ModelCapabilities(supports_cp=False,
                                           supports_pp=False, flash_head_dims=()),
-    "Gemma3ForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
+    "Gemma3ForCausalLM": ModelCapabilities(supports_cp=False,
                                            supports_pp=False, flash_head_dims=()),
     "Gemma3ForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_cp=False, supports_pp=False,
         flash_head_dims=()),
     "GptOssForCausalLM": ModelCapabilities(supports_tp=False, supports_cp=False,
                                            supports_ep=False, flash_head_dims=()),
-    "NemotronForCausalLM": ModelCapabilities(supports_tp=False,
-                                             supports_cp=False,
+    "NemotronForCausalLM": ModelCapabilities(supports_cp=False,
                                              flash_head_dims=()),
     "Glm4MoeForCausalLM": ModelCapabilities(supports_tp=False,
                                             supports_cp=False,

@@ -56,10 +56,32 @@ def llama_tp_plan(sequence_parallel: bool = False) -> dict:
     return plan
 
 
+def gemma_tp_plan(sequence_parallel: bool = False) -> dict:
+    """Gemma family: same projection layout as llama; SP would need plans
+    for the four per-layer norms — not wired yet."""
+    assert not sequence_parallel, "sequence_parallel not supported for gemma"
+    return llama_tp_plan(False)
+
+
+def nemotron_tp_plan(sequence_parallel: bool = False) -> dict:
+    assert not sequence_parallel, "sequence_parallel not supported for nemotron"
+    return {
+        "model.layers.*.self_attn.q_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.k_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.v_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.o_proj": RowwiseParallel(),
+        "model.layers.*.mlp.up_proj": ColwiseParallel(),
+        "model.layers.*.mlp.down_proj": RowwiseParallel(),
+    }
+
+
 _MODEL_PLANS = {
     "LlamaForCausalLM": llama_tp_plan,
     "Qwen2ForCausalLM": llama_tp_plan,
     "MistralForCausalLM": llama_tp_plan,
+    "GemmaForCausalLM": gemma_tp_plan,
+    "Gemma3ForCausalLM": gemma_tp_plan,
+    "NemotronForCausalLM": nemotron_tp_plan,
 }
 
 

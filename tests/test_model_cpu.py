@@ -232,13 +232,21 @@ def test_capabilities_sliding_window_cp_guard():
     assert not validate_model_against_mesh(m, {"cp": 1})
 
 
-def test_capabilities_new_families_reject_tp():
+def test_capabilities_new_families_reject_unsupported_axes():
     from automodel_amd.models.common.capabilities import validate_model_against_mesh
     from automodel_amd.models.gemma.model import GemmaForCausalLM
+    from automodel_amd.models.gpt_oss.model import GptOssForCausalLM
 
     m = GemmaForCausalLM(dict(vocab_size=100, hidden_size=32, intermediate_size=64,
                               num_hidden_layers=1, num_attention_heads=2,
                               num_key_value_heads=1, head_dim=16,
                               max_position_embeddings=64))
-    probs = validate_model_against_mesh(m, {"tp": 2})
-    assert any("TP" in p for p in probs)
+    # gemma now HAS a TP plan; PP/CP stay rejected
+    assert not any("TP" in p for p in validate_model_against_mesh(m, {"tp": 2}))
+    assert any("PP" in p for p in validate_model_against_mesh(m, {"pp": 2}))
+    o = GptOssForCausalLM(dict(vocab_size=100, hidden_size=32, intermediate_size=48,
+                               num_hidden_layers=2, num_attention_heads=2,
+                               num_key_value_heads=1, head_dim=16,
+                               num_local_experts=2, num_experts_per_tok=1,
+                               max_position_embeddings=64, sliding_window=8))
+    assert any("TP" in p for p in validate_model_against_mesh(o, {"tp": 2}))
