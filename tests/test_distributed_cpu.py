@@ -508,3 +508,31 @@ def _tp_gemma_worker(rank, world):
 
 def test_tp2_gemma_forward_parity():
     run_distributed(_tp_gemma_worker, world=2)
+
+
+def _tp_nemotron_worker(rank, world):
+    from automodel_amd.models.nemotron.model import NemotronForCausalLM
+    from automodel_amd.parallel.mesh import build_mesh
+    from automodel_amd.parallel.tp import apply_tp
+
+    cfg = dict(vocab_size=128, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=4,
+               num_key_value_heads=2, max_position_embeddings=64)
+    torch.manual_seed(0)
+    ref = NemotronForCausalLM(cfg)
+    ref.init_weights(device="cpu")
+    ids = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(3))
+    with torch.no_grad():
+        ref_logits = ref(ids)
+    ctx = build_mesh(tp=world, device_type="cpu")
+    torch.manual_seed(0)
+    model = NemotronForCausalLM(cfg)
+    model.init_weights(device="cpu")
+    apply_tp(model, ctx.mesh["tp"])
+    with torch.no_grad():
+        out = model(ids)
+    torch.testing.assert_close(out, ref_logits, atol=1e-5, rtol=1e-5)
+
+
+def test_tp2_nemotron_forward_parity():
+    run_distributed(_tp_nemotron_worker, world=2)
