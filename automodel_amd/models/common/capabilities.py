@@ -22,7 +22,8 @@ class ModelCapabilities:
 
 _CAPS = {
     "LlamaForCausalLM": ModelCapabilities(),
-    "MoEForCausalLM": ModelCapabilities(supports_ep=True, supports_cp=False),
+    "MoEForCausalLM": ModelCapabilities(supports_ep=True, supports_cp=False,
+                                       supports_tp=True),
     "DeepseekV3ForCausalLM": ModelCapabilities(supports_ep=True, supports_tp=False,
                                                supports_cp=False, flash_head_dims=()),
     "VLMForConditionalGeneration": ModelCapabilities(supports_pp=False,

@@ -75,6 +75,19 @@ def nemotron_tp_plan(sequence_parallel: bool = False) -> dict:
     }
 
 
+def moe_attention_tp_plan(sequence_parallel: bool = False) -> dict:
+    """MoE families: shard attention only; the stacked expert parameters and
+    the router stay replicated (EP is the expert-scaling axis — reference
+    parallelizer.py:219 safe replicated-router plan)."""
+    assert not sequence_parallel, "sequence_parallel not supported for MoE TP"
+    return {
+        "model.layers.*.self_attn.q_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.k_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.v_proj": ColwiseParallel(),
+        "model.layers.*.self_attn.o_proj": RowwiseParallel(),
+    }
+
+
 _MODEL_PLANS = {
     "LlamaForCausalLM": llama_tp_plan,
     "Qwen2ForCausalLM": llama_tp_plan,
@@ -82,6 +95,7 @@ _MODEL_PLANS = {
     "GemmaForCausalLM": gemma_tp_plan,
     "Gemma3ForCausalLM": gemma_tp_plan,
     "NemotronForCausalLM": nemotron_tp_plan,
+    "MoEForCausalLM": moe_attention_tp_plan,
 }
 
 
