@@ -355,3 +355,44 @@ def test_tp2_moe_attention_only():
     from tests.dist_utils import run_distributed
 
     run_distributed(_tp_moe_worker, world=2)
+
+
+def _tp_ep_worker(rank, world):
+    """world 4 = TP2 x EP2: attention TP-sharded, experts sliced over the
+    EP axis with the a2a dispatcher — forward matches single-process."""
+    import torch
+
+    from automodel_amd.models.registry import build_model
+    from automodel_amd.moe.parallelizer import apply_ep
+    from automodel_amd.parallel.mesh import build_mesh
+    from automodel_amd.parallel.tp import apply_tp
+
+    cfg = dict(vocab_size=128, hidden_size=32, intermediate_size=64,
+               num_hidden_layers=2, num_attention_heads=4,
+               num_key_value_heads=2, max_position_embeddings=64,
+               moe=dict(n_routed_experts=4, n_activated_experts=2,
+                        moe_intermediate_size=48))
+    torch.manual_seed(0)
+    ref = build_model(config=cfg, architecture="Qwen3MoeForCausalLM",
+                      dtype="float32", meta_init=False, device="cpu")
+    ref.init_weights(device="cpu")
+    ids = torch.randint(0, 128, (2, 16), generator=torch.Generator().manual_seed(3))
+    with torch.no_grad():
+        ref_logits = ref(ids)
+
+    ctx = build_mesh(dp_shard=2, tp=2, device_type="cpu")
+    torch.manual_seed(0)
+    model = build_model(config=cfg, architecture="Qwen3MoeForCausalLM",
+                        dtype="float32", meta_init=False, device="cpu")
+    model.init_weights(device="cpu")
+    apply_tp(model, ctx.mesh["tp"])
+    apply_ep(model, ctx.mesh["dp_shard"], dispatcher="a2a")
+    with torch.no_grad():
+        out = model(ids)
+    torch.testing.assert_close(out, ref_logits, atol=1e-5, rtol=1e-5)
+
+
+def test_tp2_x_ep2_world4():
+    from tests.dist_utils import run_distributed
+
+    run_distributed(_tp_ep_worker, world=4)
