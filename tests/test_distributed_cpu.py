@@ -536,3 +536,37 @@ def _tp_nemotron_worker(rank, world):
 
 def test_tp2_nemotron_forward_parity():
     run_distributed(_tp_nemotron_worker, world=2)
+
+
+def _recipe_pp_dp_fn(rank, world, tmpdir):
+    """world 4 = PP2 x dp_shard2: the recipe composes PP stage split with
+    FSDP over dp and completes steps."""
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.llm.train_ft import (
+        TrainFinetuneRecipeForNextTokenPrediction,
+    )
+
+    cfg = ConfigNode({
+        "seed": 42,
+        "model": {"config": TINY, "dtype": "float32"},
+        "loss_fn": {"backend": "chunked", "chunk_size": 32},
+        "optimizer": {"lr": 1e-3, "weight_decay": 0.0},
+        "step_scheduler": {"grad_acc_steps": 1, "max_steps": 2},
+        "distributed": {"dp_shard": 2, "pp": 2,
+                        "pipeline": {"schedule": "gpipe", "microbatches": 2}},
+        "dataloader": {
+            "dataset": {"kind": "mock", "num_samples": 16, "seq_len": 16,
+                        "vocab_size": 128},
+            "batch_size": 4,
+        },
+        "output_dir": f"{tmpdir}/out_ppdp",
+    })
+    r = TrainFinetuneRecipeForNextTokenPrediction(cfg)
+    r.setup()
+    r.run_train_validation_loop()
+    return r.step_scheduler.step
+
+
+def test_recipe_pp2_x_dp2_world4(tmp_path):
+    out = run_distributed(_recipe_pp_dp_fn, world=4, args=(str(tmp_path),))
+    assert all(v == 2 for v in out.values())
