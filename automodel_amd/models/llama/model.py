@@ -44,6 +44,7 @@ class LlamaConfig:
     mlp_bias: bool = False
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
     qk_norm_full: bool = False     # OLMoE-style full-projection q/k RMSNorm
+    olmo2_layout: bool = False     # OLMo-2: norms on sublayer OUTPUTS only
     bidirectional: bool = False    # no causal mask (retrieval embedding models,
                                    # reference models/llama_bidirectional/)
     no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
@@ -72,6 +73,8 @@ class LlamaConfig:
             scaling = rp
         return cls(
             qk_norm="Qwen3" in archs,
+            qk_norm_full="Olmo" in archs,
+            olmo2_layout="Olmo2" in archs,
             # Phi-3 ships fused qkv_proj / gate_up_proj weights — exactly this
             # family's fused layout (keys and math match; parity-tested)
             fused_qkv="Phi3" in archs,
@@ -232,12 +235,22 @@ class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: LlamaConfig, backend: BackendConfig,
                  use_rope: bool = True):
         super().__init__()
+        self.olmo2 = cfg.olmo2_layout
         self.self_attn = LlamaAttention(cfg, backend, use_rope=use_rope)
         self.mlp = LlamaMLP(cfg, backend)
-        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
-        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        if cfg.olmo2_layout:   # norms on outputs (OLMo-2)
+            self.post_attention_layernorm = RMSNorm(cfg.hidden_size,
+                                                    cfg.rms_norm_eps, backend.rms_norm)
+            self.post_feedforward_layernorm = RMSNorm(cfg.hidden_size,
+                                                      cfg.rms_norm_eps, backend.rms_norm)
+        else:
+            self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+            self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        if self.olmo2:
+            x = x + self.post_attention_layernorm(self.self_attn(x, cos, sin))
+            return x + self.post_feedforward_layernorm(self.mlp(x))
         x = x + self.self_attn(self.input_layernorm(x), cos, sin)
         x = x + self.mlp(self.post_attention_layernorm(x))
         return x
@@ -279,7 +292,7 @@ class LlamaModel(nn.Module):
 class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
                         "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
-                        "Ernie4_5ForCausalLM")
+                        "Ernie4_5ForCausalLM", "Olmo2ForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

@@ -653,3 +653,26 @@ def test_hf_logits_parity_mistral3():
                  image_sizes=torch.tensor([[16, 16]])).logits
         out = mine(seq, pixel_values=pix)
     torch.testing.assert_close(out, ref, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_olmo2():
+    """OLMo-2: full-width qk-norm + norms on sublayer OUTPUTS."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(80)
+    hf_cfg = transformers.Olmo2Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attn_implementation="eager", tie_word_embeddings=False,
+        eos_token_id=2, pad_token_id=0)
+    hf = transformers.Olmo2ForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Olmo2ForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.olmo2_layout and mine.config.qk_norm_full
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
