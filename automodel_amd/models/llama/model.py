@@ -45,6 +45,11 @@ class LlamaConfig:
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
     qk_norm_full: bool = False     # OLMoE-style full-projection q/k RMSNorm
     olmo2_layout: bool = False     # OLMo-2: norms on sublayer OUTPUTS only
+    # granite-style scalar multipliers (neutral defaults)
+    attention_multiplier: float | None = None   # attention scale override
+    residual_multiplier: float = 1.0
+    embedding_multiplier: float = 1.0
+    logits_scaling: float = 1.0
     bidirectional: bool = False    # no causal mask (retrieval embedding models,
                                    # reference models/llama_bidirectional/)
     no_rope_layers: list | None = None  # SmolLM3 NoPE: per-layer 1=rope, 0=skip
@@ -100,6 +105,12 @@ class LlamaConfig:
             # HF Qwen2 hardcodes qkv bias=True regardless of config fields
             attention_bias=("Qwen2ForCausalLM" in archs or "Qwen2MoeForCausalLM" in archs
                             or get("attention_bias", get("qkv_bias", False))),
+            # granite multipliers (neutral when absent); granite's attention
+            # scale IS the multiplier (no implicit 1/sqrt(d))
+            attention_multiplier=get("attention_multiplier"),
+            residual_multiplier=get("residual_multiplier", 1.0),
+            embedding_multiplier=get("embedding_multiplier", 1.0),
+            logits_scaling=get("logits_scaling", 1.0),
             mlp_bias=get("mlp_bias", False),
             initializer_range=get("initializer_range", 0.02),
         )
@@ -164,6 +175,7 @@ class LlamaAttention(nn.Module):
                 q = q.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
                 k = k.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
             q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
+        attn_scale = self.cfg.attention_multiplier  # None -> 1/sqrt(D)
         if self.cfg.sliding_window is not None:
             # windowed causal (mistral): explicit mask via sdpa — the flash
             # kernel's full-causal tiling doesn't window (round-2 kernel)
@@ -174,11 +186,11 @@ class LlamaAttention(nn.Module):
                 & (i[None, :] > i[:, None] - self.cfg.sliding_window)
             mask = torch.where(keep, 0.0, float("-inf")) \
                 .to(q.dtype).reshape(1, 1, S, S)
-            o = sdpa_masked(q, k, v, mask)
+            o = sdpa_masked(q, k, v, mask, scale=attn_scale)
         elif self.cfg.bidirectional:
             from automodel_amd.ops.attention import _sdpa
 
-            o = _sdpa(q, k, v, causal=False, scale=None)
+            o = _sdpa(q, k, v, causal=False, scale=attn_scale)
         elif active_cp() is not None:
             from automodel_amd.ops import attention as attn_mod
 
@@ -186,9 +198,11 @@ class LlamaAttention(nn.Module):
                 # packed documents under CP: block-diagonal-correct path
                 from automodel_amd.parallel.cp import cp_blockdiag_attention
 
-                o = cp_blockdiag_attention(q, k, v, attn_mod._VARLEN_CU)
+                o = cp_blockdiag_attention(q, k, v, attn_mod._VARLEN_CU,
+                                           scale=attn_scale)
             else:
-                o = cp_flash_attention(q, k, v, causal=True, backend=self.backend.attn)
+                o = cp_flash_attention(q, k, v, causal=True, scale=attn_scale,
+                                       backend=self.backend.attn)
         else:
             from automodel_amd.utils.kv_cache import active_kv_cache, maybe_update_kv
 
@@ -199,15 +213,15 @@ class LlamaAttention(nn.Module):
                 from automodel_amd.ops.attention import sdpa_masked
 
                 k, v, _ = maybe_update_kv(k, v)
-                o = sdpa_masked(q, k, v, cache.attn_mask())
+                o = sdpa_masked(q, k, v, cache.attn_mask(), scale=attn_scale)
             else:
                 k, v, cache_pos = maybe_update_kv(k, v)
                 # cached decode: q is short (often 1) and prompts are
                 # arbitrary lengths — sdpa is the right tool (GEMV-bound;
                 # the flash kernel's 128-row tiling wants training shapes)
                 attn_backend = "sdpa" if cache is not None else self.backend.attn
-                o = flash_attention(q, k, v, causal=True, backend=attn_backend,
-                                    q_start=cache_pos)
+                o = flash_attention(q, k, v, causal=True, scale=attn_scale,
+                                    backend=attn_backend, q_start=cache_pos)
         return self.o_proj(o.reshape(B, S, -1))
 
 
@@ -236,6 +250,7 @@ class LlamaDecoderLayer(nn.Module):
                  use_rope: bool = True):
         super().__init__()
         self.olmo2 = cfg.olmo2_layout
+        self.residual_multiplier = cfg.residual_multiplier
         self.self_attn = LlamaAttention(cfg, backend, use_rope=use_rope)
         self.mlp = LlamaMLP(cfg, backend)
         if cfg.olmo2_layout:   # norms on outputs (OLMo-2)
@@ -251,8 +266,9 @@ class LlamaDecoderLayer(nn.Module):
         if self.olmo2:
             x = x + self.post_attention_layernorm(self.self_attn(x, cos, sin))
             return x + self.post_feedforward_layernorm(self.mlp(x))
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
-        x = x + self.mlp(self.post_attention_layernorm(x))
+        r = self.residual_multiplier
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin) * r
+        x = x + self.mlp(self.post_attention_layernorm(x)) * r
         return x
 
 
@@ -276,6 +292,8 @@ class LlamaModel(nn.Module):
 
     def forward(self, input_ids: torch.Tensor, position_ids: torch.Tensor | None = None) -> torch.Tensor:
         x = self.embed_tokens(input_ids)
+        if self.cfg.embedding_multiplier != 1.0:
+            x = x * self.cfg.embedding_multiplier
         S = input_ids.shape[1]
         if position_ids is None:
             cos, sin = self.rope_cos[:S], self.rope_sin[:S]
@@ -292,7 +310,7 @@ class LlamaModel(nn.Module):
 class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
                         "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
-                        "Ernie4_5ForCausalLM", "Olmo2ForCausalLM")
+                        "Ernie4_5ForCausalLM", "Olmo2ForCausalLM", "GraniteForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod
@@ -337,6 +355,9 @@ class LlamaForCausalLM(nn.Module):
         position t), matching the reference's dataset convention.
         """
         hidden = self.model(input_ids, position_ids)
+        if self.config.logits_scaling != 1.0 and not return_hidden:
+            # dividing hidden pre-head == dividing logits (linear head)
+            hidden = hidden / self.config.logits_scaling
         if labels is not None:
             assert self.loss_fn is not None, "set model.loss_fn before passing labels"
             return self.loss_fn(hidden, self.lm_head.weight, labels)

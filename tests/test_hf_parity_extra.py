@@ -676,3 +676,29 @@ def test_hf_logits_parity_olmo2():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_granite():
+    """Granite: llama layout + four scalar multipliers (attention scale IS
+    the multiplier — no implicit 1/sqrt(d)); tested with NON-neutral values
+    so every multiplier is exercised."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(81)
+    hf_cfg = transformers.GraniteConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, rms_norm_eps=1e-6,
+        attention_multiplier=0.2, residual_multiplier=0.7,
+        embedding_multiplier=3.0, logits_scaling=2.5,
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.GraniteForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["GraniteForCausalLM"]
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.attention_multiplier == 0.2
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
