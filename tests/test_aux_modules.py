@@ -112,3 +112,22 @@ def test_generation_server():
     out = r.json()["output_ids"]
     ref = generate(model, torch.tensor([prompt]), max_new_tokens=8)
     assert out == ref[0, 4:].tolist()   # server(KV-cached) == plain greedy
+
+
+def _sigterm_worker(rank, world):
+    """SIGTERM on ONE rank propagates to all via the MAX all-reduce."""
+    import os
+    import signal
+
+    from automodel_amd.training.signal_handler import DistributedSignalHandler
+
+    with DistributedSignalHandler(signal.SIGUSR1) as h:
+        assert not h.signals_received()
+        if rank == 0:
+            os.kill(os.getpid(), signal.SIGUSR1)
+        # all ranks must now see it
+        assert h.signals_received()
+
+
+def test_signal_handler_propagates():
+    run_distributed(_sigterm_worker, world=2)
