@@ -50,7 +50,15 @@ class Gate(nn.Module):
                 ~mask.unsqueeze(-1).expand_as(gs).reshape(T, E), float("-inf")
             )
 
-        _, indices = select_scores.topk(cfg.n_activated_experts, dim=-1)
+        from automodel_amd.moe.router_replay import active_router_replay
+
+        rr = active_router_replay()
+        if rr is not None and rr.mode == "replay":
+            indices = rr.next_replay().to(x.device)   # R3: replayed routing
+        else:
+            _, indices = select_scores.topk(cfg.n_activated_experts, dim=-1)
+            if rr is not None and rr.mode == "record":
+                rr.record(indices)
         probs = scores.gather(1, indices)
         if cfg.norm_topk_prob and cfg.n_activated_experts > 1:
             probs = probs / probs.sum(dim=-1, keepdim=True).clamp_min(1e-20)

@@ -396,3 +396,36 @@ def test_tp2_x_ep2_world4():
     from tests.dist_utils import run_distributed
 
     run_distributed(_tp_ep_worker, world=4)
+
+
+def test_router_replay_r3():
+    """R3: replayed forward uses the RECORDED expert indices even after the
+    gate weights change; probabilities recompute against current weights."""
+    import torch
+
+    from automodel_amd.moe.config import MoEConfig
+    from automodel_amd.moe.layers import Gate
+    from automodel_amd.moe.router_replay import RouterReplay, router_replay_context
+
+    torch.manual_seed(0)
+    g = Gate(16, MoEConfig(n_routed_experts=8, n_activated_experts=2))
+    torch.nn.init.normal_(g.weight)
+    x = torch.randn(6, 16)
+    rr = RouterReplay()
+    with router_replay_context(rr):
+        rr.start_record()
+        p0, i0 = g(x)
+        rr.stop()
+    assert len(rr.records) == 1
+
+    with torch.no_grad():          # change routing weights drastically
+        g.weight.add_(torch.randn_like(g.weight) * 5)
+    p1, i1 = g(x)                  # natural routing now differs
+    assert not torch.equal(i0, i1)
+
+    with router_replay_context(rr):
+        rr.start_replay()
+        p2, i2 = g(x)
+        rr.stop()
+    assert torch.equal(i2, i0)     # replayed decisions
+    assert not torch.allclose(p2, p0)  # probs from CURRENT weights
