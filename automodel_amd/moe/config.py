@@ -21,4 +21,5 @@ class MoEConfig:
     moe_intermediate_size: int | None = None
     shared_expert_intermediate_size: int | None = None
     shared_expert_gate: bool = False   # qwen2-moe: sigmoid-gated shared expert
+    topk_then_softmax: bool = False    # granite-moe: topk raw logits, softmax over k
     fake_balanced_gate: bool = False      # benchmark ideal routing

@@ -56,10 +56,14 @@ class Gate(nn.Module):
         if rr is not None and rr.mode == "replay":
             indices = rr.next_replay().to(x.device)   # R3: replayed routing
         else:
-            _, indices = select_scores.topk(cfg.n_activated_experts, dim=-1)
+            base = logits if cfg.topk_then_softmax else select_scores
+            _, indices = base.topk(cfg.n_activated_experts, dim=-1)
             if rr is not None and rr.mode == "record":
                 rr.record(indices)
-        probs = scores.gather(1, indices)
+        if cfg.topk_then_softmax:   # granite-moe: softmax over the k logits
+            probs = logits.gather(1, indices).softmax(dim=-1)
+        else:
+            probs = scores.gather(1, indices)
         if cfg.norm_topk_prob and cfg.n_activated_experts > 1:
             probs = probs / probs.sum(dim=-1, keepdim=True).clamp_min(1e-20)
         probs = probs * cfg.route_scale

@@ -57,6 +57,15 @@ class MoEModelConfig(LlamaConfig):
             )
             base["qk_norm_full"] = True
             flavor = "qwen3_moe"   # same stacked-expert key layout
+        elif "GraniteMoe" in arch:
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_local_experts", 8),
+                n_activated_experts=hf.get("num_experts_per_tok", 2),
+                moe_intermediate_size=hf.get("intermediate_size", 1024),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                topk_then_softmax=True,
+            )
+            flavor = "qwen3_moe"   # stacked-key layout (+router rename)
         elif "Qwen2Moe" in arch:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 60),
@@ -85,20 +94,22 @@ class MoEModelConfig(LlamaConfig):
 class MoEDecoderLayer(nn.Module):
     def __init__(self, cfg: MoEModelConfig, backend: BackendConfig):
         super().__init__()
+        self.cfg = cfg
         self.self_attn = LlamaAttention(cfg, backend)
         self.mlp = MoE(cfg.hidden_size, cfg.moe)
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
 
     def forward(self, x, cos, sin):
-        x = x + self.self_attn(self.input_layernorm(x), cos, sin)
-        x = x + self.mlp(self.post_attention_layernorm(x))
+        r = self.cfg.residual_multiplier
+        x = x + self.self_attn(self.input_layernorm(x), cos, sin) * r
+        x = x + self.mlp(self.post_attention_layernorm(x)) * r
         return x
 
 
 class MoEForCausalLM(nn.Module):
     hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
-                        "OlmoeForCausalLM")
+                        "OlmoeForCausalLM", "GraniteMoeForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod
@@ -133,6 +144,8 @@ class MoEForCausalLM(nn.Module):
 
     def forward(self, input_ids, labels=None, position_ids=None, return_hidden=False, **_):
         x = self.model.embed_tokens(input_ids)
+        if self.config.embedding_multiplier != 1.0:    # granite-moe
+            x = x * self.config.embedding_multiplier
         S = input_ids.shape[1]
         cos, sin = self.model.rope_cos[:S], self.model.rope_sin[:S]
         if cos.dtype != torch.float32:
@@ -140,6 +153,8 @@ class MoEForCausalLM(nn.Module):
         for layer in self.model.layers:
             x = layer(x, cos, sin)
         x = self.model.norm(x)
+        if self.config.logits_scaling != 1.0 and not return_hidden:
+            x = x / self.config.logits_scaling
         if labels is not None:
             assert self.loss_fn is not None
             loss = self.loss_fn(x, self.lm_head.weight, labels)

@@ -42,14 +42,18 @@ class MoEStateDictAdapter:
                 layer, e, proj = m.group(1), int(m.group(2)), m.group(3)
                 proj = _MIXTRAL_MAP.get(proj, proj)
                 pending.setdefault(f"{layer}.mlp.experts.{proj}", {})[e] = t
-            elif key.endswith(".mlp.experts.gate_up_proj"):
+            elif key.endswith("experts.gate_up_proj"):
                 # transformers >= 4.56 stacked-expert layout: [E, 2I, H]
-                # fused as [gate; up] along dim 1 (chunk(2, dim=1))
+                # fused as [gate; up] along dim 1 (chunk(2, dim=1));
+                # granite-moe stores them under block_sparse_moe
+                base_key = key.replace(".block_sparse_moe.", ".mlp.")
                 gate, up = t.chunk(2, dim=1)
-                out[key.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
-                out[key.replace("gate_up_proj", "up_proj")] = up.contiguous()
+                out[base_key.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
+                out[base_key.replace("gate_up_proj", "up_proj")] = up.contiguous()
             elif ".mlp.shared_expert." in key:
                 out[key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")] = t
+            elif ".block_sparse_moe.router." in key:   # granite-moe
+                out[key.replace(".block_sparse_moe.router.", ".mlp.gate.")] = t
             elif ".block_sparse_moe.gate." in key:
                 out[key.replace(".block_sparse_moe.gate.", ".mlp.gate.")] = t
             elif ".block_sparse_moe." in key:

@@ -702,3 +702,31 @@ def test_hf_logits_parity_granite():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_granite_moe():
+    """GraniteMoe: granite multipliers + topk-then-softmax routing +
+    block_sparse_moe stacked-expert key layout."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(90)
+    hf_cfg = transformers.GraniteMoeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2, max_position_embeddings=128,
+        rope_theta=10000.0, rms_norm_eps=1e-6,
+        attention_multiplier=0.25, residual_multiplier=0.8,
+        embedding_multiplier=2.0, logits_scaling=1.5,
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.GraniteMoeForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["GraniteMoeForCausalLM"]
+    mine = MoEForCausalLM(MoEModelConfig.from_hf_config(d)).eval()
+    assert mine.config.moe.topk_then_softmax
+    sd = MoEStateDictAdapter(mine.config).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
