@@ -101,6 +101,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.gpt2.model import GPT2LMHeadModel
+
+        for name in GPT2LMHeadModel.hf_architectures:
+            _REGISTRY[name] = GPT2LMHeadModel
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.mistral3.model import Mistral3ForConditionalGeneration
 
         for name in Mistral3ForConditionalGeneration.hf_architectures:

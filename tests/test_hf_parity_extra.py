@@ -730,3 +730,29 @@ def test_hf_logits_parity_granite_moe():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_gpt2():
+    """GPT-2: learned absolute positions, Conv1D transposed weights,
+    biased LayerNorms, tanh-GELU."""
+    from automodel_amd.models.gpt2.model import GPT2LMHeadModel
+
+    torch.manual_seed(91)
+    hf_cfg = transformers.GPT2Config(vocab_size=300, n_embd=64, n_layer=2,
+                                     n_head=4, n_positions=128,
+                                     bos_token_id=0, eos_token_id=1,
+                                     tie_word_embeddings=False)
+    hf = transformers.GPT2LMHeadModel(hf_cfg).eval()
+    mine = GPT2LMHeadModel(GPT2LMHeadModel.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not missing and not unexpected, (missing, unexpected)
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+    # trains with the framework loss convention
+    from automodel_amd.loss.masked_ce import MaskedCrossEntropy
+
+    mine.loss_fn = lambda h, w, l: MaskedCrossEntropy()(h @ w.t(), l)
+    loss = mine(ids, labels=ids.clone())
+    loss.backward()
+    assert mine.transformer.h[0].attn.c_attn.weight.grad is not None
