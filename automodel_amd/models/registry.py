@@ -101,6 +101,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.falcon.model import FalconForCausalLM
+
+        for name in FalconForCausalLM.hf_architectures:
+            _REGISTRY[name] = FalconForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.gpt2.model import GPT2LMHeadModel
 
         for name in GPT2LMHeadModel.hf_architectures:

@@ -756,3 +756,25 @@ def test_hf_logits_parity_gpt2():
     loss = mine(ids, labels=ids.clone())
     loss.backward()
     assert mine.transformer.h[0].attn.c_attn.weight.grad is not None
+
+
+def test_hf_logits_parity_falcon():
+    """Falcon (new decoder arch): parallel attention+MLP residual, per-KV-
+    group interleaved fused qkv, rotary, biased LNs."""
+    from automodel_amd.models.falcon.model import FalconForCausalLM
+
+    torch.manual_seed(92)
+    hf_cfg = transformers.FalconConfig(
+        vocab_size=300, hidden_size=64, num_hidden_layers=2,
+        num_attention_heads=4, new_decoder_architecture=True, num_kv_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.FalconForCausalLM(hf_cfg).eval()
+    mine = FalconForCausalLM(FalconForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
