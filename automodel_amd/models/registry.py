@@ -101,6 +101,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.phi.model import PhiForCausalLM
+
+        for name in PhiForCausalLM.hf_architectures:
+            _REGISTRY[name] = PhiForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.falcon.model import FalconForCausalLM
 
         for name in FalconForCausalLM.hf_architectures:

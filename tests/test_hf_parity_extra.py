@@ -778,3 +778,25 @@ def test_hf_logits_parity_falcon():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_phi2():
+    """Phi-1/2: parallel attention+MLP over ONE shared LN, partial rotary,
+    biased everything incl. lm_head."""
+    from automodel_amd.models.phi.model import PhiForCausalLM
+
+    torch.manual_seed(93)
+    hf_cfg = transformers.PhiConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, partial_rotary_factor=0.5,
+        rope_theta=10000.0, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.PhiForCausalLM(hf_cfg).eval()
+    mine = PhiForCausalLM(PhiForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
