@@ -800,3 +800,24 @@ def test_hf_logits_parity_phi2():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_cohere():
+    """Cohere Command-R: parallel residual over one bias-free LayerNorm,
+    interleaved rope, logit_scale, tied embeddings."""
+    from automodel_amd.models.cohere.model import CohereForCausalLM
+
+    torch.manual_seed(94)
+    hf_cfg = transformers.CohereConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0, logit_scale=0.0625,
+        bos_token_id=0, eos_token_id=1, attn_implementation="eager")
+    hf = transformers.CohereForCausalLM(hf_cfg).eval()
+    mine = CohereForCausalLM(CohereForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
