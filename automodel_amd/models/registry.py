@@ -101,6 +101,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.starcoder2.model import Starcoder2ForCausalLM
+
+        for name in Starcoder2ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Starcoder2ForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.cohere.model import CohereForCausalLM
 
         for name in CohereForCausalLM.hf_architectures:
