@@ -108,6 +108,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.stablelm.model import StableLmForCausalLM
+
+        for name in StableLmForCausalLM.hf_architectures:
+            _REGISTRY[name] = StableLmForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.cohere.model import CohereForCausalLM
 
         for name in CohereForCausalLM.hf_architectures:
