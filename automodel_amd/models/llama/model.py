@@ -100,7 +100,8 @@ class LlamaConfig:
             # use_sliding_window
             sliding_window=(get("sliding_window")
                             if get("use_sliding_window", True) else None),
-            rope_interleaved="Ernie" in archs,
+            # ernie + helium use pair-interleaved rotate_half
+            rope_interleaved=("Ernie" in archs) or ("Helium" in archs),
             tie_word_embeddings=get("tie_word_embeddings", False),
             # HF Qwen2 hardcodes qkv bias=True regardless of config fields
             attention_bias=("Qwen2ForCausalLM" in archs or "Qwen2MoeForCausalLM" in archs
@@ -310,7 +311,8 @@ class LlamaModel(nn.Module):
 class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
                         "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
-                        "Ernie4_5ForCausalLM", "Olmo2ForCausalLM", "GraniteForCausalLM")
+                        "Ernie4_5ForCausalLM", "Olmo2ForCausalLM", "GraniteForCausalLM",
+                        "HeliumForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod
