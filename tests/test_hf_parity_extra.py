@@ -892,3 +892,28 @@ def test_hf_logits_parity_helium():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_exaone4():
+    """EXAONE-4: POST-norm residual blocks + per-head qk-norm + hybrid
+    sliding(rope)/full(NoPE) attention layers."""
+    from automodel_amd.models.exaone4.model import Exaone4ForCausalLM
+
+    torch.manual_seed(98)
+    hf_cfg = transformers.Exaone4Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128,
+        sliding_window=8, sliding_window_pattern=2,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    assert "full_attention" in hf_cfg.layer_types, hf_cfg.layer_types
+    hf = transformers.Exaone4ForCausalLM(hf_cfg).eval()
+    mine = Exaone4ForCausalLM(
+        Exaone4ForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
