@@ -312,7 +312,7 @@ class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
                         "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
                         "Ernie4_5ForCausalLM", "Olmo2ForCausalLM", "GraniteForCausalLM",
-                        "HeliumForCausalLM")
+                        "HeliumForCausalLM", "SeedOssForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod

@@ -917,3 +917,27 @@ def test_hf_logits_parity_exaone4():
     ids = torch.randint(0, 300, (2, 24))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_seed_oss():
+    """Seed-OSS: llama-shaped with biased qkv (unbiased o_proj) and explicit
+    head_dim — rides the llama family directly."""
+    from automodel_amd.models.llama.model import LlamaForCausalLM
+
+    torch.manual_seed(99)
+    hf_cfg = transformers.SeedOssConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["SeedOssForCausalLM"]
+    hf = transformers.SeedOssForCausalLM(hf_cfg).eval()
+    mine = LlamaForCausalLM(LlamaForCausalLM.config_from_hf(d)).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
