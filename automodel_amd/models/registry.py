@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.glm.model import GlmForCausalLM
+
+        for name in GlmForCausalLM.hf_architectures:
+            _REGISTRY[name] = GlmForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.exaone4.model import Exaone4ForCausalLM
 
         for name in Exaone4ForCausalLM.hf_architectures:

@@ -941,3 +941,31 @@ def test_hf_logits_parity_seed_oss():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+@pytest.mark.parametrize("arch", ["GlmForCausalLM", "Glm4ForCausalLM"])
+def test_hf_logits_parity_glm(arch):
+    """GLM-4 dense: fused gate_up SwiGLU + biased qkv + partial (0.5)
+    pair-interleaved rotary; Glm4 adds sandwich sublayer norms."""
+    from automodel_amd.models.glm.model import GlmForCausalLM
+
+    torch.manual_seed(100)
+    cfg_cls = transformers.GlmConfig if arch == "GlmForCausalLM" \
+        else transformers.Glm4Config
+    hf_cfg = cfg_cls(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, pad_token_id=0,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = getattr(transformers, arch)(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = [arch]
+    mine = GlmForCausalLM(GlmForCausalLM.config_from_hf(d)).eval()
+    assert mine.config.sandwich_norms == (arch == "Glm4ForCausalLM")
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
