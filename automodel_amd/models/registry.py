@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.arcee.model import ArceeForCausalLM
+
+        for name in ArceeForCausalLM.hf_architectures:
+            _REGISTRY[name] = ArceeForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.glm.model import GlmForCausalLM
 
         for name in GlmForCausalLM.hf_architectures:

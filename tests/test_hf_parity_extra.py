@@ -969,3 +969,26 @@ def test_hf_logits_parity_glm(arch):
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_arcee():
+    """Arcee/AFM: llama-shaped RMS pre-norm blocks with a GATELESS relu²
+    MLP (down(relu(up(x))²))."""
+    from automodel_amd.models.arcee.model import ArceeForCausalLM
+
+    torch.manual_seed(101)
+    hf_cfg = transformers.ArceeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.ArceeForCausalLM(hf_cfg).eval()
+    mine = ArceeForCausalLM(
+        ArceeForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
