@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.gpt_neox.model import GPTNeoXForCausalLM
+
+        for name in GPTNeoXForCausalLM.hf_architectures:
+            _REGISTRY[name] = GPTNeoXForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.arcee.model import ArceeForCausalLM
 
         for name in ArceeForCausalLM.hf_architectures:

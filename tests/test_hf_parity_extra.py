@@ -992,3 +992,27 @@ def test_hf_logits_parity_arcee():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+@pytest.mark.parametrize("parallel", [True, False])
+def test_hf_logits_parity_gpt_neox(parallel):
+    """GPT-NeoX/Pythia: per-head fused qkv, partial rotary 0.25, biased
+    LayerNorms, parallel (or sequential) attn+MLP residual."""
+    from automodel_amd.models.gpt_neox.model import GPTNeoXForCausalLM
+
+    torch.manual_seed(102)
+    hf_cfg = transformers.GPTNeoXConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        max_position_embeddings=128, use_parallel_residual=parallel,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.GPTNeoXForCausalLM(hf_cfg).eval()
+    mine = GPTNeoXForCausalLM(
+        GPTNeoXForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
