@@ -115,6 +115,15 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.olmo.model import Olmo3ForCausalLM, OlmoForCausalLM
+
+        for name in OlmoForCausalLM.hf_architectures:
+            _REGISTRY[name] = OlmoForCausalLM
+        for name in Olmo3ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Olmo3ForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.gpt_neox.model import GPTNeoXForCausalLM
 
         for name in GPTNeoXForCausalLM.hf_architectures:

@@ -1016,3 +1016,47 @@ def test_hf_logits_parity_gpt_neox(parallel):
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_olmo_v1():
+    """OLMo v1: non-parametric fp32 LayerNorms + clip_qkv clamp."""
+    from automodel_amd.models.olmo.model import OlmoForCausalLM
+
+    torch.manual_seed(103)
+    hf_cfg = transformers.OlmoConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        clip_qkv=0.2, max_position_embeddings=128,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.OlmoForCausalLM(hf_cfg).eval()
+    mine = OlmoForCausalLM(OlmoForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_olmo3():
+    """OLMo-3: olmo2-style post-norm sublayers + full-width qk-norm +
+    hybrid sliding/full attention with per-kind rope tables."""
+    from automodel_amd.models.olmo.model import Olmo3ForCausalLM
+
+    torch.manual_seed(104)
+    hf_cfg = transformers.Olmo3Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, sliding_window=8,
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    assert "full_attention" in hf_cfg.layer_types
+    hf = transformers.Olmo3ForCausalLM(hf_cfg).eval()
+    mine = Olmo3ForCausalLM(Olmo3ForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=2e-4)
