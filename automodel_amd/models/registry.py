@@ -115,6 +115,27 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.opt.model import OPTForCausalLM
+
+        for name in OPTForCausalLM.hf_architectures:
+            _REGISTRY[name] = OPTForCausalLM
+    except ImportError:
+        pass
+    try:
+        from automodel_amd.models.bloom.model import BloomForCausalLM
+
+        for name in BloomForCausalLM.hf_architectures:
+            _REGISTRY[name] = BloomForCausalLM
+    except ImportError:
+        pass
+    try:
+        from automodel_amd.models.mpt.model import MptForCausalLM
+
+        for name in MptForCausalLM.hf_architectures:
+            _REGISTRY[name] = MptForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.olmo.model import Olmo3ForCausalLM, OlmoForCausalLM
 
         for name in OlmoForCausalLM.hf_architectures:

@@ -1060,3 +1060,65 @@ def test_hf_logits_parity_olmo3():
     ids = torch.randint(0, 300, (2, 24))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=2e-4)
+
+
+@pytest.mark.parametrize("pre_ln", [True, False])
+def test_hf_logits_parity_opt(pre_ln):
+    """OPT: learned positions (+2 table offset), relu MLP, pre/post-LN."""
+    from automodel_amd.models.opt.model import OPTForCausalLM
+
+    torch.manual_seed(105)
+    hf_cfg = transformers.OPTConfig(
+        vocab_size=300, hidden_size=64, ffn_dim=128, num_hidden_layers=2,
+        num_attention_heads=4, max_position_embeddings=128,
+        do_layer_norm_before=pre_ln, word_embed_proj_dim=64,
+        bos_token_id=0, eos_token_id=1, pad_token_id=2,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.OPTForCausalLM(hf_cfg).eval()
+    mine = OPTForCausalLM(OPTForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected and not missing, (missing, unexpected)
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+@pytest.mark.parametrize("n_head", [4, 6])
+def test_hf_logits_parity_bloom(n_head):
+    """BLOOM: ALiBi bias + per-head fused qkv + post-embedding LayerNorm.
+    n_head=6 exercises the non-power-of-2 slope ladder."""
+    from automodel_amd.models.bloom.model import BloomForCausalLM
+
+    torch.manual_seed(106)
+    hf_cfg = transformers.BloomConfig(
+        vocab_size=300, hidden_size=48 if n_head == 6 else 64, n_layer=2,
+        n_head=n_head, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.BloomForCausalLM(hf_cfg).eval()
+    mine = BloomForCausalLM(BloomForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("alibi" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+@pytest.mark.parametrize("n_heads", [4, 6])
+def test_hf_logits_parity_mpt(n_heads):
+    """MPT: MosaicML ALiBi ladder (odd/even reorder for non-pow2 heads),
+    fused Wqkv, bias-free norms/projections, exact GELU."""
+    from automodel_amd.models.mpt.model import MptForCausalLM
+
+    torch.manual_seed(107)
+    hf_cfg = transformers.MptConfig(
+        vocab_size=300, d_model=48 if n_heads == 6 else 64, n_layers=2,
+        n_heads=n_heads, tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.MptForCausalLM(hf_cfg).eval()
+    mine = MptForCausalLM(MptForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("alibi" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
