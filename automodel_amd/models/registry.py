@@ -115,6 +115,20 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.persimmon.model import PersimmonForCausalLM
+
+        for name in PersimmonForCausalLM.hf_architectures:
+            _REGISTRY[name] = PersimmonForCausalLM
+    except ImportError:
+        pass
+    try:
+        from automodel_amd.models.gptj.model import GPTJForCausalLM
+
+        for name in GPTJForCausalLM.hf_architectures:
+            _REGISTRY[name] = GPTJForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.opt.model import OPTForCausalLM
 
         for name in OPTForCausalLM.hf_architectures:

@@ -1122,3 +1122,45 @@ def test_hf_logits_parity_mpt(n_heads):
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_persimmon():
+    """Persimmon: per-head fused qkv + per-head biased qk LayerNorms before
+    rope + partial rotary 0.5 + relu² MLP."""
+    from automodel_amd.models.persimmon.model import PersimmonForCausalLM
+
+    torch.manual_seed(108)
+    hf_cfg = transformers.PersimmonConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4,
+        max_position_embeddings=128, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.PersimmonForCausalLM(hf_cfg).eval()
+    mine = PersimmonForCausalLM(
+        PersimmonForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_gptj():
+    """GPT-J: parallel attn+MLP over shared ln_1, partial pair-interleaved
+    rotary (rotate_every_two), biased lm_head."""
+    from automodel_amd.models.gptj.model import GPTJForCausalLM
+
+    torch.manual_seed(109)
+    hf_cfg = transformers.GPTJConfig(
+        vocab_size=300, n_embd=64, n_layer=2, n_head=4, rotary_dim=8,
+        n_positions=128, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.GPTJForCausalLM(hf_cfg).eval()
+    mine = GPTJForCausalLM(GPTJForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
