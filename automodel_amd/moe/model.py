@@ -16,7 +16,7 @@ import torch.nn as nn
 from automodel_amd.models.common.backend import BackendConfig
 from automodel_amd.models.llama.model import LlamaAttention, LlamaConfig
 from automodel_amd.moe.config import MoEConfig
-from automodel_amd.moe.layers import MoE, Gate
+from automodel_amd.moe.layers import MoE, Gate, SharedExpert
 from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
 from automodel_amd.ops.rms_norm import RMSNorm
 from automodel_amd.ops.rope import build_rope_cache
@@ -25,7 +25,8 @@ from automodel_amd.ops.rope import build_rope_cache
 @dataclass
 class MoEModelConfig(LlamaConfig):
     moe: MoEConfig = field(default_factory=MoEConfig)
-    hf_flavor: str = "qwen3_moe"   # qwen3_moe | mixtral
+    hf_flavor: str = "qwen3_moe"   # qwen3_moe | mixtral | qwen2_moe | ernie
+    first_k_dense: int = 0         # leading dense (non-MoE) decoder layers
 
     def __post_init__(self):
         super().__post_init__()
@@ -79,6 +80,24 @@ class MoEModelConfig(LlamaConfig):
                 norm_topk_prob=hf.get("norm_topk_prob", False),
             )
             flavor = "qwen2_moe"
+        elif "Ernie4_5_Moe" in arch:
+            # softmax scores; selection biased by moe_statics correction
+            # bias, weights gathered unbiased then renormalized
+            n_shared = hf.get("moe_num_shared_experts", 0)
+            moe = MoEConfig(
+                n_routed_experts=hf.get("moe_num_experts", 64),
+                n_activated_experts=hf.get("moe_k", 8),
+                moe_intermediate_size=hf.get("moe_intermediate_size", 768),
+                n_shared_experts=1 if n_shared else 0,
+                shared_expert_intermediate_size=(
+                    hf.get("moe_intermediate_size", 768) * n_shared or None),
+                expert_bias=True,
+                norm_topk_prob=True,
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+            )
+            flavor = "ernie"
+            return cls(**base, moe=moe, hf_flavor=flavor,
+                       first_k_dense=hf.get("moe_layer_start_index", 0))
         else:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 64),
@@ -92,11 +111,13 @@ class MoEModelConfig(LlamaConfig):
 
 
 class MoEDecoderLayer(nn.Module):
-    def __init__(self, cfg: MoEModelConfig, backend: BackendConfig):
+    def __init__(self, cfg: MoEModelConfig, backend: BackendConfig, dense: bool = False):
         super().__init__()
         self.cfg = cfg
         self.self_attn = LlamaAttention(cfg, backend)
-        self.mlp = MoE(cfg.hidden_size, cfg.moe)
+        # leading dense layers (ernie/deepseek first_k_dense): plain SwiGLU
+        self.mlp = (SharedExpert(cfg.hidden_size, cfg.intermediate_size)
+                    if dense else MoE(cfg.hidden_size, cfg.moe))
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
 
@@ -109,7 +130,8 @@ class MoEDecoderLayer(nn.Module):
 
 class MoEForCausalLM(nn.Module):
     hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
-                        "OlmoeForCausalLM", "GraniteMoeForCausalLM")
+                        "OlmoeForCausalLM", "GraniteMoeForCausalLM",
+                        "Ernie4_5_MoeForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod
@@ -129,7 +151,8 @@ class MoEForCausalLM(nn.Module):
         self.model = nn.Module()
         self.model.embed_tokens = nn.Embedding(config.vocab_size, config.hidden_size)
         self.model.layers = nn.ModuleList(
-            MoEDecoderLayer(config, backend) for _ in range(config.num_hidden_layers)
+            MoEDecoderLayer(config, backend, dense=i < config.first_k_dense)
+            for i in range(config.num_hidden_layers)
         )
         self.model.norm = RMSNorm(config.hidden_size, config.rms_norm_eps, backend.rms_norm)
         cos, sin = build_rope_cache(config.head_dim, config.max_position_embeddings,
@@ -167,7 +190,8 @@ class MoEForCausalLM(nn.Module):
     def collect_aux_losses(self):
         total = None
         for layer in self.model.layers:
-            aux = getattr(layer.mlp.gate, "last_aux_loss", None)
+            gate = getattr(layer.mlp, "gate", None)
+            aux = getattr(gate, "last_aux_loss", None)
             if aux is not None:
                 total = aux if total is None else total + aux
                 layer.mlp.gate.last_aux_loss = None
@@ -178,7 +202,7 @@ class MoEForCausalLM(nn.Module):
         """Aux-free balancing bias update after each optim step
         (reference train_ft.py update_moe_gate_bias)."""
         for layer in self.model.layers:
-            load = layer.mlp.last_expert_load
+            load = getattr(layer.mlp, "last_expert_load", None)
             if load is not None:
                 layer.mlp.gate.update_bias(load)
 

@@ -50,6 +50,10 @@ class MoEStateDictAdapter:
                 gate, up = t.chunk(2, dim=1)
                 out[base_key.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
                 out[base_key.replace("gate_up_proj", "up_proj")] = up.contiguous()
+            elif key.endswith(".gate.moe_statics.e_score_correction_bias"):
+                # ernie: [1, E] fp32 selection bias -> flat buffer
+                out[key.replace(".gate.moe_statics.e_score_correction_bias",
+                                ".gate.e_score_correction_bias")] = t.reshape(-1)
             elif ".mlp.shared_expert." in key:
                 out[key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")] = t
             elif ".block_sparse_moe.router." in key:   # granite-moe
@@ -71,6 +75,23 @@ class MoEStateDictAdapter:
             sd = {k.replace(".mlp.shared_experts.", ".mlp.shared_expert."): v
                   for k, v in sd.items()}
         pat = re.compile(r"^(model\.layers\.\d+)\.mlp\.experts\.(gate_proj|up_proj|down_proj)$")
+        if self.flavor == "ernie":
+            # HF ernie stores experts STACKED: recombine gate+up -> gate_up
+            for key, t in sd.items():
+                m = pat.match(key)
+                if m and m.group(2) == "gate_proj":
+                    up = sd[key.replace("gate_proj", "up_proj")]
+                    out[key.replace("gate_proj", "gate_up_proj")] = \
+                        torch.cat([t, up], dim=1)
+                elif m and m.group(2) == "up_proj":
+                    continue
+                elif key.endswith(".gate.e_score_correction_bias"):
+                    out[key.replace(".gate.e_score_correction_bias",
+                                    ".gate.moe_statics.e_score_correction_bias")] = \
+                        t.reshape(1, -1)
+                else:
+                    out[key] = t
+            return out
         for key, t in sd.items():
             m = pat.match(key)
             if m:

@@ -1164,3 +1164,41 @@ def test_hf_logits_parity_gptj():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_ernie45_moe():
+    """Ernie-4.5-MoE: interleaved rope + leading dense layers + softmax
+    routing with selection-only correction bias (moe_statics) + stacked
+    HF expert tensors + wide shared expert."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    torch.manual_seed(110)
+    hf_cfg = transformers.Ernie4_5_MoeConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        moe_intermediate_size=48, num_hidden_layers=3,
+        num_attention_heads=4, num_key_value_heads=2,
+        moe_num_experts=4, moe_k=2, moe_num_shared_experts=2,
+        moe_layer_start_index=1, max_position_embeddings=128,
+        bos_token_id=0, eos_token_id=1, pad_token_id=2,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.Ernie4_5_MoeForCausalLM(hf_cfg).eval()
+    for n, p in hf.named_parameters():
+        if "e_score_correction_bias" in n:   # exercise biased selection
+            p.data = torch.randn_like(p.data) * 0.5
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Ernie4_5_MoeForCausalLM"]
+    mc = MoEModelConfig.from_hf_config(d)
+    assert mc.rope_interleaved and mc.first_k_dense == 1
+    mine = MoEForCausalLM(mc).eval().float()
+    adapter = MoEStateDictAdapter(mc)
+    missing, unexpected = mine.load_state_dict(
+        adapter.from_hf(hf.state_dict()), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+    # to_hf round-trips the stacked-expert + moe_statics layout exactly
+    back = adapter.to_hf(mine.state_dict())
+    assert set(hf.state_dict().keys()) - set(back.keys()) == set()
