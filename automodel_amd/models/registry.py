@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.dbrx.model import DbrxForCausalLM
+
+        for name in DbrxForCausalLM.hf_architectures:
+            _REGISTRY[name] = DbrxForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.persimmon.model import PersimmonForCausalLM
 
         for name in PersimmonForCausalLM.hf_architectures:

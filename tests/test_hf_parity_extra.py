@@ -1202,3 +1202,25 @@ def test_hf_logits_parity_ernie45_moe():
     # to_hf round-trips the stacked-expert + moe_statics layout exactly
     back = adapter.to_hf(mine.state_dict())
     assert set(hf.state_dict().keys()) - set(back.keys()) == set()
+
+
+def test_hf_logits_parity_dbrx():
+    """DBRX: norm_attn_norm blocks, clip-clamped fused Wqkv, softmax router
+    with p-norm weight normalization, flattened expert tensors."""
+    from automodel_amd.models.dbrx.model import DbrxForCausalLM
+
+    torch.manual_seed(111)
+    hf_cfg = transformers.DbrxConfig(
+        d_model=64, n_heads=4, n_layers=2, max_seq_len=128, vocab_size=300,
+        ffn_config=dict(ffn_hidden_size=48, moe_num_experts=4, moe_top_k=2),
+        attn_config=dict(kv_n_heads=2, rope_theta=10000.0, clip_qkv=8.0),
+        bos_token_id=0, eos_token_id=1, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.DbrxForCausalLM(hf_cfg).eval()
+    mine = DbrxForCausalLM(DbrxForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
