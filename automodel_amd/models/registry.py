@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.minimax.model import MiniMaxForCausalLM
+
+        for name in MiniMaxForCausalLM.hf_architectures:
+            _REGISTRY[name] = MiniMaxForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.dbrx.model import DbrxForCausalLM
 
         for name in DbrxForCausalLM.hf_architectures:

@@ -1224,3 +1224,29 @@ def test_hf_logits_parity_dbrx():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_minimax():
+    """MiniMax: hybrid lightning (block-decay linear) / full attention
+    layers, post-LN weighted residuals, mixtral-class stacked MoE. Seq
+    length spans multiple lightning blocks to exercise the KV-state
+    recurrence."""
+    from automodel_amd.models.minimax.model import MiniMaxForCausalLM
+
+    torch.manual_seed(112)
+    hf_cfg = transformers.MiniMaxConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2, block_size=8,
+        max_position_embeddings=128, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    assert "linear_attention" in hf_cfg.layer_types
+    hf = transformers.MiniMaxForCausalLM(hf_cfg).eval()
+    mine = MiniMaxForCausalLM(
+        MiniMaxForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 20))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
