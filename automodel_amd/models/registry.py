@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.hunyuan.model import HunYuanMoEV1ForCausalLM
+
+        for name in HunYuanMoEV1ForCausalLM.hf_architectures:
+            _REGISTRY[name] = HunYuanMoEV1ForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.minimax.model import MiniMaxForCausalLM
 
         for name in MiniMaxForCausalLM.hf_architectures:
