@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.bitnet.model import BitNetForCausalLM
+
+        for name in BitNetForCausalLM.hf_architectures:
+            _REGISTRY[name] = BitNetForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.hunyuan.model import HunYuanMoEV1ForCausalLM
 
         for name in HunYuanMoEV1ForCausalLM.hf_architectures:

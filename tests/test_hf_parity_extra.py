@@ -1273,3 +1273,24 @@ def test_hf_logits_parity_hunyuan_moe():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_bitnet():
+    """BitNet-b1.58 (bf16 master-weight path): llama skeleton + relu² gate
+    + sub-norms before o_proj/down_proj."""
+    from automodel_amd.models.bitnet.model import BitNetForCausalLM
+
+    torch.manual_seed(114)
+    hf_cfg = transformers.BitNetConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.BitNetForCausalLM(hf_cfg).eval()
+    mine = BitNetForCausalLM(BitNetForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
