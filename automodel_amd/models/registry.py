@@ -115,6 +115,13 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
+        from automodel_amd.models.codegen.model import CodeGenForCausalLM
+
+        for name in CodeGenForCausalLM.hf_architectures:
+            _REGISTRY[name] = CodeGenForCausalLM
+    except ImportError:
+        pass
+    try:
         from automodel_amd.models.bitnet.model import BitNetForCausalLM
 
         for name in BitNetForCausalLM.hf_architectures:

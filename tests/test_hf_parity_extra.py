@@ -1294,3 +1294,24 @@ def test_hf_logits_parity_bitnet():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_codegen():
+    """CodeGen: GPT-J parallel-residual block with mp_num=4-grouped fused
+    qkv in [q, v, k] slice order and partial interleaved rotary."""
+    from automodel_amd.models.codegen.model import CodeGenForCausalLM
+
+    torch.manual_seed(115)
+    hf_cfg = transformers.CodeGenConfig(
+        vocab_size=300, n_embd=64, n_layer=2, n_head=8, rotary_dim=4,
+        n_positions=128, bos_token_id=0, eos_token_id=1,
+        tie_word_embeddings=False, attn_implementation="eager")
+    hf = transformers.CodeGenForCausalLM(hf_cfg).eval()
+    mine = CodeGenForCausalLM(
+        CodeGenForCausalLM.config_from_hf(hf_cfg.to_dict())).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
