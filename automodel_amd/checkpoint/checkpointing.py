@@ -111,7 +111,9 @@ class Checkpointer:
                 set_optimizer_state_dict(model, optimizer, state["optimizer"])
         aux_path = os.path.join(path, "aux_state.pt")
         if os.path.exists(aux_path):
-            return torch.load(aux_path, weights_only=False)
+            # aux state is restricted to tensors/primitives (see StatefulRNG
+            # et al.) so a tampered checkpoint dir cannot execute code on load
+            return torch.load(aux_path, weights_only=True)
         return {}
 
     # ------------------------------------------------- consolidated HF export
@@ -130,15 +132,27 @@ class Checkpointer:
         if adapter is not None:
             full_sd = adapter.to_hf(full_sd)
         full_sd = {k: v.contiguous() for k, v in full_sd.items() if isinstance(v, torch.Tensor)}
-        # drop tied aliases (safetensors refuses shared storage; HF convention
-        # omits the tied lm_head — loaders re-tie from config)
-        seen: dict[int, str] = {}
+        # Drop only EXACT tied aliases (same data_ptr/shape/stride/dtype — the
+        # tied-lm_head case; HF convention omits it and loaders re-tie from
+        # config). Distinct views of one storage (fused-qkv splits, stacked
+        # MoE expert slices from to_hf adapters) are real weights: keep them,
+        # cloning so safetensors accepts the shared storage.
+        seen: dict[tuple, str] = {}
         for k in list(full_sd):
-            ptr = full_sd[k].untyped_storage().data_ptr()
-            if ptr in seen:
+            t = full_sd[k]
+            key = (t.data_ptr(), tuple(t.shape), tuple(t.stride()), t.dtype)
+            if key in seen:
                 del full_sd[k]
             else:
-                seen[ptr] = k
+                seen[key] = k
+        kept_storages: set[int] = set()
+        for k, t in full_sd.items():
+            sp = t.untyped_storage().data_ptr()
+            oversized = t.untyped_storage().nbytes() != t.numel() * t.element_size()
+            if sp in kept_storages or oversized:
+                full_sd[k] = t.clone()
+            else:
+                kept_storages.add(sp)
         total_bytes = sum(v.numel() * v.element_size() for v in full_sd.values())
         max_shard = 4 * 2**30
         if total_bytes <= max_shard:

@@ -314,5 +314,10 @@ def build_model(
             model = cls(model_cfg, backend=backend)
     else:
         model = cls(model_cfg, backend=backend)
+        # nn.Linear self-initializes but raw torch.empty Parameters (stacked
+        # MoE experts, gates) do not — run the model's init so a non-meta
+        # build never carries uninitialized memory.
+        if hasattr(model, "init_weights"):
+            model.init_weights()
     model = model.to(dtype=torch_dtype)
     return model

@@ -73,7 +73,7 @@ class BaseRecipe:
                             optimizer=getattr(self, "optimizer", None), rank=self.rank)
         else:
             aux_path = os.path.join(path, "aux_state.pt")
-            aux = torch.load(aux_path, weights_only=False) if os.path.exists(aux_path) else {}
+            aux = torch.load(aux_path, weights_only=True) if os.path.exists(aux_path) else {}
         for name, state in (aux or {}).items():
             if name in self._statefuls:
                 self._statefuls[name].load_state_dict(state)

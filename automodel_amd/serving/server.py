@@ -13,6 +13,7 @@ Test: build_app(model, tokenizer) + fastapi.testclient (CPU-tested).
 # NOTE: no `from __future__ import annotations` — pydantic must resolve the
 # locally-defined request model's unions at class-creation time.
 import argparse
+import threading
 
 import torch
 
@@ -28,6 +29,10 @@ def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False)
     )
 
     app = FastAPI(title="automodel_amd generation server")
+    # FastAPI runs sync endpoints in a threadpool; the kv-cache context
+    # (utils/kv_cache._ACTIVE) is process-global, so only one generation may
+    # be in flight at a time.
+    gen_lock = threading.Lock()
 
     class GenRequest(BaseModel):
         prompt: str | None = None
@@ -52,7 +57,7 @@ def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False)
             ids = tokenizer.encode(req.prompt)
         dev = next(model.parameters()).device
         x = torch.tensor([ids], dtype=torch.long, device=dev)
-        with torch.no_grad():
+        with gen_lock, torch.no_grad():
             if use_graph and req.temperature == 0.0:
                 out = generate_graphed(model, x, req.max_new_tokens)
             elif use_cache:

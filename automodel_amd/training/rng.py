@@ -31,9 +31,14 @@ class StatefulRNG:
             torch.cuda.manual_seed_all(seed)
 
     def state_dict(self) -> dict:
+        # Plain tensors/primitives only, so checkpoints load with
+        # torch.load(weights_only=True) (no arbitrary pickled objects).
+        py = random.getstate()
+        np_name, np_keys, np_pos, np_has_gauss, np_gauss = np.random.get_state()
         state = {
-            "python": random.getstate(),
-            "numpy": np.random.get_state(),
+            "python": [py[0], list(py[1]), py[2]],
+            "numpy": [np_name, torch.from_numpy(np_keys.copy()),
+                      int(np_pos), int(np_has_gauss), float(np_gauss)],
             "torch": torch.get_rng_state(),
         }
         if torch.cuda.is_available():
@@ -41,8 +46,10 @@ class StatefulRNG:
         return state
 
     def load_state_dict(self, state: dict) -> None:
-        random.setstate(state["python"])
-        np.random.set_state(state["numpy"])
+        py = state["python"]
+        random.setstate((py[0], tuple(py[1]), py[2]))
+        n = state["numpy"]
+        np.random.set_state((n[0], n[1].numpy().astype(np.uint32), n[2], n[3], n[4]))
         torch.set_rng_state(torch.as_tensor(state["torch"], dtype=torch.uint8, device="cpu"))
         if "cuda" in state and torch.cuda.is_available():
             torch.cuda.set_rng_state_all(state["cuda"])

@@ -321,3 +321,73 @@ def test_vlm_export_reload_roundtrip(tmp_path):
     ids = torch.randint(0, 300, (1, 8))
     with torch.no_grad():
         torch.testing.assert_close(m(ids), m2(ids))
+
+
+def test_hf_export_fused_qkv_keeps_all_projections(tmp_path):
+    """Regression: consolidated export must not dedup k/v/up projections that
+    are storage-sharing views of the fused qkv/gate_up weights."""
+    import os
+
+    import torch
+    from safetensors.torch import load_file
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.models.registry import build_model
+
+    torch.manual_seed(0)
+    m = build_model(config=dict(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64, fused_qkv=True, fused_gate_up=True),
+        architecture="LlamaForCausalLM", dtype="float32",
+        meta_init=False, device="cpu")
+    out = tmp_path / "hf"
+    Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
+    sd = {}
+    for f in os.listdir(out):
+        if f.endswith(".safetensors"):
+            sd.update(load_file(str(out / f)))
+    for i in range(2):
+        for proj in ("q_proj", "k_proj", "v_proj"):
+            assert f"model.layers.{i}.self_attn.{proj}.weight" in sd, proj
+        for proj in ("gate_proj", "up_proj", "down_proj"):
+            assert f"model.layers.{i}.mlp.{proj}.weight" in sd, proj
+    # values must match the adapter's split of the live fused weights
+    ref = m.state_dict_adapter.to_hf(m.state_dict())
+    for k in ("model.layers.0.self_attn.k_proj.weight",
+              "model.layers.1.mlp.up_proj.weight"):
+        torch.testing.assert_close(sd[k], ref[k])
+
+
+def test_hf_export_moe_keeps_all_experts(tmp_path):
+    """Regression: per-expert slices of the stacked expert weight share one
+    storage; export must keep every expert, not just expert 0."""
+    import os
+
+    import torch
+    from safetensors.torch import load_file
+
+    from automodel_amd.checkpoint.checkpointing import Checkpointer
+    from automodel_amd.models.registry import build_model
+
+    torch.manual_seed(0)
+    m = build_model(config=dict(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=1, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64,
+        moe=dict(n_routed_experts=4, n_activated_experts=2,
+                 moe_intermediate_size=48)),
+        architecture="Qwen3MoeForCausalLM", dtype="float32",
+        meta_init=False, device="cpu")
+    out = tmp_path / "hf"
+    Checkpointer(checkpoint_dir=str(tmp_path)).export_hf_safetensors(m, str(out))
+    sd = {}
+    for f in os.listdir(out):
+        if f.endswith(".safetensors"):
+            sd.update(load_file(str(out / f)))
+    expert_keys = [k for k in sd if ".mlp.experts." in k]
+    # 4 experts x 3 projections
+    assert len(expert_keys) == 12, sorted(expert_keys)
+    ref = m.state_dict_adapter.to_hf(m.state_dict())
+    for k in expert_keys:
+        torch.testing.assert_close(sd[k], ref[k])
