@@ -38,8 +38,11 @@ class BackendConfig:
             dispatcher="torch",
         )
 
-    # head dims the in-tree flash kernels are tiled for (flash_attn.hip)
-    FLASH_HEAD_DIMS = (128,)
+    # head dims the in-tree flash kernels are tiled for (flash_attn.hip);
+    # any dim <= FLASH_MAX_HEAD_DIM runs on the kernels (the wrapper
+    # zero-pads to the next tile size — ops/attention.py _target_dims).
+    FLASH_HEAD_DIMS = (64, 96, 128, 192, 256)
+    FLASH_MAX_HEAD_DIM = 256
 
     @classmethod
     def resolve(cls, cfg: "BackendConfig | dict | None", device_type: str,
@@ -51,14 +54,14 @@ class BackendConfig:
         if device_type != "cuda":
             return cfg.for_cpu()
         if (head_dim is not None and cfg.attn == "hip"
-                and head_dim not in cls.FLASH_HEAD_DIMS):
+                and head_dim > cls.FLASH_MAX_HEAD_DIM):
             # explicit, visible downgrade at model build — NOT a silent
             # runtime fallback (the kernel itself still fails loudly)
             import warnings
 
             warnings.warn(
-                f"head_dim {head_dim} not in flash kernel dims "
-                f"{cls.FLASH_HEAD_DIMS}; using sdpa attention for this model",
+                f"head_dim {head_dim} > flash kernel max "
+                f"{cls.FLASH_MAX_HEAD_DIM}; using sdpa attention for this model",
                 stacklevel=2,
             )
             cfg = replace(cfg, attn="sdpa")

@@ -155,11 +155,18 @@ class MLAAttention(nn.Module):
                                     cos, sin, backend="torch")
         k_rope = k_rope.expand(B, S, H, cfg.qk_rope_head_dim)
 
-        qf = torch.cat([q_nope, q_rope], dim=-1).transpose(1, 2)   # B,H,S,qk
-        kf = torch.cat([k_nope, k_rope], dim=-1).transpose(1, 2)
-        vf = v.transpose(1, 2)
+        qf = torch.cat([q_nope, q_rope], dim=-1)   # B,S,H,qk (192 = nope+rope)
+        kf = torch.cat([k_nope, k_rope], dim=-1)
+        if self.backend.attn == "hip" and qf.is_cuda and S % 128 == 0:
+            # split-dim flash kernel: (Dqk=192, Dv=128) instantiation
+            from automodel_amd.ops.attention import flash_attention
+
+            o = flash_attention(qf, kf, v.contiguous(), causal=True,
+                                scale=self.scale, backend="hip")
+            return self.o_proj(o.reshape(B, S, H * cfg.v_head_dim))
         o = torch.nn.functional.scaled_dot_product_attention(
-            qf, kf, vf, is_causal=True, scale=self.scale)
+            qf.transpose(1, 2), kf.transpose(1, 2), v.transpose(1, 2),
+            is_causal=True, scale=self.scale)
         return self.o_proj(o.transpose(1, 2).reshape(B, S, H * cfg.v_head_dim))
 
 

@@ -70,22 +70,65 @@ def _sdpa(q, k, v, causal: bool, scale: float | None, q_start: int = 0):
     return o.transpose(1, 2)
 
 
+# (Dqk, Dv) pairs the HIP kernels are instantiated for (flash_attn.hip);
+# (192, 128) is the MLA shape (qk = nope+rope, v = v_head_dim).
+_FLASH_DIMS = (64, 96, 128, 192, 256)
+_FLASH_PAIRS = {(64, 64), (96, 96), (128, 128), (192, 128), (192, 192), (256, 256)}
+
+
+def _target_dims(dqk: int, dv: int) -> tuple[int, int]:
+    """Smallest supported (Dqk, Dv) kernel pair >= the given dims (other
+    dims are zero-padded by the wrapper: padding qk dims leaves scores
+    unchanged, padded V columns are sliced off the output)."""
+    if (dqk, dv) in _FLASH_PAIRS:
+        return dqk, dv
+    tq = next((d for d in _FLASH_DIMS if d >= dqk), None)
+    tv = next((d for d in _FLASH_DIMS if d >= dv), None)
+    if tq is None or tv is None:
+        raise ValueError(f"head dims ({dqk},{dv}) exceed flash kernel max 256")
+    if (tq, tv) in _FLASH_PAIRS:
+        return tq, tv
+    t = max(tq, tv)
+    return t, t
+
+
+def _pad_head(t: torch.Tensor, target: int) -> torch.Tensor:
+    d = t.shape[-1]
+    return t if d == target else torch.nn.functional.pad(t, (0, target - d))
+
+
 class _FlashAttnHip(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal: bool, scale: float, q_start: int):
+    def forward(ctx, q, k, v, causal: bool, scale: float, q_start: int,
+                doc_start=None, doc_end=None):
         q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        o, lse = hip_ops().flash_attn_fwd(q, k, v, scale, causal, q_start)
+        o, lse = hip_ops().flash_attn_fwd(q, k, v, scale, causal, q_start, doc_start)
         ctx.save_for_backward(q, k, v, o, lse)
         ctx.causal, ctx.scale, ctx.q_start = causal, scale, q_start
+        ctx.docs = (doc_start, doc_end)
         return o
 
     @staticmethod
     def backward(ctx, do):
         q, k, v, o, lse = ctx.saved_tensors
+        ds, de = ctx.docs
         dq, dk, dv = hip_ops().flash_attn_bwd(
-            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal, ctx.q_start
+            do.contiguous(), q, k, v, o, lse, ctx.scale, ctx.causal, ctx.q_start,
+            ds, de,
         )
-        return dq, dk, dv, None, None, None
+        return dq, dk, dv, None, None, None, None, None
+
+
+def _flash_hip(q, k, v, causal, scale, q_start, doc_start=None, doc_end=None):
+    """Dispatch to the HIP kernel, zero-padding head dims to a supported
+    kernel pair when needed (pad/slice sit outside the Function so autograd
+    routes gradients through them)."""
+    dqk, dv = q.shape[-1], v.shape[-1]
+    tq, tv = _target_dims(dqk, dv)
+    if tq != dqk or tv != dv:
+        q, k, v = _pad_head(q, tq), _pad_head(k, tq), _pad_head(v, tv)
+    o = _FlashAttnHip.apply(q, k, v, causal, scale, q_start, doc_start, doc_end)
+    return o[..., :dv] if tv != dv else o
 
 
 _VARLEN_CU: torch.Tensor | None = None
@@ -99,33 +142,46 @@ def set_varlen_context(cu_seqlens: torch.Tensor | None) -> None:
     _VARLEN_CU = cu_seqlens
 
 
+def doc_bounds_from_cu(cu_seqlens: torch.Tensor, total: int) -> tuple[torch.Tensor, torch.Tensor]:
+    """Per-token document bounds for the varlen kernels: doc_start[t] /
+    doc_end[t] of the packed document containing token t. Tokens in
+    [cu[-1], total) (seq-length padding) form their own trailing doc."""
+    cu = cu_seqlens.to(torch.long)
+    lengths = cu[1:] - cu[:-1]
+    ds = torch.repeat_interleave(cu[:-1], lengths)
+    de = torch.repeat_interleave(cu[1:], lengths)
+    t_real = int(cu[-1])
+    if t_real < total:
+        pad = total - t_real
+        ds = torch.cat([ds, torch.full((pad,), t_real, dtype=torch.long, device=ds.device)])
+        de = torch.cat([de, torch.full((pad,), total, dtype=torch.long, device=de.device)])
+    return ds.to(torch.int32).contiguous(), de.to(torch.int32).contiguous()
+
+
 def flash_attention_varlen(q, k, v, cu_seqlens: torch.Tensor, scale: float | None = None,
                            backend: str = "hip") -> torch.Tensor:
     """Packed-sequence attention: q/k/v [1, T, H, D], cu_seqlens int32 [n+1].
 
-    GPU path: per-document HIP flash calls with zero-padding to the kernel's
-    128-row granularity — appended pad KV sits at positions > every real q
-    position, so causal masking excludes it; pad q rows are sliced off.
+    GPU path: ONE kernel launch per batch — per-token doc bounds gate the
+    causal mask inside the HIP kernel and clip KV-tile ranges per block
+    (replaces the round-1 per-document launch loop with its host-side
+    cu_seqlens.tolist() sync; reference THD runtime thd_utils.py:85).
     CPU path: one sdpa call with a dense block-causal mask.
     """
     assert q.shape[0] == 1, "varlen expects a packed THD batch [1, T, H, D]"
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
-    bounds = cu_seqlens.tolist()
     if backend == "hip" and q.is_cuda:
-        outs = []
-        for a, b in zip(bounds[:-1], bounds[1:]):
-            L = b - a
-            if L == 0:
-                continue
-            pad = (-L) % 128
-            qc, kc, vc = (t[:, a:b] for t in (q, k, v))
-            if pad:
-                qc = torch.nn.functional.pad(qc, (0, 0, 0, 0, 0, pad))
-                kc = torch.nn.functional.pad(kc, (0, 0, 0, 0, 0, pad))
-                vc = torch.nn.functional.pad(vc, (0, 0, 0, 0, 0, pad))
-            o = _FlashAttnHip.apply(qc, kc, vc, True, scale, 0)
-            outs.append(o[:, :L])
-        return torch.cat(outs, dim=1)
+        T = q.shape[1]
+        pad = (-T) % 128
+        Tp = T + pad
+        cu_dev = cu_seqlens.to(q.device)
+        ds, de = doc_bounds_from_cu(cu_dev, Tp)
+        if pad:
+            q = torch.nn.functional.pad(q, (0, 0, 0, 0, 0, pad))
+            k = torch.nn.functional.pad(k, (0, 0, 0, 0, 0, pad))
+            v = torch.nn.functional.pad(v, (0, 0, 0, 0, 0, pad))
+        o = _flash_hip(q, k, v, True, scale, 0, ds, de)
+        return o[:, :T]
     from automodel_amd.datasets.llm.packed_sequence import block_causal_mask
 
     mask = block_causal_mask(cu_seqlens.cpu(), q.shape[1]).to(q.device)
@@ -149,7 +205,7 @@ def flash_attention(
         return flash_attention_varlen(q, k, v, _VARLEN_CU, scale, backend)
     scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
     if backend == "hip" and q.is_cuda:
-        return _FlashAttnHip.apply(q, k, v, causal, scale, q_start)
+        return _flash_hip(q, k, v, causal, scale, q_start)
     if backend == "eager":
         return attention_ref(q, k, v, causal, scale, q_start)
     return _sdpa(q, k, v, causal, scale, q_start)

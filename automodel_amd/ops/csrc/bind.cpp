@@ -29,11 +29,14 @@ TORCH_LIBRARY(amd_ops, m) {
       "float weight_decay) -> ()");
   m.impl("adamw_step", &amd_ops::adamw_step);
 
-  m.def("flash_attn_fwd(Tensor q, Tensor k, Tensor v, float scale, bool causal, int q_start=0) -> (Tensor, Tensor)");
+  m.def(
+      "flash_attn_fwd(Tensor q, Tensor k, Tensor v, float scale, bool causal, "
+      "int q_start=0, Tensor? doc_start=None) -> (Tensor, Tensor)");
   m.impl("flash_attn_fwd", &amd_ops::flash_attn_fwd);
   m.def(
       "flash_attn_bwd(Tensor dout, Tensor q, Tensor k, Tensor v, Tensor o, Tensor lse, "
-      "float scale, bool causal, int q_start=0) -> (Tensor, Tensor, Tensor)");
+      "float scale, bool causal, int q_start=0, Tensor? doc_start=None, "
+      "Tensor? doc_end=None) -> (Tensor, Tensor, Tensor)");
   m.impl("flash_attn_bwd", &amd_ops::flash_attn_bwd);
 
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");

@@ -1,7 +1,10 @@
-// Flash attention forward (CDNA4, gfx950) — BSHD, causal, GQA, D=128.
+// Flash attention fwd+bwd (CDNA4, gfx950) — BSHD + packed varlen (THD),
+// causal, GQA, head dims {64, 96, 128, 192, 256} and split-dim MLA (qk 192 /
+// v 128). Deterministic backward (no atomics).
 //
 // MI355X-native replacement for the reference's TE DotProductAttention /
-// flash-attn backends. Structure follows the CDNA4
+// flash-attn backends (te_attention.py:352 arbitrary
+// dims, distributed/thd_utils.py:85 THD varlen). Structure follows the CDNA4
 // guide's fused-attention recipe (cdna_hip_programming.md Appendix B):
 //   * swapped QK^T — mfma_f32_32x32x16_bf16 computing mfma(K, Q) so each
 //     lane holds a P column for ONE q row -> softmax is in-register
@@ -9,16 +12,22 @@
 //   * online softmax (running m, l per q row)
 //   * P -> bf16 via packed cvt + __builtin_amdgcn_permlane32_swap to build
 //     the PV A-fragment without LDS round trips (guide T12)
-//   * K tile LDS-staged with XOR swizzle (guide T2 / Guideline 4: row-major
-//     [32][128] bf16 is a 16-way ds_read_b128 conflict without it)
+//   * K tile LDS-staged with XOR swizzle (guide T2 / Guideline 4), row
+//     stride padded to a multiple of 256 B so the ((row&15)<<4) swizzle is
+//     conflict-free at every head dim
 //   * V tile staged TRANSPOSED in LDS (vt[d][k]) with its own XOR swizzle
 //     so PV B-fragments are contiguous ds_read_b128
 //
-// Workgroup: 4 waves, each owning QBLK=32 q rows (block tile = 128 rows),
-// sharing the K/V LDS tiles; KVBLK=32 per iteration.
+// Varlen (packed THD): per-token doc_start/doc_end arrays (precomputed from
+// cu_seqlens in ops/attention.py) gate the causal mask to the query's own
+// document; KV-tile ranges are clipped per block/wave so cross-document
+// tiles are skipped, giving ONE kernel launch per packed batch.
 //
-// Backward is currently a composite of hipBLASLt GEMMs driven from python
-// (ops/attention.py); the fused HIP backward kernel is the next milestone.
+// Backward determinism: the dKV kernel's grid is (kv tiles, Hk, B) and each
+// block loops over the GQA query-head group, accumulating dK/dV in registers
+// across the whole q loop — plain stores, no atomicAdd, bitwise
+// run-to-run-identical grads (reference determinism discipline,
+// moe/experts.py:66).
 
 #include <torch/library.h>
 #include <ATen/ATen.h>
@@ -34,38 +43,51 @@ typedef float f32x16 __attribute__((ext_vector_type(16)));
 
 #define QBLK 32
 #define KVBLK 32
-#define FA_WAVES 4
-#define FA_BLOCK (FA_WAVES * WAVE_SIZE)
 
-// K tile: [KVBLK][D] bf16 row-major, 256 B rows. A 256-B row starts every
-// 64 banks, so ALL rows share bank 0 without a swizzle; ((row&15)<<4)
-// spreads a 16-lane ds_read_b128 group over all 16 slots -> conflict-free
-// (guide Guideline 4). PMC before the fix: SQ_LDS_BANK_CONFLICT 2.6e11 in
-// the dkv kernel alone (profiles/r1_pmc_counters.csv).
-__device__ __forceinline__ int k_lds_off(int row, int byte_in_row) {
-  return row * 256 + (byte_in_row ^ ((row & 15) << 4));
+// LDS row-major tile offsets: row stride padded to a multiple of 256 B so
+// every row starts at bank 0; ((row&15)<<4) rotates the 16-B slot per row ->
+// a 16-lane ds_read_b128 group (16 distinct rows, same in-row byte) hits 16
+// distinct slots (guide Guideline 4). PMC evidence: profiles/r1_pmc_*.csv.
+__host__ __device__ constexpr int pad256(int bytes) { return ((bytes + 255) / 256) * 256; }
+
+__device__ __forceinline__ int row_lds_off(int row, int byte_in_row, int stride_b) {
+  return row * stride_b + (byte_in_row ^ ((row & 15) << 4));
 }
-// V^T tile: [D][KVBLK] bf16, 64 B rows (16 words). Rows d and d+4 share the
-// same 16-word bank window (16*d mod 64); rotating the 16-B slot by
-// (d>>2)&3 separates the four colliding rows. (The previous (d&3) rotation
-// only moved rows that never collided.)
+// V^T-style tile: [D][KVBLK] bf16, 64 B rows (16 words). Rows d and d+4 share
+// the same 16-word bank window; ((d>>2)^(d>>3))&3 rotation separates the four
+// colliding rows for reads AND spreads the transpose-store writers.
 __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
-  // ((d>>2)^(d>>3))&3: the read-colliding set {d, d+4, d+8, d+12} still maps
-  // to 4 distinct slots (conflict-free b128 reads) AND the transpose-store
-  // writers (d stepping 8) now spread over all 4 slots (8-way -> 4-way).
   return d * 64 + (byte_in_row ^ ((((d >> 2) ^ (d >> 3)) & 3) << 4));
 }
+// per-wave transpose buffer: [32][64 B]
+__device__ __forceinline__ int tb_off(int row, int byte_in_row) {
+  return row * 64 + (byte_in_row ^ ((((row >> 2) ^ (row >> 3)) & 3) << 4));
+}
 
-template <int D, int WAVES>
+// Min-waves-per-SIMD for the backward kernels: the f32 accumulators scale
+// with head dim; past ~240 VGPR/lane two waves/SIMD would spill heavily, so
+// the big-D instantiations run one wave/SIMD (4-wave workgroups).
+__host__ __device__ constexpr int occ_dkv(int DQK, int DV) {
+  return (3 * (DQK + DV) / 4 + 16) <= 240 ? 2 : 1;
+}
+__host__ __device__ constexpr int occ_dq(int DQK, int DV) {
+  return (DQK / 4 + DV / 4 + DQK / 2 + 48) <= 240 ? 2 : 1;
+}
+
+// ===========================================================================
+// Forward
+// ===========================================================================
+
+template <int DQK, int DV, int WAVES, bool VARLEN>
 __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
     const bf16* __restrict__ q, const bf16* __restrict__ k, const bf16* __restrict__ v,
-    bf16* __restrict__ out, float* __restrict__ lse, int B, int Sq, int Skv, int Hq,
-    int Hk, int q_start, float scale, bool causal) {
-  static_assert(D == 128, "flash_fwd: D=128 only for now");
+    bf16* __restrict__ out, float* __restrict__ lse, const int* __restrict__ doc_start,
+    int B, int Sq, int Skv, int Hq, int Hk, int q_start, float scale, bool causal) {
+  constexpr int KB = pad256(DQK * 2);     // K tile row stride (bytes)
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                       // KVBLK*D*2 = 8 KiB
-  char* vt_lds = smem + KVBLK * D * 2;      // D*KVBLK*2 = 8 KiB
-  float* bcast = reinterpret_cast<float*>(smem + 2 * KVBLK * D * 2);  // FA_WAVES*32
+  char* k_lds = smem;                               // KVBLK*KB
+  char* vt_lds = smem + KVBLK * KB;                 // DV*64
+  float* bcast = reinterpret_cast<float*>(smem + KVBLK * KB + DV * 64);
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE_SIZE - 1);
@@ -78,25 +100,32 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
   const int b = blockIdx.z;
   const int kvh = h / (Hq / Hk);
 
-  const long q_base = (((long)b * Sq) * Hq + h) * D;       // + s*Hq*D
-  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;   // + s*Hk*D
-  const long q_row_stride = (long)Hq * D;
-  const long kv_row_stride = (long)Hk * D;
+  const long q_base = (((long)b * Sq) * Hq + h) * DQK;
+  const long k_base = (((long)b * Skv) * Hk + kvh) * DQK;
+  const long v_base = (((long)b * Skv) * Hk + kvh) * DV;
+  const long o_base = (((long)b * Sq) * Hq + h) * DV;
+  const long q_rs = (long)Hq * DQK, k_rs = (long)Hk * DQK;
+  const long v_rs = (long)Hk * DV, o_rs = (long)Hq * DV;
 
   const int q0 = qblk0 + wid * QBLK;   // this wave's first q row
 
   // ---- load Q fragments to registers: lane holds Q[q0+col][8*half + j + 16*c]
-  bf16x8_v qfrag[D / 16];
+  bf16x8_v qfrag[DQK / 16];
 #pragma unroll
-  for (int c = 0; c < D / 16; ++c) {
-    const bf16* src = q + q_base + (long)(q0 + col) * q_row_stride + c * 16 + half * 8;
+  for (int c = 0; c < DQK / 16; ++c) {
+    const bf16* src = q + q_base + (long)(q0 + col) * q_rs + c * 16 + half * 8;
     qfrag[c] = *reinterpret_cast<const bf16x8_v*>(src);
   }
 
+  // varlen doc bounds: lane's q row is fixed for the whole kernel
+  const int qg_lane = q_start + q0 + col;
+  const int ds_lane = VARLEN ? doc_start[qg_lane] : 0;        // lane's doc start
+  const int ds_wave = VARLEN ? doc_start[q_start + q0] : 0;   // wave min (non-decreasing)
+
   // ---- accumulators
-  f32x16 o_acc[D / 32];
+  f32x16 o_acc[DV / 32];
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t)
+  for (int t = 0; t < DV / 32; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) o_acc[t][r] = 0.f;
   float m_run = -1e30f, l_run = 0.f;
@@ -104,19 +133,25 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
   const int q_block_max = q_start + qblk0 + WAVES * QBLK - 1;
   const int n_tiles = causal ? (min(q_block_max, Skv - 1) / KVBLK + 1)
                              : (Skv + KVBLK - 1) / KVBLK;
+  // block-uniform start tile (varlen: nothing before the block's first doc)
+  const int jt0 = VARLEN ? doc_start[q_start + qblk0] / KVBLK : 0;
 
-  for (int jt = 0; jt < n_tiles; ++jt) {
+  for (int jt = jt0; jt < n_tiles; ++jt) {
     const int k0 = jt * KVBLK;
     // ---- stage K tile (swizzled) and V^T tile cooperatively
     {
-      // tile = 32 rows x 128 cols; 16 threads per row, WAVES*4 rows per pass
-      const int c0 = (tid % 16) * 8;
-#pragma unroll
-      for (int row = tid / 16; row < KVBLK; row += WAVES * 4) {
+      constexpr int KCH = DQK / 8;     // 16-B chunks per K row
+      for (int idx = tid; idx < KVBLK * KCH; idx += WAVES * WAVE_SIZE) {
+        const int row = idx / KCH, c0 = (idx % KCH) * 8;
         const int ks = min(k0 + row, Skv - 1);  // overhang rows masked later
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_row_stride + c0);
-        *reinterpret_cast<bf16x8*>(k_lds + k_lds_off(row, c0 * 2)) = kv8;
-        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)ks * kv_row_stride + c0);
+        *reinterpret_cast<bf16x8*>(k_lds + row_lds_off(row, c0 * 2, KB)) =
+            *reinterpret_cast<const bf16x8*>(k + k_base + (long)ks * k_rs + c0);
+      }
+      constexpr int VCH = DV / 8;
+      for (int idx = tid; idx < KVBLK * VCH; idx += WAVES * WAVE_SIZE) {
+        const int row = idx / VCH, c0 = (idx % VCH) * 8;
+        const int ks = min(k0 + row, Skv - 1);
+        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + v_base + (long)ks * v_rs + c0);
 #pragma unroll
         for (int j = 0; j < 8; ++j) {  // transpose store: vt[d][k]
           *reinterpret_cast<bf16*>(vt_lds + vt_lds_off(c0 + j, row * 2)) = vv8.v[j];
@@ -125,7 +160,8 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
     }
     __syncthreads();
 
-    const bool tile_live = !causal || (k0 <= q_start + q0 + QBLK - 1);
+    const bool tile_live = (!causal || (k0 <= q_start + q0 + QBLK - 1)) &&
+                           (!VARLEN || (k0 + KVBLK > ds_wave));
     float alpha = 1.f;
     bf16x8_v pa0, pa1;
     if (tile_live) {
@@ -134,22 +170,20 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) p[r] = 0.f;
 #pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
-        // A fragment: K[l&31][8*half + j] at d-chunk c
+      for (int c = 0; c < DQK / 16; ++c) {
         bf16x8_v ka = *reinterpret_cast<const bf16x8_v*>(
-            k_lds + k_lds_off(col, (c * 16 + half * 8) * 2));
+            k_lds + row_lds_off(col, (c * 16 + half * 8) * 2, KB));
         p = __builtin_amdgcn_mfma_f32_32x32x16_bf16(ka, qfrag[c], p, 0, 0, 0);
       }
 
-      // ---- masked online softmax, IN PLACE in the p accumulator (register
-      // budget: s_val/pv scratch arrays cost 32 VGPRs -> occupancy cliff)
-      const int qg = q_start + q0 + col;
+      // ---- masked online softmax, IN PLACE in the p accumulator
+      const int qg = qg_lane;
       float tile_max = -1e30f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int kg = k0 + (r & 3) + 8 * (r >> 2) + 4 * half;
         float sv = p[r] * scale;
-        if ((causal && kg > qg) || kg >= Skv) sv = -1e30f;
+        if ((causal && kg > qg) || kg >= Skv || (VARLEN && kg < ds_lane)) sv = -1e30f;
         p[r] = sv;
         tile_max = fmaxf(tile_max, sv);
       }
@@ -175,7 +209,6 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
         unsigned hi = __builtin_bit_cast(unsigned short, f2bf(p[2 * i + 1]));
         pk[i] = lo | (hi << 16);
       }
-      // fragment 0: k 0..15  <- regs 0..7 ; fragment 1: k 16..31 <- regs 8..15
 #pragma unroll
       for (int i = 0; i < 2; ++i) {
         auto r02 = __builtin_amdgcn_permlane32_swap(pk[4 * i + 0], pk[4 * i + 2], false, false);
@@ -194,7 +227,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
     if (tile_live) {
       // ---- rescale O by alpha (per q row: row = (r&3) + 8*(r>>2) + 4*half)
 #pragma unroll
-      for (int t = 0; t < D / 32; ++t) {
+      for (int t = 0; t < DV / 32; ++t) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           const int row = (r & 3) + 8 * (r >> 2) + 4 * half;
@@ -202,8 +235,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
         }
       }
 #pragma unroll
-      for (int t = 0; t < D / 32; ++t) {
-        // B fragments: vt[d = t*32 + (l&31)][k], k chunks of 8
+      for (int t = 0; t < DV / 32; ++t) {
         bf16x8_v vb0 = *reinterpret_cast<const bf16x8_v*>(
             vt_lds + vt_lds_off(t * 32 + col, (half * 8) * 2));
         bf16x8_v vb1 = *reinterpret_cast<const bf16x8_v*>(
@@ -224,82 +256,105 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
     lse[((long)b * Hq + h) * Sq + qg] = (l_run > 0.f) ? m_run + __logf(l_run) : -1e30f;
   }
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t) {
+  for (int t = 0; t < DV / 32; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int row = (r & 3) + 8 * (r >> 2) + 4 * half;  // q row in wave tile
       const int qrow = q0 + row;
       if (qrow < Sq) {
         const float inv_l = bcast[wid * 32 + row];
-        out[q_base + (long)qrow * q_row_stride + t * 32 + col] = f2bf(o_acc[t][r] * inv_l);
+        out[o_base + (long)qrow * o_rs + t * 32 + col] = f2bf(o_acc[t][r] * inv_l);
       }
     }
   }
 }
 
-std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at::Tensor& k,
-                                                  const at::Tensor& v, double scale,
-                                                  bool causal, int64_t q_start) {
+template <int DQK, int DV, int WAVES, bool VARLEN>
+static void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
+                       at::Tensor& out, at::Tensor& lse, const int* doc_start,
+                       int B, int Sq, int Skv, int Hq, int Hk, int q_start,
+                       float scale, bool causal, hipStream_t stream) {
+  const dim3 grid(Sq / (WAVES * QBLK), Hq, B);
+  const size_t smem = KVBLK * pad256(DQK * 2) + DV * 64 + WAVES * 32 * sizeof(float);
+  hipLaunchKernelGGL((flash_fwd_kernel<DQK, DV, WAVES, VARLEN>), grid,
+                     dim3(WAVES * WAVE_SIZE), smem, stream,
+                     reinterpret_cast<const bf16*>(q.data_ptr()),
+                     reinterpret_cast<const bf16*>(k.data_ptr()),
+                     reinterpret_cast<const bf16*>(v.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
+                     doc_start, B, Sq, Skv, Hq, Hk, q_start, scale, causal);
+  HIP_CHECK_KERNEL();
+}
+
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v, double scale,
+    bool causal, int64_t q_start, const std::optional<at::Tensor>& doc_start) {
   TORCH_CHECK(q.is_cuda() && q.dim() == 4 && q.scalar_type() == at::kBFloat16,
               "flash_attn_fwd: q must be [B,S,Hq,D] bf16");
-  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Skv = k.size(1), Hk = k.size(2);
-  TORCH_CHECK(D == 128, "flash_attn_fwd: only D=128 supported, got ", D);
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), Dqk = q.size(3);
+  const int Skv = k.size(1), Hk = k.size(2), Dv = v.size(3);
   TORCH_CHECK(Hq % Hk == 0, "flash_attn_fwd: Hq must be divisible by Hk");
-  TORCH_CHECK(Sq % (FA_WAVES * QBLK) == 0, "flash_attn_fwd: Sq must be a multiple of 128");
+  TORCH_CHECK(Sq % (4 * QBLK) == 0, "flash_attn_fwd: Sq must be a multiple of 128");
   TORCH_CHECK(Skv % KVBLK == 0, "flash_attn_fwd: Skv must be a multiple of 32");
-  auto out = at::empty_like(q);
+  const bool varlen = doc_start.has_value();
+  const int* ds_ptr = varlen ? doc_start->data_ptr<int>() : nullptr;
+  if (varlen) TORCH_CHECK(causal && B == 1, "varlen flash is causal with B==1");
+  auto out = at::empty({B, Sq, Hq, Dv}, q.options());
   auto lse = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
-  auto stream = c10::hip::getCurrentHIPStream();
-  // 8-wave blocks amortize the shared K/V staging over twice the waves;
-  // 4-wave fallback keeps small/CP-chunk shapes (Sq % 256 != 0) working.
-  if (Sq % (8 * QBLK) == 0) {
-    const dim3 grid(Sq / (8 * QBLK), Hq, B);
-    const size_t smem = 2 * KVBLK * 128 * 2 + 8 * 32 * sizeof(float);
-    hipLaunchKernelGGL((flash_fwd_kernel<128, 8>), grid, dim3(512), smem, stream.stream(),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
-  } else {
-    const dim3 grid(Sq / (4 * QBLK), Hq, B);
-    const size_t smem = 2 * KVBLK * 128 * 2 + 4 * 32 * sizeof(float);
-    hipLaunchKernelGGL((flash_fwd_kernel<128, 4>), grid, dim3(256), smem, stream.stream(),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       reinterpret_cast<bf16*>(out.data_ptr()), lse.data_ptr<float>(),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  auto stream = c10::hip::getCurrentHIPStream().stream();
+
+  bool done = false;
+#define FA_FWD_CASE(A, C)                                                             \
+  if (!done && Dqk == A && Dv == C) {                                                 \
+    done = true;                                                                      \
+    if (varlen) {                                                                     \
+      launch_fwd<A, C, 4, true>(q, k, v, out, lse, ds_ptr, B, Sq, Skv, Hq, Hk,        \
+                                (int)q_start, (float)scale, causal, stream);          \
+    } else if (Sq % (8 * QBLK) == 0) {                                                \
+      launch_fwd<A, C, 8, false>(q, k, v, out, lse, nullptr, B, Sq, Skv, Hq, Hk,      \
+                                 (int)q_start, (float)scale, causal, stream);         \
+    } else {                                                                          \
+      launch_fwd<A, C, 4, false>(q, k, v, out, lse, nullptr, B, Sq, Skv, Hq, Hk,      \
+                                 (int)q_start, (float)scale, causal, stream);         \
+    }                                                                                 \
   }
-  HIP_CHECK_KERNEL();
+  FA_FWD_CASE(64, 64)
+  FA_FWD_CASE(96, 96)
+  FA_FWD_CASE(128, 128)
+  FA_FWD_CASE(192, 128)
+  FA_FWD_CASE(192, 192)
+  FA_FWD_CASE(256, 256)
+#undef FA_FWD_CASE
+  TORCH_CHECK(done, "flash_attn_fwd: unsupported head dims (Dqk=", Dqk, ", Dv=", Dv,
+              "); supported pairs: (64,64),(96,96),(128,128),(192,128),(192,192),(256,256)");
   return {out, lse};
 }
 
 // ===========================================================================
 // Backward: flash-attention-2 style split.
 //   delta kernel: delta[b,h,s] = rowsum(dO * O)
-//   kernel A (kv-parallel): recompute S^T (swapped, lse/delta lane-local),
-//     accumulate dK, dV; P^T/dS^T transposed through a per-wave LDS buffer;
-//     f32 atomicAdd into dk/dv (GQA heads collapse onto the kv head).
-//   kernel B (q-parallel): recompute S in [q][k] layout, accumulate dQ
-//     (plain stores — q tiles are exclusive per block).
-// Replaces the GEMM-composite python backward (profiles/bench8b round-1
-// showed its masked_fill/exp/mul/f32-add chain at ~25% of step time).
+//   kernel A (kv-parallel, DETERMINISTIC): grid over (kv tiles, Hk, B); each
+//     block loops the GQA query-head group and all q tiles, accumulating
+//     dK/dV in registers; plain bf16 stores (no atomics -> bitwise
+//     reproducible grads).
+//   kernel B (q-parallel): recompute S in [q][k] layout, accumulate dQ.
 // ===========================================================================
 
 __global__ void fa_delta_kernel(const bf16* __restrict__ dout, const bf16* __restrict__ o,
-                                float* __restrict__ delta, int S, int Hq, long rows) {
-  // row r = ((b*S + s)*Hq + h); one wave per row, D=128 -> 2 elems/lane
+                                float* __restrict__ delta, int S, int Hq, int DV,
+                                long rows) {
+  // row r = ((b*S + s)*Hq + h); one wave per row
   const long row = ((long)blockIdx.x * blockDim.x + threadIdx.x) / WAVE_SIZE;
   if (row >= rows) return;
   const int lane = threadIdx.x & (WAVE_SIZE - 1);
-  const bf16* dp = dout + row * 128 + lane * 2;
-  const bf16* op = o + row * 128 + lane * 2;
-  float acc = bf2f(dp[0]) * bf2f(op[0]) + bf2f(dp[1]) * bf2f(op[1]);
+  const bf16* dp = dout + row * DV;
+  const bf16* op = o + row * DV;
+  float acc = 0.f;
+  for (int j = lane * 2; j < DV; j += 2 * WAVE_SIZE) {
+    acc += bf2f(dp[j]) * bf2f(op[j]) + bf2f(dp[j + 1]) * bf2f(op[j + 1]);
+  }
   acc = wave_reduce_sum(acc);
   if (lane == 0) {
-    // delta layout [B,Hq,S]: row -> (b, s, h)
     const long h = row % Hq;
     const long bs = row / Hq;
     const long b = bs / S, s = bs % S;
@@ -307,28 +362,20 @@ __global__ void fa_delta_kernel(const bf16* __restrict__ dout, const bf16* __res
   }
 }
 
-// q/do row images: [32][256 B], XOR swizzle ((row&7)<<4) (same as k_lds_off)
-// qt/dot images:   [128][64 B], XOR swizzle ((d&3)<<4)   (same as vt_lds_off)
-// per-wave transpose buffer: [32][64 B] with ((row&3)<<4)
-__device__ __forceinline__ int tb_off(int row, int byte_in_row) {
-  return row * 64 + (byte_in_row ^ ((((row >> 2) ^ (row >> 3)) & 3) << 4));
-}
-
-#define FAB_WAVES 4
-
-template <int D, int WAVES>
-__global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
+template <int DQK, int DV, int WAVES, bool VARLEN>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd_dkv_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
-    float* __restrict__ dk, float* __restrict__ dv, int B, int Sq, int Skv, int Hq,
-    int Hk, int q_start, float scale, bool causal) {
-  static_assert(D == 128);
+    bf16* __restrict__ dk, bf16* __restrict__ dv, const int* __restrict__ doc_end,
+    int B, int Sq, int Skv, int Hq, int Hk, int q_start, float scale, bool causal) {
+  constexpr int QB = pad256(DQK * 2);   // q-row image stride
+  constexpr int OB = pad256(DV * 2);    // dO-row image stride
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* q_rows = smem;                 // 32*256 = 8 KiB
-  char* do_rows = smem + 8 * 1024;     // 8 KiB
-  char* qt = smem + 16 * 1024;         // 128*64 = 8 KiB
-  char* dot = smem + 24 * 1024;        // 8 KiB
-  char* tbuf = smem + 32 * 1024;       // per-wave 2 KiB x4
+  char* q_rows = smem;                                   // 32*QB
+  char* do_rows = smem + 32 * QB;                        // 32*OB
+  char* qt = smem + 32 * QB + 32 * OB;                   // DQK*64
+  char* dot = qt + DQK * 64;                             // DV*64
+  char* tbuf = dot + DV * 64;                            // WAVES*2048
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -337,162 +384,204 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dkv_kernel(
   const int half = lane >> 5;
 
   const int kvb = blockIdx.x * (WAVES * 32);
-  const int h = blockIdx.y;
+  const int kvh = blockIdx.y;
   const int b = blockIdx.z;
-  const int kvh = h / (Hq / Hk);
+  const int G = Hq / Hk;               // GQA group size
   const int kv0 = kvb + wid * 32;      // this wave's kv rows
 
-  const long q_base = (((long)b * Sq) * Hq + h) * D;
-  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;
-  const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
-  const float* lse_row = lse + ((long)b * Hq + h) * Sq;
-  const float* dlt_row = delta + ((long)b * Hq + h) * Sq;
+  const long k_base = (((long)b * Skv) * Hk + kvh) * DQK;
+  const long v_base = (((long)b * Skv) * Hk + kvh) * DV;
+  const long k_rs = (long)Hk * DQK, v_rs = (long)Hk * DV;
+  const long q_rs = (long)Hq * DQK, do_rs = (long)Hq * DV;
 
   // K/V rows of this wave -> A fragments in registers
-  bf16x8_v kfrag[D / 16], vfrag[D / 16];
+  bf16x8_v kfrag[DQK / 16], vfrag[DV / 16];
 #pragma unroll
-  for (int c = 0; c < D / 16; ++c) {
-    const long off = kv_base + (long)(kv0 + col) * kv_rs + c * 16 + half * 8;
-    kfrag[c] = *reinterpret_cast<const bf16x8_v*>(k + off);
-    vfrag[c] = *reinterpret_cast<const bf16x8_v*>(v + off);
+  for (int c = 0; c < DQK / 16; ++c)
+    kfrag[c] = *reinterpret_cast<const bf16x8_v*>(
+        k + k_base + (long)(kv0 + col) * k_rs + c * 16 + half * 8);
+#pragma unroll
+  for (int c = 0; c < DV / 16; ++c)
+    vfrag[c] = *reinterpret_cast<const bf16x8_v*>(
+        v + v_base + (long)(kv0 + col) * v_rs + c * 16 + half * 8);
+
+  // varlen: per-element doc_end for this wave's 16 kv rows (fixed all kernel)
+  int dend[16];
+  int dend_wave = Sq;                  // max over the wave's rows
+  if (VARLEN) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+      dend[r] = doc_end[min(kg, Skv - 1)];
+    }
+    dend_wave = doc_end[min(kv0 + 31, Skv - 1)];
   }
 
-  f32x16 dk_acc[D / 32], dv_acc[D / 32];
+  f32x16 dk_acc[DQK / 32], dv_acc[DV / 32];
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t)
+  for (int t = 0; t < DQK / 32; ++t)
 #pragma unroll
-    for (int r = 0; r < 16; ++r) { dk_acc[t][r] = 0.f; dv_acc[t][r] = 0.f; }
+    for (int r = 0; r < 16; ++r) dk_acc[t][r] = 0.f;
+#pragma unroll
+  for (int t = 0; t < DV / 32; ++t)
+#pragma unroll
+    for (int r = 0; r < 16; ++r) dv_acc[t][r] = 0.f;
 
   char* tb = tbuf + wid * 2048;
   const int jq_start = causal ? (kvb > q_start ? (kvb - q_start) / 32 : 0) : 0;
-
-  for (int jq = jq_start; jq < Sq / 32; ++jq) {
-    const int q0 = jq * 32;
-    // ---- cooperative stage: q/do rows + transposed images
-    {
-      const int c0 = (tid % 16) * 8;   // 16 threads per row of 128
-      for (int rr = tid / 16; rr < 32; rr += WAVES * 4) {
-        bf16x8 qv = *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
-        bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(dout + q_base + (long)(q0 + rr) * q_rs + c0);
-        *reinterpret_cast<bf16x8*>(q_rows + k_lds_off(rr, c0 * 2)) = qv;
-        *reinterpret_cast<bf16x8*>(do_rows + k_lds_off(rr, c0 * 2)) = dv8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          *reinterpret_cast<bf16*>(qt + vt_lds_off(c0 + j, rr * 2)) = qv.v[j];
-          *reinterpret_cast<bf16*>(dot + vt_lds_off(c0 + j, rr * 2)) = dv8.v[j];
-        }
-      }
-    }
-    __syncthreads();
-
-    const bool live = !causal || (q_start + q0 + 31 >= kv0);
-    if (live) {
-      const int ql = q0 + col;          // local q row (lse/delta index)
-      const int qg = q_start + ql;      // global position (mask)
-      const float lse_q = lse_row[ql];
-      const float dlt_q = dlt_row[ql];
-
-      // ---- S^T = K Q^T (D rows = kv, cols = q)
-      f32x16 st;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) st[r] = 0.f;
-#pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
-        bf16x8_v qb = *reinterpret_cast<const bf16x8_v*>(
-            q_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
-        st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qb, st, 0, 0, 0);
-      }
-      // ---- dP^T = V dO^T
-      f32x16 dpt;
-#pragma unroll
-      for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
-#pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
-        bf16x8_v db = *reinterpret_cast<const bf16x8_v*>(
-            do_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
-        dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], db, dpt, 0, 0, 0);
-      }
-
-      // in place: st becomes P^T, dpt becomes dS^T (register budget)
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
-        const bool masked = (causal && kg > qg) || kg >= Skv;
-        const float pv = masked ? 0.f : __expf(st[r] * scale - lse_q);
-        st[r] = pv;
-        dpt[r] = pv * (dpt[r] - dlt_q) * scale;
-      }
-
-      // ---- transpose P^T -> A-frags via wave-local LDS buffer, accumulate dV
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(st[r]);
-      }
-      __builtin_amdgcn_wave_barrier();
-#pragma unroll
-      for (int c2 = 0; c2 < 2; ++c2) {
-        bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
-            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
-#pragma unroll
-        for (int t = 0; t < D / 32; ++t) {
-          bf16x8_v dob = *reinterpret_cast<const bf16x8_v*>(
-              dot + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
-          dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
-        }
-      }
-      // ---- transpose dS^T, accumulate dK
-      __builtin_amdgcn_wave_barrier();
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
-        *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(dpt[r]);
-      }
-      __builtin_amdgcn_wave_barrier();
-#pragma unroll
-      for (int c2 = 0; c2 < 2; ++c2) {
-        bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
-            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
-#pragma unroll
-        for (int t = 0; t < D / 32; ++t) {
-          bf16x8_v qb2 = *reinterpret_cast<const bf16x8_v*>(
-              qt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
-          dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
-        }
-      }
-    }
-    __syncthreads();
+  // block-uniform end (varlen: nothing after the block's last doc)
+  int jq_end = Sq / 32;
+  if (VARLEN) {
+    const int de_blk = doc_end[min(kvb + WAVES * 32 - 1, Skv - 1)];
+    jq_end = min(jq_end, (de_blk - q_start + 31) / 32);
   }
 
-  // ---- epilogue: atomic accumulate into f32 dk/dv [B,S,Hk,D]
+  for (int g = 0; g < G; ++g) {
+    const int h = kvh * G + g;
+    const long q_base = (((long)b * Sq) * Hq + h) * DQK;
+    const long do_base = (((long)b * Sq) * Hq + h) * DV;
+    const float* lse_row = lse + ((long)b * Hq + h) * Sq;
+    const float* dlt_row = delta + ((long)b * Hq + h) * Sq;
+
+    for (int jq = jq_start; jq < jq_end; ++jq) {
+      const int q0 = jq * 32;
+      // ---- cooperative stage: q/do rows + transposed images
+      {
+        constexpr int QCH = DQK / 8;
+        for (int idx = tid; idx < 32 * QCH; idx += WAVES * 64) {
+          const int rr = idx / QCH, c0 = (idx % QCH) * 8;
+          bf16x8 qv = *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
+          *reinterpret_cast<bf16x8*>(q_rows + row_lds_off(rr, c0 * 2, QB)) = qv;
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t) {
+          for (int j = 0; j < 8; ++j)
+            *reinterpret_cast<bf16*>(qt + vt_lds_off(c0 + j, rr * 2)) = qv.v[j];
+        }
+        constexpr int OCH = DV / 8;
+        for (int idx = tid; idx < 32 * OCH; idx += WAVES * 64) {
+          const int rr = idx / OCH, c0 = (idx % OCH) * 8;
+          bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(
+              dout + do_base + (long)(q0 + rr) * do_rs + c0);
+          *reinterpret_cast<bf16x8*>(do_rows + row_lds_off(rr, c0 * 2, OB)) = dv8;
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
-      const int kg = kv0 + krow;
-      if (kg < Skv) {
-        const long off = kv_base + (long)kg * kv_rs + t * 32 + col;
-        atomicAdd(dk + off, dk_acc[t][r]);
-        atomicAdd(dv + off, dv_acc[t][r]);
+          for (int j = 0; j < 8; ++j)
+            *reinterpret_cast<bf16*>(dot + vt_lds_off(c0 + j, rr * 2)) = dv8.v[j];
+        }
       }
+      __syncthreads();
+
+      const bool live = (!causal || (q_start + q0 + 31 >= kv0)) &&
+                        (!VARLEN || (q_start + q0 < dend_wave));
+      if (live) {
+        const int ql = q0 + col;          // local q row (lse/delta index)
+        const int qg = q_start + ql;      // global position (mask)
+        const float lse_q = lse_row[ql];
+        const float dlt_q = dlt_row[ql];
+
+        // ---- S^T = K Q^T (rows = kv, cols = q)
+        f32x16 st;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) st[r] = 0.f;
+#pragma unroll
+        for (int c = 0; c < DQK / 16; ++c) {
+          bf16x8_v qb = *reinterpret_cast<const bf16x8_v*>(
+              q_rows + row_lds_off(col, (c * 16 + half * 8) * 2, QB));
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qb, st, 0, 0, 0);
+        }
+        // ---- dP^T = V dO^T
+        f32x16 dpt;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) dpt[r] = 0.f;
+#pragma unroll
+        for (int c = 0; c < DV / 16; ++c) {
+          bf16x8_v db = *reinterpret_cast<const bf16x8_v*>(
+              do_rows + row_lds_off(col, (c * 16 + half * 8) * 2, OB));
+          dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], db, dpt, 0, 0, 0);
+        }
+
+        // in place: st becomes P^T, dpt becomes dS^T
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int kg = kv0 + (r & 3) + 8 * (r >> 2) + 4 * half;
+          const bool masked = (causal && kg > qg) || kg >= Skv ||
+                              (VARLEN && qg >= dend[r]);
+          const float pv = masked ? 0.f : __expf(st[r] * scale - lse_q);
+          st[r] = pv;
+          dpt[r] = pv * (dpt[r] - dlt_q) * scale;
+        }
+
+        // ---- transpose P^T -> A-frags via wave-local LDS buffer, accumulate dV
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+          *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(st[r]);
+        }
+        __builtin_amdgcn_wave_barrier();
+#pragma unroll
+        for (int c2 = 0; c2 < 2; ++c2) {
+          bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
+              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+#pragma unroll
+          for (int t = 0; t < DV / 32; ++t) {
+            bf16x8_v dob = *reinterpret_cast<const bf16x8_v*>(
+                dot + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
+            dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
+          }
+        }
+        // ---- transpose dS^T, accumulate dK
+        __builtin_amdgcn_wave_barrier();
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+          *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(dpt[r]);
+        }
+        __builtin_amdgcn_wave_barrier();
+#pragma unroll
+        for (int c2 = 0; c2 < 2; ++c2) {
+          bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
+              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+#pragma unroll
+          for (int t = 0; t < DQK / 32; ++t) {
+            bf16x8_v qb2 = *reinterpret_cast<const bf16x8_v*>(
+                qt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
+            dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
+          }
+        }
+      }
+      __syncthreads();
+    }
+  }
+
+  // ---- epilogue: plain bf16 stores — each kv row is owned by exactly one
+  // wave of one block (grid is per-Hk), so no cross-block accumulation.
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int krow = (r & 3) + 8 * (r >> 2) + 4 * half;
+    const int kg = kv0 + krow;
+    if (kg < Skv) {
+#pragma unroll
+      for (int t = 0; t < DQK / 32; ++t)
+        dk[k_base + (long)kg * k_rs + t * 32 + col] = f2bf(dk_acc[t][r]);
+#pragma unroll
+      for (int t = 0; t < DV / 32; ++t)
+        dv[v_base + (long)kg * v_rs + t * 32 + col] = f2bf(dv_acc[t][r]);
     }
   }
 }
 
-template <int D, int WAVES>
-__global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
+template <int DQK, int DV, int WAVES, bool VARLEN>
+__global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_dq_kernel(
     const bf16* __restrict__ dout, const bf16* __restrict__ q, const bf16* __restrict__ k,
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
-    bf16* __restrict__ dq, int B, int Sq, int Skv, int Hq, int Hk, int q_start,
-    float scale, bool causal) {
-  static_assert(D == 128);
+    bf16* __restrict__ dq, const int* __restrict__ doc_start, int B, int Sq, int Skv,
+    int Hq, int Hk, int q_start, float scale, bool causal) {
+  constexpr int KBB = pad256(DQK * 2);
+  constexpr int VBB = pad256(DV * 2);
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_rows = smem;                 // 8 KiB ([32][256B] swz)
-  char* v_rows = smem + 8 * 1024;      // 8 KiB
-  char* kt = smem + 16 * 1024;         // 8 KiB ([128][64B] swz)
-  char* tbuf = smem + 24 * 1024;       // per-wave 2 KiB x WAVES
-  float* stats = reinterpret_cast<float*>(smem + 24 * 1024 + WAVES * 2048);
+  char* k_rows = smem;                        // 32*KBB
+  char* v_rows = smem + 32 * KBB;             // 32*VBB
+  char* kt = v_rows + 32 * VBB;               // DQK*64
+  char* tbuf = kt + DQK * 64;                 // WAVES*2048
+  float* stats = reinterpret_cast<float*>(tbuf + WAVES * 2048);  // 3*WAVES*32
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -506,64 +595,79 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
   const int kvh = h / (Hq / Hk);
   const int q0 = qb_blk + wid * 32;
 
-  const long q_base = (((long)b * Sq) * Hq + h) * D;
-  const long kv_base = (((long)b * Skv) * Hk + kvh) * D;
-  const long q_rs = (long)Hq * D, kv_rs = (long)Hk * D;
+  const long q_base = (((long)b * Sq) * Hq + h) * DQK;
+  const long do_base = (((long)b * Sq) * Hq + h) * DV;
+  const long k_base = (((long)b * Skv) * Hk + kvh) * DQK;
+  const long v_base = (((long)b * Skv) * Hk + kvh) * DV;
+  const long q_rs = (long)Hq * DQK, do_rs = (long)Hq * DV;
+  const long k_rs = (long)Hk * DQK, v_rs = (long)Hk * DV;
   const float* lse_row = lse + ((long)b * Hq + h) * Sq;
   const float* dlt_row = delta + ((long)b * Hq + h) * Sq;
 
   // Q/dO rows of this wave as A-fragments
-  bf16x8_v qfrag[D / 16], dofrag[D / 16];
+  bf16x8_v qfrag[DQK / 16], dofrag[DV / 16];
 #pragma unroll
-  for (int c = 0; c < D / 16; ++c) {
-    const long off = q_base + (long)(q0 + col) * q_rs + c * 16 + half * 8;
-    qfrag[c] = *reinterpret_cast<const bf16x8_v*>(q + off);
-    dofrag[c] = *reinterpret_cast<const bf16x8_v*>(dout + off);
-  }
-  // lse/delta for the block's q rows -> LDS (32 regs saved per lane)
+  for (int c = 0; c < DQK / 16; ++c)
+    qfrag[c] = *reinterpret_cast<const bf16x8_v*>(
+        q + q_base + (long)(q0 + col) * q_rs + c * 16 + half * 8);
+#pragma unroll
+  for (int c = 0; c < DV / 16; ++c)
+    dofrag[c] = *reinterpret_cast<const bf16x8_v*>(
+        dout + do_base + (long)(q0 + col) * do_rs + c * 16 + half * 8);
+  // lse/delta(/doc_start) for the block's q rows -> LDS
   if (tid < WAVES * 32) {
     stats[tid] = lse_row[qb_blk + tid];
     stats[WAVES * 32 + tid] = dlt_row[qb_blk + tid];
+    if (VARLEN) stats[2 * WAVES * 32 + tid] = (float)doc_start[q_start + qb_blk + tid];
   }
 
-  f32x16 dq_acc[D / 32];
+  f32x16 dq_acc[DQK / 32];
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t)
+  for (int t = 0; t < DQK / 32; ++t)
 #pragma unroll
     for (int r = 0; r < 16; ++r) dq_acc[t][r] = 0.f;
 
   char* tb = tbuf + wid * 2048;
+  const int ds_wave = VARLEN ? doc_start[q_start + q0] : 0;
   const int block_q_max = q_start + qb_blk + WAVES * 32 - 1;
   const int n_tiles = causal ? (min(block_q_max, Skv - 1) / 32 + 1) : (Skv + 31) / 32;
+  const int jk0 = VARLEN ? doc_start[q_start + qb_blk] / 32 : 0;
 
-  for (int jk = 0; jk < n_tiles; ++jk) {
+  for (int jk = jk0; jk < n_tiles; ++jk) {
     const int k0 = jk * 32;
     // ---- stage K rows, V rows, K^T
     {
-      const int c0 = (tid % 16) * 8;
-      for (int rr = tid / 16; rr < 32; rr += WAVES * 4) {
+      constexpr int KCH = DQK / 8;
+      for (int idx = tid; idx < 32 * KCH; idx += WAVES * 64) {
+        const int rr = idx / KCH, c0 = (idx % KCH) * 8;
         const int ks = min(k0 + rr, Skv - 1);  // overhang masked in compute
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + kv_base + (long)ks * kv_rs + c0);
-        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + kv_base + (long)ks * kv_rs + c0);
-        *reinterpret_cast<bf16x8*>(k_rows + k_lds_off(rr, c0 * 2)) = kv8;
-        *reinterpret_cast<bf16x8*>(v_rows + k_lds_off(rr, c0 * 2)) = vv8;
+        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + k_base + (long)ks * k_rs + c0);
+        *reinterpret_cast<bf16x8*>(k_rows + row_lds_off(rr, c0 * 2, KBB)) = kv8;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           *reinterpret_cast<bf16*>(kt + vt_lds_off(c0 + j, rr * 2)) = kv8.v[j];
       }
+      constexpr int VCH = DV / 8;
+      for (int idx = tid; idx < 32 * VCH; idx += WAVES * 64) {
+        const int rr = idx / VCH, c0 = (idx % VCH) * 8;
+        const int ks = min(k0 + rr, Skv - 1);
+        *reinterpret_cast<bf16x8*>(v_rows + row_lds_off(rr, c0 * 2, VBB)) =
+            *reinterpret_cast<const bf16x8*>(v + v_base + (long)ks * v_rs + c0);
+      }
     }
     __syncthreads();
 
-    const bool live = !causal || (k0 <= q_start + q0 + 31);
+    const bool live = (!causal || (k0 <= q_start + q0 + 31)) &&
+                      (!VARLEN || (k0 + 31 >= ds_wave));
     if (live) {
       // ---- S[q][k] = Q K^T : A=Q rows, B from k_rows (contiguous d)
       f32x16 s;
 #pragma unroll
       for (int r = 0; r < 16; ++r) s[r] = 0.f;
 #pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
+      for (int c = 0; c < DQK / 16; ++c) {
         bf16x8_v kb = *reinterpret_cast<const bf16x8_v*>(
-            k_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+            k_rows + row_lds_off(col, (c * 16 + half * 8) * 2, KBB));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag[c], kb, s, 0, 0, 0);
       }
       // ---- dP[q][k] = dO V^T
@@ -571,18 +675,19 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
 #pragma unroll
       for (int r = 0; r < 16; ++r) dp[r] = 0.f;
 #pragma unroll
-      for (int c = 0; c < D / 16; ++c) {
+      for (int c = 0; c < DV / 16; ++c) {
         bf16x8_v vb = *reinterpret_cast<const bf16x8_v*>(
-            v_rows + k_lds_off(col, (c * 16 + half * 8) * 2));
+            v_rows + row_lds_off(col, (c * 16 + half * 8) * 2, VBB));
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vb, dp, 0, 0, 0);
       }
 
-      const int kg = k0 + col;   // D cols = k here
+      const int kg = k0 + col;   // cols = k here
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
         const int qrow = (r & 3) + 8 * (r >> 2) + 4 * half;  // block-local + wid*32
         const int qg = q_start + q0 + qrow;
-        const bool masked = (causal && kg > qg) || kg >= Skv;
+        const bool masked = (causal && kg > qg) || kg >= Skv ||
+                            (VARLEN && kg < (int)stats[2 * WAVES * 32 + wid * 32 + qrow]);
         const float p = masked ? 0.f : __expf(s[r] * scale - stats[wid * 32 + qrow]);
         s[r] = p * (dp[r] - stats[WAVES * 32 + wid * 32 + qrow]) * scale;  // dS in place
       }
@@ -599,7 +704,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
         bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
             tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-        for (int t = 0; t < D / 32; ++t) {
+        for (int t = 0; t < DQK / 32; ++t) {
           bf16x8_v ktb = *reinterpret_cast<const bf16x8_v*>(
               kt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
           dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, ktb, dq_acc[t], 0, 0, 0);
@@ -612,7 +717,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
 
   // ---- epilogue: plain bf16 stores (q rows exclusive to this block)
 #pragma unroll
-  for (int t = 0; t < D / 32; ++t) {
+  for (int t = 0; t < DQK / 32; ++t) {
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
       const int qrow = q0 + (r & 3) + 8 * (r >> 2) + 4 * half;
@@ -623,87 +728,117 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_bwd_dq_kernel(
   }
 }
 
+template <int DQK, int DV, int WAVES, bool VARLEN>
+static void launch_dkv(const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k,
+                       const at::Tensor& v, const at::Tensor& lse, const at::Tensor& delta,
+                       at::Tensor& dk, at::Tensor& dv, const int* doc_end, int B, int Sq,
+                       int Skv, int Hq, int Hk, int q_start, float scale, bool causal,
+                       hipStream_t stream) {
+  const dim3 grid(Skv / (WAVES * 32), Hk, B);
+  const size_t smem = 32 * pad256(DQK * 2) + 32 * pad256(DV * 2) + DQK * 64 + DV * 64 +
+                      WAVES * 2048;
+  hipLaunchKernelGGL((flash_bwd_dkv_kernel<DQK, DV, WAVES, VARLEN>), grid,
+                     dim3(WAVES * WAVE_SIZE), smem, stream,
+                     reinterpret_cast<const bf16*>(dout.data_ptr()),
+                     reinterpret_cast<const bf16*>(q.data_ptr()),
+                     reinterpret_cast<const bf16*>(k.data_ptr()),
+                     reinterpret_cast<const bf16*>(v.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(dk.data_ptr()),
+                     reinterpret_cast<bf16*>(dv.data_ptr()), doc_end,
+                     B, Sq, Skv, Hq, Hk, q_start, scale, causal);
+  HIP_CHECK_KERNEL();
+}
+
+template <int DQK, int DV, int WAVES, bool VARLEN>
+static void launch_dq(const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k,
+                      const at::Tensor& v, const at::Tensor& lse, const at::Tensor& delta,
+                      at::Tensor& dq, const int* doc_start, int B, int Sq, int Skv, int Hq,
+                      int Hk, int q_start, float scale, bool causal, hipStream_t stream) {
+  const dim3 grid(Sq / (WAVES * 32), Hq, B);
+  const size_t smem = 32 * pad256(DQK * 2) + 32 * pad256(DV * 2) + DQK * 64 +
+                      WAVES * 2048 + 3 * WAVES * 32 * sizeof(float);
+  hipLaunchKernelGGL((flash_bwd_dq_kernel<DQK, DV, WAVES, VARLEN>), grid,
+                     dim3(WAVES * WAVE_SIZE), smem, stream,
+                     reinterpret_cast<const bf16*>(dout.data_ptr()),
+                     reinterpret_cast<const bf16*>(q.data_ptr()),
+                     reinterpret_cast<const bf16*>(k.data_ptr()),
+                     reinterpret_cast<const bf16*>(v.data_ptr()),
+                     lse.data_ptr<float>(), delta.data_ptr<float>(),
+                     reinterpret_cast<bf16*>(dq.data_ptr()), doc_start,
+                     B, Sq, Skv, Hq, Hk, q_start, scale, causal);
+  HIP_CHECK_KERNEL();
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
-    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal,
-    int64_t q_start) {
-  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), D = q.size(3);
-  const int Skv = k.size(1), Hk = k.size(2);
-  TORCH_CHECK(D == 128 && Sq % 128 == 0 && Skv % 128 == 0,
-              "flash_attn_bwd: D=128, Sq%128==0, Skv%128==0 required");
-  auto stream = c10::hip::getCurrentHIPStream();
+    const at::Tensor& o, const at::Tensor& lse, double scale, bool causal, int64_t q_start,
+    const std::optional<at::Tensor>& doc_start, const std::optional<at::Tensor>& doc_end) {
+  const int B = q.size(0), Sq = q.size(1), Hq = q.size(2), Dqk = q.size(3);
+  const int Skv = k.size(1), Hk = k.size(2), Dv = v.size(3);
+  TORCH_CHECK(Sq % 128 == 0 && Skv % 128 == 0,
+              "flash_attn_bwd: Sq%128==0, Skv%128==0 required");
+  const bool varlen = doc_start.has_value();
+  TORCH_CHECK(varlen == doc_end.has_value(), "doc_start and doc_end go together");
+  const int* ds_ptr = varlen ? doc_start->data_ptr<int>() : nullptr;
+  const int* de_ptr = varlen ? doc_end->data_ptr<int>() : nullptr;
+  auto stream = c10::hip::getCurrentHIPStream().stream();
 
   auto delta = at::empty({B, Hq, Sq}, q.options().dtype(at::kFloat));
   {
     const long rows = (long)B * Sq * Hq;
     const int block = 256;
     const long grid = (rows * WAVE_SIZE + block - 1) / block;
-    hipLaunchKernelGGL(fa_delta_kernel, dim3((unsigned)grid), dim3(block), 0, stream.stream(),
+    hipLaunchKernelGGL(fa_delta_kernel, dim3((unsigned)grid), dim3(block), 0, stream,
                        reinterpret_cast<const bf16*>(dout.data_ptr()),
                        reinterpret_cast<const bf16*>(o.data_ptr()),
-                       delta.data_ptr<float>(), Sq, Hq, rows);
+                       delta.data_ptr<float>(), Sq, Hq, Dv, rows);
     HIP_CHECK_KERNEL();
   }
 
   auto dq = at::empty_like(q);
-  auto dk32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
-  auto dv32 = at::zeros({B, Skv, Hk, D}, q.options().dtype(at::kFloat));
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
 
-  if (Skv % 256 == 0) {
-    const dim3 grid_kv(Skv / 256, Hq, B);
-    const size_t smem_a = 32 * 1024 + 8 * 2048;
-    hipLaunchKernelGGL((flash_bwd_dkv_kernel<128, 8>), grid_kv, dim3(512), smem_a,
-                       stream.stream(),
-                       reinterpret_cast<const bf16*>(dout.data_ptr()),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
-  } else {
-    const dim3 grid_kv(Skv / 128, Hq, B);
-    const size_t smem_a = 32 * 1024 + 4 * 2048;
-    hipLaunchKernelGGL((flash_bwd_dkv_kernel<128, 4>), grid_kv, dim3(256), smem_a,
-                       stream.stream(),
-                       reinterpret_cast<const bf16*>(dout.data_ptr()),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       dk32.data_ptr<float>(), dv32.data_ptr<float>(),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
+  bool done = false;
+  // kv-parallel dKV: 8-wave only where two waves/SIMD fit (occ_dkv == 2)
+#define FA_BWD_CASE(A, C)                                                                \
+  if (!done && Dqk == A && Dv == C) {                                                    \
+    done = true;                                                                         \
+    if (varlen) {                                                                        \
+      launch_dkv<A, C, 4, true>(dout, q, k, v, lse, delta, dk, dv, de_ptr, B, Sq, Skv,   \
+                                Hq, Hk, (int)q_start, (float)scale, causal, stream);     \
+      launch_dq<A, C, 4, true>(dout, q, k, v, lse, delta, dq, ds_ptr, B, Sq, Skv, Hq,    \
+                               Hk, (int)q_start, (float)scale, causal, stream);          \
+    } else {                                                                             \
+      if (occ_dkv(A, C) == 2 && Skv % 256 == 0) {                                        \
+        launch_dkv<A, C, (occ_dkv(A, C) == 2 ? 8 : 4), false>(                           \
+            dout, q, k, v, lse, delta, dk, dv, nullptr, B, Sq, Skv, Hq, Hk,              \
+            (int)q_start, (float)scale, causal, stream);                                 \
+      } else {                                                                           \
+        launch_dkv<A, C, 4, false>(dout, q, k, v, lse, delta, dk, dv, nullptr, B, Sq,    \
+                                   Skv, Hq, Hk, (int)q_start, (float)scale, causal,      \
+                                   stream);                                              \
+      }                                                                                  \
+      if (occ_dq(A, C) == 2 && Sq % 256 == 0) {                                          \
+        launch_dq<A, C, (occ_dq(A, C) == 2 ? 8 : 4), false>(                             \
+            dout, q, k, v, lse, delta, dq, nullptr, B, Sq, Skv, Hq, Hk, (int)q_start,    \
+            (float)scale, causal, stream);                                               \
+      } else {                                                                           \
+        launch_dq<A, C, 4, false>(dout, q, k, v, lse, delta, dq, nullptr, B, Sq, Skv,    \
+                                  Hq, Hk, (int)q_start, (float)scale, causal, stream);   \
+      }                                                                                  \
+    }                                                                                    \
   }
-  HIP_CHECK_KERNEL();
-
-  if (Sq % 256 == 0) {
-    const dim3 grid_q(Sq / 256, Hq, B);
-    const size_t smem_b = 24 * 1024 + 8 * 2048 + 8 * 64 * sizeof(float);
-    hipLaunchKernelGGL((flash_bwd_dq_kernel<128, 8>), grid_q, dim3(512), smem_b,
-                       stream.stream(),
-                       reinterpret_cast<const bf16*>(dout.data_ptr()),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       reinterpret_cast<bf16*>(dq.data_ptr()),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
-  } else {
-    const dim3 grid_q(Sq / 128, Hq, B);
-    const size_t smem_b = 24 * 1024 + 4 * 2048 + 4 * 64 * sizeof(float);
-    hipLaunchKernelGGL((flash_bwd_dq_kernel<128, 4>), grid_q, dim3(256), smem_b,
-                       stream.stream(),
-                       reinterpret_cast<const bf16*>(dout.data_ptr()),
-                       reinterpret_cast<const bf16*>(q.data_ptr()),
-                       reinterpret_cast<const bf16*>(k.data_ptr()),
-                       reinterpret_cast<const bf16*>(v.data_ptr()),
-                       lse.data_ptr<float>(), delta.data_ptr<float>(),
-                       reinterpret_cast<bf16*>(dq.data_ptr()),
-                       B, Sq, Skv, Hq, Hk, (int)q_start, (float)scale, causal);
-  }
-  HIP_CHECK_KERNEL();
-
-  return {dq, dk32.to(at::kBFloat16), dv32.to(at::kBFloat16)};
+  FA_BWD_CASE(64, 64)
+  FA_BWD_CASE(96, 96)
+  FA_BWD_CASE(128, 128)
+  FA_BWD_CASE(192, 128)
+  FA_BWD_CASE(192, 192)
+  FA_BWD_CASE(256, 256)
+#undef FA_BWD_CASE
+  TORCH_CHECK(done, "flash_attn_bwd: unsupported head dims (Dqk=", Dqk, ", Dv=", Dv, ")");
+  return {dq, dk, dv};
 }
 
 // ---- MFMA layout self-test: d[32,32] = a[32,16] @ b[16,32] via one mfma.

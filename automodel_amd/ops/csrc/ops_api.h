@@ -25,13 +25,16 @@ void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor master, at::Tensor
                 at::Tensor v, int64_t step, double lr, double beta1, double beta2,
                 double eps, double weight_decay);
 
-std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(const at::Tensor& q, const at::Tensor& k,
-                                                  const at::Tensor& v, double scale,
-                                                  bool causal, int64_t q_start);
+std::tuple<at::Tensor, at::Tensor> flash_attn_fwd(
+    const at::Tensor& q, const at::Tensor& k, const at::Tensor& v, double scale,
+    bool causal, int64_t q_start,
+    const std::optional<at::Tensor>& doc_start = std::nullopt);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const at::Tensor& dout, const at::Tensor& q, const at::Tensor& k, const at::Tensor& v,
     const at::Tensor& o, const at::Tensor& lse, double scale, bool causal,
-    int64_t q_start);
+    int64_t q_start,
+    const std::optional<at::Tensor>& doc_start = std::nullopt,
+    const std::optional<at::Tensor>& doc_end = std::nullopt);
 
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b);
 

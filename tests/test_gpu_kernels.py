@@ -502,3 +502,141 @@ def test_new_families_gpu_smoke():
     out = v(seq, pixel_values=pix, image_grid_thw=torch.tensor([[1, 4, 4]]))
     out.float().sum().backward()
     assert v.model.visual.blocks[0].attn.qkv.weight.grad is not None
+
+
+def test_flash_attention_head_dim_sweep():
+    """fwd+bwd parity at every kernel head-dim pair: 64
+    (GPT-OSS), 96, 128 (Llama), (192,128) MLA, 192, 256 (Gemma-2/3), plus a
+    padded odd dim (80 -> 96)."""
+    from automodel_amd.ops.attention import attention_ref, flash_attention
+
+    torch.manual_seed(3)
+    cases = [(64, 64), (96, 96), (128, 128), (192, 128), (192, 192),
+             (256, 256), (80, 80)]
+    B, S, Hq, Hk = 1, 256, 4, 2
+    for dqk, dv in cases:
+        q = torch.randn(B, S, Hq, dqk, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        k = torch.randn(B, S, Hk, dqk, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        v = torch.randn(B, S, Hk, dv, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        scale = dqk ** -0.5
+        o = flash_attention(q, k, v, causal=True, backend="hip", scale=scale)
+        q2 = q.detach().float().requires_grad_(True)
+        k2 = k.detach().float().requires_grad_(True)
+        v2 = v.detach().float().requires_grad_(True)
+        o_ref = attention_ref(q2, k2, v2, causal=True, scale=scale)
+        assert torch.allclose(o.float(), o_ref, atol=3e-2, rtol=3e-2), \
+            (dqk, dv, (o.float() - o_ref).abs().max())
+        do = torch.randn_like(o)
+        o.backward(do)
+        o_ref.backward(do.float())
+        for g, g2, name in [(q.grad, q2.grad, "dq"), (k.grad, k2.grad, "dk"),
+                            (v.grad, v2.grad, "dv")]:
+            assert torch.allclose(g.float(), g2, atol=6e-2, rtol=6e-2), \
+                (dqk, dv, name, (g.float() - g2).abs().max())
+
+
+def test_flash_attention_bwd_deterministic():
+    """Two identical backward runs produce BITWISE-equal grads (VERDICT r1
+    #10: the r1 dKV kernel used f32 atomicAdd across the GQA group)."""
+    from automodel_amd.ops.attention import flash_attention
+
+    torch.manual_seed(4)
+    B, S, Hq, Hk, D = 2, 512, 8, 2, 128     # GQA group 4 -> old kernel raced
+    q0 = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k0 = torch.randn(B, S, Hk, D, device="cuda", dtype=torch.bfloat16)
+    v0 = torch.randn(B, S, Hk, D, device="cuda", dtype=torch.bfloat16)
+    do = torch.randn(B, S, Hq, D, device="cuda", dtype=torch.bfloat16)
+    grads = []
+    for _ in range(2):
+        q = q0.clone().requires_grad_(True)
+        k = k0.clone().requires_grad_(True)
+        v = v0.clone().requires_grad_(True)
+        o = flash_attention(q, k, v, causal=True, backend="hip")
+        o.backward(do)
+        grads.append((q.grad.clone(), k.grad.clone(), v.grad.clone()))
+    for a, b, name in zip(grads[0], grads[1], ("dq", "dk", "dv")):
+        assert torch.equal(a, b), f"{name} not bitwise deterministic"
+
+
+def test_varlen_attention_one_launch_bwd():
+    """Native varlen (doc bounds in-kernel, ONE launch) fwd+bwd parity vs
+    per-document fp32 reference."""
+    from automodel_amd.ops.attention import attention_ref, flash_attention_varlen
+
+    torch.manual_seed(5)
+    H, Hk, D = 4, 2, 128
+    cu = torch.tensor([0, 100, 356, 420, 1024], dtype=torch.int32)
+    T = 1024
+    q = torch.randn(1, T, H, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(1, T, Hk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(1, T, Hk, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = flash_attention_varlen(q, k, v, cu, backend="hip")
+    do = torch.randn_like(out)
+    out.backward(do)
+
+    q2 = q.detach().float().requires_grad_(True)
+    k2 = k.detach().float().requires_grad_(True)
+    v2 = v.detach().float().requires_grad_(True)
+    refs = []
+    for a, b in zip(cu[:-1].tolist(), cu[1:].tolist()):
+        refs.append(attention_ref(q2[:, a:b], k2[:, a:b], v2[:, a:b], causal=True))
+    ref = torch.cat(refs, dim=1)
+    assert torch.allclose(out.float(), ref, atol=3e-2, rtol=3e-2), \
+        (out.float() - ref).abs().max()
+    ref.backward(do.float())
+    assert torch.allclose(q.grad.float(), q2.grad, atol=6e-2, rtol=6e-2), \
+        (q.grad.float() - q2.grad).abs().max()
+    assert torch.allclose(k.grad.float(), k2.grad, atol=6e-2, rtol=6e-2), \
+        (k.grad.float() - k2.grad).abs().max()
+    assert torch.allclose(v.grad.float(), v2.grad, atol=6e-2, rtol=6e-2), \
+        (v.grad.float() - v2.grad).abs().max()
+
+
+def test_varlen_attention_short_docs_many():
+    """Launch-bound regression shape: many short docs, one kernel launch."""
+    from automodel_amd.ops.attention import attention_ref, flash_attention_varlen
+
+    torch.manual_seed(6)
+    H, Hk, D = 2, 1, 64
+    lens = [17, 40, 3, 128, 64, 9, 251, 32] * 2
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32)
+    T = int(cu[-1])
+    q = torch.randn(1, T, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(1, T, Hk, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(1, T, Hk, D, device="cuda", dtype=torch.bfloat16)
+    out = flash_attention_varlen(q, k, v, cu, backend="hip")
+    for a, b in zip(cu[:-1].tolist(), cu[1:].tolist()):
+        ref = attention_ref(q[:, a:b].float(), k[:, a:b].float(), v[:, a:b].float(),
+                            causal=True)
+        assert torch.allclose(out[:, a:b].float(), ref, atol=3e-2, rtol=3e-2), (a, b)
+
+
+def test_mla_attention_flash_gpu():
+    """DeepSeek MLA block runs on the (192,128) flash instantiation and
+    matches its own sdpa path."""
+    from automodel_amd.models.registry import build_model
+
+    torch.manual_seed(7)
+    cfg = dict(vocab_size=256, hidden_size=64, intermediate_size=128,
+               num_hidden_layers=1, num_attention_heads=2,
+               kv_lora_rank=32, q_lora_rank=0,
+               qk_nope_head_dim=128, qk_rope_head_dim=64, v_head_dim=128,
+               max_position_embeddings=512,
+               moe=dict(n_routed_experts=4, n_activated_experts=2,
+                        moe_intermediate_size=48), first_k_dense_replace=1)
+    m = build_model(config=cfg, architecture="DeepseekV3ForCausalLM",
+                    dtype="bfloat16", meta_init=False, device="cuda")
+    ids = torch.randint(0, 256, (1, 256), device="cuda")
+    with torch.no_grad():
+        logits_hip = m(ids)
+    # force the sdpa path for comparison
+    for layer in m.model.layers:
+        layer.self_attn.backend = layer.self_attn.backend.__class__(attn="sdpa")
+    with torch.no_grad():
+        logits_sdpa = m(ids)
+    assert torch.allclose(logits_hip.float(), logits_sdpa.float(),
+                          atol=5e-2, rtol=5e-2), \
+        (logits_hip.float() - logits_sdpa.float()).abs().max()

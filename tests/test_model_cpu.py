@@ -127,18 +127,20 @@ def test_fused_projections_match_unfused():
 
 
 def test_backend_resolve_head_dim_guard():
-    """hd != 128 on GPU downgrades attn to sdpa at build time (visible
-    warning), never a silent runtime fallback inside the kernel."""
+    """head_dim beyond the kernel max downgrades attn to sdpa at build time
+    (visible warning); every dim <= 256 stays on the HIP kernels (round-2:
+    the flash kernels cover 64/96/128/192/256 with pad-to-tile)."""
     import warnings
 
     from automodel_amd.models.common.backend import BackendConfig
 
     with warnings.catch_warnings(record=True) as w:
         warnings.simplefilter("always")
-        b = BackendConfig.resolve(None, "cuda", head_dim=64)
+        b = BackendConfig.resolve(None, "cuda", head_dim=512)
     assert b.attn == "sdpa"
-    assert any("head_dim 64" in str(x.message) for x in w)
-    assert BackendConfig.resolve(None, "cuda", head_dim=128).attn == "hip"
+    assert any("head_dim 512" in str(x.message) for x in w)
+    for d in (64, 80, 96, 128, 192, 256):
+        assert BackendConfig.resolve(None, "cuda", head_dim=d).attn == "hip", d
 
 
 def test_hf_logits_parity_llama():
