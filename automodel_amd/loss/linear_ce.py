@@ -4,12 +4,13 @@ MI355X-native equivalent of the reference's cut-cross-entropy path
 (nemo_automodel/components/loss/linear_ce.py:130-265): the [T, V] logits tensor
 (V≈128k for Llama-3 → 2 GB per microbatch at T=8k, bf16) is never stored.
 
-Two implementations behind one autograd Function interface:
-  * "hip_fused": csrc/fused_ce.hip — vocab-tiled MFMA GEMM + online softmax
-    fwd; bwd recomputes tile softmax and accumulates dH / dW.
-  * "chunked": token-chunked torch path (CPU tests + fallback) — forward under
-    no_grad per chunk, backward recomputes the chunk's softmax. Peak extra
-    memory is chunk_size × V.
+Three implementations behind one autograd Function interface:
+  * "hip_fused": csrc/fused_ce.hip — single-pass vocab-tiled MFMA stats GEMM
+    forward (no logits tensor at any point); backward recomputes chunked
+    through hipBLASLt + in-place HIP CE epilogue, reusing the forward's lse.
+  * "hybrid": chunked hipBLASLt GEMM + HIP CE-epilogue kernels fwd AND bwd
+    (one [chunk, V] bf16 buffer alive at a time).
+  * "chunked": token-chunked torch path (CPU tests + fallback).
 
 Both return the SUM of per-token losses over labels != ignore_index.
 """
@@ -111,18 +112,39 @@ class _HybridLinearCE(torch.autograd.Function):
 
 
 class _FusedLinearCEHip(torch.autograd.Function):
+    """Single-pass fused forward (csrc/fused_ce.hip): vocab-tiled MFMA stats
+    GEMM — NO [T, V] logits tensor is ever materialized, only [nV, T]
+    online-softmax partials (512 B of logits per tile-row become 8 B).
+    Backward recomputes logits chunked through hipBLASLt + the in-place HIP
+    CE epilogue, reusing the forward's lse (so backward skips the lse pass
+    the hybrid forward needs). Deterministic end to end (no atomics)."""
+
     @staticmethod
-    def forward(ctx, hidden, weight, labels):
+    def forward(ctx, hidden, weight, labels, chunk_size):
         hidden = hidden.contiguous()
         loss, lse = hip_ops().fused_ce_fwd(hidden, weight, labels)
         ctx.save_for_backward(hidden, weight, labels, lse)
+        ctx.chunk_size = chunk_size
         return loss
 
     @staticmethod
     def backward(ctx, dloss):
-        hidden, weight, labels, lse = ctx.saved_tensors
-        dh, dw = hip_ops().fused_ce_bwd(hidden, weight, labels, lse, dloss)
-        return dh, dw, None
+        ops = hip_ops()
+        hidden, weight, labels, lse_all = ctx.saved_tensors
+        chunk_size = ctx.chunk_size
+        T = hidden.shape[0]
+        dh = torch.empty_like(hidden)
+        dw = torch.zeros_like(weight, dtype=torch.float32)
+        d = dloss.reshape(1).float().contiguous()
+        wt = weight.t()
+        for s in range(0, T, chunk_size):
+            h = hidden[s : s + chunk_size]
+            y = labels[s : s + chunk_size]
+            logits = h @ wt                       # recompute (bf16 GEMM)
+            ops.ce_bwd_logits(logits, y, lse_all[s : s + chunk_size], d, 0)
+            dh[s : s + chunk_size] = logits @ weight
+            dw += (logits.t() @ h).float()
+        return dh, dw.to(weight.dtype), None, None
 
 
 def fused_linear_cross_entropy(
@@ -135,8 +157,9 @@ def fused_linear_cross_entropy(
     """hidden [*, H] -> flattened [T, H]; labels [*] -> [T]. Returns loss SUM."""
     hidden = hidden.reshape(-1, hidden.shape[-1])
     labels = labels.reshape(-1)
-    if hidden.is_cuda and backend in ("hybrid", "hip_fused"):
-        # hip_fused (single-kernel GEMM+CE) falls back to hybrid until built
+    if hidden.is_cuda and backend == "hip_fused":
+        return _FusedLinearCEHip.apply(hidden, weight, labels.to(torch.long), chunk_size)
+    if hidden.is_cuda and backend == "hybrid":
         return _HybridLinearCE.apply(hidden, weight, labels, chunk_size)
     return _ChunkedLinearCE.apply(hidden, weight, labels, chunk_size)
 

@@ -319,5 +319,7 @@ def build_model(
         # build never carries uninitialized memory.
         if hasattr(model, "init_weights"):
             model.init_weights()
+        if device is not None and str(device) != "meta":
+            model = model.to(device)
     model = model.to(dtype=torch_dtype)
     return model

@@ -51,10 +51,6 @@ TORCH_LIBRARY(amd_ops, m) {
 
   m.def("fused_ce_fwd(Tensor hidden, Tensor weight, Tensor labels) -> (Tensor, Tensor)");
   m.impl("fused_ce_fwd", &amd_ops::fused_ce_fwd);
-  m.def(
-      "fused_ce_bwd(Tensor hidden, Tensor weight, Tensor labels, Tensor lse, Tensor dloss)"
-      " -> (Tensor, Tensor)");
-  m.impl("fused_ce_bwd", &amd_ops::fused_ce_bwd);
 
   m.def("soft_ce_fwd(Tensor s, Tensor t) -> (Tensor, Tensor, Tensor)");
   m.impl("soft_ce_fwd", &amd_ops::soft_ce_fwd);

@@ -48,11 +48,6 @@ std::tuple<at::Tensor, at::Tensor> ce_stats_logits(const at::Tensor& logits);
 std::tuple<at::Tensor, at::Tensor> fused_ce_fwd(const at::Tensor& hidden,
                                                 const at::Tensor& weight,
                                                 const at::Tensor& labels);
-std::tuple<at::Tensor, at::Tensor> fused_ce_bwd(const at::Tensor& hidden,
-                                                const at::Tensor& weight,
-                                                const at::Tensor& labels,
-                                                const at::Tensor& lse,
-                                                const at::Tensor& dloss);
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> soft_ce_fwd(const at::Tensor& s,
                                                            const at::Tensor& t);

@@ -127,8 +127,19 @@ def main():
         return float(loss.detach()) / n_tok
 
     n_steps = args.profile_steps if args.profile_steps > 0 else args.steps
+    loss_step0 = None
     for i in range(args.warmup):
-        step(i)
+        l = step(i)
+        if i == 0:
+            # Sanity-anchor the trajectory: random-init model on random labels
+            # must start at ~ln(V). A broken loss path (under-computation,
+            # cached outputs) would show up here, so throughput numbers can't
+            # be confused with skipped work.
+            loss_step0 = l
+            import math as _math
+            lnv = _math.log(cfg.vocab_size)
+            assert 0.5 * lnv < l < 1.5 * lnv, \
+                f"step-0 loss {l:.3f} not near ln(V)={lnv:.3f} — loss path broken"
     if use_cuda:
         torch.cuda.synchronize()
     if dist.is_initialized():
@@ -182,6 +193,9 @@ def main():
                 "mfu": round(achieved_mfu, 4) if achieved_mfu is not None else None,
                 "flops_per_token": fpt,
                 "loss_per_token": last_loss,
+                "loss_step0": round(loss_step0, 4) if loss_step0 is not None else None,
+                "note_loss": "8 reused synthetic batches memorize quickly; "
+                             "step0 is asserted at ~ln(V)",
                 "attn_backend": args.attn,
                 "loss_backend": args.loss,
             },

@@ -628,7 +628,7 @@ def test_mla_attention_flash_gpu():
                moe=dict(n_routed_experts=4, n_activated_experts=2,
                         moe_intermediate_size=48), first_k_dense_replace=1)
     m = build_model(config=cfg, architecture="DeepseekV3ForCausalLM",
-                    dtype="bfloat16", meta_init=False, device="cuda")
+                    dtype="bfloat16", meta_init=False).to("cuda")
     ids = torch.randint(0, 256, (1, 256), device="cuda")
     with torch.no_grad():
         logits_hip = m(ids)
