@@ -443,6 +443,18 @@ class LlamaFusedStateDictAdapter:
     def __init__(self, config: LlamaConfig):
         self.cfg = config
 
+    def hf_key_targets(self, key: str) -> list[str]:
+        """Internal param name(s) an HF key feeds (streaming HF loader)."""
+        import re
+
+        m = re.match(r"^(model\.layers\.\d+)\.self_attn\.[qkv]_proj\.(weight|bias)$", key)
+        if m and self.cfg.fused_qkv:
+            return [f"{m.group(1)}.self_attn.qkv_proj.{m.group(2)}"]
+        m = re.match(r"^(model\.layers\.\d+)\.mlp\.(?:gate|up)_proj\.weight$", key)
+        if m and self.cfg.fused_gate_up:
+            return [f"{m.group(1)}.mlp.gate_up_proj.weight"]
+        return [key]
+
     def from_hf(self, sd: dict) -> dict:
         import re
 

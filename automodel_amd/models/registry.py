@@ -291,7 +291,17 @@ def build_model(
         with open(os.path.join(pretrained_path, "config.json")) as f:
             hf_cfg = json.load(f)
         architecture = architecture or hf_cfg.get("architectures", ["LlamaForCausalLM"])[0]
-        cls = get_model_class(architecture)
+        try:
+            cls = get_model_class(architecture)
+        except KeyError:
+            # generic HF-transformers fallback: any architecture without an
+            # in-tree implementation still trains (reference
+            # auto_model.py:380-643)
+            from automodel_amd.models.hf_fallback import build_hf_fallback
+
+            return build_hf_fallback(pretrained_path=pretrained_path,
+                                     architecture=architecture, dtype=dtype,
+                                     device=device)
         model_cfg = cls.config_from_hf(hf_cfg) if hasattr(cls, "config_from_hf") else None
         if model_cfg is None:
             from automodel_amd.models.llama.model import LlamaConfig
@@ -299,7 +309,13 @@ def build_model(
             model_cfg = LlamaConfig.from_hf_config(hf_cfg)
     else:
         assert config is not None, "build_model needs config= or pretrained_path="
-        cls = get_model_class(architecture or "LlamaForCausalLM")
+        try:
+            cls = get_model_class(architecture or "LlamaForCausalLM")
+        except KeyError:
+            from automodel_amd.models.hf_fallback import build_hf_fallback
+
+            return build_hf_fallback(config=dict(config), architecture=architecture,
+                                     dtype=dtype, device=device)
         cfg_cls = getattr(cls, "config_class", None)
         if cfg_cls is None:
             from automodel_amd.models.llama.model import LlamaConfig

@@ -30,6 +30,31 @@ class MoEStateDictAdapter:
         self.flavor = getattr(config, "hf_flavor", "qwen3_moe")
         self.n_experts = config.moe.n_routed_experts
 
+    def hf_key_targets(self, key: str) -> list[str]:
+        """Internal param name(s) an HF key feeds — lets the streaming HF
+        loader (checkpoint/hf_loader.py) free consumed shard keys."""
+        m = re.match(r"^(model\.layers\.\d+)\.(?:mlp|block_sparse_moe)"
+                     r"\.experts\.(\d+)\.(\w+)\.weight$", key)
+        if m:
+            proj = _MIXTRAL_MAP.get(m.group(3), m.group(3))
+            return [f"{m.group(1)}.mlp.experts.{proj}"]
+        if key.endswith("experts.gate_up_proj"):
+            base = key.replace(".block_sparse_moe.", ".mlp.")
+            return [base.replace("gate_up_proj", "gate_proj"),
+                    base.replace("gate_up_proj", "up_proj")]
+        if key.endswith(".gate.moe_statics.e_score_correction_bias"):
+            return [key.replace(".gate.moe_statics.e_score_correction_bias",
+                                ".gate.e_score_correction_bias")]
+        if ".mlp.shared_expert." in key:
+            return [key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")]
+        if ".block_sparse_moe.router." in key:
+            return [key.replace(".block_sparse_moe.router.", ".mlp.gate.")]
+        if ".block_sparse_moe.gate." in key:
+            return [key.replace(".block_sparse_moe.gate.", ".mlp.gate.")]
+        if ".block_sparse_moe." in key:
+            return [key.replace(".block_sparse_moe.", ".mlp.")]
+        return [key]
+
     # ---- HF -> internal (stack per-expert tensors)
     def from_hf(self, sd: dict[str, torch.Tensor]) -> dict[str, torch.Tensor]:
         out: dict[str, torch.Tensor] = {}

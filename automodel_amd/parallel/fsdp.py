@@ -19,6 +19,21 @@ from torch.distributed.device_mesh import DeviceMesh
 from torch.distributed.fsdp import MixedPrecisionPolicy, fully_shard
 
 
+def detect_decoder_layers(model: nn.Module) -> list[nn.Module]:
+    """Generic transformer-block detection for models whose layer class is
+    not in the known list (the HF-transformers fallback path): pick the
+    largest nn.ModuleList whose children all share one type — that is the
+    decoder stack in every HF causal LM (reference parallelizer.py's
+    per-model strategies fall back to the same per-layer granularity)."""
+    best: list[nn.Module] = []
+    for m in model.modules():
+        if isinstance(m, nn.ModuleList) and len(m) >= 2:
+            kinds = {type(c) for c in m}
+            if len(kinds) == 1 and len(m) > len(best):
+                best = list(m)
+    return best
+
+
 def apply_fsdp(
     model: nn.Module,
     mesh: DeviceMesh,
@@ -42,6 +57,8 @@ def apply_fsdp(
     layers = [
         m for m in model.modules() if type(m).__name__ in layer_cls_names
     ]
+    if not layers:
+        layers = detect_decoder_layers(model)
     for layer in layers:
         fully_shard(layer, mesh=mesh, mp_policy=mp, reshard_after_forward=reshard_after_forward)
     fully_shard(model, mesh=mesh, mp_policy=mp, reshard_after_forward=reshard_after_forward)
