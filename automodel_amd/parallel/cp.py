@@ -159,9 +159,14 @@ def cp_blockdiag_attention(q, k, v, cu_seqlens: torch.Tensor,
     """Packed-document-correct CP attention (reference blockdiag_cp/
     exchange.py all-gather strategy + runtime.py cp_blockdiag_sdpa): each
     local (zigzag-sharded) q row attends only within its document AND
-    causally, against the all-gathered global K/V. Mask is built from the
-    GLOBAL cu_seqlens, so document boundaries are exact regardless of how
-    the shard cuts documents."""
+    causally, against the all-gathered global K/V. Doc boundaries come from
+    the GLOBAL cu_seqlens, so they are exact regardless of how the shard
+    cuts documents.
+
+    GPU path: the native varlen flash kernel with per-token doc bounds and
+    ``q_start`` — O(T) metadata, no dense mask, one launch per chunk
+    (round-1 materialized an O(S_local x T) mask).
+    CPU path keeps the dense-mask sdpa reference (parity tests)."""
     import math
 
     cp = _ACTIVE_CP
@@ -173,6 +178,18 @@ def cp_blockdiag_attention(q, k, v, cu_seqlens: torch.Tensor,
     vg = _GatherSeqZigzag.apply(v, cp)
     T = kg.shape[1]
     dev = q.device
+
+    if q.is_cuda and B == 1 and C % 128 == 0 and T % 32 == 0:
+        from automodel_amd.ops.attention import _flash_hip, doc_bounds_from_cu
+
+        ds, de = doc_bounds_from_cu(cu_seqlens.to(dev), T)
+        g0, g1 = zigzag_chunk_ids(cp.rank, cp.world)
+        outs = []
+        for slot, g in ((0, g0), (1, g1)):
+            qc = q.narrow(1, slot * C, C)
+            outs.append(_flash_hip(qc, kg, vg, True, scale, g * C, ds, de))
+        return torch.cat(outs, dim=1)
+
     bounds = cu_seqlens.to(dev)[1:-1]
     doc_all = torch.bucketize(torch.arange(T, device=dev), bounds, right=True)
     gpos = local_global_positions(cp.rank, cp.world, C, device=dev)
