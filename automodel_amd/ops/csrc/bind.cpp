@@ -65,6 +65,10 @@ TORCH_LIBRARY(amd_ops, m) {
 
   m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
   m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
+  m.def("grouped_gemm_nn(Tensor g, Tensor w, Tensor offs, Tensor tile_map) -> Tensor");
+  m.impl("grouped_gemm_nn", &amd_ops::grouped_gemm_nn);
+  m.def("grouped_gemm_tn(Tensor g, Tensor x, Tensor offs, int E) -> Tensor");
+  m.impl("grouped_gemm_tn", &amd_ops::grouped_gemm_tn);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");
   m.impl("permute_gather", &amd_ops::permute_gather);
   m.def("unpermute_combine(Tensor yp, Tensor pos, Tensor probs) -> Tensor");

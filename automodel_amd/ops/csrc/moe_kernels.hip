@@ -154,6 +154,249 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
   return y;
 }
 
+// ===========================================================================
+// Grouped GEMM backward (the round-1 backward was a
+// per-expert hipBLASLt loop — launch-bound at DeepSeek-style expert counts).
+//   grouped_gemm_nn (dx): dx[m, k] = sum_n g[m, n] * w[e(m), n, k]
+//     same tile structure as the forward, with the W tile staged TRANSPOSED
+//     in LDS (wt[k][n]) so the B-fragment (8 consecutive n at fixed k) is a
+//     contiguous ds_read_b128.
+//   grouped_gemm_tn (dw): dw[e, n, k] = sum_{m in group e} g[m, n] * x[m, k]
+//     grid (E, N/128, K/128); the m-loop stays inside the block so each dw
+//     tile is written once — deterministic, no atomics.
+// ===========================================================================
+
+__global__ __launch_bounds__(256) void grouped_gemm_nn_kernel(
+    const bf16* __restrict__ g, const bf16* __restrict__ w, bf16* __restrict__ dx,
+    const int* __restrict__ tile_map, const int* __restrict__ offs, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* ga = smem;                       // [128 m][64 n]  16 KiB
+  char* wt = smem + GG_BM * GG_BK * 2;   // [128 k][64 n]  16 KiB
+
+  const int e = tile_map[2 * blockIdx.x];
+  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int m_end = offs[e + 1];
+  const int k0 = blockIdx.y * GG_BN;     // output-k tile
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;
+
+  const long wbase = (long)e * N * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  for (int n0 = 0; n0 < N; n0 += GG_BK) {
+    {
+      // g tile [128 m][64 n]: 128 rows x 8 vec-chunks = 1024 vectors
+      const int row0 = tid / 8;
+      const int c0 = (tid % 8) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + rr * 32;
+        const int m = m0 + row;
+        bf16x8 gv;
+        if (m < m_end) {
+          gv = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) gv.v[j] = f2bf(0.f);
+        }
+        *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = gv;
+      }
+      // w tile [64 n rows][128 k]: read w[n][k-chunk], scatter into wt[k][n]
+      const int nrow0 = tid / 16;
+      const int kc0 = (tid % 16) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int nrow = nrow0 + rr * 16;
+        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+            w + wbase + (long)(n0 + nrow) * K + k0 + kc0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<bf16*>(wt + gg_off(kc0 + j, nrow * 2)) = wv.v[j];
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < GG_BK / 32; ++kk) {
+      bf16x8v a[4], b[4];
+      const int arow = (wid >> 1) * 64;
+      const int brow = (wid & 1) * 64;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        a[i] = *reinterpret_cast<const bf16x8v*>(
+            ga + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+        b[i] = *reinterpret_cast<const bf16x8v*>(
+            wt + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int mw = m0 + (wid >> 1) * 64;
+  const int kw = k0 + (wid & 1) * 64;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mw + i * 16 + kq * 4 + r;
+      if (m < m_end) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          dx[(long)m * K + kw + j * 16 + l16] = f2bf(acc[i][j][r]);
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
+    const bf16* __restrict__ g, const bf16* __restrict__ x, bf16* __restrict__ dw,
+    const int* __restrict__ offs, int N, int K) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* gt = smem;                       // [128 n][64 m]  16 KiB
+  char* xt = smem + GG_BM * GG_BK * 2;   // [128 k][64 m]  16 KiB
+
+  const int e = blockIdx.x;
+  const int n0 = blockIdx.y * GG_BN;
+  const int k0 = blockIdx.z * GG_BN;
+  const int m_start = offs[e], m_end = offs[e + 1];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  for (int mc = m_start; mc < m_end; mc += GG_BK) {
+    {
+      // stage 64 m-rows of g[:, n0:n0+128] and x[:, k0:k0+128], transposed:
+      // 256 threads x (64 m x 16 vec-chunks / 256) = 4 vectors each per tensor
+      const int mrow0 = tid / 16;          // 16 rows per pass
+      const int c0 = (tid % 16) * 8;       // 16 chunks of 8 cover 128 cols
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int mrow = mrow0 + rr * 16;
+        const int m = mc + mrow;
+        bf16x8 gv, xv;
+        if (m < m_end) {
+          gv = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
+          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) { gv.v[j] = f2bf(0.f); xv.v[j] = f2bf(0.f); }
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          *reinterpret_cast<bf16*>(gt + gg_off(c0 + j, mrow * 2)) = gv.v[j];
+          *reinterpret_cast<bf16*>(xt + gg_off(c0 + j, mrow * 2)) = xv.v[j];
+        }
+      }
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < GG_BK / 32; ++kk) {
+      bf16x8v a[4], b[4];
+      const int arow = (wid >> 1) * 64;    // n sub-tile
+      const int brow = (wid & 1) * 64;     // k sub-tile
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        a[i] = *reinterpret_cast<const bf16x8v*>(
+            gt + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+        b[i] = *reinterpret_cast<const bf16x8v*>(
+            xt + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const long wbase = (long)e * N * K;
+  const int nw = n0 + (wid >> 1) * 64;
+  const int kw = k0 + (wid & 1) * 64;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int n = nw + i * 16 + kq * 4 + r;
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        dw[wbase + (long)n * K + kw + j * 16 + l16] = f2bf(acc[i][j][r]);
+    }
+  }
+}
+
+at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
+                           const at::Tensor& offs, const at::Tensor& tile_map) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 2 && g.scalar_type() == at::kBFloat16,
+              "grouped_gemm_nn: g [M,N] bf16");
+  TORCH_CHECK(w.dim() == 3 && w.scalar_type() == at::kBFloat16, "w [E,N,K] bf16");
+  const long M = g.size(0);
+  const int N = g.size(1), K = w.size(2);
+  TORCH_CHECK(N == w.size(1), "N mismatch");
+  TORCH_CHECK(N % GG_BK == 0 && K % GG_BN == 0, "need N%64==0, K%128==0");
+  auto dx = at::empty({M, (long)K}, g.options());
+  const int n_mtiles = tile_map.size(0);
+  if (n_mtiles == 0 || M == 0) return dx;
+  const dim3 grid(n_mtiles, K / GG_BN);
+  const size_t smem = 2 * GG_BM * GG_BK * 2;
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(grouped_gemm_nn_kernel, grid, dim3(256), smem, stream.stream(),
+                     reinterpret_cast<const bf16*>(g.data_ptr()),
+                     reinterpret_cast<const bf16*>(w.data_ptr()),
+                     reinterpret_cast<bf16*>(dx.data_ptr()),
+                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), N, K);
+  HIP_CHECK_KERNEL();
+  return dx;
+}
+
+at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
+                           const at::Tensor& offs, int64_t E) {
+  TORCH_CHECK(g.is_cuda() && g.dim() == 2 && x.dim() == 2 &&
+                  g.scalar_type() == at::kBFloat16,
+              "grouped_gemm_tn: g [M,N], x [M,K] bf16");
+  const int N = g.size(1), K = x.size(1);
+  TORCH_CHECK(N % GG_BN == 0 && K % GG_BN == 0, "need N%128==0, K%128==0");
+  auto dw = at::empty({E, (long)N, (long)K}, g.options());
+  const dim3 grid((unsigned)E, N / GG_BN, K / GG_BN);
+  const size_t smem = 2 * GG_BM * GG_BK * 2;
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(grouped_gemm_tn_kernel, grid, dim3(256), smem, stream.stream(),
+                     reinterpret_cast<const bf16*>(g.data_ptr()),
+                     reinterpret_cast<const bf16*>(x.data_ptr()),
+                     reinterpret_cast<bf16*>(dw.data_ptr()),
+                     offs.data_ptr<int>(), N, K);
+  HIP_CHECK_KERNEL();
+  return dw;
+}
+
 // ---- fused permute / unpermute -------------------------------------------
 // gather: y[i, :] = x[src[i], :]
 __global__ void permute_gather_kernel(const bf16* __restrict__ x, bf16* __restrict__ y,

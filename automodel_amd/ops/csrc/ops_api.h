@@ -63,6 +63,10 @@ at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
 
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map);
+at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
+                           const at::Tensor& offs, const at::Tensor& tile_map);
+at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
+                           const at::Tensor& offs, int64_t E);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);
 at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,
                              const at::Tensor& probs);

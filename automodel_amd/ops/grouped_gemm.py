@@ -48,11 +48,13 @@ class _GroupedLinear(torch.autograd.Function):
             x.is_cuda and x.dtype == torch.bfloat16
             and w.shape[2] % 64 == 0 and w.shape[1] % 128 == 0
         )
+        ctx.use_hip = use_hip
         if use_hip:
             tile_map, offs = _build_tile_map(list(counts))
+            ctx.tile_offs = (tile_map.to(x.device), offs.to(x.device))
             return hip_ops().grouped_gemm_nt(
                 x.contiguous(), w.contiguous(),
-                offs.to(x.device), tile_map.to(x.device),
+                ctx.tile_offs[1], ctx.tile_offs[0],
             )
         return _loop_gemm_nt(x, w, counts)
 
@@ -60,10 +62,19 @@ class _GroupedLinear(torch.autograd.Function):
     def backward(ctx, g):
         x, w = ctx.saved_tensors
         counts = ctx.counts
+        g = g.contiguous()
+        # single-kernel grouped backward (the per-expert
+        # hipBLASLt loop is launch-bound at 128+ small experts); dw's TN
+        # kernel needs N%128 and K%128
+        if ctx.use_hip and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0:
+            tile_map, offs = ctx.tile_offs
+            ops = hip_ops()
+            dx = ops.grouped_gemm_nn(g, w.contiguous(), offs, tile_map)
+            dw = ops.grouped_gemm_tn(g, x.contiguous(), offs, w.shape[0])
+            return dx, dw, None
         dx = torch.empty_like(x)
         dw = torch.zeros_like(w)
         start = 0
-        g = g.contiguous()
         for e, n in enumerate(counts):
             if n:
                 ge = g[start : start + n]

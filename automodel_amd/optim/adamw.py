@@ -31,8 +31,13 @@ class FusedAdamW(torch.optim.Optimizer):
         betas: tuple[float, float] = (0.9, 0.999),
         eps: float = 1e-8,
         weight_decay: float = 0.01,
+        state_dtype: torch.dtype = torch.float32,
     ):
+        # state_dtype=torch.bfloat16 halves optimizer memory (m/v bf16, no
+        # fp32 master) for single-GPU benches of 30B-class models; multi-GPU
+        # training keeps the default fp32 states (sharded by FSDP2)
         defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        self.state_dtype = state_dtype
         super().__init__(params, defaults)
 
     @torch.no_grad()
@@ -51,11 +56,13 @@ class FusedAdamW(torch.optim.Optimizer):
                 p = _local(p_)
                 grad = _local(p_.grad)
                 state = self.state[p_]
-                use_hip = p.is_cuda and p.dtype == torch.bfloat16 and p.numel() % 4 == 0
+                use_hip = (p.is_cuda and p.dtype == torch.bfloat16
+                           and p.numel() % 4 == 0
+                           and self.state_dtype == torch.float32)
                 if len(state) == 0:
                     state["step"] = 0
-                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
-                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg"] = torch.zeros_like(p, dtype=self.state_dtype)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=self.state_dtype)
                     if use_hip:
                         state["master"] = p.detach().float().clone()
                 state["step"] += 1

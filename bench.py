@@ -32,7 +32,8 @@ def parse_args():
     p.add_argument("--mbs", type=int, default=8, help="micro-batch size per GPU")
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3_8b",
-                   choices=["llama3_8b", "llama3_1b_proxy", "tiny_proxy"])
+                   choices=["llama3_8b", "llama3_1b_proxy", "tiny_proxy",
+                            "qwen3_moe_30b", "moe_tiny_proxy"])
     p.add_argument("--attn", type=str, default="hip")
     p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--loss-chunk", type=int, default=4096)
@@ -63,7 +64,24 @@ MODEL_CONFIGS = {
         head_dim=128,
         max_position_embeddings=8192, rope_theta=500000.0, rms_norm_eps=1e-5,
     ),
+    # Qwen3-30B-A3B (BASELINE.md row 7 anchor: 12,040 tok/s/GPU at 8xH100)
+    "qwen3_moe_30b": dict(
+        vocab_size=151936, hidden_size=2048, intermediate_size=6144,
+        num_hidden_layers=48, num_attention_heads=32, num_key_value_heads=4,
+        head_dim=128, max_position_embeddings=8192, rope_theta=1000000.0,
+        rms_norm_eps=1e-6,
+        moe=dict(n_routed_experts=128, n_activated_experts=8,
+                 moe_intermediate_size=768),
+    ),
+    "moe_tiny_proxy": dict(
+        vocab_size=512, hidden_size=256, intermediate_size=512,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=2,
+        head_dim=128, max_position_embeddings=512, rope_theta=10000.0,
+        moe=dict(n_routed_experts=4, n_activated_experts=2,
+                 moe_intermediate_size=128),
+    ),
 }
+MOE_MODELS = {"qwen3_moe_30b", "moe_tiny_proxy"}
 
 
 def main():
@@ -84,25 +102,45 @@ def main():
     device = torch.device("cuda") if use_cuda else torch.device("cpu")
     mesh = build_mesh(dp_shard=-1)
 
-    cfg = LlamaConfig(**MODEL_CONFIGS[args.model],
-                      fused_qkv=not args.no_fused_proj,
-                      fused_gate_up=not args.no_fused_proj)
+    is_moe = args.model in MOE_MODELS
     backend = BackendConfig(attn=args.attn, loss=args.loss)
     torch.manual_seed(1234 + rank)
-    with torch.device("meta"):
-        model = LlamaForCausalLM(cfg, backend=backend)
-    model = model.to(dtype=torch.bfloat16)
+    if is_moe:
+        from automodel_amd.models.registry import build_model
+
+        model = build_model(config=MODEL_CONFIGS[args.model],
+                            architecture="Qwen3MoeForCausalLM",
+                            backend={"attn": args.attn, "loss": args.loss},
+                            dtype="bfloat16", meta_init=True)
+        cfg = model.config
+    else:
+        cfg = LlamaConfig(**MODEL_CONFIGS[args.model],
+                          fused_qkv=not args.no_fused_proj,
+                          fused_gate_up=not args.no_fused_proj)
+        with torch.device("meta"):
+            model = LlamaForCausalLM(cfg, backend=backend)
+        model = model.to(dtype=torch.bfloat16)
     model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=args.loss_chunk)
     if args.fp8:
         from automodel_amd.quantization.fp8 import apply_fp8_to_model
         n = apply_fp8_to_model(model)
         if rank == 0:
             print(f"fp8: swapped {n} linears")
+    if is_moe and args.model == "qwen3_moe_30b":
+        # 30B at one GPU needs activation checkpointing; 8-GPU runs keep it
+        # for parity across N (weak scaling holds per-GPU work fixed)
+        from automodel_amd.parallel.activation_checkpointing import apply_ac
+        apply_ac(model, mode="full")
     if world > 1:
         apply_fsdp(model, mesh["dp_shard"], reshard_after_forward=False)
     model.init_weights(device=device)
     model.train()
-    opt = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.0)
+    # 30B-class single-GPU: bf16 optimizer states (fp32 master+m+v for 30.5B
+    # params is 366 GB > 288 GB HBM); n>=2 shards fp32 states via FSDP2
+    state_dtype = torch.bfloat16 if (is_moe and args.model == "qwen3_moe_30b"
+                                     and world == 1) else torch.float32
+    opt = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.0,
+                     state_dtype=state_dtype)
 
     # proxy configs have short contexts; clamp so --model tiny_proxy works
     # without an explicit --seq-len (driver default path is unaffected)
@@ -163,17 +201,31 @@ def main():
 
     tokens_per_step = args.mbs * args.seq_len * world
     tps = tokens_per_step * n_steps / elapsed
-    fpt = llama_flops_per_token(
-        cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
-        cfg.vocab_size, args.seq_len, cfg.num_attention_heads,
-        cfg.num_key_value_heads, cfg.head_dim,
-    )
+    if is_moe:
+        from automodel_amd.utils.flops import moe_flops_per_token
+
+        fpt = moe_flops_per_token(
+            cfg.hidden_size, cfg.num_hidden_layers, cfg.vocab_size,
+            args.seq_len, cfg.num_attention_heads, cfg.num_key_value_heads,
+            cfg.moe.moe_intermediate_size, cfg.moe.n_activated_experts,
+            head_dim=cfg.head_dim,
+        )
+        baseline_tok_per_gpu = 12040.0   # BASELINE.md row 7 (8xH100, GBS 512)
+    else:
+        fpt = llama_flops_per_token(
+            cfg.hidden_size, cfg.intermediate_size, cfg.num_hidden_layers,
+            cfg.vocab_size, args.seq_len, cfg.num_attention_heads,
+            cfg.num_key_value_heads, cfg.head_dim,
+        )
+        baseline_tok_per_gpu = 12472.87
     achieved_mfu = mfu(tps / world, fpt) if use_cuda else None
-    baseline_tok_per_gpu = 12472.87
 
     if rank == 0 and args.profile_steps == 0:
         print(json.dumps({
-            "metric": "tokens/sec (whole node) + MFU, Llama-3-8B SFT FSDP2 at 1/2/4/8 MI355X",
+            "metric": ("tokens/sec (whole node) + MFU, Qwen3-MoE-30B-A3B SFT "
+                       "FSDP2+grouped-GEMM at MI355X" if is_moe else
+                       "tokens/sec (whole node) + MFU, Llama-3-8B SFT FSDP2 "
+                       "at 1/2/4/8 MI355X"),
             "value": round(tps, 1),
             "unit": "tokens/s",
             "n_gpus": world,
@@ -186,7 +238,10 @@ def main():
             "dtype": "fp8" if args.fp8 else "bf16",
             "data": "synthetic",
             "config": {
-                "model": "llama3-8b" if args.model == "llama3_8b" else args.model,
+                "model": ("llama3-8b" if args.model == "llama3_8b"
+                          else "qwen3-moe-30b-a3b" if args.model == "qwen3_moe_30b"
+                          else args.model),
+                "optimizer_state": str(state_dtype).replace("torch.", ""),
                 "global_batch": args.mbs * world,
                 "seq_len": args.seq_len,
                 "parallelism": f"fsdp{world}" if world > 1 else "single",
