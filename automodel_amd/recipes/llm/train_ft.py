@@ -109,8 +109,6 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         self.pipeline = None
         if self.mesh.mesh is not None and self.mesh.pp_size > 1:
             from automodel_amd.parallel.pp import AutoPipeline, PipelineConfig
-            assert cfg.get("step_scheduler", ConfigNode()).get("grad_acc_steps", 1) == 1, \
-                "PP handles microbatching via the schedule; use pp microbatches"
             pp_cfg = dist_cfg.get("pipeline", ConfigNode())
             if any(p.is_meta for p in self.model.parameters()):
                 self.model.init_weights(device="cpu")
@@ -118,7 +116,8 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
                 self.model, self.mesh["pp"],
                 PipelineConfig(pp_size=self.mesh.pp_size,
                                schedule=pp_cfg.get("schedule", "1f1b"),
-                               microbatches=pp_cfg.get("microbatches", self.mesh.pp_size)),
+                               microbatches=pp_cfg.get("microbatches", self.mesh.pp_size),
+                               virtual_stages=pp_cfg.get("virtual_stages", 1)),
                 loss_fn=self.loss_fn, device=self.device,
             )
             self.model = self.pipeline.stage_module
@@ -276,13 +275,14 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
         The schedule splits the batch into microbatches; the loss is computed
         on the last stage and broadcast for metrics (reference
         train_ft.py:1188 PP loss broadcast)."""
-        batch = batches[0]
-        input_ids = batch["input_ids"].to(self.device, non_blocking=True)
-        labels = batch["labels"].to(self.device, non_blocking=True)
-        losses = self.pipeline.step(input_ids=input_ids, target=labels)
         total = torch.zeros((), dtype=torch.float32, device=self.device)
-        if self.pipeline.is_last and losses:
-            total = sum(l.float() for l in losses)
+        # grad accumulation: each schedule.step accumulates into .grad
+        for batch in batches:
+            input_ids = batch["input_ids"].to(self.device, non_blocking=True)
+            labels = batch["labels"].to(self.device, non_blocking=True)
+            losses = self.pipeline.step(input_ids=input_ids, target=labels)
+            if self.pipeline.is_last and losses:
+                total = total + sum(l.float() for l in losses)
         # scale grads: schedule backwards sum-of-microbatch losses; match the
         # non-PP loss_scale by scaling grads post-hoc
         for p in self.model.parameters():
