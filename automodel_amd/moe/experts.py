@@ -27,6 +27,75 @@ def permute_tokens(x: torch.Tensor, indices: torch.Tensor, n_experts: int):
     return x_rep[sort_idx], sort_idx, counts
 
 
+class _FusedPermuteGather(torch.autograd.Function):
+    """x [T,H] -> x_perm [T*K,H] via the HIP gather kernel (no
+    repeat_interleave copy). Backward: per-token sum of replica grads via
+    the combine kernel with unit probs."""
+
+    @staticmethod
+    def forward(ctx, x, src, pos, K):
+        from automodel_amd.ops._backend import hip_ops
+
+        ctx.save_for_backward(pos)
+        ctx.K = K
+        return hip_ops().permute_gather(x.contiguous(), src)
+
+    @staticmethod
+    def backward(ctx, g):
+        from automodel_amd.ops._backend import hip_ops
+
+        (pos,) = ctx.saved_tensors
+        T = pos.numel() // ctx.K
+        ones = torch.ones(T, ctx.K, dtype=torch.float32, device=g.device)
+        dx = hip_ops().unpermute_combine(g.contiguous(), pos, ones)
+        return dx, None, None, None
+
+
+def fused_dispatch_combine(x, probs, indices, n_experts, expert_fn):
+    """GPU MoE glue: argsort routing, HIP gather, expert_fn over permuted
+    rows, HIP combine — one gather + one combine instead of
+    repeat_interleave + fancy-index + scatter (reference fused permute/
+    unpermute, moe_utils.py:31/115)."""
+    T, K = indices.shape
+    flat = indices.reshape(-1)
+    sort_idx = flat.argsort(stable=True)
+    counts = torch.bincount(flat, minlength=n_experts).to(torch.int32)
+    src = (sort_idx // K).to(torch.int32)
+    inv = torch.empty_like(sort_idx)
+    inv[sort_idx] = torch.arange(T * K, device=x.device)
+    pos = inv.to(torch.int32).view(T, K)
+    perm_probs = probs.reshape(-1).float()[sort_idx]     # prob per perm slot
+    x_perm = _FusedPermuteGather.apply(x, src, pos, K)
+    y_perm = expert_fn(x_perm, counts)
+    # combine with prob routing; dprobs via scatter of per-slot dots
+    return _CombineWithProbs.apply(y_perm, pos, probs, src, perm_probs, sort_idx)
+
+
+class _CombineWithProbs(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y_perm, pos, probs, src, perm_probs, sort_idx):
+        from automodel_amd.ops._backend import hip_ops
+
+        out = hip_ops().unpermute_combine(y_perm.contiguous(), pos,
+                                          probs.float().contiguous())
+        ctx.save_for_backward(y_perm, src, perm_probs, sort_idx)
+        ctx.shape_tk = probs.shape
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        from automodel_amd.ops._backend import hip_ops
+
+        y_perm, src, perm_probs, sort_idx = ctx.saved_tensors
+        g = g.contiguous()
+        g_gath = hip_ops().permute_gather(g, src)       # [T*K, H] perm order
+        dy = (g_gath.float() * perm_probs.unsqueeze(1)).to(y_perm.dtype)
+        s = (g_gath.float() * y_perm.float()).sum(-1)   # [T*K] perm order
+        dprobs = torch.empty_like(s)
+        dprobs.scatter_(0, sort_idx, s)
+        return dy, None, dprobs.view(ctx.shape_tk).to(torch.float32), None, None, None
+
+
 def unpermute_tokens(y_perm: torch.Tensor, sort_idx: torch.Tensor,
                      probs: torch.Tensor) -> torch.Tensor:
     """Scatter back and combine top-k with routing probs."""
@@ -58,12 +127,12 @@ class GroupedExperts(nn.Module):
         """Grouped projection y[t] = x[t] @ w[expert(t)].T over expert-sorted
         tokens — grouped-GEMM HIP kernel on GPU, per-expert loop on CPU.
         (Also used by peft/lora_experts.py adapters.)"""
-        cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
         if (x_perm.is_cuda and x_perm.dtype == torch.bfloat16
                 and x_perm.numel() > 0):
             from automodel_amd.ops.grouped_gemm import grouped_linear
 
-            return grouped_linear(x_perm, w, cl)
+            return grouped_linear(x_perm, w, counts)
+        cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
         outs = []
         start = 0
         for e, n in enumerate(cl):
@@ -99,16 +168,27 @@ class GroupedExperts(nn.Module):
             and x_perm.numel() > 0
         )
         if use_grouped:
-            from automodel_amd.ops.grouped_gemm import grouped_linear
+            from automodel_amd.ops.grouped_gemm import grouped_linear, make_group_plan
 
-            cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
-            g = grouped_linear(x_perm, self.gate_proj, cl)
-            u = grouped_linear(x_perm, self.up_proj, cl)
+            if torch.is_tensor(counts) and counts.is_cuda:
+                # device-side plan shared by all three projections — the
+                # routing counts never touch the host
+                plan = make_group_plan(counts, x_perm.shape[0])
+                cl = counts
+            else:
+                plan = None
+                cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+            g = grouped_linear(x_perm, self.gate_proj, cl, plan=plan)
+            u = grouped_linear(x_perm, self.up_proj, cl, plan=plan)
             h = swiglu(g, u)
-            return grouped_linear(h, self.down_proj, cl)
+            return grouped_linear(h, self.down_proj, cl, plan=plan)
         return self._expert_mlp_loop(x_perm, counts)
 
     def forward(self, x: torch.Tensor, probs: torch.Tensor, indices: torch.Tensor) -> torch.Tensor:
+        if (x.is_cuda and x.dtype == torch.bfloat16 and x.shape[-1] % 8 == 0
+                and self.backend in ("auto", "hip_grouped")):
+            return fused_dispatch_combine(x, probs, indices, self.n_experts,
+                                          self.forward_permuted)
         x_perm, sort_idx, counts = permute_tokens(x, indices, self.n_experts)
         y_perm = self.forward_permuted(x_perm, counts)
         return unpermute_tokens(y_perm, sort_idx, probs)

@@ -46,6 +46,37 @@ __global__ void adamw_bf16_kernel(bf16* __restrict__ p, const bf16* __restrict__
   }
 }
 
+// bf16-STATE variant: param/grad/m/v all bf16, NO fp32 master — halves
+// optimizer memory for single-GPU 30B-class benches (update math in fp32
+// registers; only the stored moments are quantized).
+__global__ void adamw_bf16_state_kernel(bf16* __restrict__ p, const bf16* __restrict__ g,
+                                        bf16* __restrict__ m, bf16* __restrict__ v,
+                                        long n4, float lr, float b1, float b2,
+                                        float eps, float wd_factor, float inv_bc1,
+                                        float inv_sqrt_bc2) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x4 gv = reinterpret_cast<const bf16x4*>(g)[i];
+    bf16x4 pv = reinterpret_cast<const bf16x4*>(p)[i];
+    bf16x4 mv = reinterpret_cast<const bf16x4*>(m)[i];
+    bf16x4 vv = reinterpret_cast<const bf16x4*>(v)[i];
+    bf16x4 pout, mout, vout;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float gf = bf2f(gv.v[j]);
+      float w = bf2f(pv.v[j]) * wd_factor;
+      float m_ = b1 * bf2f(mv.v[j]) + (1.f - b1) * gf;
+      float v_ = b2 * bf2f(vv.v[j]) + (1.f - b2) * gf * gf;
+      float denom = sqrtf(v_) * inv_sqrt_bc2 + eps;
+      w -= lr * inv_bc1 * m_ / denom;
+      mout.v[j] = f2bf(m_); vout.v[j] = f2bf(v_); pout.v[j] = f2bf(w);
+    }
+    reinterpret_cast<bf16x4*>(m)[i] = mout;
+    reinterpret_cast<bf16x4*>(v)[i] = vout;
+    reinterpret_cast<bf16x4*>(p)[i] = pout;
+  }
+}
+
 // fp32-param variant (no separate master copy); grad may be fp32 too.
 __global__ void adamw_f32_kernel(float* __restrict__ p, const float* __restrict__ g,
                                  float* __restrict__ m, float* __restrict__ v, long n,
@@ -73,7 +104,19 @@ void adamw_step(at::Tensor param, at::Tensor grad, at::Tensor master, at::Tensor
   const float inv_sqrt_bc2 = (float)(1.0 / std::sqrt(bc2));
   auto stream = c10::hip::getCurrentHIPStream();
   const int block = 256;
-  if (param.scalar_type() == at::kBFloat16) {
+  if (param.scalar_type() == at::kBFloat16 && m.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(param.numel() % 4 == 0, "adamw_step: numel must be multiple of 4");
+    const long n4 = param.numel() / 4;
+    const int grid = (int)std::min<long>((n4 + block - 1) / block, 4096);
+    hipLaunchKernelGGL(adamw_bf16_state_kernel, dim3(grid), dim3(block), 0,
+                       stream.stream(),
+                       reinterpret_cast<bf16*>(param.data_ptr()),
+                       reinterpret_cast<const bf16*>(grad.data_ptr()),
+                       reinterpret_cast<bf16*>(m.data_ptr()),
+                       reinterpret_cast<bf16*>(v.data_ptr()),
+                       n4, (float)lr, (float)beta1, (float)beta2, (float)eps,
+                       wd_factor, inv_bc1, inv_sqrt_bc2);
+  } else if (param.scalar_type() == at::kBFloat16) {
     TORCH_CHECK(param.numel() % 4 == 0, "adamw_step: numel must be multiple of 4");
     const long n4 = param.numel() / 4;
     const int grid = (int)std::min<long>((n4 + block - 1) / block, 4096);

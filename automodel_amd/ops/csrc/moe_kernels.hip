@@ -33,11 +33,38 @@ __device__ __forceinline__ int gg_off(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ (((row >> 1) & 7) << 4));
 }
 
+// Device-side group plan: offsets + (expert, row0) tile list built from the
+// on-device counts so the hot path never syncs counts to the host
+// (the round-1 tile map was host-built from counts.tolist() —
+// one device sync per MoE layer per direction). One block, serial scan over
+// E <= 4096 groups (microseconds).
+__global__ void build_group_plan_kernel(const int* __restrict__ counts, int E,
+                                        int* __restrict__ offs,
+                                        int* __restrict__ tile_map,
+                                        int* __restrict__ n_tiles) {
+  if (threadIdx.x != 0 || blockIdx.x != 0) return;
+  int acc = 0, t = 0;
+  offs[0] = 0;
+  for (int e = 0; e < E; ++e) {
+    const int c = counts[e];
+    for (int r = 0; r < c; r += GG_BM) {
+      tile_map[2 * t] = e;
+      tile_map[2 * t + 1] = acc + r;
+      ++t;
+    }
+    acc += c;
+    offs[e + 1] = acc;
+  }
+  n_tiles[0] = t;
+}
+
 __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w, bf16* __restrict__ y,
     const int* __restrict__ tile_map,   // [n_mtiles][2]: (expert, row0)
     const int* __restrict__ offs,       // [E+1] group row offsets
+    const int* __restrict__ n_tiles,    // device tile count (null = grid-sized)
     int K, int N) {
+  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xa = smem;                  // 16 KiB
   char* wb = smem + GG_BM * GG_BK * 2;
@@ -128,8 +155,27 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
   }
 }
 
+std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
+    const at::Tensor& counts, int64_t M) {
+  TORCH_CHECK(counts.is_cuda() && counts.scalar_type() == at::kInt,
+              "build_group_plan: counts int32 on GPU");
+  const int E = counts.size(0);
+  const long max_tiles = M / GG_BM + E + 1;
+  auto opts = counts.options();
+  auto offs = at::empty({E + 1}, opts);
+  auto tile_map = at::empty({max_tiles, 2}, opts);
+  auto n_tiles = at::empty({1}, opts);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(build_group_plan_kernel, dim3(1), dim3(64), 0, stream.stream(),
+                     counts.data_ptr<int>(), E, offs.data_ptr<int>(),
+                     tile_map.data_ptr<int>(), n_tiles.data_ptr<int>());
+  HIP_CHECK_KERNEL();
+  return {offs, tile_map, n_tiles};
+}
+
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
-                           const at::Tensor& offs, const at::Tensor& tile_map) {
+                           const at::Tensor& offs, const at::Tensor& tile_map,
+                           const std::optional<at::Tensor>& n_tiles) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kBFloat16,
               "grouped_gemm_nt: x [M,K] bf16");
   TORCH_CHECK(w.dim() == 3 && w.scalar_type() == at::kBFloat16, "w [E,N,K] bf16");
@@ -145,11 +191,12 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
   const dim3 grid(n_mtiles, N / GG_BN);
   const size_t smem = 2 * GG_BM * GG_BK * 2;
   auto stream = c10::hip::getCurrentHIPStream();
+  const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
   hipLaunchKernelGGL(grouped_gemm_nt_kernel, grid, dim3(256), smem, stream.stream(),
                      reinterpret_cast<const bf16*>(x.data_ptr()),
                      reinterpret_cast<const bf16*>(w.data_ptr()),
                      reinterpret_cast<bf16*>(y.data_ptr()),
-                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), K, N);
+                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
   HIP_CHECK_KERNEL();
   return y;
 }
@@ -168,7 +215,9 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
 
 __global__ __launch_bounds__(256) void grouped_gemm_nn_kernel(
     const bf16* __restrict__ g, const bf16* __restrict__ w, bf16* __restrict__ dx,
-    const int* __restrict__ tile_map, const int* __restrict__ offs, int N, int K) {
+    const int* __restrict__ tile_map, const int* __restrict__ offs,
+    const int* __restrict__ n_tiles, int N, int K) {
+  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* ga = smem;                       // [128 m][64 n]  16 KiB
   char* wt = smem + GG_BM * GG_BK * 2;   // [128 k][64 n]  16 KiB
@@ -354,7 +403,8 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
 }
 
 at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
-                           const at::Tensor& offs, const at::Tensor& tile_map) {
+                           const at::Tensor& offs, const at::Tensor& tile_map,
+                           const std::optional<at::Tensor>& n_tiles) {
   TORCH_CHECK(g.is_cuda() && g.dim() == 2 && g.scalar_type() == at::kBFloat16,
               "grouped_gemm_nn: g [M,N] bf16");
   TORCH_CHECK(w.dim() == 3 && w.scalar_type() == at::kBFloat16, "w [E,N,K] bf16");
@@ -368,11 +418,12 @@ at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
   const dim3 grid(n_mtiles, K / GG_BN);
   const size_t smem = 2 * GG_BM * GG_BK * 2;
   auto stream = c10::hip::getCurrentHIPStream();
+  const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
   hipLaunchKernelGGL(grouped_gemm_nn_kernel, grid, dim3(256), smem, stream.stream(),
                      reinterpret_cast<const bf16*>(g.data_ptr()),
                      reinterpret_cast<const bf16*>(w.data_ptr()),
                      reinterpret_cast<bf16*>(dx.data_ptr()),
-                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), N, K);
+                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, N, K);
   HIP_CHECK_KERNEL();
   return dx;
 }

@@ -61,10 +61,14 @@ at::Tensor nf4_dequant(const at::Tensor& packed, const at::Tensor& absmax,
 at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias);
 
+std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
+    const at::Tensor& counts, int64_t M);
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
-                           const at::Tensor& offs, const at::Tensor& tile_map);
+                           const at::Tensor& offs, const at::Tensor& tile_map,
+                           const std::optional<at::Tensor>& n_tiles = std::nullopt);
 at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
-                           const at::Tensor& offs, const at::Tensor& tile_map);
+                           const at::Tensor& offs, const at::Tensor& tile_map,
+                           const std::optional<at::Tensor>& n_tiles = std::nullopt);
 at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
                            const at::Tensor& offs, int64_t E);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);

@@ -1,9 +1,14 @@
-"""Grouped GEMM for stacked expert weights: HIP MFMA kernel + fallbacks.
+"""Grouped GEMM for stacked expert weights: HIP MFMA kernels + fallbacks.
 
 ``grouped_linear(x_perm, w, counts)`` computes y[rows of group e] =
 x[rows] @ w[e].T with autograd. Forward runs the in-tree CDNA4 kernel
-(csrc/moe_kernels.hip grouped_gemm_nt); backward uses per-group hipBLASLt
-GEMMs (dx = g @ w[e], dw[e] = g_e^T @ x_e).
+(csrc/moe_kernels.hip grouped_gemm_nt); backward runs the single-kernel
+grouped dx (NN) and dw (TN) kernels — no per-expert loop, no atomics.
+
+With ``counts`` as a DEVICE int tensor the (offs, tile_map, n_tiles) plan is
+built by a device kernel (build_group_plan) so the hot path never syncs the
+routing counts to the host (the round-1 counts.tolist() cost one device sync
+per MoE layer per direction).
 
 Replaces the reference's grouped_gemm / torch._grouped_mm expert backends
 (nemo_automodel/components/moe/experts.py:656).
@@ -31,6 +36,11 @@ def _build_tile_map(counts: list[int]) -> tuple[torch.Tensor, torch.Tensor]:
     return tile_map, torch.tensor(offs, dtype=torch.int32)
 
 
+def make_group_plan(counts: torch.Tensor, M: int):
+    """Device-side plan (offs, tile_map, n_tiles) from on-device counts."""
+    return hip_ops().build_group_plan(counts.to(torch.int32), M)
+
+
 def _loop_gemm_nt(x, w, counts):
     outs, start = [], 0
     for e, n in enumerate(counts):
@@ -41,51 +51,63 @@ def _loop_gemm_nt(x, w, counts):
 
 class _GroupedLinear(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, counts: tuple):
+    def forward(ctx, x, w, counts, offs, tile_map, n_tiles):
         ctx.save_for_backward(x, w)
-        ctx.counts = counts
         use_hip = (
             x.is_cuda and x.dtype == torch.bfloat16
             and w.shape[2] % 64 == 0 and w.shape[1] % 128 == 0
         )
         ctx.use_hip = use_hip
+        ctx.counts = counts
+        ctx.plan = (offs, tile_map, n_tiles)
         if use_hip:
-            tile_map, offs = _build_tile_map(list(counts))
-            ctx.tile_offs = (tile_map.to(x.device), offs.to(x.device))
             return hip_ops().grouped_gemm_nt(
-                x.contiguous(), w.contiguous(),
-                ctx.tile_offs[1], ctx.tile_offs[0],
-            )
-        return _loop_gemm_nt(x, w, counts)
+                x.contiguous(), w.contiguous(), offs, tile_map, n_tiles)
+        cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+        return _loop_gemm_nt(x, w, cl)
 
     @staticmethod
     def backward(ctx, g):
         x, w = ctx.saved_tensors
-        counts = ctx.counts
         g = g.contiguous()
-        # single-kernel grouped backward (the per-expert
-        # hipBLASLt loop is launch-bound at 128+ small experts); dw's TN
+        offs, tile_map, n_tiles = ctx.plan
+        # single-kernel grouped backward; dw's TN
         # kernel needs N%128 and K%128
         if ctx.use_hip and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0:
-            tile_map, offs = ctx.tile_offs
             ops = hip_ops()
-            dx = ops.grouped_gemm_nn(g, w.contiguous(), offs, tile_map)
+            dx = ops.grouped_gemm_nn(g, w.contiguous(), offs, tile_map, n_tiles)
             dw = ops.grouped_gemm_tn(g, x.contiguous(), offs, w.shape[0])
-            return dx, dw, None
+            return dx, dw, None, None, None, None
+        cl = ctx.counts.tolist() if torch.is_tensor(ctx.counts) else list(ctx.counts)
         dx = torch.empty_like(x)
         dw = torch.zeros_like(w)
         start = 0
-        for e, n in enumerate(counts):
+        for e, n in enumerate(cl):
             if n:
                 ge = g[start : start + n]
                 dx[start : start + n] = ge @ w[e]
                 dw[e] = (ge.t() @ x[start : start + n]).to(w.dtype)
             start += n
-        return dx, dw, None
+        return dx, dw, None, None, None, None
 
 
-def grouped_linear(x_perm: torch.Tensor, w: torch.Tensor, counts) -> torch.Tensor:
-    """x_perm [M, K] sorted by group; w [E, N, K]; counts per group (host)."""
-    if not isinstance(counts, (tuple, list)):
-        counts = counts.tolist()
-    return _GroupedLinear.apply(x_perm, w, tuple(int(c) for c in counts))
+def grouped_linear(x_perm: torch.Tensor, w: torch.Tensor, counts,
+                   plan=None) -> torch.Tensor:
+    """x_perm [M, K] sorted by group; w [E, N, K]; counts per group.
+
+    ``counts`` may be a DEVICE int tensor (no host sync — plan built on
+    device) or a host list/tuple. ``plan`` lets callers share one
+    (offs, tile_map, n_tiles) across several projections of the same
+    routing."""
+    hip_eligible = (x_perm.is_cuda and x_perm.dtype == torch.bfloat16
+                    and w.shape[2] % 64 == 0 and w.shape[1] % 128 == 0)
+    if plan is None and hip_eligible:
+        if torch.is_tensor(counts) and counts.is_cuda:
+            plan = make_group_plan(counts, x_perm.shape[0])
+        else:
+            cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+            tile_map, offs = _build_tile_map([int(c) for c in cl])
+            plan = (offs.to(x_perm.device), tile_map.to(x_perm.device), None)
+    if plan is None:
+        plan = (None, None, None)
+    return _GroupedLinear.apply(x_perm, w, counts, *plan)

@@ -57,20 +57,22 @@ class FusedAdamW(torch.optim.Optimizer):
                 grad = _local(p_.grad)
                 state = self.state[p_]
                 use_hip = (p.is_cuda and p.dtype == torch.bfloat16
-                           and p.numel() % 4 == 0
-                           and self.state_dtype == torch.float32)
+                           and p.numel() % 4 == 0)
                 if len(state) == 0:
                     state["step"] = 0
                     state["exp_avg"] = torch.zeros_like(p, dtype=self.state_dtype)
                     state["exp_avg_sq"] = torch.zeros_like(p, dtype=self.state_dtype)
-                    if use_hip:
+                    if use_hip and self.state_dtype == torch.float32:
                         state["master"] = p.detach().float().clone()
                 state["step"] += 1
                 t = state["step"]
                 m, v = state["exp_avg"], state["exp_avg_sq"]
                 if use_hip:
+                    # fp32 states ride the master-weight kernel; bf16 states
+                    # ride the master-free kernel (m passed as placeholder
+                    # for the unused master arg)
                     hip_ops().adamw_step(
-                        p, grad.to(torch.bfloat16), state["master"], m, v,
+                        p, grad.to(torch.bfloat16), state.get("master", m), m, v,
                         t, lr, beta1, beta2, eps, wd,
                     )
                 else:

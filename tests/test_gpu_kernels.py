@@ -783,3 +783,66 @@ def test_grouped_gemm_bwd_deterministic():
         grads.append((x.grad.clone(), w.grad.clone()))
     assert torch.equal(grads[0][0], grads[1][0])
     assert torch.equal(grads[0][1], grads[1][1])
+
+
+def test_fused_dispatch_combine_parity():
+    """HIP gather/combine MoE glue (no repeat_interleave, no host sync) vs
+    the torch permute path: fwd + grads through x, probs, and experts."""
+    from automodel_amd.moe.experts import (
+        GroupedExperts,
+        fused_dispatch_combine,
+        permute_tokens,
+        unpermute_tokens,
+    )
+
+    torch.manual_seed(0)
+    T, H, E, K, inter = 512, 256, 8, 2, 128
+    x0 = torch.randn(T, H, device="cuda", dtype=torch.bfloat16)
+    probs0 = torch.rand(T, K, device="cuda", dtype=torch.float32)
+    probs0 = probs0 / probs0.sum(-1, keepdim=True)
+    indices = torch.randint(0, E, (T, K), device="cuda")
+    experts = GroupedExperts(E, H, inter).to("cuda", torch.bfloat16)
+    experts.init_weights()
+
+    x = x0.clone().requires_grad_(True)
+    probs = probs0.clone().requires_grad_(True)
+    out = fused_dispatch_combine(x, probs, indices, E, experts.forward_permuted)
+    gout = torch.randn_like(out)
+    out.backward(gout)
+
+    x2 = x0.clone().requires_grad_(True)
+    probs2 = probs0.clone().requires_grad_(True)
+    x_perm, sort_idx, counts = permute_tokens(x2, indices, E)
+    y_perm = experts.forward_permuted(x_perm, counts)
+    ref = unpermute_tokens(y_perm, sort_idx, probs2)
+    ref.backward(gout)
+
+    assert torch.allclose(out.float(), ref.float(), atol=3e-2, rtol=3e-2), \
+        (out.float() - ref.float()).abs().max()
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=5e-2, rtol=5e-2), \
+        (x.grad.float() - x2.grad.float()).abs().max()
+    assert torch.allclose(probs.grad, probs2.grad, atol=5e-2, rtol=5e-2), \
+        (probs.grad - probs2.grad).abs().max()
+
+
+def test_adamw_bf16_state_kernel():
+    """Master-free bf16-state AdamW kernel follows the fp32 trajectory to
+    bf16 quantization."""
+    from automodel_amd.optim.adamw import FusedAdamW
+
+    torch.manual_seed(0)
+    w0 = torch.randn(8192, device="cuda")
+    p_bf = torch.nn.Parameter(w0.clone().to(torch.bfloat16))
+    p_ref = torch.nn.Parameter(w0.clone())
+    opt_bf = FusedAdamW([p_bf], lr=1e-2, betas=(0.9, 0.95), weight_decay=0.1,
+                        state_dtype=torch.bfloat16)
+    opt_ref = torch.optim.AdamW([p_ref], lr=1e-2, betas=(0.9, 0.95), eps=1e-8,
+                                weight_decay=0.1)
+    for _ in range(5):
+        g = torch.randn(8192, device="cuda")
+        p_bf.grad = g.to(torch.bfloat16)
+        p_ref.grad = g.clone()
+        opt_bf.step()
+        opt_ref.step()
+    assert torch.allclose(p_bf.float(), p_ref.detach(), atol=5e-2, rtol=5e-2), \
+        (p_bf.float() - p_ref.detach()).abs().max()
