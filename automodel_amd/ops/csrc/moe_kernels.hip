@@ -90,31 +90,40 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  for (int k0 = 0; k0 < K; k0 += GG_BK) {
-    // ---- stage x tile [128 m][64 k] and w tile [128 n][64 k]
-    {
-      // 256 threads x 4 pieces: piece p -> row tid/4 + p*64? use flat:
-      // elem = (tid*4 + p) * 8 over 128*64 elems
-      const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
-      const int c0 = (tid % 8) * 8;
+  // T14 split (guide): issue global loads for chunk kc+1 before computing
+  // chunk kc from LDS — HBM latency hides under the MFMA phase.
+  const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
+  const int c0 = (tid % 8) * 8;
+  bf16x8 xreg[4], wreg[4];
+  auto issue_loads = [&](int k0) {
 #pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int row = row0 + rr * 32;
-        const int m = m0 + row;
-        bf16x8 xv;
-        if (m < m_end) {
-          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
-        } else {
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      const int m = m0 + row;
+      if (m < m_end) {
+        xreg[rr] = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+      } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
-        }
-        *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xv;
-        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
-            w + wbase + (long)(n0 + row) * K + k0 + c0);
-        *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wv;
+        for (int j = 0; j < 8; ++j) xreg[rr].v[j] = f2bf(0.f);
       }
+      wreg[rr] = *reinterpret_cast<const bf16x8*>(
+          w + wbase + (long)(n0 + row) * K + k0 + c0);
     }
+  };
+  auto write_lds = [&] {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xreg[rr];
+      *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wreg[rr];
+    }
+  };
+
+  issue_loads(0);
+  for (int k0 = 0; k0 < K; k0 += GG_BK) {
+    write_lds();
     __syncthreads();
+    if (k0 + GG_BK < K) issue_loads(k0 + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -243,38 +252,43 @@ __global__ __launch_bounds__(256) void grouped_gemm_nn_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  for (int n0 = 0; n0 < N; n0 += GG_BK) {
-    {
-      // g tile [128 m][64 n]: 128 rows x 8 vec-chunks = 1024 vectors
-      const int row0 = tid / 8;
-      const int c0 = (tid % 8) * 8;
+  const int row0 = tid / 8;
+  const int c0 = (tid % 8) * 8;
+  const int nrow0 = tid / 16;
+  const int kc0 = (tid % 16) * 8;
+  bf16x8 greg[4], wreg[4];
+  auto issue_loads = [&](int n0) {
 #pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int row = row0 + rr * 32;
-        const int m = m0 + row;
-        bf16x8 gv;
-        if (m < m_end) {
-          gv = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
-        } else {
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      const int m = m0 + row;
+      if (m < m_end) {
+        greg[rr] = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
+      } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) gv.v[j] = f2bf(0.f);
-        }
-        *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = gv;
+        for (int j = 0; j < 8; ++j) greg[rr].v[j] = f2bf(0.f);
       }
-      // w tile [64 n rows][128 k]: read w[n][k-chunk], scatter into wt[k][n]
-      const int nrow0 = tid / 16;
-      const int kc0 = (tid % 16) * 8;
-#pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int nrow = nrow0 + rr * 16;
-        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
-            w + wbase + (long)(n0 + nrow) * K + k0 + kc0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<bf16*>(wt + gg_off(kc0 + j, nrow * 2)) = wv.v[j];
-      }
+      wreg[rr] = *reinterpret_cast<const bf16x8*>(
+          w + wbase + (long)(n0 + nrow0 + rr * 16) * K + k0 + kc0);
     }
+  };
+  auto write_lds = [&] {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = greg[rr];
+      // w tile staged TRANSPOSED (wt[k][n]) for the contiguous B-fragment
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        *reinterpret_cast<bf16*>(wt + gg_off(kc0 + j, (nrow0 + rr * 16) * 2)) = wreg[rr].v[j];
+    }
+  };
+
+  issue_loads(0);
+  for (int n0 = 0; n0 < N; n0 += GG_BK) {
+    write_lds();
     __syncthreads();
+    if (n0 + GG_BK < N) issue_loads(n0 + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -314,11 +328,16 @@ __global__ __launch_bounds__(256) void grouped_gemm_nn_kernel(
 }
 
 __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
-    const bf16* __restrict__ g, const bf16* __restrict__ x, bf16* __restrict__ dw,
-    const int* __restrict__ offs, int N, int K) {
+    const bf16* __restrict__ gt, const bf16* __restrict__ xt, bf16* __restrict__ dw,
+    const int* __restrict__ offs, int N, int K, long M) {
+  // dw[e, n, k] = sum_{m in group e} gt[n, m] * xt[k, m]
+  // Operands arrive PRE-TRANSPOSED ([N, M] / [K, M], m contiguous) so the
+  // staging is plain vector loads/stores — the round-1 in-kernel scalar
+  // transpose ran at 80 TF/s (profiles/moe_r2_kernel_stats.csv); this
+  // NT-shaped structure + T14 split matches the forward kernel's pipeline.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* gt = smem;                       // [128 n][64 m]  16 KiB
-  char* xt = smem + GG_BM * GG_BK * 2;   // [128 k][64 m]  16 KiB
+  char* ga = smem;                       // [128 n][64 m]
+  char* xb = smem + GG_BM * GG_BK * 2;   // [128 k][64 m]
 
   const int e = blockIdx.x;
   const int n0 = blockIdx.y * GG_BN;
@@ -339,32 +358,45 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  for (int mc = m_start; mc < m_end; mc += GG_BK) {
-    {
-      // stage 64 m-rows of g[:, n0:n0+128] and x[:, k0:k0+128], transposed:
-      // 256 threads x (64 m x 16 vec-chunks / 256) = 4 vectors each per tensor
-      const int mrow0 = tid / 16;          // 16 rows per pass
-      const int c0 = (tid % 16) * 8;       // 16 chunks of 8 cover 128 cols
+  const int row0 = tid / 8;          // 32 rows per pass (of 128)
+  const int c0 = (tid % 8) * 8;      // 8 chunks cover 64 m
+  bf16x8 greg[4], xreg[4];
+  auto issue_loads = [&](int mc) {
+    // mc is 64-aligned (loop starts at m_start & ~63); rows outside
+    // [m_start, m_end) are zeroed — zero contraction terms are harmless
 #pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int mrow = mrow0 + rr * 16;
-        const int m = mc + mrow;
-        bf16x8 gv, xv;
-        if (m < m_end) {
-          gv = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
-          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) { gv.v[j] = f2bf(0.f); xv.v[j] = f2bf(0.f); }
-        }
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      const long m = (long)mc + c0;
+      if (m >= m_start && m + 7 < m_end) {
+        greg[rr] = *reinterpret_cast<const bf16x8*>(gt + (long)(n0 + row) * M + m);
+        xreg[rr] = *reinterpret_cast<const bf16x8*>(xt + (long)(k0 + row) * M + m);
+      } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          *reinterpret_cast<bf16*>(gt + gg_off(c0 + j, mrow * 2)) = gv.v[j];
-          *reinterpret_cast<bf16*>(xt + gg_off(c0 + j, mrow * 2)) = xv.v[j];
+          const long mm = m + j;
+          const bool ok = mm >= m_start && mm < m_end;
+          greg[rr].v[j] = ok ? gt[(long)(n0 + row) * M + mm] : f2bf(0.f);
+          xreg[rr].v[j] = ok ? xt[(long)(k0 + row) * M + mm] : f2bf(0.f);
         }
       }
     }
+  };
+  auto write_lds = [&] {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = greg[rr];
+      *reinterpret_cast<bf16x8*>(xb + gg_off(row, c0 * 2)) = xreg[rr];
+    }
+  };
+
+  const int mc0 = m_start & ~(GG_BK - 1);     // 64-aligned loop start
+  if (m_start < m_end) issue_loads(mc0);
+  for (int mc = mc0; mc < m_end; mc += GG_BK) {
+    write_lds();
     __syncthreads();
+    if (mc + GG_BK < m_end) issue_loads(mc + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -374,9 +406,9 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         a[i] = *reinterpret_cast<const bf16x8v*>(
-            gt + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+            ga + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
         b[i] = *reinterpret_cast<const bf16x8v*>(
-            xt + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+            xb + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -428,22 +460,24 @@ at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
   return dx;
 }
 
-at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
+at::Tensor grouped_gemm_tn(const at::Tensor& gt, const at::Tensor& xt,
                            const at::Tensor& offs, int64_t E) {
-  TORCH_CHECK(g.is_cuda() && g.dim() == 2 && x.dim() == 2 &&
-                  g.scalar_type() == at::kBFloat16,
-              "grouped_gemm_tn: g [M,N], x [M,K] bf16");
-  const int N = g.size(1), K = x.size(1);
+  TORCH_CHECK(gt.is_cuda() && gt.dim() == 2 && xt.dim() == 2 &&
+                  gt.scalar_type() == at::kBFloat16,
+              "grouped_gemm_tn: gt [N,M], xt [K,M] bf16 (pre-transposed)");
+  const int N = gt.size(0), K = xt.size(0);
+  const long M = gt.size(1);
+  TORCH_CHECK(xt.size(1) == M, "M mismatch");
   TORCH_CHECK(N % GG_BN == 0 && K % GG_BN == 0, "need N%128==0, K%128==0");
-  auto dw = at::empty({E, (long)N, (long)K}, g.options());
+  auto dw = at::empty({E, (long)N, (long)K}, gt.options());
   const dim3 grid((unsigned)E, N / GG_BN, K / GG_BN);
   const size_t smem = 2 * GG_BM * GG_BK * 2;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(grouped_gemm_tn_kernel, grid, dim3(256), smem, stream.stream(),
-                     reinterpret_cast<const bf16*>(g.data_ptr()),
-                     reinterpret_cast<const bf16*>(x.data_ptr()),
+                     reinterpret_cast<const bf16*>(gt.data_ptr()),
+                     reinterpret_cast<const bf16*>(xt.data_ptr()),
                      reinterpret_cast<bf16*>(dw.data_ptr()),
-                     offs.data_ptr<int>(), N, K);
+                     offs.data_ptr<int>(), N, K, M);
   HIP_CHECK_KERNEL();
   return dw;
 }

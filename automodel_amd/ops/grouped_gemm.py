@@ -76,7 +76,10 @@ class _GroupedLinear(torch.autograd.Function):
         if ctx.use_hip and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0:
             ops = hip_ops()
             dx = ops.grouped_gemm_nn(g, w.contiguous(), offs, tile_map, n_tiles)
-            dw = ops.grouped_gemm_tn(g, x.contiguous(), offs, w.shape[0])
+            # TN operands pre-transposed (m contiguous) so the kernel's
+            # staging is vectorized — see csrc grouped_gemm_tn_kernel
+            dw = ops.grouped_gemm_tn(g.t().contiguous(), x.t().contiguous(),
+                                     offs, w.shape[0])
             return dx, dw, None, None, None, None
         cl = ctx.counts.tolist() if torch.is_tensor(ctx.counts) else list(ctx.counts)
         dx = torch.empty_like(x)
