@@ -392,8 +392,9 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
   };
 
   const int mc0 = m_start & ~(GG_BK - 1);     // 64-aligned loop start
-  if (m_start < m_end) issue_loads(mc0);
-  for (int mc = mc0; mc < m_end; mc += GG_BK) {
+  if (m_start < m_end) {
+    issue_loads(mc0);
+    for (int mc = mc0; mc < m_end; mc += GG_BK) {
     write_lds();
     __syncthreads();
     if (mc + GG_BK < m_end) issue_loads(mc + GG_BK);
@@ -417,6 +418,7 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
     }
     __syncthreads();
+    }
   }
 
   const long wbase = (long)e * N * K;

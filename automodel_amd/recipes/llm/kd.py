@@ -1,8 +1,24 @@
 """Knowledge-distillation recipe: frozen teacher + student.
 
-Reference behavior: nemo_automodel/recipes/llm/kd.py (teacher/student KD with
-forward-KL + CE mixture). Reuses the finetune loop; builds a second (frozen)
-teacher model and overrides the loss step.
+Reference behavior: nemo_automodel/recipes/llm/kd.py (1243 LoC:
+KnowledgeDistillationRecipeForNextTokenPrediction — teacher built/sharded
+beside the student, chunked-logits KD loss, per-component loss logging,
+tokenizer/vocab compatibility validation, intermediate-layer distillation).
+Re-designed on the finetune loop rather than copied: the teacher shares the
+student's FSDP mesh, hidden states for intermediate distillation are
+captured with forward hooks, and the KD/CE components are logged separately.
+
+Config surface::
+
+    teacher:  {architecture | pretrained_path, config, dtype}
+    kd:
+      alpha: 0.5            # CE weight (1-alpha on the KL term)
+      temperature: 1.0
+      chunk_size: 2048      # tokens per chunked teacher-KL block
+      intermediate:         # optional hidden-state distillation
+        layer_map: [[2, 5], [4, 11]]     # [student_layer, teacher_layer]
+        weight: 1.0
+        mode: cosine        # cosine | mse (loss/distill_extras.py)
 """
 
 from __future__ import annotations
@@ -17,6 +33,31 @@ from automodel_amd.models.registry import build_model
 from automodel_amd.recipes.llm.train_ft import TrainFinetuneRecipeForNextTokenPrediction
 
 
+def _decoder_layers(model) -> list[torch.nn.Module]:
+    core = getattr(model, "model", None) or getattr(model, "transformer", None)
+    return list(core.layers) if core is not None and hasattr(core, "layers") else []
+
+
+class _HiddenTap:
+    """Forward hooks capturing selected decoder-layer outputs."""
+
+    def __init__(self, model, layer_ids: list[int]):
+        self.acts: dict[int, torch.Tensor] = {}
+        self.handles = []
+        layers = _decoder_layers(model)
+        for li in layer_ids:
+            def hook(mod, inp, out, _li=li):
+                self.acts[_li] = out[0] if isinstance(out, tuple) else out
+            self.handles.append(layers[li].register_forward_hook(hook))
+
+    def clear(self):
+        self.acts.clear()
+
+    def remove(self):
+        for h in self.handles:
+            h.remove()
+
+
 class KDRecipeForNextTokenPrediction(TrainFinetuneRecipeForNextTokenPrediction):
     def setup(self) -> None:
         super().setup()
@@ -27,30 +68,84 @@ class KDRecipeForNextTokenPrediction(TrainFinetuneRecipeForNextTokenPrediction):
             architecture=tcfg.get("architecture"),
             dtype=tcfg.get("dtype", "bfloat16"),
         )
+        # vocab compatibility (reference _verify_tokenizer_compatibility):
+        # student and teacher must score the same token space
+        sv = getattr(self.model.config, "vocab_size", None)
+        tv = getattr(self.teacher.config, "vocab_size", None)
+        if sv is not None and tv is not None and sv != tv:
+            raise ValueError(
+                f"KD teacher vocab_size {tv} != student {sv} — same-tokenizer "
+                "distillation requires matching vocabularies")
+
+        # teacher shares the student's DP mesh: sharded forward-only weights
+        if self.mesh.mesh is not None and self.mesh.dims.get("dp_shard", 1) > 1:
+            from automodel_amd.parallel.fsdp import apply_fsdp
+            apply_fsdp(self.teacher, self.mesh["dp_shard"],
+                       reshard_after_forward=True)
         if tcfg.get("pretrained_path"):
             from automodel_amd.checkpoint.hf_loader import load_hf_weights
             load_hf_weights(self.teacher, tcfg.pretrained_path, device=self.device)
+            self.teacher.to(self.device)
         else:
             self.teacher.init_weights(device=self.device)
         self.teacher.eval()
         for p in self.teacher.parameters():
             p.requires_grad_(False)
+
         kd_cfg = self.cfg.get("kd", ConfigNode())
         self.kd_loss = KDLoss(
             alpha=kd_cfg.get("alpha", 0.5),
             temperature=kd_cfg.get("temperature", 1.0),
             chunk_size=kd_cfg.get("chunk_size", 2048),
         )
+        # optional intermediate-layer distillation (reference recipe +
+        # loss/intermediate_distill equivalents)
+        self._inter_cfg = kd_cfg.get("intermediate")
+        self._taps = None
+        if self._inter_cfg:
+            from automodel_amd.loss.distill_extras import EmbeddingDistillLoss
+
+            layer_map = [tuple(x) for x in self._inter_cfg.layer_map]
+            self._layer_map = layer_map
+            self._inter_weight = self._inter_cfg.get("weight", 1.0)
+            self._taps = (
+                _HiddenTap(self.model, [s for s, _ in layer_map]),
+                _HiddenTap(self.teacher, [t for _, t in layer_map]),
+            )
+            self._inter_loss = EmbeddingDistillLoss(
+                self.model.config.hidden_size,
+                self.teacher.config.hidden_size,
+                mode=self._inter_cfg.get("mode", "cosine"),
+            ).to(self.device)
+            # projection params train with the student
+            self.optimizer.add_param_group(
+                {"params": list(self._inter_loss.parameters())})
+        self._kd_components: dict[str, float] = {}
 
     def _forward_backward_step(self, batch: dict, loss_scale: float) -> torch.Tensor:
         input_ids = batch["input_ids"].to(self.device, non_blocking=True)
         labels = batch["labels"].to(self.device, non_blocking=True)
+        if self._taps:
+            self._taps[0].clear()
+            self._taps[1].clear()
         with torch.no_grad():
             teacher_logits = self.teacher(input_ids)
         student_logits = self.model(input_ids)
         loss = self.kd_loss(student_logits, teacher_logits, labels)
+        self._kd_components = {"kd_total": float(loss.detach())}
+        if self._taps:
+            s_tap, t_tap = self._taps
+            inter = student_logits.new_zeros(())
+            for s_li, t_li in self._layer_map:
+                inter = inter + self._inter_loss(s_tap.acts[s_li],
+                                                 t_tap.acts[t_li])
+            loss = loss + self._inter_weight * inter
+            self._kd_components["kd_intermediate"] = float(inter.detach())
         (loss * loss_scale).backward()
         return loss.detach()
+
+    def extra_log_fields(self) -> dict:
+        return dict(self._kd_components)
 
 
 def main(argv=None):
