@@ -90,40 +90,31 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  // T14 split (guide): issue global loads for chunk kc+1 before computing
-  // chunk kc from LDS — HBM latency hides under the MFMA phase.
-  const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
-  const int c0 = (tid % 8) * 8;
-  bf16x8 xreg[4], wreg[4];
-  auto issue_loads = [&](int k0) {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int row = row0 + rr * 32;
-      const int m = m0 + row;
-      if (m < m_end) {
-        xreg[rr] = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) xreg[rr].v[j] = f2bf(0.f);
-      }
-      wreg[rr] = *reinterpret_cast<const bf16x8*>(
-          w + wbase + (long)(n0 + row) * K + k0 + c0);
-    }
-  };
-  auto write_lds = [&] {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int row = row0 + rr * 32;
-      *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xreg[rr];
-      *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wreg[rr];
-    }
-  };
-
-  issue_loads(0);
   for (int k0 = 0; k0 < K; k0 += GG_BK) {
-    write_lds();
+    // ---- stage x tile [128 m][64 k] and w tile [128 n][64 k]
+    {
+      // 256 threads x 4 pieces: piece p -> row tid/4 + p*64? use flat:
+      // elem = (tid*4 + p) * 8 over 128*64 elems
+      const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
+      const int c0 = (tid % 8) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + rr * 32;
+        const int m = m0 + row;
+        bf16x8 xv;
+        if (m < m_end) {
+          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
+        }
+        *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xv;
+        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+            w + wbase + (long)(n0 + row) * K + k0 + c0);
+        *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wv;
+      }
+    }
     __syncthreads();
-    if (k0 + GG_BK < K) issue_loads(k0 + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -252,43 +243,38 @@ __global__ __launch_bounds__(256) void grouped_gemm_nn_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  const int row0 = tid / 8;
-  const int c0 = (tid % 8) * 8;
-  const int nrow0 = tid / 16;
-  const int kc0 = (tid % 16) * 8;
-  bf16x8 greg[4], wreg[4];
-  auto issue_loads = [&](int n0) {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int row = row0 + rr * 32;
-      const int m = m0 + row;
-      if (m < m_end) {
-        greg[rr] = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) greg[rr].v[j] = f2bf(0.f);
-      }
-      wreg[rr] = *reinterpret_cast<const bf16x8*>(
-          w + wbase + (long)(n0 + nrow0 + rr * 16) * K + k0 + kc0);
-    }
-  };
-  auto write_lds = [&] {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int row = row0 + rr * 32;
-      *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = greg[rr];
-      // w tile staged TRANSPOSED (wt[k][n]) for the contiguous B-fragment
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        *reinterpret_cast<bf16*>(wt + gg_off(kc0 + j, (nrow0 + rr * 16) * 2)) = wreg[rr].v[j];
-    }
-  };
-
-  issue_loads(0);
   for (int n0 = 0; n0 < N; n0 += GG_BK) {
-    write_lds();
+    {
+      // g tile [128 m][64 n]: 128 rows x 8 vec-chunks = 1024 vectors
+      const int row0 = tid / 8;
+      const int c0 = (tid % 8) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int row = row0 + rr * 32;
+        const int m = m0 + row;
+        bf16x8 gv;
+        if (m < m_end) {
+          gv = *reinterpret_cast<const bf16x8*>(g + (long)m * N + n0 + c0);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) gv.v[j] = f2bf(0.f);
+        }
+        *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = gv;
+      }
+      // w tile [64 n rows][128 k]: read w[n][k-chunk], scatter into wt[k][n]
+      const int nrow0 = tid / 16;
+      const int kc0 = (tid % 16) * 8;
+#pragma unroll
+      for (int rr = 0; rr < 4; ++rr) {
+        const int nrow = nrow0 + rr * 16;
+        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+            w + wbase + (long)(n0 + nrow) * K + k0 + kc0);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          *reinterpret_cast<bf16*>(wt + gg_off(kc0 + j, nrow * 2)) = wv.v[j];
+      }
+    }
     __syncthreads();
-    if (n0 + GG_BK < N) issue_loads(n0 + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -334,7 +320,7 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
   // Operands arrive PRE-TRANSPOSED ([N, M] / [K, M], m contiguous) so the
   // staging is plain vector loads/stores — the round-1 in-kernel scalar
   // transpose ran at 80 TF/s (profiles/moe_r2_kernel_stats.csv); this
-  // NT-shaped structure + T14 split matches the forward kernel's pipeline.
+  // NT-shaped plain-staged structure matches the forward kernel.
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* ga = smem;                       // [128 n][64 m]
   char* xb = smem + GG_BM * GG_BK * 2;   // [128 k][64 m]
@@ -360,44 +346,31 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
 
   const int row0 = tid / 8;          // 32 rows per pass (of 128)
   const int c0 = (tid % 8) * 8;      // 8 chunks cover 64 m
-  bf16x8 greg[4], xreg[4];
-  auto issue_loads = [&](int mc) {
-    // mc is 64-aligned (loop starts at m_start & ~63); rows outside
-    // [m_start, m_end) are zeroed — zero contraction terms are harmless
+  const int mc0 = m_start & ~(GG_BK - 1);     // 64-aligned loop start
+  for (int mc = mc0; mc < m_end; mc += GG_BK) {
+    // plain staging (round-1 A/B: source-level double-buffering is a net
+    // loss here — the extra register set costs occupancy; see NOTES_ROUND2)
 #pragma unroll
     for (int rr = 0; rr < 4; ++rr) {
       const int row = row0 + rr * 32;
       const long m = (long)mc + c0;
+      bf16x8 gv, xv;
       if (m >= m_start && m + 7 < m_end) {
-        greg[rr] = *reinterpret_cast<const bf16x8*>(gt + (long)(n0 + row) * M + m);
-        xreg[rr] = *reinterpret_cast<const bf16x8*>(xt + (long)(k0 + row) * M + m);
+        gv = *reinterpret_cast<const bf16x8*>(gt + (long)(n0 + row) * M + m);
+        xv = *reinterpret_cast<const bf16x8*>(xt + (long)(k0 + row) * M + m);
       } else {
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const long mm = m + j;
           const bool ok = mm >= m_start && mm < m_end;
-          greg[rr].v[j] = ok ? gt[(long)(n0 + row) * M + mm] : f2bf(0.f);
-          xreg[rr].v[j] = ok ? xt[(long)(k0 + row) * M + mm] : f2bf(0.f);
+          gv.v[j] = ok ? gt[(long)(n0 + row) * M + mm] : f2bf(0.f);
+          xv.v[j] = ok ? xt[(long)(k0 + row) * M + mm] : f2bf(0.f);
         }
       }
+      *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = gv;
+      *reinterpret_cast<bf16x8*>(xb + gg_off(row, c0 * 2)) = xv;
     }
-  };
-  auto write_lds = [&] {
-#pragma unroll
-    for (int rr = 0; rr < 4; ++rr) {
-      const int row = row0 + rr * 32;
-      *reinterpret_cast<bf16x8*>(ga + gg_off(row, c0 * 2)) = greg[rr];
-      *reinterpret_cast<bf16x8*>(xb + gg_off(row, c0 * 2)) = xreg[rr];
-    }
-  };
-
-  const int mc0 = m_start & ~(GG_BK - 1);     // 64-aligned loop start
-  if (m_start < m_end) {
-    issue_loads(mc0);
-    for (int mc = mc0; mc < m_end; mc += GG_BK) {
-    write_lds();
     __syncthreads();
-    if (mc + GG_BK < m_end) issue_loads(mc + GG_BK);
 
 #pragma unroll
     for (int kk = 0; kk < GG_BK / 32; ++kk) {
@@ -418,7 +391,6 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
     }
     __syncthreads();
-    }
   }
 
   const long wbase = (long)e * N * K;

@@ -1,0 +1,33 @@
+"""Grouped-GEMM microbench at Qwen3-30B bench shapes (gpurun)."""
+import torch, time
+from automodel_amd.ops._backend import hip_ops
+from automodel_amd.ops.grouped_gemm import make_group_plan
+
+ops = hip_ops()
+E, T, K8 = 128, 16384, 8
+M = T * K8
+H, I = 2048, 768
+counts = torch.full((E,), M // E, dtype=torch.int32, device="cuda")
+x = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+g = torch.randn(M, I, device="cuda", dtype=torch.bfloat16)
+w = torch.randn(E, I, H, device="cuda", dtype=torch.bfloat16) * 0.02
+offs, tm, ntl = make_group_plan(counts, M)
+gt, xt = g.t().contiguous(), x.t().contiguous()
+
+def timeit(name, fn, flops, iters=10):
+    for _ in range(3): fn()
+    torch.cuda.synchronize(); t0 = time.perf_counter()
+    for _ in range(iters): fn()
+    torch.cuda.synchronize()
+    dt = (time.perf_counter() - t0) / iters
+    print(f"{name}: {dt*1e3:.2f} ms  {flops/dt/1e12:.0f} TF/s")
+
+fl = 2 * M * I * H
+timeit("nt  (y=x@wT)", lambda: ops.grouped_gemm_nt(x, w, offs, tm, ntl), fl)
+timeit("nn  (dx=g@w)", lambda: ops.grouped_gemm_nn(g, w, offs, tm, ntl), fl)
+timeit("tn  (dw=gTx)", lambda: ops.grouped_gemm_tn(gt, xt, offs, E), fl)
+timeit("tn+transpose", lambda: ops.grouped_gemm_tn(g.t().contiguous(), x.t().contiguous(), offs, E), fl)
+# hipBLASLt ceiling: one dense GEMM of the same FLOPs
+xb = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
+wb = torch.randn(I, H, device="cuda", dtype=torch.bfloat16)
+timeit("hipBLASLt NT same-FLOPs", lambda: xb @ wb.t(), fl)
