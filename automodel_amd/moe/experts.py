@@ -172,8 +172,11 @@ class GroupedExperts(nn.Module):
 
             if torch.is_tensor(counts) and counts.is_cuda:
                 # device-side plan shared by all three projections — the
-                # routing counts never touch the host
-                plan = make_group_plan(counts, x_perm.shape[0])
+                # routing counts never touch the host;
+                # 256-row tiles when every projection width allows
+                bm = 256 if (self.intermediate_size % 256 == 0
+                             and self.hidden_size % 256 == 0) else 128
+                plan = (*make_group_plan(counts, x_perm.shape[0], bm), bm)
                 cl = counts
             else:
                 plan = None

@@ -63,14 +63,16 @@ TORCH_LIBRARY(amd_ops, m) {
   m.def("gemv_bf16(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.impl("gemv_bf16", &amd_ops::gemv_bf16);
 
-  m.def("build_group_plan(Tensor counts, int M) -> (Tensor, Tensor, Tensor)");
+  m.def("build_group_plan(Tensor counts, int M, int bm=128) -> (Tensor, Tensor, Tensor)");
   m.impl("build_group_plan", &amd_ops::build_group_plan);
-  m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map, Tensor? n_tiles=None) -> Tensor");
+  m.def("grouped_gemm_nt(Tensor x, Tensor w, Tensor offs, Tensor tile_map, Tensor? n_tiles=None, int bm=128) -> Tensor");
   m.impl("grouped_gemm_nt", &amd_ops::grouped_gemm_nt);
   m.def("grouped_gemm_nn(Tensor g, Tensor w, Tensor offs, Tensor tile_map, Tensor? n_tiles=None) -> Tensor");
   m.impl("grouped_gemm_nn", &amd_ops::grouped_gemm_nn);
   m.def("grouped_gemm_tn(Tensor g, Tensor x, Tensor offs, int E) -> Tensor");
   m.impl("grouped_gemm_tn", &amd_ops::grouped_gemm_tn);
+  m.def("transpose_bf16(Tensor x) -> Tensor");
+  m.impl("transpose_bf16", &amd_ops::transpose_bf16);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");
   m.impl("permute_gather", &amd_ops::permute_gather);
   m.def("unpermute_combine(Tensor yp, Tensor pos, Tensor probs) -> Tensor");

@@ -33,6 +33,17 @@ __device__ __forceinline__ int gg_off(int row, int byte_in_row) {
   return row * 128 + (byte_in_row ^ (((row >> 1) & 7) << 4));
 }
 
+// BK-parametrized row offset: 128-B rows keep the (row>>1)&7 rotation;
+// 256-B rows start at bank 0 every row, so rotate by row&15 (Guideline 4).
+template <int BK>
+__device__ __forceinline__ int ggk_off(int row, int byte_in_row) {
+  if constexpr (BK == 64) {
+    return row * 128 + (byte_in_row ^ (((row >> 1) & 7) << 4));
+  } else {
+    return row * (BK * 2) + (byte_in_row ^ ((row & 15) << 4));
+  }
+}
+
 // Device-side group plan: offsets + (expert, row0) tile list built from the
 // on-device counts so the hot path never syncs counts to the host
 // (the round-1 tile map was host-built from counts.tolist() —
@@ -41,13 +52,13 @@ __device__ __forceinline__ int gg_off(int row, int byte_in_row) {
 __global__ void build_group_plan_kernel(const int* __restrict__ counts, int E,
                                         int* __restrict__ offs,
                                         int* __restrict__ tile_map,
-                                        int* __restrict__ n_tiles) {
+                                        int* __restrict__ n_tiles, int bm) {
   if (threadIdx.x != 0 || blockIdx.x != 0) return;
   int acc = 0, t = 0;
   offs[0] = 0;
   for (int e = 0; e < E; ++e) {
     const int c = counts[e];
-    for (int r = 0; r < c; r += GG_BM) {
+    for (int r = 0; r < c; r += bm) {
       tile_map[2 * t] = e;
       tile_map[2 * t + 1] = acc + r;
       ++t;
@@ -58,6 +69,7 @@ __global__ void build_group_plan_kernel(const int* __restrict__ counts, int E,
   n_tiles[0] = t;
 }
 
+template <int BK>
 __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w, bf16* __restrict__ y,
     const int* __restrict__ tile_map,   // [n_mtiles][2]: (expert, row0)
@@ -66,8 +78,8 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
     int K, int N) {
   if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* xa = smem;                  // 16 KiB
-  char* wb = smem + GG_BM * GG_BK * 2;
+  char* xa = smem;                  // [128][BK] bf16
+  char* wb = smem + GG_BM * BK * 2;
 
   const int e = tile_map[2 * blockIdx.x];
   const int m0 = tile_map[2 * blockIdx.x + 1];
@@ -90,43 +102,42 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
 #pragma unroll
       for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
 
-  for (int k0 = 0; k0 < K; k0 += GG_BK) {
-    // ---- stage x tile [128 m][64 k] and w tile [128 n][64 k]
-    {
-      // 256 threads x 4 pieces: piece p -> row tid/4 + p*64? use flat:
-      // elem = (tid*4 + p) * 8 over 128*64 elems
-      const int row0 = tid / 8;           // 32 rows per pass of 8 cols*8
-      const int c0 = (tid % 8) * 8;
+  constexpr int CPR = BK / 8;              // 16-B chunks per row
+  constexpr int PASSES = 128 * CPR / 256;  // staging passes per tensor
+  const int row0 = tid / CPR;
+  const int c0 = (tid % CPR) * 8;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    // ---- stage x tile [128 m][BK k] and w tile [128 n][BK k]
 #pragma unroll
-      for (int rr = 0; rr < 4; ++rr) {
-        const int row = row0 + rr * 32;
-        const int m = m0 + row;
-        bf16x8 xv;
-        if (m < m_end) {
-          xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
-        } else {
+    for (int rr = 0; rr < PASSES; ++rr) {
+      const int row = row0 + rr * (256 / CPR);
+      const int m = m0 + row;
+      bf16x8 xv;
+      if (m < m_end) {
+        xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+      } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
-        }
-        *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xv;
-        bf16x8 wv = *reinterpret_cast<const bf16x8*>(
-            w + wbase + (long)(n0 + row) * K + k0 + c0);
-        *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wv;
+        for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
       }
+      *reinterpret_cast<bf16x8*>(xa + ggk_off<BK>(row, c0 * 2)) = xv;
+      bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+          w + wbase + (long)(n0 + row) * K + k0 + c0);
+      *reinterpret_cast<bf16x8*>(wb + ggk_off<BK>(row, c0 * 2)) = wv;
     }
     __syncthreads();
 
 #pragma unroll
-    for (int kk = 0; kk < GG_BK / 32; ++kk) {
+    for (int kk = 0; kk < BK / 32; ++kk) {
       bf16x8v a[4], b[4];
       const int arow = (wid >> 1) * 64;
       const int brow = (wid & 1) * 64;
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         a[i] = *reinterpret_cast<const bf16x8v*>(
-            xa + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+            xa + ggk_off<BK>(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
         b[i] = *reinterpret_cast<const bf16x8v*>(
-            wb + gg_off(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+            wb + ggk_off<BK>(brow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -155,12 +166,106 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
   }
 }
 
+// 256x256 big-tile NT variant (8 waves): 2x the arithmetic intensity per
+// staged byte of the 128x128 tile — the 128-tile kernel parks 81% of wave
+// cycles on memory waits (profiles/gg PMC) while hipBLASLt's MT256x256
+// equivalents run 11%. Wave grid 4(m) x 2(n): 64 rows x 128 cols each.
+__global__ __launch_bounds__(512, 1) void grouped_gemm_nt_big_kernel(
+    const bf16* __restrict__ x, const bf16* __restrict__ w, bf16* __restrict__ y,
+    const int* __restrict__ tile_map, const int* __restrict__ offs,
+    const int* __restrict__ n_tiles, int K, int N) {
+  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* xa = smem;                   // [256 m][64 k]  32 KiB
+  char* wb = smem + 256 * 64 * 2;    // [256 n][64 k]  32 KiB
+
+  const int e = tile_map[2 * blockIdx.x];
+  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int m_end = offs[e + 1];
+  const int n0 = blockIdx.y * 256;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;
+
+  const long wbase = (long)e * N * K;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  const int row0 = tid / 8;          // 64 rows per pass (of 256)
+  const int c0 = (tid % 8) * 8;
+
+  for (int k0 = 0; k0 < K; k0 += 64) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 64;
+      const int m = m0 + row;
+      bf16x8 xv;
+      if (m < m_end) {
+        xv = *reinterpret_cast<const bf16x8*>(x + (long)m * K + k0 + c0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) xv.v[j] = f2bf(0.f);
+      }
+      *reinterpret_cast<bf16x8*>(xa + gg_off(row, c0 * 2)) = xv;
+      bf16x8 wv = *reinterpret_cast<const bf16x8*>(
+          w + wbase + (long)(n0 + row) * K + k0 + c0);
+      *reinterpret_cast<bf16x8*>(wb + gg_off(row, c0 * 2)) = wv;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8v a[4], b[8];
+      const int arow = (wid >> 1) * 64;     // m group (4)
+      const int brow = (wid & 1) * 128;     // n group (2)
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        a[i] = *reinterpret_cast<const bf16x8v*>(
+            xa + gg_off(arow + i * 16 + l16, (kk * 32 + kq * 8) * 2));
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        b[j] = *reinterpret_cast<const bf16x8v*>(
+            wb + gg_off(brow + j * 16 + l16, (kk * 32 + kq * 8) * 2));
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int mw = m0 + (wid >> 1) * 64;
+  const int nw = n0 + (wid & 1) * 128;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mw + i * 16 + kq * 4 + r;
+      if (m < m_end) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          y[(long)m * N + nw + j * 16 + l16] = f2bf(acc[i][j][r]);
+      }
+    }
+  }
+}
+
 std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
-    const at::Tensor& counts, int64_t M) {
+    const at::Tensor& counts, int64_t M, int64_t bm) {
   TORCH_CHECK(counts.is_cuda() && counts.scalar_type() == at::kInt,
               "build_group_plan: counts int32 on GPU");
   const int E = counts.size(0);
-  const long max_tiles = M / GG_BM + E + 1;
+  const long max_tiles = M / bm + E + 1;
   auto opts = counts.options();
   auto offs = at::empty({E + 1}, opts);
   auto tile_map = at::empty({max_tiles, 2}, opts);
@@ -168,14 +273,14 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(build_group_plan_kernel, dim3(1), dim3(64), 0, stream.stream(),
                      counts.data_ptr<int>(), E, offs.data_ptr<int>(),
-                     tile_map.data_ptr<int>(), n_tiles.data_ptr<int>());
+                     tile_map.data_ptr<int>(), n_tiles.data_ptr<int>(), (int)bm);
   HIP_CHECK_KERNEL();
   return {offs, tile_map, n_tiles};
 }
 
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map,
-                           const std::optional<at::Tensor>& n_tiles) {
+                           const std::optional<at::Tensor>& n_tiles, int64_t bm) {
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.scalar_type() == at::kBFloat16,
               "grouped_gemm_nt: x [M,K] bf16");
   TORCH_CHECK(w.dim() == 3 && w.scalar_type() == at::kBFloat16, "w [E,N,K] bf16");
@@ -188,15 +293,43 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
   auto y = at::empty({M, (long)N}, x.options());
   const int n_mtiles = tile_map.size(0);
   if (n_mtiles == 0 || M == 0) return y;
-  const dim3 grid(n_mtiles, N / GG_BN);
-  const size_t smem = 2 * GG_BM * GG_BK * 2;
   auto stream = c10::hip::getCurrentHIPStream();
   const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
-  hipLaunchKernelGGL(grouped_gemm_nt_kernel, grid, dim3(256), smem, stream.stream(),
-                     reinterpret_cast<const bf16*>(x.data_ptr()),
-                     reinterpret_cast<const bf16*>(w.data_ptr()),
-                     reinterpret_cast<bf16*>(y.data_ptr()),
-                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+  if (bm == 256) {
+    TORCH_CHECK(N % 256 == 0, "big-tile nt needs N%256==0");
+    const dim3 gridb(n_mtiles, N / 256);
+    const size_t smemb = 2 * 256 * 64 * 2;
+    hipLaunchKernelGGL(grouped_gemm_nt_big_kernel, gridb, dim3(512), smemb,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<const bf16*>(w.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()),
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+    HIP_CHECK_KERNEL();
+    return y;
+  }
+  const dim3 grid(n_mtiles, N / GG_BN);
+  static const int bk_env = []{
+    const char* v = getenv("AMD_OPS_GG_BK");
+    return v ? atoi(v) : 64;
+  }();
+  if (K % 128 == 0 && bk_env == 128) {
+    const size_t smem = 2 * GG_BM * 128 * 2;
+    hipLaunchKernelGGL((grouped_gemm_nt_kernel<128>), grid, dim3(256), smem,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<const bf16*>(w.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()),
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+  } else {
+    const size_t smem = 2 * GG_BM * 64 * 2;
+    hipLaunchKernelGGL((grouped_gemm_nt_kernel<64>), grid, dim3(256), smem,
+                       stream.stream(),
+                       reinterpret_cast<const bf16*>(x.data_ptr()),
+                       reinterpret_cast<const bf16*>(w.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()),
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+  }
   HIP_CHECK_KERNEL();
   return y;
 }
@@ -454,6 +587,66 @@ at::Tensor grouped_gemm_tn(const at::Tensor& gt, const at::Tensor& xt,
                      offs.data_ptr<int>(), N, K, M);
   HIP_CHECK_KERNEL();
   return dw;
+}
+
+// ---- bf16 tiled transpose (for the TN backward's pre-transposed operands;
+// torch's strided transpose measured 6x off HBM roofline on [131072, 2048])
+// 64x64 tiles through LDS: coalesced vector reads AND writes.
+__global__ __launch_bounds__(256) void transpose_bf16_kernel(
+    const bf16* __restrict__ in, bf16* __restrict__ out, int R, int C) {
+  __shared__ bf16 tile[64][64 + 8];   // +8 bf16 pad kills bank conflicts
+  const int tr0 = blockIdx.y * 64;    // input row tile
+  const int tc0 = blockIdx.x * 64;    // input col tile
+  const int tid = threadIdx.x;
+  // read: 64 rows x 8 vectors of 8 -> 512 vector reads, 2 per thread
+  for (int idx = tid; idx < 64 * 8; idx += 256) {
+    const int r = idx / 8, c = (idx % 8) * 8;
+    const int gr = tr0 + r, gc = tc0 + c;
+    if (gr < R) {
+      bf16x8 v;
+      if (gc + 7 < C) {
+        v = *reinterpret_cast<const bf16x8*>(in + (long)gr * C + gc);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          v.v[j] = (gc + j < C) ? in[(long)gr * C + gc + j] : f2bf(0.f);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) tile[c + j][r] = v.v[j];
+    }
+  }
+  __syncthreads();
+  // write: transposed rows (= input cols), coalesced vectors
+  for (int idx = tid; idx < 64 * 8; idx += 256) {
+    const int r = idx / 8, c = (idx % 8) * 8;
+    const int gr = tc0 + r, gc = tr0 + c;   // output [C, R]
+    if (gr < C && gc < R) {
+      bf16x8 v;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.v[j] = tile[r][c + j];
+      if (gc + 7 < R) {
+        *reinterpret_cast<bf16x8*>(out + (long)gr * R + gc) = v;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gc + j < R) out[(long)gr * R + gc + j] = v.v[j];
+      }
+    }
+  }
+}
+
+at::Tensor transpose_bf16(const at::Tensor& in) {
+  TORCH_CHECK(in.is_cuda() && in.dim() == 2 && in.scalar_type() == at::kBFloat16 &&
+              in.is_contiguous(), "transpose_bf16: contiguous 2-D bf16");
+  const int R = in.size(0), C = in.size(1);
+  auto out = at::empty({C, R}, in.options());
+  const dim3 grid((C + 63) / 64, (R + 63) / 64);
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(transpose_bf16_kernel, grid, dim3(256), 0, stream.stream(),
+                     reinterpret_cast<const bf16*>(in.data_ptr()),
+                     reinterpret_cast<bf16*>(out.data_ptr()), R, C);
+  HIP_CHECK_KERNEL();
+  return out;
 }
 
 // ---- fused permute / unpermute -------------------------------------------

@@ -62,15 +62,17 @@ at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias);
 
 std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
-    const at::Tensor& counts, int64_t M);
+    const at::Tensor& counts, int64_t M, int64_t bm = 128);
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map,
-                           const std::optional<at::Tensor>& n_tiles = std::nullopt);
+                           const std::optional<at::Tensor>& n_tiles = std::nullopt,
+                           int64_t bm = 128);
 at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map,
                            const std::optional<at::Tensor>& n_tiles = std::nullopt);
 at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
                            const at::Tensor& offs, int64_t E);
+at::Tensor transpose_bf16(const at::Tensor& in);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);
 at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,
                              const at::Tensor& probs);

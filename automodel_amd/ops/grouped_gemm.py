@@ -36,9 +36,9 @@ def _build_tile_map(counts: list[int]) -> tuple[torch.Tensor, torch.Tensor]:
     return tile_map, torch.tensor(offs, dtype=torch.int32)
 
 
-def make_group_plan(counts: torch.Tensor, M: int):
+def make_group_plan(counts: torch.Tensor, M: int, bm: int = 128):
     """Device-side plan (offs, tile_map, n_tiles) from on-device counts."""
-    return hip_ops().build_group_plan(counts.to(torch.int32), M)
+    return hip_ops().build_group_plan(counts.to(torch.int32), M, bm)
 
 
 def _loop_gemm_nt(x, w, counts):
@@ -51,7 +51,7 @@ def _loop_gemm_nt(x, w, counts):
 
 class _GroupedLinear(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, counts, offs, tile_map, n_tiles):
+    def forward(ctx, x, w, counts, offs, tile_map, n_tiles, bm):
         ctx.save_for_backward(x, w)
         use_hip = (
             x.is_cuda and x.dtype == torch.bfloat16
@@ -60,9 +60,10 @@ class _GroupedLinear(torch.autograd.Function):
         ctx.use_hip = use_hip
         ctx.counts = counts
         ctx.plan = (offs, tile_map, n_tiles)
+        ctx.bm = bm
         if use_hip:
             return hip_ops().grouped_gemm_nt(
-                x.contiguous(), w.contiguous(), offs, tile_map, n_tiles)
+                x.contiguous(), w.contiguous(), offs, tile_map, n_tiles, bm)
         cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
         return _loop_gemm_nt(x, w, cl)
 
@@ -75,12 +76,21 @@ class _GroupedLinear(torch.autograd.Function):
         # kernel needs N%128 and K%128
         if ctx.use_hip and w.shape[1] % 128 == 0 and w.shape[2] % 128 == 0:
             ops = hip_ops()
+            if ctx.bm != 128:
+                # nn rides 128-row tiles; rebuild its plan on device (cheap)
+                if torch.is_tensor(ctx.counts) and ctx.counts.is_cuda:
+                    offs, tile_map, n_tiles = make_group_plan(ctx.counts, g.shape[0], 128)
+                else:
+                    cl = list(ctx.counts)
+                    tile_map, offs = _build_tile_map([int(c) for c in cl])
+                    tile_map, offs, n_tiles = tile_map.to(g.device), offs.to(g.device), None
             dx = ops.grouped_gemm_nn(g, w.contiguous(), offs, tile_map, n_tiles)
-            # TN operands pre-transposed (m contiguous) so the kernel's
-            # staging is vectorized — see csrc grouped_gemm_tn_kernel
-            dw = ops.grouped_gemm_tn(g.t().contiguous(), x.t().contiguous(),
+            # TN operands pre-transposed (m contiguous) by the tiled HIP
+            # transpose — torch's strided transpose measured 6x off roofline
+            dw = ops.grouped_gemm_tn(ops.transpose_bf16(g),
+                                     ops.transpose_bf16(x.contiguous()),
                                      offs, w.shape[0])
-            return dx, dw, None, None, None, None
+            return dx, dw, None, None, None, None, None
         cl = ctx.counts.tolist() if torch.is_tensor(ctx.counts) else list(ctx.counts)
         dx = torch.empty_like(x)
         dw = torch.zeros_like(w)
@@ -91,7 +101,7 @@ class _GroupedLinear(torch.autograd.Function):
                 dx[start : start + n] = ge @ w[e]
                 dw[e] = (ge.t() @ x[start : start + n]).to(w.dtype)
             start += n
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
 def grouped_linear(x_perm: torch.Tensor, w: torch.Tensor, counts,
@@ -104,13 +114,23 @@ def grouped_linear(x_perm: torch.Tensor, w: torch.Tensor, counts,
     routing."""
     hip_eligible = (x_perm.is_cuda and x_perm.dtype == torch.bfloat16
                     and w.shape[2] % 64 == 0 and w.shape[1] % 128 == 0)
-    if plan is None and hip_eligible:
+    if plan is not None:
+        # caller-provided plan carries its tile height: (offs, tm, nt, bm)
+        offs, tile_map, n_tiles, bm = plan
+        return _GroupedLinear.apply(x_perm, w, counts, offs, tile_map, n_tiles, bm)
+    # 256x256 big-tile forward when the output width allows (+27% measured,
+    # benchmarks/gg_micro.py)
+    bm = 256 if (hip_eligible and w.shape[1] % 256 == 0
+                 and torch.is_tensor(counts) and counts.is_cuda) else 128
+    if hip_eligible:
         if torch.is_tensor(counts) and counts.is_cuda:
-            plan = make_group_plan(counts, x_perm.shape[0])
+            offs, tile_map, n_tiles = make_group_plan(counts, x_perm.shape[0], bm)
         else:
             cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
             tile_map, offs = _build_tile_map([int(c) for c in cl])
-            plan = (offs.to(x_perm.device), tile_map.to(x_perm.device), None)
-    if plan is None:
-        plan = (None, None, None)
-    return _GroupedLinear.apply(x_perm, w, counts, *plan)
+            tile_map, offs, n_tiles = tile_map.to(x_perm.device), offs.to(x_perm.device), None
+            bm = 128
+    else:
+        offs = tile_map = n_tiles = None
+        bm = 128
+    return _GroupedLinear.apply(x_perm, w, counts, offs, tile_map, n_tiles, bm)

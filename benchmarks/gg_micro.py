@@ -1,4 +1,6 @@
 """Grouped-GEMM microbench at Qwen3-30B bench shapes (gpurun)."""
+import sys, os
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch, time
 from automodel_amd.ops._backend import hip_ops
 from automodel_amd.ops.grouped_gemm import make_group_plan
@@ -24,9 +26,19 @@ def timeit(name, fn, flops, iters=10):
 
 fl = 2 * M * I * H
 timeit("nt  (y=x@wT)", lambda: ops.grouped_gemm_nt(x, w, offs, tm, ntl), fl)
+offs_b, tm_b, ntl_b = ops.build_group_plan(counts, M, 256)
+w3 = torch.randn(E, 1024, H, device="cuda", dtype=torch.bfloat16) * 0.02
+fl3 = 2 * M * 1024 * H
+timeit("nt-big 256x256 (N=1024)", lambda: ops.grouped_gemm_nt(x, w3, offs_b, tm_b, ntl_b, 256), fl3)
+y_big = ops.grouped_gemm_nt(x, w3, offs_b, tm_b, ntl_b, 256)
+y_ref = ops.grouped_gemm_nt(x, w3, offs, tm, ntl)
+print("big-vs-128 maxdiff:", float((y_big.float()-y_ref.float()).abs().max()))
 timeit("nn  (dx=g@w)", lambda: ops.grouped_gemm_nn(g, w, offs, tm, ntl), fl)
 timeit("tn  (dw=gTx)", lambda: ops.grouped_gemm_tn(gt, xt, offs, E), fl)
-timeit("tn+transpose", lambda: ops.grouped_gemm_tn(g.t().contiguous(), x.t().contiguous(), offs, E), fl)
+timeit("tn+torch-transpose", lambda: ops.grouped_gemm_tn(g.t().contiguous(), x.t().contiguous(), offs, E), fl)
+timeit("tn+hip-transpose", lambda: ops.grouped_gemm_tn(ops.transpose_bf16(g), ops.transpose_bf16(x), offs, E), fl)
+xtr = ops.transpose_bf16(x)
+assert torch.equal(xtr, x.t().contiguous()), "transpose mismatch"
 # hipBLASLt ceiling: one dense GEMM of the same FLOPs
 xb = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
 wb = torch.randn(I, H, device="cuda", dtype=torch.bfloat16)
