@@ -41,6 +41,8 @@ TORCH_LIBRARY(amd_ops, m) {
 
   m.def("mfma_probe(Tensor a, Tensor b) -> Tensor");
   m.impl("mfma_probe", &amd_ops::mfma_probe);
+  m.def("tr16_probe(Tensor pattern) -> Tensor");
+  m.impl("tr16_probe", &amd_ops::tr16_probe);
 
   m.def("ce_fwd_logits(Tensor logits, Tensor labels, Tensor(a!) loss_sum) -> (Tensor, Tensor)");
   m.impl("ce_fwd_logits", &amd_ops::ce_fwd_logits);

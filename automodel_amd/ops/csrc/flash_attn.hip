@@ -862,6 +862,42 @@ __global__ void mfma_probe_kernel(const bf16* a, const bf16* b, float* d) {
   }
 }
 
+// ---- ds_read_b64_tr_b16 layout probe: fill LDS with a linear pattern and
+// dump what each lane receives for reads at stride-8B lane addresses.
+// Guide T10: the gather is hardware-defined; verify on silicon before
+// building kernels on it (cdna_hip_programming.md §5.4 rule 27).
+__global__ void tr16_probe_kernel(const bf16* __restrict__ in, float* __restrict__ out,
+                                  int n) {
+  __shared__ __attribute__((aligned(16))) bf16 lds[2048];
+  const int l = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < n; i += 64) lds[i] = in[i];
+  __syncthreads();
+  // two reads cover 1024 B: lanes at 8-B stride
+  for (int rd = 0; rd < 2; ++rd) {
+    unsigned off = (unsigned)(l * 8 + rd * 512);
+    unsigned long long v;
+    asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
+                 : "=v"(v) : "v"(off));
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      unsigned short u = (unsigned short)(v >> (16 * j));
+      out[(rd * 64 + l) * 4 + j] = bf2f(__builtin_bit_cast(bf16, u));
+    }
+  }
+}
+
+at::Tensor tr16_probe(const at::Tensor& pattern) {
+  TORCH_CHECK(pattern.is_cuda() && pattern.scalar_type() == at::kBFloat16 &&
+              pattern.numel() <= 2048, "tr16_probe: bf16, <=2048 elems");
+  auto out = at::empty({2, 64, 4}, pattern.options().dtype(at::kFloat));
+  auto stream = c10::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0, stream.stream(),
+                     reinterpret_cast<const bf16*>(pattern.data_ptr()),
+                     out.data_ptr<float>(), (int)pattern.numel());
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b) {
   TORCH_CHECK(a.is_cuda() && a.sizes() == at::IntArrayRef({32, 16}), "a must be [32,16]");
   TORCH_CHECK(b.sizes() == at::IntArrayRef({16, 32}), "b must be [16,32]");

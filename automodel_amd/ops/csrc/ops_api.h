@@ -37,6 +37,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> flash_attn_bwd(
     const std::optional<at::Tensor>& doc_end = std::nullopt);
 
 at::Tensor mfma_probe(const at::Tensor& a, const at::Tensor& b);
+at::Tensor tr16_probe(const at::Tensor& pattern);
 
 std::tuple<at::Tensor, at::Tensor> ce_fwd_logits(const at::Tensor& logits,
                                                  const at::Tensor& labels,
