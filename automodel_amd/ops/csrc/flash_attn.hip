@@ -59,6 +59,76 @@ __device__ __forceinline__ int row_lds_off(int row, int byte_in_row, int stride_
 __device__ __forceinline__ int vt_lds_off(int d, int byte_in_row) {
   return d * 64 + (byte_in_row ^ ((((d >> 2) ^ (d >> 3)) & 3) << 4));
 }
+
+// ---- gfx950 hardware-transpose (ds_read_b64_tr_b16) tile images -----------
+// Probe-verified semantics (benchmarks/tr16_map.py, guide T10): each 16-lane
+// group reads a contiguous 128-B region as a [4 row][16 col] bf16 block and
+// lane g receives COLUMN g (one value per row). A [32 k][D] tile stored as
+// [D/16 subtiles][32 k rows][16 d cols] therefore serves BOTH access
+// patterns with no scalar transposes:
+//   * row-chunk b128 reads (A-style: lane = k row, 8 consecutive d)
+//   * transposed B-fragments via two tr16 reads (lane = d col, 8 k rows)
+// Subtile stride 1152 B (not 1024): adjacent subtiles land 32 banks apart,
+// so the two 16-lane halves of a wave (reading subtiles 2t and 2t+1) never
+// share a bank window (b64 conflict groups are 32 lanes).
+__device__ __forceinline__ int sub_off(int k, int d) {
+  return (d >> 4) * 1152 + k * 32 + (d & 15) * 2;
+}
+template <int D>
+__host__ __device__ constexpr int sub_img_bytes() { return (D / 16) * 1152; }
+
+__device__ __forceinline__ unsigned lds_addr(const void* p) {
+  return (unsigned)(unsigned long long)(uintptr_t)p;
+}
+
+// B-fragment for mfma_32x32x16: element j = T[k = 8*khalf + j][col =
+// tile32*32 + (lane&31)] from a subtiled image, via two hardware-transpose
+// reads (k rows 8h..8h+3 and 8h+4..8h+7). The trailing s_waitcnt rides in
+// the second asm so the consuming mfma (which needs both halves) cannot be
+// scheduled before the data lands.
+// pipelined form: issue one fragment's two reads; wait-and-take later with
+// a counted lgkmcnt (the "+v" operands order the consuming MFMAs after it)
+__device__ __forceinline__ void tr16_issue2(unsigned addr, unsigned long long& r0,
+                                            unsigned long long& r1) {
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\tds_read_b64_tr_b16 %1, %2 offset:128"
+               : "=v"(r0), "=v"(r1) : "v"(addr));
+}
+__device__ __forceinline__ bf16x8_v tr16_take(unsigned long long& r0,
+                                              unsigned long long& r1, bool newer2) {
+  if (newer2) {
+    asm volatile("s_waitcnt lgkmcnt(2)" : "+v"(r0), "+v"(r1));
+  } else {
+    asm volatile("s_waitcnt lgkmcnt(0)" : "+v"(r0), "+v"(r1));
+  }
+  ulonglong2 u{r0, r1};
+  return __builtin_bit_cast(bf16x8_v, u);
+}
+
+__device__ __forceinline__ bf16x8_v tr16_frag(unsigned img_base, int tile32,
+                                              int khalf, int lane) {
+  const unsigned a0 = img_base + (unsigned)((tile32 * 2 + ((lane >> 4) & 1)) * 1152 +
+                                            (8 * khalf) * 32 + (lane & 15) * 8);
+  unsigned long long v0, v1;
+  // single asm block: outputs are fully landed when it retires, so the
+  // register allocator can never spill an in-flight read (a real silent-
+  // corruption hazard caught at (192,192) under register pressure)
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\t"
+               "ds_read_b64_tr_b16 %1, %2 offset:128\n\t"
+               "s_waitcnt lgkmcnt(0)"
+               : "=v"(v0), "=v"(v1) : "v"(a0));
+  ulonglong2 u{v0, v1};
+  return __builtin_bit_cast(bf16x8_v, u);
+}
+
+// Pipelining across other code is only safe when nothing spills: gate the
+// pipelined tr16 paths to instantiations comfortably inside the register
+// budget; the big-D variants take the self-contained tr16_frag instead.
+__host__ __device__ constexpr bool pipe_ok_dkv(int DQK, int DV) {
+  return 3 * (DQK + DV) / 4 + 16 <= 208;
+}
+__host__ __device__ constexpr bool pipe_ok_dq(int DQK, int DV) {
+  return DQK / 4 + DV / 4 + DQK / 2 + 48 <= 208;
+}
 // per-wave transpose buffer: [32][64 B]
 __device__ __forceinline__ int tb_off(int row, int byte_in_row) {
   return row * 64 + (byte_in_row ^ ((((row >> 2) ^ (row >> 3)) & 3) << 4));
@@ -86,8 +156,8 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
   constexpr int KB = pad256(DQK * 2);     // K tile row stride (bytes)
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                               // KVBLK*KB
-  char* vt_lds = smem + KVBLK * KB;                 // DV*64
-  float* bcast = reinterpret_cast<float*>(smem + KVBLK * KB + DV * 64);
+  char* v_img = smem + KVBLK * KB;                  // subtiled [DV/16][32][16]
+  float* bcast = reinterpret_cast<float*>(smem + KVBLK * KB + sub_img_bytes<DV>());
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE_SIZE - 1);
@@ -151,11 +221,10 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
       for (int idx = tid; idx < KVBLK * VCH; idx += WAVES * WAVE_SIZE) {
         const int row = idx / VCH, c0 = (idx % VCH) * 8;
         const int ks = min(k0 + row, Skv - 1);
-        bf16x8 vv8 = *reinterpret_cast<const bf16x8*>(v + v_base + (long)ks * v_rs + c0);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {  // transpose store: vt[d][k]
-          *reinterpret_cast<bf16*>(vt_lds + vt_lds_off(c0 + j, row * 2)) = vv8.v[j];
-        }
+        // vectorized subtile store (the round-1 path scatter-stored 8
+        // scalars per vector to build a transposed image)
+        *reinterpret_cast<bf16x8*>(v_img + sub_off(row, c0)) =
+            *reinterpret_cast<const bf16x8*>(v + v_base + (long)ks * v_rs + c0);
       }
     }
     __syncthreads();
@@ -234,14 +303,57 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
           o_acc[t][r] *= bcast[wid * 32 + row];
         }
       }
+      // software-pipelined tr16 PV: tile t+1's four transpose reads are in
+      // flight during tile t's MFMAs; the counted s_waitcnt takes the
+      // fragment registers as "+v" operands so the consuming MFMAs cannot
+      // be scheduled past it. lgkmcnt(4) is conservative-safe: compiler-
+      // issued LDS ops between ours only strengthen the wait.
+      const unsigned v_base_a = lds_addr(v_img);
+      if constexpr (DV > 128) {
+#pragma unroll
+        for (int t = 0; t < DV / 32; ++t) {
+          bf16x8_v vb0 = tr16_frag(v_base_a, t, half, lane);
+          bf16x8_v vb1 = tr16_frag(v_base_a, t, half + 2, lane);
+          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, vb0, o_acc[t], 0, 0, 0);
+          o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, vb1, o_acc[t], 0, 0, 0);
+        }
+      } else {
+      unsigned long long f0a, f0b, f1a, f1b, g0a, g0b, g1a, g1b;
+      auto tr_issue = [&](int t, unsigned long long& r0a, unsigned long long& r0b,
+                          unsigned long long& r1a, unsigned long long& r1b) {
+        const unsigned base = v_base_a +
+            (unsigned)((t * 2 + ((lane >> 4) & 1)) * 1152 + (lane & 15) * 8);
+        const unsigned klo = (unsigned)(8 * half) * 32;
+        const unsigned khi = (unsigned)(8 * (half + 2)) * 32;
+        asm volatile("ds_read_b64_tr_b16 %0, %4\n\t"
+                     "ds_read_b64_tr_b16 %1, %4 offset:128\n\t"
+                     "ds_read_b64_tr_b16 %2, %5\n\t"
+                     "ds_read_b64_tr_b16 %3, %5 offset:128"
+                     : "=v"(r0a), "=v"(r0b), "=v"(r1a), "=v"(r1b)
+                     : "v"(base + klo), "v"(base + khi));
+      };
+      tr_issue(0, f0a, f0b, f1a, f1b);
 #pragma unroll
       for (int t = 0; t < DV / 32; ++t) {
-        bf16x8_v vb0 = *reinterpret_cast<const bf16x8_v*>(
-            vt_lds + vt_lds_off(t * 32 + col, (half * 8) * 2));
-        bf16x8_v vb1 = *reinterpret_cast<const bf16x8_v*>(
-            vt_lds + vt_lds_off(t * 32 + col, (16 + half * 8) * 2));
+        unsigned long long& c0a = (t & 1) ? g0a : f0a;
+        unsigned long long& c0b = (t & 1) ? g0b : f0b;
+        unsigned long long& c1a = (t & 1) ? g1a : f1a;
+        unsigned long long& c1b = (t & 1) ? g1b : f1b;
+        if (t + 1 < DV / 32) {
+          tr_issue(t + 1, (t & 1) ? f0a : g0a, (t & 1) ? f0b : g0b,
+                   (t & 1) ? f1a : g1a, (t & 1) ? f1b : g1b);
+          asm volatile("s_waitcnt lgkmcnt(4)"
+                       : "+v"(c0a), "+v"(c0b), "+v"(c1a), "+v"(c1b));
+        } else {
+          asm volatile("s_waitcnt lgkmcnt(0)"
+                       : "+v"(c0a), "+v"(c0b), "+v"(c1a), "+v"(c1b));
+        }
+        ulonglong2 u0{c0a, c0b}, u1{c1a, c1b};
+        bf16x8_v vb0 = __builtin_bit_cast(bf16x8_v, u0);
+        bf16x8_v vb1 = __builtin_bit_cast(bf16x8_v, u1);
         o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa0, vb0, o_acc[t], 0, 0, 0);
         o_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa1, vb1, o_acc[t], 0, 0, 0);
+      }
       }
     }
     __syncthreads();
@@ -275,7 +387,8 @@ static void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tenso
                        int B, int Sq, int Skv, int Hq, int Hk, int q_start,
                        float scale, bool causal, hipStream_t stream) {
   const dim3 grid(Sq / (WAVES * QBLK), Hq, B);
-  const size_t smem = KVBLK * pad256(DQK * 2) + DV * 64 + WAVES * 32 * sizeof(float);
+  const size_t smem = KVBLK * pad256(DQK * 2) + sub_img_bytes<DV>() +
+                      WAVES * 32 * sizeof(float);
   hipLaunchKernelGGL((flash_fwd_kernel<DQK, DV, WAVES, VARLEN>), grid,
                      dim3(WAVES * WAVE_SIZE), smem, stream,
                      reinterpret_cast<const bf16*>(q.data_ptr()),
@@ -368,14 +481,10 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dk, bf16* __restrict__ dv, const int* __restrict__ doc_end,
     int B, int Sq, int Skv, int Hq, int Hk, int q_start, float scale, bool causal) {
-  constexpr int QB = pad256(DQK * 2);   // q-row image stride
-  constexpr int OB = pad256(DV * 2);    // dO-row image stride
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* q_rows = smem;                                   // 32*QB
-  char* do_rows = smem + 32 * QB;                        // 32*OB
-  char* qt = smem + 32 * QB + 32 * OB;                   // DQK*64
-  char* dot = qt + DQK * 64;                             // DV*64
-  char* tbuf = dot + DV * 64;                            // WAVES*2048
+  char* q_img = smem;                                    // subtiled [DQK/16][32][16]
+  char* do_img = smem + sub_img_bytes<DQK>();            // subtiled [DV/16][32][16]
+  char* tbuf = do_img + sub_img_bytes<DV>();             // WAVES*2048
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -445,26 +554,22 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
 
     for (int jq = jq_start; jq < jq_end; ++jq) {
       const int q0 = jq * 32;
-      // ---- cooperative stage: q/do rows + transposed images
+      // ---- cooperative stage: ONE subtiled image per tensor (serves both
+      // the row-chunk B-frags and the tr16 transposed B-frags — the round-1
+      // kernel kept a row image AND a scalar-scattered transposed image)
       {
         constexpr int QCH = DQK / 8;
         for (int idx = tid; idx < 32 * QCH; idx += WAVES * 64) {
           const int rr = idx / QCH, c0 = (idx % QCH) * 8;
-          bf16x8 qv = *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
-          *reinterpret_cast<bf16x8*>(q_rows + row_lds_off(rr, c0 * 2, QB)) = qv;
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            *reinterpret_cast<bf16*>(qt + vt_lds_off(c0 + j, rr * 2)) = qv.v[j];
+          *reinterpret_cast<bf16x8*>(q_img + sub_off(rr, c0)) =
+              *reinterpret_cast<const bf16x8*>(q + q_base + (long)(q0 + rr) * q_rs + c0);
         }
         constexpr int OCH = DV / 8;
         for (int idx = tid; idx < 32 * OCH; idx += WAVES * 64) {
           const int rr = idx / OCH, c0 = (idx % OCH) * 8;
-          bf16x8 dv8 = *reinterpret_cast<const bf16x8*>(
-              dout + do_base + (long)(q0 + rr) * do_rs + c0);
-          *reinterpret_cast<bf16x8*>(do_rows + row_lds_off(rr, c0 * 2, OB)) = dv8;
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            *reinterpret_cast<bf16*>(dot + vt_lds_off(c0 + j, rr * 2)) = dv8.v[j];
+          *reinterpret_cast<bf16x8*>(do_img + sub_off(rr, c0)) =
+              *reinterpret_cast<const bf16x8*>(
+                  dout + do_base + (long)(q0 + rr) * do_rs + c0);
         }
       }
       __syncthreads();
@@ -484,7 +589,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
 #pragma unroll
         for (int c = 0; c < DQK / 16; ++c) {
           bf16x8_v qb = *reinterpret_cast<const bf16x8_v*>(
-              q_rows + row_lds_off(col, (c * 16 + half * 8) * 2, QB));
+              q_img + sub_off(col, c * 16 + half * 8));
           st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag[c], qb, st, 0, 0, 0);
         }
         // ---- dP^T = V dO^T
@@ -494,7 +599,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
 #pragma unroll
         for (int c = 0; c < DV / 16; ++c) {
           bf16x8_v db = *reinterpret_cast<const bf16x8_v*>(
-              do_rows + row_lds_off(col, (c * 16 + half * 8) * 2, OB));
+              do_img + sub_off(col, c * 16 + half * 8));
           dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vfrag[c], db, dpt, 0, 0, 0);
         }
 
@@ -516,15 +621,39 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
           *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(st[r]);
         }
         __builtin_amdgcn_wave_barrier();
+        const unsigned do_base_a = lds_addr(do_img);
+        if constexpr (!pipe_ok_dkv(DQK, DV)) {
 #pragma unroll
-        for (int c2 = 0; c2 < 2; ++c2) {
-          bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
-              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+          for (int c2 = 0; c2 < 2; ++c2) {
+            bf16x8_v pa = *reinterpret_cast<const bf16x8_v*>(
+                tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-          for (int t = 0; t < DV / 32; ++t) {
-            bf16x8_v dob = *reinterpret_cast<const bf16x8_v*>(
-                dot + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
-            dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
+            for (int t = 0; t < DV / 32; ++t) {
+              bf16x8_v dob = tr16_frag(do_base_a, t, c2 * 2 + half, lane);
+              dv_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(pa, dob, dv_acc[t], 0, 0, 0);
+            }
+          }
+        } else {
+          bf16x8_v paf[2];
+#pragma unroll
+          for (int c2 = 0; c2 < 2; ++c2)
+            paf[c2] = *reinterpret_cast<const bf16x8_v*>(
+                tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+          constexpr int NT = DV / 32;
+          unsigned long long pA, pB, qA, qB;
+          auto addr_f = [&](int i) {
+            const int c2 = i / NT, t = i % NT;
+            return do_base_a + (unsigned)((t * 2 + ((lane >> 4) & 1)) * 1152 +
+                                          (8 * (c2 * 2 + half)) * 32 + (lane & 15) * 8);
+          };
+          tr16_issue2(addr_f(0), pA, pB);
+#pragma unroll
+          for (int i = 0; i < 2 * NT; ++i) {
+            const bool last = (i + 1 == 2 * NT);
+            if (!last) tr16_issue2(addr_f(i + 1), (i & 1) ? pA : qA, (i & 1) ? pB : qB);
+            bf16x8_v dob = tr16_take((i & 1) ? qA : pA, (i & 1) ? qB : pB, !last);
+            dv_acc[i % NT] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                paf[i / NT], dob, dv_acc[i % NT], 0, 0, 0);
           }
         }
         // ---- transpose dS^T, accumulate dK
@@ -535,15 +664,39 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dkv(DQK, DV)) void flash_bwd
           *reinterpret_cast<bf16*>(tb + tb_off(krow, col * 2)) = f2bf(dpt[r]);
         }
         __builtin_amdgcn_wave_barrier();
+        const unsigned q_base_a = lds_addr(q_img);
+        if constexpr (!pipe_ok_dkv(DQK, DV)) {
 #pragma unroll
-        for (int c2 = 0; c2 < 2; ++c2) {
-          bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
-              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+          for (int c2 = 0; c2 < 2; ++c2) {
+            bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
+                tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-          for (int t = 0; t < DQK / 32; ++t) {
-            bf16x8_v qb2 = *reinterpret_cast<const bf16x8_v*>(
-                qt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
-            dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
+            for (int t = 0; t < DQK / 32; ++t) {
+              bf16x8_v qb2 = tr16_frag(q_base_a, t, c2 * 2 + half, lane);
+              dk_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, qb2, dk_acc[t], 0, 0, 0);
+            }
+          }
+        } else {
+          bf16x8_v daf[2];
+#pragma unroll
+          for (int c2 = 0; c2 < 2; ++c2)
+            daf[c2] = *reinterpret_cast<const bf16x8_v*>(
+                tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+          constexpr int NT = DQK / 32;
+          unsigned long long pA, pB, qA, qB;
+          auto addr_f = [&](int i) {
+            const int c2 = i / NT, t = i % NT;
+            return q_base_a + (unsigned)((t * 2 + ((lane >> 4) & 1)) * 1152 +
+                                         (8 * (c2 * 2 + half)) * 32 + (lane & 15) * 8);
+          };
+          tr16_issue2(addr_f(0), pA, pB);
+#pragma unroll
+          for (int i = 0; i < 2 * NT; ++i) {
+            const bool last = (i + 1 == 2 * NT);
+            if (!last) tr16_issue2(addr_f(i + 1), (i & 1) ? pA : qA, (i & 1) ? pB : qB);
+            bf16x8_v qb2 = tr16_take((i & 1) ? qA : pA, (i & 1) ? qB : pB, !last);
+            dk_acc[i % NT] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+                daf[i / NT], qb2, dk_acc[i % NT], 0, 0, 0);
           }
         }
       }
@@ -574,13 +727,10 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_
     const bf16* __restrict__ v, const float* __restrict__ lse, const float* __restrict__ delta,
     bf16* __restrict__ dq, const int* __restrict__ doc_start, int B, int Sq, int Skv,
     int Hq, int Hk, int q_start, float scale, bool causal) {
-  constexpr int KBB = pad256(DQK * 2);
-  constexpr int VBB = pad256(DV * 2);
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_rows = smem;                        // 32*KBB
-  char* v_rows = smem + 32 * KBB;             // 32*VBB
-  char* kt = v_rows + 32 * VBB;               // DQK*64
-  char* tbuf = kt + DQK * 64;                 // WAVES*2048
+  char* k_img = smem;                                  // subtiled [DQK/16][32][16]
+  char* v_img = smem + sub_img_bytes<DQK>();           // subtiled [DV/16][32][16]
+  char* tbuf = v_img + sub_img_bytes<DV>();            // WAVES*2048
   float* stats = reinterpret_cast<float*>(tbuf + WAVES * 2048);  // 3*WAVES*32
 
   const int tid = threadIdx.x;
@@ -635,23 +785,21 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_
 
   for (int jk = jk0; jk < n_tiles; ++jk) {
     const int k0 = jk * 32;
-    // ---- stage K rows, V rows, K^T
+    // ---- stage K and V as subtiled images (one image serves both the
+    // row-chunk and the tr16 transposed reads)
     {
       constexpr int KCH = DQK / 8;
       for (int idx = tid; idx < 32 * KCH; idx += WAVES * 64) {
         const int rr = idx / KCH, c0 = (idx % KCH) * 8;
         const int ks = min(k0 + rr, Skv - 1);  // overhang masked in compute
-        bf16x8 kv8 = *reinterpret_cast<const bf16x8*>(k + k_base + (long)ks * k_rs + c0);
-        *reinterpret_cast<bf16x8*>(k_rows + row_lds_off(rr, c0 * 2, KBB)) = kv8;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          *reinterpret_cast<bf16*>(kt + vt_lds_off(c0 + j, rr * 2)) = kv8.v[j];
+        *reinterpret_cast<bf16x8*>(k_img + sub_off(rr, c0)) =
+            *reinterpret_cast<const bf16x8*>(k + k_base + (long)ks * k_rs + c0);
       }
       constexpr int VCH = DV / 8;
       for (int idx = tid; idx < 32 * VCH; idx += WAVES * 64) {
         const int rr = idx / VCH, c0 = (idx % VCH) * 8;
         const int ks = min(k0 + rr, Skv - 1);
-        *reinterpret_cast<bf16x8*>(v_rows + row_lds_off(rr, c0 * 2, VBB)) =
+        *reinterpret_cast<bf16x8*>(v_img + sub_off(rr, c0)) =
             *reinterpret_cast<const bf16x8*>(v + v_base + (long)ks * v_rs + c0);
       }
     }
@@ -667,7 +815,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_
 #pragma unroll
       for (int c = 0; c < DQK / 16; ++c) {
         bf16x8_v kb = *reinterpret_cast<const bf16x8_v*>(
-            k_rows + row_lds_off(col, (c * 16 + half * 8) * 2, KBB));
+            k_img + sub_off(col, c * 16 + half * 8));
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfrag[c], kb, s, 0, 0, 0);
       }
       // ---- dP[q][k] = dO V^T
@@ -677,7 +825,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_
 #pragma unroll
       for (int c = 0; c < DV / 16; ++c) {
         bf16x8_v vb = *reinterpret_cast<const bf16x8_v*>(
-            v_rows + row_lds_off(col, (c * 16 + half * 8) * 2, VBB));
+            v_img + sub_off(col, c * 16 + half * 8));
         dp = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofrag[c], vb, dp, 0, 0, 0);
       }
 
@@ -699,15 +847,39 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, occ_dq(DQK, DV)) void flash_bwd_
         *reinterpret_cast<bf16*>(tb + tb_off(qrow, col * 2)) = f2bf(s[r]);
       }
       __builtin_amdgcn_wave_barrier();
+      const unsigned k_base_a = lds_addr(k_img);
+      if constexpr (!pipe_ok_dq(DQK, DV)) {
 #pragma unroll
-      for (int c2 = 0; c2 < 2; ++c2) {
-        bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
-            tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+        for (int c2 = 0; c2 < 2; ++c2) {
+          bf16x8_v da = *reinterpret_cast<const bf16x8_v*>(
+              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
 #pragma unroll
-        for (int t = 0; t < DQK / 32; ++t) {
-          bf16x8_v ktb = *reinterpret_cast<const bf16x8_v*>(
-              kt + vt_lds_off(t * 32 + col, (c2 * 16 + half * 8) * 2));
-          dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, ktb, dq_acc[t], 0, 0, 0);
+          for (int t = 0; t < DQK / 32; ++t) {
+            bf16x8_v ktb = tr16_frag(k_base_a, t, c2 * 2 + half, lane);
+            dq_acc[t] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(da, ktb, dq_acc[t], 0, 0, 0);
+          }
+        }
+      } else {
+        bf16x8_v daf[2];
+#pragma unroll
+        for (int c2 = 0; c2 < 2; ++c2)
+          daf[c2] = *reinterpret_cast<const bf16x8_v*>(
+              tb + tb_off(col, (c2 * 16 + half * 8) * 2));
+        constexpr int NT = DQK / 32;
+        unsigned long long pA, pB, qA, qB;
+        auto addr_f = [&](int i) {
+          const int c2 = i / NT, t = i % NT;
+          return k_base_a + (unsigned)((t * 2 + ((lane >> 4) & 1)) * 1152 +
+                                       (8 * (c2 * 2 + half)) * 32 + (lane & 15) * 8);
+        };
+        tr16_issue2(addr_f(0), pA, pB);
+#pragma unroll
+        for (int i = 0; i < 2 * NT; ++i) {
+          const bool last = (i + 1 == 2 * NT);
+          if (!last) tr16_issue2(addr_f(i + 1), (i & 1) ? pA : qA, (i & 1) ? pB : qB);
+          bf16x8_v ktb = tr16_take((i & 1) ? qA : pA, (i & 1) ? qB : pB, !last);
+          dq_acc[i % NT] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              daf[i / NT], ktb, dq_acc[i % NT], 0, 0, 0);
         }
       }
       __builtin_amdgcn_wave_barrier();
@@ -735,8 +907,7 @@ static void launch_dkv(const at::Tensor& dout, const at::Tensor& q, const at::Te
                        int Skv, int Hq, int Hk, int q_start, float scale, bool causal,
                        hipStream_t stream) {
   const dim3 grid(Skv / (WAVES * 32), Hk, B);
-  const size_t smem = 32 * pad256(DQK * 2) + 32 * pad256(DV * 2) + DQK * 64 + DV * 64 +
-                      WAVES * 2048;
+  const size_t smem = sub_img_bytes<DQK>() + sub_img_bytes<DV>() + WAVES * 2048;
   hipLaunchKernelGGL((flash_bwd_dkv_kernel<DQK, DV, WAVES, VARLEN>), grid,
                      dim3(WAVES * WAVE_SIZE), smem, stream,
                      reinterpret_cast<const bf16*>(dout.data_ptr()),
@@ -756,7 +927,7 @@ static void launch_dq(const at::Tensor& dout, const at::Tensor& q, const at::Ten
                       at::Tensor& dq, const int* doc_start, int B, int Sq, int Skv, int Hq,
                       int Hk, int q_start, float scale, bool causal, hipStream_t stream) {
   const dim3 grid(Sq / (WAVES * 32), Hq, B);
-  const size_t smem = 32 * pad256(DQK * 2) + 32 * pad256(DV * 2) + DQK * 64 +
+  const size_t smem = sub_img_bytes<DQK>() + sub_img_bytes<DV>() +
                       WAVES * 2048 + 3 * WAVES * 32 * sizeof(float);
   hipLaunchKernelGGL((flash_bwd_dq_kernel<DQK, DV, WAVES, VARLEN>), grid,
                      dim3(WAVES * WAVE_SIZE), smem, stream,
@@ -872,9 +1043,10 @@ __global__ void tr16_probe_kernel(const bf16* __restrict__ in, float* __restrict
   const int l = threadIdx.x & 63;
   for (int i = threadIdx.x; i < n; i += 64) lds[i] = in[i];
   __syncthreads();
-  // two reads cover 1024 B: lanes at 8-B stride
+  // two reads cover 1024 B: lanes at 8-B stride (address = LDS base + offset)
+  const unsigned base = (unsigned)(unsigned long long)(uintptr_t)(&lds[0]);
   for (int rd = 0; rd < 2; ++rd) {
-    unsigned off = (unsigned)(l * 8 + rd * 512);
+    unsigned off = base + (unsigned)(l * 8 + rd * 512);
     unsigned long long v;
     asm volatile("ds_read_b64_tr_b16 %0, %1\n\ts_waitcnt lgkmcnt(0)"
                  : "=v"(v) : "v"(off));
