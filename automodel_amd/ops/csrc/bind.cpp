@@ -61,6 +61,10 @@ TORCH_LIBRARY(amd_ops, m) {
 
   m.def("nf4_dequant(Tensor packed, Tensor absmax, int block_size, int rows, int cols) -> Tensor");
   m.impl("nf4_dequant", &amd_ops::nf4_dequant);
+  m.def("fp8_cast(Tensor x, Tensor scale, Tensor(a!) amax_out, bool e5m2) -> Tensor");
+  m.impl("fp8_cast", &amd_ops::fp8_cast);
+  m.def("fp8_transpose(Tensor x8) -> Tensor");
+  m.impl("fp8_transpose", &amd_ops::fp8_transpose);
 
   m.def("gemv_bf16(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.impl("gemv_bf16", &amd_ops::gemv_bf16);

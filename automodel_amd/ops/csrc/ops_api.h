@@ -58,6 +58,9 @@ at::Tensor soft_ce_bwd(const at::Tensor& s, const at::Tensor& t,
 
 at::Tensor nf4_dequant(const at::Tensor& packed, const at::Tensor& absmax,
                        int64_t block_size, int64_t rows, int64_t cols);
+at::Tensor fp8_cast(const at::Tensor& x, const at::Tensor& scale, at::Tensor amax_out,
+                    bool e5m2);
+at::Tensor fp8_transpose(const at::Tensor& x8);
 
 at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias);
