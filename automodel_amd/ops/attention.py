@@ -121,13 +121,30 @@ class _FlashAttnHip(torch.autograd.Function):
 
 def _flash_hip(q, k, v, causal, scale, q_start, doc_start=None, doc_end=None):
     """Dispatch to the HIP kernel, zero-padding head dims to a supported
-    kernel pair when needed (pad/slice sit outside the Function so autograd
-    routes gradients through them)."""
+    kernel pair and (for plain causal) sequence lengths to the kernel's
+    tile granularity — pad/slice sit outside the Function so autograd
+    routes gradients through them."""
     dqk, dv = q.shape[-1], v.shape[-1]
     tq, tv = _target_dims(dqk, dv)
     if tq != dqk or tv != dv:
         q, k, v = _pad_head(q, tq), _pad_head(k, tq), _pad_head(v, tv)
+    sq, skv = q.shape[1], k.shape[1]
+    pad_q = (-sq) % 128
+    pad_kv = (-skv) % 128 if doc_start is None else (-skv) % 32
+    if pad_q or pad_kv:
+        # safe only for plain causal from position 0: padded KV rows sit at
+        # positions above every real q row, so the causal mask excludes them
+        assert causal and q_start == 0 and doc_start is None, \
+            "flash: Sq%128/Skv%32 required for q_start/varlen/non-causal calls"
+        F = torch.nn.functional.pad
+        if pad_q:
+            q = F(q, (0, 0, 0, 0, 0, pad_q))
+        if pad_kv:
+            k = F(k, (0, 0, 0, 0, 0, pad_kv))
+            v = F(v, (0, 0, 0, 0, 0, pad_kv))
     o = _FlashAttnHip.apply(q, k, v, causal, scale, q_start, doc_start, doc_end)
+    if pad_q:
+        o = o[:, :sq]
     return o[..., :dv] if tv != dv else o
 
 
