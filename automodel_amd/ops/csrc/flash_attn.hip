@@ -158,6 +158,7 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
   char* k_lds = smem;                               // KVBLK*KB
   char* v_img = smem + KVBLK * KB;                  // subtiled [DV/16][32][16]
   float* bcast = reinterpret_cast<float*>(smem + KVBLK * KB + sub_img_bytes<DV>());
+  int* bcast_skip = reinterpret_cast<int*>(bcast + WAVES * 32);
 
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE_SIZE - 1);
@@ -257,8 +258,14 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
         tile_max = fmaxf(tile_max, sv);
       }
       tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32));
-      const float m_new = fmaxf(m_run, tile_max);
-      alpha = __expf(m_run - m_new);
+      // defer-max (guide T13, THR=8): if no row's max grew past m+THR, keep
+      // the old max and skip the O-wide rescale pass — P is then bounded by
+      // e^8 instead of 1, which the f32 accumulator tolerates. Wave-uniform
+      // decision taken BEFORE this tile's P is exponentiated (the textbook-
+      // safe order; guide T13 correctness hazard).
+      const bool need_rescale = !__all(tile_max <= m_run + 8.0f);
+      const float m_new = need_rescale ? fmaxf(m_run, tile_max) : m_run;
+      alpha = need_rescale ? __expf(m_run - m_new) : 1.f;
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
@@ -287,20 +294,24 @@ __global__ __launch_bounds__(WAVES * WAVE_SIZE, 2) void flash_fwd_kernel(
         else pa1 = __builtin_bit_cast(bf16x8_v, *reinterpret_cast<ulonglong2*>(frag));
       }
 
-      // ---- broadcast alpha to O layout via LDS
+      // ---- broadcast alpha to O layout via LDS (sign bit of slot 0 set
+      // when the whole wave deferred, so the rescale pass can be skipped)
       if (lane < 32) bcast[wid * 32 + lane] = alpha;
+      if (lane == 0) bcast_skip[wid] = need_rescale ? 1 : 0;
     }
     // (waves with dead tiles skip compute but still hit the barriers)
     __syncthreads();
 
     if (tile_live) {
-      // ---- rescale O by alpha (per q row: row = (r&3) + 8*(r>>2) + 4*half)
+      // ---- rescale O by alpha (skipped on deferred tiles, guide T13)
+      if (bcast_skip[wid]) {
 #pragma unroll
-      for (int t = 0; t < DV / 32; ++t) {
+        for (int t = 0; t < DV / 32; ++t) {
 #pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          const int row = (r & 3) + 8 * (r >> 2) + 4 * half;
-          o_acc[t][r] *= bcast[wid * 32 + row];
+          for (int r = 0; r < 16; ++r) {
+            const int row = (r & 3) + 8 * (r >> 2) + 4 * half;
+            o_acc[t][r] *= bcast[wid * 32 + row];
+          }
         }
       }
       // software-pipelined tr16 PV: tile t+1's four transpose reads are in
@@ -388,7 +399,7 @@ static void launch_fwd(const at::Tensor& q, const at::Tensor& k, const at::Tenso
                        float scale, bool causal, hipStream_t stream) {
   const dim3 grid(Sq / (WAVES * QBLK), Hq, B);
   const size_t smem = KVBLK * pad256(DQK * 2) + sub_img_bytes<DV>() +
-                      WAVES * 32 * sizeof(float);
+                      WAVES * 33 * sizeof(float);
   hipLaunchKernelGGL((flash_fwd_kernel<DQK, DV, WAVES, VARLEN>), grid,
                      dim3(WAVES * WAVE_SIZE), smem, stream,
                      reinterpret_cast<const bf16*>(q.data_ptr()),
