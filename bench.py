@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--loss-chunk", type=int, default=4096)
     p.add_argument("--fp8", action="store_true", help="swap linears to Float8Linear")
+    p.add_argument("--lora", action="store_true",
+                   help="LoRA r=32 SFT (the BASELINE.md row 11 anchor is LoRA: "
+                        "12,472.87 tok/s on 1xH100)")
     p.add_argument("--no-fused-proj", action="store_true",
                    help="disable fused qkv/gate_up projections")
     p.add_argument("--profile-steps", type=int, default=0,
@@ -121,6 +124,17 @@ def main():
             model = LlamaForCausalLM(cfg, backend=backend)
         model = model.to(dtype=torch.bfloat16)
     model.loss_fn = FusedLinearCrossEntropy(backend=args.loss, chunk_size=args.loss_chunk)
+    if args.lora:
+        # match the reference anchor's recipe shape (LoRA adapters on the
+        # attention+mlp projections, frozen base)
+        from automodel_amd.peft.lora import PeftConfig, apply_lora_to_linear_modules
+        n = apply_lora_to_linear_modules(model, PeftConfig(
+            target_modules=["*q_proj", "*k_proj", "*v_proj", "*o_proj",
+                            "*qkv_proj", "*gate_up_proj",
+                            "*gate_proj", "*up_proj", "*down_proj"],
+            dim=32, alpha=64.0))
+        if rank == 0:
+            print(f"lora: wrapped {n} linears")
     if args.fp8:
         from automodel_amd.quantization.fp8 import apply_fp8_to_model
         n = apply_fp8_to_model(model)
@@ -134,12 +148,18 @@ def main():
     if world > 1:
         apply_fsdp(model, mesh["dp_shard"], reshard_after_forward=False)
     model.init_weights(device=device)
+    if args.lora:
+        from automodel_amd.peft.lora import LinearLoRA
+        for m in model.modules():
+            if isinstance(m, LinearLoRA):
+                m.reset_lora_parameters()      # B=0 after materialization
     model.train()
     # 30B-class single-GPU: bf16 optimizer states (fp32 master+m+v for 30.5B
     # params is 366 GB > 288 GB HBM); n>=2 shards fp32 states via FSDP2
     state_dtype = torch.bfloat16 if (is_moe and args.model == "qwen3_moe_30b"
                                      and world == 1) else torch.float32
-    opt = FusedAdamW(model.parameters(), lr=2e-5, weight_decay=0.0,
+    params = [p for p in model.parameters() if p.requires_grad]
+    opt = FusedAdamW(params, lr=2e-5, weight_decay=0.0,
                      state_dtype=state_dtype)
 
     # proxy configs have short contexts; clamp so --model tiny_proxy works
@@ -245,6 +265,7 @@ def main():
                 "global_batch": args.mbs * world,
                 "seq_len": args.seq_len,
                 "parallelism": f"fsdp{world}" if world > 1 else "single",
+                "peft": "lora_r32" if args.lora else "full",
                 "mfu": round(achieved_mfu, 4) if achieved_mfu is not None else None,
                 "flops_per_token": fpt,
                 "loss_per_token": last_loss,
