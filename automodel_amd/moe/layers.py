@@ -126,6 +126,17 @@ class SharedExpert(nn.Module):
         return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
+_SHARED_STREAMS: dict = {}
+
+
+def _shared_stream(device) -> "torch.cuda.Stream":
+    """Per-device side stream for shared-expert overlap."""
+    key = device.index if hasattr(device, "index") else int(device)
+    if key not in _SHARED_STREAMS:
+        _SHARED_STREAMS[key] = torch.cuda.Stream(device=device)
+    return _SHARED_STREAMS[key]
+
+
 class MoE(nn.Module):
     """Gate -> dispatch -> grouped experts -> combine (+ shared experts)."""
 
@@ -156,6 +167,21 @@ class MoE(nn.Module):
             self.last_expert_load = load
         if self.dispatcher is not None:
             y = self.dispatcher(xf, probs, indices, self.experts)
+        elif self.shared_experts is not None and xf.is_cuda:
+            # shared-expert overlap (reference token_dispatcher.py shared-
+            # expert stream): the dense shared MLP runs on a side stream
+            # concurrently with the routed dispatch+grouped-GEMM path —
+            # stream-aware autograd keeps backward ordering correct
+            ss = _shared_stream(xf.device)
+            ss.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(ss):
+                shared = self.shared_experts(xf)
+                if getattr(self, "shared_expert_gate", None) is not None:
+                    shared = torch.sigmoid(self.shared_expert_gate(xf)) * shared
+            y = self.experts(xf, probs, indices)
+            torch.cuda.current_stream().wait_stream(ss)
+            shared.record_stream(torch.cuda.current_stream())
+            return (y + shared).view(B, S, H)
         else:
             y = self.experts(xf, probs, indices)
         if self.shared_experts is not None:

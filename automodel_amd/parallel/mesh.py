@@ -86,9 +86,15 @@ def build_mesh(
     pp: int = 1,
     cp: int = 1,
     device_type: str | None = None,
+    axis_timeouts: dict | None = None,
 ) -> MeshContext:
     """Build the named root mesh (pp, dp_replicate, dp_shard, cp, tp) and
-    flatten dp / dp_shard_cp / dp_cp, mirroring the reference's axis names."""
+    flatten dp / dp_shard_cp / dp_cp, mirroring the reference's axis names.
+
+    ``axis_timeouts``: per-axis collective timeout overrides in minutes
+    (reference mesh_utils.py:173 _nccl_backend_override — e.g. a long pp
+    timeout for uneven pipeline stages, short dp for fast failure
+    detection).  Applied to the axis process groups after construction."""
     world = dist.get_world_size() if dist.is_initialized() else 1
     if dp_shard == -1:
         denom = dp_replicate * tp * pp * cp
@@ -108,4 +114,13 @@ def build_mesh(
     mesh[("dp_replicate", "dp_shard")]._flatten("dp")
     mesh[("dp_replicate", "dp_shard", "cp")]._flatten("dp_cp")
     mesh[("dp_shard", "cp")]._flatten("dp_shard_cp")
+    if axis_timeouts:
+        import datetime
+
+        from torch.distributed.distributed_c10d import _set_pg_timeout
+
+        for axis, minutes in axis_timeouts.items():
+            if dims.get(axis, 1) > 1:
+                _set_pg_timeout(datetime.timedelta(minutes=float(minutes)),
+                                mesh[axis].get_group())
     return MeshContext(mesh=mesh, world_size=world, dims=dims)

@@ -55,6 +55,8 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             tp=dist_cfg.get("tp", 1),
             pp=dist_cfg.get("pp", 1),
             cp=dist_cfg.get("cp", 1),
+            axis_timeouts=(dist_cfg.get("axis_timeouts")
+                           and dict(dist_cfg.axis_timeouts.items())),
         )
         self.cp_size = self.mesh.dims.get("cp", 1)
         if self.cp_size > 1:

@@ -131,3 +131,37 @@ def _sigterm_worker(rank, world):
 
 def test_signal_handler_propagates():
     run_distributed(_sigterm_worker, world=2)
+
+
+def test_compile_config_regional():
+    """Regional torch.compile wraps decoder layers (ROCm-safe aot_eager)."""
+    import torch
+
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    from automodel_amd.utils.compile import CompileConfig, apply_compile
+
+    m = LlamaForCausalLM(LlamaConfig(
+        vocab_size=128, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=1,
+        max_position_embeddings=64))
+    m.init_weights()
+    n = apply_compile(m, CompileConfig(enabled=True, backend="eager"))
+    assert n == 2
+    ids = torch.randint(0, 128, (1, 8))
+    assert m(ids).shape == (1, 8, 128)
+
+
+def _mesh_timeout_fn(rank, world):
+    from automodel_amd.parallel.mesh import build_mesh
+
+    ctx = build_mesh(dp_shard=2, device_type="cpu",
+                     axis_timeouts={"dp_shard": 7, "tp": 1})
+    return ctx.dims["dp_shard"]
+
+
+def test_mesh_axis_timeouts_cpu():
+    """axis_timeouts applies per-axis pg timeout overrides (world 2 gloo)."""
+    from tests.dist_utils import run_distributed
+
+    out = run_distributed(_mesh_timeout_fn, world=2)
+    assert out[0] == 2
