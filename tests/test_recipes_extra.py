@@ -462,3 +462,30 @@ def test_kd_recipe_vocab_mismatch_rejected(tmp_path):
     r = KDRecipeForNextTokenPrediction(cfg)
     with pytest.raises(ValueError, match="vocab"):
         r.setup()
+
+
+def test_diffusion_flow_matching_recipe(tmp_path):
+    """Rectified-flow DiT recipe: loss finite and decreasing over steps
+    (reference recipes/diffusion/train.py domain)."""
+    import torch
+
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.recipes.diffusion.train import TrainDiffusionRecipe
+
+    cfg = ConfigNode({
+        "seed": 0,
+        "model": {"config": dict(image_size=16, patch_size=4, in_channels=3,
+                                 hidden_size=64, num_hidden_layers=2,
+                                 num_attention_heads=2)},
+        "optimizer": {"lr": 3e-3},
+        "step_scheduler": {"max_steps": 12},
+        "dataloader": {"dataset": {"num_samples": 8}, "batch_size": 4},
+        "output_dir": str(tmp_path),
+    })
+    r = TrainDiffusionRecipe(cfg)
+    r.setup()
+    r.run_train_validation_loop()
+    assert len(r.losses) >= 10
+    assert all(torch.isfinite(torch.tensor(r.losses)).tolist())
+    # rectified flow on a tiny memorizable set: loss should drop
+    assert sum(r.losses[-3:]) < sum(r.losses[:3]), r.losses
