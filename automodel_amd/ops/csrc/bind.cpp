@@ -65,6 +65,8 @@ TORCH_LIBRARY(amd_ops, m) {
   m.impl("fp8_cast", &amd_ops::fp8_cast);
   m.def("fp8_transpose(Tensor x8) -> Tensor");
   m.impl("fp8_transpose", &amd_ops::fp8_transpose);
+  m.def("lora_fused_fwd(Tensor x, Tensor A, Tensor B, float scale) -> Tensor");
+  m.impl("lora_fused_fwd", &amd_ops::lora_fused_fwd);
 
   m.def("gemv_bf16(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.impl("gemv_bf16", &amd_ops::gemv_bf16);

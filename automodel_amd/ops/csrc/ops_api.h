@@ -61,6 +61,8 @@ at::Tensor nf4_dequant(const at::Tensor& packed, const at::Tensor& absmax,
 at::Tensor fp8_cast(const at::Tensor& x, const at::Tensor& scale, at::Tensor amax_out,
                     bool e5m2);
 at::Tensor fp8_transpose(const at::Tensor& x8);
+at::Tensor lora_fused_fwd(const at::Tensor& x, const at::Tensor& A,
+                          const at::Tensor& B, double scale);
 
 at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias);

@@ -38,6 +38,33 @@ class PeftConfig:
         return cls(**{k: (list(v) if k == "target_modules" else v) for k, v in d.items()})
 
 
+class _FusedLoRAFn(torch.autograd.Function):
+    """Fused delta = (x @ A^T) @ B^T * scale (csrc/lora.hip — one kernel,
+    the [M, r] intermediate never hits HBM; reference _peft/lora_kernel.py
+    lora_forward_kernel). Backward recomputes t with two skinny GEMMs."""
+
+    @staticmethod
+    def forward(ctx, x2, wa, wb, scale):
+        from automodel_amd.ops._backend import hip_ops
+
+        delta = hip_ops().lora_fused_fwd(x2, wa, wb, scale)
+        ctx.save_for_backward(x2, wa, wb)
+        ctx.scale = scale
+        return delta
+
+    @staticmethod
+    def backward(ctx, g):
+        x2, wa, wb = ctx.saved_tensors
+        s = ctx.scale
+        g = g.contiguous()
+        t = x2 @ wa.t()                  # [M, r] recompute
+        dt = (g @ wb) * s
+        dx = dt.to(x2.dtype) @ wa
+        da = dt.t().to(x2.dtype) @ x2
+        db = (g.t() @ t.to(g.dtype)) * s
+        return dx, da.to(wa.dtype), db.to(wb.dtype), None
+
+
 class LinearLoRA(nn.Module):
     """y = base(x) + (dropout(x) @ A^T) @ B^T * (alpha / r); base frozen.
 
@@ -90,7 +117,18 @@ class LinearLoRA(nn.Module):
             w = w * (self.lora_magnitude / col_norm).unsqueeze(1)
             y = torch.nn.functional.linear(x, w, self.base.bias)
             return y
+        wa, wb = self.lora_A.weight, self.lora_B.weight
+        if (x.is_cuda and x.dtype == torch.bfloat16 and not self.training_dropout()
+                and wa.shape[0] in (32, 64) and wa.shape[1] % 64 == 0
+                and wb.shape[0] % 64 == 0
+                and (x.numel() // x.shape[-1]) % 8 == 0):
+            x2 = x.reshape(-1, x.shape[-1]).contiguous()
+            delta = _FusedLoRAFn.apply(x2, wa, wb, self.scale)
+            return self.base(x) + delta.view(*x.shape[:-1], wb.shape[0])
         return self.base(x) + self.lora_B(self.lora_A(self.dropout(x))) * self.scale
+
+    def training_dropout(self) -> bool:
+        return self.training and not isinstance(self.dropout, nn.Identity)
 
     @torch.no_grad()
     def merge(self) -> nn.Linear:
