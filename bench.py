@@ -33,7 +33,8 @@ def parse_args():
     p.add_argument("--seq-len", type=int, default=4096)
     p.add_argument("--model", type=str, default="llama3_8b",
                    choices=["llama3_8b", "llama3_1b_proxy", "tiny_proxy",
-                            "qwen3_moe_30b", "moe_tiny_proxy"])
+                            "qwen3_moe_30b", "moe_tiny_proxy",
+                            "deepseek_v3_16b"])
     p.add_argument("--attn", type=str, default="hip")
     p.add_argument("--loss", type=str, default="hybrid")
     p.add_argument("--loss-chunk", type=int, default=4096)
@@ -76,6 +77,21 @@ MODEL_CONFIGS = {
         moe=dict(n_routed_experts=128, n_activated_experts=8,
                  moe_intermediate_size=768),
     ),
+    # DeepSeek-V3-STYLE MLA+MoE pretrain config (BASELINE.json config #5:
+    # "MLA + fp8 MFMA + EP, 288 GB shard sizing") at a 16B-class size that
+    # one GPU holds with full optimizer state; MLA dims are the real V3's
+    # (qk nope 128 + rope 64, v 128, kv_lora 512)
+    "deepseek_v3_16b": dict(
+        vocab_size=129280, hidden_size=2048, intermediate_size=10944,
+        num_hidden_layers=27, num_attention_heads=16,
+        q_lora_rank=0, kv_lora_rank=512,
+        qk_nope_head_dim=128, qk_rope_head_dim=64, v_head_dim=128,
+        max_position_embeddings=8192, first_k_dense_replace=1,
+        moe=dict(n_routed_experts=64, n_activated_experts=6,
+                 moe_intermediate_size=1408, n_shared_experts=2,
+                 shared_expert_intermediate_size=2816,
+                 score_func="sigmoid", expert_bias=True, route_scale=2.5),
+    ),
     "moe_tiny_proxy": dict(
         vocab_size=512, hidden_size=256, intermediate_size=512,
         num_hidden_layers=2, num_attention_heads=2, num_key_value_heads=2,
@@ -84,7 +100,10 @@ MODEL_CONFIGS = {
                  moe_intermediate_size=128),
     ),
 }
-MOE_MODELS = {"qwen3_moe_30b", "moe_tiny_proxy"}
+MOE_MODELS = {"qwen3_moe_30b", "moe_tiny_proxy", "deepseek_v3_16b"}
+ARCH_BY_MODEL = {"qwen3_moe_30b": "Qwen3MoeForCausalLM",
+                 "moe_tiny_proxy": "Qwen3MoeForCausalLM",
+                 "deepseek_v3_16b": "DeepseekV3ForCausalLM"}
 
 
 def main():
@@ -112,7 +131,7 @@ def main():
         from automodel_amd.models.registry import build_model
 
         model = build_model(config=MODEL_CONFIGS[args.model],
-                            architecture="Qwen3MoeForCausalLM",
+                            architecture=ARCH_BY_MODEL[args.model],
                             backend={"attn": args.attn, "loss": args.loss},
                             dtype="bfloat16", meta_init=True)
         cfg = model.config
@@ -140,7 +159,7 @@ def main():
         n = apply_fp8_to_model(model)
         if rank == 0:
             print(f"fp8: swapped {n} linears")
-    if is_moe and args.model == "qwen3_moe_30b":
+    if is_moe and args.model in ("qwen3_moe_30b", "deepseek_v3_16b"):
         # 30B at one GPU needs activation checkpointing; 8-GPU runs keep it
         # for parity across N (weak scaling holds per-GPU work fixed)
         from automodel_amd.parallel.activation_checkpointing import apply_ac
@@ -221,7 +240,12 @@ def main():
 
     tokens_per_step = args.mbs * args.seq_len * world
     tps = tokens_per_step * n_steps / elapsed
-    if is_moe:
+    if is_moe and args.model == "deepseek_v3_16b":
+        from automodel_amd.utils.flops import deepseek_v3_flops_per_token
+
+        fpt = deepseek_v3_flops_per_token(cfg, args.seq_len)
+        baseline_tok_per_gpu = 1002.0   # BASELINE.md row 5 (256xH100, PP4 EP64)
+    elif is_moe:
         from automodel_amd.utils.flops import moe_flops_per_token
 
         fpt = moe_flops_per_token(
@@ -242,7 +266,9 @@ def main():
 
     if rank == 0 and args.profile_steps == 0:
         print(json.dumps({
-            "metric": ("tokens/sec (whole node) + MFU, Qwen3-MoE-30B-A3B SFT "
+            "metric": ("tokens/sec (whole node) + MFU, DeepSeek-V3-style MLA+MoE "
+                       "pretrain at MI355X" if args.model == "deepseek_v3_16b" else
+                       "tokens/sec (whole node) + MFU, Qwen3-MoE-30B-A3B SFT "
                        "FSDP2+grouped-GEMM at MI355X" if is_moe else
                        "tokens/sec (whole node) + MFU, Llama-3-8B SFT FSDP2 "
                        "at 1/2/4/8 MI355X"),
@@ -260,6 +286,7 @@ def main():
             "config": {
                 "model": ("llama3-8b" if args.model == "llama3_8b"
                           else "qwen3-moe-30b-a3b" if args.model == "qwen3_moe_30b"
+                          else "deepseek-v3-style-16b" if args.model == "deepseek_v3_16b"
                           else args.model),
                 "optimizer_state": str(state_dtype).replace("torch.", ""),
                 "global_batch": args.mbs * world,
