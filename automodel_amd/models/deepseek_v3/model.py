@@ -5,10 +5,11 @@ Reference behavior: nemo_automodel/components/models/deepseek_v3/
 Attention: queries and KV are projected through low-rank bottlenecks; RoPE is
 applied to a decoupled rope sub-dimension (shared single k_rope head).
 
-Attention head dims differ from 128 (qk = nope+rope, v = v_head_dim), so the
-attention core runs torch-SDPA for now; the D-generalized HIP flash kernel is
-a planned follow-up. All other hot ops (RMSNorm, SwiGLU, grouped experts,
-fused CE) run the in-tree HIP kernels.
+Attention head dims differ from 128 (qk = nope+rope, v = v_head_dim): the
+attention core runs the split-dim (Dqk=192, Dv=128) instantiation of the
+in-tree HIP flash kernel (round 2), with torch-SDPA as the CPU/odd-shape
+fallback. All other hot ops (RMSNorm, SwiGLU, grouped experts, fused CE)
+also run the in-tree HIP kernels.
 """
 
 from __future__ import annotations
@@ -157,7 +158,7 @@ class MLAAttention(nn.Module):
 
         qf = torch.cat([q_nope, q_rope], dim=-1)   # B,S,H,qk (192 = nope+rope)
         kf = torch.cat([k_nope, k_rope], dim=-1)
-        if self.backend.attn == "hip" and qf.is_cuda and S % 128 == 0:
+        if self.backend.attn == "hip" and qf.is_cuda:
             # split-dim flash kernel: (Dqk=192, Dv=128) instantiation
             from automodel_amd.ops.attention import flash_attention
 
@@ -200,7 +201,11 @@ class DeepseekV3DecoderLayer(nn.Module):
 
 
 class DeepseekV3ForCausalLM(nn.Module):
-    hf_architectures = ("DeepseekV3ForCausalLM", "DeepseekV2ForCausalLM")
+    # Kimi-K2 is DeepSeek-V3-architecture-compatible (reference
+    # models/kimi_k2/config.py: KimiK2Config(DeepseekV3Config)); Moonlight
+    # ships the same architecture string
+    hf_architectures = ("DeepseekV3ForCausalLM", "DeepseekV2ForCausalLM",
+                        "KimiK2ForCausalLM")
     config_class = DeepseekV3Config
 
     @staticmethod
