@@ -43,7 +43,9 @@ class LlamaConfig:
     attention_bias: bool = False
     mlp_bias: bool = False
     qk_norm: bool = False          # Qwen3-style per-head q/k RMSNorm
+    qk_norm_after_rope: bool = False   # HunYuan: per-head norm AFTER rope
     qk_norm_full: bool = False     # OLMoE-style full-projection q/k RMSNorm
+    mlp_type: str = "swiglu"       # "swiglu" | "xielu" (Apertus: up->xIELU->down)
     olmo2_layout: bool = False     # OLMo-2: norms on sublayer OUTPUTS only
     # granite-style scalar multipliers (neutral defaults)
     attention_multiplier: float | None = None   # attention scale override
@@ -57,6 +59,8 @@ class LlamaConfig:
     rope_interleaved: bool = False      # GPT-NeoX/Ernie pair-interleaved rope
     fused_qkv: bool = False        # one qkv GEMM (state_dict_adapter keeps HF keys)
     fused_gate_up: bool = False    # one gate|up GEMM + concatenated swiglu
+    # HF-key substring renames (hunyuan/apertus name their norms differently)
+    hf_key_renames: dict | None = None
     initializer_range: float = 0.02
 
     def __post_init__(self):
@@ -77,7 +81,16 @@ class LlamaConfig:
         if scaling is None and rp.get("rope_type", "default") != "default":
             scaling = rp
         return cls(
-            qk_norm="Qwen3" in archs,
+            qk_norm=("Qwen3" in archs) or ("Apertus" in archs)
+                or ("HunYuanDense" in archs),
+            qk_norm_after_rope="HunYuanDense" in archs,
+            mlp_type="xielu" if "Apertus" in archs else "swiglu",
+            hf_key_renames=(
+                {"query_layernorm": "q_norm", "key_layernorm": "k_norm"}
+                if "HunYuanDense" in archs else
+                {"attention_layernorm": "input_layernorm",
+                 "feedforward_layernorm": "post_attention_layernorm"}
+                if "Apertus" in archs else None),
             qk_norm_full="Olmo" in archs,
             olmo2_layout="Olmo2" in archs,
             # Phi-3 ships fused qkv_proj / gate_up_proj weights — exactly this
@@ -163,7 +176,7 @@ class LlamaAttention(nn.Module):
             q = q.view(B, S, -1, self.head_dim)
             k = k.view(B, S, -1, self.head_dim)
             v = self.v_proj(x).view(B, S, -1, self.head_dim)
-        if self.cfg.qk_norm:
+        if self.cfg.qk_norm and not self.cfg.qk_norm_after_rope:
             q = self.q_norm(q)
             k = self.k_norm(k)
         if self.use_rope:
@@ -176,6 +189,10 @@ class LlamaAttention(nn.Module):
                 q = q.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
                 k = k.reshape(B, S, -1, d2, 2).transpose(-1, -2).reshape(B, S, -1, self.head_dim)
             q, k = apply_rope(q, k, cos, sin, backend=self.backend.rope)
+        if self.cfg.qk_norm and self.cfg.qk_norm_after_rope:
+            # HunYuan-Dense: per-head q/k RMSNorm applied AFTER rope
+            q = self.q_norm(q)
+            k = self.k_norm(k)
         attn_scale = self.cfg.attention_multiplier  # None -> 1/sqrt(D)
         if self.cfg.sliding_window is not None:
             # windowed causal (mistral): explicit mask via sdpa — the flash
@@ -246,6 +263,43 @@ class LlamaMLP(nn.Module):
         return self.down_proj(swiglu(self.gate_proj(x), self.up_proj(x)))
 
 
+class XIELUActivation(nn.Module):
+    """xIELU (arXiv:2411.13010, Apertus): learnable alpha_p/alpha_n stored
+    in softplus-inverse form; matches HF transformers.activations math."""
+
+    def __init__(self):
+        super().__init__()
+        import math as _m
+
+        inv_sp = lambda v: _m.log(_m.expm1(v))
+        self.alpha_p = nn.Parameter(torch.tensor([inv_sp(0.8)]))
+        self.alpha_n = nn.Parameter(torch.tensor([inv_sp(0.8 - 0.5)]))
+        self.register_buffer("beta", torch.tensor(0.5))
+        self.register_buffer("eps", torch.tensor(-1e-6))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        ap = torch.nn.functional.softplus(self.alpha_p)
+        an = self.beta + torch.nn.functional.softplus(self.alpha_n)
+        return torch.where(
+            x > 0,
+            ap * x * x + self.beta * x,
+            (torch.expm1(torch.min(x, self.eps.to(x.dtype))) - x) * an + self.beta * x,
+        )
+
+
+class XIELUMLP(nn.Module):
+    """Apertus MLP: up_proj -> xIELU -> down_proj (no gate)."""
+
+    def __init__(self, cfg: LlamaConfig, backend: BackendConfig):
+        super().__init__()
+        self.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=cfg.mlp_bias)
+        self.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=cfg.mlp_bias)
+        self.act_fn = XIELUActivation()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(self.act_fn(self.up_proj(x)))
+
+
 class LlamaDecoderLayer(nn.Module):
     def __init__(self, cfg: LlamaConfig, backend: BackendConfig,
                  use_rope: bool = True):
@@ -253,7 +307,7 @@ class LlamaDecoderLayer(nn.Module):
         self.olmo2 = cfg.olmo2_layout
         self.residual_multiplier = cfg.residual_multiplier
         self.self_attn = LlamaAttention(cfg, backend, use_rope=use_rope)
-        self.mlp = LlamaMLP(cfg, backend)
+        self.mlp = (XIELUMLP if cfg.mlp_type == "xielu" else LlamaMLP)(cfg, backend)
         if cfg.olmo2_layout:   # norms on outputs (OLMo-2)
             self.post_attention_layernorm = RMSNorm(cfg.hidden_size,
                                                     cfg.rms_norm_eps, backend.rms_norm)
@@ -312,7 +366,8 @@ class LlamaForCausalLM(nn.Module):
     hf_architectures = ("LlamaForCausalLM", "Qwen2ForCausalLM", "MistralForCausalLM",
                         "Qwen3ForCausalLM", "Phi3ForCausalLM", "SmolLM3ForCausalLM",
                         "Ernie4_5ForCausalLM", "Olmo2ForCausalLM", "GraniteForCausalLM",
-                        "HeliumForCausalLM", "SeedOssForCausalLM")
+                        "HeliumForCausalLM", "SeedOssForCausalLM",
+                        "HunYuanDenseV1ForCausalLM", "ApertusForCausalLM")
     config_class = LlamaConfig
 
     @staticmethod
@@ -336,6 +391,8 @@ class LlamaForCausalLM(nn.Module):
             self.lm_head.weight = self.model.embed_tokens.weight
         if config.fused_qkv or config.fused_gate_up:
             self.state_dict_adapter = LlamaFusedStateDictAdapter(config)
+        elif config.hf_key_renames:
+            self.state_dict_adapter = LlamaRenameStateDictAdapter(config.hf_key_renames)
         # set by the recipe; called as loss_fn(hidden, lm_head_weight, labels).
         # Computing the loss INSIDE forward keeps lm_head.weight unsharded
         # under FSDP2 (reference computes it via lm-weight gather instead,
@@ -435,6 +492,36 @@ class LlamaForCausalLM(nn.Module):
                 seen.add(id(p))
                 total += p.numel()
         return total
+
+
+class LlamaRenameStateDictAdapter:
+    """Substring-rename adapter for llama-family archs whose HF key names
+    differ (HunYuan-Dense norm names, Apertus pre-norm names)."""
+
+    def __init__(self, renames: dict):
+        self.renames = dict(renames)
+
+    def _apply(self, sd: dict, mapping: dict) -> dict:
+        out = {}
+        for k, v in sd.items():
+            for a, b in mapping.items():
+                if a in k:
+                    k = k.replace(a, b)
+            out[k] = v
+        return out
+
+    def from_hf(self, sd: dict) -> dict:
+        return self._apply(sd, self.renames)
+
+    def to_hf(self, sd: dict) -> dict:
+        return self._apply(sd, {b: a for a, b in self.renames.items()})
+
+    def hf_key_targets(self, key: str) -> list:
+        k = key
+        for a, b in self.renames.items():
+            if a in k:
+                k = k.replace(a, b)
+        return [k]
 
 
 class LlamaFusedStateDictAdapter:

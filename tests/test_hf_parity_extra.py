@@ -1315,3 +1315,96 @@ def test_hf_logits_parity_codegen():
     ids = torch.randint(0, 300, (2, 16))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=2e-4, rtol=2e-4)
+
+
+def test_hf_logits_parity_hunyuan_dense():
+    """HunYuan-Dense-V1: llama-shaped + per-head qk RMSNorm AFTER rope."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.registry import build_model
+
+    hf_cfg = transformers.HunYuanDenseV1Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+        attention_bias=False, tie_word_embeddings=False,
+        attn_implementation="eager")
+    torch.manual_seed(0)
+    hf = transformers.HunYuanDenseV1ForCausalLM(hf_cfg).eval()
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    cfg = LlamaConfig.from_hf_config(
+        {**hf_cfg.to_dict(), "architectures": ["HunYuanDenseV1ForCausalLM"]})
+    mine = LlamaForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    # HF names the qk norms query_layernorm/key_layernorm
+    remap = {k.replace("query_layernorm", "q_norm").replace("key_layernorm", "k_norm"): v
+             for k, v in hf.state_dict().items()}
+    missing, unexpected = mine.load_state_dict(remap, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_apertus():
+    """Apertus: qk-norm before rope + gate-free xIELU MLP."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+
+    hf_cfg = transformers.ApertusConfig(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=128, rope_theta=10000.0,
+        tie_word_embeddings=False, attn_implementation="eager",
+        hidden_act="xielu")
+    torch.manual_seed(1)
+    hf = transformers.ApertusForCausalLM(hf_cfg).eval().float()
+    cfg = LlamaConfig.from_hf_config(
+        {**hf_cfg.to_dict(), "architectures": ["ApertusForCausalLM"]})
+    mine = LlamaForCausalLM(cfg).eval()
+    # Apertus names its pre-norms attention_layernorm / feedforward_layernorm
+    # (same pre-norm positions as llama's input/post_attention norms)
+    remap = {k.replace("attention_layernorm", "input_layernorm")
+              .replace("feedforward_layernorm", "post_attention_layernorm"): v
+             for k, v in hf.state_dict().items()}
+    missing, unexpected = mine.load_state_dict(remap, strict=False)
+    assert not unexpected, unexpected
+    assert all(("rope" in k) or ("beta" in k) or ("eps" in k) for k in missing), missing
+    ids = torch.randint(0, 300, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=3e-4, rtol=3e-4)
+
+
+def test_hunyuan_apertus_checkpoint_roundtrip(tmp_path):
+    """The rename adapters make real-checkpoint load paths work: save an
+    HF-keyed checkpoint, load via build_model(pretrained_path)."""
+    import json
+    import os
+
+    transformers = pytest.importorskip("transformers")
+    from safetensors.torch import save_file
+
+    from automodel_amd.checkpoint.hf_loader import load_hf_weights
+    from automodel_amd.models.registry import build_model
+
+    torch.manual_seed(2)
+    hf_cfg = transformers.HunYuanDenseV1Config(
+        vocab_size=300, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, max_position_embeddings=128, rope_theta=10000.0,
+        attention_bias=False, tie_word_embeddings=False,
+        attn_implementation="eager")
+    hf = transformers.HunYuanDenseV1ForCausalLM(hf_cfg).eval()
+    ckpt = tmp_path / "hy"
+    os.makedirs(ckpt)
+    save_file({k: v.contiguous() for k, v in hf.state_dict().items()},
+              str(ckpt / "model.safetensors"))
+    with open(ckpt / "config.json", "w") as f:
+        json.dump({**hf_cfg.to_dict(),
+                   "architectures": ["HunYuanDenseV1ForCausalLM"]}, f)
+    m = build_model(pretrained_path=str(ckpt), dtype="float32")
+    m.init_weights(device="cpu")
+    load_hf_weights(m, str(ckpt))
+    ids = torch.randint(0, 300, (2, 16))
+    with torch.no_grad():
+        torch.testing.assert_close(m.eval()(ids), hf(ids).logits,
+                                   atol=3e-4, rtol=3e-4)
