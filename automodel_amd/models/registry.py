@@ -50,6 +50,10 @@ def _ensure_builtin() -> None:
         pass
     try:
         from automodel_amd.models.deepseek_v3.model import DeepseekV3ForCausalLM
+        from automodel_amd.models.llama4.model import Llama4ForCausalLM
+
+        for name in Llama4ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Llama4ForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1408,3 +1408,35 @@ def test_hunyuan_apertus_checkpoint_roundtrip(tmp_path):
     with torch.no_grad():
         torch.testing.assert_close(m.eval()(ids), hf(ids).logits,
                                    atol=3e-4, rtol=3e-4)
+
+
+def test_hf_logits_parity_llama4():
+    """Llama4 text: interleaved MoE/dense layers, input-scaled sigmoid
+    routing + shared expert, NoPE layers with temperature tuning, L2
+    qk-norm after (interleaved) rope."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.llama4.model import Llama4Config, Llama4ForCausalLM
+
+    hf_cfg = transformers.Llama4TextConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        intermediate_size_mlp=160, num_hidden_layers=4, num_attention_heads=4,
+        num_key_value_heads=2, head_dim=16, num_local_experts=4,
+        num_experts_per_tok=2, interleave_moe_layer_step=2,
+        max_position_embeddings=128, rope_theta=10000.0,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(5)
+    hf = transformers.Llama4ForCausalLM(hf_cfg).eval()
+    cfg = Llama4Config.from_hf_config(hf_cfg.to_dict())
+    assert cfg.no_rope_layers == hf_cfg.no_rope_layers
+    mine = Llama4ForCausalLM(cfg).eval()
+    sd = mine.state_dict_adapter.from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 24))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=4e-4, rtol=4e-4)
+    # export roundtrip through the adapter
+    back = mine.state_dict_adapter.to_hf(mine.state_dict())
+    for k, v in hf.state_dict().items():
+        torch.testing.assert_close(back[k], v, atol=0, rtol=0)
