@@ -55,6 +55,11 @@ _CAPS = {
         supports_tp=False, supports_cp=False, supports_pp=False),
     "Mistral3ForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_cp=False, supports_pp=False),
+    "NemotronHForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
+    "Llama4ForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)
     "HFFallbackForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,

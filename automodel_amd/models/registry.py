@@ -54,6 +54,10 @@ def _ensure_builtin() -> None:
 
         for name in Llama4ForCausalLM.hf_architectures:
             _REGISTRY[name] = Llama4ForCausalLM
+        from automodel_amd.models.nemotron_h.model import NemotronHForCausalLM
+
+        for name in NemotronHForCausalLM.hf_architectures:
+            _REGISTRY[name] = NemotronHForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

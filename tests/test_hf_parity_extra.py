@@ -1440,3 +1440,32 @@ def test_hf_logits_parity_llama4():
     back = mine.state_dict_adapter.to_hf(mine.state_dict())
     for k, v in hf.state_dict().items():
         torch.testing.assert_close(back[k], v, atol=0, rtol=0)
+
+
+def test_hf_logits_parity_nemotron_h():
+    """Nemotron-H hybrid: Mamba2 chunked-SSD mixer + NoPE GQA attention +
+    relu^2 MLP layers, pattern-driven; parity vs the HF torch fallback."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.nemotron_h.model import (
+        NemotronHConfig,
+        NemotronHForCausalLM,
+    )
+
+    hf_cfg = transformers.NemotronHConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, mamba_num_heads=4, mamba_head_dim=16, ssm_state_size=8,
+        conv_kernel=4, n_groups=2, chunk_size=8,
+        hybrid_override_pattern="M*M-", max_position_embeddings=128,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(7)
+    hf = transformers.NemotronHForCausalLM(hf_cfg).eval()
+    cfg = NemotronHConfig.from_hf_config(hf_cfg.to_dict())
+    mine = NemotronHForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    # seq length deliberately NOT a multiple of chunk_size (pad path)
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
