@@ -32,7 +32,7 @@ class NemotronHConfig:
     hidden_size: int = 4096
     intermediate_size: int = 21504
     num_hidden_layers: int = 52
-    hybrid_override_pattern: str = "M" * 52
+    hybrid_override_pattern: str | None = None
     num_attention_heads: int = 32
     num_key_value_heads: int = 8
     head_dim: int = 128
@@ -51,6 +51,10 @@ class NemotronHConfig:
     max_position_embeddings: int = 8192
     tie_word_embeddings: bool = False
     initializer_range: float = 0.02
+
+    def __post_init__(self):
+        if self.hybrid_override_pattern is None:
+            self.hybrid_override_pattern = "M" * self.num_hidden_layers
 
     @property
     def mamba_intermediate(self) -> int:
