@@ -183,46 +183,55 @@ class GatedRMSNorm(nn.Module):
 
 
 class Mamba2Mixer(nn.Module):
-    def __init__(self, cfg: NemotronHConfig):
+    """Standalone Mamba2 mixer — shared by every hybrid family (Nemotron-H,
+    Bamba, and future Zamba2/FalconH1-style stacks)."""
+
+    def __init__(self, hidden_size: int, n_heads: int, head_dim: int,
+                 state_size: int, n_groups: int, conv_kernel: int = 4,
+                 chunk_size: int = 128, eps: float = 1e-5,
+                 use_bias: bool = False, use_conv_bias: bool = True,
+                 norm_group_size: int | None = None,
+                 time_step_limit: tuple = (0.0, float("inf"))):
         super().__init__()
-        inter = cfg.mamba_intermediate
-        self.cfg = cfg
-        self.conv_dim = inter + 2 * cfg.n_groups * cfg.ssm_state_size
-        self.in_proj = nn.Linear(
-            cfg.hidden_size, inter + self.conv_dim + cfg.mamba_num_heads,
-            bias=cfg.use_bias)
-        self.conv1d = nn.Conv1d(self.conv_dim, self.conv_dim, cfg.conv_kernel,
-                                groups=self.conv_dim, padding=cfg.conv_kernel - 1,
-                                bias=cfg.use_conv_bias)
-        self.dt_bias = nn.Parameter(torch.ones(cfg.mamba_num_heads))
-        self.A_log = nn.Parameter(torch.zeros(cfg.mamba_num_heads))
-        self.D = nn.Parameter(torch.ones(cfg.mamba_num_heads))
-        self.norm = GatedRMSNorm(inter, cfg.layer_norm_epsilon,
-                                 group_size=inter // cfg.n_groups)
-        self.out_proj = nn.Linear(inter, cfg.hidden_size, bias=cfg.use_bias)
+        inter = n_heads * head_dim
+        self.n_heads, self.head_dim = n_heads, head_dim
+        self.state_size, self.n_groups = state_size, n_groups
+        self.chunk_size = chunk_size
+        self.time_step_limit = time_step_limit
+        self.inter = inter
+        self.conv_dim = inter + 2 * n_groups * state_size
+        self.in_proj = nn.Linear(hidden_size, inter + self.conv_dim + n_heads,
+                                 bias=use_bias)
+        self.conv1d = nn.Conv1d(self.conv_dim, self.conv_dim, conv_kernel,
+                                groups=self.conv_dim, padding=conv_kernel - 1,
+                                bias=use_conv_bias)
+        self.dt_bias = nn.Parameter(torch.ones(n_heads))
+        self.A_log = nn.Parameter(torch.zeros(n_heads))
+        self.D = nn.Parameter(torch.ones(n_heads))
+        self.norm = GatedRMSNorm(inter, eps, group_size=norm_group_size)
+        self.out_proj = nn.Linear(inter, hidden_size, bias=use_bias)
 
     def forward(self, h: torch.Tensor) -> torch.Tensor:
-        cfg = self.cfg
         b, s, _ = h.shape
-        inter = cfg.mamba_intermediate
+        inter = self.inter
         z, xBC, dt = self.in_proj(h).split(
-            [inter, self.conv_dim, cfg.mamba_num_heads], dim=-1)
+            [inter, self.conv_dim, self.n_heads], dim=-1)
         xBC = torch.nn.functional.silu(
             self.conv1d(xBC.transpose(1, 2))[..., :s].transpose(1, 2))
         x, B, C = xBC.split(
-            [inter, cfg.n_groups * cfg.ssm_state_size,
-             cfg.n_groups * cfg.ssm_state_size], dim=-1)
+            [inter, self.n_groups * self.state_size,
+             self.n_groups * self.state_size], dim=-1)
         dt = torch.nn.functional.softplus(dt.float() + self.dt_bias.float())
-        lo, hi = cfg.time_step_limit
+        lo, hi = self.time_step_limit
         if lo > 0 or hi != float("inf"):
             dt = dt.clamp(min=lo, max=hi)
         A = -torch.exp(self.A_log.float())
         y = mamba2_chunked_scan(
-            x.view(b, s, cfg.mamba_num_heads, cfg.mamba_head_dim), dt, A,
-            B.view(b, s, cfg.n_groups, cfg.ssm_state_size),
-            C.view(b, s, cfg.n_groups, cfg.ssm_state_size), cfg.chunk_size)
+            x.view(b, s, self.n_heads, self.head_dim), dt, A,
+            B.view(b, s, self.n_groups, self.state_size),
+            C.view(b, s, self.n_groups, self.state_size), self.chunk_size)
         y = y + self.D.float().view(1, 1, -1, 1) * \
-            x.view(b, s, cfg.mamba_num_heads, cfg.mamba_head_dim).float()
+            x.view(b, s, self.n_heads, self.head_dim).float()
         y = y.to(h.dtype).reshape(b, s, inter)
         return self.out_proj(self.norm(y, z))
 
@@ -265,7 +274,13 @@ class NemotronHBlock(nn.Module):
         super().__init__()
         self.norm = RMSNorm(cfg.hidden_size, cfg.layer_norm_epsilon, backend.rms_norm)
         if kind == "M":
-            self.mixer = Mamba2Mixer(cfg)
+            self.mixer = Mamba2Mixer(
+                cfg.hidden_size, cfg.mamba_num_heads, cfg.mamba_head_dim,
+                cfg.ssm_state_size, cfg.n_groups, cfg.conv_kernel,
+                cfg.chunk_size, cfg.layer_norm_epsilon, cfg.use_bias,
+                cfg.use_conv_bias,
+                norm_group_size=cfg.mamba_intermediate // cfg.n_groups,
+                time_step_limit=cfg.time_step_limit)
         elif kind == "*":
             self.mixer = NemotronHAttentionMixer(cfg, backend)
         elif kind == "-":

@@ -58,6 +58,10 @@ def _ensure_builtin() -> None:
 
         for name in NemotronHForCausalLM.hf_architectures:
             _REGISTRY[name] = NemotronHForCausalLM
+        from automodel_amd.models.bamba.model import BambaForCausalLM
+
+        for name in BambaForCausalLM.hf_architectures:
+            _REGISTRY[name] = BambaForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -893,3 +893,37 @@ def test_lora_module_fused_path_grads():
     assert torch.allclose(gA.float(), lora.lora_A.weight.grad.float(), atol=5e-2, rtol=5e-2)
     assert torch.allclose(gB.float(), lora.lora_B.weight.grad.float(), atol=5e-2, rtol=5e-2)
     assert torch.allclose(gx.float(), x2.grad.float(), atol=5e-2, rtol=5e-2)
+
+
+def test_hybrid_and_llama4_gpu_smoke():
+    """Round-2 families forward+backward on MI355X bf16: Nemotron-H hybrid
+    (Mamba2 chunked scan) and Llama-4 (MoE + NoPE temperature tuning)."""
+    from automodel_amd.models.llama4.model import Llama4Config, Llama4ForCausalLM
+    from automodel_amd.models.nemotron_h.model import (
+        NemotronHConfig,
+        NemotronHForCausalLM,
+    )
+
+    torch.manual_seed(0)
+    nh = NemotronHForCausalLM(NemotronHConfig(
+        vocab_size=256, hidden_size=128, intermediate_size=256,
+        num_hidden_layers=3, hybrid_override_pattern="M*-",
+        num_attention_heads=2, num_key_value_heads=1, head_dim=64,
+        mamba_num_heads=4, mamba_head_dim=32, ssm_state_size=16,
+        n_groups=2, chunk_size=64))
+    nh.init_weights(device="cuda")
+    nh = nh.to(torch.bfloat16)
+    ids = torch.randint(0, 256, (2, 128), device="cuda")
+    nh(ids).float().sum().backward()
+    assert nh.model.layers[0].mixer.in_proj.weight.grad is not None
+
+    l4 = Llama4ForCausalLM(Llama4Config(
+        vocab_size=256, hidden_size=128, intermediate_size=128,
+        intermediate_size_mlp=256, num_hidden_layers=2,
+        num_attention_heads=2, num_key_value_heads=1, head_dim=64,
+        num_local_experts=4, num_experts_per_tok=2,
+        interleave_moe_layer_step=2, max_position_embeddings=256))
+    l4.init_weights(device="cuda")
+    l4 = l4.to(torch.bfloat16)
+    l4(ids).float().sum().backward()
+    assert l4.model.layers[1].feed_forward.router.weight.grad is not None

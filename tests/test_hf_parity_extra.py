@@ -1469,3 +1469,28 @@ def test_hf_logits_parity_nemotron_h():
     ids = torch.randint(0, 200, (2, 21))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_bamba():
+    """Bamba hybrid: shared Mamba2 mixer (full-dim gated norm) + partial-
+    rotary GQA attention layers, per-layer SwiGLU FFN."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.bamba.model import BambaConfig, BambaForCausalLM
+
+    hf_cfg = transformers.BambaConfig(
+        vocab_size=200, hidden_size=64, mamba_expand=2, intermediate_size=160,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        mamba_n_heads=8, mamba_d_head=16, mamba_d_state=8, mamba_n_groups=1,
+        mamba_chunk_size=8, attn_layer_indices=[1],
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(9)
+    hf = transformers.BambaForCausalLM(hf_cfg).eval()
+    cfg = BambaConfig.from_hf_config(hf_cfg.to_dict())
+    mine = BambaForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
