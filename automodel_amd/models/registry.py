@@ -62,6 +62,12 @@ def _ensure_builtin() -> None:
 
         for name in BambaForCausalLM.hf_architectures:
             _REGISTRY[name] = BambaForCausalLM
+        from automodel_amd.models.granitemoe_hybrid.model import (
+            GraniteMoeHybridForCausalLM,
+        )
+
+        for name in GraniteMoeHybridForCausalLM.hf_architectures:
+            _REGISTRY[name] = GraniteMoeHybridForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1494,3 +1494,45 @@ def test_hf_logits_parity_bamba():
     ids = torch.randint(0, 200, (2, 21))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_granitemoe_hybrid():
+    """GraniteMoeHybrid (Granite 4.0): hybrid mamba/attention mixers +
+    granite multipliers + NoPE + topk-then-softmax MoE + shared MLP."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.granitemoe_hybrid.model import (
+        GraniteMoeHybridConfig,
+        GraniteMoeHybridForCausalLM,
+        GraniteMoeHybridStateDictAdapter,
+    )
+
+    hf_cfg = transformers.GraniteMoeHybridConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=48,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2, shared_intermediate_size=96,
+        mamba_n_heads=8, mamba_d_head=16, mamba_d_state=8, mamba_n_groups=1,
+        mamba_chunk_size=8, mamba_expand=2,
+        layer_types=["mamba", "attention", "mamba"],
+        attention_multiplier=0.25, residual_multiplier=0.8,
+        embedding_multiplier=2.0, logits_scaling=1.5,
+        position_embedding_type=None,
+        max_position_embeddings=64, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(10)
+    hf = transformers.GraniteMoeHybridForCausalLM(hf_cfg).eval()
+    cfg = GraniteMoeHybridConfig.from_hf_config(hf_cfg.to_dict())
+    mine = GraniteMoeHybridForCausalLM(cfg).eval()
+    sd = GraniteMoeHybridStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+    # adapter roundtrip
+    rt = GraniteMoeHybridStateDictAdapter().to_hf(
+        {k: v for k, v in mine.state_dict().items() if "rope" not in k})
+    hf_sd = hf.state_dict()
+    assert set(rt) == set(hf_sd)
+    for k in rt:
+        torch.testing.assert_close(rt[k], hf_sd[k])
