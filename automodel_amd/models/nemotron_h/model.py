@@ -172,9 +172,11 @@ class GatedRMSNorm(nn.Module):
         self.eps = eps
         self.group_size = group_size or size
 
-    def forward(self, y: torch.Tensor, z: torch.Tensor) -> torch.Tensor:
+    def forward(self, y: torch.Tensor, z: torch.Tensor | None) -> torch.Tensor:
         dt = y.dtype
-        y = y.float() * torch.nn.functional.silu(z.float())
+        y = y.float()
+        if z is not None:
+            y = y * torch.nn.functional.silu(z.float())
         *lead, last = y.shape
         yg = y.view(*lead, last // self.group_size, self.group_size)
         v = yg.pow(2).mean(-1, keepdim=True)
@@ -191,7 +193,15 @@ class Mamba2Mixer(nn.Module):
                  chunk_size: int = 128, eps: float = 1e-5,
                  use_bias: bool = False, use_conv_bias: bool = True,
                  norm_group_size: int | None = None,
-                 time_step_limit: tuple = (0.0, float("inf"))):
+                 time_step_limit: tuple = (0.0, float("inf")),
+                 gate_mode: str = "norm_gate", in_scale: float = 1.0,
+                 zxbcdt_multipliers: tuple | None = None):
+        """gate_mode: "norm_gate" = GroupRMSNorm(y*silu(z)) (Mamba2/NemotronH/
+        Bamba); "norm_then_gate" = GroupRMSNorm(y)*silu(z) (FalconH1
+        norm_before_gate); "silu_only" = y*silu(z), no norm weight (FalconH1
+        mamba_rms_norm=False). in_scale / zxbcdt_multipliers are FalconH1's
+        muP scalars (ssm_in_multiplier; per-section z,x,B,C,dt scales applied
+        to the in_proj output before the conv)."""
         super().__init__()
         inter = n_heads * head_dim
         self.n_heads, self.head_dim = n_heads, head_dim
@@ -208,14 +218,31 @@ class Mamba2Mixer(nn.Module):
         self.dt_bias = nn.Parameter(torch.ones(n_heads))
         self.A_log = nn.Parameter(torch.zeros(n_heads))
         self.D = nn.Parameter(torch.ones(n_heads))
-        self.norm = GatedRMSNorm(inter, eps, group_size=norm_group_size)
+        self.gate_mode = gate_mode
+        self.in_scale = in_scale
+        self.zxbcdt_multipliers = (tuple(zxbcdt_multipliers)
+                                   if zxbcdt_multipliers else None)
+        if gate_mode == "silu_only":
+            self.norm = None
+        else:
+            self.norm = GatedRMSNorm(inter, eps, group_size=norm_group_size)
         self.out_proj = nn.Linear(inter, hidden_size, bias=use_bias)
 
     def forward(self, h: torch.Tensor) -> torch.Tensor:
         b, s, _ = h.shape
         inter = self.inter
+        if self.in_scale != 1.0:
+            h = h * self.in_scale
         z, xBC, dt = self.in_proj(h).split(
             [inter, self.conv_dim, self.n_heads], dim=-1)
+        if self.zxbcdt_multipliers is not None:
+            m = self.zxbcdt_multipliers
+            gts = self.n_groups * self.state_size
+            z = z * m[0]
+            dt = dt * m[4]
+            xBC = torch.cat([xBC[..., :inter] * m[1],
+                             xBC[..., inter:inter + gts] * m[2],
+                             xBC[..., inter + gts:] * m[3]], dim=-1)
         xBC = torch.nn.functional.silu(
             self.conv1d(xBC.transpose(1, 2))[..., :s].transpose(1, 2))
         x, B, C = xBC.split(
@@ -233,7 +260,14 @@ class Mamba2Mixer(nn.Module):
         y = y + self.D.float().view(1, 1, -1, 1) * \
             x.view(b, s, self.n_heads, self.head_dim).float()
         y = y.to(h.dtype).reshape(b, s, inter)
-        return self.out_proj(self.norm(y, z))
+        if self.gate_mode == "silu_only":
+            y = y * torch.nn.functional.silu(z.float()).to(y.dtype)
+        elif self.gate_mode == "norm_then_gate":
+            y = (self.norm(y, None) *
+                 torch.nn.functional.silu(z.float())).to(y.dtype)
+        else:
+            y = self.norm(y, z)
+        return self.out_proj(y)
 
 
 class NemotronHAttentionMixer(nn.Module):

@@ -68,6 +68,10 @@ def _ensure_builtin() -> None:
 
         for name in GraniteMoeHybridForCausalLM.hf_architectures:
             _REGISTRY[name] = GraniteMoeHybridForCausalLM
+        from automodel_amd.models.falcon_h1.model import FalconH1ForCausalLM
+
+        for name in FalconH1ForCausalLM.hf_architectures:
+            _REGISTRY[name] = FalconH1ForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

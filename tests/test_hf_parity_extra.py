@@ -1536,3 +1536,62 @@ def test_hf_logits_parity_granitemoe_hybrid():
     assert set(rt) == set(hf_sd)
     for k in rt:
         torch.testing.assert_close(rt[k], hf_sd[k])
+
+
+def test_hf_logits_parity_falcon_h1():
+    """FalconH1: parallel mamba+attention per layer, muP multipliers
+    (key/attn-in-out/ssm-in-out/mlp/zxbcdt/embedding/lm-head) applied at
+    runtime, silu-only mamba gate (mamba_rms_norm=False)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.falcon_h1.model import (
+        FalconH1Config,
+        FalconH1ForCausalLM,
+    )
+
+    hf_cfg = transformers.FalconH1Config(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        mamba_n_heads=8, mamba_d_head=16, mamba_d_state=8, mamba_n_groups=2,
+        mamba_chunk_size=8, mamba_expand=2, mamba_d_ssm=128,
+        max_position_embeddings=64,
+        embedding_multiplier=2.0, lm_head_multiplier=0.5, key_multiplier=1.5,
+        attention_in_multiplier=0.9, attention_out_multiplier=1.1,
+        ssm_in_multiplier=0.8, ssm_out_multiplier=1.2,
+        mlp_multipliers=[1.3, 0.7], ssm_multipliers=[0.9, 1.1, 0.8, 1.2, 1.05],
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(11)
+    hf = transformers.FalconH1ForCausalLM(hf_cfg).eval()
+    cfg = FalconH1Config.from_hf_config(hf_cfg.to_dict())
+    mine = FalconH1ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_falcon_h1_gated_norm():
+    """FalconH1 with mamba_rms_norm=True: group norm BEFORE the silu gate."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.falcon_h1.model import (
+        FalconH1Config,
+        FalconH1ForCausalLM,
+    )
+
+    hf_cfg = transformers.FalconH1Config(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        mamba_n_heads=8, mamba_d_head=16, mamba_d_state=8, mamba_n_groups=2,
+        mamba_chunk_size=8, mamba_expand=2, mamba_d_ssm=128,
+        mamba_rms_norm=True, max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(12)
+    hf = transformers.FalconH1ForCausalLM(hf_cfg).eval()
+    cfg = FalconH1Config.from_hf_config(hf_cfg.to_dict())
+    mine = FalconH1ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 200, (2, 17))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
