@@ -67,6 +67,9 @@ _CAPS = {
     "FalconH1ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_packed_sequences=False, flash_head_dims=()),
+    "Qwen3NextForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

@@ -72,6 +72,10 @@ def _ensure_builtin() -> None:
 
         for name in FalconH1ForCausalLM.hf_architectures:
             _REGISTRY[name] = FalconH1ForCausalLM
+        from automodel_amd.models.qwen3_next.model import Qwen3NextForCausalLM
+
+        for name in Qwen3NextForCausalLM.hf_architectures:
+            _REGISTRY[name] = Qwen3NextForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1595,3 +1595,51 @@ def test_hf_logits_parity_falcon_h1_gated_norm():
     ids = torch.randint(0, 200, (2, 17))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_qwen3_next():
+    """Qwen3-Next: GatedDeltaNet linear attention (chunked delta rule via
+    triangular solve), gated GQA with partial rotary + zero-centered norms,
+    softmax-topk MoE with sigmoid-gated shared expert."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.qwen3_next.model import (
+        Qwen3NextConfig,
+        Qwen3NextForCausalLM,
+        Qwen3NextStateDictAdapter,
+    )
+
+    hf_cfg = transformers.Qwen3NextConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, linear_num_value_heads=4, linear_num_key_heads=2,
+        linear_key_head_dim=8, linear_value_head_dim=16,
+        linear_conv_kernel_dim=3, num_experts=4, num_experts_per_tok=2,
+        moe_intermediate_size=32, shared_expert_intermediate_size=48,
+        decoder_sparse_step=1, mlp_only_layers=[1],
+        layer_types=["linear_attention", "full_attention",
+                     "linear_attention", "full_attention"],
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(13)
+    hf = transformers.Qwen3NextForCausalLM(hf_cfg).eval()
+    cfg = Qwen3NextConfig.from_hf_config(hf_cfg.to_dict())
+    mine = Qwen3NextForCausalLM(cfg).eval()
+    sd = Qwen3NextStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+    # multi-chunk sequence (chunk_size 64 -> 2 chunks + padding)
+    ids_long = torch.randint(0, 200, (1, 120))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids_long), hf(ids_long).logits,
+                                   atol=8e-4, rtol=8e-4)
+    # adapter roundtrip
+    rt = Qwen3NextStateDictAdapter().to_hf(
+        {k: v for k, v in mine.state_dict().items() if "rope" not in k})
+    hf_sd = hf.state_dict()
+    assert set(rt) == set(hf_sd)
+    for k in rt:
+        torch.testing.assert_close(rt[k], hf_sd[k])
