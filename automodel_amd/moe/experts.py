@@ -117,6 +117,11 @@ class GroupedExperts(nn.Module):
         self.gate_proj = nn.Parameter(torch.empty(n_experts, intermediate_size, hidden_size))
         self.up_proj = nn.Parameter(torch.empty(n_experts, intermediate_size, hidden_size))
         self.down_proj = nn.Parameter(torch.empty(n_experts, hidden_size, intermediate_size))
+        # opt-in fp8 expert forward (set by quantization.fp8.apply_fp8 /
+        # bench --fp8): fp8-e4m3 grouped NT forward with delayed tensorwise
+        # scaling, bf16 grouped backward
+        self.fp8 = False
+        self._fp8_states = None
 
     def init_weights(self, std: float = 0.02) -> None:
         for p in (self.gate_proj, self.up_proj, self.down_proj):
@@ -181,6 +186,22 @@ class GroupedExperts(nn.Module):
             else:
                 plan = None
                 cl = counts.tolist() if torch.is_tensor(counts) else list(counts)
+            if (self.fp8 and plan is not None
+                    and self.intermediate_size % 128 == 0
+                    and self.hidden_size % 128 == 0):
+                from automodel_amd.ops.grouped_gemm import (
+                    Fp8GroupedState,
+                    grouped_linear_fp8,
+                )
+
+                if self._fp8_states is None:
+                    self._fp8_states = {k: Fp8GroupedState()
+                                        for k in ("gate", "up", "down")}
+                st = self._fp8_states
+                g = grouped_linear_fp8(x_perm, self.gate_proj, cl, plan, st["gate"])
+                u = grouped_linear_fp8(x_perm, self.up_proj, cl, plan, st["up"])
+                h = swiglu(g, u)
+                return grouped_linear_fp8(h, self.down_proj, cl, plan, st["down"])
             g = grouped_linear(x_perm, self.gate_proj, cl, plan=plan)
             u = grouped_linear(x_perm, self.up_proj, cl, plan=plan)
             h = swiglu(g, u)

@@ -79,6 +79,8 @@ TORCH_LIBRARY(amd_ops, m) {
   m.impl("grouped_gemm_nn", &amd_ops::grouped_gemm_nn);
   m.def("grouped_gemm_tn(Tensor g, Tensor x, Tensor offs, int E) -> Tensor");
   m.impl("grouped_gemm_tn", &amd_ops::grouped_gemm_tn);
+  m.def("grouped_gemm_nt_fp8(Tensor x8, Tensor w8, Tensor offs, Tensor tile_map, Tensor scale, Tensor? n_tiles=None, int bm=128) -> Tensor");
+  m.impl("grouped_gemm_nt_fp8", &amd_ops::grouped_gemm_nt_fp8);
   m.def("transpose_bf16(Tensor x) -> Tensor");
   m.impl("transpose_bf16", &amd_ops::transpose_bf16);
   m.def("permute_gather(Tensor x, Tensor src) -> Tensor");

@@ -713,4 +713,245 @@ at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,
   return out;
 }
 
+// ===========================================================================
+// fp8 grouped GEMM forward (SURVEY §2.5 last gap: the reference's MXFP8
+// grouped path, moe/fp8_utils.py). Tensorwise-scaled OCP e4m3 operands,
+// __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8 (bf16 MFMA rate, HALF the
+// HBM/LDS bytes — the bf16 grouped kernels are memory-wait-bound per
+// profiles/gg_r2_pmc_counters.csv, so fp8 wins on bandwidth, not rate).
+// BK = 128 fp8 bytes per stage: rows are 128 B exactly like the bf16
+// tiles, so the proven gg_off swizzle carries over unchanged.
+// ===========================================================================
+
+typedef long fp8x8;   // 8 packed e4m3 bytes = one MFMA operand
+
+__global__ __launch_bounds__(256) void grouped_gemm_nt_fp8_kernel(
+    const unsigned char* __restrict__ x, const unsigned char* __restrict__ w,
+    bf16* __restrict__ y, const int* __restrict__ tile_map,
+    const int* __restrict__ offs, const int* __restrict__ n_tiles,
+    int K, int N, const float* __restrict__ scale_p) {
+  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+  const float scale = scale_p[0];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* xa = smem;                  // [128 m][128 k-bytes] 16 KiB
+  char* wb = smem + GG_BM * 128;
+
+  const int e = tile_map[2 * blockIdx.x];
+  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int m_end = offs[e + 1];
+  const int n0 = blockIdx.y * GG_BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;
+
+  const long wbase = (long)e * N * K;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  const int row0 = tid / 8;          // 8 x 16-B chunks per 128-B row
+  const int c0 = (tid % 8) * 16;
+
+  for (int k0 = 0; k0 < K; k0 += 128) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 32;
+      const int m = m0 + row;
+      int4 xv;
+      if (m < m_end) {
+        xv = *reinterpret_cast<const int4*>(x + (long)m * K + k0 + c0);
+      } else {
+        xv = make_int4(0, 0, 0, 0);
+      }
+      *reinterpret_cast<int4*>(xa + gg_off(row, c0)) = xv;
+      int4 wv = *reinterpret_cast<const int4*>(
+          w + wbase + (long)(n0 + row) * K + k0 + c0);
+      *reinterpret_cast<int4*>(wb + gg_off(row, c0)) = wv;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      fp8x8 a[4], b[4];
+      const int arow = (wid >> 1) * 64;
+      const int brow = (wid & 1) * 64;
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        a[i] = *reinterpret_cast<const fp8x8*>(
+            xa + gg_off(arow + i * 16 + l16, kk * 32 + kq * 8));
+        b[i] = *reinterpret_cast<const fp8x8*>(
+            wb + gg_off(brow + i * 16 + l16, kk * 32 + kq * 8));
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int mw = m0 + (wid >> 1) * 64;
+  const int nw = n0 + (wid & 1) * 64;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mw + i * 16 + kq * 4 + r;
+      if (m < m_end) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          y[(long)m * N + nw + j * 16 + l16] = f2bf(acc[i][j][r] * scale);
+      }
+    }
+  }
+}
+
+// 256x256 fp8 big-tile (8 waves, 64 KiB LDS): the production MoE tile.
+__global__ __launch_bounds__(512, 1) void grouped_gemm_nt_fp8_big_kernel(
+    const unsigned char* __restrict__ x, const unsigned char* __restrict__ w,
+    bf16* __restrict__ y, const int* __restrict__ tile_map,
+    const int* __restrict__ offs, const int* __restrict__ n_tiles,
+    int K, int N, const float* __restrict__ scale_p) {
+  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+  const float scale = scale_p[0];
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* xa = smem;                    // [256 m][128 k-bytes] 32 KiB
+  char* wb = smem + 256 * 128;
+
+  const int e = tile_map[2 * blockIdx.x];
+  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int m_end = offs[e + 1];
+  const int n0 = blockIdx.y * 256;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wid = tid >> 6;
+  const int l16 = lane & 15;
+  const int kq = lane >> 4;
+
+  const long wbase = (long)e * N * K;
+
+  f32x4 acc[4][8];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[i][j][r] = 0.f;
+
+  const int row0 = tid / 8;            // 64 rows per pass (of 256)
+  const int c0 = (tid % 8) * 16;
+
+  for (int k0 = 0; k0 < K; k0 += 128) {
+#pragma unroll
+    for (int rr = 0; rr < 4; ++rr) {
+      const int row = row0 + rr * 64;
+      const int m = m0 + row;
+      int4 xv;
+      if (m < m_end) {
+        xv = *reinterpret_cast<const int4*>(x + (long)m * K + k0 + c0);
+      } else {
+        xv = make_int4(0, 0, 0, 0);
+      }
+      *reinterpret_cast<int4*>(xa + gg_off(row, c0)) = xv;
+      int4 wv = *reinterpret_cast<const int4*>(
+          w + wbase + (long)(n0 + row) * K + k0 + c0);
+      *reinterpret_cast<int4*>(wb + gg_off(row, c0)) = wv;
+    }
+    __syncthreads();
+
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      fp8x8 a[4], b[8];
+      const int arow = (wid >> 1) * 64;
+      const int brow = (wid & 1) * 128;
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        a[i] = *reinterpret_cast<const fp8x8*>(
+            xa + gg_off(arow + i * 16 + l16, kk * 32 + kq * 8));
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        b[j] = *reinterpret_cast<const fp8x8*>(
+            wb + gg_off(brow + j * 16 + l16, kk * 32 + kq * 8));
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              a[i], b[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const int mw = m0 + (wid >> 1) * 64;
+  const int nw = n0 + (wid & 1) * 128;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = mw + i * 16 + kq * 4 + r;
+      if (m < m_end) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          y[(long)m * N + nw + j * 16 + l16] = f2bf(acc[i][j][r] * scale);
+      }
+    }
+  }
+}
+
+at::Tensor grouped_gemm_nt_fp8(const at::Tensor& x8, const at::Tensor& w8,
+                               const at::Tensor& offs, const at::Tensor& tile_map,
+                               const at::Tensor& scale,
+                               const std::optional<at::Tensor>& n_tiles,
+                               int64_t bm) {
+  TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == at::kFloat && scale.numel() == 1,
+              "scale: 1-elem float32 device tensor (combined sx*sw dequant)");
+  TORCH_CHECK(x8.is_cuda() && x8.dim() == 2 && x8.element_size() == 1,
+              "grouped_gemm_nt_fp8: x8 [M,K] e4m3 bytes");
+  TORCH_CHECK(w8.dim() == 3 && w8.element_size() == 1, "w8 [E,N,K] e4m3 bytes");
+  const long M = x8.size(0);
+  const int K = x8.size(1), N = w8.size(1);
+  TORCH_CHECK(K == w8.size(2), "K mismatch");
+  TORCH_CHECK(K % 128 == 0 && N % GG_BN == 0, "need K%128==0, N%128==0");
+  auto y = at::empty({M, (long)N},
+                     x8.options().dtype(at::kBFloat16));
+  const int n_mtiles = tile_map.size(0);
+  if (n_mtiles == 0 || M == 0) return y;
+  auto stream = c10::hip::getCurrentHIPStream();
+  const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
+  if (bm == 256) {
+    TORCH_CHECK(N % 256 == 0, "big-tile fp8 nt needs N%256==0");
+    const dim3 gridb(n_mtiles, N / 256);
+    hipLaunchKernelGGL(grouped_gemm_nt_fp8_big_kernel, gridb, dim3(512),
+                       2 * 256 * 128, stream.stream(),
+                       reinterpret_cast<const unsigned char*>(x8.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(w8.data_ptr()),
+                       reinterpret_cast<bf16*>(y.data_ptr()),
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
+                       scale.data_ptr<float>());
+    HIP_CHECK_KERNEL();
+    return y;
+  }
+  const dim3 grid(n_mtiles, N / GG_BN);
+  hipLaunchKernelGGL(grouped_gemm_nt_fp8_kernel, grid, dim3(256),
+                     2 * GG_BM * 128, stream.stream(),
+                     reinterpret_cast<const unsigned char*>(x8.data_ptr()),
+                       reinterpret_cast<const unsigned char*>(w8.data_ptr()),
+                     reinterpret_cast<bf16*>(y.data_ptr()),
+                     tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
+                     scale.data_ptr<float>());
+  HIP_CHECK_KERNEL();
+  return y;
+}
+
 }  // namespace amd_ops

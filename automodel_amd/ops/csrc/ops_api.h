@@ -78,6 +78,11 @@ at::Tensor grouped_gemm_nn(const at::Tensor& g, const at::Tensor& w,
                            const std::optional<at::Tensor>& n_tiles = std::nullopt);
 at::Tensor grouped_gemm_tn(const at::Tensor& g, const at::Tensor& x,
                            const at::Tensor& offs, int64_t E);
+at::Tensor grouped_gemm_nt_fp8(const at::Tensor& x8, const at::Tensor& w8,
+                               const at::Tensor& offs, const at::Tensor& tile_map,
+                               const at::Tensor& scale,
+                               const std::optional<at::Tensor>& n_tiles = std::nullopt,
+                               int64_t bm = 128);
 at::Tensor transpose_bf16(const at::Tensor& in);
 at::Tensor permute_gather(const at::Tensor& x, const at::Tensor& src);
 at::Tensor unpermute_combine(const at::Tensor& yp, const at::Tensor& pos,

@@ -134,3 +134,99 @@ def grouped_linear(x_perm: torch.Tensor, w: torch.Tensor, counts,
         offs = tile_map = n_tiles = None
         bm = 128
     return _GroupedLinear.apply(x_perm, w, counts, offs, tile_map, n_tiles, bm)
+
+
+# ===========================================================================
+# fp8 grouped forward (SURVEY §2.5 MXFP8-grouped equivalent): tensorwise
+# delayed scaling exactly like quantization/fp8.py Float8Linear — the step's
+# activation scale comes from LAST step's recorded amax (margin 1.25x), the
+# whole expert stack shares one cached e4m3 cast refreshed when the
+# optimizer bumps w._version. Backward stays bf16 (grouped nn/tn kernels)
+# from the saved bf16 operands: fp8 pays on the forward's bandwidth (the
+# bf16 grouped kernels are memory-wait-bound per profiles/gg PMC), grads
+# keep full grouped-bwd accuracy.
+# ===========================================================================
+
+E4M3_MAX = 448.0
+
+
+class Fp8GroupedState:
+    """Delayed-scaling state for one grouped projection (x amax window +
+    cached expert-stack cast)."""
+
+    MARGIN = 1.25
+
+    def __init__(self):
+        self.amax_x = None
+        self.amax_x_prev = None
+        self._wcache = None
+        self._steps = 0
+
+    def _lazy(self, dev):
+        if self.amax_x is None:
+            self.amax_x = torch.zeros(1, device=dev)
+            self.amax_x_prev = torch.zeros(1, device=dev)
+
+    def x_scale(self, x) -> torch.Tensor:
+        self._lazy(x.device)
+        if self._steps == 0:
+            self.amax_x_prev.copy_(
+                x.abs().amax().float().clamp(min=1e-12).reshape(1))
+        return (E4M3_MAX / (self.amax_x_prev.clamp(min=1e-12) * self.MARGIN)
+                ).clamp(max=1e12).reshape(1)
+
+    def cached_w8(self, w):
+        self._lazy(w.device)
+        ver = w._version
+        if self._wcache is not None and self._wcache[0] == ver:
+            return self._wcache[1], self._wcache[2]
+        # weight changed -> step boundary: roll the delayed window
+        self.amax_x_prev.copy_(self.amax_x.clamp(min=1e-12))
+        self.amax_x.zero_()
+        if self._wcache is not None:
+            self._steps += 1
+        wd = w.detach().contiguous()
+        amax = wd.abs().amax().float().clamp(min=1e-12).reshape(1)
+        sw = (E4M3_MAX / amax).clamp(max=1e12)
+        dummy = torch.zeros(1, device=w.device)
+        w8 = hip_ops().fp8_cast(wd.view(-1, wd.shape[-1]), sw, dummy,
+                                False).view(wd.shape)
+        inv_w = sw.reciprocal()
+        self._wcache = (ver, w8, inv_w)
+        return w8, inv_w
+
+
+class _GroupedLinearFp8(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, w, counts, offs, tile_map, n_tiles, bm, state):
+        ops = hip_ops()
+        x = x.contiguous()
+        sx = state.x_scale(x)
+        x8 = ops.fp8_cast(x, sx, state.amax_x, False)
+        w8, inv_w = state.cached_w8(w)
+        dequant = sx.reciprocal() * inv_w
+        y = ops.grouped_gemm_nt_fp8(x8, w8, offs, tile_map, dequant,
+                                    n_tiles, bm)
+        ctx.save_for_backward(x, w)
+        ctx.use_hip = True
+        ctx.counts = counts
+        ctx.plan = (offs, tile_map, n_tiles)
+        ctx.bm = bm
+        return y
+
+    @staticmethod
+    def backward(ctx, g):
+        dx, dw, *_ = _GroupedLinear.backward(ctx, g)
+        return dx, dw, None, None, None, None, None, None
+
+
+def grouped_linear_fp8(x_perm: torch.Tensor, w: torch.Tensor, counts,
+                       plan, state: Fp8GroupedState) -> torch.Tensor:
+    """fp8-forward grouped linear. Caller guarantees bf16 CUDA operands with
+    K%128==0, N%128==0 and a device plan (falls back to bf16 otherwise)."""
+    if (not x_perm.is_cuda or x_perm.dtype != torch.bfloat16
+            or w.shape[2] % 128 or w.shape[1] % 128 or plan is None):
+        return grouped_linear(x_perm, w, counts, plan)
+    offs, tile_map, n_tiles, bm = plan
+    return _GroupedLinearFp8.apply(x_perm, w, counts, offs, tile_map,
+                                   n_tiles, bm, state)

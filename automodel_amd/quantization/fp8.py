@@ -175,4 +175,14 @@ def apply_fp8_to_model(
             ):
                 setattr(module, child_name, Float8Linear.from_linear(child))
                 n += 1
+    # grouped expert stacks: flip the fp8 forward flag (fp8-e4m3 grouped NT
+    # kernel with delayed tensorwise scaling; see ops/grouped_gemm.py)
+    from automodel_amd.moe.experts import GroupedExperts
+
+    for module in model.modules():
+        if (isinstance(module, GroupedExperts)
+                and module.hidden_size % 128 == 0
+                and module.intermediate_size % 128 == 0):
+            module.fp8 = True
+            n += 1
     return n

@@ -43,3 +43,24 @@ assert torch.equal(xtr, x.t().contiguous()), "transpose mismatch"
 xb = torch.randn(M, H, device="cuda", dtype=torch.bfloat16)
 wb = torch.randn(I, H, device="cuda", dtype=torch.bfloat16)
 timeit("hipBLASLt NT same-FLOPs", lambda: xb @ wb.t(), fl)
+
+# ---- fp8 grouped forward (tensorwise-scaled e4m3, BK=128 bytes) ----
+sx = torch.ones(1, device="cuda")
+dummy = torch.zeros(1, device="cuda")
+amax = 448.0 / float(x.abs().amax())
+sx8 = torch.full((1,), amax, device="cuda")
+x8 = ops.fp8_cast(x, sx8, dummy, False)
+wamax = 448.0 / float(w.abs().amax())
+sw8 = torch.full((1,), wamax, device="cuda")
+w8 = ops.fp8_cast(w.view(-1, H), sw8, dummy, False).view(E, I, H)
+deq = (sx8 * sw8).reciprocal()
+timeit("nt-fp8 128 (y=x8@w8T)", lambda: ops.grouped_gemm_nt_fp8(x8, w8, offs, tm, deq, ntl), fl)
+w38 = ops.fp8_cast(w3.view(-1, H), sw8, dummy, False).view(E, 1024, H)
+timeit("nt-fp8 256x256 (N=1024)", lambda: ops.grouped_gemm_nt_fp8(x8, w38, offs_b, tm_b, deq, ntl_b, 256), fl3)
+# numerics vs fp32 reference on a slice
+y8 = ops.grouped_gemm_nt_fp8(x8, w8, offs, tm, deq, ntl)
+ybf = ops.grouped_gemm_nt(x, w, offs, tm, ntl)
+ref = (x[:128].float() @ w[0].float().t())
+err8 = (y8[:128].float() - ref).abs().max() / ref.abs().max()
+errb = (ybf[:128].float() - ref).abs().max() / ref.abs().max()
+print(f"fp8 rel-err {float(err8):.4f} (bf16 path {float(errb):.4f})")
