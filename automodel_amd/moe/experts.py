@@ -198,8 +198,15 @@ class GroupedExperts(nn.Module):
                     self._fp8_states = {k: Fp8GroupedState()
                                         for k in ("gate", "up", "down")}
                 st = self._fp8_states
-                g = grouped_linear_fp8(x_perm, self.gate_proj, cl, plan, st["gate"])
-                u = grouped_linear_fp8(x_perm, self.up_proj, cl, plan, st["up"])
+                from automodel_amd.ops._backend import hip_ops
+
+                xc = x_perm.contiguous()
+                sx = st["gate"].x_scale(xc)
+                x8 = hip_ops().fp8_cast(xc, sx, st["gate"].amax_x, False)
+                g = grouped_linear_fp8(xc, self.gate_proj, cl, plan,
+                                       st["gate"], x8, sx)
+                u = grouped_linear_fp8(xc, self.up_proj, cl, plan,
+                                       st["up"], x8, sx)
                 h = swiglu(g, u)
                 return grouped_linear_fp8(h, self.down_proj, cl, plan, st["down"])
             g = grouped_linear(x_perm, self.gate_proj, cl, plan=plan)

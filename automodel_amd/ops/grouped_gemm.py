@@ -198,11 +198,13 @@ class Fp8GroupedState:
 
 class _GroupedLinearFp8(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, w, counts, offs, tile_map, n_tiles, bm, state):
+    def forward(ctx, x, w, counts, offs, tile_map, n_tiles, bm, state,
+                x8=None, sx=None):
         ops = hip_ops()
         x = x.contiguous()
-        sx = state.x_scale(x)
-        x8 = ops.fp8_cast(x, sx, state.amax_x, False)
+        if x8 is None:
+            sx = state.x_scale(x)
+            x8 = ops.fp8_cast(x, sx, state.amax_x, False)
         w8, inv_w = state.cached_w8(w)
         dequant = sx.reciprocal() * inv_w
         y = ops.grouped_gemm_nt_fp8(x8, w8, offs, tile_map, dequant,
@@ -217,16 +219,19 @@ class _GroupedLinearFp8(torch.autograd.Function):
     @staticmethod
     def backward(ctx, g):
         dx, dw, *_ = _GroupedLinear.backward(ctx, g)
-        return dx, dw, None, None, None, None, None, None
+        return dx, dw, None, None, None, None, None, None, None, None
 
 
 def grouped_linear_fp8(x_perm: torch.Tensor, w: torch.Tensor, counts,
-                       plan, state: Fp8GroupedState) -> torch.Tensor:
+                       plan, state: Fp8GroupedState, x8=None,
+                       sx=None) -> torch.Tensor:
     """fp8-forward grouped linear. Caller guarantees bf16 CUDA operands with
-    K%128==0, N%128==0 and a device plan (falls back to bf16 otherwise)."""
+    K%128==0, N%128==0 and a device plan (falls back to bf16 otherwise).
+    ``x8``/``sx``: share one activation cast across projections of the same
+    input (gate/up)."""
     if (not x_perm.is_cuda or x_perm.dtype != torch.bfloat16
             or w.shape[2] % 128 or w.shape[1] % 128 or plan is None):
         return grouped_linear(x_perm, w, counts, plan)
     offs, tile_map, n_tiles, bm = plan
     return _GroupedLinearFp8.apply(x_perm, w, counts, offs, tile_map,
-                                   n_tiles, bm, state)
+                                   n_tiles, bm, state, x8, sx)

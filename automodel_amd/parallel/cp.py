@@ -201,3 +201,138 @@ def cp_blockdiag_attention(q, k, v, cu_seqlens: torch.Tensor,
     from automodel_amd.ops.attention import sdpa_masked
 
     return sdpa_masked(q, kg, vg, mask, scale)
+
+
+# ===========================================================================
+# Ring-P2P KV context parallelism (reference's ring exchange strategy,
+# context_parallel/exchange.py ring mode; VERDICT r1 §2.4: "no ring-P2P KV
+# rotate"). Peak KV memory is O(S/P) per step instead of the all-gather
+# path's O(S): each rank's zigzag KV block rotates around the cp ring
+# (P2P isend/irecv — on xGMI these ride the direct per-pair links), and
+# attention partials merge by online softmax in fp32. Backward makes the
+# same P-step rotation, accumulating dq locally while dK/dV accumulators
+# travel WITH their KV block and hop home on a final rotation. Exact (same
+# math as flash): parity-tested against cp_flash_attention on gloo world 2.
+# ===========================================================================
+
+
+def _ring_sendrecv(bufs: list[torch.Tensor], cp: CPContext) -> list[torch.Tensor]:
+    """Rotate tensors one hop: send to rank+1, receive from rank-1."""
+    dst = (cp.rank + 1) % cp.world
+    src = (cp.rank - 1) % cp.world
+    outs = [torch.empty_like(b) for b in bufs]
+    reqs = []
+    for b, o in zip(bufs, outs):
+        reqs.append(dist.P2POp(dist.isend, b.contiguous(), dst, group=cp.group))
+        reqs.append(dist.P2POp(dist.irecv, o, src, group=cp.group))
+    for w in dist.batch_isend_irecv(reqs):
+        w.wait()
+    return outs
+
+
+def _ring_pairs(qg: tuple[int, int], kg: tuple[int, int], causal: bool):
+    for qi in (0, 1):
+        for ki in (0, 1):
+            if causal and kg[ki] > qg[qi]:
+                continue
+            yield qi, ki, causal and kg[ki] == qg[qi]
+
+
+class _RingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, cp, causal, scale):
+        P, r = cp.world, cp.rank
+        B, S2, H, D = q.shape
+        C = S2 // 2
+        qg = zigzag_chunk_ids(r, P)
+        qf = q.float().view(B, 2, C, H, D)
+        m = torch.full((B, 2, C, H), float("-inf"), device=q.device)
+        den = torch.zeros_like(m)
+        num = torch.zeros(B, 2, C, H, D, device=q.device)
+        Hk = k.shape[2]
+        rep = H // Hk
+        kb, vb = k.float().view(B, 2, C, Hk, D), v.float().view(B, 2, C, Hk, D)
+        diag = torch.ones(C, C, dtype=torch.bool, device=q.device).triu(1)
+        for step in range(P):
+            src = (r - step) % P
+            kg = zigzag_chunk_ids(src, P)
+            for qi, ki, is_diag in _ring_pairs(qg, kg, causal):
+                ke = kb[:, ki].repeat_interleave(rep, dim=2) if rep > 1 else kb[:, ki]
+                ve = vb[:, ki].repeat_interleave(rep, dim=2) if rep > 1 else vb[:, ki]
+                s = torch.einsum("bqhd,bkhd->bqkh", qf[:, qi], ke) * scale
+                if is_diag:
+                    s = s.masked_fill(diag.view(1, C, C, 1), float("-inf"))
+                bm = s.amax(dim=2)
+                newm = torch.maximum(m[:, qi], bm)
+                p = torch.exp(s - newm.unsqueeze(2))
+                corr = torch.exp(m[:, qi] - newm)
+                den[:, qi] = den[:, qi] * corr + p.sum(dim=2)
+                num[:, qi] = (num[:, qi] * corr.unsqueeze(-1)
+                              + torch.einsum("bqkh,bkhd->bqhd", p, ve))
+                m[:, qi] = newm
+            if step < P - 1:
+                kb, vb = _ring_sendrecv([kb, vb], cp)
+        o = num / den.clamp_min(1e-30).unsqueeze(-1)
+        lse = m + den.clamp_min(1e-30).log()
+        ctx.save_for_backward(q, k, v, o.to(q.dtype), lse)
+        ctx.cp, ctx.causal, ctx.scale = cp, causal, scale
+        return o.view(B, S2, H, D).to(q.dtype)
+
+    @staticmethod
+    def backward(ctx, do):
+        q, k, v, o, lse = ctx.saved_tensors
+        cp, causal, scale = ctx.cp, ctx.causal, ctx.scale
+        P, r = cp.world, cp.rank
+        B, S2, H, D = q.shape
+        C = S2 // 2
+        qg = zigzag_chunk_ids(r, P)
+        qf = q.float().view(B, 2, C, H, D)
+        dof = do.float().view(B, 2, C, H, D)
+        of = o.float().view(B, 2, C, H, D)
+        Dm = (dof * of).sum(-1)                    # [B, 2, C, H]
+        dq = torch.zeros_like(qf)
+        Hk = k.shape[2]
+        rep = H // Hk
+        kb, vb = k.float().view(B, 2, C, Hk, D), v.float().view(B, 2, C, Hk, D)
+        dkb, dvb = torch.zeros_like(kb), torch.zeros_like(vb)
+        diag = torch.ones(C, C, dtype=torch.bool, device=q.device).triu(1)
+
+        def fold(t):   # [B, C, H, D] -> [B, C, Hk, D] (sum over GQA reps)
+            return t if rep == 1 else t.view(B, C, Hk, rep, D).sum(3)
+
+        for step in range(P):
+            src = (r - step) % P
+            kg = zigzag_chunk_ids(src, P)
+            for qi, ki, is_diag in _ring_pairs(qg, kg, causal):
+                ke = kb[:, ki].repeat_interleave(rep, dim=2) if rep > 1 else kb[:, ki]
+                ve = vb[:, ki].repeat_interleave(rep, dim=2) if rep > 1 else vb[:, ki]
+                s = torch.einsum("bqhd,bkhd->bqkh", qf[:, qi], ke) * scale
+                if is_diag:
+                    s = s.masked_fill(diag.view(1, C, C, 1), float("-inf"))
+                p = torch.exp(s - lse[:, qi].unsqueeze(2))
+                dvb[:, ki] += fold(torch.einsum("bqkh,bqhd->bkhd", p, dof[:, qi]))
+                dp = torch.einsum("bqhd,bkhd->bqkh", dof[:, qi], ve)
+                ds = p * (dp - Dm[:, qi].unsqueeze(2)) * scale
+                dq[:, qi] += torch.einsum("bqkh,bkhd->bqhd", ds, ke)
+                dkb[:, ki] += fold(torch.einsum("bqkh,bqhd->bkhd", ds, qf[:, qi]))
+            # rotate kv + their grad accumulators together; the final hop
+            # returns each block's dK/dV to its owner
+            kb, vb, dkb, dvb = _ring_sendrecv([kb, vb, dkb, dvb], cp)
+        return (dq.view(B, S2, H, D).to(q.dtype),
+                dkb.view(B, S2, Hk, D).to(k.dtype),
+                dvb.view(B, S2, Hk, D).to(v.dtype), None, None, None)
+
+
+def cp_ring_attention(q, k, v, causal: bool = True,
+                      scale: float | None = None) -> torch.Tensor:
+    """Ring-P2P CP attention over zigzag shards [B, 2C, H, D] -> local O.
+    Memory-bounded alternative to cp_flash_attention (KV never gathered)."""
+    import math
+
+    cp = _ACTIVE_CP
+    assert cp is not None, "cp_ring_attention called without enable_cp"
+    scale = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+    if cp.world == 1:
+        return flash_attention(q, k, v, causal=causal, scale=scale,
+                               backend="torch")
+    return _RingAttention.apply(q, k, v, cp, causal, scale)

