@@ -27,16 +27,19 @@ class CPContext:
     group: object
     rank: int
     world: int
+    mechanism: str = "allgather"   # "allgather" (xGMI-preferred) | "ring"
 
 
 _ACTIVE_CP: CPContext | None = None
 
 
-def enable_cp(mesh_axis) -> CPContext:
+def enable_cp(mesh_axis, mechanism: str = "allgather") -> CPContext:
     global _ACTIVE_CP
+    assert mechanism in ("allgather", "ring"), mechanism
     group = mesh_axis.get_group() if hasattr(mesh_axis, "get_group") else mesh_axis
     _ACTIVE_CP = CPContext(group=group, rank=dist.get_rank(group),
-                           world=dist.get_world_size(group))
+                           world=dist.get_world_size(group),
+                           mechanism=mechanism)
     return _ACTIVE_CP
 
 
@@ -126,6 +129,8 @@ def cp_flash_attention(q, k, v, causal: bool = True, scale: float | None = None,
     """q/k/v local zigzag shards [B, 2C, H, D]; returns local O shard."""
     cp = _ACTIVE_CP
     assert cp is not None, "cp_flash_attention called without enable_cp"
+    if cp.mechanism == "ring":
+        return cp_ring_attention(q, k, v, causal=causal, scale=scale)
     P = cp.world
     B, S2 = q.shape[0], q.shape[1]
     C = S2 // 2

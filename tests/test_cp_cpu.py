@@ -292,3 +292,34 @@ def test_cp2_ring_attention_parity():
     """Ring-P2P KV rotation (zigzag, batch_isend_irecv) matches dense
     attention forward AND backward — incl. GQA head folding."""
     run_distributed(_cp_ring_fn, world=2)
+
+
+def _cp_mech_fn(rank, world):
+    import torch.distributed as dist
+
+    from automodel_amd.ops.attention import attention_ref
+    from automodel_amd.parallel.cp import cp_flash_attention, disable_cp, enable_cp
+
+    torch.manual_seed(4)
+    B, S, H, D = 1, 16, 2, 8
+    q = torch.randn(B, S, H, D)
+    k = torch.randn(B, S, H, D)
+    v = torch.randn(B, S, H, D)
+    o_ref = attention_ref(q, k, v, causal=True)
+    enable_cp(dist.group.WORLD, mechanism="ring")
+    try:
+        C = S // (2 * world)
+        g0, g1 = zigzag_chunk_ids(rank, world)
+        sel = torch.cat([torch.arange(g0 * C, (g0 + 1) * C),
+                         torch.arange(g1 * C, (g1 + 1) * C)])
+        o = cp_flash_attention(q[:, sel].contiguous(), k[:, sel].contiguous(),
+                               v[:, sel].contiguous(), causal=True)
+        assert torch.allclose(o, o_ref[:, sel], atol=1e-4)
+    finally:
+        disable_cp()
+    return True
+
+
+def test_cp2_mechanism_ring_dispatch():
+    """enable_cp(mechanism='ring') routes cp_flash_attention to the ring."""
+    run_distributed(_cp_mech_fn, world=2)
