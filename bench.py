@@ -20,6 +20,11 @@ import json
 import os
 import time
 
+# fp8 expert-weight caches push the 30B MoE config near the 288 GB line;
+# expandable segments avoid the allocator fragmentation that tips it over.
+# Must be set before the first CUDA allocation.
+os.environ.setdefault("PYTORCH_ALLOC_CONF", "expandable_segments:True")
+
 import torch
 import torch.distributed as dist
 
