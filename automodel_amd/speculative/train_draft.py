@@ -86,6 +86,12 @@ class TrainEagleDraftRecipe:
             lr=opt_cfg.get("lr", 1e-4), weight_decay=opt_cfg.get("weight_decay", 0.01),
         )
         self.hard_weight = cfg.get("hard_label_weight", 0.1)
+        # EAGLE-3 training-time multi-step unroll (TTT): after the first
+        # step the draft conditions on its OWN previous hidden state, so
+        # training sees the same drift it will face at decode time
+        # (reference train_eagle3.py recipe_args.ttt_steps)
+        self.ttt_steps = cfg.get("ttt_steps", 1)
+        self.ttt_decay = cfg.get("ttt_decay", 1.0)
         self.max_steps = cfg.get_by_dotted("step_scheduler.max_steps", 100)
 
         dl_cfg = cfg.get("dataloader", ConfigNode())
@@ -103,18 +109,32 @@ class TrainEagleDraftRecipe:
         with torch.no_grad():
             z, aux = self.target.forward_with_aux(input_ids)
         # fuse_aux is the draft's trainable fc — OUTSIDE no_grad
-        carry = self.draft.fuse_aux([a.detach() for a in aux])
-        d_logits = self.draft(input_ids[:, 1:], carry[:, :-1])
-        loss = eagle_distill_loss(d_logits, z[:, 1:], input_ids[:, 1:],
-                                  self.hard_weight)
-        # one-step draft accuracy vs the target's greedy choice (the number
-        # that predicts decode acceptance rate)
-        with torch.no_grad():
-            acc = (d_logits.argmax(-1) == z[:, 1:].argmax(-1)).float().mean()
+        carry = self.draft.fuse_aux([a.detach() for a in aux])[:, :-1]
+        ids, tgt, hard = input_ids[:, 1:], z[:, 1:], input_ids[:, 1:]
+        total = None
+        accs = []
+        for s in range(self.ttt_steps):
+            h = self.draft.backbone(ids, carry)
+            d_logits = self.draft.lm_head(h)
+            step_loss = eagle_distill_loss(d_logits, tgt, hard, self.hard_weight)
+            w = self.ttt_decay ** s
+            total = step_loss * w if total is None else total + step_loss * w
+            # per-unroll-depth draft accuracy vs the target's greedy choice
+            # (the number that predicts decode acceptance at depth s+1)
+            with torch.no_grad():
+                accs.append(float((d_logits.argmax(-1) == tgt.argmax(-1))
+                                  .float().mean()))
+            if s + 1 < self.ttt_steps:
+                # next depth: the draft rides its OWN hidden, shifted one
+                carry = h[:, :-1]
+                ids, tgt, hard = ids[:, 1:], tgt[:, 1:], hard[:, 1:]
         self.optimizer.zero_grad(set_to_none=True)
-        loss.backward()
+        total.backward()
         self.optimizer.step()
-        return {"loss": float(loss.detach()), "draft_top1_agreement": float(acc)}
+        out = {"loss": float(total.detach()), "draft_top1_agreement": accs[0]}
+        for s, a in enumerate(accs[1:], start=2):
+            out[f"draft_top1_agreement_depth{s}"] = a
+        return out
 
     def run_train_validation_loop(self) -> list[dict]:
         """CLI/launcher entry point (same contract as the other recipes)."""

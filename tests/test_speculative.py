@@ -215,3 +215,30 @@ def test_generate_cached_rejects_cache_blind_model():
 
     with pytest.raises(AssertionError, match="consult the KV cache"):
         generate_cached(Blind(), torch.randint(0, 50, (1, 4)), max_new_tokens=2)
+
+
+def test_eagle_ttt_multistep_training():
+    """EAGLE-3 training-time unroll: ttt_steps=3 trains through the draft's
+    own hiddens; loss decreases and per-depth agreement is reported."""
+    from automodel_amd.config.loader import ConfigNode
+    from automodel_amd.speculative.train_draft import TrainEagleDraftRecipe
+
+    cfg = ConfigNode({
+        "seed": 5, "ttt_steps": 3, "ttt_decay": 0.8,
+        "model": {"config": {"vocab_size": 96, "hidden_size": 32,
+                             "intermediate_size": 48, "num_hidden_layers": 3,
+                             "num_attention_heads": 4, "num_key_value_heads": 2,
+                             "max_position_embeddings": 128}},
+        "draft": {"num_layers": 1},
+        "optimizer": {"lr": 3e-3},
+        "dataloader": {"batch_size": 2,
+                       "dataset": {"num_samples": 8, "seq_len": 32}},
+        "step_scheduler": {"max_steps": 6},
+    })
+    r = TrainEagleDraftRecipe(cfg)
+    r.setup()
+    logs = r.run()
+    assert len(logs) == 6
+    assert "draft_top1_agreement_depth2" in logs[0]
+    assert "draft_top1_agreement_depth3" in logs[0]
+    assert logs[-1]["loss"] < logs[0]["loss"]
