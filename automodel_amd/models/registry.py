@@ -76,6 +76,10 @@ def _ensure_builtin() -> None:
 
         for name in Qwen3NextForCausalLM.hf_architectures:
             _REGISTRY[name] = Qwen3NextForCausalLM
+        from automodel_amd.models.lfm2.model import Lfm2ForCausalLM
+
+        for name in Lfm2ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Lfm2ForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -180,8 +180,11 @@ def main():
     model.train()
     # 30B-class single-GPU: bf16 optimizer states (fp32 master+m+v for 30.5B
     # params is 366 GB > 288 GB HBM); n>=2 shards fp32 states via FSDP2
-    state_dtype = torch.bfloat16 if (is_moe and args.model == "qwen3_moe_30b"
-                                     and world == 1) else torch.float32
+    # fp8 runs additionally carry the cached e4m3 weight casts (~0.5 GB per
+    # MoE layer) — bf16 states buy that headroom back at n=1
+    state_dtype = torch.bfloat16 if (is_moe and world == 1
+                                     and (args.model == "qwen3_moe_30b"
+                                          or args.fp8)) else torch.float32
     params = [p for p in model.parameters() if p.requires_grad]
     opt = FusedAdamW(params, lr=2e-5, weight_decay=0.0,
                      state_dtype=state_dtype)

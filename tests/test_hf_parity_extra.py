@@ -1643,3 +1643,27 @@ def test_hf_logits_parity_qwen3_next():
     assert set(rt) == set(hf_sd)
     for k in rt:
         torch.testing.assert_close(rt[k], hf_sd[k])
+
+
+def test_hf_logits_parity_lfm2():
+    """LFM2: gated short-conv layers (no activation) + q/k-normed GQA
+    attention, auto-adjusted SwiGLU width (w1/w3/w2)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.lfm2.model import Lfm2Config, Lfm2ForCausalLM
+
+    hf_cfg = transformers.Lfm2Config(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        conv_L_cache=3, layer_types=["conv", "full_attention", "conv"],
+        max_position_embeddings=64, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(14)
+    hf = transformers.Lfm2ForCausalLM(hf_cfg).eval()
+    cfg = Lfm2Config.from_hf_config(hf_cfg.to_dict())
+    mine = Lfm2ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
