@@ -201,8 +201,12 @@ class GroupedExperts(nn.Module):
                 from automodel_amd.ops._backend import hip_ops
 
                 xc = x_perm.contiguous()
-                sx = st["gate"].x_scale(xc)
-                x8 = hip_ops().fp8_cast(xc, sx, st["gate"].amax_x, False)
+                # non-differentiable byproducts: cast under no_grad so the
+                # fp8 bytes don't drag an autograd-fallback node into BOTH
+                # gate and up Function graphs
+                with torch.no_grad():
+                    sx = st["gate"].x_scale(xc)
+                    x8 = hip_ops().fp8_cast(xc, sx, st["gate"].amax_x, False)
                 g = grouped_linear_fp8(xc, self.gate_proj, cl, plan,
                                        st["gate"], x8, sx)
                 u = grouped_linear_fp8(xc, self.up_proj, cl, plan,
