@@ -80,6 +80,10 @@ def _ensure_builtin() -> None:
 
         for name in Lfm2ForCausalLM.hf_architectures:
             _REGISTRY[name] = Lfm2ForCausalLM
+        from automodel_amd.models.jamba.model import JambaForCausalLM
+
+        for name in JambaForCausalLM.hf_architectures:
+            _REGISTRY[name] = JambaForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1667,3 +1667,35 @@ def test_hf_logits_parity_lfm2():
     ids = torch.randint(0, 200, (2, 21))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_jamba():
+    """Jamba: Mamba-1 selective scan (chunked segsum) with dt/B/C norms,
+    NoPE attention, period/offset-scheduled MoE (softmax-topk, no renorm)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.jamba.model import (
+        JambaConfig,
+        JambaForCausalLM,
+        JambaStateDictAdapter,
+    )
+
+    hf_cfg = transformers.JambaConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        mamba_d_state=8, mamba_d_conv=3, mamba_expand=2, mamba_dt_rank=8,
+        num_experts=4, num_experts_per_tok=2,
+        expert_layer_period=2, expert_layer_offset=1,
+        attn_layer_period=4, attn_layer_offset=2,
+        max_position_embeddings=64, attn_implementation="eager",
+        use_mamba_kernels=False, tie_word_embeddings=False)
+    torch.manual_seed(15)
+    hf = transformers.JambaForCausalLM(hf_cfg).eval()
+    cfg = JambaConfig.from_hf_config(hf_cfg.to_dict())
+    mine = JambaForCausalLM(cfg).eval()
+    sd = JambaStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    ids = torch.randint(0, 200, (2, 37))   # crosses the chunk-16 boundary
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
