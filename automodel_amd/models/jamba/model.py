@@ -46,10 +46,11 @@ def mamba1_chunked_scan(x, dt, A, B, C, chunk: int = 16):
          ).view(Bb, n, chunk, D, N)
     Cc = C.float().view(Bb, n, chunk, N)
     cum = la.cumsum(2)
-    # pairwise decay within the chunk: exp(cum_t - cum_s), t >= s (<= 1)
-    dec = (cum.unsqueeze(3) - cum.unsqueeze(2)).exp()
+    # pairwise decay within the chunk: exp(cum_t - cum_s), t >= s (<= 1);
+    # mask BEFORE exp — the t<s half is exp(positive) and would inf*0=NaN
     mask = torch.ones(chunk, chunk, dtype=torch.bool, device=x.device).tril()
-    dec = dec * mask.view(1, 1, chunk, chunk, 1, 1)
+    dec = (cum.unsqueeze(3) - cum.unsqueeze(2)).masked_fill(
+        ~mask.view(1, 1, chunk, chunk, 1, 1), float("-inf")).exp()
     h = torch.einsum("bctsdn,bcsdn->bctdn", dec, b)
     state = x.new_zeros(Bb, D, N, dtype=torch.float32)
     ys = []
