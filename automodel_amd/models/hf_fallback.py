@@ -142,6 +142,11 @@ def build_hf_fallback(
         model = _try(lambda impl: AutoModelForCausalLM.from_config(
             hf_cfg, attn_implementation=impl))
         model = model.to(torch_dtype)
+    # kernel-patch ladder (reference model_init.py:1431): swap matching HF
+    # modules for the MI355X-native ops (HIP RMSNorm, fused SwiGLU)
+    from automodel_amd.models.hf_patches import apply_kernel_patches
+
+    apply_kernel_patches(model)
     wrapped = HFFallbackForCausalLM(model, hf_cfg)
     if device is not None and str(device) != "meta":
         wrapped = wrapped.to(device)

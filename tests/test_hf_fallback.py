@@ -84,3 +84,49 @@ def test_fallback_fsdp_layer_detection():
     layers = detect_decoder_layers(m)
     assert len(layers) == 2
     assert all(type(l).__name__ == "XGLMDecoderLayer" for l in layers)
+
+
+def test_kernel_patch_ladder_preserves_logits():
+    """apply_kernel_patches swaps HF RMSNorms + silu MLPs for the native ops
+    without changing logits, reusing the SAME weight Parameters."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.hf_patches import apply_kernel_patches
+    from automodel_amd.ops.rms_norm import RMSNorm
+
+    cfg = transformers.LlamaConfig(
+        vocab_size=128, hidden_size=32, intermediate_size=48,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        max_position_embeddings=64, attn_implementation="eager")
+    torch.manual_seed(0)
+    m = transformers.LlamaForCausalLM(cfg).eval()
+    ids = torch.randint(0, 128, (2, 12))
+    with torch.no_grad():
+        ref = m(ids).logits
+    params_before = {n: p for n, p in m.named_parameters()}
+    counts = apply_kernel_patches(m)
+    assert counts["rms_norm"] >= 5 and counts["swiglu_mlp"] == 2, counts
+    assert any(isinstance(mod, RMSNorm) for mod in m.modules())
+    with torch.no_grad():
+        out = m(ids).logits
+    torch.testing.assert_close(out, ref, atol=2e-5, rtol=2e-5)
+    # same Parameter objects (optimizer/checkpoint identity preserved)
+    for n, p in m.named_parameters():
+        assert params_before[n] is p, n
+
+
+def test_fallback_build_applies_patches():
+    pytest.importorskip("transformers")
+    from automodel_amd.models.hf_fallback import build_hf_fallback
+    from automodel_amd.ops.rms_norm import RMSNorm
+
+    m = build_hf_fallback(
+        config=dict(model_type="llama", vocab_size=96, hidden_size=32,
+                    intermediate_size=48, num_hidden_layers=2,
+                    num_attention_heads=4, num_key_value_heads=2,
+                    max_position_embeddings=64),
+        architecture="LlamaForCausalLM", dtype="float32")
+    assert any(isinstance(mod, RMSNorm) for mod in m.modules())
+    ids = torch.randint(0, 96, (1, 8))
+    labels = ids.clone()
+    loss = m(ids, labels=labels)
+    assert torch.isfinite(loss)
