@@ -46,8 +46,9 @@ def _is_silu_gate_mlp(mod: nn.Module) -> bool:
         if not isinstance(child, nn.Linear):
             return False
     act = getattr(mod, "act_fn", None)
-    return isinstance(act, nn.SiLU) or (
-        act is not None and getattr(act, "__name__", "") == "silu")
+    return (isinstance(act, nn.SiLU)
+            or type(act).__name__ == "SiLUActivation"
+            or getattr(act, "__name__", "") == "silu")
 
 
 def _fused_mlp_forward(self, x):
