@@ -570,3 +570,36 @@ def _recipe_pp_dp_fn(rank, world, tmpdir):
 def test_recipe_pp2_x_dp2_world4(tmp_path):
     out = run_distributed(_recipe_pp_dp_fn, world=4, args=(str(tmp_path),))
     assert all(v == 2 for v in out.values())
+
+
+def test_selective_op_ac_grad_parity():
+    """selective_ops AC: GEMM outputs saved, elementwise recomputed — grads
+    bitwise-match the no-AC model (reference selective-op AC mode)."""
+    import copy
+
+    from automodel_amd.models.llama.model import LlamaConfig, LlamaForCausalLM
+    from automodel_amd.parallel.activation_checkpointing import apply_ac
+
+    torch.manual_seed(0)
+    cfg = LlamaConfig(vocab_size=96, hidden_size=32, intermediate_size=48,
+                      num_hidden_layers=2, num_attention_heads=4,
+                      num_key_value_heads=2, max_position_embeddings=64)
+    from automodel_amd.loss.linear_ce import FusedLinearCrossEntropy
+
+    m1 = LlamaForCausalLM(cfg)
+    m1.init_weights()
+    m1.loss_fn = FusedLinearCrossEntropy(backend="torch")
+    m2 = copy.deepcopy(m1)
+    m2.loss_fn = FusedLinearCrossEntropy(backend="torch")
+    apply_ac(m2, mode="selective_ops")
+    ids = torch.randint(0, 96, (2, 16))
+    labels = ids.clone()
+    l1 = m1(ids, labels=labels)
+    l1.backward()
+    l2 = m2(ids, labels=labels)
+    l2.backward()
+    assert torch.equal(l1.detach(), l2.detach())
+    g1 = {n: p.grad for n, p in m1.named_parameters()}
+    for n, p in m2.named_parameters():
+        n_clean = n.replace("_checkpoint_wrapped_module.", "")
+        assert torch.equal(g1[n_clean], p.grad), n_clean
