@@ -67,6 +67,8 @@ TORCH_LIBRARY(amd_ops, m) {
   m.impl("fp8_transpose", &amd_ops::fp8_transpose);
   m.def("lora_fused_fwd(Tensor x, Tensor A, Tensor B, float scale) -> Tensor");
   m.impl("lora_fused_fwd", &amd_ops::lora_fused_fwd);
+  m.def("sgmv_fused_fwd(Tensor x, Tensor A, Tensor B, Tensor scales, Tensor offs, Tensor tile_map, Tensor? n_tiles=None) -> Tensor");
+  m.impl("sgmv_fused_fwd", &amd_ops::sgmv_fused_fwd);
 
   m.def("gemv_bf16(Tensor x, Tensor w, Tensor? bias) -> Tensor");
   m.impl("gemv_bf16", &amd_ops::gemv_bf16);

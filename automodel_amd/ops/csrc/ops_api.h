@@ -63,6 +63,10 @@ at::Tensor fp8_cast(const at::Tensor& x, const at::Tensor& scale, at::Tensor ama
 at::Tensor fp8_transpose(const at::Tensor& x8);
 at::Tensor lora_fused_fwd(const at::Tensor& x, const at::Tensor& A,
                           const at::Tensor& B, double scale);
+at::Tensor sgmv_fused_fwd(const at::Tensor& x, const at::Tensor& A,
+                          const at::Tensor& B, const at::Tensor& scales,
+                          const at::Tensor& offs, const at::Tensor& tile_map,
+                          const std::optional<at::Tensor>& n_tiles = std::nullopt);
 
 at::Tensor gemv_bf16(const at::Tensor& x, const at::Tensor& w,
                      const c10::optional<at::Tensor>& bias);
