@@ -38,8 +38,14 @@ def apply_ep(model: nn.Module, ep_mesh, dispatcher: str = "a2a") -> nn.Module:
                     local = nn.Parameter(full[offset : offset + n_local].clone())
                     setattr(m.experts, name, local)
             m.experts.n_experts = n_local
-            disp_cls = AllToAllDispatcher if dispatcher == "a2a" else AllGatherDispatcher
-            m.dispatcher = disp_cls(group, E, offset, n_local)
+            if dispatcher == "a2a_pipelined":
+                from automodel_amd.moe.dispatch import PipelinedAllToAllDispatcher
+
+                m.dispatcher = PipelinedAllToAllDispatcher(group, E, offset, n_local)
+            else:
+                disp_cls = (AllToAllDispatcher if dispatcher == "a2a"
+                            else AllGatherDispatcher)
+                m.dispatcher = disp_cls(group, E, offset, n_local)
     return model
 
 

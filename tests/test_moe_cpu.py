@@ -184,7 +184,7 @@ def _ep_fn(rank, world, dispatcher):
     return float(y_ep.sum())
 
 
-@pytest.mark.parametrize("dispatcher", ["a2a", "allgather"])
+@pytest.mark.parametrize("dispatcher", ["a2a", "allgather", "a2a_pipelined"])
 def test_ep2_dispatcher_matches_dense(dispatcher):
     run_distributed(_ep_fn, world=2, args=(dispatcher,))
 
