@@ -84,6 +84,10 @@ def _ensure_builtin() -> None:
 
         for name in JambaForCausalLM.hf_architectures:
             _REGISTRY[name] = JambaForCausalLM
+        from automodel_amd.models.zamba2.model import Zamba2ForCausalLM
+
+        for name in Zamba2ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Zamba2ForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1699,3 +1699,86 @@ def test_hf_logits_parity_jamba():
     ids = torch.randint(0, 200, (2, 37))   # crosses the chunk-16 boundary
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_zamba2():
+    """Zamba2: shared transformer blocks over concat(hidden, embeds) with
+    per-use gate_up adapters, (head_dim/2)^-0.5 scale, Mamba2 backbone."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.zamba2.model import Zamba2Config, Zamba2ForCausalLM
+
+    hf_cfg = transformers.Zamba2Config(
+        vocab_size=200, hidden_size=64, num_hidden_layers=6,
+        layers_block_type=["mamba", "mamba", "hybrid", "mamba", "mamba", "hybrid"],
+        attention_head_dim=16, num_attention_heads=8, num_key_value_heads=8,
+        mamba_d_state=8, mamba_d_conv=4, mamba_expand=2, mamba_ngroups=1,
+        mamba_headdim=16, intermediate_size=128, adapter_rank=8,
+        num_mem_blocks=2, max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(16)
+    hf = transformers.Zamba2ForCausalLM(hf_cfg).eval()
+    cfg = Zamba2Config.from_hf_config(hf_cfg.to_dict())
+    mine = Zamba2ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_zamba2_rope_shared_adapters():
+    """Zamba2 with use_mem_rope + shared attention adapters + ONE mem block
+    reused by both hybrid layers (true weight sharing)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.zamba2.model import Zamba2Config, Zamba2ForCausalLM
+
+    hf_cfg = transformers.Zamba2Config(
+        vocab_size=200, hidden_size=64, num_hidden_layers=5,
+        layers_block_type=["mamba", "hybrid", "mamba", "hybrid", "mamba"],
+        attention_head_dim=16, num_attention_heads=8, num_key_value_heads=8,
+        mamba_d_state=8, mamba_d_conv=4, mamba_expand=2, mamba_ngroups=1,
+        mamba_headdim=16, intermediate_size=128, adapter_rank=8,
+        num_mem_blocks=1, use_mem_rope=True, use_shared_attention_adapter=True,
+        max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(17)
+    hf = transformers.Zamba2ForCausalLM(hf_cfg).eval()
+    cfg = Zamba2Config.from_hf_config(hf_cfg.to_dict())
+    mine = Zamba2ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 17))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_zamba2_tie_shared_blocks():
+    """tie_shared_blocks: hybrid layers of the same block_id share ONE
+    module; per-use adapters survive grafting; forward stays finite."""
+    from automodel_amd.models.zamba2.model import Zamba2Config, Zamba2ForCausalLM
+
+    cfg = Zamba2Config(
+        vocab_size=100, hidden_size=32, intermediate_size=64,
+        num_hidden_layers=5,
+        layers_block_type=["mamba", "hybrid", "mamba", "hybrid", "mamba"],
+        num_attention_heads=4, num_key_value_heads=4, attention_head_dim=16,
+        mamba_d_state=8, mamba_headdim=8, mamba_ngroups=1, adapter_rank=4,
+        num_mem_blocks=1, max_position_embeddings=64, tie_word_embeddings=False)
+    torch.manual_seed(0)
+    m = Zamba2ForCausalLM(cfg)
+    m.init_weights()
+    l1, l3 = m.model.layers[1], m.model.layers[3]
+    assert l1.shared_transformer is not l3.shared_transformer
+    # make use-1's adapter distinguishable before tying
+    a1 = l3.shared_transformer.feed_forward.gate_up_proj_adapter_list[1]
+    assert not isinstance(a1, torch.nn.Identity)
+    ties = m.tie_shared_blocks()
+    assert ties == 1
+    assert l1.shared_transformer is l3.shared_transformer
+    assert l1.shared_transformer.feed_forward.gate_up_proj_adapter_list[1] is a1
+    ids = torch.randint(0, 100, (1, 12))
+    with torch.no_grad():
+        out = m(ids)
+    assert torch.isfinite(out).all()
