@@ -79,6 +79,15 @@ _CAPS = {
     "Zamba2ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_packed_sequences=False, flash_head_dims=()),
+    "MambaForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
+    "Mamba2ForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
+    "FalconMambaForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

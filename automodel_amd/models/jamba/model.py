@@ -66,23 +66,34 @@ class JambaMambaMixer(nn.Module):
 
     def __init__(self, hidden_size: int, intermediate: int, state_size: int,
                  dt_rank: int, conv_kernel: int = 4, eps: float = 1e-6,
-                 use_bias: bool = False, use_conv_bias: bool = True):
+                 use_bias: bool = False, use_conv_bias: bool = True,
+                 dt_bc_norms: str = "weighted"):
+        """dt_bc_norms: "weighted" (Jamba), "weightless" (FalconMamba),
+        "none" (vanilla Mamba)."""
         super().__init__()
         self.inter = intermediate
         self.state_size = state_size
         self.dt_rank = dt_rank
+        self.dt_bc_norms = dt_bc_norms
+        self.eps = eps
         self.in_proj = nn.Linear(hidden_size, 2 * intermediate, bias=use_bias)
         self.conv1d = nn.Conv1d(intermediate, intermediate, conv_kernel,
                                 groups=intermediate, padding=conv_kernel - 1,
                                 bias=use_conv_bias)
         self.x_proj = nn.Linear(intermediate, dt_rank + 2 * state_size, bias=False)
         self.dt_proj = nn.Linear(dt_rank, intermediate, bias=True)
-        self.dt_layernorm = RMSNorm(dt_rank, eps, "torch")
-        self.b_layernorm = RMSNorm(state_size, eps, "torch")
-        self.c_layernorm = RMSNorm(state_size, eps, "torch")
+        if dt_bc_norms == "weighted":
+            self.dt_layernorm = RMSNorm(dt_rank, eps, "torch")
+            self.b_layernorm = RMSNorm(state_size, eps, "torch")
+            self.c_layernorm = RMSNorm(state_size, eps, "torch")
         self.A_log = nn.Parameter(torch.zeros(intermediate, state_size))
         self.D = nn.Parameter(torch.ones(intermediate))
         self.out_proj = nn.Linear(intermediate, hidden_size, bias=use_bias)
+
+    @staticmethod
+    def _rms(t: torch.Tensor, eps: float) -> torch.Tensor:
+        tf = t.float()
+        return (tf * torch.rsqrt(tf.pow(2).mean(-1, keepdim=True) + eps)).to(t.dtype)
 
     def forward(self, h: torch.Tensor) -> torch.Tensor:
         S = h.shape[1]
@@ -90,9 +101,14 @@ class JambaMambaMixer(nn.Module):
         x = F.silu(self.conv1d(x.transpose(1, 2))[..., :S].transpose(1, 2))
         dt_r, B, C = self.x_proj(x).split(
             [self.dt_rank, self.state_size, self.state_size], dim=-1)
-        dt_r = self.dt_layernorm(dt_r)
-        B = self.b_layernorm(B)
-        C = self.c_layernorm(C)
+        if self.dt_bc_norms == "weighted":
+            dt_r = self.dt_layernorm(dt_r)
+            B = self.b_layernorm(B)
+            C = self.c_layernorm(C)
+        elif self.dt_bc_norms == "weightless":
+            dt_r = self._rms(dt_r, self.eps)
+            B = self._rms(B, self.eps)
+            C = self._rms(C, self.eps)
         dt = F.softplus(F.linear(dt_r, self.dt_proj.weight).float()
                         + self.dt_proj.bias.float())
         A = -torch.exp(self.A_log.float())

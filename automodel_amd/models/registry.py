@@ -88,6 +88,15 @@ def _ensure_builtin() -> None:
 
         for name in Zamba2ForCausalLM.hf_architectures:
             _REGISTRY[name] = Zamba2ForCausalLM
+        from automodel_amd.models.mamba_lm.model import (
+            FalconMambaForCausalLM,
+            Mamba2ForCausalLM,
+            MambaForCausalLM,
+        )
+
+        for cls2 in (MambaForCausalLM, Mamba2ForCausalLM, FalconMambaForCausalLM):
+            for name in cls2.hf_architectures:
+                _REGISTRY[name] = cls2
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

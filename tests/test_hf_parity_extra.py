@@ -1782,3 +1782,47 @@ def test_zamba2_tie_shared_blocks():
     with torch.no_grad():
         out = m(ids)
     assert torch.isfinite(out).all()
+
+
+@pytest.mark.parametrize("kind", ["mamba", "mamba2", "falcon_mamba"])
+def test_hf_logits_parity_pure_mamba(kind):
+    """Pure SSM LMs (Mamba / Mamba2 / FalconMamba) on the shared mixers;
+    FalconMamba exercises the weightless dt/B/C stabilizer norms."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.mamba_lm.model import (
+        FalconMambaForCausalLM,
+        Mamba2ForCausalLM,
+        MambaForCausalLM,
+        MambaLMConfig,
+    )
+
+    torch.manual_seed(18)
+    if kind == "mamba2":
+        hf_cfg = transformers.Mamba2Config(
+            vocab_size=200, hidden_size=64, num_hidden_layers=2, state_size=8,
+            num_heads=8, head_dim=16, n_groups=1, expand=2, conv_kernel=4,
+            chunk_size=8, tie_word_embeddings=False)
+        hf = transformers.Mamba2ForCausalLM(hf_cfg).eval()
+        mine = Mamba2ForCausalLM(MambaLMConfig.from_hf_config(hf_cfg.to_dict(),
+                                                              "mamba2")).eval()
+    elif kind == "mamba":
+        hf_cfg = transformers.MambaConfig(
+            vocab_size=200, hidden_size=64, num_hidden_layers=2, state_size=8,
+            expand=2, conv_kernel=4, time_step_rank=8, tie_word_embeddings=False,
+            use_mambapy=False)
+        hf = transformers.MambaForCausalLM(hf_cfg).eval()
+        mine = MambaForCausalLM(MambaLMConfig.from_hf_config(hf_cfg.to_dict(),
+                                                             "mamba")).eval()
+    else:
+        hf_cfg = transformers.FalconMambaConfig(
+            vocab_size=200, hidden_size=64, num_hidden_layers=2, state_size=8,
+            expand=2, conv_kernel=4, time_step_rank=8, tie_word_embeddings=False)
+        hf = transformers.FalconMambaForCausalLM(hf_cfg).eval()
+        mine = FalconMambaForCausalLM(
+            MambaLMConfig.from_hf_config(hf_cfg.to_dict(), "falcon_mamba")).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
