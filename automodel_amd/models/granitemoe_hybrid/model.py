@@ -71,6 +71,10 @@ class GraniteMoeHybridConfig:
         layer_types = g("layers_block_type") or g("layer_types") or []
         layer_types = ["mamba" if t in ("mamba", "linear_attention") else "attention"
                        for t in layer_types]
+        arch = (g("architectures") or [""])[0]
+        if arch == "GraniteMoeSharedForCausalLM":
+            # all-attention granite with a shared MLP; rope always on
+            layer_types = ["attention"] * g("num_hidden_layers", 32)
         return cls(
             vocab_size=g("vocab_size", 32000),
             hidden_size=g("hidden_size", 1536),
@@ -95,7 +99,8 @@ class GraniteMoeHybridConfig:
             rms_norm_eps=g("rms_norm_eps", 1e-5),
             rope_theta=rp.get("rope_theta", g("rope_theta", 10000.0)),
             rope_scaling=g("rope_scaling"),
-            position_embedding_type=g("position_embedding_type") or "nope",
+            position_embedding_type=("rope" if arch == "GraniteMoeSharedForCausalLM"
+                                     else g("position_embedding_type") or "nope"),
             embedding_multiplier=g("embedding_multiplier", 1.0),
             attention_multiplier=g("attention_multiplier", 1.0),
             residual_multiplier=g("residual_multiplier", 1.0),
@@ -254,7 +259,8 @@ class GraniteMoeHybridStateDictAdapter:
 
 
 class GraniteMoeHybridForCausalLM(nn.Module):
-    hf_architectures = ("GraniteMoeHybridForCausalLM",)
+    hf_architectures = ("GraniteMoeHybridForCausalLM",
+                        "GraniteMoeSharedForCausalLM")
     config_class = GraniteMoeHybridConfig
     state_dict_adapter = GraniteMoeHybridStateDictAdapter
 
