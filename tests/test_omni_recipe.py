@@ -36,15 +36,19 @@ def test_omni_stage2_joint_losses_decrease():
     from automodel_amd.recipes.multimodal.finetune import FinetuneRecipeForMultimodal
 
     torch.manual_seed(0)
-    r = FinetuneRecipeForMultimodal(_cfg(stage=2, max_steps=4))
+    cfg = _cfg(stage=2, max_steps=10)
+    cfg["optimizer"] = ConfigNode({"lr": 3e-3, "warmup_steps": 1})
+    r = FinetuneRecipeForMultimodal(cfg)
     r.setup()
     r.run_train_validation_loop()
-    assert len(r.metrics) == 4
+    assert len(r.metrics) == 10
     for m in r.metrics:
         assert m["ce_tokens"] > 0 and m["mse_tokens"] > 0
         assert torch.isfinite(torch.tensor(m["ce"]))
         assert torch.isfinite(torch.tensor(m["mse"]))
-    assert r.metrics[-1]["ce"] < r.metrics[0]["ce"]
+    # average the flow-matching noise out: mean of last 3 vs first 3 steps
+    ce = [m["ce"] for m in r.metrics]
+    assert sum(ce[-3:]) / 3 < sum(ce[:3]) / 3, ce
 
 
 def test_omni_stage1_ce_only():
