@@ -154,6 +154,13 @@ class VLMForConditionalGeneration(nn.Module):
             elif isinstance(m, nn.LayerNorm):
                 nn.init.ones_(m.weight)
                 nn.init.zeros_(m.bias)
+            elif isinstance(m, nn.MultiheadAttention):
+                # raw in_proj parameters are NOT nn.Linear — without this
+                # they survive to_empty() as garbage (found via omni recipe
+                # NaNs; the out_proj IS a Linear and is covered above)
+                nn.init.normal_(m.in_proj_weight, std=0.02)
+                if m.in_proj_bias is not None:
+                    nn.init.zeros_(m.in_proj_bias)
         if not self.visual.pos_embed.is_meta:
             nn.init.normal_(self.visual.pos_embed, std=0.02)
         for m in self.projector.modules():
