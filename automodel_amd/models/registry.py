@@ -97,6 +97,12 @@ def _ensure_builtin() -> None:
         for cls2 in (MambaForCausalLM, Mamba2ForCausalLM, FalconMambaForCausalLM):
             for name in cls2.hf_architectures:
                 _REGISTRY[name] = cls2
+        from automodel_amd.models.recurrent_gemma.model import (
+            RecurrentGemmaForCausalLM,
+        )
+
+        for name in RecurrentGemmaForCausalLM.hf_architectures:
+            _REGISTRY[name] = RecurrentGemmaForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

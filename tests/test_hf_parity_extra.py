@@ -1859,3 +1859,33 @@ def test_hf_logits_parity_granitemoe_shared():
     ids = torch.randint(0, 200, (2, 17))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_recurrent_gemma():
+    """RecurrentGemma (Griffin): RG-LRU blocks (block-diagonal gates,
+    sqrt(1-a^2) normalization, chunked segsum scan), windowed partial-rotary
+    attention, biased GeGLU at intermediate//2, bf16 sqrt(H) normalizer,
+    tanh logits soft-cap."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.recurrent_gemma.model import (
+        RecurrentGemmaConfig,
+        RecurrentGemmaForCausalLM,
+    )
+
+    hf_cfg = transformers.RecurrentGemmaConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=128,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=1,
+        head_dim=16, lru_width=64, conv1d_width=3, attention_window_size=64,
+        block_types=("recurrent", "recurrent", "attention"),
+        max_position_embeddings=128, attn_implementation="eager",
+        tie_word_embeddings=False)
+    torch.manual_seed(20)
+    hf = transformers.RecurrentGemmaForCausalLM(hf_cfg).eval()
+    cfg = RecurrentGemmaConfig.from_hf_config(hf_cfg.to_dict())
+    mine = RecurrentGemmaForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 77))   # crosses the scan chunk boundary
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
