@@ -103,6 +103,10 @@ def _ensure_builtin() -> None:
 
         for name in RecurrentGemmaForCausalLM.hf_architectures:
             _REGISTRY[name] = RecurrentGemmaForCausalLM
+        from automodel_amd.models.phimoe.model import PhimoeForCausalLM
+
+        for name in PhimoeForCausalLM.hf_architectures:
+            _REGISTRY[name] = PhimoeForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

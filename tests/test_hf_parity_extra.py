@@ -1889,3 +1889,38 @@ def test_hf_logits_parity_recurrent_gemma():
     ids = torch.randint(0, 200, (2, 77))   # crosses the scan chunk boundary
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_phimoe():
+    """Phimoe (Phi-3.5-MoE): LayerNorm pre-norms + SparseMixer-v2 routing
+    (threshold-banded argmax top-2, eval path)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.phimoe.model import (
+        PhimoeConfig,
+        PhimoeForCausalLM,
+        PhimoeStateDictAdapter,
+    )
+
+    hf_cfg = transformers.PhimoeConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        num_local_experts=4, num_experts_per_tok=2, attention_bias=True,
+        lm_head_bias=True, max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(21)
+    hf = transformers.PhimoeForCausalLM(hf_cfg).eval()
+    cfg = PhimoeConfig.from_hf_config(hf_cfg.to_dict())
+    mine = PhimoeForCausalLM(cfg).eval()
+    sd = PhimoeStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+    # training path runs and produces finite sparse-mixer gradients
+    mine.train()
+    loss = mine(ids, labels=ids.clone())
+    loss.backward()
+    g = mine.model.layers[0].mlp.router.weight.grad
+    assert g is not None and torch.isfinite(g).all()
