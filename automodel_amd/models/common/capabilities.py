@@ -94,6 +94,9 @@ _CAPS = {
     "PhimoeForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_ep=True, flash_head_dims=()),
+    "GPTBigCodeForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

@@ -107,6 +107,10 @@ def _ensure_builtin() -> None:
 
         for name in PhimoeForCausalLM.hf_architectures:
             _REGISTRY[name] = PhimoeForCausalLM
+        from automodel_amd.models.gpt_bigcode.model import GPTBigCodeForCausalLM
+
+        for name in GPTBigCodeForCausalLM.hf_architectures:
+            _REGISTRY[name] = GPTBigCodeForCausalLM
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -1924,3 +1924,28 @@ def test_hf_logits_parity_phimoe():
     loss.backward()
     g = mine.model.layers[0].mlp.router.weight.grad
     assert g is not None and torch.isfinite(g).all()
+
+
+def test_hf_logits_parity_gpt_bigcode():
+    """GPTBigCode (StarCoder-1): fused MQA c_attn, learned positions,
+    tanh-GELU, biased LayerNorms."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.gpt_bigcode.model import (
+        GPTBigCodeConfig,
+        GPTBigCodeForCausalLM,
+    )
+
+    hf_cfg = transformers.GPTBigCodeConfig(
+        vocab_size=200, n_embd=64, n_layer=2, n_head=4, n_positions=64,
+        multi_query=True, bos_token_id=0, eos_token_id=0,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(22)
+    hf = transformers.GPTBigCodeForCausalLM(hf_cfg).eval()
+    cfg = GPTBigCodeConfig.from_hf_config(hf_cfg.to_dict())
+    mine = GPTBigCodeForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
