@@ -31,6 +31,8 @@ class CohereConfig:
     rope_theta: float = 10000.0
     layer_norm_eps: float = 1e-5
     logit_scale: float = 0.0625
+    sliding_window: int | None = None          # Cohere2 (Command-R7B)
+    layer_types: tuple = ()                    # "sliding_attention"/"full_attention"
     tie_word_embeddings: bool = True
     initializer_range: float = 0.02
 
@@ -55,6 +57,8 @@ class CohereConfig:
             rope_theta=rp.get("rope_theta", g("rope_theta", 10000.0)),
             layer_norm_eps=g("layer_norm_eps", 1e-5),
             logit_scale=g("logit_scale", 0.0625),
+            sliding_window=g("sliding_window"),
+            layer_types=tuple(g("layer_types") or ()),
             tie_word_embeddings=g("tie_word_embeddings", True),
         )
 
@@ -75,10 +79,14 @@ class CohereLayerNorm(nn.Module):
 
 
 class CohereDecoderLayer(nn.Module):
-    def __init__(self, cfg: CohereConfig):
+    def __init__(self, cfg: CohereConfig, layer_idx: int = 0):
         super().__init__()
         H, Hk, D = cfg.num_attention_heads, cfg.num_key_value_heads, cfg.head_dim
         self.H, self.Hk, self.D = H, Hk, D
+        # Cohere2: rope ONLY on sliding_attention layers (full layers NoPE)
+        kind = cfg.layer_types[layer_idx] if cfg.layer_types else "full_attention"
+        self.sliding = cfg.sliding_window if kind == "sliding_attention" else None
+        self.use_rope = not cfg.layer_types or kind == "sliding_attention"
         self.input_layernorm = CohereLayerNorm(cfg.hidden_size, cfg.layer_norm_eps)
         attn = nn.Module()
         attn.q_proj = nn.Linear(cfg.hidden_size, H * D, bias=False)
@@ -101,13 +109,22 @@ class CohereDecoderLayer(nn.Module):
         v = a.v_proj(h).view(B, S, self.Hk, self.D)
         # cohere's rope is interleaved-pair; de-interleave to half-split
         # order (scores invariant to a shared head-dim permutation)
-        d2 = self.D // 2
-        q = q.reshape(B, S, self.H, d2, 2).transpose(-1, -2).reshape(B, S, self.H, self.D)
-        k = k.reshape(B, S, self.Hk, d2, 2).transpose(-1, -2).reshape(B, S, self.Hk, self.D)
-        q, k = apply_rope_ref(q, k, cos, sin)
-        o = F.scaled_dot_product_attention(
-            q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
-            is_causal=True, enable_gqa=self.H != self.Hk)
+        if self.use_rope:
+            d2 = self.D // 2
+            q = q.reshape(B, S, self.H, d2, 2).transpose(-1, -2).reshape(B, S, self.H, self.D)
+            k = k.reshape(B, S, self.Hk, d2, 2).transpose(-1, -2).reshape(B, S, self.Hk, self.D)
+            q, k = apply_rope_ref(q, k, cos, sin)
+        if self.sliding is not None and S > self.sliding:
+            i = torch.arange(S, device=x.device)
+            allowed = (i[None, :] <= i[:, None]) & (i[None, :] > i[:, None] - self.sliding)
+            mask = torch.where(allowed, 0.0, float("-inf")).to(q.dtype)
+            o = F.scaled_dot_product_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                attn_mask=mask, enable_gqa=self.H != self.Hk)
+        else:
+            o = F.scaled_dot_product_attention(
+                q.transpose(1, 2), k.transpose(1, 2), v.transpose(1, 2),
+                is_causal=True, enable_gqa=self.H != self.Hk)
         attn_out = a.o_proj(o.transpose(1, 2).reshape(B, S, -1))
         mlp_out = self.mlp.down_proj(
             F.silu(self.mlp.gate_proj(h)) * self.mlp.up_proj(h))
@@ -129,8 +146,8 @@ class CohereForCausalLM(nn.Module):
         self.config = config
         inner = nn.Module()
         inner.embed_tokens = nn.Embedding(config.vocab_size, config.hidden_size)
-        inner.layers = nn.ModuleList(CohereDecoderLayer(config)
-                                     for _ in range(config.num_hidden_layers))
+        inner.layers = nn.ModuleList(CohereDecoderLayer(config, i)
+                                     for i in range(config.num_hidden_layers))
         inner.norm = CohereLayerNorm(config.hidden_size, config.layer_norm_eps)
         cos, sin = build_rope_cache(config.head_dim,
                                     config.max_position_embeddings,
@@ -180,3 +197,12 @@ class CohereForCausalLM(nn.Module):
                 nn.init.ones_(mod.weight)
         if self.config.tie_word_embeddings:
             self.lm_head.weight = self.model.embed_tokens.weight
+
+
+class Cohere2ForCausalLM(CohereForCausalLM):
+    """Cohere2 (Command-R7B): same parallel-residual block with sliding-
+    window layers (rope) interleaved with NoPE full-attention layers
+    (reference transformers.models.cohere2 — rope applied only when
+    layer_type == "sliding_attention")."""
+
+    hf_architectures = ("Cohere2ForCausalLM",)

@@ -97,6 +97,9 @@ _CAPS = {
     "GPTBigCodeForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         flash_head_dims=()),
+    "Cohere2ForCausalLM": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

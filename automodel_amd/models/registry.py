@@ -283,10 +283,15 @@ def _ensure_builtin() -> None:
     except ImportError:
         pass
     try:
-        from automodel_amd.models.cohere.model import CohereForCausalLM
+        from automodel_amd.models.cohere.model import (
+            Cohere2ForCausalLM,
+            CohereForCausalLM,
+        )
 
         for name in CohereForCausalLM.hf_architectures:
             _REGISTRY[name] = CohereForCausalLM
+        for name in Cohere2ForCausalLM.hf_architectures:
+            _REGISTRY[name] = Cohere2ForCausalLM
     except ImportError:
         pass
     try:

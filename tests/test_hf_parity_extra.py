@@ -1949,3 +1949,27 @@ def test_hf_logits_parity_gpt_bigcode():
     ids = torch.randint(0, 200, (2, 21))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_cohere2():
+    """Cohere2 (Command-R7B): sliding-window rope layers interleaved with
+    NoPE full-attention, parallel residual, logit scale."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.cohere.model import Cohere2ForCausalLM, CohereConfig
+
+    hf_cfg = transformers.Cohere2Config(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=5, num_attention_heads=4, num_key_value_heads=2,
+        sliding_window=8, max_position_embeddings=64, eos_token_id=0,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(23)
+    hf = transformers.Cohere2ForCausalLM(hf_cfg).eval()
+    cfg = CohereConfig.from_hf_config(hf_cfg.to_dict())
+    assert "full_attention" in cfg.layer_types
+    mine = Cohere2ForCausalLM(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))   # S=21 > window=8 exercises the band
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
