@@ -173,16 +173,22 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
 __global__ __launch_bounds__(512, 1) void grouped_gemm_nt_big_kernel(
     const bf16* __restrict__ x, const bf16* __restrict__ w, bf16* __restrict__ y,
     const int* __restrict__ tile_map, const int* __restrict__ offs,
-    const int* __restrict__ n_tiles, int K, int N) {
-  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+    const int* __restrict__ n_tiles, int K, int N, int nfirst) {
+  // nfirst: interpret blockIdx.x as the N tile (consecutive blocks then
+  // share one (expert, m0) row tile — XCD round-robin dispatch puts the
+  // sibling n-tiles of one x tile on different XCDs; A/B via
+  // AMD_OPS_GG_NFIRST, benchmarks/gg_micro.py)
+  const int tb = nfirst ? (int)blockIdx.y : (int)blockIdx.x;
+  const int nb = nfirst ? (int)blockIdx.x : (int)blockIdx.y;
+  if (n_tiles != nullptr && tb >= n_tiles[0]) return;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xa = smem;                   // [256 m][64 k]  32 KiB
   char* wb = smem + 256 * 64 * 2;    // [256 n][64 k]  32 KiB
 
-  const int e = tile_map[2 * blockIdx.x];
-  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int e = tile_map[2 * tb];
+  const int m0 = tile_map[2 * tb + 1];
   const int m_end = offs[e + 1];
-  const int n0 = blockIdx.y * 256;
+  const int n0 = nb * 256;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -297,14 +303,19 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
   const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
   if (bm == 256) {
     TORCH_CHECK(N % 256 == 0, "big-tile nt needs N%256==0");
-    const dim3 gridb(n_mtiles, N / 256);
+    static const int nfirst = []{
+      const char* v = getenv("AMD_OPS_GG_NFIRST");
+      return v ? atoi(v) : 0;
+    }();
+    const dim3 gridb = nfirst ? dim3(N / 256, n_mtiles) : dim3(n_mtiles, N / 256);
     const size_t smemb = 2 * 256 * 64 * 2;
     hipLaunchKernelGGL(grouped_gemm_nt_big_kernel, gridb, dim3(512), smemb,
                        stream.stream(),
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<const bf16*>(w.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
-                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
+                       nfirst);
     HIP_CHECK_KERNEL();
     return y;
   }
