@@ -75,16 +75,18 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_kernel(
     const int* __restrict__ tile_map,   // [n_mtiles][2]: (expert, row0)
     const int* __restrict__ offs,       // [E+1] group row offsets
     const int* __restrict__ n_tiles,    // device tile count (null = grid-sized)
-    int K, int N) {
-  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+    int K, int N, int nfirst) {
+  const int tb = nfirst ? (int)blockIdx.y : (int)blockIdx.x;
+  const int nb = nfirst ? (int)blockIdx.x : (int)blockIdx.y;
+  if (n_tiles != nullptr && tb >= n_tiles[0]) return;
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xa = smem;                  // [128][BK] bf16
   char* wb = smem + GG_BM * BK * 2;
 
-  const int e = tile_map[2 * blockIdx.x];
-  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int e = tile_map[2 * tb];
+  const int m0 = tile_map[2 * tb + 1];
   const int m_end = offs[e + 1];
-  const int n0 = blockIdx.y * GG_BN;
+  const int n0 = nb * GG_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -284,6 +286,15 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> build_group_plan(
   return {offs, tile_map, n_tiles};
 }
 
+
+static int gg_nfirst() {
+  static const int v = []{
+    const char* e = getenv("AMD_OPS_GG_NFIRST");
+    return e ? atoi(e) : 1;
+  }();
+  return v;
+}
+
 at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                            const at::Tensor& offs, const at::Tensor& tile_map,
                            const std::optional<at::Tensor>& n_tiles, int64_t bm) {
@@ -303,10 +314,10 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
   const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
   if (bm == 256) {
     TORCH_CHECK(N % 256 == 0, "big-tile nt needs N%256==0");
-    static const int nfirst = []{
-      const char* v = getenv("AMD_OPS_GG_NFIRST");
-      return v ? atoi(v) : 0;
-    }();
+    // n-first default: consecutive blocks share one (expert, m0) x-tile
+    // and their sibling n-tiles round-robin across XCDs — measured +6-9%
+    // (benchmarks/gg_grid_ab.py: 478->519 / 505->534 / 503->546 TF/s)
+    const int nfirst = gg_nfirst();
     const dim3 gridb = nfirst ? dim3(N / 256, n_mtiles) : dim3(n_mtiles, N / 256);
     const size_t smemb = 2 * 256 * 64 * 2;
     hipLaunchKernelGGL(grouped_gemm_nt_big_kernel, gridb, dim3(512), smemb,
@@ -319,7 +330,8 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
     HIP_CHECK_KERNEL();
     return y;
   }
-  const dim3 grid(n_mtiles, N / GG_BN);
+  const int nfirst = gg_nfirst();
+  const dim3 grid = nfirst ? dim3(N / GG_BN, n_mtiles) : dim3(n_mtiles, N / GG_BN);
   static const int bk_env = []{
     const char* v = getenv("AMD_OPS_GG_BK");
     return v ? atoi(v) : 64;
@@ -331,7 +343,8 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<const bf16*>(w.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
-                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
+                       nfirst);
   } else {
     const size_t smem = 2 * GG_BM * 64 * 2;
     hipLaunchKernelGGL((grouped_gemm_nt_kernel<64>), grid, dim3(256), smem,
@@ -339,7 +352,8 @@ at::Tensor grouped_gemm_nt(const at::Tensor& x, const at::Tensor& w,
                        reinterpret_cast<const bf16*>(x.data_ptr()),
                        reinterpret_cast<const bf16*>(w.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
-                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N);
+                       tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
+                       nfirst);
   }
   HIP_CHECK_KERNEL();
   return y;
@@ -740,17 +754,19 @@ __global__ __launch_bounds__(256) void grouped_gemm_nt_fp8_kernel(
     const unsigned char* __restrict__ x, const unsigned char* __restrict__ w,
     bf16* __restrict__ y, const int* __restrict__ tile_map,
     const int* __restrict__ offs, const int* __restrict__ n_tiles,
-    int K, int N, const float* __restrict__ scale_p) {
-  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+    int K, int N, const float* __restrict__ scale_p, int nfirst) {
+  const int tb = nfirst ? (int)blockIdx.y : (int)blockIdx.x;
+  const int nb = nfirst ? (int)blockIdx.x : (int)blockIdx.y;
+  if (n_tiles != nullptr && tb >= n_tiles[0]) return;
   const float scale = scale_p[0];
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xa = smem;                  // [128 m][128 k-bytes] 16 KiB
   char* wb = smem + GG_BM * 128;
 
-  const int e = tile_map[2 * blockIdx.x];
-  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int e = tile_map[2 * tb];
+  const int m0 = tile_map[2 * tb + 1];
   const int m_end = offs[e + 1];
-  const int n0 = blockIdx.y * GG_BN;
+  const int n0 = nb * GG_BN;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -832,17 +848,19 @@ __global__ __launch_bounds__(512, 1) void grouped_gemm_nt_fp8_big_kernel(
     const unsigned char* __restrict__ x, const unsigned char* __restrict__ w,
     bf16* __restrict__ y, const int* __restrict__ tile_map,
     const int* __restrict__ offs, const int* __restrict__ n_tiles,
-    int K, int N, const float* __restrict__ scale_p) {
-  if (n_tiles != nullptr && (int)blockIdx.x >= n_tiles[0]) return;
+    int K, int N, const float* __restrict__ scale_p, int nfirst) {
+  const int tb = nfirst ? (int)blockIdx.y : (int)blockIdx.x;
+  const int nb = nfirst ? (int)blockIdx.x : (int)blockIdx.y;
+  if (n_tiles != nullptr && tb >= n_tiles[0]) return;
   const float scale = scale_p[0];
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* xa = smem;                    // [256 m][128 k-bytes] 32 KiB
   char* wb = smem + 256 * 128;
 
-  const int e = tile_map[2 * blockIdx.x];
-  const int m0 = tile_map[2 * blockIdx.x + 1];
+  const int e = tile_map[2 * tb];
+  const int m0 = tile_map[2 * tb + 1];
   const int m_end = offs[e + 1];
-  const int n0 = blockIdx.y * 256;
+  const int n0 = nb * 256;
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -942,25 +960,27 @@ at::Tensor grouped_gemm_nt_fp8(const at::Tensor& x8, const at::Tensor& w8,
   const int* ntp = n_tiles.has_value() ? n_tiles->data_ptr<int>() : nullptr;
   if (bm == 256) {
     TORCH_CHECK(N % 256 == 0, "big-tile fp8 nt needs N%256==0");
-    const dim3 gridb(n_mtiles, N / 256);
+    const int nfirstb = gg_nfirst();
+    const dim3 gridb = nfirstb ? dim3(N / 256, n_mtiles) : dim3(n_mtiles, N / 256);
     hipLaunchKernelGGL(grouped_gemm_nt_fp8_big_kernel, gridb, dim3(512),
                        2 * 256 * 128, stream.stream(),
                        reinterpret_cast<const unsigned char*>(x8.data_ptr()),
                        reinterpret_cast<const unsigned char*>(w8.data_ptr()),
                        reinterpret_cast<bf16*>(y.data_ptr()),
                        tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
-                       scale.data_ptr<float>());
+                       scale.data_ptr<float>(), nfirstb);
     HIP_CHECK_KERNEL();
     return y;
   }
-  const dim3 grid(n_mtiles, N / GG_BN);
+  const int nfirst = gg_nfirst();
+  const dim3 grid = nfirst ? dim3(N / GG_BN, n_mtiles) : dim3(n_mtiles, N / GG_BN);
   hipLaunchKernelGGL(grouped_gemm_nt_fp8_kernel, grid, dim3(256),
                      2 * GG_BM * 128, stream.stream(),
                      reinterpret_cast<const unsigned char*>(x8.data_ptr()),
                        reinterpret_cast<const unsigned char*>(w8.data_ptr()),
                      reinterpret_cast<bf16*>(y.data_ptr()),
                      tile_map.data_ptr<int>(), offs.data_ptr<int>(), ntp, K, N,
-                     scale.data_ptr<float>());
+                     scale.data_ptr<float>(), nfirst);
   HIP_CHECK_KERNEL();
   return y;
 }
