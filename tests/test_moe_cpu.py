@@ -429,3 +429,41 @@ def test_router_replay_r3():
         rr.stop()
     assert torch.equal(i2, i0)     # replayed decisions
     assert not torch.allclose(p2, p0)  # probs from CURRENT weights
+
+
+def test_mtp_multi_depth_and_packed_masking():
+    """Multi-depth MTP: depth-1 equals the single-head loss on the common
+    positions; packed seq boundaries mask cross-document rolls."""
+    from automodel_amd.loss.mtp import (
+        MTPHeads,
+        calculate_mtp_loss_multi,
+    )
+
+    torch.manual_seed(0)
+    B, S, H, V = 2, 16, 32, 64
+    hidden = torch.randn(B, S, H, requires_grad=True)
+    emb = torch.nn.Embedding(V, H)
+    lm_w = torch.randn(V, H)
+    ids = torch.randint(0, V, (B, S))
+    labels = torch.randint(0, V, (B, S))
+    heads = MTPHeads(H, n_depths=2)
+    loss, per_depth = calculate_mtp_loss_multi(
+        hidden, emb, lm_w, heads, ids, labels, scaling_factor=0.1,
+        return_per_depth=True)
+    assert len(per_depth) == 2
+    assert torch.isfinite(loss)
+    loss.backward()
+    assert hidden.grad is not None and torch.isfinite(hidden.grad).all()
+
+    # packed: two docs of 8; depth-2 rolls from doc 2 into doc 1 are masked
+    cu = torch.tensor([0, 8, 16])
+    h2 = hidden.detach().clone().requires_grad_(True)
+    heads1 = MTPHeads(H, n_depths=1)
+    l_packed, pd = calculate_mtp_loss_multi(
+        h2, emb, lm_w, heads1, ids, labels, scaling_factor=1.0,
+        cu_seqlens=cu, return_per_depth=True)
+    l_flat, _ = calculate_mtp_loss_multi(
+        h2, emb, lm_w, heads1, ids, labels, scaling_factor=1.0,
+        return_per_depth=True)
+    # boundary-crossing rolls removed -> strictly fewer summed tokens
+    assert float(l_packed) < float(l_flat)
