@@ -18,7 +18,8 @@ import threading
 import torch
 
 
-def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False):
+def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False,
+              proposer=None):
     from fastapi import FastAPI
     from pydantic import BaseModel
 
@@ -39,6 +40,8 @@ def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False)
         prompt_ids: list[int] | None = None
         max_new_tokens: int = 64
         temperature: float = 0.0
+        speculative: bool = False      # greedy draft-verify (needs a proposer
+                                       # or falls back to the ngram proposer)
 
     @app.get("/health")
     def health():
@@ -58,6 +61,22 @@ def build_app(model, tokenizer, use_cache: bool = True, use_graph: bool = False)
         dev = next(model.parameters()).device
         x = torch.tensor([ids], dtype=torch.long, device=dev)
         with gen_lock, torch.no_grad():
+            if req.speculative and req.temperature == 0.0:
+                from automodel_amd.speculative.decode import (
+                    NgramProposer,
+                    speculative_generate,
+                )
+
+                prop = proposer if proposer is not None else NgramProposer()
+                out, stats = speculative_generate(model, prop, x,
+                                                  req.max_new_tokens)
+                new = out[0, len(ids):].tolist()
+                resp = {"prompt_len": len(ids), "output_ids": new,
+                        "spec_acceptance_rate": stats.acceptance_rate,
+                        "spec_tokens_per_call": stats.tokens_per_target_call}
+                if tokenizer is not None:
+                    resp["text"] = tokenizer.decode(new)
+                return resp
             if use_graph and req.temperature == 0.0:
                 out = generate_graphed(model, x, req.max_new_tokens)
             elif use_cache:
@@ -88,6 +107,9 @@ def main(argv=None):
     ap.add_argument("--no-cache", action="store_true")
     ap.add_argument("--graph", action="store_true",
                     help="hipGraph-captured decode (greedy only)")
+    ap.add_argument("--draft", default=None,
+                    help="EAGLE draft checkpoint dir: serve speculative "
+                         "decoding (reference serve_target.py role)")
     args = ap.parse_args(argv)
 
     cfg = load_yaml_config(args.config)
@@ -105,8 +127,14 @@ def main(argv=None):
         from automodel_amd.models.auto_tokenizer import build_tokenizer
 
         tokenizer = build_tokenizer(tok_path)
+    proposer = None
+    if args.draft:
+        from automodel_amd.speculative.decode import EagleProposer
+        from automodel_amd.speculative.draft import load_draft
+
+        proposer = EagleProposer(load_draft(args.draft, model, device), model)
     app = build_app(model, tokenizer, use_cache=not args.no_cache,
-                    use_graph=args.graph)
+                    use_graph=args.graph, proposer=proposer)
     uvicorn.run(app, host=args.host, port=args.port)
 
 

@@ -121,3 +121,26 @@ class EagleDraftModel(nn.Module):
 
     def forward(self, input_ids: torch.Tensor, carry: torch.Tensor) -> torch.Tensor:
         return self.lm_head(self.backbone(input_ids, carry))
+
+
+def load_draft(path: str, target, device="cpu") -> "EagleDraftModel":
+    """Load a trained EAGLE draft from a checkpoint dir (draft.pt +
+    draft_config.json written by TrainEagleDraftRecipe / torch.save)."""
+    import json
+    import os
+
+    import torch
+
+    cfg_path = os.path.join(path, "draft_config.json")
+    if os.path.exists(cfg_path):
+        with open(cfg_path) as f:
+            cfg = EagleDraftConfig(**json.load(f))
+    else:
+        cfg = EagleDraftConfig.from_target(target.config)
+    draft = EagleDraftModel(cfg)
+    sd_path = os.path.join(path, "draft.pt")
+    if os.path.exists(sd_path):
+        draft.load_state_dict(torch.load(sd_path, map_location="cpu",
+                                         weights_only=True), strict=False)
+    draft.tie_to_target(target)
+    return draft.to(device)

@@ -113,6 +113,17 @@ def test_generation_server():
     ref = generate(model, torch.tensor([prompt]), max_new_tokens=8)
     assert out == ref[0, 4:].tolist()   # server(KV-cached) == plain greedy
 
+    # speculative endpoint: greedy-equivalence guarantee holds over the
+    # ngram fallback proposer, and acceptance stats are reported
+    r = client.post("/generate", json={"prompt_ids": prompt,
+                                       "max_new_tokens": 8,
+                                       "speculative": True})
+    assert r.status_code == 200
+    body = r.json()
+    assert body["output_ids"] == ref[0, 4:].tolist()
+    assert 0.0 <= body["spec_acceptance_rate"] <= 1.0
+    assert body["spec_tokens_per_call"] >= 1.0
+
 
 def _sigterm_worker(rank, world):
     """SIGTERM on ONE rank propagates to all via the MAX all-reduce."""
