@@ -67,6 +67,21 @@ class MoEModelConfig(LlamaConfig):
                 topk_then_softmax=True,
             )
             flavor = "qwen3_moe"   # stacked-key layout (+router rename)
+        elif "MiniMaxM2" in arch:
+            # sigmoid scores + aux-free correction bias (selection biased,
+            # gathered weights unbiased, top-k renormalized) + FULL-width
+            # q/k projection norms — rides the OLMoE attention variant
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_local_experts", 256),
+                n_activated_experts=hf.get("num_experts_per_tok", 8),
+                moe_intermediate_size=hf.get("intermediate_size", 1536),
+                score_func="sigmoid",
+                expert_bias=True,
+                norm_topk_prob=True,
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+            )
+            base["qk_norm_full"] = True
+            flavor = "qwen3_moe"
         elif "Qwen2Moe" in arch:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 60),
@@ -131,7 +146,7 @@ class MoEDecoderLayer(nn.Module):
 class MoEForCausalLM(nn.Module):
     hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
                         "OlmoeForCausalLM", "GraniteMoeForCausalLM",
-                        "Ernie4_5_MoeForCausalLM")
+                        "Ernie4_5_MoeForCausalLM", "MiniMaxM2ForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod

@@ -45,6 +45,10 @@ class MoEStateDictAdapter:
         if key.endswith(".gate.moe_statics.e_score_correction_bias"):
             return [key.replace(".gate.moe_statics.e_score_correction_bias",
                                 ".gate.e_score_correction_bias")]
+        if key.endswith(".block_sparse_moe.e_score_correction_bias") or \
+                key.endswith(".mlp.e_score_correction_bias"):
+            # MiniMax-M2 keeps the aux-free bias at the MoE-block level
+            return [key.rsplit(".", 1)[0] + ".gate.e_score_correction_bias"]
         if ".mlp.shared_expert." in key:
             return [key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")]
         if ".block_sparse_moe.router." in key:
@@ -79,6 +83,11 @@ class MoEStateDictAdapter:
                 # ernie: [1, E] fp32 selection bias -> flat buffer
                 out[key.replace(".gate.moe_statics.e_score_correction_bias",
                                 ".gate.e_score_correction_bias")] = t.reshape(-1)
+            elif (key.endswith(".mlp.e_score_correction_bias")
+                  or key.endswith(".block_sparse_moe.e_score_correction_bias")):
+                # minimax-m2: block-level aux-free bias -> gate buffer
+                out[key.rsplit(".", 1)[0].replace(".block_sparse_moe", ".mlp")
+                    + ".gate.e_score_correction_bias"] = t.reshape(-1)
             elif ".mlp.shared_expert." in key:
                 out[key.replace(".mlp.shared_expert.", ".mlp.shared_experts.")] = t
             elif ".block_sparse_moe.router." in key:   # granite-moe

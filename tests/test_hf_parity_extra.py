@@ -1973,3 +1973,36 @@ def test_hf_logits_parity_cohere2():
     ids = torch.randint(0, 200, (2, 21))   # S=21 > window=8 exercises the band
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_minimax_m2():
+    """MiniMax-M2: sigmoid routing + aux-free correction bias + top-k
+    renorm + full-width q/k norms (generic MoE model flavor)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+    from automodel_amd.moe.state_dict_adapter import MoEStateDictAdapter
+
+    hf_cfg = transformers.MiniMaxM2Config(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=2, num_attention_heads=4, num_key_value_heads=2,
+        head_dim=16, num_local_experts=4, num_experts_per_tok=2,
+        bos_token_id=0, eos_token_id=0, max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(24)
+    hf = transformers.MiniMaxM2ForCausalLM(hf_cfg).eval()
+    # make the correction bias non-trivial so selection-vs-weight asymmetry
+    # is exercised
+    with torch.no_grad():
+        for lyr in hf.model.layers:
+            lyr.mlp.e_score_correction_bias.uniform_(-0.5, 0.5)
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["MiniMaxM2ForCausalLM"]
+    cfg = MoEModelConfig.from_hf_config(d)
+    assert cfg.moe.score_func == "sigmoid" and cfg.moe.expert_bias
+    mine = MoEForCausalLM(cfg).eval()
+    sd = MoEStateDictAdapter(cfg).from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    ids = torch.randint(0, 200, (2, 17))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
