@@ -483,9 +483,12 @@ __global__ __launch_bounds__(256) void grouped_gemm_tn_kernel(
   char* ga = smem;                       // [128 n][64 m]
   char* xb = smem + GG_BM * GG_BK * 2;   // [128 k][64 m]
 
-  const int e = blockIdx.x;
+  // k-first grid (z = expert): consecutive blocks share the expert's g/x
+  // row range, so its m-stream stays hot in the XCD L2s (same reasoning as
+  // the NT n-first order; A/B via AMD_OPS_GG_TN_EFIRST=1 for the old order)
+  const int e = blockIdx.z;
   const int n0 = blockIdx.y * GG_BN;
-  const int k0 = blockIdx.z * GG_BN;
+  const int k0 = blockIdx.x * GG_BN;
   const int m_start = offs[e], m_end = offs[e + 1];
 
   const int tid = threadIdx.x;
@@ -602,7 +605,7 @@ at::Tensor grouped_gemm_tn(const at::Tensor& gt, const at::Tensor& xt,
   TORCH_CHECK(xt.size(1) == M, "M mismatch");
   TORCH_CHECK(N % GG_BN == 0 && K % GG_BN == 0, "need N%128==0, K%128==0");
   auto dw = at::empty({E, (long)N, (long)K}, gt.options());
-  const dim3 grid((unsigned)E, N / GG_BN, K / GG_BN);
+  const dim3 grid(K / GG_BN, N / GG_BN, (unsigned)E);
   const size_t smem = 2 * GG_BM * GG_BK * 2;
   auto stream = c10::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(grouped_gemm_tn_kernel, grid, dim3(256), smem, stream.stream(),
