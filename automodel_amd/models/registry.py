@@ -111,6 +111,12 @@ def _ensure_builtin() -> None:
 
         for name in GPTBigCodeForCausalLM.hf_architectures:
             _REGISTRY[name] = GPTBigCodeForCausalLM
+        from automodel_amd.models.qwen3_vl.model import (
+            Qwen3VLForConditionalGeneration,
+        )
+
+        for name in Qwen3VLForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen3VLForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

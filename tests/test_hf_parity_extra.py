@@ -2006,3 +2006,53 @@ def test_hf_logits_parity_minimax_m2():
     ids = torch.randint(0, 200, (2, 17))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_qwen3_vl():
+    """Qwen3-VL: DeepStack ViT (learned pos-table bilinear resample in
+    merge order, 2-axis rotary, per-image full attention, postshuffle-norm
+    deepstack mergers into early LLM layers) + interleaved 3D MRoPE text.
+    Text-only AND image parity."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.qwen3_vl.model import (
+        Qwen3VLConfig,
+        Qwen3VLForConditionalGeneration,
+    )
+
+    tcfg = dict(vocab_size=200, hidden_size=64, intermediate_size=96,
+                num_hidden_layers=3, num_attention_heads=4,
+                num_key_value_heads=2, head_dim=16,
+                rope_scaling={"rope_type": "default", "mrope_section": [4, 2, 2]},
+                max_position_embeddings=128)
+    vcfg = dict(depth=2, hidden_size=32, intermediate_size=64, num_heads=2,
+                patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                out_hidden_size=64, num_position_embeddings=36,
+                deepstack_visual_indexes=[0, 1])
+    hf_cfg = transformers.Qwen3VLConfig(
+        text_config=tcfg, vision_config=vcfg, image_token_id=3,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(25)
+    hf = transformers.Qwen3VLForConditionalGeneration(hf_cfg).eval()
+    cfg = Qwen3VLConfig.from_hf_config(hf_cfg.to_dict())
+    mine = Qwen3VLForConditionalGeneration(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+
+    # text-only parity
+    ids = torch.randint(5, 200, (2, 15))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=8e-4, rtol=8e-4)
+
+    # one 8x8 image -> grid (1, 2, 2) after patch 4: 4 patches -> 1 merged tok
+    grid = torch.tensor([[1, 4, 4]])   # 16 patches -> 4 merged tokens
+    n_patches = 16
+    pixels = torch.randn(n_patches, 3 * 1 * 4 * 4)
+    ids = torch.randint(5, 200, (1, 18))
+    ids[0, 6:10] = 3                    # 4 merged-image slots
+    mm_type = (ids == 3).to(torch.int32)     # text 0, image 1
+    with torch.no_grad():
+        ref = hf(ids, pixel_values=pixels, image_grid_thw=grid,
+                 mm_token_type_ids=mm_type).logits
+        out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
