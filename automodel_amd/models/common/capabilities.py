@@ -103,6 +103,9 @@ _CAPS = {
     "Qwen3VLForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_packed_sequences=False, flash_head_dims=()),
+    "Qwen3VLMoeForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, supports_ep=True, flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

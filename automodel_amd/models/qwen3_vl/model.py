@@ -62,6 +62,12 @@ class Qwen3VLTextConfig:
     attention_bias: bool = False
     max_position_embeddings: int = 128000
     tie_word_embeddings: bool = False
+    # MoE variant (Qwen3VLMoe): softmax-topk-renorm routing, fused experts
+    num_experts: int = 0
+    num_experts_per_tok: int = 8
+    moe_intermediate_size: int = 768
+    decoder_sparse_step: int = 1
+    mlp_only_layers: tuple = ()
 
 
 @dataclass
@@ -98,6 +104,11 @@ class Qwen3VLConfig:
             max_position_embeddings=t.get("max_position_embeddings", 128000),
             tie_word_embeddings=hf.get("tie_word_embeddings",
                                        t.get("tie_word_embeddings", False)),
+            num_experts=t.get("num_experts") or t.get("num_local_experts") or 0,
+            num_experts_per_tok=t.get("num_experts_per_tok", 8),
+            moe_intermediate_size=t.get("moe_intermediate_size", 768),
+            decoder_sparse_step=t.get("decoder_sparse_step", 1),
+            mlp_only_layers=tuple(t.get("mlp_only_layers") or ()),
         )
         vision = Qwen3VLVisionConfig(
             depth=v.get("depth", 27),
@@ -289,14 +300,29 @@ class Qwen3VLTextAttention(nn.Module):
 
 
 class Qwen3VLTextLayer(nn.Module):
-    def __init__(self, cfg: Qwen3VLTextConfig, backend: BackendConfig):
+    def __init__(self, cfg: Qwen3VLTextConfig, backend: BackendConfig,
+                 layer_idx: int = 0):
         super().__init__()
         self.self_attn = Qwen3VLTextAttention(cfg, backend)
-        mlp = nn.Module()
-        mlp.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        mlp.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
-        mlp.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
-        self.mlp = mlp
+        sparse = (cfg.num_experts > 0
+                  and layer_idx not in (cfg.mlp_only_layers or ())
+                  and (layer_idx + 1) % cfg.decoder_sparse_step == 0)
+        self.is_moe = sparse
+        if sparse:
+            from automodel_amd.moe.config import MoEConfig
+            from automodel_amd.moe.layers import MoE
+
+            self.mlp = MoE(cfg.hidden_size, MoEConfig(
+                n_routed_experts=cfg.num_experts,
+                n_activated_experts=cfg.num_experts_per_tok,
+                moe_intermediate_size=cfg.moe_intermediate_size,
+                norm_topk_prob=True))
+        else:
+            mlp = nn.Module()
+            mlp.gate_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+            mlp.up_proj = nn.Linear(cfg.hidden_size, cfg.intermediate_size, bias=False)
+            mlp.down_proj = nn.Linear(cfg.intermediate_size, cfg.hidden_size, bias=False)
+            self.mlp = mlp
         self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps,
                                                 backend.rms_norm)
@@ -304,6 +330,8 @@ class Qwen3VLTextLayer(nn.Module):
     def forward(self, x, cos, sin):
         x = x + self.self_attn(self.input_layernorm(x), cos, sin)
         h = self.post_attention_layernorm(x)
+        if self.is_moe:
+            return x + self.mlp(h)
         return x + self.mlp.down_proj(swiglu(self.mlp.gate_proj(h), self.mlp.up_proj(h)))
 
 
@@ -313,7 +341,7 @@ class Qwen3VLTextModel(nn.Module):
         self.cfg = cfg
         self.embed_tokens = nn.Embedding(cfg.vocab_size, cfg.hidden_size)
         self.layers = nn.ModuleList(
-            Qwen3VLTextLayer(cfg, backend) for _ in range(cfg.num_hidden_layers))
+            Qwen3VLTextLayer(cfg, backend, i) for i in range(cfg.num_hidden_layers))
         self.norm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
         D = cfg.head_dim
         inv = 1.0 / (cfg.rope_theta ** (torch.arange(0, D, 2).float() / D))
@@ -462,3 +490,37 @@ class Qwen3VLForConditionalGeneration(nn.Module):
 
     def num_parameters(self) -> int:
         return sum(p.numel() for p in self.parameters())
+
+class Qwen3VLMoeStateDictAdapter:
+    """HF fused experts.gate_up_proj [E,2I,H] <-> stacked gate/up [E,I,H]."""
+
+    def from_hf(self, sd: dict) -> dict:
+        out = {}
+        for k, v in sd.items():
+            if k.endswith("mlp.experts.gate_up_proj"):
+                gate, up = v.chunk(2, dim=1)
+                out[k.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
+                out[k.replace("gate_up_proj", "up_proj")] = up.contiguous()
+            else:
+                out[k] = v
+        return out
+
+    def to_hf(self, sd: dict) -> dict:
+        out = {}
+        for k, v in sd.items():
+            if k.endswith("mlp.experts.gate_proj"):
+                up = sd[k.replace("gate_proj", "up_proj")]
+                out[k.replace("gate_proj", "gate_up_proj")] = torch.cat([v, up], dim=1)
+            elif k.endswith("mlp.experts.up_proj"):
+                continue
+            else:
+                out[k] = v
+        return out
+
+
+class Qwen3VLMoeForConditionalGeneration(Qwen3VLForConditionalGeneration):
+    """Qwen3-VL-MoE: same DeepStack vision + MRoPE text with softmax-topk-
+    renorm MoE FFNs (reference transformers.models.qwen3_vl_moe)."""
+
+    hf_architectures = ("Qwen3VLMoeForConditionalGeneration",)
+    state_dict_adapter = Qwen3VLMoeStateDictAdapter

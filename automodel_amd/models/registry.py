@@ -113,10 +113,13 @@ def _ensure_builtin() -> None:
             _REGISTRY[name] = GPTBigCodeForCausalLM
         from automodel_amd.models.qwen3_vl.model import (
             Qwen3VLForConditionalGeneration,
+            Qwen3VLMoeForConditionalGeneration,
         )
 
         for name in Qwen3VLForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen3VLForConditionalGeneration
+        for name in Qwen3VLMoeForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen3VLMoeForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -2056,3 +2056,47 @@ def test_hf_logits_parity_qwen3_vl():
                  mm_token_type_ids=mm_type).logits
         out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_qwen3_vl_moe():
+    """Qwen3-VL-MoE: DeepStack vision + MoE text FFNs (softmax-topk renorm,
+    fused experts split by the adapter)."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.qwen3_vl.model import (
+        Qwen3VLConfig,
+        Qwen3VLMoeForConditionalGeneration,
+        Qwen3VLMoeStateDictAdapter,
+    )
+
+    tcfg = dict(vocab_size=200, hidden_size=64, intermediate_size=96,
+                num_hidden_layers=2, num_attention_heads=4,
+                num_key_value_heads=2, head_dim=16,
+                num_experts=4, num_experts_per_tok=2, moe_intermediate_size=32,
+                rope_scaling={"rope_type": "default", "mrope_section": [4, 2, 2]},
+                max_position_embeddings=128)
+    vcfg = dict(depth=2, hidden_size=32, intermediate_size=64, num_heads=2,
+                patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                out_hidden_size=64, num_position_embeddings=36,
+                deepstack_visual_indexes=[0, 1])
+    hf_cfg = transformers.Qwen3VLMoeConfig(
+        text_config=tcfg, vision_config=vcfg, image_token_id=3,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(26)
+    hf = transformers.Qwen3VLMoeForConditionalGeneration(hf_cfg).eval()
+    cfg = Qwen3VLConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.text.num_experts == 4
+    mine = Qwen3VLMoeForConditionalGeneration(cfg).eval()
+    sd = Qwen3VLMoeStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+    grid = torch.tensor([[1, 4, 4]])
+    pixels = torch.randn(16, 3 * 1 * 4 * 4)
+    ids = torch.randint(5, 200, (1, 18))
+    ids[0, 6:10] = 3
+    mm_type = (ids == 3).to(torch.int32)
+    with torch.no_grad():
+        ref = hf(ids, pixel_values=pixels, image_grid_thw=grid,
+                 mm_token_type_ids=mm_type).logits
+        out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
