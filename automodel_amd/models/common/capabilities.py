@@ -106,6 +106,9 @@ _CAPS = {
     "Qwen3VLMoeForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_packed_sequences=False, supports_ep=True, flash_head_dims=()),
+    "Glm4vForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

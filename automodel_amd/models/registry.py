@@ -120,6 +120,10 @@ def _ensure_builtin() -> None:
             _REGISTRY[name] = Qwen3VLForConditionalGeneration
         for name in Qwen3VLMoeForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen3VLMoeForConditionalGeneration
+        from automodel_amd.models.glm4v.model import Glm4vForConditionalGeneration
+
+        for name in Glm4vForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Glm4vForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

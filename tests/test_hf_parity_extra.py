@@ -2100,3 +2100,50 @@ def test_hf_logits_parity_qwen3_vl_moe():
                  mm_token_type_ids=mm_type).logits
         out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_glm4v():
+    """GLM-4V: EVA-style ViT (grid_sample bicubic pos resample, RMS norms,
+    SwiGLU vision MLP, Conv2d merge downsample, GELU/SwiGLU merger) + GLM-4
+    text (sandwich norms, interleaved partial rope) on chunked 3D MRoPE."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.glm4v.model import (
+        Glm4vConfig,
+        Glm4vForConditionalGeneration,
+    )
+
+    tcfg = dict(vocab_size=200, hidden_size=64, intermediate_size=96,
+                num_hidden_layers=2, num_attention_heads=4,
+                num_key_value_heads=2,
+                rope_scaling={"rope_type": "default",
+                              "partial_rotary_factor": 0.5,
+                              "mrope_section": [2, 1, 1]},
+                max_position_embeddings=128)
+    vcfg = dict(depth=2, hidden_size=32, intermediate_size=64, num_heads=2,
+                patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                out_hidden_size=64, image_size=16)
+    hf_cfg = transformers.Glm4vConfig(
+        text_config=tcfg, vision_config=vcfg, image_token_id=3,
+        attn_implementation="eager", tie_word_embeddings=False)
+    torch.manual_seed(27)
+    hf = transformers.Glm4vForConditionalGeneration(hf_cfg).eval()
+    cfg = Glm4vConfig.from_hf_config(hf_cfg.to_dict())
+    mine = Glm4vForConditionalGeneration(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert not missing, missing
+
+    ids = torch.randint(5, 200, (2, 15))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=8e-4, rtol=8e-4)
+
+    grid = torch.tensor([[1, 4, 4]])
+    pixels = torch.randn(16, 3 * 1 * 4 * 4)
+    ids = torch.randint(5, 200, (1, 18))
+    ids[0, 6:10] = 3
+    mm_type = (ids == 3).to(torch.int32)
+    with torch.no_grad():
+        ref = hf(ids, pixel_values=pixels, image_grid_thw=grid,
+                 mm_token_type_ids=mm_type).logits
+        out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
