@@ -485,6 +485,11 @@ class Qwen3VLForConditionalGeneration(nn.Module):
                 nn.init.zeros_(m.bias)
             elif type(m).__name__ == "RMSNorm":
                 nn.init.ones_(m.weight)
+            elif type(m).__name__ == "MoE":          # Qwen3VLMoe text FFNs
+                nn.init.normal_(m.gate.weight, std=std)
+                nn.init.normal_(m.experts.gate_proj, std=std)
+                nn.init.normal_(m.experts.up_proj, std=std)
+                nn.init.normal_(m.experts.down_proj, std=std)
         if self.config.text.tie_word_embeddings:
             self.lm_head.weight = self.model.language_model.embed_tokens.weight
 
