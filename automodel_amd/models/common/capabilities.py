@@ -109,6 +109,9 @@ _CAPS = {
     "Glm4vForConditionalGeneration": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False,
         supports_packed_sequences=False, flash_head_dims=()),
+    "Qwen2AudioForConditionalGeneration": ModelCapabilities(
+        supports_tp=False, supports_pp=False, supports_cp=False,
+        supports_packed_sequences=False, flash_head_dims=()),
     "Llama4ForCausalLM": ModelCapabilities(
         supports_tp=False, supports_pp=False, supports_cp=False),
     # generic transformers fallback: DP/FSDP only (reference non-custom path)

@@ -124,6 +124,12 @@ def _ensure_builtin() -> None:
 
         for name in Glm4vForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Glm4vForConditionalGeneration
+        from automodel_amd.models.qwen2_audio.model import (
+            Qwen2AudioForConditionalGeneration,
+        )
+
+        for name in Qwen2AudioForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen2AudioForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

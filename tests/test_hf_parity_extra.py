@@ -2147,3 +2147,45 @@ def test_hf_logits_parity_glm4v():
                  mm_token_type_ids=mm_type).logits
         out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_qwen2_audio():
+    """Qwen2-Audio: Whisper-style encoder (conv+stride2, pre-LN, avg-pool 2)
+    + linear projector splicing audio embeddings at audio_token_id."""
+    transformers = pytest.importorskip("transformers")
+    from automodel_amd.models.qwen2_audio.model import (
+        Qwen2AudioConfig,
+        Qwen2AudioForConditionalGeneration,
+    )
+
+    acfg = dict(d_model=32, encoder_layers=2, encoder_attention_heads=2,
+                encoder_ffn_dim=48, num_mel_bins=16, max_source_positions=10)
+    tcfg = dict(model_type="qwen2", vocab_size=120, hidden_size=64,
+                intermediate_size=96, num_hidden_layers=2,
+                num_attention_heads=4, num_key_value_heads=2,
+                max_position_embeddings=128)
+    hf_cfg = transformers.Qwen2AudioConfig(
+        audio_config=acfg, text_config=tcfg, audio_token_id=3,
+        attn_implementation="eager")
+    torch.manual_seed(28)
+    hf = transformers.Qwen2AudioForConditionalGeneration(hf_cfg).eval()
+    cfg = Qwen2AudioConfig.from_hf_config(hf_cfg.to_dict())
+    mine = Qwen2AudioForConditionalGeneration(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+
+    # mel input padded to max_source_positions * 2 = 20 -> conv2 stride 2 ->
+    # 10 frames -> avg pool 2 -> 5 audio tokens
+    feats = torch.randn(1, 16, 20)
+    ids = torch.randint(5, 120, (1, 14))
+    ids[0, 3:8] = 3
+    fmask = torch.ones(1, 20, dtype=torch.long)
+    with torch.no_grad():
+        ref = hf(ids, input_features=feats, feature_attention_mask=fmask).logits
+        out = mine(ids, input_features=feats)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+    # text-only
+    ids = torch.randint(5, 120, (2, 11))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits, atol=8e-4, rtol=8e-4)
