@@ -50,6 +50,8 @@ class VisionConfig:
     intermediate_size: int | None = None     # v2_5 explicit mlp width
     window_size: int = 112
     fullatt_block_indexes: tuple = (7, 15, 23, 31)
+    # omni towers keep separate q/k/v linears instead of a fused qkv
+    qkv_separate: bool = False
 
 
 @dataclass
@@ -135,14 +137,25 @@ class VisionAttention(nn.Module):
         super().__init__()
         self.num_heads = cfg.num_heads
         self.head_dim = cfg.embed_dim // cfg.num_heads
-        self.qkv = nn.Linear(cfg.embed_dim, cfg.embed_dim * 3, bias=True)
+        self.separate = cfg.qkv_separate
+        if self.separate:       # omni naming: attn.q / attn.k / attn.v
+            self.q = nn.Linear(cfg.embed_dim, cfg.embed_dim, bias=True)
+            self.k = nn.Linear(cfg.embed_dim, cfg.embed_dim, bias=True)
+            self.v = nn.Linear(cfg.embed_dim, cfg.embed_dim, bias=True)
+        else:
+            self.qkv = nn.Linear(cfg.embed_dim, cfg.embed_dim * 3, bias=True)
         self.proj = nn.Linear(cfg.embed_dim, cfg.embed_dim)
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
                 cu_seqlens: torch.Tensor) -> torch.Tensor:
         N = x.shape[0]
-        q, k, v = self.qkv(x).reshape(N, 3, self.num_heads, -1) \
-            .permute(1, 0, 2, 3).unbind(0)
+        if self.separate:
+            q = self.q(x).reshape(N, self.num_heads, -1)
+            k = self.k(x).reshape(N, self.num_heads, -1)
+            v = self.v(x).reshape(N, self.num_heads, -1)
+        else:
+            q, k, v = self.qkv(x).reshape(N, 3, self.num_heads, -1) \
+                .permute(1, 0, 2, 3).unbind(0)
         c, s = cos[:, None, :].float(), sin[:, None, :].float()
 
         def rot(t):

@@ -130,6 +130,12 @@ def _ensure_builtin() -> None:
 
         for name in Qwen2AudioForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen2AudioForConditionalGeneration
+        from automodel_amd.models.qwen2_5_omni.model import (
+            Qwen2_5OmniThinkerForConditionalGeneration,
+        )
+
+        for name in Qwen2_5OmniThinkerForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen2_5OmniThinkerForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

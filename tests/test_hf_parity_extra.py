@@ -2189,3 +2189,79 @@ def test_hf_logits_parity_qwen2_audio():
     ids = torch.randint(5, 120, (2, 11))
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_qwen2_5_omni_thinker():
+    """Qwen2.5-Omni thinker: windowed Whisper-style audio encoder (chunked
+    attention, sinusoidal positions, per-sample pooling), the Qwen2.5-VL
+    windowed ViT with separate q/k/v, and TMRoPE position assignment —
+    text-only, +audio, and +audio+image paths."""
+    from automodel_amd.models.qwen2_5_omni.model import (
+        Qwen2_5OmniThinkerConfig,
+        Qwen2_5OmniThinkerForConditionalGeneration,
+    )
+
+    torch.manual_seed(60)
+    hf_cfg = transformers.Qwen2_5OmniThinkerConfig(
+        audio_config=dict(d_model=32, encoder_layers=2,
+                          encoder_attention_heads=2, encoder_ffn_dim=48,
+                          num_mel_bins=16, max_source_positions=16,
+                          n_window=4, output_dim=64),
+        vision_config=dict(hidden_size=32, depth=2, num_heads=2,
+                           out_hidden_size=64, patch_size=4,
+                           temporal_patch_size=2, spatial_merge_size=2,
+                           in_channels=3, intermediate_size=64,
+                           window_size=16, fullatt_block_indexes=[1]),
+        text_config=dict(vocab_size=300, hidden_size=64, intermediate_size=128,
+                         num_hidden_layers=2, num_attention_heads=4,
+                         num_key_value_heads=2, max_position_embeddings=256,
+                         rms_norm_eps=1e-6,
+                         rope_parameters={"rope_type": "default",
+                                          "rope_theta": 10000.0,
+                                          "mrope_section": [2, 3, 3]},
+                         tie_word_embeddings=False),
+        audio_token_index=3, image_token_index=4, video_token_index=5,
+        vision_start_token_id=298, audio_start_token_id=297,
+        attn_implementation="eager")
+    hf = transformers.Qwen2_5OmniThinkerForConditionalGeneration(hf_cfg).eval()
+    cfg = Qwen2_5OmniThinkerConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.audio_token_id == 3 and cfg.image_token_id == 4
+    mine = Qwen2_5OmniThinkerForConditionalGeneration(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k or "rot_inv" in k or "positional_embedding" in k
+               for k in missing), missing
+
+    # ---- text-only
+    ids = torch.randint(6, 290, (2, 15))
+    with torch.no_grad():
+        ref = hf(input_ids=ids, attention_mask=torch.ones_like(ids)).logits
+        torch.testing.assert_close(mine(ids), ref, atol=8e-4, rtol=8e-4)
+
+    # ---- audio: 20 mel frames -> chunks [8,8,4] -> 10 post-conv -> 5 tokens
+    feats = torch.randn(1, 16, 20)
+    fmask = torch.ones(1, 20, dtype=torch.long)
+    seq = torch.cat([torch.randint(6, 290, (1, 3)), torch.tensor([[297]]),
+                     torch.full((1, 5), 3), torch.randint(6, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, input_features=feats,
+                 feature_attention_mask=fmask,
+                 attention_mask=torch.ones_like(seq)).logits
+        out = mine(seq, input_features=feats, feature_attention_mask=fmask)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+    # ---- audio + image in one sequence (TMRoPE interleaved assignment)
+    grid = torch.tensor([[1, 8, 8]])
+    pix = torch.randn(64, 3 * 2 * 4 * 4)
+    seq = torch.cat([torch.randint(6, 290, (1, 2)), torch.tensor([[297]]),
+                     torch.full((1, 5), 3), torch.randint(6, 290, (1, 2)),
+                     torch.tensor([[298]]), torch.full((1, 16), 4),
+                     torch.randint(6, 290, (1, 3))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, input_features=feats,
+                 feature_attention_mask=fmask, pixel_values=pix,
+                 image_grid_thw=grid,
+                 attention_mask=torch.ones_like(seq)).logits
+        out = mine(seq, input_features=feats, feature_attention_mask=fmask,
+                   pixel_values=pix, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
