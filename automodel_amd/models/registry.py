@@ -136,6 +136,12 @@ def _ensure_builtin() -> None:
 
         for name in Qwen2_5OmniThinkerForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen2_5OmniThinkerForConditionalGeneration
+        from automodel_amd.models.qwen3_omni_moe.model import (
+            Qwen3OmniMoeThinkerForConditionalGeneration,
+        )
+
+        for name in Qwen3OmniMoeThinkerForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Qwen3OmniMoeThinkerForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

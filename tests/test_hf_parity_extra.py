@@ -2265,3 +2265,81 @@ def test_hf_logits_parity_qwen2_5_omni_thinker():
         out = mine(seq, input_features=feats, feature_attention_mask=fmask,
                    pixel_values=pix, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_qwen3_omni_moe_thinker():
+    """Qwen3-Omni-MoE thinker: AuT audio encoder (3x stride-2 Conv2d
+    downsample, conv_out fold, merged inference windows), DeepStack ViT with
+    omni merger naming, MoE text under interleaved MRoPE, float TMRoPE —
+    text-only, +audio, +audio+image paths."""
+    from automodel_amd.models.qwen3_omni_moe.model import (
+        Qwen3OmniMoeConfig,
+        Qwen3OmniMoeThinkerForConditionalGeneration,
+    )
+    from automodel_amd.models.qwen3_vl.model import Qwen3VLMoeStateDictAdapter
+
+    torch.manual_seed(61)
+    tcfg = dict(vocab_size=300, hidden_size=64, intermediate_size=96,
+                num_hidden_layers=2, num_attention_heads=4,
+                num_key_value_heads=2, head_dim=16,
+                num_experts=4, num_experts_per_tok=2, moe_intermediate_size=32,
+                rope_scaling={"rope_type": "default", "mrope_section": [4, 2, 2]},
+                max_position_embeddings=256, tie_word_embeddings=False)
+    vcfg = dict(depth=2, hidden_size=32, intermediate_size=64, num_heads=2,
+                patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                out_hidden_size=64, num_position_embeddings=36,
+                deepstack_visual_indexes=[0, 1])
+    # n_window must stay 50: HF hardcodes 13 tokens per full chunk
+    acfg = dict(d_model=32, encoder_layers=2, encoder_attention_heads=2,
+                encoder_ffn_dim=48, num_mel_bins=16, max_source_positions=16,
+                n_window=50, n_window_infer=200, conv_chunksize=2,
+                downsample_hidden_size=8, output_dim=64)
+    hf_cfg = transformers.Qwen3OmniMoeThinkerConfig(
+        audio_config=acfg, vision_config=vcfg, text_config=tcfg,
+        audio_token_id=3, image_token_id=4, video_token_id=5,
+        vision_start_token_id=298, audio_start_token_id=297,
+        attn_implementation="eager")
+    hf = transformers.Qwen3OmniMoeThinkerForConditionalGeneration(hf_cfg).eval()
+    cfg = Qwen3OmniMoeConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.text.num_experts == 4 and cfg.audio.n_window == 50
+    mine = Qwen3OmniMoeThinkerForConditionalGeneration(cfg).eval()
+    sd = Qwen3VLMoeStateDictAdapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k or "rot_inv" in k or "inv_freq" in k
+               or "positional_embedding" in k for k in missing), missing
+
+    # ---- text-only
+    ids = torch.randint(6, 290, (2, 12))
+    with torch.no_grad():
+        ref = hf(input_ids=ids, attention_mask=torch.ones_like(ids)).logits
+        torch.testing.assert_close(mine(ids), ref, atol=8e-4, rtol=8e-4)
+
+    # ---- audio only: 230 frames -> chunks [100,100,30] -> 13+13+4 = 30
+    # audio tokens, attention windows of 26 (n_window_infer=200 -> 2 chunks)
+    feats = torch.randn(1, 16, 230)
+    fmask = torch.ones(1, 230, dtype=torch.long)
+    seq = torch.cat([torch.randint(6, 290, (1, 3)), torch.tensor([[297]]),
+                     torch.full((1, 30), 3), torch.randint(6, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, input_features=feats,
+                 feature_attention_mask=fmask,
+                 attention_mask=torch.ones_like(seq)).logits
+        out = mine(seq, input_features=feats, feature_attention_mask=fmask)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+    # ---- audio + image (float TMRoPE + deepstack injection)
+    grid = torch.tensor([[1, 4, 4]])
+    pix = torch.randn(16, 3 * 1 * 4 * 4)
+    seq = torch.cat([torch.randint(6, 290, (1, 2)), torch.tensor([[297]]),
+                     torch.full((1, 30), 3), torch.randint(6, 290, (1, 2)),
+                     torch.tensor([[298]]), torch.full((1, 4), 4),
+                     torch.randint(6, 290, (1, 3))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, input_features=feats,
+                 feature_attention_mask=fmask, pixel_values=pix,
+                 image_grid_thw=grid,
+                 attention_mask=torch.ones_like(seq)).logits
+        out = mine(seq, input_features=feats, feature_attention_mask=fmask,
+                   pixel_values=pix, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
