@@ -142,6 +142,12 @@ def _ensure_builtin() -> None:
 
         for name in Qwen3OmniMoeThinkerForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Qwen3OmniMoeThinkerForConditionalGeneration
+        from automodel_amd.models.llava_onevision.model import (
+            LlavaOnevisionForConditionalGeneration,
+        )
+
+        for name in LlavaOnevisionForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = LlavaOnevisionForConditionalGeneration
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM

@@ -2343,3 +2343,62 @@ def test_hf_logits_parity_qwen3_omni_moe_thinker():
         out = mine(seq, input_features=feats, feature_attention_mask=fmask,
                    pixel_values=pix, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_llava_onevision():
+    """LLaVA-OneVision: SigLIP tower (no head), anyres grid re-assembly with
+    unpad + image_newline columns, video bilinear pooling, qwen2 decoder —
+    text, anyres-image, and video paths."""
+    from automodel_amd.models.llava_onevision.model import (
+        LlavaOnevisionConfig,
+        LlavaOnevisionForConditionalGeneration,
+    )
+
+    torch.manual_seed(62)
+    hf_cfg = transformers.LlavaOnevisionConfig(
+        vision_config=dict(model_type="siglip_vision_model", hidden_size=32,
+                           intermediate_size=48, num_hidden_layers=2,
+                           num_attention_heads=2, image_size=8, patch_size=4,
+                           vision_use_head=False),
+        text_config=dict(model_type="qwen2", vocab_size=300, hidden_size=64,
+                         intermediate_size=96, num_hidden_layers=2,
+                         num_attention_heads=4, num_key_value_heads=2,
+                         max_position_embeddings=256, rope_theta=10000.0,
+                         tie_word_embeddings=False),
+        image_grid_pinpoints=[[8, 8], [8, 16], [16, 8], [16, 16]],
+        image_token_index=3, video_token_index=4,
+        vision_feature_layer=-1, vision_feature_select_strategy="full",
+        vision_aspect_ratio="anyres_max_9", attn_implementation="eager")
+    hf = transformers.LlavaOnevisionForConditionalGeneration(hf_cfg).eval()
+    cfg = LlavaOnevisionConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.image_token_id == 3 and cfg.vision_feature_layer == -1
+    mine = LlavaOnevisionForConditionalGeneration(cfg).eval()
+    missing, unexpected = mine.load_state_dict(hf.state_dict(), strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+
+    # ---- text-only
+    ids = torch.randint(6, 290, (2, 11))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(input_ids=ids).logits,
+                                   atol=8e-4, rtol=8e-4)
+
+    # ---- anyres image: 12x16 -> best (16,16) -> 2x2 grid + base = 5 crops,
+    # base 4 tokens + 4x(4+1 newline) = 24 image tokens
+    pix = torch.randn(1, 5, 3, 8, 8)
+    sizes = torch.tensor([[12, 16]])
+    seq = torch.cat([torch.randint(6, 290, (1, 3)), torch.full((1, 24), 3),
+                     torch.randint(6, 290, (1, 4))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values=pix, image_sizes=sizes).logits
+        out = mine(seq, pixel_values=pix, image_sizes=sizes)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+    # ---- video: 2 frames -> pooled 1 token/frame + 1 newline = 3 tokens
+    vid = torch.randn(1, 2, 3, 8, 8)
+    seq = torch.cat([torch.randint(6, 290, (1, 2)), torch.full((1, 3), 4),
+                     torch.randint(6, 290, (1, 3))], dim=1)
+    with torch.no_grad():
+        ref = hf(input_ids=seq, pixel_values_videos=vid).logits
+        out = mine(seq, pixel_values_videos=vid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
