@@ -120,8 +120,11 @@ class Glm4MoeAttention(nn.Module):
         if self.use_qk_norm:
             q, k = self.q_norm(q), self.k_norm(k)
         r = self.rot_dim
-        c = cos[None, :, None, :].to(q.dtype)
-        s = sin[None, :, None, :].to(q.dtype)
+        # [S, rot] plain tables or [B, S, rot] mrope tables (glm4v_moe)
+        c = (cos[None, :, None, :] if cos.dim() == 2
+             else cos[:, :, None, :]).to(q.dtype)
+        s = (sin[None, :, None, :] if sin.dim() == 2
+             else sin[:, :, None, :]).to(q.dtype)
         q = torch.cat([self._rot(q[..., :r], c, s), q[..., r:]], dim=-1)
         k = torch.cat([self._rot(k[..., :r], c, s), k[..., r:]], dim=-1)
         o = flash_attention(q, k, v, causal=True, backend="sdpa")

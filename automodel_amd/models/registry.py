@@ -124,6 +124,12 @@ def _ensure_builtin() -> None:
 
         for name in Glm4vForConditionalGeneration.hf_architectures:
             _REGISTRY[name] = Glm4vForConditionalGeneration
+        from automodel_amd.models.glm4v_moe.model import (
+            Glm4vMoeForConditionalGeneration,
+        )
+
+        for name in Glm4vMoeForConditionalGeneration.hf_architectures:
+            _REGISTRY[name] = Glm4vMoeForConditionalGeneration
         from automodel_amd.models.qwen2_audio.model import (
             Qwen2AudioForConditionalGeneration,
         )

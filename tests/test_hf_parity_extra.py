@@ -2402,3 +2402,54 @@ def test_hf_logits_parity_llava_onevision():
         ref = hf(input_ids=seq, pixel_values_videos=vid).logits
         out = mine(seq, pixel_values_videos=vid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_glm4v_moe():
+    """GLM-4.5V: GLM-4V ViT + GLM4-MoE text (sigmoid+bias routing, group
+    top-k, shared expert, dense-first, plain pre/post norms, contiguous-half
+    partial rotary) on chunked 3D MRoPE — text and image paths."""
+    from automodel_amd.models.glm4v_moe.model import (
+        Glm4vMoeConfig,
+        Glm4vMoeForConditionalGeneration,
+    )
+
+    torch.manual_seed(63)
+    tcfg = dict(vocab_size=200, hidden_size=64, intermediate_size=96,
+                num_hidden_layers=3, num_attention_heads=4,
+                num_key_value_heads=2, head_dim=16,
+                n_routed_experts=8, n_shared_experts=1, num_experts_per_tok=2,
+                moe_intermediate_size=48, first_k_dense_replace=1,
+                norm_topk_prob=True, routed_scaling_factor=1.0,
+                attention_bias=True,
+                rope_scaling={"rope_type": "default",
+                              "partial_rotary_factor": 0.5,
+                              "mrope_section": [2, 1, 1]},
+                max_position_embeddings=128, tie_word_embeddings=False)
+    vcfg = dict(depth=2, hidden_size=32, intermediate_size=64, num_heads=2,
+                patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                out_hidden_size=64, image_size=16)
+    hf_cfg = transformers.Glm4vMoeConfig(
+        text_config=tcfg, vision_config=vcfg, image_token_id=3,
+        attn_implementation="eager")
+    hf = transformers.Glm4vMoeForConditionalGeneration(hf_cfg).eval()
+    cfg = Glm4vMoeConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.text.moe.n_routed_experts == 8 and cfg.text.attention_bias
+    mine = Glm4vMoeForConditionalGeneration(cfg).eval()
+    sd = mine.state_dict_adapter.from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k or "inv_freq" in k for k in missing), missing
+
+    ids = torch.randint(5, 200, (2, 15))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits,
+                                   atol=8e-4, rtol=8e-4)
+    grid = torch.tensor([[1, 4, 4]])
+    pixels = torch.randn(16, 3 * 1 * 4 * 4)
+    ids = torch.randint(5, 200, (1, 18))
+    ids[0, 6:10] = 3
+    with torch.no_grad():
+        ref = hf(ids, pixel_values=pixels, image_grid_thw=grid,
+                 mm_token_type_ids=(ids == 3).int()).logits
+        out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
+    torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
