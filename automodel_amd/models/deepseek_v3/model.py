@@ -127,7 +127,8 @@ class MLAAttention(nn.Module):
         self.o_proj = nn.Linear(H * cfg.v_head_dim, cfg.hidden_size, bias=False)
         self.scale = cfg.qk_head_dim**-0.5
 
-    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+    def _qkv(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor):
+        """Project to the decoupled-rope (qf, kf, v) triple [B,S,H,D*]."""
         cfg = self.cfg
         B, S, _ = x.shape
         H = self.num_heads
@@ -158,6 +159,13 @@ class MLAAttention(nn.Module):
 
         qf = torch.cat([q_nope, q_rope], dim=-1)   # B,S,H,qk (192 = nope+rope)
         kf = torch.cat([k_nope, k_rope], dim=-1)
+        return qf, kf, v
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor) -> torch.Tensor:
+        cfg = self.cfg
+        B, S, _ = x.shape
+        H = self.num_heads
+        qf, kf, v = self._qkv(x, cos, sin)
         if self.backend.attn == "hip" and qf.is_cuda:
             # split-dim flash kernel: (Dqk=192, Dv=128) instantiation
             from automodel_amd.ops.attention import flash_attention

@@ -157,6 +157,12 @@ def _ensure_builtin() -> None:
 
         for name in DeepseekV3ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV3ForCausalLM
+        from automodel_amd.models.deepseek_v32.model import (
+            DeepseekV32ForCausalLM,
+        )
+
+        for name in DeepseekV32ForCausalLM.hf_architectures:
+            _REGISTRY[name] = DeepseekV32ForCausalLM
     except ImportError:
         pass
     try:
