@@ -163,6 +163,10 @@ def _ensure_builtin() -> None:
 
         for name in DeepseekV32ForCausalLM.hf_architectures:
             _REGISTRY[name] = DeepseekV32ForCausalLM
+        from automodel_amd.models.kimi_linear.model import KimiLinearForCausalLM
+
+        for name in KimiLinearForCausalLM.hf_architectures:
+            _REGISTRY[name] = KimiLinearForCausalLM
     except ImportError:
         pass
     try:
