@@ -68,6 +68,10 @@ class _Bucket:
         if self.gather_work is not None:
             return
         src = self.shard.data.to(self.compute_dtype)
+        if not dist.is_initialized() or dist.get_world_size() == 1:
+            self.gather_buf.copy_(src)
+            self.gather_work = _DONE
+            return
         if dist.get_backend() == "gloo":
             chunks = list(self.gather_buf.chunk(dist.get_world_size()))
             self.gather_work = dist.all_gather(chunks, src, async_op=async_op)
@@ -125,8 +129,9 @@ class MegatronFSDPEngine:
                  **optim_kwargs):
         self.model = model
         self.group = process_group
-        self.world = dist.get_world_size(process_group)
-        self.rank = dist.get_rank(process_group)
+        self.world = (dist.get_world_size(process_group)
+                      if dist.is_initialized() else 1)
+        self.rank = dist.get_rank(process_group) if dist.is_initialized() else 0
         self.prefetch = prefetch
         if layers is None:
             layers = [m for m in model.modules()
@@ -137,14 +142,18 @@ class MegatronFSDPEngine:
         self.buckets: list[_Bucket] = []
         pairs: list[tuple[nn.Module, _Bucket]] = []
         for i, layer in enumerate(layers):
-            ps = [p for p in layer.parameters() if id(p) not in seen]
+            # frozen params (e.g. PEFT base weights) stay replicated plain
+            # tensors: AdamW weight-decay must never touch them
+            ps = [p for p in layer.parameters()
+                  if id(p) not in seen and p.requires_grad]
             for p in ps:
                 seen.add(id(p))
             if ps:
                 b = _Bucket(f"layer{i}", ps, self.world, self.rank, dtype)
                 self.buckets.append(b)
                 pairs.append((layer, b))
-        rest = [p for p in model.parameters() if id(p) not in seen]
+        rest = [p for p in model.parameters()
+                if id(p) not in seen and p.requires_grad]
         if rest:
             self.buckets.append(_Bucket("rest", rest, self.world,
                                         self.rank, dtype))
@@ -188,6 +197,12 @@ class MegatronFSDPEngine:
     def reduce_grads(self) -> None:
         """Reduce-scatter each bucket's grads (SUM) into fp32 main-grad
         shards; accumulates across micro-batches until ``step``."""
+        if self.world == 1:
+            for b in self.buckets:
+                b.main_grad += b.grad_flat()[:b.shard_n]
+                for p in b.params:
+                    p.grad = None
+            return
         pending = []
         for b in self.buckets:
             flat = b.grad_flat()
@@ -214,7 +229,8 @@ class MegatronFSDPEngine:
         sq = torch.zeros(1, device=self.buckets[0].device)
         for b in self.buckets:
             sq += b.main_grad.square().sum()
-        dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.group)
+        if self.world > 1:
+            dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=self.group)
         norm = sq.sqrt()
         scale = max_norm / (float(norm) + 1e-6)
         if scale < 1.0:

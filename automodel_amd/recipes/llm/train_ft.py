@@ -37,6 +37,8 @@ from automodel_amd.training.utils import clip_grad_norm_, count_label_tokens, pr
 
 
 class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
+    engine = None          # megatron_fsdp second engine (distributed.engine)
+
     def __init__(self, cfg: ConfigNode):
         super().__init__(cfg)
         self.logger = setup_logging()
@@ -128,7 +130,13 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             from automodel_amd.parallel.tp import apply_tp
             apply_tp(self.model, self.mesh["tp"],
                      sequence_parallel=dist_cfg.get("sequence_parallel", False))
-        if self.mesh.mesh is not None and (
+        # second sharding engine (reference megatron_fsdp.py:46): flat-shard
+        # DP with jointly-sharded optimizer state; excludes FSDP2/PP/EP
+        self._mfsdp = dist_cfg.get("engine") == "megatron_fsdp"
+        if self._mfsdp:
+            assert self.mesh.pp_size == 1 and self.mesh.dims.get("tp", 1) == 1 \
+                and self.cp_size == 1, "megatron_fsdp engine is DP-only"
+        if not self._mfsdp and self.mesh.mesh is not None and (
             self.mesh.dims["dp_shard"] > 1 or self.cp_size > 1
         ):
             fsdp_axis = "dp_shard_cp" if self.cp_size > 1 else "dp_shard"
@@ -168,12 +176,23 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
 
         # ---- optimizer / schedulers
         opt_cfg = cfg.get("optimizer", ConfigNode())
-        self.optimizer = opt_cfg.maybe_instantiate(model=self.model) or build_adamw(
-            self.model,
-            lr=opt_cfg.get("lr", 2e-5),
-            weight_decay=opt_cfg.get("weight_decay", 0.01),
-            betas=tuple(opt_cfg.get("betas", (0.9, 0.999))),
-        )
+        self.engine = None
+        if self._mfsdp:
+            from automodel_amd.parallel.megatron_fsdp import MegatronFSDPEngine
+            self.engine = MegatronFSDPEngine(
+                self.model,
+                lr=opt_cfg.get("lr", 2e-5),
+                weight_decay=opt_cfg.get("weight_decay", 0.01),
+                betas=tuple(opt_cfg.get("betas", (0.9, 0.999))),
+            )
+            self.optimizer = self.engine.optimizer
+        else:
+            self.optimizer = opt_cfg.maybe_instantiate(model=self.model) or build_adamw(
+                self.model,
+                lr=opt_cfg.get("lr", 2e-5),
+                weight_decay=opt_cfg.get("weight_decay", 0.01),
+                betas=tuple(opt_cfg.get("betas", (0.9, 0.999))),
+            )
         sched_cfg = cfg.get("step_scheduler", ConfigNode())
         self.step_scheduler = StepScheduler(
             grad_acc_steps=sched_cfg.get("grad_acc_steps", 1),
@@ -330,8 +349,10 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
 
         # loss_sum / global_tokens * dp_cp_world compensates FSDP's mean-reduce
         # (with CP, each rank sees a token subset and FSDP reduces over
-        # dp_shard_cp — reference train_ft.py:1186 scales by dp_cp_size)
-        loss_scale = self.mesh.dp_cp_size / global_tokens
+        # dp_shard_cp — reference train_ft.py:1186 scales by dp_cp_size);
+        # the megatron_fsdp engine reduce-scatters with SUM, so no dp factor
+        loss_scale = ((1.0 if self.engine is not None else self.mesh.dp_cp_size)
+                      / global_tokens)
         total_loss = torch.zeros((), dtype=torch.float32, device=device)
         if self.pipeline is not None:
             total_loss = self._pp_step(batches, loss_scale).float()
@@ -339,12 +360,20 @@ class TrainFinetuneRecipeForNextTokenPrediction(BaseRecipe):
             for i, batch in enumerate(batches):
                 prepare_for_grad_accumulation(self.model, is_final_microbatch=(i == len(batches) - 1))
                 total_loss += self._forward_backward_step(batch, loss_scale).float()
+                if self.engine is not None:
+                    self.engine.reduce_grads()
 
-        grad_norm = clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
+        if self.engine is not None:
+            grad_norm = self.engine.clip_grad_norm(self.max_grad_norm)
+        else:
+            grad_norm = clip_grad_norm_(self.model.parameters(), self.max_grad_norm)
         if hasattr(self, "checkpointer") and hasattr(self.checkpointer, "maybe_wait_for_staging"):
             self.checkpointer.maybe_wait_for_staging()
-        self.optimizer.step()
-        self.optimizer.zero_grad(set_to_none=True)
+        if self.engine is not None:
+            self.engine.step()
+        else:
+            self.optimizer.step()
+            self.optimizer.zero_grad(set_to_none=True)
         self.lr_scheduler.step()
         # MoE: aux-free gate-bias update + load metrics (reference
         # train_ft.py update_moe_gate_bias / load_balance_metrics)
