@@ -68,3 +68,129 @@ def test_glm4v_builds_and_steps():
     ids[0, 2:6] = 3
     pixels = torch.randn(16, 3 * 16)
     _step(m, ids, pixel_values=pixels, image_grid_thw=torch.tensor([[1, 4, 4]]))
+
+
+def test_qwen2_5_omni_builds_and_steps():
+    from automodel_amd.models.qwen2_5_omni.model import (
+        Qwen2_5OmniThinkerConfig,
+        Qwen2_5OmniThinkerForConditionalGeneration,
+    )
+
+    cfg = Qwen2_5OmniThinkerConfig(
+        text=dict(vocab_size=320, hidden_size=32, intermediate_size=48,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, attention_bias=True,
+                  max_position_embeddings=64, rope_theta=10000.0),
+        audio=dict(d_model=16, encoder_layers=1, encoder_attention_heads=2,
+                   encoder_ffn_dim=24, num_mel_bins=8, max_source_positions=8,
+                   n_window=4, output_dim=32),
+        vision=dict(variant="v2_5", qkv_separate=True, embed_dim=16, depth=1,
+                    num_heads=2, intermediate_size=24, patch_size=4,
+                    temporal_patch_size=2, spatial_merge_size=2,
+                    hidden_size=32, window_size=16, fullatt_block_indexes=(0,)),
+        mrope_section=(2, 1, 1), audio_token_id=3, image_token_id=4,
+        vision_start_token_id=298, audio_start_token_id=297)
+    torch.manual_seed(0)
+    m = Qwen2_5OmniThinkerForConditionalGeneration(cfg)
+    m.init_weights()
+    # 20 mel frames -> chunks [8,8,4] -> 10 -> 5 audio tokens
+    feats = torch.randn(1, 8, 20)
+    fmask = torch.ones(1, 20, dtype=torch.long)
+    ids = torch.cat([torch.randint(6, 290, (1, 2)), torch.tensor([[297]]),
+                     torch.full((1, 5), 3), torch.randint(6, 290, (1, 2)),
+                     torch.tensor([[298]]), torch.full((1, 16), 4),
+                     torch.randint(6, 290, (1, 2))], dim=1)
+    pix = torch.randn(64, 3 * 2 * 4 * 4)
+    _step(m, ids, input_features=feats, feature_attention_mask=fmask,
+          pixel_values=pix, image_grid_thw=torch.tensor([[1, 8, 8]]))
+
+
+def test_qwen3_omni_moe_builds_and_steps():
+    from automodel_amd.models.qwen3_omni_moe.model import (
+        Qwen3OmniMoeConfig,
+        Qwen3OmniMoeThinkerForConditionalGeneration,
+    )
+
+    cfg = Qwen3OmniMoeConfig(
+        text=dict(vocab_size=320, hidden_size=32, intermediate_size=48,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, head_dim=8, mrope_section=(2, 1, 1),
+                  num_experts=4, num_experts_per_tok=2,
+                  moe_intermediate_size=16, max_position_embeddings=64),
+        vision=dict(depth=1, hidden_size=16, intermediate_size=32, num_heads=2,
+                    patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                    out_hidden_size=32, num_position_embeddings=16,
+                    deepstack_visual_indexes=[0]),
+        audio=dict(d_model=16, encoder_layers=1, encoder_attention_heads=2,
+                   encoder_ffn_dim=24, num_mel_bins=8, max_source_positions=16,
+                   n_window=50, n_window_infer=200, conv_chunksize=2,
+                   downsample_hidden_size=8, output_dim=32),
+        audio_token_id=3, image_token_id=4,
+        vision_start_token_id=298, audio_start_token_id=297)
+    torch.manual_seed(1)
+    m = Qwen3OmniMoeThinkerForConditionalGeneration(cfg)
+    m.init_weights()
+    feats = torch.randn(1, 8, 230)        # -> 30 audio tokens
+    fmask = torch.ones(1, 230, dtype=torch.long)
+    ids = torch.cat([torch.randint(6, 290, (1, 2)), torch.tensor([[297]]),
+                     torch.full((1, 30), 3), torch.randint(6, 290, (1, 2)),
+                     torch.tensor([[298]]), torch.full((1, 4), 4),
+                     torch.randint(6, 290, (1, 2))], dim=1)
+    pix = torch.randn(16, 3 * 16)
+    _step(m, ids, input_features=feats, feature_attention_mask=fmask,
+          pixel_values=pix, image_grid_thw=torch.tensor([[1, 4, 4]]))
+
+
+def test_llava_onevision_builds_and_steps():
+    from automodel_amd.models.llava_onevision.model import (
+        LlavaOnevisionConfig,
+        LlavaOnevisionForConditionalGeneration,
+    )
+
+    cfg = LlavaOnevisionConfig(
+        text=dict(vocab_size=320, hidden_size=32, intermediate_size=48,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, attention_bias=True,
+                  max_position_embeddings=64, rope_theta=10000.0),
+        vision=dict(hidden_size=16, intermediate_size=24, num_hidden_layers=1,
+                    num_attention_heads=2, image_size=8, patch_size=4),
+        image_token_id=3, video_token_id=4,
+        image_grid_pinpoints=[[8, 8], [16, 16]],
+        vision_feature_layer=-1, vision_feature_select_strategy="full")
+    torch.manual_seed(2)
+    m = LlavaOnevisionForConditionalGeneration(cfg)
+    m.init_weights()
+    pix = torch.randn(1, 5, 3, 8, 8)
+    ids = torch.cat([torch.randint(6, 290, (1, 2)), torch.full((1, 24), 3),
+                     torch.randint(6, 290, (1, 2))], dim=1)
+    _step(m, ids, pixel_values=pix, image_sizes=torch.tensor([[12, 16]]))
+
+
+def test_glm4v_moe_builds_and_steps():
+    from automodel_amd.models.glm4v_moe.model import (
+        Glm4vMoeConfig,
+        Glm4vMoeForConditionalGeneration,
+    )
+
+    cfg = Glm4vMoeConfig(
+        text=dict(vocab_size=320, hidden_size=32, intermediate_size=48,
+                  num_hidden_layers=2, num_attention_heads=4,
+                  num_key_value_heads=2, head_dim=8,
+                  partial_rotary_factor=0.5, first_k_dense_replace=1,
+                  attention_bias=True, max_position_embeddings=64,
+                  moe=dict(n_routed_experts=4, n_shared_experts=1,
+                           n_activated_experts=2, moe_intermediate_size=16,
+                           score_func="sigmoid", expert_bias=True,
+                           norm_topk_prob=True,
+                           shared_expert_intermediate_size=16)),
+        vision=dict(depth=1, hidden_size=16, intermediate_size=32, num_heads=2,
+                    patch_size=4, temporal_patch_size=1, spatial_merge_size=2,
+                    out_hidden_size=32, image_size=16),
+        mrope_section=(1, 1, 0), image_token_id=3)
+    torch.manual_seed(3)
+    m = Glm4vMoeForConditionalGeneration(cfg)
+    m.init_weights()
+    ids = torch.randint(5, 300, (1, 14))
+    ids[0, 4:8] = 3
+    pix = torch.randn(16, 3 * 1 * 4 * 4)
+    _step(m, ids, pixel_values=pix, image_grid_thw=torch.tensor([[1, 4, 4]]))
