@@ -66,9 +66,11 @@ def kda_recurrent(q, k, v, g, beta):
     return torch.stack(outs, dim=1)                           # [B,S,H,Dv]
 
 
-def kda_chunked(q, k, v, g, beta, chunk_size: int = 16):
+def kda_chunked(q, k, v, g, beta, chunk_size: int = 16,
+                initial_state=None, return_final_state=False):
     """Chunked KDA (float32): per-channel-decay generalization of the
-    qwen3_next chunked gated delta rule. Same I/O as ``kda_recurrent``."""
+    qwen3_next chunked gated delta rule. Same I/O as ``kda_recurrent``;
+    initial/final state support the CP chunk relay."""
     q = (_l2norm(q.transpose(1, 2).float()))
     k = _l2norm(k.transpose(1, 2).float())
     v = v.transpose(1, 2).float()
@@ -93,7 +95,8 @@ def kda_chunked(q, k, v, g, beta, chunk_size: int = 16):
 
     v_beta = v * beta.unsqueeze(-1)
     k_beta = k * beta.unsqueeze(-1)
-    state = q.new_zeros(b, h, dk, dv)
+    state = (q.new_zeros(b, h, dk, dv) if initial_state is None
+             else initial_state.to(q.dtype))
     out = torch.empty(b, h, n, c, dv, dtype=q.dtype, device=q.device)
     low = torch.ones(c, c, dtype=torch.bool, device=q.device).tril()
     for i in range(n):
@@ -116,7 +119,10 @@ def kda_chunked(q, k, v, g, beta, chunk_size: int = 16):
         state = (state * glast.exp().unsqueeze(-1) +
                  (k[:, :, i] * (glast.unsqueeze(-2) - Gi).exp()
                   ).transpose(-1, -2) @ v_new)
-    return out.reshape(b, h, -1, dv)[:, :, :s].transpose(1, 2)
+    out = out.reshape(b, h, -1, dv)[:, :, :s].transpose(1, 2)
+    if return_final_state:
+        return out, state
+    return out
 
 
 @dataclass

@@ -144,7 +144,8 @@ def _l2norm(x, eps: float = 1e-6):
     return x * torch.rsqrt((x * x).sum(-1, keepdim=True) + eps)
 
 
-def gated_delta_rule_chunked(q, k, v, g, beta, chunk_size: int = 64):
+def gated_delta_rule_chunked(q, k, v, g, beta, chunk_size: int = 64,
+                             initial_state=None, return_final_state=False):
     """Chunked gated delta rule (float32). q/k/v [B,S,Hv,D*]; g/beta [B,S,Hv].
 
     Math identical to the reference's torch_chunk_gated_delta_rule
@@ -152,6 +153,9 @@ def gated_delta_rule_chunked(q, k, v, g, beta, chunk_size: int = 64):
     updates with per-head log decay g, processed in chunks. The per-row
     forward-substitution loop building T = (I - tril(Kb K^T decay, -1))^{-1}
     is replaced by one batched triangular solve.
+
+    ``initial_state`` [B,H,Dk,Dv] / ``return_final_state`` support the
+    linear-attention context-parallel chunk relay (parallel/cp_linear.py).
     """
     q = _l2norm(q.transpose(1, 2).float())
     k = _l2norm(k.transpose(1, 2).float())
@@ -185,7 +189,8 @@ def gated_delta_rule_chunked(q, k, v, g, beta, chunk_size: int = 64):
     v_in = T @ v_beta
     k_cumdecay = T @ (k_beta * g.exp().unsqueeze(-1))
 
-    state = q.new_zeros(b, h, dk, dv)
+    state = (q.new_zeros(b, h, dk, dv) if initial_state is None
+             else initial_state.to(q.dtype))
     out = torch.empty_like(v_in)
     attn_mask = torch.ones(c, c, dtype=torch.bool, device=q.device).triu(1)
     for i in range(n):
@@ -198,7 +203,10 @@ def gated_delta_rule_chunked(q, k, v, g, beta, chunk_size: int = 64):
         state = (state * g[:, :, i, -1, None, None].exp() +
                  (k_i * (g[:, :, i, -1, None] - g[:, :, i]).exp()[..., None]
                   ).transpose(-1, -2) @ v_new)
-    return out.reshape(b, h, -1, dv)[:, :, :s].transpose(1, 2)
+    out = out.reshape(b, h, -1, dv)[:, :, :s].transpose(1, 2)
+    if return_final_state:
+        return out, state
+    return out
 
 
 class GatedDeltaNet(nn.Module):
