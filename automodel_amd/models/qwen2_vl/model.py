@@ -262,7 +262,8 @@ class VisionTransformer(nn.Module):
         freqs = (pos.unsqueeze(-1).float() * self.rot_inv_freq).flatten(1)
         emb = torch.cat([freqs, freqs], dim=-1)
         cos, sin = emb.cos(), emb.sin()
-        cu = torch.cat([torch.zeros(1, dtype=torch.long),
+        cu = torch.cat([torch.zeros(1, dtype=torch.long,
+                                    device=grid_thw.device),
                         grid_thw.prod(-1).cumsum(0)]).to(x.device)
         if c.variant == "v2_5":
             # window reorder (qwen2.5): merge-unit groups permuted so each
