@@ -49,6 +49,7 @@ def main():
                  n_activated_experts=2, moe_intermediate_size=32,
                  score_func="sigmoid", expert_bias=True, norm_topk_prob=True)))
     v32.init_weights(device=dev)
+    v32 = v32.to(torch.bfloat16)
     ids = torch.randint(0, 160, (1, 16), device=dev)
     out = v32(ids)
     assert torch.isfinite(out.float()).all(); results["deepseek_v32_sparse"] = "ok"
@@ -69,6 +70,7 @@ def main():
                  score_func="sigmoid", expert_bias=True, norm_topk_prob=True,
                  shared_expert_intermediate_size=16)))
     kl.init_weights(device=dev)
+    kl = kl.to(torch.bfloat16)
     ids = torch.randint(0, 120, (2, 24), device=dev)
     loss = kl(ids, labels=ids.clone())
     loss.backward()
