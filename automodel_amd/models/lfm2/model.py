@@ -243,3 +243,114 @@ class Lfm2ForCausalLM(nn.Module):
 
     def num_parameters(self) -> int:
         return sum(p.numel() for p in self.parameters())
+
+
+# ------------------------------------------------------------- LFM2-MoE
+@dataclass
+class Lfm2MoeConfig(Lfm2Config):
+    """LFM2-MoE: the same conv/attention hybrid with sigmoid-routed MoE
+    FFNs (aux-free expert bias, DeepSeek-style) after ``num_dense_layers``
+    dense SwiGLU layers. Dense layers use ``intermediate_size`` directly
+    (no 2/3 auto-adjust in the MoE variant)."""
+
+    num_experts: int = 0
+    num_experts_per_tok: int = 4
+    moe_intermediate_size: int = 1792
+    num_dense_layers: int = 2
+    use_expert_bias: bool = True
+    routed_scaling_factor: float = 1.0
+    norm_topk_prob: bool = True
+
+    @classmethod
+    def from_hf_config(cls, hf: Any) -> "Lfm2MoeConfig":
+        if hasattr(hf, "to_dict"):
+            hf = hf.to_dict()
+        base = Lfm2Config.from_hf_config(hf)
+        cfg = cls(**{f: getattr(base, f)
+                     for f in Lfm2Config.__dataclass_fields__})
+        g = hf.get
+        cfg.block_auto_adjust_ff_dim = False
+        cfg.num_experts = g("num_experts", 32)
+        cfg.num_experts_per_tok = g("num_experts_per_tok", 4)
+        cfg.moe_intermediate_size = g("moe_intermediate_size", 1792)
+        cfg.num_dense_layers = g("num_dense_layers", 2)
+        cfg.use_expert_bias = g("use_expert_bias", True)
+        cfg.routed_scaling_factor = g("routed_scaling_factor", 1.0)
+        cfg.norm_topk_prob = g("norm_topk_prob", True)
+        return cfg
+
+
+class Lfm2MoeStateDictAdapter:
+    """HF fused experts + block-level expert_bias <-> in-tree MoE keys."""
+
+    def from_hf(self, sd: dict) -> dict:
+        out = {}
+        for k, v in sd.items():
+            if k.endswith("feed_forward.expert_bias"):
+                out[k.replace("expert_bias",
+                              "gate.e_score_correction_bias")] = v
+            elif k.endswith("experts.gate_up_proj"):
+                gate, up = v.chunk(2, dim=1)
+                out[k.replace("gate_up_proj", "gate_proj")] = gate.contiguous()
+                out[k.replace("gate_up_proj", "up_proj")] = up.contiguous()
+            else:
+                out[k] = v
+        return out
+
+    def to_hf(self, sd: dict) -> dict:
+        out = {}
+        for k, v in sd.items():
+            if k.endswith("gate.e_score_correction_bias"):
+                out[k.replace("gate.e_score_correction_bias",
+                              "expert_bias")] = v
+            elif k.endswith("experts.gate_proj"):
+                up = sd[k.replace("gate_proj", "up_proj")]
+                out[k.replace("gate_proj", "gate_up_proj")] = \
+                    torch.cat([v, up], dim=1)
+            elif k.endswith("experts.up_proj"):
+                continue
+            else:
+                out[k] = v
+        return out
+
+
+class Lfm2MoeForCausalLM(Lfm2ForCausalLM):
+    hf_architectures = ("Lfm2MoeForCausalLM",)
+    config_class = Lfm2MoeConfig
+    state_dict_adapter = Lfm2MoeStateDictAdapter
+
+    @staticmethod
+    def config_from_hf(hf_cfg) -> Lfm2MoeConfig:
+        return Lfm2MoeConfig.from_hf_config(hf_cfg)
+
+    def __init__(self, config: Lfm2MoeConfig | dict, backend=None):
+        if isinstance(config, dict):
+            config = Lfm2MoeConfig(**config)
+        super().__init__(config, backend)
+        from automodel_amd.moe.config import MoEConfig
+        from automodel_amd.moe.layers import MoE
+
+        moe_cfg = MoEConfig(
+            n_routed_experts=config.num_experts,
+            n_shared_experts=0,
+            n_activated_experts=config.num_experts_per_tok,
+            moe_intermediate_size=config.moe_intermediate_size,
+            score_func="sigmoid", expert_bias=config.use_expert_bias,
+            norm_topk_prob=config.norm_topk_prob,
+            route_scale=config.routed_scaling_factor)
+        for i, layer in enumerate(self.model.layers):
+            if i >= config.num_dense_layers:
+                layer.feed_forward = MoE(config.hidden_size, moe_cfg)
+
+    @torch.no_grad()
+    def init_weights(self, device=None) -> None:
+        super().init_weights(device=device)
+        from automodel_amd.moe.layers import MoE
+
+        std = self.config.initializer_range
+        for m in self.modules():
+            if isinstance(m, MoE):
+                m.experts.init_weights(std)
+                nn.init.normal_(m.gate.weight, std=std)
+                if getattr(m.gate, "e_score_correction_bias", None) is not None:
+                    m.gate.e_score_correction_bias.zero_()

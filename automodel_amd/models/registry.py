@@ -76,10 +76,15 @@ def _ensure_builtin() -> None:
 
         for name in Qwen3NextForCausalLM.hf_architectures:
             _REGISTRY[name] = Qwen3NextForCausalLM
-        from automodel_amd.models.lfm2.model import Lfm2ForCausalLM
+        from automodel_amd.models.lfm2.model import (
+            Lfm2ForCausalLM,
+            Lfm2MoeForCausalLM,
+        )
 
         for name in Lfm2ForCausalLM.hf_architectures:
             _REGISTRY[name] = Lfm2ForCausalLM
+        for name in Lfm2MoeForCausalLM.hf_architectures:
+            _REGISTRY[name] = Lfm2MoeForCausalLM
         from automodel_amd.models.jamba.model import JambaForCausalLM
 
         for name in JambaForCausalLM.hf_architectures:

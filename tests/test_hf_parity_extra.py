@@ -2453,3 +2453,36 @@ def test_hf_logits_parity_glm4v_moe():
                  mm_token_type_ids=(ids == 3).int()).logits
         out = mine(ids, pixel_values=pixels, image_grid_thw=grid)
     torch.testing.assert_close(out, ref, atol=8e-4, rtol=8e-4)
+
+
+def test_hf_logits_parity_lfm2_moe():
+    """LFM2-MoE: the conv/attention hybrid with sigmoid + aux-free-bias
+    routed MoE FFNs after num_dense_layers dense layers."""
+    from automodel_amd.models.lfm2.model import Lfm2MoeConfig, Lfm2MoeForCausalLM
+
+    torch.manual_seed(64)
+    hf_cfg = transformers.Lfm2MoeConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=4, num_attention_heads=4, num_key_value_heads=2,
+        conv_L_cache=3,
+        layer_types=["conv", "full_attention", "conv", "full_attention"],
+        num_experts=8, num_experts_per_tok=2, moe_intermediate_size=32,
+        num_dense_layers=1, use_expert_bias=True, norm_topk_prob=True,
+        routed_scaling_factor=1.0, max_position_embeddings=64,
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.Lfm2MoeForCausalLM(hf_cfg).eval()
+    # exercise non-zero aux-free bias routing too
+    with torch.no_grad():
+        for layer in hf.model.layers[1:]:
+            layer.feed_forward.expert_bias.uniform_(-0.05, 0.05)
+    cfg = Lfm2MoeConfig.from_hf_config(hf_cfg.to_dict())
+    assert cfg.num_experts == 8 and cfg.num_dense_layers == 1
+    mine = Lfm2MoeForCausalLM(cfg).eval()
+    sd = mine.state_dict_adapter().from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 21))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits,
+                                   atol=5e-4, rtol=5e-4)
