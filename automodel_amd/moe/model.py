@@ -82,6 +82,28 @@ class MoEModelConfig(LlamaConfig):
             )
             base["qk_norm_full"] = True
             flavor = "qwen3_moe"
+        elif "Dots1" in arch:
+            # dots.llm1: DeepSeek-style sigmoid + aux-free-bias routing
+            # (group top-k, shared expert, dense-first layers) over a
+            # qwen3-style per-head qk-norm attention
+            n_shared = hf.get("n_shared_experts") or 0
+            moe = MoEConfig(
+                n_routed_experts=hf.get("n_routed_experts") or 64,
+                n_activated_experts=hf.get("num_experts_per_tok") or 8,
+                moe_intermediate_size=hf.get("moe_intermediate_size", 1408),
+                n_shared_experts=n_shared,
+                shared_expert_intermediate_size=(
+                    hf.get("moe_intermediate_size", 1408) * n_shared or None),
+                score_func="sigmoid",
+                expert_bias=True,
+                norm_topk_prob=hf.get("norm_topk_prob", False),
+                route_scale=hf.get("routed_scaling_factor", 1.0),
+                n_expert_groups=hf.get("n_group", 1) or 1,
+                n_limited_groups=hf.get("topk_group", 1) or 1,
+            )
+            base["qk_norm"] = True
+            return cls(**base, moe=moe, hf_flavor="qwen3_moe",
+                       first_k_dense=hf.get("first_k_dense_replace", 0))
         elif "Qwen2Moe" in arch:
             moe = MoEConfig(
                 n_routed_experts=hf.get("num_experts", 60),
@@ -146,7 +168,8 @@ class MoEDecoderLayer(nn.Module):
 class MoEForCausalLM(nn.Module):
     hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
                         "OlmoeForCausalLM", "GraniteMoeForCausalLM",
-                        "Ernie4_5_MoeForCausalLM", "MiniMaxM2ForCausalLM")
+                        "Ernie4_5_MoeForCausalLM", "MiniMaxM2ForCausalLM",
+                        "Dots1ForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod

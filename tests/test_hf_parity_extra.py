@@ -2486,3 +2486,39 @@ def test_hf_logits_parity_lfm2_moe():
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits,
                                    atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_dots1():
+    """dots.llm1: DeepSeek-style sigmoid + aux-free-bias routing (shared
+    expert, dense-first) on qwen3-style per-head qk-norm attention —
+    rides the generic MoE model."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+
+    torch.manual_seed(65)
+    hf_cfg = transformers.Dots1Config(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        n_routed_experts=8, n_shared_experts=1, num_experts_per_tok=2,
+        moe_intermediate_size=32, first_k_dense_replace=1,
+        norm_topk_prob=True, routed_scaling_factor=1.0,
+        n_group=1, topk_group=1, max_position_embeddings=64,
+        rope_theta=10000.0, attn_implementation="eager",
+        tie_word_embeddings=False)
+    hf = transformers.Dots1ForCausalLM(hf_cfg).eval()
+    with torch.no_grad():   # exercise non-zero aux-free routing bias
+        for layer in hf.model.layers[1:]:
+            layer.mlp.gate.e_score_correction_bias.uniform_(-0.05, 0.05)
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["Dots1ForCausalLM"]
+    cfg = MoEModelConfig.from_hf_config(d)
+    assert cfg.qk_norm and cfg.first_k_dense == 1
+    assert cfg.moe.score_func == "sigmoid" and cfg.moe.n_shared_experts == 1
+    mine = MoEForCausalLM(cfg).eval()
+    sd = mine.state_dict_adapter.from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 17))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits,
+                                   atol=5e-4, rtol=5e-4)
