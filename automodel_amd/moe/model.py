@@ -82,6 +82,18 @@ class MoEModelConfig(LlamaConfig):
             )
             base["qk_norm_full"] = True
             flavor = "qwen3_moe"
+        elif "FlexOlmo" in arch:
+            # OLMo-2 POST-norm layout + full-width q/k norms + softmax-topk
+            moe = MoEConfig(
+                n_routed_experts=hf.get("num_experts", 7),
+                n_activated_experts=hf.get("num_experts_per_tok", 5),
+                moe_intermediate_size=hf.get("intermediate_size", 11008),
+                aux_loss_coeff=hf.get("router_aux_loss_coef", 0.0),
+                norm_topk_prob=hf.get("norm_topk_prob", False),
+            )
+            base["qk_norm_full"] = True
+            base["olmo2_layout"] = True
+            return cls(**base, moe=moe, hf_flavor="qwen3_moe")
         elif "Dots1" in arch:
             # dots.llm1: DeepSeek-style sigmoid + aux-free-bias routing
             # (group top-k, shared expert, dense-first layers) over a
@@ -155,11 +167,22 @@ class MoEDecoderLayer(nn.Module):
         # leading dense layers (ernie/deepseek first_k_dense): plain SwiGLU
         self.mlp = (SharedExpert(cfg.hidden_size, cfg.intermediate_size)
                     if dense else MoE(cfg.hidden_size, cfg.moe))
-        self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
-        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        if cfg.olmo2_layout:   # FlexOlmo: POST-norm after each sublayer
+            self.post_attention_layernorm = RMSNorm(
+                cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+            self.post_feedforward_layernorm = RMSNorm(
+                cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
+        else:
+            self.input_layernorm = RMSNorm(cfg.hidden_size, cfg.rms_norm_eps,
+                                           backend.rms_norm)
+            self.post_attention_layernorm = RMSNorm(
+                cfg.hidden_size, cfg.rms_norm_eps, backend.rms_norm)
 
     def forward(self, x, cos, sin):
         r = self.cfg.residual_multiplier
+        if self.cfg.olmo2_layout:
+            x = x + self.post_attention_layernorm(self.self_attn(x, cos, sin)) * r
+            return x + self.post_feedforward_layernorm(self.mlp(x)) * r
         x = x + self.self_attn(self.input_layernorm(x), cos, sin) * r
         x = x + self.mlp(self.post_attention_layernorm(x)) * r
         return x
@@ -169,7 +192,7 @@ class MoEForCausalLM(nn.Module):
     hf_architectures = ("Qwen3MoeForCausalLM", "Qwen2MoeForCausalLM", "MixtralForCausalLM",
                         "OlmoeForCausalLM", "GraniteMoeForCausalLM",
                         "Ernie4_5_MoeForCausalLM", "MiniMaxM2ForCausalLM",
-                        "Dots1ForCausalLM")
+                        "Dots1ForCausalLM", "FlexOlmoForCausalLM")
     config_class = MoEModelConfig
 
     @staticmethod

@@ -2522,3 +2522,32 @@ def test_hf_logits_parity_dots1():
     with torch.no_grad():
         torch.testing.assert_close(mine(ids), hf(ids).logits,
                                    atol=5e-4, rtol=5e-4)
+
+
+def test_hf_logits_parity_flex_olmo():
+    """FlexOlmo: OLMo-2 POST-norm layout (post_attention/post_feedforward
+    norms, full-width q/k norms) with softmax-topk MoE FFNs."""
+    from automodel_amd.moe.model import MoEForCausalLM, MoEModelConfig
+
+    torch.manual_seed(66)
+    hf_cfg = transformers.FlexOlmoConfig(
+        vocab_size=200, hidden_size=64, intermediate_size=96,
+        num_hidden_layers=3, num_attention_heads=4, num_key_value_heads=2,
+        num_experts=6, num_experts_per_tok=2, norm_topk_prob=False,
+        max_position_embeddings=64, rope_theta=10000.0,
+        pad_token_id=0, eos_token_id=1,
+        attn_implementation="eager", tie_word_embeddings=False)
+    hf = transformers.FlexOlmoForCausalLM(hf_cfg).eval()
+    d = hf_cfg.to_dict()
+    d["architectures"] = ["FlexOlmoForCausalLM"]
+    cfg = MoEModelConfig.from_hf_config(d)
+    assert cfg.olmo2_layout and cfg.qk_norm_full
+    mine = MoEForCausalLM(cfg).eval()
+    sd = mine.state_dict_adapter.from_hf(hf.state_dict())
+    missing, unexpected = mine.load_state_dict(sd, strict=False)
+    assert not unexpected, unexpected
+    assert all("rope" in k for k in missing), missing
+    ids = torch.randint(0, 200, (2, 19))
+    with torch.no_grad():
+        torch.testing.assert_close(mine(ids), hf(ids).logits,
+                                   atol=5e-4, rtol=5e-4)
